@@ -1,0 +1,150 @@
+"""PandasDataFrame: the local-CPU workhorse frame.
+
+Reference parity: ``fugue/dataframe/pandas_dataframe.py``.
+"""
+from typing import Any, Dict, Iterable, List, Optional
+
+import pandas as pd
+import pyarrow as pa
+
+from fugue_amd.dataframe.dataframe import (
+    DataFrame,
+    LocalBoundedDataFrame,
+)
+from fugue_amd.exceptions import (
+    FugueDataFrameEmptyError,
+    FugueDataFrameInitError,
+    FugueDataFrameOperationError,
+)
+from fugue_amd.schema import Schema
+from fugue_amd.utils.pandas_like import cast_pandas, pandas_to_schema
+
+
+class PandasDataFrame(LocalBoundedDataFrame):
+    def __init__(
+        self,
+        df: Any = None,
+        schema: Any = None,
+        pandas_df_wrapper: bool = False,
+    ):
+        try:
+            if df is None:
+                schema = Schema(schema).assert_not_empty()
+                df = schema.create_empty_pandas()
+                pdf = df
+            elif isinstance(df, pd.DataFrame):
+                if schema is None:
+                    pdf = df.reset_index(drop=True)
+                    schema = Schema(pandas_to_schema(pdf))
+                elif pandas_df_wrapper:
+                    pdf = df
+                    schema = Schema(schema).assert_not_empty()
+                else:
+                    schema = Schema(schema).assert_not_empty()
+                    pdf = cast_pandas(df.reset_index(drop=True), schema)
+            elif isinstance(df, pd.Series):
+                pdf = df.to_frame().reset_index(drop=True)
+                schema = (
+                    Schema(pandas_to_schema(pdf))
+                    if schema is None
+                    else Schema(schema).assert_not_empty()
+                )
+                pdf = cast_pandas(pdf, schema)
+            elif isinstance(df, (list, tuple)) or isinstance(df, Iterable):
+                schema = Schema(schema).assert_not_empty()
+                rows = [{c: row[i] for i, c in enumerate(schema.names)} for row in df]
+                pdf = pa.Table.from_pylist(rows, schema=schema.pa_schema).to_pandas()
+            else:
+                raise ValueError(f"{type(df)} is incompatible with PandasDataFrame")
+        except FugueDataFrameInitError:
+            raise
+        except Exception as e:
+            raise FugueDataFrameInitError(str(e)) from e
+        self._native: pd.DataFrame = pdf
+        super().__init__(schema)
+
+    @property
+    def native(self) -> pd.DataFrame:
+        return self._native
+
+    def native_as_df(self) -> pd.DataFrame:
+        return self._native
+
+    @property
+    def empty(self) -> bool:
+        return len(self._native) == 0
+
+    def count(self) -> int:
+        return len(self._native)
+
+    def peek_array(self) -> List[Any]:
+        if self.empty:
+            raise FugueDataFrameEmptyError("dataframe is empty")
+        return list(self._arrow_slice(0, 1).to_pylist()[0].values())
+
+    def as_pandas(self) -> pd.DataFrame:
+        return self._native
+
+    def as_arrow(self, type_safe: bool = False) -> pa.Table:
+        return pa.Table.from_pandas(
+            self._native.reset_index(drop=True),
+            schema=self.schema.pa_schema,
+            preserve_index=False,
+            safe=type_safe,
+        )
+
+    def _arrow_slice(self, start: int, length: int) -> pa.Table:
+        return pa.Table.from_pandas(
+            self._native.iloc[start : start + length].reset_index(drop=True),
+            schema=self.schema.pa_schema,
+            preserve_index=False,
+        )
+
+    def as_array(
+        self, columns: Optional[List[str]] = None, type_safe: bool = False
+    ) -> List[Any]:
+        tbl = self.as_arrow()
+        if columns is not None:
+            tbl = tbl.select(columns)
+        return [list(d.values()) for d in tbl.to_pylist()]
+
+    def as_array_iterable(
+        self, columns: Optional[List[str]] = None, type_safe: bool = False
+    ) -> Iterable[Any]:
+        yield from self.as_array(columns, type_safe=type_safe)
+
+    def _drop_cols(self, cols: List[str]) -> DataFrame:
+        schema = self.schema.exclude(cols)
+        return PandasDataFrame(
+            self._native[schema.names], schema, pandas_df_wrapper=True
+        )
+
+    def _select_cols(self, cols: List[Any]) -> DataFrame:
+        schema = self.schema.extract(cols)
+        return PandasDataFrame(
+            self._native[schema.names], schema, pandas_df_wrapper=True
+        )
+
+    def rename(self, columns: Dict[str, str]) -> DataFrame:
+        try:
+            schema = self.schema.rename(columns)
+        except Exception as e:
+            raise FugueDataFrameOperationError(str(e)) from e
+        return PandasDataFrame(
+            self._native.rename(columns=columns), schema, pandas_df_wrapper=True
+        )
+
+    def alter_columns(self, columns: Any) -> DataFrame:
+        schema = self._get_altered_schema(columns)
+        if schema == self.schema:
+            return self
+        return PandasDataFrame(self._native, schema)
+
+    def head(
+        self, n: int, columns: Optional[List[str]] = None
+    ) -> LocalBoundedDataFrame:
+        sub = self._native if columns is None else self._native[columns]
+        schema = self.schema if columns is None else self.schema.extract(columns)
+        return PandasDataFrame(
+            sub.head(n).reset_index(drop=True), schema, pandas_df_wrapper=True
+        )

@@ -1,0 +1,62 @@
+"""Deterministic uuid hashing of nested python values.
+
+Replaces ``triad.utils.hash.to_uuid`` (reference usage: spec-UUID
+determinism for workflow tasks, ``fugue/workflow/_tasks.py:85``).  The hash
+is stable across processes and runs: it never uses object ids.  Objects may
+customize by defining ``__uuid__``.
+"""
+import uuid
+from typing import Any, Iterable
+
+_NAMESPACE = uuid.UUID("6b1f84e2-13a1-4a73-9a33-0d67b6f0a1c5")
+
+
+def _feed(obj: Any, parts: list) -> None:
+    if obj is None:
+        parts.append("\0N")
+    elif hasattr(obj, "__uuid__"):
+        parts.append("\0U" + str(obj.__uuid__()))
+    elif isinstance(obj, bool):
+        parts.append("\0b" + str(obj))
+    elif isinstance(obj, (int, float, complex)):
+        parts.append("\0n" + repr(obj))
+    elif isinstance(obj, str):
+        parts.append("\0s" + obj)
+    elif isinstance(obj, bytes):
+        parts.append("\0y" + obj.hex())
+    elif isinstance(obj, dict):
+        parts.append("\0d{")
+        for k in obj:  # preserve insertion order (it is part of identity)
+            _feed(k, parts)
+            _feed(obj[k], parts)
+        parts.append("}")
+    elif isinstance(obj, (list, tuple)):
+        parts.append("\0l[")
+        for x in obj:
+            _feed(x, parts)
+        parts.append("]")
+    elif isinstance(obj, set):
+        parts.append("\0S[")
+        for x in sorted(str(i) for i in obj):
+            parts.append(x + ",")
+        parts.append("]")
+    elif isinstance(obj, Iterable):
+        parts.append("\0l[")
+        for x in obj:
+            _feed(x, parts)
+        parts.append("]")
+    elif callable(obj):
+        name = getattr(obj, "__qualname__", getattr(obj, "__name__", None))
+        mod = getattr(obj, "__module__", "")
+        if name is None:
+            name = type(obj).__qualname__
+        parts.append("\0f" + mod + "." + name)
+    else:
+        parts.append("\0o" + repr(obj))
+
+
+def to_uuid(*args: Any) -> str:
+    parts: list = []
+    for a in args:
+        _feed(a, parts)
+    return str(uuid.uuid5(_NAMESPACE, "".join(parts)))

@@ -1,0 +1,132 @@
+"""File IO: load/save parquet/csv/json with glob support.
+
+Reference behavior: ``fugue/_utils/io.py:107,126`` (format inference from
+extension, glob loading, column pruning on load).  Implementation is new,
+on pyarrow datasets + pandas.
+"""
+import glob as _glob
+import os
+from typing import Any, List, Optional, Tuple, Union
+
+import pandas as pd
+import pyarrow as pa
+import pyarrow.parquet as pq
+
+from fugue_amd.schema import Schema
+
+_FORMATS = {".parquet": "parquet", ".csv": "csv", ".json": "json"}
+
+
+def infer_format(path: str, fmt: Optional[str] = None) -> str:
+    if fmt is not None and fmt != "":
+        return fmt
+    base = path
+    if "*" in base:
+        base = base.replace("*", "x")
+    _, ext = os.path.splitext(base)
+    if ext in _FORMATS:
+        return _FORMATS[ext]
+    raise ValueError(f"can't infer file format of {path}")
+
+
+def _expand(path: Union[str, List[str]]) -> List[str]:
+    paths = [path] if isinstance(path, str) else list(path)
+    res: List[str] = []
+    for p in paths:
+        if "*" in p:
+            matches = sorted(_glob.glob(p))
+            if len(matches) == 0:
+                raise FileNotFoundError(f"no files match {p}")
+            res.extend(matches)
+        else:
+            res.append(p)
+    return res
+
+
+def load_df(
+    path: Union[str, List[str]],
+    format_hint: Optional[str] = None,
+    columns: Optional[Any] = None,
+    **kwargs: Any,
+) -> Tuple[pd.DataFrame, Optional[Schema]]:
+    """Load file(s) into a pandas frame. Returns (df, schema or None)."""
+    first = path[0] if isinstance(path, list) else path
+    fmt = infer_format(first, format_hint)
+    files = _expand(path)
+    col_names: Optional[List[str]] = None
+    schema: Optional[Schema] = None
+    if columns is not None:
+        if isinstance(columns, list):
+            col_names = columns
+        else:
+            schema = Schema(columns)
+            col_names = schema.names
+    if fmt == "parquet":
+        tables = [pq.read_table(f, columns=col_names) for f in files]
+        table = pa.concat_tables(tables) if len(tables) > 1 else tables[0]
+        df = table.to_pandas()
+        if schema is None:
+            schema = Schema(table.schema)
+    elif fmt == "csv":
+        header = kwargs.pop("header", True)
+        infer_schema = kwargs.pop("infer_schema", False)
+        dfs = []
+        for f in files:
+            if header:
+                d = pd.read_csv(f, dtype=None if infer_schema else str, **kwargs)
+            else:
+                if col_names is None:
+                    raise ValueError("columns required to load headerless csv")
+                d = pd.read_csv(
+                    f, header=None, names=col_names,
+                    dtype=None if infer_schema else str, **kwargs
+                )
+            dfs.append(d)
+        df = pd.concat(dfs, ignore_index=True) if len(dfs) > 1 else dfs[0]
+        if col_names is not None:
+            df = df[col_names]
+    elif fmt == "json":
+        dfs = [pd.read_json(f, orient="records", lines=True, **kwargs) for f in files]
+        df = pd.concat(dfs, ignore_index=True) if len(dfs) > 1 else dfs[0]
+        if col_names is not None:
+            df = df[col_names]
+    else:
+        raise ValueError(f"unsupported format {fmt}")
+    if schema is not None and fmt != "parquet":
+        from fugue_amd.utils.pandas_like import cast_pandas
+
+        df = cast_pandas(df, schema)
+    return df.reset_index(drop=True), schema
+
+
+def save_df(
+    df: pd.DataFrame,
+    schema: Schema,
+    path: str,
+    format_hint: Optional[str] = None,
+    mode: str = "overwrite",
+    **kwargs: Any,
+) -> None:
+    fmt = infer_format(path, format_hint)
+    if os.path.exists(path):
+        if mode == "error":
+            raise FileExistsError(path)
+        if mode == "overwrite":
+            if os.path.isdir(path):
+                import shutil
+
+                shutil.rmtree(path)
+            else:
+                os.remove(path)
+    parent = os.path.dirname(os.path.abspath(path))
+    os.makedirs(parent, exist_ok=True)
+    if fmt == "parquet":
+        table = pa.Table.from_pandas(df, schema=schema.pa_schema, preserve_index=False)
+        pq.write_table(table, path)
+    elif fmt == "csv":
+        header = kwargs.pop("header", False)
+        df.to_csv(path, index=False, header=header, **kwargs)
+    elif fmt == "json":
+        df.to_json(path, orient="records", lines=True, **kwargs)
+    else:
+        raise ValueError(f"unsupported format {fmt}")

@@ -1,0 +1,47 @@
+import pandas as pd
+import pytest
+
+from fugue_amd.utils.hash import to_uuid
+from fugue_amd.utils.params import ParamDict
+from fugue_amd.utils.convert import to_function, to_instance, to_type
+from fugue_amd.utils.interfaceless import parse_output_schema_from_comment
+
+
+def test_to_uuid():
+    assert to_uuid(1) == to_uuid(1)
+    assert to_uuid(1) != to_uuid("1")
+    assert to_uuid([1, 2]) == to_uuid([1, 2])
+    assert to_uuid([1, 2]) != to_uuid([2, 1])
+    assert to_uuid(dict(a=1, b=[1, 2])) == to_uuid(dict(a=1, b=[1, 2]))
+    assert to_uuid(None) == to_uuid(None)
+    assert to_uuid(to_uuid) == to_uuid(to_uuid)
+
+
+def test_param_dict():
+    p = ParamDict(dict(a=1, b="2", c="true"))
+    assert p.get("a", 0) == 1
+    assert p.get("b", 0) == 2
+    assert p.get("c", False) is True
+    assert p.get("missing", "x") == "x"
+    with pytest.raises(KeyError):
+        p.get_or_throw("missing")
+    assert p.get_or_none("missing") is None
+    assert p.get_or_throw("a", int) == 1
+
+
+def test_convert():
+    assert to_type("fugue_amd.schema.Schema") is not None
+    f = to_function("fugue_amd.utils.hash.to_uuid")
+    assert f(1) == to_uuid(1)
+
+
+# schema: a:int,b:str
+def _schema_fn(df: pd.DataFrame) -> pd.DataFrame:
+    return df
+
+
+def test_comment_schema():
+    assert parse_output_schema_from_comment(_schema_fn) == "a:int,b:str"
+    def no_comment(df):
+        return df
+    assert parse_output_schema_from_comment(no_comment) is None
