@@ -423,6 +423,19 @@ class _DataFramesParam(AnnotatedParam):
         return df
 
 
+class ExecutionEngineParam(AnnotatedParam):
+    """An ExecutionEngine (or subclass) annotation — the engine instance is
+    injected by the caller (code ``e``)."""
+
+    code = "e"
+
+    @staticmethod
+    def matches(anno: Any) -> bool:
+        from fugue_amd.execution.execution_engine import ExecutionEngine
+
+        return isinstance(anno, type) and issubclass(anno, ExecutionEngine)
+
+
 class _CallableParam(AnnotatedParam):
     code = "F"
 
@@ -493,6 +506,7 @@ for _c in [
     _PyArrowTableParam,
     _IterableArrowParam,
     _DataFramesParam,
+    ExecutionEngineParam,
     _OptionalCallableParam,
     _CallableParam,
 ]:

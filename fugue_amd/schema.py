@@ -556,6 +556,98 @@ class Schema:
             ]
         )
 
+    def transform(self, *args: Any) -> "Schema":
+        """Schema transformation expressions (triad parity), e.g.::
+
+            schema.transform("*")            # identity
+            schema.transform("*,c:int")      # append column
+            schema.transform("*-b")          # remove column b
+            schema.transform("*~b,c")        # remove b and c if they exist
+            schema.transform(lambda s: ...)  # callable form
+        """
+        result_fields: List[pa.Field] = []
+        for a in args:
+            if callable(a):
+                result_fields.extend(Schema(a(self)).fields)
+                continue
+            if isinstance(a, Schema):
+                result_fields.extend(a.fields)
+                continue
+            if not isinstance(a, str):
+                result_fields.extend(self._parse(a))
+                continue
+            for token in self._split_top_level(a):
+                token = token.strip()
+                if token == "":
+                    continue
+                if token.startswith("*"):
+                    rest = token[1:]
+                    removed: List[str] = []
+                    soft_removed: List[str] = []
+                    while rest != "":
+                        if rest[0] == "-":
+                            rest = rest[1:]
+                            name, rest = self._take_name(rest)
+                            removed.append(name)
+                        elif rest[0] == "~":
+                            rest = rest[1:]
+                            name, rest = self._take_name(rest)
+                            soft_removed.append(name)
+                        else:
+                            raise SchemaError(
+                                f"invalid transform expression {token!r}"
+                            )
+                    for r in removed:
+                        if r not in self._index:
+                            raise SchemaError(f"{r} not in schema {self}")
+                    drop = set(removed) | {
+                        s for s in soft_removed if s in self._index
+                    }
+                    result_fields.extend(
+                        f for f in self.fields if f.name not in drop
+                    )
+                elif token.startswith("-") or token.startswith("~"):
+                    hard = token[0] == "-"
+                    names = [
+                        t.strip() for t in token[1:].split("+") if t.strip() != ""
+                    ]
+                    keep = list(result_fields)
+                    for n in names:
+                        if hard and n not in [f.name for f in keep]:
+                            raise SchemaError(f"{n} not in schema")
+                        keep = [f for f in keep if f.name != n]
+                    result_fields = keep
+                else:
+                    result_fields.extend(list(expression_to_schema(token)))
+        return Schema(result_fields)
+
+    @staticmethod
+    def _take_name(s: str) -> Tuple[str, str]:
+        i = 0
+        while i < len(s) and (s[i].isalnum() or s[i] == "_"):
+            i += 1
+        if i == 0:
+            raise SchemaError(f"expected name in {s!r}")
+        return s[:i], s[i:]
+
+    @staticmethod
+    def _split_top_level(s: str) -> List[str]:
+        parts: List[str] = []
+        depth = 0
+        cur = ""
+        for ch in s:
+            if ch in "[{<(":
+                depth += 1
+            elif ch in "]}>)":
+                depth -= 1
+            if ch == "," and depth == 0:
+                parts.append(cur)
+                cur = ""
+            else:
+                cur += ch
+        parts.append(cur)
+        return parts
+
     def _as_keys(self, obj: Any) -> List[Any]:
         if obj is None:
             return []
