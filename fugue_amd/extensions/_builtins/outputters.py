@@ -78,6 +78,7 @@ class RunOutputTransformer(Outputter):
             tf._has_rpc_client = True
         ie = self.params.get("ignore_errors", [])
         self._ignore_errors = [to_type(x, Exception) for x in ie]
+        tf.validate_on_compile()
         tf.validate_on_runtime(df)
         if isinstance(tf, Transformer):
             self.transform(df, tf)

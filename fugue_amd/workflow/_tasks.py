@@ -41,8 +41,7 @@ class FugueTask:
         self._partition_spec = partition_spec or PartitionSpec()
         self._checkpoint = Checkpoint()
         self._broadcast = False
-        self._yield_name: Optional[str] = None
-        self._yield_obj: Optional[Any] = None
+        self._yields: List[Any] = []
         self._result: Optional[DataFrame] = None
         self._executed = False
         self._traceback = None
@@ -94,8 +93,7 @@ class FugueTask:
         return self
 
     def set_yield(self, name: str, obj: Any) -> None:
-        self._yield_name = name
-        self._yield_obj = obj
+        self._yields.append(obj)
 
     @property
     def executed(self) -> bool:
@@ -132,8 +130,7 @@ class FugueTask:
             df = ctx.execution_engine.broadcast(df)
         self._result = df
         self._executed = True
-        if self._yield_name is not None:
-            obj = self._yield_obj
+        for obj in self._yields:
             if isinstance(obj, PhysicalYielded):
                 if obj.storage_type == "file":
                     path = ctx.checkpoint_path.get_temp_file(self.__uuid__(), True)

@@ -64,6 +64,13 @@ def transform(
         return result
     if isinstance(df, DataFrame):
         return result
+    import pandas as pd
+    import pyarrow as pa
+
+    if isinstance(df, pd.DataFrame):
+        return result.as_pandas()
+    if isinstance(df, pa.Table):
+        return result.as_arrow()
     return result.native_as_df()
 
 

@@ -1,0 +1,212 @@
+import os
+import tempfile
+from typing import Any, Callable, Dict, Iterable, List
+
+import pandas as pd
+import pytest
+
+import fugue_amd.api as fa
+from fugue_amd import ArrayDataFrame, DataFrame, LocalDataFrame, PandasDataFrame
+from fugue_amd.column.expressions import col
+from fugue_amd.column import functions as f
+from fugue_amd.constants import FUGUE_CONF_WORKFLOW_CHECKPOINT_PATH
+from fugue_amd.exceptions import (
+    FugueInterfacelessError,
+    FugueWorkflowCompileValidationError,
+)
+from fugue_amd.workflow import FugueWorkflow, transform, out_transform
+
+
+def test_basic_dag_run():
+    dag = FugueWorkflow()
+    a = dag.df(pd.DataFrame(dict(x=[1, 2, 3])))
+    b = a.filter(col("x") > 1)
+    b.yield_dataframe_as("r")
+    res = dag.run()
+    assert res["r"].result.as_array() == [[2], [3]]
+
+
+def test_dag_determinism():
+    def build():
+        dag = FugueWorkflow()
+        a = dag.df([[1, "a"]], "x:long,y:str")
+        b = a.transform(_double, schema="*")
+        b.yield_dataframe_as("r")
+        return [t.__uuid__() for t in dag._task_order]
+
+    assert build() == build()
+
+
+# schema: *
+def _double(df: pd.DataFrame) -> pd.DataFrame:
+    df["x"] = df["x"] * 2
+    return df
+
+
+def test_transform_styles():
+    pdf = pd.DataFrame(dict(x=[1, 2], g=["a", "b"]))
+
+    # plain function with schema comment
+    res = transform(pdf, _double)
+    assert res["x"].tolist() == [2, 4]
+
+    # plain function with explicit schema, list output
+    def to_arr(df: List[List[Any]]) -> List[List[Any]]:
+        return [[r[0] * 10] for r in df]
+
+    res2 = transform(pdf[["x"]], to_arr, schema="x:long")
+    assert res2["x"].tolist() == [10, 20]
+
+    # iterable of dicts
+    def gen(df: Iterable[Dict[str, Any]]) -> Iterable[Dict[str, Any]]:
+        for row in df:
+            row["x"] += 1
+            yield row
+
+    res3 = transform(pdf, gen, schema="*")
+    assert res3["x"].tolist() == [2, 3]
+
+    # transformer class instance via decorator
+    from fugue_amd.extensions import transformer
+
+    @transformer("*,z:long")
+    def with_z(df: pd.DataFrame) -> pd.DataFrame:
+        df["z"] = 1
+        return df
+
+    res4 = transform(pdf, with_z)
+    assert "z" in res4.columns
+
+
+def test_transform_partition():
+    pdf = pd.DataFrame(dict(g=["a", "a", "b"], v=[3, 1, 2]))
+
+    # schema: g:str,first_v:long
+    def first_v(df: pd.DataFrame) -> pd.DataFrame:
+        return pd.DataFrame(dict(g=[df["g"].iloc[0]], first_v=[df["v"].iloc[0]]))
+
+    res = transform(
+        pdf, first_v, partition=dict(by=["g"], presort="v")
+    )
+    assert sorted(res.values.tolist()) == [["a", 1], ["b", 2]]
+
+
+def test_out_transform_and_callback():
+    collected = []
+
+    def cb(x: int) -> None:
+        collected.append(x)
+
+    def sink(df: pd.DataFrame, callback: Callable) -> None:
+        callback(len(df))
+
+    out_transform(pd.DataFrame(dict(a=[1, 2])), sink, callback=cb)
+    assert collected == [2]
+
+
+def test_ignore_errors():
+    def bad(df: pd.DataFrame) -> pd.DataFrame:
+        if df["g"].iloc[0] == "a":
+            raise ValueError("boom")
+        return df
+
+    pdf = pd.DataFrame(dict(g=["a", "b"], v=[1, 2]))
+    res = transform(
+        pdf,
+        bad,
+        schema="*",
+        partition=dict(by=["g"]),
+        ignore_errors=[ValueError],
+    )
+    assert res.values.tolist() == [["b", 2]]
+
+
+def test_workflow_join_setops():
+    dag = FugueWorkflow()
+    a = dag.df([[1, "x"], [2, "y"]], "k:long,a:str")
+    b = dag.df([[2, 5.0]], "k:long,b:double")
+    j = a.inner_join(b)
+    j.yield_dataframe_as("j")
+    u = a.union(a, distinct=False)
+    u.yield_dataframe_as("u")
+    res = dag.run()
+    assert res["j"].result.as_array() == [[2, "y", 5.0]]
+    assert res["u"].result.count() == 4
+
+
+def test_checkpoints_and_persist():
+    with tempfile.TemporaryDirectory() as tmp:
+        dag = FugueWorkflow()
+        a = dag.df([[1]], "x:long").persist()
+        b = a.strong_checkpoint()
+        b.yield_dataframe_as("r")
+        res = dag.run(None, {FUGUE_CONF_WORKFLOW_CHECKPOINT_PATH: tmp})
+        assert res["r"].result.as_array() == [[1]]
+
+
+def test_deterministic_checkpoint_reuse():
+    with tempfile.TemporaryDirectory() as tmp:
+        conf = {FUGUE_CONF_WORKFLOW_CHECKPOINT_PATH: tmp}
+
+        def run_once():
+            dag = FugueWorkflow()
+            a = dag.df([[1]], "x:long")
+            b = a.transform(_double_x, schema="*").deterministic_checkpoint()
+            b.yield_dataframe_as("r")
+            return dag.run(None, conf)["r"].result.as_array()
+
+        assert run_once() == [[2]]
+        files = os.listdir(tmp)
+        assert run_once() == [[2]]
+        assert os.listdir(tmp) == files  # second run reused the checkpoint
+
+
+def _double_x(df: pd.DataFrame) -> pd.DataFrame:
+    df["x"] = df["x"] * 2
+    return df
+
+
+def test_validation_rules():
+    # partitionby_has: g
+    def needs_g(df: pd.DataFrame) -> pd.DataFrame:
+        return df
+
+    with pytest.raises(FugueWorkflowCompileValidationError):
+        transform(
+            pd.DataFrame(dict(g=[1], v=[2])),
+            needs_g,
+            schema="*",
+        )
+
+
+def test_zip_comap_workflow():
+    dag = FugueWorkflow()
+    a = dag.df([[1, "a"], [2, "b"]], "k:long,x:str")
+    b = dag.df([[1, 1.0], [1, 2.0]], "k:long,y:double")
+    z = dag.zip(a, b)
+
+    def merge_counts(df1: pd.DataFrame, df2: pd.DataFrame) -> List[List[Any]]:
+        return [[len(df1), len(df2)]]
+
+    r = z.transform(merge_counts, schema="n1:long,n2:long")
+    r.yield_dataframe_as("r")
+    res = dag.run()
+    assert res["r"].result.as_array() == [[1, 2]]
+
+
+def test_workflow_parallelism():
+    dag = FugueWorkflow()
+    a = dag.df([[1]], "x:long")
+    for i in range(5):
+        a.transform(_double_x, schema="*").yield_dataframe_as(f"r{i}")
+    res = dag.run(None, {"fugue.workflow.concurrency": 4})
+    for i in range(5):
+        assert res[f"r{i}"].result.as_array() == [[2]]
+
+
+def test_runtime_exception_traceback():
+    def bad(df: pd.DataFrame) -> pd.DataFrame:
+        raise RuntimeError("inner failure")
+
+    with pytest.raises(RuntimeError, match="inner failure"):
+        transform(pd.DataFrame(dict(a=[1])), bad, schema="*")
