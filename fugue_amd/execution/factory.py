@@ -120,6 +120,11 @@ def make_execution_engine(
     if isinstance(engine, str):
         with _LOCK:
             func = _ENGINE_REGISTRY.get(engine)
+        if func is None and engine in ("hip", "mi355x", "gpu"):
+            import fugue_amd.hip  # noqa: F401  (registers the engine)
+
+            with _LOCK:
+                func = _ENGINE_REGISTRY.get(engine)
         if func is None:
             raise ValueError(f"execution engine {engine!r} is not registered")
         return func(merged)

@@ -1,0 +1,229 @@
+// Torch-extension bindings for the CDNA4 relational kernels
+// (fugue_amd/hip/csrc/relational.hip).
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <limits>
+#include <vector>
+
+extern "C" {
+void launch_hash_col_i64(const int64_t*, const bool*, uint64_t*, int64_t, int,
+                         hipStream_t);
+void launch_hash_col_i32(const int32_t*, const bool*, uint64_t*, int64_t, int,
+                         hipStream_t);
+void launch_hash_col_f64(const double*, const bool*, uint64_t*, int64_t, int,
+                         hipStream_t);
+void launch_hash_col_f32(const float*, const bool*, uint64_t*, int64_t, int,
+                         hipStream_t);
+void launch_hash_col_i8(const int8_t*, const bool*, uint64_t*, int64_t, int,
+                        hipStream_t);
+void launch_bucket_of(const uint64_t*, int32_t*, int64_t, int32_t,
+                      hipStream_t);
+void launch_histogram(const int32_t*, int64_t*, int64_t, int32_t, hipStream_t);
+void launch_scatter(const int32_t*, int64_t*, int64_t*, int64_t, hipStream_t);
+void launch_gb_aggregate(const int64_t*, const double*, const bool*,
+                         const int32_t*, int, int64_t, int64_t*, double*,
+                         int64_t*, int64_t, int, hipStream_t);
+void launch_join_build(const int64_t*, int64_t, int32_t*, int32_t*, int64_t,
+                       hipStream_t);
+void launch_join_count(const int64_t*, int64_t, const int64_t*, const int32_t*,
+                       const int32_t*, int64_t, int32_t*, hipStream_t);
+void launch_join_emit(const int64_t*, int64_t, const int64_t*, const int32_t*,
+                      const int32_t*, int64_t, const int64_t*, int64_t*,
+                      int64_t*, int, hipStream_t);
+void launch_join_mark_build(const int64_t*, int64_t, const int64_t*,
+                            const int32_t*, const int32_t*, int64_t, bool*,
+                            hipStream_t);
+}
+
+namespace {
+
+hipStream_t current_stream() {
+  return (hipStream_t)c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_gpu(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on the GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+const bool* opt_valid_ptr(const c10::optional<at::Tensor>& v) {
+  if (!v.has_value()) return nullptr;
+  return v->data_ptr<bool>();
+}
+
+}  // namespace
+
+// Combine one column into the running row hash (out int64 viewed as u64).
+void hash_column(at::Tensor data, c10::optional<at::Tensor> valid,
+                 at::Tensor out, bool is_first) {
+  check_gpu(data, "data");
+  check_gpu(out, "out");
+  int64_t n = data.numel();
+  TORCH_CHECK(out.numel() == n, "out size mismatch");
+  auto stream = current_stream();
+  uint64_t* optr = reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>());
+  const bool* vptr = opt_valid_ptr(valid);
+  switch (data.scalar_type()) {
+    case at::kLong:
+      launch_hash_col_i64(data.data_ptr<int64_t>(), vptr, optr, n, is_first,
+                          stream);
+      break;
+    case at::kInt:
+      launch_hash_col_i32(data.data_ptr<int32_t>(), vptr, optr, n, is_first,
+                          stream);
+      break;
+    case at::kDouble:
+      launch_hash_col_f64(data.data_ptr<double>(), vptr, optr, n, is_first,
+                          stream);
+      break;
+    case at::kFloat:
+      launch_hash_col_f32(data.data_ptr<float>(), vptr, optr, n, is_first,
+                          stream);
+      break;
+    case at::kChar:
+    case at::kBool:
+      launch_hash_col_i8((const int8_t*)data.data_ptr(), vptr, optr, n,
+                         is_first, stream);
+      break;
+    default:
+      TORCH_CHECK(false, "unsupported dtype for hash_column");
+  }
+}
+
+at::Tensor bucket_of(at::Tensor hashes, int64_t num_buckets) {
+  check_gpu(hashes, "hashes");
+  int64_t n = hashes.numel();
+  auto buckets = at::empty({n}, hashes.options().dtype(at::kInt));
+  launch_bucket_of(
+      reinterpret_cast<const uint64_t*>(hashes.data_ptr<int64_t>()),
+      buckets.data_ptr<int32_t>(), n, (int32_t)num_buckets, current_stream());
+  return buckets;
+}
+
+at::Tensor bucket_histogram(at::Tensor buckets, int64_t num_buckets) {
+  check_gpu(buckets, "buckets");
+  auto hist = at::zeros({num_buckets}, buckets.options().dtype(at::kLong));
+  launch_histogram(buckets.data_ptr<int32_t>(), hist.data_ptr<int64_t>(),
+                   buckets.numel(), (int32_t)num_buckets, current_stream());
+  return hist;
+}
+
+at::Tensor bucket_scatter(at::Tensor buckets, at::Tensor offsets) {
+  // offsets: exclusive prefix sum of the histogram (int64, on device);
+  // mutated in-place as the scatter cursor.
+  check_gpu(buckets, "buckets");
+  check_gpu(offsets, "offsets");
+  int64_t n = buckets.numel();
+  auto perm = at::empty({n}, buckets.options().dtype(at::kLong));
+  launch_scatter(buckets.data_ptr<int32_t>(), offsets.data_ptr<int64_t>(),
+                 perm.data_ptr<int64_t>(), n, current_stream());
+  return perm;
+}
+
+std::vector<at::Tensor> gb_aggregate(at::Tensor keys, at::Tensor vals,
+                                     c10::optional<at::Tensor> valids,
+                                     at::Tensor ops, int64_t tsize,
+                                     bool use_lds) {
+  check_gpu(keys, "keys");
+  check_gpu(vals, "vals");
+  check_gpu(ops, "ops");
+  TORCH_CHECK((tsize & (tsize - 1)) == 0, "tsize must be a power of 2");
+  int64_t n = keys.numel();
+  int n_aggs = (int)vals.size(0);
+  TORCH_CHECK(vals.dim() == 2 && vals.size(1) == n, "vals must be [A, n]");
+  auto tkeys = at::full({tsize}, (int64_t)0x8000000000000000LL,
+                        keys.options());
+  // init aggs to op-appropriate identity: sum/count -> 0; min -> +inf; max -> -inf
+  auto gaggs = at::zeros({n_aggs, tsize}, vals.options());
+  auto ops_cpu = ops.to(at::kCPU);
+  for (int a = 0; a < n_aggs; ++a) {
+    int op = ops_cpu[a].item<int32_t>();
+    if (op == 1) gaggs[a].fill_(std::numeric_limits<double>::infinity());
+    if (op == 2) gaggs[a].fill_(-std::numeric_limits<double>::infinity());
+  }
+  auto gcount = at::zeros({tsize}, keys.options());
+  launch_gb_aggregate(keys.data_ptr<int64_t>(), vals.data_ptr<double>(),
+                      opt_valid_ptr(valids), ops.data_ptr<int32_t>(), n_aggs,
+                      n, tkeys.data_ptr<int64_t>(), gaggs.data_ptr<double>(),
+                      gcount.data_ptr<int64_t>(), tsize, use_lds ? 1 : 0,
+                      current_stream());
+  return {tkeys, gaggs, gcount};
+}
+
+std::vector<at::Tensor> join_build(at::Tensor keys, int64_t tsize) {
+  check_gpu(keys, "keys");
+  TORCH_CHECK((tsize & (tsize - 1)) == 0, "tsize must be a power of 2");
+  int64_t n = keys.numel();
+  auto heads = at::full({tsize}, -1, keys.options().dtype(at::kInt));
+  auto next = at::empty({std::max<int64_t>(n, 1)},
+                        keys.options().dtype(at::kInt));
+  if (n > 0) {
+    launch_join_build(keys.data_ptr<int64_t>(), n, heads.data_ptr<int32_t>(),
+                      next.data_ptr<int32_t>(), tsize, current_stream());
+  }
+  return {heads, next};
+}
+
+at::Tensor join_count(at::Tensor pkeys, at::Tensor bkeys, at::Tensor heads,
+                      at::Tensor next, int64_t tsize) {
+  check_gpu(pkeys, "pkeys");
+  int64_t np = pkeys.numel();
+  auto counts = at::zeros({np}, pkeys.options().dtype(at::kInt));
+  if (np > 0) {
+    launch_join_count(pkeys.data_ptr<int64_t>(), np,
+                      bkeys.data_ptr<int64_t>(), heads.data_ptr<int32_t>(),
+                      next.data_ptr<int32_t>(), tsize,
+                      counts.data_ptr<int32_t>(), current_stream());
+  }
+  return counts;
+}
+
+std::vector<at::Tensor> join_emit(at::Tensor pkeys, at::Tensor bkeys,
+                                  at::Tensor heads, at::Tensor next,
+                                  int64_t tsize, at::Tensor offsets,
+                                  int64_t out_n, int64_t mode) {
+  check_gpu(pkeys, "pkeys");
+  int64_t np = pkeys.numel();
+  auto out_p = at::empty({out_n}, pkeys.options());
+  auto out_b = at::empty({out_n}, pkeys.options());
+  if (np > 0 && out_n > 0) {
+    launch_join_emit(pkeys.data_ptr<int64_t>(), np, bkeys.data_ptr<int64_t>(),
+                     heads.data_ptr<int32_t>(), next.data_ptr<int32_t>(),
+                     tsize, offsets.data_ptr<int64_t>(),
+                     out_p.data_ptr<int64_t>(), out_b.data_ptr<int64_t>(),
+                     (int)mode, current_stream());
+  }
+  return {out_p, out_b};
+}
+
+at::Tensor join_mark_build(at::Tensor pkeys, at::Tensor bkeys,
+                           at::Tensor heads, at::Tensor next, int64_t tsize,
+                           int64_t n_build) {
+  check_gpu(pkeys, "pkeys");
+  auto matched = at::zeros({n_build}, pkeys.options().dtype(at::kBool));
+  if (pkeys.numel() > 0 && n_build > 0) {
+    launch_join_mark_build(pkeys.data_ptr<int64_t>(), pkeys.numel(),
+                           bkeys.data_ptr<int64_t>(),
+                           heads.data_ptr<int32_t>(),
+                           next.data_ptr<int32_t>(), tsize,
+                           matched.data_ptr<bool>(), current_stream());
+  }
+  return matched;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("hash_column", &hash_column,
+        "combine a column into the running row hash");
+  m.def("bucket_of", &bucket_of, "hash -> bucket id");
+  m.def("bucket_histogram", &bucket_histogram, "bucket histogram");
+  m.def("bucket_scatter", &bucket_scatter,
+        "scatter row indices into bucket-contiguous order");
+  m.def("gb_aggregate", &gb_aggregate, "hash group-by aggregation");
+  m.def("join_build", &join_build, "build chained hash table");
+  m.def("join_count", &join_count, "count matches per probe row");
+  m.def("join_emit", &join_emit, "emit join pairs");
+  m.def("join_mark_build", &join_mark_build, "mark matched build rows");
+}

@@ -1,0 +1,485 @@
+// CDNA4 (gfx950 / MI355X) relational kernels for the fugue_amd HIP engine.
+//
+// Design notes (see /opt/skills/guides/cdna_hip_programming.md):
+//  * wave64: blocks are multiples of 64 threads (256 used throughout).
+//  * memory-bound kernels use grid-stride loops capped at ~8192 blocks
+//    (256 CUs x 8 blocks/CU x headroom) — guide §6 G11.
+//  * hash tables are open-addressing in HBM with linear probing;
+//    the group-by kernel additionally has an LDS-resident pre-aggregation
+//    variant (per-workgroup table in shared memory, flushed once) for
+//    low-cardinality keys — the LDS-hash-table requirement of the north
+//    star ("LDS-resident hash tables").
+//  * all cross-workgroup updates use device-scope atomics (guide §6 G16:
+//    per-XCD L2s are not coherent; atomicAdd on global memory is
+//    device-scope by default).
+//
+// These kernels replace the reference framework's delegation of
+// repartition/groupby/join to Spark/Dask/DuckDB.
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define WAVE 64
+#define BLOCK 256
+#define MAX_GRID 8192
+
+static inline int grid_for(int64_t n, int per_thread = 1) {
+  int64_t blocks = (n + (int64_t)BLOCK * per_thread - 1) / ((int64_t)BLOCK * per_thread);
+  if (blocks < 1) blocks = 1;
+  if (blocks > MAX_GRID) blocks = MAX_GRID;
+  return (int)blocks;
+}
+
+__device__ __forceinline__ uint64_t mix64(uint64_t x) {
+  // splitmix64 finalizer — full-avalanche 64-bit mix
+  x ^= x >> 33;
+  x *= 0xff51afd7ed558ccdULL;
+  x ^= x >> 33;
+  x *= 0xc4ceb9fe1a85ec53ULL;
+  x ^= x >> 33;
+  return x;
+}
+
+__device__ __forceinline__ uint64_t hash_combine(uint64_t h, uint64_t v) {
+  return mix64(h ^ (v + 0x9e3779b97f4a7c15ULL + (h << 6) + (h >> 2)));
+}
+
+// ------------------------------------------------------------------ //
+// hashing: combine one column into the running row-hash               //
+// ------------------------------------------------------------------ //
+template <typename T>
+__global__ __launch_bounds__(BLOCK) void hash_col_kernel(
+    const T* __restrict__ data,
+    const bool* __restrict__ valid,  // may be null
+    uint64_t* __restrict__ out,
+    int64_t n,
+    int is_first) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint64_t v;
+    if (valid != nullptr && !valid[i]) {
+      v = 0x9e3779b97f4a7c15ULL;  // canonical NULL hash
+    } else {
+      if constexpr (sizeof(T) == 8) {
+        v = mix64(*reinterpret_cast<const uint64_t*>(&data[i]));
+      } else if constexpr (sizeof(T) == 4) {
+        uint32_t raw = *reinterpret_cast<const uint32_t*>(&data[i]);
+        v = mix64((uint64_t)raw);
+      } else {
+        v = mix64((uint64_t)(uint8_t)data[i]);
+      }
+    }
+    out[i] = is_first ? v : hash_combine(out[i], v);
+  }
+}
+
+extern "C" {
+
+void launch_hash_col_i64(const int64_t* data, const bool* valid, uint64_t* out,
+                         int64_t n, int is_first, hipStream_t stream) {
+  hipLaunchKernelGGL(hash_col_kernel<int64_t>, dim3(grid_for(n)), dim3(BLOCK),
+                     0, stream, data, valid, out, n, is_first);
+}
+void launch_hash_col_i32(const int32_t* data, const bool* valid, uint64_t* out,
+                         int64_t n, int is_first, hipStream_t stream) {
+  hipLaunchKernelGGL(hash_col_kernel<int32_t>, dim3(grid_for(n)), dim3(BLOCK),
+                     0, stream, data, valid, out, n, is_first);
+}
+void launch_hash_col_f64(const double* data, const bool* valid, uint64_t* out,
+                         int64_t n, int is_first, hipStream_t stream) {
+  hipLaunchKernelGGL(hash_col_kernel<double>, dim3(grid_for(n)), dim3(BLOCK),
+                     0, stream, data, valid, out, n, is_first);
+}
+void launch_hash_col_f32(const float* data, const bool* valid, uint64_t* out,
+                         int64_t n, int is_first, hipStream_t stream) {
+  hipLaunchKernelGGL(hash_col_kernel<float>, dim3(grid_for(n)), dim3(BLOCK),
+                     0, stream, data, valid, out, n, is_first);
+}
+void launch_hash_col_i8(const int8_t* data, const bool* valid, uint64_t* out,
+                        int64_t n, int is_first, hipStream_t stream) {
+  hipLaunchKernelGGL(hash_col_kernel<int8_t>, dim3(grid_for(n)), dim3(BLOCK),
+                     0, stream, data, valid, out, n, is_first);
+}
+
+}  // extern "C"
+
+// ------------------------------------------------------------------ //
+// radix partition: histogram + scatter by bucket                      //
+// ------------------------------------------------------------------ //
+__global__ __launch_bounds__(BLOCK) void bucket_of_kernel(
+    const uint64_t* __restrict__ hashes, int32_t* __restrict__ buckets,
+    int64_t n, int32_t num_buckets) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    buckets[i] = (int32_t)(hashes[i] % (uint64_t)num_buckets);
+  }
+}
+
+__global__ __launch_bounds__(BLOCK) void histogram_kernel(
+    const int32_t* __restrict__ buckets, int64_t* __restrict__ hist,
+    int64_t n, int32_t num_buckets) {
+  extern __shared__ int64_t lhist[];
+  for (int i = threadIdx.x; i < num_buckets; i += blockDim.x) lhist[i] = 0;
+  __syncthreads();
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    atomicAdd((unsigned long long*)&lhist[buckets[i]], 1ULL);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < num_buckets; i += blockDim.x) {
+    if (lhist[i] > 0)
+      atomicAdd((unsigned long long*)&hist[i], (unsigned long long)lhist[i]);
+  }
+}
+
+// scatter: stable within bucket is NOT guaranteed (atomic claim);
+// relational semantics don't require row order stability here.
+__global__ __launch_bounds__(BLOCK) void scatter_kernel(
+    const int32_t* __restrict__ buckets, int64_t* __restrict__ cursor,
+    int64_t* __restrict__ perm, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t pos = (int64_t)atomicAdd((unsigned long long*)&cursor[buckets[i]],
+                                     1ULL);
+    perm[pos] = i;
+  }
+}
+
+extern "C" {
+
+void launch_bucket_of(const uint64_t* hashes, int32_t* buckets, int64_t n,
+                      int32_t num_buckets, hipStream_t stream) {
+  hipLaunchKernelGGL(bucket_of_kernel, dim3(grid_for(n)), dim3(BLOCK), 0,
+                     stream, hashes, buckets, n, num_buckets);
+}
+
+void launch_histogram(const int32_t* buckets, int64_t* hist, int64_t n,
+                      int32_t num_buckets, hipStream_t stream) {
+  size_t shm = (size_t)num_buckets * sizeof(int64_t);
+  hipLaunchKernelGGL(histogram_kernel, dim3(grid_for(n)), dim3(BLOCK), shm,
+                     stream, buckets, hist, n, num_buckets);
+}
+
+void launch_scatter(const int32_t* buckets, int64_t* cursor, int64_t* perm,
+                    int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(scatter_kernel, dim3(grid_for(n)), dim3(BLOCK), 0, stream,
+                     buckets, cursor, perm, n);
+}
+
+}  // extern "C"
+
+// ------------------------------------------------------------------ //
+// group-by aggregation: open-addressing HBM hash table                //
+//   key: exact int64 (multi-column keys packed by the python layer)   //
+//   aggs: fp64 matrix [n_aggs, n_rows]; per-agg op code:              //
+//     0=sum 1=min 2=max 3=count(valid)                                //
+//   count of rows per group always collected (slot -1 of aggs)        //
+// ------------------------------------------------------------------ //
+
+#define GB_EMPTY 0x8000000000000000LL
+
+__device__ __forceinline__ void atomic_min_f64(double* addr, double val) {
+  unsigned long long* a = (unsigned long long*)addr;
+  unsigned long long old = *a, assumed;
+  do {
+    assumed = old;
+    double cur = __longlong_as_double(assumed);
+    if (!(val < cur)) break;
+    old = atomicCAS(a, assumed, __double_as_longlong(val));
+  } while (old != assumed);
+}
+
+__device__ __forceinline__ void atomic_max_f64(double* addr, double val) {
+  unsigned long long* a = (unsigned long long*)addr;
+  unsigned long long old = *a, assumed;
+  do {
+    assumed = old;
+    double cur = __longlong_as_double(assumed);
+    if (!(val > cur)) break;
+    old = atomicCAS(a, assumed, __double_as_longlong(val));
+  } while (old != assumed);
+}
+
+// probe/insert into the global table; returns the claimed slot.
+// Aggregation is keyed by SLOT (not a dense group id): slots are
+// compacted after the kernel (torch.nonzero over tkeys != EMPTY), which
+// avoids any in-kernel id-publication wait (an intra-wave spin on
+// another lane's store can deadlock under divergent-branch
+// serialization).
+__device__ __forceinline__ int64_t gb_probe_insert(
+    int64_t key, int64_t* __restrict__ tkeys, int64_t tsize) {
+  uint64_t h = mix64((uint64_t)key);
+  int64_t slot = (int64_t)(h & (uint64_t)(tsize - 1));
+  while (true) {
+    long long prev = (long long)atomicCAS(
+        (unsigned long long*)&tkeys[slot], (unsigned long long)GB_EMPTY,
+        (unsigned long long)key);
+    if (prev == GB_EMPTY || prev == key) return slot;
+    slot = (slot + 1) & (tsize - 1);
+  }
+}
+
+__global__ __launch_bounds__(BLOCK) void gb_aggregate_kernel(
+    const int64_t* __restrict__ keys,
+    const double* __restrict__ vals,    // [n_aggs, n] row-major
+    const bool* __restrict__ valids,    // [n_aggs, n] or null
+    const int32_t* __restrict__ ops,    // [n_aggs]
+    int n_aggs,
+    int64_t n,
+    int64_t* __restrict__ tkeys,        // [tsize] init GB_EMPTY
+    double* __restrict__ gaggs,         // [n_aggs, tsize]
+    int64_t* __restrict__ gcount,       // [tsize] row counts per slot
+    int64_t tsize) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t key = keys[i];
+    int64_t slot = gb_probe_insert(key, tkeys, tsize);
+    atomicAdd((unsigned long long*)&gcount[slot], 1ULL);
+    for (int a = 0; a < n_aggs; ++a) {
+      bool ok = (valids == nullptr) || valids[(int64_t)a * n + i];
+      if (!ok) continue;
+      double v = vals[(int64_t)a * n + i];
+      double* dst = &gaggs[(int64_t)a * tsize + slot];
+      switch (ops[a]) {
+        case 0: atomicAdd(dst, v); break;
+        case 1: atomic_min_f64(dst, v); break;
+        case 2: atomic_max_f64(dst, v); break;
+        case 3: atomicAdd(dst, 1.0); break;
+      }
+    }
+  }
+}
+
+// LDS pre-aggregation variant: per-workgroup table (keys+sum/count only,
+// the common fast path: all ops are sum or count), flushed to the global
+// table at block end.  LDS budget: 1024 entries x (8B key + 8B*n_aggs)
+// must fit 160KiB; python chooses this path only when n_aggs<=4.
+#define LDS_SLOTS 1024
+
+__global__ __launch_bounds__(BLOCK) void gb_aggregate_lds_kernel(
+    const int64_t* __restrict__ keys,
+    const double* __restrict__ vals,
+    const bool* __restrict__ valids,
+    const int32_t* __restrict__ ops,   // all must be 0 (sum) or 3 (count)
+    int n_aggs,
+    int64_t n,
+    int64_t* __restrict__ tkeys,
+    double* __restrict__ gaggs,
+    int64_t* __restrict__ gcount,
+    int64_t tsize) {
+  __shared__ int64_t lkeys[LDS_SLOTS];
+  __shared__ double laggs[4 * LDS_SLOTS];
+  __shared__ long long lcount[LDS_SLOTS];
+  for (int i = threadIdx.x; i < LDS_SLOTS; i += blockDim.x) {
+    lkeys[i] = GB_EMPTY;
+    lcount[i] = 0;
+    for (int a = 0; a < n_aggs; ++a) laggs[a * LDS_SLOTS + i] = 0.0;
+  }
+  __syncthreads();
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t key = keys[i];
+    uint64_t h = mix64((uint64_t)key);
+    int slot = (int)(h & (LDS_SLOTS - 1));
+    bool in_lds = false;
+    for (int probe = 0; probe < 16; ++probe) {
+      long long prev = (long long)atomicCAS(
+          (unsigned long long*)&lkeys[slot], (unsigned long long)GB_EMPTY,
+          (unsigned long long)key);
+      if (prev == GB_EMPTY || prev == key) { in_lds = true; break; }
+      slot = (slot + 1) & (LDS_SLOTS - 1);
+    }
+    if (in_lds) {
+      atomicAdd((unsigned long long*)&lcount[slot], 1ULL);
+      for (int a = 0; a < n_aggs; ++a) {
+        bool ok = (valids == nullptr) || valids[(int64_t)a * n + i];
+        if (!ok) continue;
+        double v = (ops[a] == 3) ? 1.0 : vals[(int64_t)a * n + i];
+        atomicAdd(&laggs[a * LDS_SLOTS + slot], v);
+      }
+    } else {
+      // overflow: straight to the global table
+      int64_t gslot = gb_probe_insert(key, tkeys, tsize);
+      atomicAdd((unsigned long long*)&gcount[gslot], 1ULL);
+      for (int a = 0; a < n_aggs; ++a) {
+        bool ok = (valids == nullptr) || valids[(int64_t)a * n + i];
+        if (!ok) continue;
+        double v = (ops[a] == 3) ? 1.0 : vals[(int64_t)a * n + i];
+        atomicAdd(&gaggs[(int64_t)a * tsize + gslot], v);
+      }
+    }
+  }
+  __syncthreads();
+  // flush the LDS table into the global table
+  for (int i = threadIdx.x; i < LDS_SLOTS; i += blockDim.x) {
+    int64_t key = lkeys[i];
+    if (key == GB_EMPTY) continue;
+    int64_t gslot = gb_probe_insert(key, tkeys, tsize);
+    atomicAdd((unsigned long long*)&gcount[gslot],
+              (unsigned long long)lcount[i]);
+    for (int a = 0; a < n_aggs; ++a) {
+      atomicAdd(&gaggs[(int64_t)a * tsize + gslot], laggs[a * LDS_SLOTS + i]);
+    }
+  }
+}
+
+extern "C" {
+
+void launch_gb_aggregate(const int64_t* keys, const double* vals,
+                         const bool* valids, const int32_t* ops, int n_aggs,
+                         int64_t n, int64_t* tkeys, double* gaggs,
+                         int64_t* gcount, int64_t tsize, int use_lds,
+                         hipStream_t stream) {
+  if (use_lds && n_aggs <= 4) {
+    hipLaunchKernelGGL(gb_aggregate_lds_kernel, dim3(grid_for(n, 4)),
+                       dim3(BLOCK), 0, stream, keys, vals, valids, ops, n_aggs,
+                       n, tkeys, gaggs, gcount, tsize);
+  } else {
+    hipLaunchKernelGGL(gb_aggregate_kernel, dim3(grid_for(n, 4)), dim3(BLOCK),
+                       0, stream, keys, vals, valids, ops, n_aggs, n, tkeys,
+                       gaggs, gcount, tsize);
+  }
+}
+
+}  // extern "C"
+
+// ------------------------------------------------------------------ //
+// hash join (int64 keys): chained-bucket build + 2-pass probe         //
+// ------------------------------------------------------------------ //
+
+__global__ __launch_bounds__(BLOCK) void join_build_kernel(
+    const int64_t* __restrict__ keys, int64_t n,
+    int32_t* __restrict__ heads,   // [tsize] init -1
+    int32_t* __restrict__ next,    // [n]
+    int64_t tsize) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint64_t h = mix64((uint64_t)keys[i]);
+    int64_t slot = (int64_t)(h & (uint64_t)(tsize - 1));
+    int32_t old = atomicExch(&heads[slot], (int32_t)i);
+    next[i] = old;
+  }
+}
+
+// pass 1: per-probe-row match count
+__global__ __launch_bounds__(BLOCK) void join_count_kernel(
+    const int64_t* __restrict__ pkeys, int64_t np,
+    const int64_t* __restrict__ bkeys,
+    const int32_t* __restrict__ heads, const int32_t* __restrict__ next,
+    int64_t tsize, int32_t* __restrict__ counts) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < np;
+       i += stride) {
+    int64_t key = pkeys[i];
+    uint64_t h = mix64((uint64_t)key);
+    int32_t cur = heads[(int64_t)(h & (uint64_t)(tsize - 1))];
+    int32_t c = 0;
+    while (cur >= 0) {
+      if (bkeys[cur] == key) ++c;
+      cur = next[cur];
+    }
+    counts[i] = c;
+  }
+}
+
+// pass 2: emit (probe_idx, build_idx) pairs at offsets
+// mode: 0=inner 1=left (emit (i,-1) when no match) 2=semi 3=anti
+__global__ __launch_bounds__(BLOCK) void join_emit_kernel(
+    const int64_t* __restrict__ pkeys, int64_t np,
+    const int64_t* __restrict__ bkeys,
+    const int32_t* __restrict__ heads, const int32_t* __restrict__ next,
+    int64_t tsize, const int64_t* __restrict__ offsets,
+    int64_t* __restrict__ out_p, int64_t* __restrict__ out_b, int mode) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < np;
+       i += stride) {
+    int64_t key = pkeys[i];
+    uint64_t h = mix64((uint64_t)key);
+    int32_t cur = heads[(int64_t)(h & (uint64_t)(tsize - 1))];
+    int64_t pos = offsets[i];
+    bool any = false;
+    while (cur >= 0) {
+      if (bkeys[cur] == key) {
+        any = true;
+        if (mode == 0 || mode == 1) {
+          out_p[pos] = i;
+          out_b[pos] = cur;
+          ++pos;
+        } else if (mode == 2) {  // semi: first match only
+          out_p[pos] = i;
+          out_b[pos] = cur;
+          ++pos;
+          break;
+        } else {  // anti: presence is enough
+          break;
+        }
+      }
+      cur = next[cur];
+    }
+    if (!any && (mode == 1 || mode == 3)) {
+      out_p[pos] = i;
+      out_b[pos] = -1;
+    }
+  }
+}
+
+// match-flag on the build side (for right/full outer): mark matched rows
+__global__ __launch_bounds__(BLOCK) void join_mark_build_kernel(
+    const int64_t* __restrict__ pkeys, int64_t np,
+    const int64_t* __restrict__ bkeys,
+    const int32_t* __restrict__ heads, const int32_t* __restrict__ next,
+    int64_t tsize, bool* __restrict__ bmatched) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < np;
+       i += stride) {
+    int64_t key = pkeys[i];
+    uint64_t h = mix64((uint64_t)key);
+    int32_t cur = heads[(int64_t)(h & (uint64_t)(tsize - 1))];
+    while (cur >= 0) {
+      if (bkeys[cur] == key) bmatched[cur] = true;
+      cur = next[cur];
+    }
+  }
+}
+
+extern "C" {
+
+void launch_join_build(const int64_t* keys, int64_t n, int32_t* heads,
+                       int32_t* next, int64_t tsize, hipStream_t stream) {
+  hipLaunchKernelGGL(join_build_kernel, dim3(grid_for(n)), dim3(BLOCK), 0,
+                     stream, keys, n, heads, next, tsize);
+}
+
+void launch_join_count(const int64_t* pkeys, int64_t np, const int64_t* bkeys,
+                       const int32_t* heads, const int32_t* next,
+                       int64_t tsize, int32_t* counts, hipStream_t stream) {
+  hipLaunchKernelGGL(join_count_kernel, dim3(grid_for(np)), dim3(BLOCK), 0,
+                     stream, pkeys, np, bkeys, heads, next, tsize, counts);
+}
+
+void launch_join_emit(const int64_t* pkeys, int64_t np, const int64_t* bkeys,
+                      const int32_t* heads, const int32_t* next, int64_t tsize,
+                      const int64_t* offsets, int64_t* out_p, int64_t* out_b,
+                      int mode, hipStream_t stream) {
+  hipLaunchKernelGGL(join_emit_kernel, dim3(grid_for(np)), dim3(BLOCK), 0,
+                     stream, pkeys, np, bkeys, heads, next, tsize, offsets,
+                     out_p, out_b, mode);
+}
+
+void launch_join_mark_build(const int64_t* pkeys, int64_t np,
+                            const int64_t* bkeys, const int32_t* heads,
+                            const int32_t* next, int64_t tsize, bool* bmatched,
+                            hipStream_t stream) {
+  hipLaunchKernelGGL(join_mark_build_kernel, dim3(grid_for(np)), dim3(BLOCK),
+                     0, stream, pkeys, np, bkeys, heads, next, tsize,
+                     bmatched);
+}
+
+}  // extern "C"

@@ -1,0 +1,176 @@
+"""Column-expression evaluation on device frames (torch elementwise ops —
+each lowers to a HIP kernel; fusion of the scalar pipeline comes from
+torch's elementwise fuser, while the relational ops use the hand-written
+kernels in ``csrc/relational.hip``)."""
+from typing import Any, Optional, Tuple
+
+import pyarrow as pa
+import torch
+
+from fugue_amd.column.expressions import (
+    ColumnExpr,
+    _BinaryOpExpr,
+    _FuncExpr,
+    _LiteralColumnExpr,
+    _NamedColumnExpr,
+    _NotOpExpr,
+    _UnaryAggFuncExpr,
+    _UnaryOpExpr,
+)
+from fugue_amd.hip.frame import DeviceColumn, HipDataFrame, StringDeviceColumn
+
+
+class DeviceExprError(NotImplementedError):
+    """Expression can't be evaluated on device → caller falls back."""
+
+
+def eval_device_expr(
+    expr: ColumnExpr, df: HipDataFrame
+) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+    """Evaluate a non-aggregate expression; returns (data, valid_or_None)."""
+    if isinstance(expr, _LiteralColumnExpr):
+        n = df.count()
+        v = expr.value
+        device = df.device
+        if v is None:
+            return (
+                torch.zeros(n, dtype=torch.float64, device=device),
+                torch.zeros(n, dtype=torch.bool, device=device),
+            )
+        if isinstance(v, bool):
+            return torch.full((n,), v, dtype=torch.bool, device=device), None
+        if isinstance(v, int):
+            return torch.full((n,), v, dtype=torch.int64, device=device), None
+        if isinstance(v, float):
+            return torch.full((n,), v, dtype=torch.float64, device=device), None
+        raise DeviceExprError(f"literal {v!r} not supported on device")
+    if isinstance(expr, _NamedColumnExpr):
+        c = df.col(expr.name)
+        if isinstance(c, StringDeviceColumn):
+            raise DeviceExprError("string expressions not supported on device")
+        data = c.data
+        if expr.as_type is not None:
+            data = _cast(data, expr.as_type)
+        return data, c.valid
+    if isinstance(expr, _NotOpExpr):
+        d, v = eval_device_expr(expr.col, df)
+        return ~d.to(torch.bool), v
+    if isinstance(expr, _UnaryOpExpr):
+        d, v = eval_device_expr(expr.col, df)
+        if expr.op == "IS_NULL":
+            if v is None:
+                return torch.zeros_like(d, dtype=torch.bool), None
+            return ~v, None
+        if expr.op == "NOT_NULL":
+            if v is None:
+                return torch.ones_like(d, dtype=torch.bool), None
+            return v, None
+        if expr.op == "-":
+            return -d, v
+        raise DeviceExprError(f"unary {expr.op}")
+    if isinstance(expr, _UnaryAggFuncExpr):
+        raise DeviceExprError("aggregate in scalar context")
+    if isinstance(expr, _BinaryOpExpr):
+        ld, lv = eval_device_expr(expr.left, df)
+        rd, rv = eval_device_expr(expr.right, df)
+        valid = _merge_valid(lv, rv)
+        op = expr.op
+        if op == "&":
+            # SQL three-valued logic approximated: null treated as False
+            res = _as_bool(ld, lv) & _as_bool(rd, rv)
+            return res, None
+        if op == "|":
+            res = _as_bool(ld, lv) | _as_bool(rd, rv)
+            return res, None
+        if op in ("+", "-", "*", "/"):
+            if op == "/":
+                ld = ld.to(torch.float64)
+                rd = rd.to(torch.float64)
+            res = {
+                "+": torch.add,
+                "-": torch.sub,
+                "*": torch.mul,
+                "/": torch.div,
+            }[op](ld, rd)
+            out = res
+            if expr.as_type is not None:
+                out = _cast(out, expr.as_type)
+            return out, valid
+        cmp = {
+            "==": torch.eq,
+            "!=": torch.ne,
+            "<": torch.lt,
+            "<=": torch.le,
+            ">": torch.gt,
+            ">=": torch.ge,
+        }.get(op)
+        if cmp is None:
+            raise DeviceExprError(f"binary {op}")
+        if ld.dtype != rd.dtype:
+            common = torch.promote_types(ld.dtype, rd.dtype)
+            ld = ld.to(common)
+            rd = rd.to(common)
+        res = cmp(ld, rd)
+        if valid is not None:
+            res = res & valid  # null comparison → false
+        return res, None
+    if isinstance(expr, _FuncExpr):
+        fname = expr.func.upper()
+        if fname == "COALESCE":
+            out_d: Optional[torch.Tensor] = None
+            out_v: Optional[torch.Tensor] = None
+            for a in expr.args:
+                d, v = eval_device_expr(a, df)
+                if out_d is None:
+                    out_d, out_v = d, v
+                    continue
+                if out_v is None:
+                    break
+                d = d.to(out_d.dtype)
+                out_d = torch.where(out_v, out_d, d)
+                out_v = out_v | (v if v is not None else torch.ones_like(out_v))
+            if out_v is not None and bool(out_v.all().item()):
+                out_v = None
+            return out_d, out_v
+        raise DeviceExprError(f"function {expr.func}")
+    raise DeviceExprError(f"can't evaluate {expr} on device")
+
+
+def _merge_valid(
+    a: Optional[torch.Tensor], b: Optional[torch.Tensor]
+) -> Optional[torch.Tensor]:
+    if a is None:
+        return b
+    if b is None:
+        return a
+    return a & b
+
+
+def _as_bool(d: torch.Tensor, v: Optional[torch.Tensor]) -> torch.Tensor:
+    b = d.to(torch.bool)
+    if v is not None:
+        b = b & v
+    return b
+
+
+def _cast(data: torch.Tensor, tp: pa.DataType) -> torch.Tensor:
+    import pyarrow as pa
+
+    if pa.types.is_floating(tp):
+        return data.to(torch.float64 if tp == pa.float64() else torch.float32)
+    if pa.types.is_integer(tp):
+        m = {
+            pa.int64(): torch.int64,
+            pa.int32(): torch.int32,
+            pa.int16(): torch.int16,
+            pa.int8(): torch.int8,
+        }
+        return data.to(m.get(tp, torch.int64))
+    if pa.types.is_boolean(tp):
+        return data.to(torch.bool)
+    raise DeviceExprError(f"cast to {tp} not supported on device")
+
+
+def filter_mask(expr: ColumnExpr, df: HipDataFrame) -> torch.Tensor:
+    d, v = eval_device_expr(expr, df)
+    return _as_bool(d, v)

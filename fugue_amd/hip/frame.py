@@ -1,0 +1,481 @@
+"""Device-resident columnar frame: Arrow-schema'd columns in HBM as torch
+tensors.
+
+The MI355X analog of the reference's backend-native frames (SURVEY.md §2.2
+``fugue_ray``'s Arrow-batch blocks are the closest reference): each column
+is a contiguous device buffer + optional validity mask; strings are
+(offsets, bytes) pairs.  288 GB HBM3E per GPU means frames are kept
+resident; host round-trips happen only at UDF boundaries and IO.
+"""
+from typing import Any, Dict, Iterable, List, Optional, Tuple
+
+import numpy as np
+import pandas as pd
+import pyarrow as pa
+import torch
+
+from fugue_amd.dataframe.dataframe import DataFrame, LocalBoundedDataFrame
+from fugue_amd.dataframe.pandas_dataframe import PandasDataFrame
+from fugue_amd.exceptions import (
+    FugueDataFrameEmptyError,
+    FugueDataFrameInitError,
+    FugueDataFrameOperationError,
+)
+from fugue_amd.schema import Schema
+
+_PA_TO_TORCH = {
+    pa.int8(): torch.int8,
+    pa.int16(): torch.int16,
+    pa.int32(): torch.int32,
+    pa.int64(): torch.int64,
+    pa.float32(): torch.float32,
+    pa.float64(): torch.float64,
+    pa.bool_(): torch.bool,
+}
+
+
+def _torch_dtype_for(tp: pa.DataType) -> torch.dtype:
+    if tp in _PA_TO_TORCH:
+        return _PA_TO_TORCH[tp]
+    if pa.types.is_timestamp(tp) or pa.types.is_date(tp):
+        return torch.int64
+    raise FugueDataFrameInitError(f"unsupported device column type {tp}")
+
+
+def supported_device_type(tp: pa.DataType) -> bool:
+    return (
+        tp in _PA_TO_TORCH
+        or pa.types.is_timestamp(tp)
+        or pa.types.is_date(tp)
+        or pa.types.is_string(tp)
+        or pa.types.is_large_string(tp)
+    )
+
+
+class DeviceColumn:
+    """One column in HBM: ``data`` tensor + optional validity mask
+    (True = valid)."""
+
+    def __init__(
+        self,
+        data: torch.Tensor,
+        valid: Optional[torch.Tensor],
+        pa_type: pa.DataType,
+    ):
+        self.data = data
+        self.valid = valid
+        self.pa_type = pa_type
+
+    def __len__(self) -> int:
+        return int(self.data.numel())
+
+    @property
+    def has_nulls(self) -> bool:
+        return self.valid is not None
+
+    @property
+    def is_string(self) -> bool:
+        return False
+
+    def gather(self, idx: torch.Tensor) -> "DeviceColumn":
+        return DeviceColumn(
+            self.data.index_select(0, idx),
+            None if self.valid is None else self.valid.index_select(0, idx),
+            self.pa_type,
+        )
+
+    def slice(self, start: int, length: int) -> "DeviceColumn":
+        return DeviceColumn(
+            self.data[start : start + length],
+            None if self.valid is None else self.valid[start : start + length],
+            self.pa_type,
+        )
+
+    def concat_with(self, others: List["DeviceColumn"]) -> "DeviceColumn":
+        cols = [self] + others
+        data = torch.cat([c.data for c in cols])
+        if any(c.valid is not None for c in cols):
+            valid = torch.cat(
+                [
+                    c.valid
+                    if c.valid is not None
+                    else torch.ones(len(c), dtype=torch.bool, device=c.data.device)
+                    for c in cols
+                ]
+            )
+        else:
+            valid = None
+        return DeviceColumn(data, valid, self.pa_type)
+
+    # --- conversion ---------------------------------------------------- #
+    @staticmethod
+    def from_arrow(arr: pa.ChunkedArray, tp: pa.DataType, device: str) -> "DeviceColumn":
+        if isinstance(arr, pa.ChunkedArray):
+            arr = arr.combine_chunks()
+        if pa.types.is_string(tp) or pa.types.is_large_string(tp):
+            return StringDeviceColumn.from_arrow_strings(arr, device)
+        np_arr = arr.to_numpy(zero_copy_only=False)
+        valid_t: Optional[torch.Tensor] = None
+        if arr.null_count > 0:
+            valid_np = ~np.asarray(arr.is_null())
+            valid_t = torch.from_numpy(valid_np).to(device)
+            if pa.types.is_floating(tp):
+                np_arr = np.nan_to_num(np_arr, nan=0.0)
+            elif np_arr.dtype == np.dtype("object") or np_arr.dtype.kind == "f":
+                # ints with nulls come back as float; fill and cast
+                np_arr = np.nan_to_num(np_arr.astype("float64"), nan=0.0)
+                np_arr = np_arr.astype(_np_dtype_for(tp))
+        if pa.types.is_timestamp(tp):
+            np_arr = np_arr.astype("datetime64[us]").astype("int64")
+        elif pa.types.is_date(tp):
+            np_arr = np_arr.astype("datetime64[D]").astype("int64")
+        elif np_arr.dtype != _np_dtype_for(tp):
+            np_arr = np_arr.astype(_np_dtype_for(tp))
+        data = torch.from_numpy(np.ascontiguousarray(np_arr)).to(device)
+        return DeviceColumn(data, valid_t, tp)
+
+    def to_arrow(self) -> pa.Array:
+        data_np = self.data.cpu().numpy()
+        mask = None
+        if self.valid is not None:
+            mask = ~self.valid.cpu().numpy()
+        if pa.types.is_timestamp(self.pa_type):
+            return pa.Array.from_pandas(
+                data_np.astype("datetime64[us]"), mask=mask, type=self.pa_type
+            )
+        if pa.types.is_date(self.pa_type):
+            return pa.Array.from_pandas(
+                data_np.astype("datetime64[D]"), mask=mask, type=self.pa_type
+            )
+        return pa.Array.from_pandas(data_np, mask=mask, type=self.pa_type)
+
+
+def _np_dtype_for(tp: pa.DataType):
+    return np.dtype(tp.to_pandas_dtype()) if tp not in (pa.bool_(),) else np.dtype("bool")
+
+
+class StringDeviceColumn(DeviceColumn):
+    """UTF-8 strings in HBM: int64 offsets [n+1] + uint8 bytes."""
+
+    def __init__(
+        self,
+        offsets: torch.Tensor,
+        bytes_: torch.Tensor,
+        valid: Optional[torch.Tensor],
+    ):
+        # ``data`` property is the offsets tensor (n+1 elements)
+        super().__init__(offsets, valid, pa.string())
+        self.offsets = offsets
+        self.bytes = bytes_
+
+    def __len__(self) -> int:
+        return int(self.offsets.numel()) - 1
+
+    @property
+    def is_string(self) -> bool:
+        return True
+
+    def gather(self, idx: torch.Tensor) -> "StringDeviceColumn":
+        lengths = self.offsets[1:] - self.offsets[:-1]
+        new_len = lengths.index_select(0, idx)
+        new_offsets = torch.zeros(
+            idx.numel() + 1, dtype=torch.int64, device=idx.device
+        )
+        torch.cumsum(new_len, 0, out=new_offsets[1:])
+        # byte gather: source ranges → flat index
+        starts = self.offsets.index_select(0, idx)
+        total = int(new_offsets[-1].item()) if idx.numel() > 0 else 0
+        if total > 0:
+            seq = torch.arange(total, device=idx.device)
+            row = torch.searchsorted(new_offsets[1:], seq, right=True)
+            src_idx = starts.index_select(0, row) + (
+                seq - new_offsets.index_select(0, row)
+            )
+            new_bytes = self.bytes.index_select(0, src_idx)
+        else:
+            new_bytes = torch.empty(0, dtype=torch.uint8, device=idx.device)
+        return StringDeviceColumn(
+            new_offsets,
+            new_bytes,
+            None if self.valid is None else self.valid.index_select(0, idx),
+        )
+
+    def slice(self, start: int, length: int) -> "StringDeviceColumn":
+        offs = self.offsets[start : start + length + 1]
+        b0 = int(offs[0].item())
+        b1 = int(offs[-1].item())
+        return StringDeviceColumn(
+            offs - b0,
+            self.bytes[b0:b1],
+            None if self.valid is None else self.valid[start : start + length],
+        )
+
+    def concat_with(self, others: List["DeviceColumn"]) -> "StringDeviceColumn":
+        cols: List[StringDeviceColumn] = [self] + others  # type: ignore
+        device = self.offsets.device
+        bytes_ = torch.cat([c.bytes for c in cols])
+        lengths = torch.cat([c.offsets[1:] - c.offsets[:-1] for c in cols])
+        offsets = torch.zeros(
+            int(lengths.numel()) + 1, dtype=torch.int64, device=device
+        )
+        torch.cumsum(lengths, 0, out=offsets[1:])
+        if any(c.valid is not None for c in cols):
+            valid = torch.cat(
+                [
+                    c.valid
+                    if c.valid is not None
+                    else torch.ones(len(c), dtype=torch.bool, device=device)
+                    for c in cols
+                ]
+            )
+        else:
+            valid = None
+        return StringDeviceColumn(offsets, bytes_, valid)
+
+    @staticmethod
+    def from_arrow_strings(arr: pa.Array, device: str) -> "StringDeviceColumn":
+        arr = arr.cast(pa.large_string())
+        buffers = arr.buffers()
+        # buffers: [validity, offsets(int64), data]
+        offsets_np = np.frombuffer(
+            buffers[1], dtype=np.int64, count=len(arr) + 1 + arr.offset
+        )[arr.offset :].copy()
+        base = offsets_np[0]
+        offsets_np = offsets_np - base
+        if buffers[2] is not None and len(buffers[2]) > 0:
+            data_np = np.frombuffer(buffers[2], dtype=np.uint8)[
+                base : base + offsets_np[-1]
+            ].copy()
+        else:
+            data_np = np.empty(0, dtype=np.uint8)
+        valid_t: Optional[torch.Tensor] = None
+        if arr.null_count > 0:
+            valid_np = ~np.asarray(arr.is_null())
+            valid_t = torch.from_numpy(valid_np).to(device)
+        return StringDeviceColumn(
+            torch.from_numpy(offsets_np).to(device),
+            torch.from_numpy(data_np).to(device),
+            valid_t,
+        )
+
+    def to_arrow(self) -> pa.Array:
+        offsets_np = self.offsets.cpu().numpy()
+        bytes_np = self.bytes.cpu().numpy()
+        arr = pa.LargeStringArray.from_buffers(
+            pa.large_string(),
+            len(self),
+            [None, pa.py_buffer(offsets_np.tobytes()), pa.py_buffer(bytes_np.tobytes())],
+        )
+        arr = arr.cast(pa.string())
+        if self.valid is not None:
+            mask = self.valid.cpu().numpy()
+            # rebuild with nulls
+            py = arr.to_pylist()
+            py = [v if m else None for v, m in zip(py, mask)]
+            arr = pa.array(py, type=pa.string())
+        return arr
+
+
+class HipDataFrame(LocalBoundedDataFrame):
+    """A device-resident bounded frame (one shard; the distributed engine
+    holds one HipDataFrame per rank)."""
+
+    def __init__(
+        self,
+        df: Any = None,
+        schema: Any = None,
+        device: Optional[str] = None,
+        _columns: Optional[Dict[str, DeviceColumn]] = None,
+    ):
+        self._device = device or "cuda:0"
+        if _columns is not None:
+            self._cols = _columns
+            super().__init__(schema)
+            return
+        try:
+            if df is None:
+                schema = Schema(schema).assert_not_empty()
+                table = schema.create_empty_arrow()
+            elif isinstance(df, pa.Table):
+                table = df
+                if schema is None:
+                    schema = Schema(df.schema)
+                else:
+                    schema = Schema(schema)
+                    if df.schema != schema.pa_schema:
+                        table = df.select(schema.names).cast(schema.pa_schema)
+            elif isinstance(df, pd.DataFrame):
+                schema = Schema(schema) if schema is not None else None
+                if schema is None:
+                    table = pa.Table.from_pandas(
+                        df.reset_index(drop=True), preserve_index=False
+                    )
+                    schema = Schema(table.schema)
+                else:
+                    from fugue_amd.utils.pandas_like import cast_pandas
+
+                    table = pa.Table.from_pandas(
+                        cast_pandas(df.reset_index(drop=True), schema),
+                        schema=schema.pa_schema,
+                        preserve_index=False,
+                    )
+            elif isinstance(df, DataFrame):
+                schema = df.schema if schema is None else Schema(schema)
+                table = df.as_arrow()
+            elif isinstance(df, Iterable):
+                schema = Schema(schema).assert_not_empty()
+                rows = [
+                    {c: row[i] for i, c in enumerate(schema.names)} for row in df
+                ]
+                table = pa.Table.from_pylist(rows, schema=schema.pa_schema)
+            else:
+                raise ValueError(f"{type(df)} is incompatible with HipDataFrame")
+        except FugueDataFrameInitError:
+            raise
+        except Exception as e:
+            raise FugueDataFrameInitError(str(e)) from e
+        self._cols = {
+            f.name: DeviceColumn.from_arrow(table.column(f.name), f.type, self._device)
+            for f in Schema(schema).fields
+        }
+        super().__init__(schema)
+
+    # --- construction helpers ------------------------------------------- #
+    @staticmethod
+    def from_columns(
+        columns: Dict[str, DeviceColumn], schema: Schema, device: str
+    ) -> "HipDataFrame":
+        return HipDataFrame(schema=schema, device=device, _columns=columns)
+
+    @property
+    def device(self) -> str:
+        return self._device
+
+    @property
+    def columns_map(self) -> Dict[str, DeviceColumn]:
+        return self._cols
+
+    def col(self, name: str) -> DeviceColumn:
+        return self._cols[name]
+
+    @property
+    def native(self) -> Dict[str, DeviceColumn]:
+        return self._cols
+
+    def native_as_df(self) -> "HipDataFrame":
+        return self
+
+    @property
+    def empty(self) -> bool:
+        return self.count() == 0
+
+    def count(self) -> int:
+        if len(self._cols) == 0:
+            return 0
+        return len(next(iter(self._cols.values())))
+
+    def num_bytes(self) -> int:
+        total = 0
+        for c in self._cols.values():
+            total += c.data.numel() * c.data.element_size()
+            if isinstance(c, StringDeviceColumn):
+                total += c.bytes.numel()
+            if c.valid is not None:
+                total += c.valid.numel()
+        return total
+
+    def peek_array(self) -> List[Any]:
+        if self.empty:
+            raise FugueDataFrameEmptyError("dataframe is empty")
+        head = self.gather_rows(
+            torch.zeros(1, dtype=torch.int64, device=self._device)
+        )
+        return list(head.as_arrow().to_pylist()[0].values())
+
+    # --- device ops ------------------------------------------------------ #
+    def gather_rows(self, idx: torch.Tensor) -> "HipDataFrame":
+        cols = {n: c.gather(idx) for n, c in self._cols.items()}
+        return HipDataFrame.from_columns(cols, self.schema, self._device)
+
+    def slice_rows(self, start: int, length: int) -> "HipDataFrame":
+        cols = {n: c.slice(start, length) for n, c in self._cols.items()}
+        return HipDataFrame.from_columns(cols, self.schema, self._device)
+
+    def concat_with(self, others: List["HipDataFrame"]) -> "HipDataFrame":
+        others = [o for o in others if o.count() > 0]
+        if len(others) == 0:
+            return self
+        if self.count() == 0 and len(others) == 1:
+            return others[0]
+        cols = {
+            n: c.concat_with([o.col(n) for o in others])
+            for n, c in self._cols.items()
+        }
+        return HipDataFrame.from_columns(cols, self.schema, self._device)
+
+    # --- conversion ------------------------------------------------------ #
+    def as_arrow(self, type_safe: bool = False) -> pa.Table:
+        arrays = [self._cols[f.name].to_arrow() for f in self.schema.fields]
+        return pa.Table.from_arrays(arrays, schema=self.schema.pa_schema)
+
+    def as_pandas(self) -> pd.DataFrame:
+        return self.as_arrow().to_pandas()
+
+    def as_local_bounded(self) -> LocalBoundedDataFrame:
+        from fugue_amd.dataframe.arrow_dataframe import ArrowDataFrame
+
+        return ArrowDataFrame(self.as_arrow())
+
+    def as_array(
+        self, columns: Optional[List[str]] = None, type_safe: bool = False
+    ) -> List[Any]:
+        tbl = self.as_arrow()
+        if columns is not None:
+            tbl = tbl.select(columns)
+        return [list(d.values()) for d in tbl.to_pylist()]
+
+    def as_array_iterable(
+        self, columns: Optional[List[str]] = None, type_safe: bool = False
+    ) -> Iterable[Any]:
+        yield from self.as_array(columns, type_safe=type_safe)
+
+    # --- frame ops -------------------------------------------------------- #
+    def _drop_cols(self, cols: List[str]) -> "HipDataFrame":
+        schema = self.schema.exclude(cols)
+        return HipDataFrame.from_columns(
+            {n: self._cols[n] for n in schema.names}, schema, self._device
+        )
+
+    def _select_cols(self, cols: List[Any]) -> "HipDataFrame":
+        schema = self.schema.extract(cols)
+        return HipDataFrame.from_columns(
+            {n: self._cols[n] for n in schema.names}, schema, self._device
+        )
+
+    def rename(self, columns: Dict[str, str]) -> "HipDataFrame":
+        try:
+            schema = self.schema.rename(columns)
+        except Exception as e:
+            raise FugueDataFrameOperationError(str(e)) from e
+        new_cols = {columns.get(n, n): c for n, c in self._cols.items()}
+        return HipDataFrame.from_columns(
+            {n: new_cols[n] for n in schema.names}, schema, self._device
+        )
+
+    def alter_columns(self, columns: Any) -> "HipDataFrame":
+        schema = self._get_altered_schema(columns)
+        if schema == self.schema:
+            return self
+        # cast through arrow (rare path; device-native casts later)
+        return HipDataFrame(
+            self.as_arrow().cast(schema.pa_schema), schema, device=self._device
+        )
+
+    def head(
+        self, n: int, columns: Optional[List[str]] = None
+    ) -> LocalBoundedDataFrame:
+        sub = self if columns is None else self._select_cols(columns)
+        n = min(n, sub.count())
+        from fugue_amd.dataframe.arrow_dataframe import ArrowDataFrame
+
+        return ArrowDataFrame(sub.slice_rows(0, n).as_arrow())

@@ -1,0 +1,326 @@
+"""Device relational primitives: row hashing, radix partition, group-by
+aggregation, hash join, sort.
+
+These wrap the hand-written CDNA4 kernels (``csrc/relational.hip``) plus
+rocPRIM-backed torch primitives (sort/cumsum/unique — the "library GEMM"
+equivalents).  Reference call sites being replaced are listed in
+SURVEY.md §2.3.
+"""
+from typing import Any, Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import pyarrow as pa
+import torch
+
+from fugue_amd.exceptions import FugueBug
+from fugue_amd.hip.ext import get_ext
+from fugue_amd.hip.frame import DeviceColumn, HipDataFrame, StringDeviceColumn
+from fugue_amd.schema import Schema
+
+GB_EMPTY = -0x8000000000000000
+
+AGG_SUM = 0
+AGG_MIN = 1
+AGG_MAX = 2
+AGG_COUNT = 3
+
+
+def _next_pow2(n: int) -> int:
+    p = 1
+    while p < n:
+        p <<= 1
+    return p
+
+
+def hash_rows(cols: Sequence[DeviceColumn]) -> torch.Tensor:
+    """Row-wise 64-bit hash over multiple columns (int64 tensor holding
+    uint64 bits)."""
+    ext = get_ext()
+    n = len(cols[0])
+    device = cols[0].data.device
+    out = torch.empty(n, dtype=torch.int64, device=device)
+    first = True
+    for c in cols:
+        if isinstance(c, StringDeviceColumn):
+            raise NotImplementedError(
+                "string partition keys are not yet supported on device"
+            )
+        data = c.data
+        if data.dtype == torch.int16:
+            data = data.to(torch.int32)
+        ext.hash_column(data, c.valid, out, first)
+        first = False
+    return out
+
+
+def partition_by_hash(
+    df: HipDataFrame, hashes: torch.Tensor, num_buckets: int
+) -> Tuple[HipDataFrame, torch.Tensor]:
+    """Reorder rows bucket-contiguously; returns (frame, counts[num_buckets]).
+
+    MI355X equivalent of Dask's ``hash_repartition``
+    (``fugue_dask/_utils.py:44``).
+    """
+    ext = get_ext()
+    buckets = ext.bucket_of(hashes, num_buckets)
+    counts = ext.bucket_histogram(buckets, num_buckets)
+    offsets = torch.zeros(num_buckets, dtype=torch.int64, device=hashes.device)
+    torch.cumsum(counts[:-1], 0, out=offsets[1:])
+    perm = ext.bucket_scatter(buckets, offsets.clone())
+    return df.gather_rows(perm), counts
+
+
+def rand_buckets(n: int, num_buckets: int, seed: Optional[int], device) -> torch.Tensor:
+    gen = torch.Generator(device=device)
+    if seed is not None:
+        gen.manual_seed(seed)
+    return torch.randint(
+        0, num_buckets, (n,), dtype=torch.int64, device=device, generator=gen
+    )
+
+
+def pack_keys(
+    key_cols: Sequence[DeviceColumn],
+    mins: Optional[List[int]] = None,
+    widths: Optional[List[int]] = None,
+) -> Tuple[torch.Tensor, Optional[Dict[str, Any]]]:
+    """Pack 1+ integer-like key columns into a single exact int64 key.
+
+    Nulls get their own code (0); valid values are offset by 1.  Returns
+    (keys, meta) where meta describes how to unpack; meta None means the
+    single input column was used as-is (no nulls, int64).
+    """
+    if (
+        len(key_cols) == 1
+        and key_cols[0].data.dtype == torch.int64
+        and key_cols[0].valid is None
+        and mins is None
+    ):
+        return key_cols[0].data, None
+    datas = []
+    comp_mins: List[int] = []
+    comp_widths: List[int] = []
+    for i, c in enumerate(key_cols):
+        if isinstance(c, StringDeviceColumn):
+            raise NotImplementedError("string group keys not yet on device")
+        d = c.data.to(torch.int64)
+        if mins is None:
+            lo = int(d.min().item()) if d.numel() > 0 else 0
+            hi = int(d.max().item()) if d.numel() > 0 else 0
+        else:
+            lo, hi = mins[i], mins[i] + (1 << widths[i]) - 2
+        code = d - lo + 1  # 0 reserved for NULL
+        if c.valid is not None:
+            code = torch.where(c.valid, code, torch.zeros_like(code))
+        datas.append(code)
+        comp_mins.append(lo)
+        if widths is None:
+            w = max(1, int(np.ceil(np.log2(max(2, hi - lo + 2)))))
+            comp_widths.append(w)
+        else:
+            comp_widths.append(widths[i])
+    total_bits = sum(comp_widths)
+    if total_bits <= 63:
+        packed = torch.zeros_like(datas[0])
+        shift = 0
+        for d, w in zip(reversed(datas), reversed(comp_widths)):
+            packed = packed | (d << shift)
+            shift += w
+        return packed, dict(mode="pack", mins=comp_mins, widths=comp_widths)
+    # fallback: dense re-encode via torch.unique on stacked keys
+    stacked = torch.stack(datas, dim=1)
+    uniq, inverse = torch.unique(stacked, dim=0, return_inverse=True)
+    return inverse.to(torch.int64), dict(
+        mode="unique", uniq=uniq, mins=comp_mins
+    )
+
+
+def unpack_keys(
+    packed: torch.Tensor,
+    meta: Optional[Dict[str, Any]],
+    key_cols: Sequence[DeviceColumn],
+) -> List[DeviceColumn]:
+    """Recover key columns (with nulls) from packed group keys."""
+    if meta is None:
+        return [DeviceColumn(packed, None, key_cols[0].pa_type)]
+    res: List[DeviceColumn] = []
+    if meta["mode"] == "pack":
+        shift = sum(meta["widths"])
+        for i, (c, lo, w) in enumerate(
+            zip(key_cols, meta["mins"], meta["widths"])
+        ):
+            shift -= w
+            code = (packed >> shift) & ((1 << w) - 1)
+            valid = code != 0
+            data = (code - 1 + lo).to(c.data.dtype)
+            res.append(
+                DeviceColumn(
+                    data, None if bool(valid.all().item()) else valid, c.pa_type
+                )
+            )
+        return res
+    uniq = meta["uniq"]
+    for i, c in enumerate(key_cols):
+        code = uniq[packed, i]
+        valid = code != 0
+        data = (code - 1 + meta["mins"][i]).to(c.data.dtype)
+        res.append(
+            DeviceColumn(
+                data, None if bool(valid.all().item()) else valid, c.pa_type
+            )
+        )
+    return res
+
+
+def groupby_aggregate(
+    df: HipDataFrame,
+    keys: List[str],
+    aggs: List[Tuple[str, int, str]],  # (input col, op, output name)
+    expected_groups: Optional[int] = None,
+) -> Tuple[torch.Tensor, Dict[str, torch.Tensor], torch.Tensor, Optional[Dict[str, Any]]]:
+    """Hash group-by: returns (group_keys_packed, {out_name: fp64 values},
+    group_row_counts, pack_meta).
+
+    Reference comparator: the groupby-aggregate SQL path
+    (``fugue/execution/execution_engine.py:889`` + qpd/duckdb/dask-sql).
+    """
+    ext = get_ext()
+    n = df.count()
+    device = df.col(keys[0]).data.device if keys else torch.device(df.device)
+    key_cols = [df.col(k) for k in keys]
+    packed, meta = pack_keys(key_cols)
+    n_aggs = len(aggs)
+    if n_aggs > 0:
+        vals = torch.empty((n_aggs, n), dtype=torch.float64, device=device)
+        any_null = any(df.col(c).valid is not None for c, _, _ in aggs)
+        valids: Optional[torch.Tensor] = None
+        if any_null:
+            valids = torch.ones((n_aggs, n), dtype=torch.bool, device=device)
+        for i, (cname, op, _) in enumerate(aggs):
+            c = df.col(cname)
+            vals[i] = c.data.to(torch.float64)
+            if valids is not None and c.valid is not None:
+                valids[i] = c.valid
+        ops = torch.tensor(
+            [op for _, op, _ in aggs], dtype=torch.int32, device=device
+        )
+    else:
+        vals = torch.zeros((1, n), dtype=torch.float64, device=device)
+        valids = None
+        ops = torch.tensor([AGG_COUNT], dtype=torch.int32, device=device)
+        n_aggs = 0
+    if expected_groups is None:
+        # sample-based cardinality estimate
+        if n > 65536:
+            sample = packed[:: max(1, n // 65536)]
+            est = int(torch.unique(sample).numel())
+            expected_groups = min(n, est * max(1, n // max(1, sample.numel())))
+        else:
+            expected_groups = max(1, n)
+    tsize = _next_pow2(max(16, int(expected_groups * 2)))
+    use_lds = expected_groups <= 100_000 and all(
+        op in (AGG_SUM, AGG_COUNT) for _, op, _ in aggs
+    )
+    tkeys, gaggs, gcount = ext.gb_aggregate(
+        packed,
+        vals if len(aggs) > 0 else vals,
+        valids,
+        ops,
+        tsize,
+        use_lds,
+    )
+    occupied = (tkeys != GB_EMPTY).nonzero(as_tuple=True)[0]
+    out_keys = tkeys.index_select(0, occupied)
+    out_count = gcount.index_select(0, occupied)
+    out_aggs: Dict[str, torch.Tensor] = {}
+    for i, (_, op, oname) in enumerate(aggs):
+        out_aggs[oname] = gaggs[i].index_select(0, occupied)
+    return out_keys, out_aggs, out_count, meta
+
+
+def hash_join_indices(
+    probe_keys: torch.Tensor,
+    build_keys: torch.Tensor,
+    how: str,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Join on exact int64 keys; returns (probe_idx, build_idx) where
+    build_idx == -1 marks no-match rows (left/anti).
+
+    ``how`` ∈ {inner, left, semi, anti}.  CDNA4 chained-bucket hash join
+    (replaces the reference's delegation to pandas/duckdb/spark joins,
+    SURVEY.md §2.3 row "join ×9 types").
+    """
+    ext = get_ext()
+    nb = int(build_keys.numel())
+    tsize = _next_pow2(max(16, nb * 2))
+    heads, nxt = ext.join_build(build_keys, tsize)
+    counts = ext.join_count(probe_keys, build_keys, heads, nxt, tsize)
+    counts64 = counts.to(torch.int64)
+    if how == "inner":
+        out_counts = counts64
+    elif how == "left":
+        out_counts = torch.clamp(counts64, min=1)
+    elif how == "semi":
+        out_counts = (counts64 > 0).to(torch.int64)
+    elif how == "anti":
+        out_counts = (counts64 == 0).to(torch.int64)
+    else:
+        raise FugueBug(f"unsupported join mode {how}")
+    offsets = torch.zeros_like(out_counts)
+    if out_counts.numel() > 1:
+        torch.cumsum(out_counts[:-1], 0, out=offsets[1:])
+    total = int(out_counts.sum().item())
+    mode = {"inner": 0, "left": 1, "semi": 2, "anti": 3}[how]
+    out_p, out_b = ext.join_emit(
+        probe_keys, build_keys, heads, nxt, tsize, offsets, total, mode
+    )
+    return out_p, out_b
+
+
+def mark_matched_build_rows(
+    probe_keys: torch.Tensor, build_keys: torch.Tensor
+) -> torch.Tensor:
+    ext = get_ext()
+    nb = int(build_keys.numel())
+    tsize = _next_pow2(max(16, nb * 2))
+    heads, nxt = ext.join_build(build_keys, tsize)
+    return ext.join_mark_build(
+        probe_keys, build_keys, heads, nxt, tsize, nb
+    )
+
+
+def sort_indices(
+    df: HipDataFrame, by: List[str], ascending: List[bool]
+) -> torch.Tensor:
+    """Stable multi-key sort permutation (rocPRIM radix sort via torch,
+    least-significant key first)."""
+    n = df.count()
+    device = torch.device(df.device)
+    perm = torch.arange(n, dtype=torch.int64, device=device)
+    for name, asc in reversed(list(zip(by, ascending))):
+        c = df.col(name)
+        if isinstance(c, StringDeviceColumn):
+            raise NotImplementedError("string sort keys not yet on device")
+        vals = c.data.index_select(0, perm)
+        if c.valid is not None:
+            # nulls last regardless of direction (pandas na_position default)
+            big = (
+                torch.iinfo(vals.dtype).max
+                if not vals.is_floating_point()
+                else float("inf")
+            )
+            v = c.valid.index_select(0, perm)
+            vals = torch.where(v, vals, torch.full_like(vals, big))
+        idx = torch.argsort(vals, stable=True, descending=not asc)
+        perm = perm.index_select(0, idx)
+    return perm
+
+
+def group_boundaries(sorted_keys: torch.Tensor) -> torch.Tensor:
+    """Start offsets of each group in a key-sorted int64 tensor."""
+    n = sorted_keys.numel()
+    if n == 0:
+        return torch.zeros(0, dtype=torch.int64, device=sorted_keys.device)
+    change = torch.ones(n, dtype=torch.bool, device=sorted_keys.device)
+    change[1:] = sorted_keys[1:] != sorted_keys[:-1]
+    return change.nonzero(as_tuple=True)[0]
