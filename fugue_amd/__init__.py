@@ -23,3 +23,6 @@ from fugue_amd.dataframe import (
     as_fugue_df,
 )
 from fugue_amd.dataset import Dataset
+from fugue_amd.registry import register_builtins as _register_builtins
+
+_register_builtins()

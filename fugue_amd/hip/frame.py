@@ -261,10 +261,14 @@ class StringDeviceColumn(DeviceColumn):
     def to_arrow(self) -> pa.Array:
         offsets_np = self.offsets.cpu().numpy()
         bytes_np = self.bytes.cpu().numpy()
-        arr = pa.LargeStringArray.from_buffers(
+        arr = pa.Array.from_buffers(
             pa.large_string(),
             len(self),
-            [None, pa.py_buffer(offsets_np.tobytes()), pa.py_buffer(bytes_np.tobytes())],
+            [
+                None,
+                pa.py_buffer(offsets_np.tobytes()),
+                pa.py_buffer(bytes_np.tobytes()),
+            ],
         )
         arr = arr.cast(pa.string())
         if self.valid is not None:

@@ -220,9 +220,10 @@ def test_fugue_sql_on_hip(engine):
         "SELECT k, SUM(v) AS s FROM a GROUP BY k",
         a=a,
         engine=engine,
+        as_fugue=True,
     )
     expected = a.groupby("k", as_index=False).agg(s=("v", "sum"))
-    got = pd.DataFrame(res).sort_values("k").reset_index(drop=True)
+    got = res.as_pandas().sort_values("k").reset_index(drop=True)
     np.testing.assert_allclose(
         got["s"].values, expected.sort_values("k")["s"].values, rtol=1e-9
     )
