@@ -28,6 +28,12 @@ void launch_gb_aggregate(const int64_t*, const double*, const bool*,
                          int64_t*, int64_t, int, hipStream_t);
 void launch_join_build(const int64_t*, int64_t, int32_t*, int32_t*, int64_t,
                        hipStream_t);
+void launch_gb_part_hist(const int64_t*, int64_t, int, int64_t*, hipStream_t);
+void launch_gb_part_scatter(const int64_t*, const double*, int, int64_t, int,
+                            int64_t*, int64_t*, double*, hipStream_t);
+void launch_gb_aggregate_part(const int64_t*, const double*, const int32_t*,
+                              int, int64_t, const int64_t*, int64_t, int64_t*,
+                              double*, int64_t*, int64_t, hipStream_t);
 void launch_join_count(const int64_t*, int64_t, const int64_t*, const int32_t*,
                        const int32_t*, int64_t, int32_t*, hipStream_t);
 void launch_join_emit(const int64_t*, int64_t, const int64_t*, const int32_t*,
@@ -153,6 +159,49 @@ std::vector<at::Tensor> gb_aggregate(at::Tensor keys, at::Tensor vals,
   return {tkeys, gaggs, gcount};
 }
 
+std::vector<at::Tensor> gb_aggregate_partitioned(
+    at::Tensor keys, at::Tensor vals, at::Tensor ops, int64_t num_parts,
+    int64_t tsize) {
+  check_gpu(keys, "keys");
+  check_gpu(vals, "vals");
+  TORCH_CHECK((tsize & (tsize - 1)) == 0, "tsize must be a power of 2");
+  TORCH_CHECK((num_parts & (num_parts - 1)) == 0,
+              "num_parts must be a power of 2");
+  int64_t n = keys.numel();
+  int n_aggs = (int)vals.size(0);
+  auto stream = current_stream();
+  int shift = 64;
+  {
+    int64_t p = num_parts;
+    while (p > 1) { --shift; p >>= 1; }
+  }
+  // phase 1: histogram + scan
+  auto hist = at::zeros({num_parts}, keys.options());
+  launch_gb_part_hist(keys.data_ptr<int64_t>(), n, shift,
+                      hist.data_ptr<int64_t>(), stream);
+  auto offsets = at::zeros({num_parts + 1}, keys.options());
+  offsets.narrow(0, 1, num_parts).copy_(at::cumsum(hist, 0));
+  auto cursor = offsets.narrow(0, 0, num_parts).clone();
+  // phase 2: scatter into partitioned order
+  auto pkeys = at::empty({n}, keys.options());
+  auto pvals = at::empty({n_aggs, n}, vals.options());
+  launch_gb_part_scatter(keys.data_ptr<int64_t>(), vals.data_ptr<double>(),
+                         n_aggs, n, shift, cursor.data_ptr<int64_t>(),
+                         pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),
+                         stream);
+  // phase 3: per-partition LDS aggregation into the global table
+  auto tkeys = at::full({tsize}, (int64_t)0x8000000000000000LL,
+                        keys.options());
+  auto gaggs = at::zeros({n_aggs, tsize}, vals.options());
+  auto gcount = at::zeros({tsize}, keys.options());
+  launch_gb_aggregate_part(
+      pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),
+      ops.data_ptr<int32_t>(), n_aggs, n, offsets.data_ptr<int64_t>(),
+      num_parts, tkeys.data_ptr<int64_t>(), gaggs.data_ptr<double>(),
+      gcount.data_ptr<int64_t>(), tsize, stream);
+  return {tkeys, gaggs, gcount};
+}
+
 std::vector<at::Tensor> join_build(at::Tensor keys, int64_t tsize) {
   check_gpu(keys, "keys");
   TORCH_CHECK((tsize & (tsize - 1)) == 0, "tsize must be a power of 2");
@@ -222,6 +271,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bucket_scatter", &bucket_scatter,
         "scatter row indices into bucket-contiguous order");
   m.def("gb_aggregate", &gb_aggregate, "hash group-by aggregation");
+  m.def("gb_aggregate_partitioned", &gb_aggregate_partitioned,
+        "partitioned (2-phase) hash group-by aggregation");
   m.def("join_build", &join_build, "build chained hash table");
   m.def("join_count", &join_count, "count matches per probe row");
   m.def("join_emit", &join_emit, "emit join pairs");

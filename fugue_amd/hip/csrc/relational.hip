@@ -17,6 +17,7 @@
 // repartition/groupby/join to Spark/Dask/DuckDB.
 
 #include <hip/hip_runtime.h>
+#include <algorithm>
 #include <cstdint>
 
 #define WAVE 64
@@ -328,6 +329,141 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_lds_kernel(
     }
   }
 }
+
+// ------------------------------------------------------------------ //
+// partitioned group-by (high cardinality): phase 1 scatter rows into   //
+// hash-partitioned order, phase 2 per-partition LDS aggregation.       //
+// Partition id = TOP bits of mix64(key); LDS slots use LOW bits, so    //
+// keys within one partition still spread across the LDS table.         //
+// ------------------------------------------------------------------ //
+
+__global__ __launch_bounds__(BLOCK) void gb_part_hist_kernel(
+    const int64_t* __restrict__ keys, int64_t n, int shift,
+    int64_t* __restrict__ hist) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t p = (int64_t)(mix64((uint64_t)keys[i]) >> shift);
+    atomicAdd((unsigned long long*)&hist[p], 1ULL);
+  }
+}
+
+__global__ __launch_bounds__(BLOCK) void gb_part_scatter_kernel(
+    const int64_t* __restrict__ keys,
+    const double* __restrict__ vals,   // [n_aggs, n]
+    int n_aggs, int64_t n, int shift,
+    int64_t* __restrict__ cursor,      // [P] exclusive offsets (mutated)
+    int64_t* __restrict__ out_keys,
+    double* __restrict__ out_vals) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t key = keys[i];
+    int64_t p = (int64_t)(mix64((uint64_t)key) >> shift);
+    int64_t pos =
+        (int64_t)atomicAdd((unsigned long long*)&cursor[p], 1ULL);
+    out_keys[pos] = key;
+    for (int a = 0; a < n_aggs; ++a)
+      out_vals[(int64_t)a * n + pos] = vals[(int64_t)a * n + i];
+  }
+}
+
+// one workgroup per partition (grid-stride over partitions); ops must be
+// sum (0) or count (3)
+__global__ __launch_bounds__(BLOCK) void gb_aggregate_part_kernel(
+    const int64_t* __restrict__ part_keys,
+    const double* __restrict__ part_vals,  // [n_aggs, n]
+    const int32_t* __restrict__ ops,
+    int n_aggs, int64_t n,
+    const int64_t* __restrict__ offsets,   // [P+1]
+    int64_t num_parts,
+    int64_t* __restrict__ tkeys,
+    double* __restrict__ gaggs,
+    int64_t* __restrict__ gcount,
+    int64_t tsize) {
+  __shared__ int64_t lkeys[LDS_SLOTS];
+  __shared__ double laggs[4 * LDS_SLOTS];
+  __shared__ long long lcount[LDS_SLOTS];
+  for (int64_t p = blockIdx.x; p < num_parts; p += gridDim.x) {
+    for (int i = threadIdx.x; i < LDS_SLOTS; i += blockDim.x) {
+      lkeys[i] = GB_EMPTY;
+      lcount[i] = 0;
+      for (int a = 0; a < n_aggs; ++a) laggs[a * LDS_SLOTS + i] = 0.0;
+    }
+    __syncthreads();
+    int64_t lo = offsets[p], hi = offsets[p + 1];
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+      int64_t key = part_keys[i];
+      uint64_t h = mix64((uint64_t)key);
+      int slot = (int)(h & (LDS_SLOTS - 1));
+      bool in_lds = false;
+      for (int probe = 0; probe < 32; ++probe) {
+        long long prev = (long long)atomicCAS(
+            (unsigned long long*)&lkeys[slot], (unsigned long long)GB_EMPTY,
+            (unsigned long long)key);
+        if (prev == GB_EMPTY || prev == key) { in_lds = true; break; }
+        slot = (slot + 1) & (LDS_SLOTS - 1);
+      }
+      if (in_lds) {
+        atomicAdd((unsigned long long*)&lcount[slot], 1ULL);
+        for (int a = 0; a < n_aggs; ++a) {
+          double v = (ops[a] == 3) ? 1.0 : part_vals[(int64_t)a * n + i];
+          atomicAdd(&laggs[a * LDS_SLOTS + slot], v);
+        }
+      } else {
+        int64_t gslot = gb_probe_insert(key, tkeys, tsize);
+        atomicAdd((unsigned long long*)&gcount[gslot], 1ULL);
+        for (int a = 0; a < n_aggs; ++a) {
+          double v = (ops[a] == 3) ? 1.0 : part_vals[(int64_t)a * n + i];
+          atomicAdd(&gaggs[(int64_t)a * tsize + gslot], v);
+        }
+      }
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < LDS_SLOTS; i += blockDim.x) {
+      int64_t key = lkeys[i];
+      if (key == GB_EMPTY) continue;
+      int64_t gslot = gb_probe_insert(key, tkeys, tsize);
+      atomicAdd((unsigned long long*)&gcount[gslot],
+                (unsigned long long)lcount[i]);
+      for (int a = 0; a < n_aggs; ++a)
+        atomicAdd(&gaggs[(int64_t)a * tsize + gslot],
+                  laggs[a * LDS_SLOTS + i]);
+    }
+    __syncthreads();
+  }
+}
+
+extern "C" {
+
+void launch_gb_part_hist(const int64_t* keys, int64_t n, int shift,
+                         int64_t* hist, hipStream_t stream) {
+  hipLaunchKernelGGL(gb_part_hist_kernel, dim3(grid_for(n, 4)), dim3(BLOCK),
+                     0, stream, keys, n, shift, hist);
+}
+
+void launch_gb_part_scatter(const int64_t* keys, const double* vals,
+                            int n_aggs, int64_t n, int shift, int64_t* cursor,
+                            int64_t* out_keys, double* out_vals,
+                            hipStream_t stream) {
+  hipLaunchKernelGGL(gb_part_scatter_kernel, dim3(grid_for(n, 4)),
+                     dim3(BLOCK), 0, stream, keys, vals, n_aggs, n, shift,
+                     cursor, out_keys, out_vals);
+}
+
+void launch_gb_aggregate_part(const int64_t* part_keys,
+                              const double* part_vals, const int32_t* ops,
+                              int n_aggs, int64_t n, const int64_t* offsets,
+                              int64_t num_parts, int64_t* tkeys,
+                              double* gaggs, int64_t* gcount, int64_t tsize,
+                              hipStream_t stream) {
+  int grid = (int)std::min<int64_t>(num_parts, 8192);
+  hipLaunchKernelGGL(gb_aggregate_part_kernel, dim3(grid), dim3(BLOCK), 0,
+                     stream, part_keys, part_vals, ops, n_aggs, n, offsets,
+                     num_parts, tkeys, gaggs, gcount, tsize);
+}
+
+}  // extern "C"
 
 extern "C" {
 

@@ -1110,11 +1110,17 @@ class HipExecutionEngine(ExecutionEngine):
                 if c.is_distinct:
                     raise DeviceExprError("count distinct: fallback")
                 if fname == "COUNT":
-                    if isinstance(arg, _NamedColumnExpr) and arg.name != "*":
+                    if (
+                        isinstance(arg, _NamedColumnExpr)
+                        and arg.name != "*"
+                        and d.col(arg.name).valid is not None
+                    ):
                         src = arg.name
                         tmp = _add_partial(src, dops.AGG_COUNT)
                         plans.append((name, "count", tmp))
                     else:
+                        # COUNT(*) or COUNT(col) on a no-null column is the
+                        # per-group row count — no agg column traffic needed
                         plans.append((name, "rowcount", None))
                     continue
                 if not isinstance(arg, _NamedColumnExpr):

@@ -190,7 +190,18 @@ def groupby_aggregate(
     key_cols = [df.col(k) for k in keys]
     packed, meta = pack_keys(key_cols)
     n_aggs = len(aggs)
-    if n_aggs > 0:
+    if (
+        n_aggs == 1
+        and df.col(aggs[0][0]).data.dtype == torch.float64
+        and df.col(aggs[0][0]).valid is None
+    ):
+        # zero-copy: single fp64 agg column used in place
+        vals = df.col(aggs[0][0]).data.unsqueeze(0)
+        valids = None
+        ops = torch.tensor(
+            [aggs[0][1]], dtype=torch.int32, device=device
+        )
+    elif n_aggs > 0:
         vals = torch.empty((n_aggs, n), dtype=torch.float64, device=device)
         any_null = any(df.col(c).valid is not None for c, _, _ in aggs)
         valids: Optional[torch.Tensor] = None
@@ -218,17 +229,19 @@ def groupby_aggregate(
         else:
             expected_groups = max(1, n)
     tsize = _next_pow2(max(16, int(expected_groups * 2)))
-    use_lds = expected_groups <= 100_000 and all(
-        op in (AGG_SUM, AGG_COUNT) for _, op, _ in aggs
-    )
-    tkeys, gaggs, gcount = ext.gb_aggregate(
-        packed,
-        vals if len(aggs) > 0 else vals,
-        valids,
-        ops,
-        tsize,
-        use_lds,
-    )
+    sum_count_only = all(op in (AGG_SUM, AGG_COUNT) for _, op, _ in aggs)
+    if expected_groups > 100_000 and valids is None and sum_count_only:
+        # high cardinality: 2-phase partitioned aggregation (hash-partition
+        # rows so each partition's groups fit the per-workgroup LDS table)
+        num_parts = min(1 << 20, _next_pow2(max(16, expected_groups // 512)))
+        tkeys, gaggs, gcount = ext.gb_aggregate_partitioned(
+            packed, vals, ops, num_parts, tsize
+        )
+    else:
+        use_lds = expected_groups <= 100_000 and sum_count_only
+        tkeys, gaggs, gcount = ext.gb_aggregate(
+            packed, vals, valids, ops, tsize, use_lds
+        )
     occupied = (tkeys != GB_EMPTY).nonzero(as_tuple=True)[0]
     out_keys = tkeys.index_select(0, occupied)
     out_count = gcount.index_select(0, occupied)
