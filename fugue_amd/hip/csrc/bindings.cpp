@@ -28,9 +28,10 @@ void launch_gb_aggregate(const int64_t*, const double*, const bool*,
                          int64_t*, int64_t, int, hipStream_t);
 void launch_join_build(const int64_t*, int64_t, int32_t*, int32_t*, int64_t,
                        hipStream_t);
-void launch_gb_part_hist(const int64_t*, int64_t, int, int64_t*, hipStream_t);
+void launch_gb_part_hist(const int64_t*, int64_t, int, int64_t*, int,
+                         hipStream_t);
 void launch_gb_part_scatter(const int64_t*, const double*, int, int64_t, int,
-                            int64_t*, int64_t*, double*, hipStream_t);
+                            int64_t*, int64_t*, double*, int, hipStream_t);
 void launch_gb_aggregate_part(const int64_t*, const double*, const int32_t*,
                               int, int64_t, const int64_t*, int64_t, int64_t*,
                               double*, int64_t*, int64_t, hipStream_t);
@@ -177,8 +178,9 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
   }
   // phase 1: histogram + scan
   auto hist = at::zeros({num_parts}, keys.options());
+  TORCH_CHECK(num_parts <= 4096, "num_parts must be <= 4096");
   launch_gb_part_hist(keys.data_ptr<int64_t>(), n, shift,
-                      hist.data_ptr<int64_t>(), stream);
+                      hist.data_ptr<int64_t>(), (int)num_parts, stream);
   auto offsets = at::zeros({num_parts + 1}, keys.options());
   offsets.narrow(0, 1, num_parts).copy_(at::cumsum(hist, 0));
   auto cursor = offsets.narrow(0, 0, num_parts).clone();
@@ -188,7 +190,7 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
   launch_gb_part_scatter(keys.data_ptr<int64_t>(), vals.data_ptr<double>(),
                          n_aggs, n, shift, cursor.data_ptr<int64_t>(),
                          pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),
-                         stream);
+                         (int)num_parts, stream);
   // phase 3: per-partition LDS aggregation into the global table
   auto tkeys = at::full({tsize}, (int64_t)0x8000000000000000LL,
                         keys.options());

@@ -337,16 +337,34 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_lds_kernel(
 // keys within one partition still spread across the LDS table.         //
 // ------------------------------------------------------------------ //
 
+#define PART_MAX 4096
+
+// LDS-privatized histogram (global atomics only on the per-block flush)
 __global__ __launch_bounds__(BLOCK) void gb_part_hist_kernel(
     const int64_t* __restrict__ keys, int64_t n, int shift,
-    int64_t* __restrict__ hist) {
+    int64_t* __restrict__ hist, int num_parts) {
+  __shared__ int lhist[PART_MAX];
+  for (int i = threadIdx.x; i < num_parts; i += blockDim.x) lhist[i] = 0;
+  __syncthreads();
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
-    int64_t p = (int64_t)(mix64((uint64_t)keys[i]) >> shift);
-    atomicAdd((unsigned long long*)&hist[p], 1ULL);
+    int p = (int)(mix64((uint64_t)keys[i]) >> shift);
+    atomicAdd(&lhist[p], 1);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < num_parts; i += blockDim.x) {
+    if (lhist[i] > 0)
+      atomicAdd((unsigned long long*)&hist[i], (unsigned long long)lhist[i]);
   }
 }
+
+// Chunked scatter with per-block LDS range reservation: each block
+// histograms a chunk in LDS, reserves contiguous per-partition ranges
+// with ONE global atomic per touched partition, then scatters the chunk
+// (the chunk's keys re-read from L2).  Cuts global atomics from one per
+// row to one per (block, partition).
+#define SCATTER_CHUNK (BLOCK * 16)
 
 __global__ __launch_bounds__(BLOCK) void gb_part_scatter_kernel(
     const int64_t* __restrict__ keys,
@@ -354,17 +372,40 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_kernel(
     int n_aggs, int64_t n, int shift,
     int64_t* __restrict__ cursor,      // [P] exclusive offsets (mutated)
     int64_t* __restrict__ out_keys,
-    double* __restrict__ out_vals) {
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
-    int64_t key = keys[i];
-    int64_t p = (int64_t)(mix64((uint64_t)key) >> shift);
-    int64_t pos =
-        (int64_t)atomicAdd((unsigned long long*)&cursor[p], 1ULL);
-    out_keys[pos] = key;
-    for (int a = 0; a < n_aggs; ++a)
-      out_vals[(int64_t)a * n + pos] = vals[(int64_t)a * n + i];
+    double* __restrict__ out_vals,
+    int num_parts) {
+  __shared__ int lhist[PART_MAX];
+  __shared__ int64_t lbase[PART_MAX];
+  for (int64_t start = (int64_t)blockIdx.x * SCATTER_CHUNK; start < n;
+       start += (int64_t)gridDim.x * SCATTER_CHUNK) {
+    int64_t end = start + SCATTER_CHUNK;
+    if (end > n) end = n;
+    for (int i = threadIdx.x; i < num_parts; i += blockDim.x) lhist[i] = 0;
+    __syncthreads();
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+      int p = (int)(mix64((uint64_t)keys[i]) >> shift);
+      atomicAdd(&lhist[p], 1);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < num_parts; i += blockDim.x) {
+      int c = lhist[i];
+      lbase[i] =
+          c > 0
+              ? (int64_t)atomicAdd((unsigned long long*)&cursor[i],
+                                   (unsigned long long)c)
+              : 0;
+      lhist[i] = 0;
+    }
+    __syncthreads();
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+      int64_t key = keys[i];
+      int p = (int)(mix64((uint64_t)key) >> shift);
+      int64_t pos = lbase[p] + atomicAdd(&lhist[p], 1);
+      out_keys[pos] = key;
+      for (int a = 0; a < n_aggs; ++a)
+        out_vals[(int64_t)a * n + pos] = vals[(int64_t)a * n + i];
+    }
+    __syncthreads();
   }
 }
 
@@ -437,18 +478,21 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_kernel(
 extern "C" {
 
 void launch_gb_part_hist(const int64_t* keys, int64_t n, int shift,
-                         int64_t* hist, hipStream_t stream) {
+                         int64_t* hist, int num_parts, hipStream_t stream) {
   hipLaunchKernelGGL(gb_part_hist_kernel, dim3(grid_for(n, 4)), dim3(BLOCK),
-                     0, stream, keys, n, shift, hist);
+                     0, stream, keys, n, shift, hist, num_parts);
 }
 
 void launch_gb_part_scatter(const int64_t* keys, const double* vals,
                             int n_aggs, int64_t n, int shift, int64_t* cursor,
                             int64_t* out_keys, double* out_vals,
-                            hipStream_t stream) {
-  hipLaunchKernelGGL(gb_part_scatter_kernel, dim3(grid_for(n, 4)),
-                     dim3(BLOCK), 0, stream, keys, vals, n_aggs, n, shift,
-                     cursor, out_keys, out_vals);
+                            int num_parts, hipStream_t stream) {
+  int64_t blocks = (n + SCATTER_CHUNK - 1) / SCATTER_CHUNK;
+  if (blocks > MAX_GRID) blocks = MAX_GRID;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(gb_part_scatter_kernel, dim3((int)blocks), dim3(BLOCK),
+                     0, stream, keys, vals, n_aggs, n, shift, cursor,
+                     out_keys, out_vals, num_parts);
 }
 
 void launch_gb_aggregate_part(const int64_t* part_keys,
