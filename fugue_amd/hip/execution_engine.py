@@ -433,16 +433,10 @@ class HipExecutionEngine(ExecutionEngine):
             buckets = dops.rand_buckets(
                 hdf.count(), self.world_size, None, torch.device(self._device)
             )
-            hashes = buckets  # already bucket ids; reuse partition path
-            ext = dops.get_ext()
-            b32 = buckets.to(torch.int32)
-            counts = ext.bucket_histogram(b32, self.world_size)
-            offsets = torch.zeros(
-                self.world_size, dtype=torch.int64, device=b32.device
+            part, counts = dops.partition_by_bucket_ids(
+                hdf, buckets, self.world_size
             )
-            torch.cumsum(counts[:-1], 0, out=offsets[1:])
-            perm = ext.bucket_scatter(b32, offsets.clone())
-            return self._exchange(hdf.gather_rows(perm), counts)
+            return self._exchange(part, counts)
         if algo == "even":
             # equalize row counts across ranks
             local_n = hdf.count()
@@ -471,15 +465,11 @@ class HipExecutionEngine(ExecutionEngine):
                     device=torch.device(self._device),
                 ),
                 right=True,
-            ).to(torch.int32)
-            ext = dops.get_ext()
-            counts = ext.bucket_histogram(dest, self.world_size)
-            offsets = torch.zeros(
-                self.world_size, dtype=torch.int64, device=dest.device
             )
-            torch.cumsum(counts[:-1], 0, out=offsets[1:])
-            perm = ext.bucket_scatter(dest, offsets.clone())
-            return self._exchange(hdf.gather_rows(perm), counts)
+            part, counts = dops.partition_by_bucket_ids(
+                hdf, dest, self.world_size
+            )
+            return self._exchange(part, counts)
         return hdf
 
     def broadcast(self, df: DataFrame) -> DataFrame:

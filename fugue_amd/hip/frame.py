@@ -131,7 +131,10 @@ class DeviceColumn:
             np_arr = np_arr.astype("datetime64[D]").astype("int64")
         elif np_arr.dtype != _np_dtype_for(tp):
             np_arr = np_arr.astype(_np_dtype_for(tp))
-        data = torch.from_numpy(np.ascontiguousarray(np_arr)).to(device)
+        np_arr = np.ascontiguousarray(np_arr)
+        if not np_arr.flags.writeable:
+            np_arr = np_arr.copy()
+        data = torch.from_numpy(np_arr).to(device)
         return DeviceColumn(data, valid_t, tp)
 
     def to_arrow(self) -> pa.Array:

@@ -32,19 +32,84 @@ def _next_pow2(n: int) -> int:
     return p
 
 
+def _is_cpu(t: torch.Tensor) -> bool:
+    return t.device.type == "cpu"
+
+
+# ------------------------------------------------------------------ #
+# CPU equivalents (numpy uint64) of the kernel hash — used when frames
+# are CPU-resident (gloo multi-process tests; no GPU in CI).  The GPU
+# path always runs the HIP kernels; on a GPU host the extension is
+# REQUIRED (get_ext raises loudly if it failed to build).
+# ------------------------------------------------------------------ #
+def _np_mix64(x: "np.ndarray") -> "np.ndarray":
+    with np.errstate(over="ignore"):
+        x = x.astype(np.uint64, copy=True)
+        x ^= x >> np.uint64(33)
+        x *= np.uint64(0xFF51AFD7ED558CCD)
+        x ^= x >> np.uint64(33)
+        x *= np.uint64(0xC4CEB9FE1A85EC53)
+        x ^= x >> np.uint64(33)
+        return x
+
+
+def _np_hash_combine(h: "np.ndarray", v: "np.ndarray") -> "np.ndarray":
+    with np.errstate(over="ignore"):
+        return _np_mix64(
+            h
+            ^ (
+                v
+                + np.uint64(0x9E3779B97F4A7C15)
+                + (h << np.uint64(6))
+                + (h >> np.uint64(2))
+            )
+        )
+
+
+def _hash_rows_cpu(cols: Sequence[DeviceColumn]) -> torch.Tensor:
+    out: Optional[np.ndarray] = None
+    null_h = np.uint64(0x9E3779B97F4A7C15)
+    for c in cols:
+        raw = c.data.numpy()
+        if raw.dtype == np.bool_:
+            raw = raw.astype(np.uint8)
+        bits = raw.view(_unsigned_view_dtype(raw.dtype)).astype(np.uint64)
+        v = _np_mix64(bits)
+        if c.valid is not None:
+            v = np.where(c.valid.numpy(), v, null_h)
+        out = v if out is None else _np_hash_combine(out, v)
+    return torch.from_numpy(out.view(np.int64).copy())
+
+
+def _unsigned_view_dtype(dt: "np.dtype"):
+    m = {
+        np.dtype("int64"): np.uint64,
+        np.dtype("float64"): np.uint64,
+        np.dtype("int32"): np.uint32,
+        np.dtype("float32"): np.uint32,
+        np.dtype("int16"): np.uint16,
+        np.dtype("int8"): np.uint8,
+        np.dtype("uint8"): np.uint8,
+    }
+    return m[np.dtype(dt)]
+
+
 def hash_rows(cols: Sequence[DeviceColumn]) -> torch.Tensor:
     """Row-wise 64-bit hash over multiple columns (int64 tensor holding
     uint64 bits)."""
+    for c in cols:
+        if isinstance(c, StringDeviceColumn):
+            raise NotImplementedError(
+                "string partition keys are not yet supported on device"
+            )
+    if _is_cpu(cols[0].data):
+        return _hash_rows_cpu(cols)
     ext = get_ext()
     n = len(cols[0])
     device = cols[0].data.device
     out = torch.empty(n, dtype=torch.int64, device=device)
     first = True
     for c in cols:
-        if isinstance(c, StringDeviceColumn):
-            raise NotImplementedError(
-                "string partition keys are not yet supported on device"
-            )
         data = c.data
         if data.dtype == torch.int16:
             data = data.to(torch.int32)
@@ -61,12 +126,45 @@ def partition_by_hash(
     MI355X equivalent of Dask's ``hash_repartition``
     (``fugue_dask/_utils.py:44``).
     """
+    if _is_cpu(hashes):
+        h = hashes.numpy().view(np.uint64)
+        buckets_np = (h % np.uint64(num_buckets)).astype(np.int64)
+        counts_np = np.bincount(buckets_np, minlength=num_buckets)
+        perm_np = np.argsort(buckets_np, kind="stable")
+        return (
+            df.gather_rows(torch.from_numpy(perm_np)),
+            torch.from_numpy(counts_np.astype(np.int64)),
+        )
     ext = get_ext()
     buckets = ext.bucket_of(hashes, num_buckets)
     counts = ext.bucket_histogram(buckets, num_buckets)
     offsets = torch.zeros(num_buckets, dtype=torch.int64, device=hashes.device)
     torch.cumsum(counts[:-1], 0, out=offsets[1:])
     perm = ext.bucket_scatter(buckets, offsets.clone())
+    return df.gather_rows(perm), counts
+
+
+def partition_by_bucket_ids(
+    df: HipDataFrame, bucket_ids: torch.Tensor, num_buckets: int
+) -> Tuple[HipDataFrame, torch.Tensor]:
+    """Reorder rows into bucket-contiguous order given explicit bucket ids
+    (rand / even repartition)."""
+    if _is_cpu(bucket_ids):
+        b = bucket_ids.numpy().astype(np.int64)
+        counts_np = np.bincount(b, minlength=num_buckets)
+        perm_np = np.argsort(b, kind="stable")
+        return (
+            df.gather_rows(torch.from_numpy(perm_np)),
+            torch.from_numpy(counts_np.astype(np.int64)),
+        )
+    ext = get_ext()
+    b32 = bucket_ids.to(torch.int32)
+    counts = ext.bucket_histogram(b32, num_buckets)
+    offsets = torch.zeros(
+        num_buckets, dtype=torch.int64, device=bucket_ids.device
+    )
+    torch.cumsum(counts[:-1], 0, out=offsets[1:])
+    perm = ext.bucket_scatter(b32, offsets.clone())
     return df.gather_rows(perm), counts
 
 
@@ -184,7 +282,6 @@ def groupby_aggregate(
     Reference comparator: the groupby-aggregate SQL path
     (``fugue/execution/execution_engine.py:889`` + qpd/duckdb/dask-sql).
     """
-    ext = get_ext()
     n = df.count()
     device = df.col(keys[0]).data.device if keys else torch.device(df.device)
     key_cols = [df.col(k) for k in keys]
@@ -220,6 +317,9 @@ def groupby_aggregate(
         valids = None
         ops = torch.tensor([AGG_COUNT], dtype=torch.int32, device=device)
         n_aggs = 0
+    if _is_cpu(packed):
+        return _groupby_aggregate_cpu(packed, aggs, df, meta)
+    ext = get_ext()
     if expected_groups is None:
         # sample-based cardinality estimate
         if n > 65536:
@@ -251,6 +351,80 @@ def groupby_aggregate(
     return out_keys, out_aggs, out_count, meta
 
 
+def _groupby_aggregate_cpu(
+    packed: torch.Tensor,
+    aggs: List[Tuple[str, int, str]],
+    df: HipDataFrame,
+    meta: Optional[Dict[str, Any]],
+) -> Tuple[torch.Tensor, Dict[str, torch.Tensor], torch.Tensor, Optional[Dict[str, Any]]]:
+    import pandas as pd
+
+    data: Dict[str, Any] = {"__key": packed.numpy()}
+    for cname, op, oname in aggs:
+        c = df.col(cname)
+        v = c.data.numpy().astype("float64")
+        if c.valid is not None:
+            v = np.where(c.valid.numpy(), v, np.nan)
+        data[oname] = v
+    pdf = pd.DataFrame(data)
+    g = pdf.groupby("__key", sort=False)
+    out_count = g.size()
+    out_keys = torch.from_numpy(out_count.index.to_numpy().astype(np.int64))
+    count_t = torch.from_numpy(out_count.to_numpy().astype(np.int64))
+    out_aggs: Dict[str, torch.Tensor] = {}
+    for cname, op, oname in aggs:
+        if op == AGG_SUM:
+            s = g[oname].sum(min_count=0)
+        elif op == AGG_MIN:
+            s = g[oname].min()
+        elif op == AGG_MAX:
+            s = g[oname].max()
+        elif op == AGG_COUNT:
+            s = g[oname].count().astype("float64")
+        else:
+            raise FugueBug(f"op {op}")
+        out_aggs[oname] = torch.from_numpy(s.to_numpy().astype("float64"))
+    return out_keys, out_aggs, count_t, meta
+
+
+def _hash_join_indices_cpu(
+    probe_keys: torch.Tensor, build_keys: torch.Tensor, how: str
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    import pandas as pd
+
+    p = pd.DataFrame({"k": probe_keys.numpy()})
+    p["pi"] = np.arange(len(p), dtype=np.int64)
+    b = pd.DataFrame({"k": build_keys.numpy()})
+    b["bi"] = np.arange(len(b), dtype=np.int64)
+    if how == "inner":
+        m = p.merge(b, on="k", how="inner")
+        return (
+            torch.from_numpy(m["pi"].to_numpy()),
+            torch.from_numpy(m["bi"].to_numpy()),
+        )
+    if how == "left":
+        m = p.merge(b, on="k", how="left")
+        bi = m["bi"].fillna(-1).astype("int64")
+        return (
+            torch.from_numpy(m["pi"].to_numpy()),
+            torch.from_numpy(bi.to_numpy()),
+        )
+    if how == "semi":
+        m = p.merge(b.drop_duplicates("k"), on="k", how="inner")
+        return (
+            torch.from_numpy(m["pi"].to_numpy()),
+            torch.from_numpy(m["bi"].to_numpy()),
+        )
+    if how == "anti":
+        m = p.merge(b.drop_duplicates("k"), on="k", how="left")
+        m = m[m["bi"].isna()]
+        return (
+            torch.from_numpy(m["pi"].to_numpy()),
+            torch.from_numpy(np.full(len(m), -1, dtype=np.int64)),
+        )
+    raise FugueBug(f"unsupported join mode {how}")
+
+
 def hash_join_indices(
     probe_keys: torch.Tensor,
     build_keys: torch.Tensor,
@@ -263,6 +437,8 @@ def hash_join_indices(
     (replaces the reference's delegation to pandas/duckdb/spark joins,
     SURVEY.md §2.3 row "join ×9 types").
     """
+    if _is_cpu(probe_keys):
+        return _hash_join_indices_cpu(probe_keys, build_keys, how)
     ext = get_ext()
     nb = int(build_keys.numel())
     tsize = _next_pow2(max(16, nb * 2))
@@ -293,6 +469,10 @@ def hash_join_indices(
 def mark_matched_build_rows(
     probe_keys: torch.Tensor, build_keys: torch.Tensor
 ) -> torch.Tensor:
+    if _is_cpu(probe_keys):
+        bk = build_keys.numpy()
+        matched = np.isin(bk, np.unique(probe_keys.numpy()))
+        return torch.from_numpy(matched)
     ext = get_ext()
     nb = int(build_keys.numel())
     tsize = _next_pow2(max(16, nb * 2))
