@@ -68,10 +68,10 @@ _AGG_FUNC_TO_OP = {
 
 
 class HipSQLEngine(SQLEngine):
-    """SQL facet: parses with the built-in SQL parser; the relational plan
-    runs on the gathered local data via the pandas executor (the
-    device-translated plan path is used by the functional ops; SQL-plan →
-    device lowering is incremental)."""
+    """SQL facet: parses with the built-in SQL parser, lowers the plan to
+    engine relational ops (device kernels + RCCL shuffle) when the shape
+    is supported, and falls back to the pandas executor on gathered data
+    otherwise."""
 
     @property
     def dialect(self) -> Optional[str]:
@@ -82,13 +82,22 @@ class HipSQLEngine(SQLEngine):
         return self.execution_engine.is_distributed
 
     def select(self, dfs: Any, statement: StructuredRawSQL) -> DataFrame:
-        from fugue_amd.sql.executor import run_sql_on_pandas
+        from fugue_amd.sql.executor import parse_select, run_sql_on_pandas
+        from fugue_amd.sql.planner import UnsupportedPlan, execute_plan
 
         engine: HipExecutionEngine = self.execution_engine  # type: ignore
         sql = statement.construct(log=self.log)
+        tables = {k: engine.to_df(v) for k, v in dfs.items()}
+        try:
+            stmt = parse_select(sql)
+            return execute_plan(stmt, tables, engine)
+        except UnsupportedPlan as e:
+            self.log.debug("SQL plan fallback to pandas executor: %s", e)
+        except NotImplementedError as e:
+            self.log.debug("SQL plan fallback to pandas executor: %s", e)
         pdfs = {}
-        for k, v in dfs.items():
-            vv = engine.to_df(v)
+        for k, v in tables.items():
+            vv = v
             if isinstance(vv, HipDataFrame):
                 vv = engine._gather_all(vv)
             pdfs[k] = vv.as_pandas()

@@ -1,0 +1,300 @@
+"""SQL plan → ExecutionEngine lowering.
+
+Takes the parsed ``SelectStmt`` AST (``fugue_amd/sql/executor.py``) and
+executes it through ``ExecutionEngine`` relational ops (join / filter /
+select / aggregate / union / take), so a FugueSQL SELECT runs on the HIP
+engine's device kernels with the distributed shuffle — instead of
+gathering to pandas.  Raises :class:`UnsupportedPlan` for shapes outside
+the lowering (CASE/LIKE, non-equi joins, correlated subqueries, ...); the
+SQL facet then falls back to the pandas executor.
+"""
+from typing import Any, Dict, List, Optional, Tuple
+
+from fugue_amd.column import functions as F
+from fugue_amd.column.expressions import ColumnExpr, col, lit
+from fugue_amd.column.sql import SelectColumns
+from fugue_amd.dataframe.dataframe import DataFrame
+from fugue_amd.schema import Schema
+from fugue_amd.sql import executor as X
+
+
+class UnsupportedPlan(Exception):
+    pass
+
+
+def _to_column_expr(e: X.Expr, schema: Schema, alias_map: Dict[str, str]) -> ColumnExpr:
+    if isinstance(e, X.ColRef):
+        name = e.name
+        if name not in schema:
+            raise UnsupportedPlan(f"column {name} not in {schema}")
+        return col(name)
+    if isinstance(e, X.Lit):
+        v = e.value
+        if v is None or isinstance(v, (bool, int, float, str)):
+            return lit(v)
+        raise UnsupportedPlan(f"literal {v!r}")
+    if isinstance(e, X.BinOp):
+        l = _to_column_expr(e.left, schema, alias_map)
+        r = _to_column_expr(e.right, schema, alias_map)
+        op = e.op
+        if op == "AND":
+            return l & r
+        if op == "OR":
+            return l | r
+        m = {
+            "+": lambda: l + r,
+            "-": lambda: l - r,
+            "*": lambda: l * r,
+            "/": lambda: l / r,
+            "=": lambda: l == r,
+            "==": lambda: l == r,
+            "<>": lambda: l != r,
+            "!=": lambda: l != r,
+            "<": lambda: l < r,
+            "<=": lambda: l <= r,
+            ">": lambda: l > r,
+            ">=": lambda: l >= r,
+        }
+        if op not in m:
+            raise UnsupportedPlan(f"operator {op}")
+        return m[op]()
+    if isinstance(e, X.UnOp):
+        inner = _to_column_expr(e.operand, schema, alias_map)
+        if e.op == "-":
+            return -inner
+        if e.op == "NOT":
+            return ~inner
+        if e.op == "ISNULL":
+            return inner.is_null()
+        if e.op == "NOTNULL":
+            return inner.not_null()
+        raise UnsupportedPlan(f"unary {e.op}")
+    if isinstance(e, X.Between):
+        inner = _to_column_expr(e.expr, schema, alias_map)
+        lo = _to_column_expr(e.low, schema, alias_map)
+        hi = _to_column_expr(e.high, schema, alias_map)
+        res = (inner >= lo) & (inner <= hi)
+        return ~res if e.negate else res
+    if isinstance(e, X.InList):
+        inner = _to_column_expr(e.expr, schema, alias_map)
+        res: Optional[ColumnExpr] = None
+        for v in e.values:
+            term = inner == _to_column_expr(v, schema, alias_map)
+            res = term if res is None else (res | term)
+        if res is None:
+            raise UnsupportedPlan("empty IN list")
+        return ~res if e.negate else res
+    if isinstance(e, X.Cast):
+        inner = _to_column_expr(e.expr, schema, alias_map)
+        try:
+            return inner.cast(e.type_name)
+        except Exception:
+            raise UnsupportedPlan(f"cast to {e.type_name}")
+    if isinstance(e, X.FuncCall):
+        name = e.name
+        if name in ("SUM", "MIN", "MAX", "AVG", "COUNT", "FIRST", "LAST"):
+            if len(e.args) != 1:
+                raise UnsupportedPlan(f"{name} with {len(e.args)} args")
+            arg = e.args[0]
+            if isinstance(arg, X.Star):
+                if name != "COUNT":
+                    raise UnsupportedPlan(f"{name}(*)")
+                inner = col("*")
+            else:
+                inner = _to_column_expr(arg, schema, alias_map)
+            fn = {
+                "SUM": F.sum,
+                "MIN": F.min,
+                "MAX": F.max,
+                "AVG": F.avg,
+                "COUNT": F.count_distinct if e.distinct else F.count,
+                "FIRST": F.first,
+                "LAST": F.last,
+            }[name]
+            return fn(inner)
+        if name == "COALESCE":
+            return F.coalesce(
+                *[_to_column_expr(a, schema, alias_map) for a in e.args]
+            )
+        raise UnsupportedPlan(f"function {name}")
+    raise UnsupportedPlan(f"expression {type(e).__name__}")
+
+
+_JOIN_MAP = {
+    "inner": "inner",
+    "left": "left_outer",
+    "right": "right_outer",
+    "outer": "full_outer",
+    "cross": "cross",
+    "semi": "semi",
+    "anti": "anti",
+}
+
+
+def _extract_on_names(on: X.Expr) -> List[Tuple[str, str]]:
+    """ON must be a conjunction of colA = colB equalities."""
+    pairs: List[Tuple[str, str]] = []
+
+    def _walk(e: X.Expr) -> None:
+        if isinstance(e, X.BinOp) and e.op == "AND":
+            _walk(e.left)
+            _walk(e.right)
+            return
+        if (
+            isinstance(e, X.BinOp)
+            and e.op in ("=", "==")
+            and isinstance(e.left, X.ColRef)
+            and isinstance(e.right, X.ColRef)
+        ):
+            pairs.append((e.left.name, e.right.name))
+            return
+        raise UnsupportedPlan("non-equi join condition")
+
+    _walk(on)
+    return pairs
+
+
+def execute_plan(
+    stmt: X.SelectStmt, tables: Dict[str, DataFrame], engine: Any
+) -> DataFrame:
+    """Execute the statement on the engine; raises UnsupportedPlan."""
+    res = _execute_core(stmt, tables, engine)
+    for op, all_, rhs in stmt.set_ops:
+        other = _execute_core(rhs, tables, engine)
+        if op == "UNION":
+            res = engine.union(res, other, distinct=not all_)
+        elif op == "INTERSECT":
+            res = engine.intersect(res, other, distinct=True)
+        elif op == "EXCEPT":
+            res = engine.subtract(res, other, distinct=True)
+    if stmt.order_by or stmt.limit is not None:
+        res = _order_limit(res, stmt, engine)
+    return res
+
+
+def _order_limit(res: DataFrame, stmt: X.SelectStmt, engine: Any) -> DataFrame:
+    for o in stmt.order_by:
+        if not isinstance(o.expr, X.ColRef):
+            raise UnsupportedPlan("ORDER BY must use output columns")
+    if stmt.limit is not None and stmt.order_by:
+        presort = ",".join(
+            f"{o.expr.name} {'asc' if o.asc else 'desc'}" for o in stmt.order_by
+        )
+        return engine.take(res, stmt.limit, presort=presort)
+    # full sort (no limit) or limit without order: do it locally (results
+    # at this point are post-projection; a full global sort is a later
+    # optimization)
+    local = res.as_local_bounded() if hasattr(res, "as_local_bounded") else res
+    pdf = local.as_pandas()
+    if stmt.order_by:
+        pdf = pdf.sort_values(
+            [o.expr.name for o in stmt.order_by],
+            ascending=[o.asc for o in stmt.order_by],
+        ).reset_index(drop=True)
+    if stmt.limit is not None:
+        pdf = pdf.head(stmt.limit)
+    from fugue_amd.dataframe.pandas_dataframe import PandasDataFrame
+
+    return engine.to_df(PandasDataFrame(pdf, res.schema))
+
+
+def _resolve_from(item: X.FromItem, tables: Dict[str, DataFrame], engine: Any) -> DataFrame:
+    if item.subquery is not None:
+        return execute_plan(item.subquery, tables, engine)
+    if item.table not in tables:
+        raise UnsupportedPlan(f"table {item.table} not found")
+    return engine.to_df(tables[item.table])
+
+
+def _execute_core(
+    stmt: X.SelectStmt, tables: Dict[str, DataFrame], engine: Any
+) -> DataFrame:
+    if stmt.from_item is None:
+        raise UnsupportedPlan("SELECT without FROM")
+    res = _resolve_from(stmt.from_item, tables, engine)
+    for j in stmt.joins:
+        right = _resolve_from(j.item, tables, engine)
+        how = _JOIN_MAP.get(j.how)
+        if how is None:
+            raise UnsupportedPlan(f"join {j.how}")
+        if j.using is not None:
+            keys = list(j.using)
+        elif j.on is not None:
+            pairs = _extract_on_names(j.on)
+            keys = []
+            for a, b in pairs:
+                if a in res.schema and b in right.schema:
+                    if a != b:
+                        raise UnsupportedPlan(
+                            "join keys with different names need rename"
+                        )
+                    keys.append(a)
+                elif b in res.schema and a in right.schema:
+                    if a != b:
+                        raise UnsupportedPlan(
+                            "join keys with different names need rename"
+                        )
+                    keys.append(a)
+                else:
+                    raise UnsupportedPlan("can't resolve join key sides")
+        elif how == "cross":
+            keys = []
+        else:
+            raise UnsupportedPlan("JOIN without ON/USING")
+        common = [n for n in res.schema.names if n in right.schema]
+        if how != "cross" and set(keys) != set(common):
+            raise UnsupportedPlan(
+                f"join keys {keys} != common columns {common} "
+                "(fugue joins use all common columns)"
+            )
+        res = engine.join(res, right, how=how, on=keys if keys else None)
+    # select columns
+    schema = res.schema
+    where_expr: Optional[ColumnExpr] = None
+    if stmt.where is not None:
+        where_expr = _to_column_expr(stmt.where, schema, {})
+    cols: List[ColumnExpr] = []
+    from fugue_amd.column.expressions import all_cols
+
+    for i, (e, alias) in enumerate(stmt.columns):
+        if isinstance(e, X.Star):
+            if e.qualifier is not None:
+                raise UnsupportedPlan("qualified * in select")
+            cols.append(all_cols())
+            continue
+        ce = _to_column_expr(e, schema, {})
+        name = alias or e.default_name() or f"_col{i}"
+        if isinstance(e, X.ColRef) and alias is None:
+            cols.append(ce)
+        else:
+            cols.append(ce.alias(name))
+    having_expr: Optional[ColumnExpr] = None
+    sc = SelectColumns(*cols, arg_distinct=stmt.distinct)
+    # GROUP BY validation: keys must match the non-agg select columns
+    if stmt.group_by:
+        if not sc.has_agg:
+            raise UnsupportedPlan("GROUP BY without aggregates")
+        group_names = set()
+        for g in stmt.group_by:
+            if isinstance(g, X.ColRef):
+                group_names.add(g.name)
+            elif isinstance(g, X.Lit) and isinstance(g.value, int):
+                e0 = stmt.columns[g.value - 1][0]
+                if isinstance(e0, X.ColRef):
+                    group_names.add(e0.name)
+                else:
+                    raise UnsupportedPlan("positional GROUP BY on expression")
+            else:
+                raise UnsupportedPlan("GROUP BY expression")
+        sel_keys = {k.output_name or k.name for k in sc.group_keys}
+        if group_names != sel_keys:
+            raise UnsupportedPlan(
+                f"GROUP BY {group_names} != selected keys {sel_keys}"
+            )
+    if stmt.having is not None:
+        # having references output aggregate aliases
+        out_schema_names = [c.output_name for c in sc.all_cols]
+        having_expr = _to_column_expr(
+            stmt.having, Schema([(n, "double") for n in out_schema_names if n]), {}
+        )
+    return engine.select(res, sc, where=where_expr, having=having_expr)
