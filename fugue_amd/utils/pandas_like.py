@@ -70,27 +70,42 @@ def pandas_join(
     on: List[str],
 ) -> pd.DataFrame:
     """Join with fugue's 9 join types. ``how`` is normalized (see
-    ``fugue_amd.dataframe.utils.parse_join_type``)."""
+    ``fugue_amd.dataframe.utils.parse_join_type``).
+
+    SQL null semantics: null keys never match (pandas ``merge`` would
+    treat NaN keys as equal, so null-key rows are split out first)."""
     if how == "cross":
         d1 = df1.assign(__fugue_cross__=1)
         d2 = df2.assign(__fugue_cross__=1)
         res = d1.merge(d2, on="__fugue_cross__").drop(columns=["__fugue_cross__"])
         return res.reset_index(drop=True)
+    null1 = df1[on].isna().any(axis=1)
+    null2 = df2[on].isna().any(axis=1)
+    d1, d1n = df1[~null1], df1[null1]
+    d2, d2n = df2[~null2], df2[null2]
     if how in ("semi", "left_semi"):
-        keys = df2[on].drop_duplicates()
-        return df1.merge(keys, on=on, how="inner").reset_index(drop=True)
+        keys = d2[on].drop_duplicates()
+        return d1.merge(keys, on=on, how="inner").reset_index(drop=True)
     if how in ("anti", "left_anti"):
-        keys = df2[on].drop_duplicates().assign(__fugue_anti__=1)
-        res = df1.merge(keys, on=on, how="left")
+        keys = d2[on].drop_duplicates().assign(__fugue_anti__=1)
+        res = d1.merge(keys, on=on, how="left")
         res = res[res["__fugue_anti__"].isna()].drop(columns=["__fugue_anti__"])
-        return res.reset_index(drop=True)
+        return pd.concat([res, d1n], ignore_index=True)
     pd_how = {
         "inner": "inner",
         "left_outer": "left",
         "right_outer": "right",
         "full_outer": "outer",
     }[how]
-    return df1.merge(df2, on=on, how=pd_how).reset_index(drop=True)
+    res = d1.merge(d2, on=on, how=pd_how)
+    parts = [res]
+    if how in ("left_outer", "full_outer") and len(d1n) > 0:
+        parts.append(d1n)
+    if how in ("right_outer", "full_outer") and len(d2n) > 0:
+        parts.append(d2n)
+    if len(parts) > 1:
+        res = pd.concat(parts, ignore_index=True)
+    return res.reset_index(drop=True)
 
 
 def pandas_union(df1: pd.DataFrame, df2: pd.DataFrame, unique: bool) -> pd.DataFrame:
