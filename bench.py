@@ -68,9 +68,11 @@ def main() -> None:
         Schema("k:long,v:double"),
         engine.device,
     )
+    dim_gen = torch.Generator(device=device)
+    dim_gen.manual_seed(7)  # identical dims on every rank (broadcast table)
     dim_k = torch.arange(0, DIM_ROWS, dtype=torch.int64, device=device)
     dim_w = torch.rand(DIM_ROWS, dtype=torch.float64, device=device,
-                       generator=gen)
+                       generator=dim_gen)
     dims = HipDataFrame.from_columns(
         {
             "k": DeviceColumn(dim_k, None, pa.int64()),

@@ -281,7 +281,7 @@ class FugueSQLParser:
     # ------------------------------------------------------------------ #
     def _parse_schema(self) -> str:
         """Consume schema tokens (after SCHEMA keyword) as raw text until a
-        boundary keyword."""
+        boundary keyword or statement start."""
         ts = self.ts
         start = ts.peek().pos
         end = start
@@ -300,6 +300,15 @@ class FugueSQLParser:
                 break
             elif depth == 0 and t.kind == "PUNCT" and t.value == ";":
                 break
+            elif (
+                depth == 0
+                and t.kind == "NAME"
+                and self._starts_line(t)
+                and ts.peek(1) is not None
+                and ts.peek(1).kind == "OP"
+                and ts.peek(1).value == "="
+            ):
+                break  # next statement: NAME = ...
             ts.next()
             end = self._token_end(ts.pos - 1)
         return self.code[start:end].strip()

@@ -35,6 +35,8 @@ def _feed(obj: Any, parts: list) -> None:
         for x in obj:
             _feed(x, parts)
         parts.append("]")
+    elif _is_raw_frame(obj):
+        parts.append("\0D" + _frame_token(obj))
     elif isinstance(obj, set):
         parts.append("\0S[")
         for x in sorted(str(i) for i in obj):
@@ -53,6 +55,40 @@ def _feed(obj: Any, parts: list) -> None:
         parts.append("\0f" + mod + "." + name)
     else:
         parts.append("\0o" + repr(obj))
+
+
+_SMALL_FRAME_ROWS = 10_000
+
+
+def _is_raw_frame(obj: Any) -> bool:
+    cls = type(obj).__module__ + "." + type(obj).__name__
+    return cls in (
+        "pandas.core.frame.DataFrame",
+        "pyarrow.lib.Table",
+    )
+
+
+def _frame_token(obj: Any) -> str:
+    """Identity token for a raw pandas/arrow frame: content-based for
+    small frames (so identical literals dedup deterministically), id-based
+    for large ones (content hashing a 1e8-row frame is not acceptable —
+    such tasks are simply not deterministic across runs)."""
+    try:
+        n = len(obj)
+    except TypeError:  # pyarrow Table has num_rows
+        n = obj.num_rows
+    if n <= _SMALL_FRAME_ROWS:
+        try:
+            import pandas as pd
+
+            if isinstance(obj, pd.DataFrame):
+                return (
+                    "c" + str(list(obj.columns)) + repr(obj.values.tolist())
+                )
+            return "c" + str(obj.schema) + repr(obj.to_pylist())
+        except Exception:
+            pass
+    return "i" + str(id(obj))
 
 
 def to_uuid(*args: Any) -> str:

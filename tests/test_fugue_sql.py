@@ -1,0 +1,192 @@
+import os
+import tempfile
+from typing import Any, Dict, Iterable, List
+
+import numpy as np
+import pandas as pd
+import pytest
+
+import fugue_amd.api as fa
+from fugue_amd.exceptions import FugueSQLError, FugueSQLSyntaxError
+from fugue_amd.sql import FugueSQLWorkflow, fugue_sql, fugue_sql_flow
+
+
+def test_select_basic():
+    a = pd.DataFrame(dict(x=[1, 2, 3]))
+    res = fa.fugue_sql("SELECT x, x*2 AS x2 FROM a WHERE x > 1", a=a)
+    assert res.values.tolist() == [[2, 4], [3, 6]]
+
+
+def test_assignment_chain():
+    a = pd.DataFrame(dict(x=[1, 2, 3]))
+    res = fa.fugue_sql(
+        """
+        b = SELECT x FROM a WHERE x > 1
+        c = SELECT x+10 AS y FROM b
+        SELECT * FROM c
+        """,
+        a=a,
+    )
+    assert sorted(res["y"].tolist()) == [12, 13]
+
+
+def test_implicit_from():
+    res = fa.fugue_sql(
+        """
+        CREATE [[1],[2],[3]] SCHEMA x:int
+        SELECT * WHERE x >= 2
+        SELECT x+1 AS y
+        """
+    )
+    assert sorted(res["y"].tolist()) == [3, 4]
+
+
+def test_transform_prepartition():
+    a = pd.DataFrame(dict(g=["a", "a", "b"], v=[3, 1, 2]))
+
+    # schema: g:str,top:long
+    def top1(df: pd.DataFrame) -> pd.DataFrame:
+        return pd.DataFrame(dict(g=[df["g"].iloc[0]], top=[df["v"].iloc[0]]))
+
+    res = fa.fugue_sql(
+        "TRANSFORM a PREPARTITION BY g PRESORT v DESC USING top1",
+        a=a,
+        top1=top1,
+    )
+    assert sorted(res.values.tolist()) == [["a", 3], ["b", 2]]
+
+
+def test_take_sample_drop_rename_alter():
+    a = pd.DataFrame(dict(x=[3.0, 1.0, None], y=["a", "b", "c"]))
+    r1 = fa.fugue_sql("TAKE 2 ROWS FROM a PRESORT x", a=a)
+    assert len(r1) == 2
+    r2 = fa.fugue_sql("SAMPLE 2 ROWS FROM a", a=a)
+    assert len(r2) == 2
+    r3 = fa.fugue_sql("DROP ROWS IF ANY NULL FROM a", a=a)
+    assert len(r3) == 2
+    r4 = fa.fugue_sql("DROP COLUMNS y FROM a", a=a)
+    assert list(r4.columns) == ["x"]
+    r5 = fa.fugue_sql("RENAME COLUMNS x:xx FROM a", a=a)
+    assert list(r5.columns) == ["xx", "y"]
+    r6 = fa.fugue_sql("ALTER COLUMNS x:str FROM a", a=a)
+    assert r6["x"].iloc[0] == "3"
+    r7 = fa.fugue_sql("FILL NULLS (x=0) FROM a", a=a)
+    assert r7["x"].tolist() == [3.0, 1.0, 0.0]
+
+
+def test_save_and_load():
+    a = pd.DataFrame(dict(x=[1, 2]))
+    with tempfile.TemporaryDirectory() as tmp:
+        pq = os.path.join(tmp, "t.parquet")
+        fa.fugue_sql_flow(f"SAVE a OVERWRITE PARQUET '{pq}'", a=a).run()
+        res = fa.fugue_sql(f"LOAD '{pq}'")
+        assert len(res) == 2
+        csv = os.path.join(tmp, "t.csv")
+        fa.fugue_sql_flow(
+            f"SAVE a OVERWRITE CSV '{csv}' (header=true)", a=a
+        ).run()
+        res2 = fa.fugue_sql(
+            f"LOAD CSV '{csv}' (header=true, infer_schema=true)"
+        )
+        assert len(res2) == 2
+
+
+def test_yield_and_checkpoint():
+    dag = fugue_sql_flow(
+        """
+        a = CREATE [[1],[2]] SCHEMA x:int
+        b = SELECT x+1 AS y FROM a PERSIST
+        YIELD DATAFRAME AS out
+        """
+    )
+    res = dag.run()
+    assert sorted(r[0] for r in res["out"].result.as_array()) == [2, 3]
+
+
+def test_print_and_outtransform(capsys):
+    side: List[int] = []
+
+    def sink(rows: List[List[Any]]) -> None:
+        side.append(len(rows))
+
+    fa.fugue_sql_flow(
+        """
+        a = CREATE [[1],[2]] SCHEMA x:int
+        PRINT a TITLE 'hello'
+        OUTTRANSFORM a USING sink
+        """,
+        sink=sink,
+    ).run()
+    out = capsys.readouterr().out
+    assert "hello" in out
+    assert sum(side) == 2
+
+
+def test_zip_cotransform_sql():
+    a = pd.DataFrame(dict(k=[1, 2], v=[10, 20]))
+    b = pd.DataFrame(dict(k=[1, 1], w=[5.0, 6.0]))
+
+    # schema: k:long,n:long
+    def count_pair(df1: pd.DataFrame, df2: pd.DataFrame) -> Iterable[Dict[str, Any]]:
+        yield dict(k=int(df1["k"].iloc[0]), n=len(df2))
+
+    res = fa.fugue_sql(
+        """
+        z = ZIP a, b LEFT OUTER BY k
+        TRANSFORM z USING count_pair
+        """,
+        a=a,
+        b=b,
+        count_pair=count_pair,
+    )
+    assert sorted(res.values.tolist()) == [[1, 2], [2, 0]]
+
+
+def test_process_output_sql():
+    a = pd.DataFrame(dict(x=[1, 2]))
+    collected: List[int] = []
+
+    def doubler(df: pd.DataFrame) -> pd.DataFrame:
+        return df * 2
+
+    def collect(rows: List[List[Any]]) -> None:
+        collected.extend(r[0] for r in rows)
+
+    res = fa.fugue_sql(
+        """
+        b = PROCESS a USING doubler SCHEMA x:long
+        OUTPUT b USING collect
+        SELECT * FROM b
+        """,
+        a=a,
+        doubler=doubler,
+        collect=collect,
+    )
+    assert sorted(collected) == [2, 4]
+    assert sorted(res["x"].tolist()) == [2, 4]
+
+
+def test_jinja_template():
+    a = pd.DataFrame(dict(x=[1, 2, 3]))
+    res = fa.fugue_sql(
+        "SELECT * FROM a WHERE x > {{threshold}}", a=a, threshold=1
+    )
+    assert len(res) == 2
+
+
+def test_union_sql():
+    a = pd.DataFrame(dict(x=[1]))
+    b = pd.DataFrame(dict(x=[2]))
+    res = fa.fugue_sql("SELECT * FROM a UNION ALL SELECT * FROM b", a=a, b=b)
+    assert sorted(res["x"].tolist()) == [1, 2]
+
+
+def test_sql_caller_var_capture():
+    captured_df = pd.DataFrame(dict(q=[42]))
+    res = fugue_sql("SELECT * FROM captured_df")
+    assert res["q"].tolist() == [42]
+
+
+def test_sql_errors():
+    with pytest.raises((FugueSQLSyntaxError, FugueSQLError)):
+        fa.fugue_sql("NONSENSE STATEMENT HERE")
