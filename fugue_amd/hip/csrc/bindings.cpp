@@ -35,6 +35,13 @@ void launch_gb_part_scatter(const int64_t*, const double*, int, int64_t, int,
 void launch_gb_aggregate_part(const int64_t*, const double*, const int32_t*,
                               int, int64_t, const int64_t*, int64_t, int64_t*,
                               double*, int64_t*, int64_t, hipStream_t);
+void launch_gb_part_scatter_staged(const int64_t*, const double*, int64_t,
+                                   int, int64_t*, int64_t*, double*,
+                                   hipStream_t);
+void launch_gb_aggregate_part_big(const int64_t*, const double*,
+                                  const int32_t*, int64_t, const int64_t*,
+                                  int64_t, int64_t*, double*, int64_t*,
+                                  int64_t, hipStream_t);
 void launch_join_count(const int64_t*, int64_t, const int64_t*, const int32_t*,
                        const int32_t*, int64_t, int32_t*, hipStream_t);
 void launch_join_emit(const int64_t*, int64_t, const int64_t*, const int32_t*,
@@ -187,20 +194,36 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
   // phase 2: scatter into partitioned order
   auto pkeys = at::empty({n}, keys.options());
   auto pvals = at::empty({n_aggs, n}, vals.options());
-  launch_gb_part_scatter(keys.data_ptr<int64_t>(), vals.data_ptr<double>(),
-                         n_aggs, n, shift, cursor.data_ptr<int64_t>(),
-                         pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),
-                         (int)num_parts, stream);
+  bool staged = (n_aggs == 1 && num_parts == 512);
+  if (staged) {
+    launch_gb_part_scatter_staged(
+        keys.data_ptr<int64_t>(), vals.data_ptr<double>(), n, shift,
+        cursor.data_ptr<int64_t>(), pkeys.data_ptr<int64_t>(),
+        pvals.data_ptr<double>(), stream);
+  } else {
+    launch_gb_part_scatter(keys.data_ptr<int64_t>(), vals.data_ptr<double>(),
+                           n_aggs, n, shift, cursor.data_ptr<int64_t>(),
+                           pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),
+                           (int)num_parts, stream);
+  }
   // phase 3: per-partition LDS aggregation into the global table
   auto tkeys = at::full({tsize}, (int64_t)0x8000000000000000LL,
                         keys.options());
   auto gaggs = at::zeros({n_aggs, tsize}, vals.options());
   auto gcount = at::zeros({tsize}, keys.options());
-  launch_gb_aggregate_part(
-      pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),
-      ops.data_ptr<int32_t>(), n_aggs, n, offsets.data_ptr<int64_t>(),
-      num_parts, tkeys.data_ptr<int64_t>(), gaggs.data_ptr<double>(),
-      gcount.data_ptr<int64_t>(), tsize, stream);
+  if (staged) {
+    launch_gb_aggregate_part_big(
+        pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),
+        ops.data_ptr<int32_t>(), n, offsets.data_ptr<int64_t>(), num_parts,
+        tkeys.data_ptr<int64_t>(), gaggs.data_ptr<double>(),
+        gcount.data_ptr<int64_t>(), tsize, stream);
+  } else {
+    launch_gb_aggregate_part(
+        pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),
+        ops.data_ptr<int32_t>(), n_aggs, n, offsets.data_ptr<int64_t>(),
+        num_parts, tkeys.data_ptr<int64_t>(), gaggs.data_ptr<double>(),
+        gcount.data_ptr<int64_t>(), tsize, stream);
+  }
   return {tkeys, gaggs, gcount};
 }
 

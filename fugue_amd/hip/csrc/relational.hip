@@ -475,6 +475,170 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_kernel(
   }
 }
 
+// LDS write-staged scatter: stage the first STAGE_E rows of each
+// partition per chunk in LDS and flush them as contiguous runs, so the
+// bulk of the scattered writes leave the CU coalesced instead of as
+// isolated 8B transactions.  Overflow rows (partition skew beyond
+// STAGE_E within a chunk) are written directly.  Single-agg-column
+// variant (the common case after the COUNT-from-rowcount optimization).
+#define STAGE_P 512
+#define STAGE_E 8
+
+__global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
+    const int64_t* __restrict__ keys,
+    const double* __restrict__ vals,   // [1, n]
+    int64_t n, int shift,
+    int64_t* __restrict__ cursor,
+    int64_t* __restrict__ out_keys,
+    double* __restrict__ out_vals) {
+  __shared__ int lhist[STAGE_P];
+  __shared__ int64_t lbase[STAGE_P];
+  __shared__ int lcnt[STAGE_P];
+  __shared__ int64_t skey[STAGE_P * STAGE_E];
+  __shared__ double sval[STAGE_P * STAGE_E];
+  for (int64_t start = (int64_t)blockIdx.x * SCATTER_CHUNK; start < n;
+       start += (int64_t)gridDim.x * SCATTER_CHUNK) {
+    int64_t end = start + SCATTER_CHUNK;
+    if (end > n) end = n;
+    for (int i = threadIdx.x; i < STAGE_P; i += blockDim.x) {
+      lhist[i] = 0;
+      lcnt[i] = 0;
+    }
+    __syncthreads();
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+      int p = (int)(mix64((uint64_t)keys[i]) >> shift);
+      atomicAdd(&lhist[p], 1);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < STAGE_P; i += blockDim.x) {
+      int c = lhist[i];
+      lbase[i] =
+          c > 0
+              ? (int64_t)atomicAdd((unsigned long long*)&cursor[i],
+                                   (unsigned long long)c)
+              : 0;
+    }
+    __syncthreads();
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+      int64_t key = keys[i];
+      int p = (int)(mix64((uint64_t)key) >> shift);
+      int pos = atomicAdd(&lcnt[p], 1);
+      if (pos < STAGE_E) {
+        skey[p * STAGE_E + pos] = key;
+        sval[p * STAGE_E + pos] = vals[i];
+      } else {
+        int64_t gpos = lbase[p] + pos;
+        out_keys[gpos] = key;
+        out_vals[gpos] = vals[i];
+      }
+    }
+    __syncthreads();
+    // flush staged entries as contiguous runs
+    for (int t = threadIdx.x; t < STAGE_P * STAGE_E; t += blockDim.x) {
+      int p = t / STAGE_E;
+      int e = t % STAGE_E;
+      int c = lcnt[p];
+      if (e < c && e < STAGE_E) {
+        int64_t gpos = lbase[p] + e;
+        out_keys[gpos] = skey[t];
+        out_vals[gpos] = sval[t];
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// phase-3 variant with a 4096-slot LDS table (single agg column) for the
+// staged path's lower partition count (more groups per partition)
+#define LDS_SLOTS_BIG 4096
+
+__global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
+    const int64_t* __restrict__ part_keys,
+    const double* __restrict__ part_vals,  // [1, n]
+    const int32_t* __restrict__ ops,
+    int64_t n,
+    const int64_t* __restrict__ offsets,
+    int64_t num_parts,
+    int64_t* __restrict__ tkeys,
+    double* __restrict__ gaggs,
+    int64_t* __restrict__ gcount,
+    int64_t tsize) {
+  __shared__ int64_t lkeys[LDS_SLOTS_BIG];
+  __shared__ double laggs[LDS_SLOTS_BIG];
+  __shared__ long long lcount[LDS_SLOTS_BIG];
+  bool is_count = ops[0] == 3;
+  for (int64_t p = blockIdx.x; p < num_parts; p += gridDim.x) {
+    for (int i = threadIdx.x; i < LDS_SLOTS_BIG; i += blockDim.x) {
+      lkeys[i] = GB_EMPTY;
+      lcount[i] = 0;
+      laggs[i] = 0.0;
+    }
+    __syncthreads();
+    int64_t lo = offsets[p], hi = offsets[p + 1];
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+      int64_t key = part_keys[i];
+      uint64_t h = mix64((uint64_t)key);
+      int slot = (int)(h & (LDS_SLOTS_BIG - 1));
+      bool in_lds = false;
+      for (int probe = 0; probe < 32; ++probe) {
+        long long prev = (long long)atomicCAS(
+            (unsigned long long*)&lkeys[slot], (unsigned long long)GB_EMPTY,
+            (unsigned long long)key);
+        if (prev == GB_EMPTY || prev == key) { in_lds = true; break; }
+        slot = (slot + 1) & (LDS_SLOTS_BIG - 1);
+      }
+      double v = is_count ? 1.0 : part_vals[i];
+      if (in_lds) {
+        atomicAdd((unsigned long long*)&lcount[slot], 1ULL);
+        atomicAdd(&laggs[slot], v);
+      } else {
+        int64_t gslot = gb_probe_insert(key, tkeys, tsize);
+        atomicAdd((unsigned long long*)&gcount[gslot], 1ULL);
+        atomicAdd(&gaggs[gslot], v);
+      }
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < LDS_SLOTS_BIG; i += blockDim.x) {
+      int64_t key = lkeys[i];
+      if (key == GB_EMPTY) continue;
+      int64_t gslot = gb_probe_insert(key, tkeys, tsize);
+      atomicAdd((unsigned long long*)&gcount[gslot],
+                (unsigned long long)lcount[i]);
+      atomicAdd(&gaggs[gslot], laggs[i]);
+    }
+    __syncthreads();
+  }
+}
+
+extern "C" {
+
+void launch_gb_part_scatter_staged(const int64_t* keys, const double* vals,
+                                   int64_t n, int shift, int64_t* cursor,
+                                   int64_t* out_keys, double* out_vals,
+                                   hipStream_t stream) {
+  int64_t blocks = (n + SCATTER_CHUNK - 1) / SCATTER_CHUNK;
+  if (blocks > MAX_GRID) blocks = MAX_GRID;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(gb_part_scatter_staged_kernel, dim3((int)blocks),
+                     dim3(BLOCK), 0, stream, keys, vals, n, shift, cursor,
+                     out_keys, out_vals);
+}
+
+void launch_gb_aggregate_part_big(const int64_t* part_keys,
+                                  const double* part_vals,
+                                  const int32_t* ops, int64_t n,
+                                  const int64_t* offsets, int64_t num_parts,
+                                  int64_t* tkeys, double* gaggs,
+                                  int64_t* gcount, int64_t tsize,
+                                  hipStream_t stream) {
+  int grid = (int)std::min<int64_t>(num_parts, 8192);
+  hipLaunchKernelGGL(gb_aggregate_part_big_kernel, dim3(grid), dim3(BLOCK),
+                     0, stream, part_keys, part_vals, ops, n, offsets,
+                     num_parts, tkeys, gaggs, gcount, tsize);
+}
+
+}  // extern "C"
+
 extern "C" {
 
 void launch_gb_part_hist(const int64_t* keys, int64_t n, int shift,

@@ -333,7 +333,12 @@ def groupby_aggregate(
     if expected_groups > 100_000 and valids is None and sum_count_only:
         # high cardinality: 2-phase partitioned aggregation (hash-partition
         # rows so each partition's groups fit the per-workgroup LDS table)
-        num_parts = min(4096, _next_pow2(max(16, expected_groups // 512)))
+        # single-agg path with moderate cardinality uses the LDS
+        # write-staged scatter (512 parts, 4096-slot phase-3 table)
+        if len(aggs) <= 1 and expected_groups <= 1_500_000:
+            num_parts = 512
+        else:
+            num_parts = min(4096, _next_pow2(max(16, expected_groups // 512)))
         tkeys, gaggs, gcount = ext.gb_aggregate_partitioned(
             packed, vals, ops, num_parts, tsize
         )
