@@ -321,11 +321,17 @@ def groupby_aggregate(
         return _groupby_aggregate_cpu(packed, aggs, df, meta)
     ext = get_ext()
     if expected_groups is None:
-        # sample-based cardinality estimate
+        # sample-based distinct-count estimate (Chao83: D ≈ d + f1²/(2·f2),
+        # robust when the sample is mostly singletons — a linear scale-up
+        # would estimate ~n for any high-ish cardinality)
         if n > 65536:
             sample = packed[:: max(1, n // 65536)]
-            est = int(torch.unique(sample).numel())
-            expected_groups = min(n, est * max(1, n // max(1, sample.numel())))
+            _, counts = torch.unique(sample, return_counts=True)
+            d = int(counts.numel())
+            f1 = int((counts == 1).sum().item())
+            f2 = int((counts == 2).sum().item())
+            est = d + (f1 * f1) // max(2 * f2, 1)
+            expected_groups = max(d, min(n, est))
         else:
             expected_groups = max(1, n)
     tsize = _next_pow2(max(16, int(expected_groups * 2)))
