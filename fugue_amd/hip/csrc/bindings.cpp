@@ -39,9 +39,8 @@ void launch_gb_part_scatter_staged(const int64_t*, const double*, int64_t,
                                    int, int64_t*, int64_t*, double*,
                                    hipStream_t);
 void launch_gb_aggregate_part_big(const int64_t*, const double*,
-                                  const int32_t*, int64_t, const int64_t*,
-                                  int64_t, int64_t*, double*, int64_t*,
-                                  int64_t, hipStream_t);
+                                  const int32_t*, int64_t, int64_t*,
+                                  double*, int64_t*, int64_t, hipStream_t);
 void launch_join_count(const int64_t*, int64_t, const int64_t*, const int32_t*,
                        const int32_t*, int64_t, int32_t*, hipStream_t);
 void launch_join_emit(const int64_t*, int64_t, const int64_t*, const int32_t*,
@@ -214,9 +213,8 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
   if (staged) {
     launch_gb_aggregate_part_big(
         pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),
-        ops.data_ptr<int32_t>(), n, offsets.data_ptr<int64_t>(), num_parts,
-        tkeys.data_ptr<int64_t>(), gaggs.data_ptr<double>(),
-        gcount.data_ptr<int64_t>(), tsize, stream);
+        ops.data_ptr<int32_t>(), n, tkeys.data_ptr<int64_t>(),
+        gaggs.data_ptr<double>(), gcount.data_ptr<int64_t>(), tsize, stream);
   } else {
     launch_gb_aggregate_part(
         pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),

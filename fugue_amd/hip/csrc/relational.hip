@@ -548,34 +548,39 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
   }
 }
 
-// phase-3 variant with a 4096-slot LDS table (single agg column) for the
-// staged path's lower partition count (more groups per partition)
+// phase-3 variant for the staged path: after partitioning, ANY
+// contiguous chunk of rows spans at most a couple of partitions (so only
+// a few thousand distinct keys) — partition boundaries are irrelevant to
+// correctness (the LDS table just absorbs duplicates; flushes are
+// atomic).  Chunk-striding restores full grid parallelism regardless of
+// the partition count.  4096 slots / 80KB LDS → 2 blocks/CU.
 #define LDS_SLOTS_BIG 4096
+#define AGG_CHUNK (BLOCK * 256)  // 65536 rows per chunk
 
 __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
     const int64_t* __restrict__ part_keys,
     const double* __restrict__ part_vals,  // [1, n]
     const int32_t* __restrict__ ops,
     int64_t n,
-    const int64_t* __restrict__ offsets,
-    int64_t num_parts,
     int64_t* __restrict__ tkeys,
     double* __restrict__ gaggs,
     int64_t* __restrict__ gcount,
     int64_t tsize) {
   __shared__ int64_t lkeys[LDS_SLOTS_BIG];
   __shared__ double laggs[LDS_SLOTS_BIG];
-  __shared__ long long lcount[LDS_SLOTS_BIG];
+  __shared__ int lcount[LDS_SLOTS_BIG];
   bool is_count = ops[0] == 3;
-  for (int64_t p = blockIdx.x; p < num_parts; p += gridDim.x) {
+  for (int64_t start = (int64_t)blockIdx.x * AGG_CHUNK; start < n;
+       start += (int64_t)gridDim.x * AGG_CHUNK) {
+    int64_t end = start + AGG_CHUNK;
+    if (end > n) end = n;
     for (int i = threadIdx.x; i < LDS_SLOTS_BIG; i += blockDim.x) {
       lkeys[i] = GB_EMPTY;
       lcount[i] = 0;
       laggs[i] = 0.0;
     }
     __syncthreads();
-    int64_t lo = offsets[p], hi = offsets[p + 1];
-    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
       int64_t key = part_keys[i];
       uint64_t h = mix64((uint64_t)key);
       int slot = (int)(h & (LDS_SLOTS_BIG - 1));
@@ -589,7 +594,7 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
       }
       double v = is_count ? 1.0 : part_vals[i];
       if (in_lds) {
-        atomicAdd((unsigned long long*)&lcount[slot], 1ULL);
+        atomicAdd(&lcount[slot], 1);
         atomicAdd(&laggs[slot], v);
       } else {
         int64_t gslot = gb_probe_insert(key, tkeys, tsize);
@@ -627,14 +632,15 @@ void launch_gb_part_scatter_staged(const int64_t* keys, const double* vals,
 void launch_gb_aggregate_part_big(const int64_t* part_keys,
                                   const double* part_vals,
                                   const int32_t* ops, int64_t n,
-                                  const int64_t* offsets, int64_t num_parts,
                                   int64_t* tkeys, double* gaggs,
                                   int64_t* gcount, int64_t tsize,
                                   hipStream_t stream) {
-  int grid = (int)std::min<int64_t>(num_parts, 8192);
-  hipLaunchKernelGGL(gb_aggregate_part_big_kernel, dim3(grid), dim3(BLOCK),
-                     0, stream, part_keys, part_vals, ops, n, offsets,
-                     num_parts, tkeys, gaggs, gcount, tsize);
+  int64_t blocks = (n + AGG_CHUNK - 1) / AGG_CHUNK;
+  if (blocks > MAX_GRID) blocks = MAX_GRID;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(gb_aggregate_part_big_kernel, dim3((int)blocks),
+                     dim3(BLOCK), 0, stream, part_keys, part_vals, ops, n,
+                     tkeys, gaggs, gcount, tsize);
 }
 
 }  // extern "C"
