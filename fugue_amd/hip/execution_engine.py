@@ -180,7 +180,6 @@ class HipMapEngine(MapEngine):
                 bounds = dops.group_boundaries(sorted_keys).cpu().tolist()
                 n = sorted_df.count()
                 bounds.append(n)
-                pdf_all = sorted_df.as_pandas()
             except NotImplementedError:
                 # string keys etc.: host-side grouping
                 from fugue_amd.utils.pandas_like import safe_groupby_apply
@@ -208,14 +207,34 @@ class HipMapEngine(MapEngine):
                 return engine.to_df(
                     PandasDataFrame(out, output_schema), shard_replicated=False
                 )
-            for gi in range(len(bounds) - 1):
-                start, end = bounds[gi], bounds[gi + 1]
-                sub = pdf_all.iloc[start:end].reset_index(drop=True)
-                input_df = PandasDataFrame(
-                    sub, Schema(df.schema), pandas_df_wrapper=True
-                )
-                cursor.set(lambda: input_df.peek_array(), gi, 0)
-                results.append(map_func(cursor, input_df).as_pandas())
+            from fugue_amd.hip.staging import (
+                can_fast_stage,
+                staged_pandas_batches,
+            )
+
+            if can_fast_stage(sorted_df):
+                # pinned, double-buffered D2H: batch k+1 copies on a side
+                # stream while the UDF processes batch k
+                for g0, g1, batch in staged_pandas_batches(sorted_df, bounds):
+                    base = bounds[g0]
+                    for gi in range(g0, g1):
+                        start, end = bounds[gi] - base, bounds[gi + 1] - base
+                        sub = batch.iloc[start:end].reset_index(drop=True)
+                        input_df = PandasDataFrame(
+                            sub, Schema(df.schema), pandas_df_wrapper=True
+                        )
+                        cursor.set(lambda: input_df.peek_array(), gi, 0)
+                        results.append(map_func(cursor, input_df).as_pandas())
+            else:
+                pdf_all = sorted_df.as_pandas()
+                for gi in range(len(bounds) - 1):
+                    start, end = bounds[gi], bounds[gi + 1]
+                    sub = pdf_all.iloc[start:end].reset_index(drop=True)
+                    input_df = PandasDataFrame(
+                        sub, Schema(df.schema), pandas_df_wrapper=True
+                    )
+                    cursor.set(lambda: input_df.peek_array(), gi, 0)
+                    results.append(map_func(cursor, input_df).as_pandas())
         if len(results) == 0:
             out_pdf = output_schema.create_empty_pandas()
         else:
