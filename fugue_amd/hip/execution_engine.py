@@ -49,6 +49,7 @@ from fugue_amd.hip.expr import DeviceExprError, eval_device_expr, filter_mask
 from fugue_amd.hip.frame import (
     DeviceColumn,
     HipDataFrame,
+    SpilledDataFrame,
     StringDeviceColumn,
     supported_device_type,
 )
@@ -298,6 +299,8 @@ class HipExecutionEngine(ExecutionEngine):
         replicated) input is sharded into contiguous row ranges so the
         global dataframe equals the input; frames produced by this engine
         are already shards (``shard_replicated=False`` path)."""
+        if isinstance(df, SpilledDataFrame):
+            return df.restore()
         if isinstance(df, HipDataFrame):
             return df
         if isinstance(df, DataFrame):
@@ -465,6 +468,33 @@ class HipExecutionEngine(ExecutionEngine):
                 hdf, buckets, self.world_size
             )
             return self._exchange(part, counts)
+        if algo == "even" and len(keys) > 0:
+            # Dask parity (``even_repartition`` by distinct groups,
+            # fugue_dask/_utils.py:133): whole key-groups are assigned
+            # round-robin over the globally-sorted distinct keys
+            key_cols = [hdf.col(k) for k in keys]
+            try:
+                packed, _ = dops.pack_keys(key_cols)
+            except NotImplementedError:
+                return self._shuffle_by_columns(hdf, keys)
+            local_uniq = torch.unique(packed)
+            import torch.distributed as dist
+
+            gathered: List[Any] = [None] * self.world_size
+            dist.all_gather_object(gathered, local_uniq.cpu())
+            global_uniq = torch.unique(torch.cat([t for t in gathered]))
+            dev = torch.device(self._device)
+            global_uniq = global_uniq.to(dev)
+            ranks = (
+                torch.arange(global_uniq.numel(), device=dev)
+                % self.world_size
+            )
+            pos = torch.searchsorted(global_uniq, packed)
+            dest = ranks.index_select(0, pos)
+            part, counts = dops.partition_by_bucket_ids(
+                hdf, dest, self.world_size
+            )
+            return self._exchange(part, counts)
         if algo == "even":
             # equalize row counts across ranks
             local_n = hdf.count()
@@ -512,7 +542,21 @@ class HipExecutionEngine(ExecutionEngine):
         return out
 
     def persist(self, df: DataFrame, lazy: bool = False, **kwargs: Any) -> DataFrame:
+        """Materialize (device frames are already materialized).  With
+        ``storage="host"`` — or when the shard exceeds
+        ``fugue.hip.spill_threshold_bytes`` — the columns spill to pinned
+        host memory (288 GB HBM is the working set; persisted-but-cold
+        frames shouldn't hold it)."""
         res = self.to_df(df)
+        if isinstance(res, HipDataFrame):
+            storage = kwargs.get("storage", "device")
+            threshold = int(
+                self.conf.get("fugue.hip.spill_threshold_bytes", 0) or 0
+            )
+            if storage == "host" or (
+                threshold > 0 and res.num_bytes() > threshold
+            ):
+                res = SpilledDataFrame(res)
         if df.has_metadata:
             res.reset_metadata(df.metadata)
         return res
@@ -1008,21 +1052,40 @@ class HipExecutionEngine(ExecutionEngine):
             if presort is not None and presort != ""
             else partition_spec.presort
         )
-        local = self._as_local(d).as_pandas()
-        if len(_presort) > 0:
-            local = local.sort_values(
-                list(_presort.keys()),
-                ascending=list(_presort.values()),
-                na_position=na_position,
+
+        def _take_pdf(pdf: pd.DataFrame) -> pd.DataFrame:
+            if len(_presort) > 0:
+                pdf = pdf.sort_values(
+                    list(_presort.keys()),
+                    ascending=list(_presort.values()),
+                    na_position=na_position,
+                )
+            if len(partition_spec.partition_by) == 0:
+                pdf = pdf.head(n)
+            else:
+                pdf = pdf.groupby(
+                    partition_spec.partition_by, dropna=False, sort=False
+                ).head(n)
+            return pdf.reset_index(drop=True)
+
+        if self.is_distributed and isinstance(d, HipDataFrame):
+            # prune to local candidates first (any global top-n row is in
+            # some rank's local top-n), then gather only the candidates
+            candidates = _take_pdf(d.as_pandas())
+            gathered = self._gather_all(
+                HipDataFrame(
+                    PandasDataFrame(candidates, d.schema).as_arrow(),
+                    d.schema,
+                    device=self._device,
+                )
             )
-        if len(partition_spec.partition_by) == 0:
-            local = local.head(n)
-        else:
-            local = local.groupby(
-                partition_spec.partition_by, dropna=False, sort=False
-            ).head(n)
+            local = _take_pdf(gathered.as_pandas())
+            return self.to_df(
+                PandasDataFrame(local, d.schema), shard_replicated=True
+            )
+        local = _take_pdf(self._as_local(d).as_pandas())
         return self.to_df(
-            PandasDataFrame(local.reset_index(drop=True), d.schema),
+            PandasDataFrame(local, d.schema),
             shard_replicated=self.is_distributed,
         )
 

@@ -283,6 +283,112 @@ class StringDeviceColumn(DeviceColumn):
         return arr
 
 
+class SpilledDataFrame(LocalBoundedDataFrame):
+    """A persisted shard spilled from HBM to (pinned) host DRAM.
+
+    ``restore()`` re-uploads with ``hipMemcpyAsync`` (pinned source) to
+    rebuild the device frame; the engine's ``to_df`` restores
+    transparently when an op touches the frame."""
+
+    def __init__(self, src: "HipDataFrame"):
+        self._device = src.device
+        self._host_cols: Dict[str, DeviceColumn] = {}
+        pin = torch.cuda.is_available()
+
+        def _to_host(t: torch.Tensor) -> torch.Tensor:
+            if t.device.type == "cpu":
+                return t
+            host = torch.empty(t.shape, dtype=t.dtype, pin_memory=pin)
+            host.copy_(t, non_blocking=False)
+            return host
+
+        for name, c in src.columns_map.items():
+            valid = None if c.valid is None else _to_host(c.valid)
+            if isinstance(c, StringDeviceColumn):
+                self._host_cols[name] = StringDeviceColumn(
+                    _to_host(c.offsets), _to_host(c.bytes), valid
+                )
+            else:
+                self._host_cols[name] = DeviceColumn(
+                    _to_host(c.data), valid, c.pa_type
+                )
+        super().__init__(src.schema)
+
+    def restore(self) -> "HipDataFrame":
+        dev = self._device
+
+        def _up(t: torch.Tensor) -> torch.Tensor:
+            return t.to(dev, non_blocking=True)
+
+        cols: Dict[str, DeviceColumn] = {}
+        for name, c in self._host_cols.items():
+            valid = None if c.valid is None else _up(c.valid)
+            if isinstance(c, StringDeviceColumn):
+                cols[name] = StringDeviceColumn(
+                    _up(c.offsets), _up(c.bytes), valid
+                )
+            else:
+                cols[name] = DeviceColumn(_up(c.data), valid, c.pa_type)
+        if dev.startswith("cuda") and torch.cuda.is_available():
+            torch.cuda.synchronize()
+        return HipDataFrame.from_columns(cols, self.schema, dev)
+
+    # --- DataFrame interface (delegates to a temporary restore-less view) --
+    @property
+    def native(self) -> Dict[str, DeviceColumn]:
+        return self._host_cols
+
+    def native_as_df(self) -> "SpilledDataFrame":
+        return self
+
+    @property
+    def empty(self) -> bool:
+        return self.count() == 0
+
+    def count(self) -> int:
+        if len(self._host_cols) == 0:
+            return 0
+        return len(next(iter(self._host_cols.values())))
+
+    def _as_host_frame(self) -> "HipDataFrame":
+        return HipDataFrame.from_columns(self._host_cols, self.schema, "cpu")
+
+    def peek_array(self) -> List[Any]:
+        return self._as_host_frame().peek_array()
+
+    def as_arrow(self, type_safe: bool = False) -> pa.Table:
+        return self._as_host_frame().as_arrow()
+
+    def as_pandas(self) -> pd.DataFrame:
+        return self.as_arrow().to_pandas()
+
+    def as_local_bounded(self) -> LocalBoundedDataFrame:
+        from fugue_amd.dataframe.arrow_dataframe import ArrowDataFrame
+
+        return ArrowDataFrame(self.as_arrow())
+
+    def as_array(self, columns=None, type_safe: bool = False) -> List[Any]:
+        return self._as_host_frame().as_array(columns, type_safe=type_safe)
+
+    def as_array_iterable(self, columns=None, type_safe: bool = False):
+        yield from self.as_array(columns, type_safe=type_safe)
+
+    def _drop_cols(self, cols: List[str]) -> DataFrame:
+        return self._as_host_frame()._drop_cols(cols)
+
+    def _select_cols(self, cols: List[Any]) -> DataFrame:
+        return self._as_host_frame()._select_cols(cols)
+
+    def rename(self, columns: Dict[str, str]) -> DataFrame:
+        return self._as_host_frame().rename(columns)
+
+    def alter_columns(self, columns: Any) -> DataFrame:
+        return self._as_host_frame().alter_columns(columns)
+
+    def head(self, n: int, columns=None) -> LocalBoundedDataFrame:
+        return self._as_host_frame().head(n, columns)
+
+
 class HipDataFrame(LocalBoundedDataFrame):
     """A device-resident bounded frame (one shard; the distributed engine
     holds one HipDataFrame per rank)."""
