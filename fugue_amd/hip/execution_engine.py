@@ -54,6 +54,7 @@ from fugue_amd.hip.frame import (
     supported_device_type,
 )
 from fugue_amd.parallel.comm import Communicator, get_communicator
+from fugue_amd.utils.tracing import op_range
 from fugue_amd.schema import Schema
 from fugue_amd.utils.params import ParamDict
 
@@ -437,9 +438,10 @@ class HipExecutionEngine(ExecutionEngine):
         return self._comm.allreduce_sum(local) > 0
 
     def _shuffle_by_columns(self, df: HipDataFrame, keys: List[str]) -> HipDataFrame:
-        hashes = dops.hash_rows([df.col(k) for k in keys])
-        part, counts = dops.partition_by_hash(df, hashes, self.world_size)
-        return self._exchange(part, counts)
+        with op_range("fugue.shuffle.hash"):
+            hashes = dops.hash_rows([df.col(k) for k in keys])
+            part, counts = dops.partition_by_hash(df, hashes, self.world_size)
+            return self._exchange(part, counts)
 
     def _shuffle_by_tensor_key(
         self, df: HipDataFrame, keys: torch.Tensor
@@ -565,6 +567,16 @@ class HipExecutionEngine(ExecutionEngine):
     # joins                                                                #
     # ------------------------------------------------------------------ #
     def join(
+        self,
+        df1: DataFrame,
+        df2: DataFrame,
+        how: str,
+        on: Optional[List[str]] = None,
+    ) -> DataFrame:
+        with op_range(f"fugue.join.{how}"):
+            return self._join_impl(df1, df2, how, on)
+
+    def _join_impl(
         self,
         df1: DataFrame,
         df2: DataFrame,
@@ -1153,7 +1165,8 @@ class HipExecutionEngine(ExecutionEngine):
                 out_cols, Schema(fields), self._device
             )
         # aggregation path
-        return self._device_aggregate(d, cols, having)
+        with op_range("fugue.aggregate"):
+            return self._device_aggregate(d, cols, having)
 
     def _device_aggregate(
         self,

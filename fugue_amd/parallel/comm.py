@@ -79,6 +79,12 @@ class Communicator:
         bucket-contiguous, ordered by destination rank)."""
         if not self.is_distributed:
             return send
+        # NCCL has no bool dtype: exchange as uint8 and cast back
+        if send.dtype == torch.bool:
+            out = self.all_to_all_v(
+                send.to(torch.uint8), send_counts, recv_counts
+            )
+            return out.to(torch.bool)
         recv_total = sum(recv_counts)
         recv = torch.empty(
             recv_total, dtype=send.dtype, device=send.device
