@@ -41,14 +41,21 @@ void launch_gb_part_scatter_staged(const int64_t*, const double*, int64_t,
 void launch_gb_aggregate_part_big(const int64_t*, const double*,
                                   const int32_t*, int64_t, int64_t*,
                                   double*, int64_t*, int64_t, hipStream_t);
-void launch_join_count(const int64_t*, int64_t, const int64_t*, const int32_t*,
+void launch_join_count(const int64_t*, int64_t, const int64_t*,
+                       const int64_t*, const int64_t*, const int32_t*,
                        const int32_t*, int64_t, int32_t*, hipStream_t);
-void launch_join_emit(const int64_t*, int64_t, const int64_t*, const int32_t*,
-                      const int32_t*, int64_t, const int64_t*, int64_t*,
-                      int64_t*, int, hipStream_t);
+void launch_join_emit(const int64_t*, int64_t, const int64_t*, const int64_t*,
+                      const int64_t*, const int32_t*, const int32_t*, int64_t,
+                      const int64_t*, int64_t*, int64_t*, int, hipStream_t);
 void launch_join_mark_build(const int64_t*, int64_t, const int64_t*,
-                            const int32_t*, const int32_t*, int64_t, bool*,
-                            hipStream_t);
+                            const int64_t*, const int64_t*, const int32_t*,
+                            const int32_t*, int64_t, bool*, hipStream_t);
+void launch_hash_string_col(const int64_t*, const uint8_t*, const bool*,
+                            uint64_t*, int64_t, int, hipStream_t);
+void launch_hash_seed(uint64_t*, int64_t, uint64_t, hipStream_t);
+void launch_gb_mark_reps(const int64_t*, const int64_t*, const int64_t*,
+                         int64_t, int64_t*, int64_t*, int64_t*, int64_t,
+                         hipStream_t);
 }
 
 namespace {
@@ -239,14 +246,24 @@ std::vector<at::Tensor> join_build(at::Tensor keys, int64_t tsize) {
   return {heads, next};
 }
 
-at::Tensor join_count(at::Tensor pkeys, at::Tensor bkeys, at::Tensor heads,
+namespace {
+const int64_t* opt_i64_ptr(const c10::optional<at::Tensor>& v) {
+  if (!v.has_value()) return nullptr;
+  return v->data_ptr<int64_t>();
+}
+}  // namespace
+
+at::Tensor join_count(at::Tensor pkeys, at::Tensor bkeys,
+                      c10::optional<at::Tensor> ph2,
+                      c10::optional<at::Tensor> bh2, at::Tensor heads,
                       at::Tensor next, int64_t tsize) {
   check_gpu(pkeys, "pkeys");
   int64_t np = pkeys.numel();
   auto counts = at::zeros({np}, pkeys.options().dtype(at::kInt));
   if (np > 0) {
     launch_join_count(pkeys.data_ptr<int64_t>(), np,
-                      bkeys.data_ptr<int64_t>(), heads.data_ptr<int32_t>(),
+                      bkeys.data_ptr<int64_t>(), opt_i64_ptr(ph2),
+                      opt_i64_ptr(bh2), heads.data_ptr<int32_t>(),
                       next.data_ptr<int32_t>(), tsize,
                       counts.data_ptr<int32_t>(), current_stream());
   }
@@ -254,6 +271,8 @@ at::Tensor join_count(at::Tensor pkeys, at::Tensor bkeys, at::Tensor heads,
 }
 
 std::vector<at::Tensor> join_emit(at::Tensor pkeys, at::Tensor bkeys,
+                                  c10::optional<at::Tensor> ph2,
+                                  c10::optional<at::Tensor> bh2,
                                   at::Tensor heads, at::Tensor next,
                                   int64_t tsize, at::Tensor offsets,
                                   int64_t out_n, int64_t mode) {
@@ -263,6 +282,7 @@ std::vector<at::Tensor> join_emit(at::Tensor pkeys, at::Tensor bkeys,
   auto out_b = at::empty({out_n}, pkeys.options());
   if (np > 0 && out_n > 0) {
     launch_join_emit(pkeys.data_ptr<int64_t>(), np, bkeys.data_ptr<int64_t>(),
+                     opt_i64_ptr(ph2), opt_i64_ptr(bh2),
                      heads.data_ptr<int32_t>(), next.data_ptr<int32_t>(),
                      tsize, offsets.data_ptr<int64_t>(),
                      out_p.data_ptr<int64_t>(), out_b.data_ptr<int64_t>(),
@@ -272,18 +292,54 @@ std::vector<at::Tensor> join_emit(at::Tensor pkeys, at::Tensor bkeys,
 }
 
 at::Tensor join_mark_build(at::Tensor pkeys, at::Tensor bkeys,
-                           at::Tensor heads, at::Tensor next, int64_t tsize,
-                           int64_t n_build) {
+                           c10::optional<at::Tensor> ph2,
+                           c10::optional<at::Tensor> bh2, at::Tensor heads,
+                           at::Tensor next, int64_t tsize, int64_t n_build) {
   check_gpu(pkeys, "pkeys");
   auto matched = at::zeros({n_build}, pkeys.options().dtype(at::kBool));
   if (pkeys.numel() > 0 && n_build > 0) {
     launch_join_mark_build(pkeys.data_ptr<int64_t>(), pkeys.numel(),
-                           bkeys.data_ptr<int64_t>(),
-                           heads.data_ptr<int32_t>(),
+                           bkeys.data_ptr<int64_t>(), opt_i64_ptr(ph2),
+                           opt_i64_ptr(bh2), heads.data_ptr<int32_t>(),
                            next.data_ptr<int32_t>(), tsize,
                            matched.data_ptr<bool>(), current_stream());
   }
   return matched;
+}
+
+void hash_string_column(at::Tensor offsets, at::Tensor bytes,
+                        c10::optional<at::Tensor> valid, at::Tensor out,
+                        bool is_first) {
+  check_gpu(offsets, "offsets");
+  check_gpu(out, "out");
+  int64_t n = out.numel();
+  launch_hash_string_col(
+      offsets.data_ptr<int64_t>(), bytes.data_ptr<uint8_t>(),
+      opt_valid_ptr(valid),
+      reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>()), n, is_first,
+      current_stream());
+}
+
+void hash_seed(at::Tensor out, int64_t seed) {
+  check_gpu(out, "out");
+  launch_hash_seed(reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>()),
+                   out.numel(), (uint64_t)seed, current_stream());
+}
+
+std::vector<at::Tensor> gb_mark_reps(at::Tensor h1, at::Tensor h2,
+                                     at::Tensor tkeys, int64_t tsize) {
+  check_gpu(h1, "h1");
+  auto rep = at::full({tsize}, -1, h1.options());
+  auto th2 = at::full({tsize}, (int64_t)0x8000000000000000LL, h1.options());
+  auto conflict = at::zeros({1}, h1.options());
+  if (h1.numel() > 0) {
+    launch_gb_mark_reps(h1.data_ptr<int64_t>(), h2.data_ptr<int64_t>(),
+                        tkeys.data_ptr<int64_t>(), tsize,
+                        rep.data_ptr<int64_t>(), th2.data_ptr<int64_t>(),
+                        conflict.data_ptr<int64_t>(), h1.numel(),
+                        current_stream());
+  }
+  return {rep, th2, conflict};
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -297,6 +353,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gb_aggregate_partitioned", &gb_aggregate_partitioned,
         "partitioned (2-phase) hash group-by aggregation");
   m.def("join_build", &join_build, "build chained hash table");
+  m.def("hash_string_column", &hash_string_column,
+        "combine a string column into the running row hash");
+  m.def("hash_seed", &hash_seed, "seed a row-hash buffer");
+  m.def("gb_mark_reps", &gb_mark_reps,
+        "representative rows + h2 verification for hashed group-by");
   m.def("join_count", &join_count, "count matches per probe row");
   m.def("join_emit", &join_emit, "emit join pairs");
   m.def("join_mark_build", &join_mark_build, "mark matched build rows");

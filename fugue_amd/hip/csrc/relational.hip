@@ -718,10 +718,13 @@ __global__ __launch_bounds__(BLOCK) void join_build_kernel(
   }
 }
 
-// pass 1: per-probe-row match count
+// pass 1: per-probe-row match count.  ph2/bh2 (optional) carry a second
+// independent 64-bit hash for string keys: a match requires both hashes
+// equal (128-bit verification).
 __global__ __launch_bounds__(BLOCK) void join_count_kernel(
     const int64_t* __restrict__ pkeys, int64_t np,
     const int64_t* __restrict__ bkeys,
+    const int64_t* __restrict__ ph2, const int64_t* __restrict__ bh2,
     const int32_t* __restrict__ heads, const int32_t* __restrict__ next,
     int64_t tsize, int32_t* __restrict__ counts) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -732,7 +735,7 @@ __global__ __launch_bounds__(BLOCK) void join_count_kernel(
     int32_t cur = heads[(int64_t)(h & (uint64_t)(tsize - 1))];
     int32_t c = 0;
     while (cur >= 0) {
-      if (bkeys[cur] == key) ++c;
+      if (bkeys[cur] == key && (ph2 == nullptr || bh2[cur] == ph2[i])) ++c;
       cur = next[cur];
     }
     counts[i] = c;
@@ -744,6 +747,7 @@ __global__ __launch_bounds__(BLOCK) void join_count_kernel(
 __global__ __launch_bounds__(BLOCK) void join_emit_kernel(
     const int64_t* __restrict__ pkeys, int64_t np,
     const int64_t* __restrict__ bkeys,
+    const int64_t* __restrict__ ph2, const int64_t* __restrict__ bh2,
     const int32_t* __restrict__ heads, const int32_t* __restrict__ next,
     int64_t tsize, const int64_t* __restrict__ offsets,
     int64_t* __restrict__ out_p, int64_t* __restrict__ out_b, int mode) {
@@ -756,7 +760,7 @@ __global__ __launch_bounds__(BLOCK) void join_emit_kernel(
     int64_t pos = offsets[i];
     bool any = false;
     while (cur >= 0) {
-      if (bkeys[cur] == key) {
+      if (bkeys[cur] == key && (ph2 == nullptr || bh2[cur] == ph2[i])) {
         any = true;
         if (mode == 0 || mode == 1) {
           out_p[pos] = i;
@@ -784,6 +788,7 @@ __global__ __launch_bounds__(BLOCK) void join_emit_kernel(
 __global__ __launch_bounds__(BLOCK) void join_mark_build_kernel(
     const int64_t* __restrict__ pkeys, int64_t np,
     const int64_t* __restrict__ bkeys,
+    const int64_t* __restrict__ ph2, const int64_t* __restrict__ bh2,
     const int32_t* __restrict__ heads, const int32_t* __restrict__ next,
     int64_t tsize, bool* __restrict__ bmatched) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -793,7 +798,8 @@ __global__ __launch_bounds__(BLOCK) void join_mark_build_kernel(
     uint64_t h = mix64((uint64_t)key);
     int32_t cur = heads[(int64_t)(h & (uint64_t)(tsize - 1))];
     while (cur >= 0) {
-      if (bkeys[cur] == key) bmatched[cur] = true;
+      if (bkeys[cur] == key && (ph2 == nullptr || bh2[cur] == ph2[i]))
+        bmatched[cur] = true;
       cur = next[cur];
     }
   }
@@ -808,28 +814,137 @@ void launch_join_build(const int64_t* keys, int64_t n, int32_t* heads,
 }
 
 void launch_join_count(const int64_t* pkeys, int64_t np, const int64_t* bkeys,
+                       const int64_t* ph2, const int64_t* bh2,
                        const int32_t* heads, const int32_t* next,
                        int64_t tsize, int32_t* counts, hipStream_t stream) {
   hipLaunchKernelGGL(join_count_kernel, dim3(grid_for(np)), dim3(BLOCK), 0,
-                     stream, pkeys, np, bkeys, heads, next, tsize, counts);
+                     stream, pkeys, np, bkeys, ph2, bh2, heads, next, tsize,
+                     counts);
 }
 
 void launch_join_emit(const int64_t* pkeys, int64_t np, const int64_t* bkeys,
+                      const int64_t* ph2, const int64_t* bh2,
                       const int32_t* heads, const int32_t* next, int64_t tsize,
                       const int64_t* offsets, int64_t* out_p, int64_t* out_b,
                       int mode, hipStream_t stream) {
   hipLaunchKernelGGL(join_emit_kernel, dim3(grid_for(np)), dim3(BLOCK), 0,
-                     stream, pkeys, np, bkeys, heads, next, tsize, offsets,
-                     out_p, out_b, mode);
+                     stream, pkeys, np, bkeys, ph2, bh2, heads, next, tsize,
+                     offsets, out_p, out_b, mode);
 }
 
 void launch_join_mark_build(const int64_t* pkeys, int64_t np,
-                            const int64_t* bkeys, const int32_t* heads,
+                            const int64_t* bkeys, const int64_t* ph2,
+                            const int64_t* bh2, const int32_t* heads,
                             const int32_t* next, int64_t tsize, bool* bmatched,
                             hipStream_t stream) {
   hipLaunchKernelGGL(join_mark_build_kernel, dim3(grid_for(np)), dim3(BLOCK),
-                     0, stream, pkeys, np, bkeys, heads, next, tsize,
-                     bmatched);
+                     0, stream, pkeys, np, bkeys, ph2, bh2, heads, next,
+                     tsize, bmatched);
+}
+
+}  // extern "C"
+
+// ------------------------------------------------------------------ //
+// string hashing: per-row xx-style mix over the UTF-8 bytes           //
+// (offsets int64 [n+1], bytes uint8).  One thread per row; fine for   //
+// typical short keys — wave-cooperative long-string variant later.    //
+// ------------------------------------------------------------------ //
+__device__ __forceinline__ uint64_t hash_bytes(
+    const uint8_t* __restrict__ data, int64_t len) {
+  uint64_t h = 0x27d4eb2f165667c5ULL ^ (uint64_t)len;
+  int64_t i = 0;
+  for (; i + 8 <= len; i += 8) {
+    uint64_t k;
+    memcpy(&k, data + i, 8);
+    h = hash_combine(h, k);
+  }
+  uint64_t tail = 0;
+  for (int64_t j = i; j < len; ++j) tail = (tail << 8) | data[j];
+  if (i < len) h = hash_combine(h, tail);
+  return mix64(h);
+}
+
+__global__ __launch_bounds__(BLOCK) void hash_string_col_kernel(
+    const int64_t* __restrict__ offsets,
+    const uint8_t* __restrict__ bytes,
+    const bool* __restrict__ valid,
+    uint64_t* __restrict__ out,
+    int64_t n,
+    int is_first) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    uint64_t v;
+    if (valid != nullptr && !valid[i]) {
+      v = 0x9e3779b97f4a7c15ULL;
+    } else {
+      int64_t lo = offsets[i];
+      v = hash_bytes(bytes + lo, offsets[i + 1] - lo);
+    }
+    out[i] = is_first ? v : hash_combine(out[i], v);
+  }
+}
+
+// seed an output hash buffer (for independent second hashes)
+__global__ __launch_bounds__(BLOCK) void hash_seed_kernel(
+    uint64_t* __restrict__ out, int64_t n, uint64_t seed) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    out[i] = mix64(seed + (uint64_t)1);
+}
+
+// representative-row + second-hash verification pass for hashed
+// (string-keyed) group-by: probes the table built by gb_aggregate over
+// h1, records one representative row per slot and flags h2 mismatches
+// (h1 collision between distinct keys → caller falls back).
+__global__ __launch_bounds__(BLOCK) void gb_mark_reps_kernel(
+    const int64_t* __restrict__ h1,
+    const int64_t* __restrict__ h2,
+    const int64_t* __restrict__ tkeys,
+    int64_t tsize,
+    int64_t* __restrict__ rep,      // [tsize] init -1
+    int64_t* __restrict__ th2,      // [tsize] init GB_EMPTY
+    int64_t* __restrict__ conflict, // [1]
+    int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t key = h1[i];
+    uint64_t h = mix64((uint64_t)key);
+    int64_t slot = (int64_t)(h & (uint64_t)(tsize - 1));
+    while (tkeys[slot] != key) slot = (slot + 1) & (tsize - 1);
+    atomicCAS((unsigned long long*)&rep[slot],
+              (unsigned long long)(int64_t)-1, (unsigned long long)i);
+    long long prev = (long long)atomicCAS((unsigned long long*)&th2[slot],
+                                          (unsigned long long)GB_EMPTY,
+                                          (unsigned long long)h2[i]);
+    if (prev != GB_EMPTY && prev != h2[i])
+      atomicAdd((unsigned long long*)conflict, 1ULL);
+  }
+}
+
+extern "C" {
+
+void launch_hash_string_col(const int64_t* offsets, const uint8_t* bytes,
+                            const bool* valid, uint64_t* out, int64_t n,
+                            int is_first, hipStream_t stream) {
+  hipLaunchKernelGGL(hash_string_col_kernel, dim3(grid_for(n)), dim3(BLOCK),
+                     0, stream, offsets, bytes, valid, out, n, is_first);
+}
+
+void launch_hash_seed(uint64_t* out, int64_t n, uint64_t seed,
+                      hipStream_t stream) {
+  hipLaunchKernelGGL(hash_seed_kernel, dim3(grid_for(n)), dim3(BLOCK), 0,
+                     stream, out, n, seed);
+}
+
+void launch_gb_mark_reps(const int64_t* h1, const int64_t* h2,
+                         const int64_t* tkeys, int64_t tsize, int64_t* rep,
+                         int64_t* th2, int64_t* conflict, int64_t n,
+                         hipStream_t stream) {
+  hipLaunchKernelGGL(gb_mark_reps_kernel, dim3(grid_for(n)), dim3(BLOCK), 0,
+                     stream, h1, h2, tkeys, tsize, rep, th2, conflict, n);
 }
 
 }  // extern "C"

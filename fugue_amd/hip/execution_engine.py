@@ -591,11 +591,6 @@ class HipExecutionEngine(ExecutionEngine):
             isinstance(d1, HipDataFrame)
             and isinstance(d2, HipDataFrame)
             and how != "cross"
-            and all(
-                not isinstance(d1.col(k), StringDeviceColumn)
-                and not isinstance(d2.col(k), StringDeviceColumn)
-                for k in key_schema.names
-            )
         ):
             try:
                 return self._device_join(
@@ -618,6 +613,25 @@ class HipExecutionEngine(ExecutionEngine):
         if isinstance(df, HipDataFrame) and self.is_distributed:
             return self._gather_all(df)
         return df.as_local()
+
+    def _join_keys(
+        self, d1: HipDataFrame, d2: HipDataFrame, keys: List[str]
+    ) -> Tuple[torch.Tensor, torch.Tensor, Optional[torch.Tensor], Optional[torch.Tensor]]:
+        """Comparable join keys for both sides: exact packed int64 when
+        the key tuple fits 63 bits; otherwise 128-bit hashed keys
+        (h1 compared in the table, h2 verified — string keys)."""
+        try:
+            k1, k2 = self._shared_key_pack(d1, d2, keys)
+            return k1, k2, None, None
+        except NotImplementedError:
+            pass
+        c1 = [d1.col(k) for k in keys]
+        c2 = [d2.col(k) for k in keys]
+        k1 = dops.hash_rows(c1)
+        k2 = dops.hash_rows(c2)
+        h21 = dops.hash_rows(c1, seed=dops._H2_SEED)
+        h22 = dops.hash_rows(c2, seed=dops._H2_SEED)
+        return k1, k2, h21, h22
 
     def _shared_key_pack(
         self, d1: HipDataFrame, d2: HipDataFrame, keys: List[str]
@@ -702,11 +716,11 @@ class HipExecutionEngine(ExecutionEngine):
                 # with d1 replicated and d2 sharded, the local joins union
                 # to the right answer for inner joins
             else:
-                k1t, _ = self._shared_key_pack(d1v, d2v, keys)
+                k1t, _, _, _ = self._join_keys(d1v, d1v, keys)
                 d1v = self._shuffle_by_tensor_key(d1v, k1t)
-                k2t2, _ = self._shared_key_pack(d2v, d2v, keys)
+                k2t2, _, _, _ = self._join_keys(d2v, d2v, keys)
                 d2v = self._shuffle_by_tensor_key(d2v, k2t2)
-        k1, k2 = self._shared_key_pack(d1v, d2v, keys)
+        k1, k2, h21, h22 = self._join_keys(d1v, d2v, keys)
         if how in ("inner", "left_outer", "semi", "anti"):
             mode = {
                 "inner": "inner",
@@ -714,7 +728,7 @@ class HipExecutionEngine(ExecutionEngine):
                 "semi": "semi",
                 "anti": "anti",
             }[how]
-            pi, bi = dops.hash_join_indices(k1, k2, mode)
+            pi, bi = dops.hash_join_indices(k1, k2, mode, h21, h22)
             res = self._emit_join_output(
                 d1v, d2v, pi, bi, keys, output_schema, probe_is_left=True
             )
@@ -723,7 +737,7 @@ class HipExecutionEngine(ExecutionEngine):
                 res = res.concat_with([nulls])
             return res
         if how == "right_outer":
-            pi, bi = dops.hash_join_indices(k2, k1, "left")
+            pi, bi = dops.hash_join_indices(k2, k1, "left", h22, h21)
             res = self._emit_join_output(
                 d2v, d1v, pi, bi, keys, output_schema, probe_is_left=False
             )
@@ -732,11 +746,11 @@ class HipExecutionEngine(ExecutionEngine):
                 res = res.concat_with([nulls])
             return res
         if how == "full_outer":
-            pi, bi = dops.hash_join_indices(k1, k2, "left")
+            pi, bi = dops.hash_join_indices(k1, k2, "left", h21, h22)
             res = self._emit_join_output(
                 d1v, d2v, pi, bi, keys, output_schema, probe_is_left=True
             )
-            matched = dops.mark_matched_build_rows(k1, k2)
+            matched = dops.mark_matched_build_rows(k1, k2, h21, h22)
             un = (~matched).nonzero(as_tuple=True)[0]
             parts: List[HipDataFrame] = []
             if un.numel() > 0:
@@ -1236,9 +1250,13 @@ class HipExecutionEngine(ExecutionEngine):
         if len(key_names) == 0:
             raise DeviceExprError("global aggregate: fallback")
         key_cols = [d.col(k) for k in key_names]
-        for kc in key_cols:
-            if isinstance(kc, StringDeviceColumn):
-                raise DeviceExprError("string group keys: fallback")
+        hashed_keys = any(
+            isinstance(kc, StringDeviceColumn) for kc in key_cols
+        )
+        if hashed_keys:
+            return self._device_aggregate_hashed(
+                d, key_names, partials, plans, having, cols
+            )
         # local partial aggregation
         out_keys, out_aggs, out_count, meta = dops.groupby_aggregate(
             d, key_names, partials
@@ -1276,6 +1294,68 @@ class HipExecutionEngine(ExecutionEngine):
                 tmp_s, tmp_c = info
                 cnt = out_aggs[tmp_c]
                 vals = out_aggs[tmp_s] / torch.clamp(cnt, min=1.0)
+                out_cols[name] = DeviceColumn(vals, None, pa.float64())
+                fields.append(pa.field(name, pa.float64()))
+        res = HipDataFrame.from_columns(out_cols, Schema(fields), self._device)
+        if having is not None:
+            mask = filter_mask(having, res)
+            res = res.gather_rows(mask.nonzero(as_tuple=True)[0])
+        if cols.is_distinct:
+            raise DeviceExprError("distinct aggregate: fallback")
+        return res
+
+    def _device_aggregate_hashed(
+        self,
+        d: HipDataFrame,
+        key_names: List[str],
+        partials: List[Tuple[str, int, str]],
+        plans: List[Tuple[str, str, Any]],
+        having: Optional[ColumnExpr],
+        cols: SelectColumns,
+    ) -> DataFrame:
+        """Group-by on string (or otherwise unpackable) key tuples: the
+        distributed case co-shuffles full rows by key hash first, so the
+        local 128-bit-hashed aggregation is globally exact."""
+        if self.is_distributed:
+            d = self._shuffle_by_columns(d, key_names)
+        try:
+            reps, out_aggs, out_count = dops.groupby_aggregate_hashed(
+                d, key_names, partials
+            )
+        except dops.HashCollisionError:
+            raise DeviceExprError("hash collision on keys: exact fallback")
+        rep_frame = d.gather_rows(reps.to(torch.device(self._device)))
+        out_cols: Dict[str, DeviceColumn] = {}
+        fields = []
+        for name, kind, info in plans:
+            if kind == "key":
+                src = rep_frame.col(info)
+                out_cols[name] = src
+                fields.append(pa.field(name, src.pa_type))
+            elif kind == "rowcount":
+                cnt = out_count.to(torch.device(self._device))
+                out_cols[name] = DeviceColumn(cnt, None, pa.int64())
+                fields.append(pa.field(name, pa.int64()))
+            elif kind == "count":
+                out_cols[name] = DeviceColumn(
+                    out_aggs[info].to(torch.device(self._device)).to(torch.int64),
+                    None,
+                    pa.int64(),
+                )
+                fields.append(pa.field(name, pa.int64()))
+            elif kind in ("sum", "min", "max"):
+                tmp, _cexpr = info
+                out_cols[name] = DeviceColumn(
+                    out_aggs[tmp].to(torch.device(self._device)),
+                    None,
+                    pa.float64(),
+                )
+                fields.append(pa.field(name, pa.float64()))
+            elif kind == "avg":
+                tmp_s, tmp_c = info
+                dev = torch.device(self._device)
+                cnt = out_aggs[tmp_c].to(dev)
+                vals = out_aggs[tmp_s].to(dev) / torch.clamp(cnt, min=1.0)
                 out_cols[name] = DeviceColumn(vals, None, pa.float64())
                 fields.append(pa.field(name, pa.float64()))
         res = HipDataFrame.from_columns(out_cols, Schema(fields), self._device)

@@ -66,17 +66,33 @@ def _np_hash_combine(h: "np.ndarray", v: "np.ndarray") -> "np.ndarray":
         )
 
 
-def _hash_rows_cpu(cols: Sequence[DeviceColumn]) -> torch.Tensor:
+def _hash_rows_cpu(
+    cols: Sequence[DeviceColumn], seed: Optional[int] = None
+) -> torch.Tensor:
     out: Optional[np.ndarray] = None
     null_h = np.uint64(0x9E3779B97F4A7C15)
+    if seed is not None:
+        n = len(cols[0])
+        out = _np_mix64(np.full(n, np.uint64(seed) + np.uint64(1)))
     for c in cols:
-        raw = c.data.numpy()
-        if raw.dtype == np.bool_:
-            raw = raw.astype(np.uint8)
-        bits = raw.view(_unsigned_view_dtype(raw.dtype)).astype(np.uint64)
-        v = _np_mix64(bits)
-        if c.valid is not None:
-            v = np.where(c.valid.numpy(), v, null_h)
+        if isinstance(c, StringDeviceColumn):
+            import pandas as pd
+
+            strings = c.to_arrow().to_pandas()
+            v = pd.util.hash_array(
+                strings.fillna("\0__null__").to_numpy(dtype=object)
+            ).astype(np.uint64)
+            v = _np_mix64(v)
+            if c.valid is not None:
+                v = np.where(c.valid.numpy(), v, null_h)
+        else:
+            raw = c.data.numpy()
+            if raw.dtype == np.bool_:
+                raw = raw.astype(np.uint8)
+            bits = raw.view(_unsigned_view_dtype(raw.dtype)).astype(np.uint64)
+            v = _np_mix64(bits)
+            if c.valid is not None:
+                v = np.where(c.valid.numpy(), v, null_h)
         out = v if out is None else _np_hash_combine(out, v)
     return torch.from_numpy(out.view(np.int64).copy())
 
@@ -94,26 +110,29 @@ def _unsigned_view_dtype(dt: "np.dtype"):
     return m[np.dtype(dt)]
 
 
-def hash_rows(cols: Sequence[DeviceColumn]) -> torch.Tensor:
-    """Row-wise 64-bit hash over multiple columns (int64 tensor holding
-    uint64 bits)."""
-    for c in cols:
-        if isinstance(c, StringDeviceColumn):
-            raise NotImplementedError(
-                "string partition keys are not yet supported on device"
-            )
+def hash_rows(
+    cols: Sequence[DeviceColumn], seed: Optional[int] = None
+) -> torch.Tensor:
+    """Row-wise 64-bit hash over multiple columns (strings included);
+    int64 tensor holding uint64 bits.  ``seed`` derives an independent
+    hash family (used for 128-bit verification of string keys)."""
     if _is_cpu(cols[0].data):
-        return _hash_rows_cpu(cols)
+        return _hash_rows_cpu(cols, seed=seed)
     ext = get_ext()
     n = len(cols[0])
     device = cols[0].data.device
     out = torch.empty(n, dtype=torch.int64, device=device)
-    first = True
+    first = seed is None
+    if seed is not None:
+        ext.hash_seed(out, seed)
     for c in cols:
-        data = c.data
-        if data.dtype == torch.int16:
-            data = data.to(torch.int32)
-        ext.hash_column(data, c.valid, out, first)
+        if isinstance(c, StringDeviceColumn):
+            ext.hash_string_column(c.offsets, c.bytes, c.valid, out, first)
+        else:
+            data = c.data
+            if data.dtype == torch.int16:
+                data = data.to(torch.int32)
+            ext.hash_column(data, c.valid, out, first)
         first = False
     return out
 
@@ -399,35 +418,43 @@ def _groupby_aggregate_cpu(
 
 
 def _hash_join_indices_cpu(
-    probe_keys: torch.Tensor, build_keys: torch.Tensor, how: str
+    probe_keys: torch.Tensor,
+    build_keys: torch.Tensor,
+    how: str,
+    probe_h2: Optional[torch.Tensor] = None,
+    build_h2: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     import pandas as pd
 
     p = pd.DataFrame({"k": probe_keys.numpy()})
-    p["pi"] = np.arange(len(p), dtype=np.int64)
     b = pd.DataFrame({"k": build_keys.numpy()})
+    if probe_h2 is not None:
+        p["k2"] = probe_h2.numpy()
+        b["k2"] = build_h2.numpy()
+    on = list(p.columns)
+    p["pi"] = np.arange(len(p), dtype=np.int64)
     b["bi"] = np.arange(len(b), dtype=np.int64)
     if how == "inner":
-        m = p.merge(b, on="k", how="inner")
+        m = p.merge(b, on=on, how="inner")
         return (
             torch.from_numpy(m["pi"].to_numpy()),
             torch.from_numpy(m["bi"].to_numpy()),
         )
     if how == "left":
-        m = p.merge(b, on="k", how="left")
+        m = p.merge(b, on=on, how="left")
         bi = m["bi"].fillna(-1).astype("int64")
         return (
             torch.from_numpy(m["pi"].to_numpy()),
             torch.from_numpy(bi.to_numpy()),
         )
     if how == "semi":
-        m = p.merge(b.drop_duplicates("k"), on="k", how="inner")
+        m = p.merge(b.drop_duplicates(on), on=on, how="inner")
         return (
             torch.from_numpy(m["pi"].to_numpy()),
             torch.from_numpy(m["bi"].to_numpy()),
         )
     if how == "anti":
-        m = p.merge(b.drop_duplicates("k"), on="k", how="left")
+        m = p.merge(b.drop_duplicates(on), on=on, how="left")
         m = m[m["bi"].isna()]
         return (
             torch.from_numpy(m["pi"].to_numpy()),
@@ -436,10 +463,115 @@ def _hash_join_indices_cpu(
     raise FugueBug(f"unsupported join mode {how}")
 
 
+class HashCollisionError(RuntimeError):
+    """h1 collision between distinct keys detected (exact fallback path
+    is taken by the caller)."""
+
+
+_H2_SEED = 0x5851F42D4C957F2D
+
+
+def groupby_aggregate_hashed(
+    df: HipDataFrame,
+    keys: List[str],
+    aggs: List[Tuple[str, int, str]],
+) -> Tuple[torch.Tensor, Dict[str, torch.Tensor], torch.Tensor]:
+    """Group-by for key tuples that can't be packed exactly (string keys):
+    groups on a 64-bit hash with an independent second hash verified
+    in-kernel.  An h1 collision between distinct keys raises
+    :class:`HashCollisionError` (caller falls back to the exact host
+    path); the undetected-failure probability is that of a full 128-bit
+    collision (~n²·2⁻¹²⁹).
+
+    Returns (representative_row_indices, {out_name: fp64}, counts).
+    """
+    key_cols = [df.col(k) for k in keys]
+    h1 = hash_rows(key_cols)
+    h2 = hash_rows(key_cols, seed=_H2_SEED)
+    n = df.count()
+    if _is_cpu(h1):
+        import pandas as pd
+
+        pdf = pd.DataFrame({"__h1": h1.numpy(), "__h2": h2.numpy()})
+        pdf["__idx"] = np.arange(n, dtype=np.int64)
+        for cname, op, oname in aggs:
+            c = df.col(cname)
+            v = c.data.numpy().astype("float64")
+            if c.valid is not None:
+                v = np.where(c.valid.numpy(), v, np.nan)
+            pdf[oname] = v
+        g = pdf.groupby("__h1", sort=False)
+        if int((g["__h2"].nunique() > 1).sum()) > 0:
+            raise HashCollisionError("h1 collision on string keys")
+        reps = torch.from_numpy(g["__idx"].first().to_numpy())
+        counts = torch.from_numpy(g.size().to_numpy().astype(np.int64))
+        out: Dict[str, torch.Tensor] = {}
+        for cname, op, oname in aggs:
+            if op == AGG_SUM:
+                sr = g[oname].sum(min_count=0)
+            elif op == AGG_MIN:
+                sr = g[oname].min()
+            elif op == AGG_MAX:
+                sr = g[oname].max()
+            elif op == AGG_COUNT:
+                sr = g[oname].count().astype("float64")
+            else:
+                raise FugueBug(f"op {op}")
+            out[oname] = torch.from_numpy(sr.to_numpy().astype("float64"))
+        return reps, out, counts
+    ext = get_ext()
+    device = h1.device
+    n_aggs = len(aggs)
+    if n_aggs > 0:
+        vals = torch.empty((n_aggs, n), dtype=torch.float64, device=device)
+        valids: Optional[torch.Tensor] = None
+        if any(df.col(c).valid is not None for c, _, _ in aggs):
+            valids = torch.ones((n_aggs, n), dtype=torch.bool, device=device)
+        for i, (cname, op, _) in enumerate(aggs):
+            c = df.col(cname)
+            vals[i] = c.data.to(torch.float64)
+            if valids is not None and c.valid is not None:
+                valids[i] = c.valid
+        ops = torch.tensor(
+            [op for _, op, _ in aggs], dtype=torch.int32, device=device
+        )
+    else:
+        vals = torch.zeros((1, n), dtype=torch.float64, device=device)
+        valids = None
+        ops = torch.tensor([AGG_COUNT], dtype=torch.int32, device=device)
+    # distinct estimate over h1
+    if n > 65536:
+        sample = h1[:: max(1, n // 65536)]
+        _, cnts = torch.unique(sample, return_counts=True)
+        d = int(cnts.numel())
+        f1 = int((cnts == 1).sum().item())
+        f2 = int((cnts == 2).sum().item())
+        est = d + (f1 * f1) // max(2 * f2, 1)
+        expected = max(d, min(n, est))
+    else:
+        expected = max(1, n)
+    tsize = _next_pow2(max(16, int(expected * 2)))
+    tkeys, gaggs, gcount = ext.gb_aggregate(
+        h1, vals, valids, ops, tsize, expected <= 100_000
+    )
+    rep, th2, conflict = ext.gb_mark_reps(h1, h2, tkeys, tsize)
+    if int(conflict.item()) > 0:
+        raise HashCollisionError("h1 collision on string keys")
+    occupied = (tkeys != GB_EMPTY).nonzero(as_tuple=True)[0]
+    reps = rep.index_select(0, occupied)
+    counts = gcount.index_select(0, occupied)
+    out_aggs: Dict[str, torch.Tensor] = {}
+    for i, (_, op, oname) in enumerate(aggs):
+        out_aggs[oname] = gaggs[i].index_select(0, occupied)
+    return reps, out_aggs, counts
+
+
 def hash_join_indices(
     probe_keys: torch.Tensor,
     build_keys: torch.Tensor,
     how: str,
+    probe_h2: Optional[torch.Tensor] = None,
+    build_h2: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Join on exact int64 keys; returns (probe_idx, build_idx) where
     build_idx == -1 marks no-match rows (left/anti).
@@ -449,12 +581,16 @@ def hash_join_indices(
     SURVEY.md §2.3 row "join ×9 types").
     """
     if _is_cpu(probe_keys):
-        return _hash_join_indices_cpu(probe_keys, build_keys, how)
+        return _hash_join_indices_cpu(
+            probe_keys, build_keys, how, probe_h2, build_h2
+        )
     ext = get_ext()
     nb = int(build_keys.numel())
     tsize = _next_pow2(max(16, nb * 2))
     heads, nxt = ext.join_build(build_keys, tsize)
-    counts = ext.join_count(probe_keys, build_keys, heads, nxt, tsize)
+    counts = ext.join_count(
+        probe_keys, build_keys, probe_h2, build_h2, heads, nxt, tsize
+    )
     counts64 = counts.to(torch.int64)
     if how == "inner":
         out_counts = counts64
@@ -472,24 +608,40 @@ def hash_join_indices(
     total = int(out_counts.sum().item())
     mode = {"inner": 0, "left": 1, "semi": 2, "anti": 3}[how]
     out_p, out_b = ext.join_emit(
-        probe_keys, build_keys, heads, nxt, tsize, offsets, total, mode
+        probe_keys, build_keys, probe_h2, build_h2, heads, nxt, tsize,
+        offsets, total, mode
     )
     return out_p, out_b
 
 
 def mark_matched_build_rows(
-    probe_keys: torch.Tensor, build_keys: torch.Tensor
+    probe_keys: torch.Tensor,
+    build_keys: torch.Tensor,
+    probe_h2: Optional[torch.Tensor] = None,
+    build_h2: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     if _is_cpu(probe_keys):
-        bk = build_keys.numpy()
-        matched = np.isin(bk, np.unique(probe_keys.numpy()))
-        return torch.from_numpy(matched)
+        if probe_h2 is None:
+            bk = build_keys.numpy()
+            matched = np.isin(bk, np.unique(probe_keys.numpy()))
+            return torch.from_numpy(matched)
+        import pandas as pd
+
+        b = pd.DataFrame(
+            {"k": build_keys.numpy(), "k2": build_h2.numpy()}
+        )
+        pset = pd.DataFrame(
+            {"k": probe_keys.numpy(), "k2": probe_h2.numpy()}
+        ).drop_duplicates()
+        pset["__m"] = True
+        m = b.merge(pset, on=["k", "k2"], how="left")
+        return torch.from_numpy(m["__m"].notna().to_numpy())
     ext = get_ext()
     nb = int(build_keys.numel())
     tsize = _next_pow2(max(16, nb * 2))
     heads, nxt = ext.join_build(build_keys, tsize)
     return ext.join_mark_build(
-        probe_keys, build_keys, heads, nxt, tsize, nb
+        probe_keys, build_keys, probe_h2, build_h2, heads, nxt, tsize, nb
     )
 
 

@@ -242,3 +242,59 @@ def test_sample_bernoulli(engine):
     pdf = _rand_df(100000, seed=19)
     s = engine.sample(engine.to_df(pdf), frac=0.1, seed=42)
     assert 8000 < s.count() < 12000
+
+
+def test_string_groupby_gpu(engine):
+    rng = np.random.default_rng(21)
+    cats = np.array(["apple", "banana", "cherry", "date", "elderberry", ""])
+    pdf = pd.DataFrame(
+        dict(k=cats[rng.integers(0, 6, 200000)], v=rng.random(200000))
+    )
+    res = fa.aggregate(
+        pdf, partition_by="k", engine=engine, s=f.sum(col("v")),
+        n=f.count(col("v")), as_fugue=True,
+    )
+    got = res.as_pandas().sort_values("k").reset_index(drop=True)
+    exp = (
+        pdf.groupby("k", as_index=False)
+        .agg(s=("v", "sum"), n=("v", "count"))
+        .sort_values("k")
+        .reset_index(drop=True)
+    )
+    assert got["k"].tolist() == exp["k"].tolist()
+    np.testing.assert_allclose(got["s"], exp["s"], rtol=1e-9)
+    assert got["n"].tolist() == exp["n"].tolist()
+
+
+def test_string_join_gpu(engine):
+    rng = np.random.default_rng(22)
+    words = np.array([f"w{i:04d}" for i in range(500)])
+    left = pd.DataFrame(
+        dict(k=words[rng.integers(0, 500, 30000)], x=rng.random(30000))
+    )
+    right = pd.DataFrame(dict(k=words[::2], y=np.arange(250).astype("f8")))
+    for how in ("inner", "left_outer", "semi", "anti"):
+        exp = fa.join(left, right, how=how, engine="native")
+        got = fa.join(left, right, how=how, engine=engine, as_fugue=True)
+        from fugue_amd.dataframe.pandas_dataframe import PandasDataFrame
+        from fugue_amd.dataframe.utils import _df_eq
+
+        assert _df_eq(got.as_local_bounded(), PandasDataFrame(exp), throw=True), how
+
+
+def test_string_groupby_high_card_gpu(engine):
+    rng = np.random.default_rng(23)
+    words = np.array([f"key_{i:06d}" for i in range(50000)])
+    pdf = pd.DataFrame(
+        dict(k=words[rng.integers(0, 50000, 500000)], v=rng.random(500000))
+    )
+    res = fa.aggregate(
+        pdf, partition_by="k", engine=engine, s=f.sum(col("v")), as_fugue=True
+    )
+    got = res.as_pandas().sort_values("k").reset_index(drop=True)
+    exp = (
+        pdf.groupby("k", as_index=False).agg(s=("v", "sum"))
+        .sort_values("k").reset_index(drop=True)
+    )
+    assert len(got) == len(exp)
+    np.testing.assert_allclose(got["s"], exp["s"], rtol=1e-9)

@@ -1,0 +1,111 @@
+"""String group/join keys on the MI355X engine (CPU tensors here; the
+same code paths run the string-hash HIP kernels on device — see
+tests/test_hip_gpu.py for the device run)."""
+import numpy as np
+import pandas as pd
+import pytest
+
+import fugue_amd.api as fa
+from fugue_amd.column.expressions import col
+from fugue_amd.column import functions as f
+from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+
+@pytest.fixture(scope="module")
+def engine():
+    return HipExecutionEngine()
+
+
+def test_string_groupby(engine):
+    rng = np.random.default_rng(0)
+    cats = np.array(["apple", "banana", "cherry", "date", ""])
+    pdf = pd.DataFrame(
+        dict(k=cats[rng.integers(0, 5, 5000)], v=rng.random(5000))
+    )
+    res = fa.aggregate(
+        pdf, partition_by="k", engine=engine,
+        s=f.sum(col("v")), n=f.count(col("v")), as_fugue=True,
+    )
+    got = res.as_pandas().sort_values("k").reset_index(drop=True)
+    exp = (
+        pdf.groupby("k", as_index=False)
+        .agg(s=("v", "sum"), n=("v", "count"))
+        .sort_values("k")
+        .reset_index(drop=True)
+    )
+    assert got["k"].tolist() == exp["k"].tolist()
+    np.testing.assert_allclose(got["s"], exp["s"], rtol=1e-9)
+    assert got["n"].tolist() == exp["n"].tolist()
+
+
+def test_string_groupby_with_nulls(engine):
+    pdf = pd.DataFrame(
+        dict(k=["a", None, "a", None, "b"], v=[1.0, 2.0, 3.0, 4.0, 5.0])
+    )
+    res = fa.aggregate(
+        pdf, partition_by="k", engine=engine, s=f.sum(col("v")), as_fugue=True
+    )
+    got = {r[0]: r[1] for r in res.as_array()}
+    assert got["a"] == 4.0
+    assert got["b"] == 5.0
+    assert got[None] == 6.0
+
+
+def test_mixed_string_int_keys(engine):
+    rng = np.random.default_rng(1)
+    pdf = pd.DataFrame(
+        dict(
+            s=np.array(["x", "y"])[rng.integers(0, 2, 1000)],
+            g=rng.integers(0, 3, 1000),
+            v=rng.random(1000),
+        )
+    )
+    res = fa.aggregate(
+        pdf, partition_by=["s", "g"], engine=engine,
+        total=f.sum(col("v")), as_fugue=True,
+    )
+    got = (
+        res.as_pandas().sort_values(["s", "g"]).reset_index(drop=True)
+    )
+    exp = (
+        pdf.groupby(["s", "g"], as_index=False)
+        .agg(total=("v", "sum"))
+        .sort_values(["s", "g"])
+        .reset_index(drop=True)
+    )
+    assert got["s"].tolist() == exp["s"].tolist()
+    assert got["g"].tolist() == exp["g"].tolist()
+    np.testing.assert_allclose(got["total"], exp["total"], rtol=1e-9)
+
+
+@pytest.mark.parametrize("how", ["inner", "left_outer", "semi", "anti", "full_outer"])
+def test_string_join(engine, how):
+    left = pd.DataFrame(
+        dict(k=["a", "b", "c", "a", None], x=[1.0, 2.0, 3.0, 4.0, 5.0])
+    )
+    right = pd.DataFrame(dict(k=["a", "c", "d"], y=[10.0, 30.0, 40.0]))
+    exp = fa.join(left, right, how=how, engine="native")
+    got = fa.join(left, right, how=how, engine=engine, as_fugue=True)
+    from fugue_amd.dataframe.pandas_dataframe import PandasDataFrame
+    from fugue_amd.dataframe.utils import _df_eq
+
+    assert _df_eq(got.as_local_bounded(), PandasDataFrame(exp), throw=True)
+
+
+def test_string_shuffle_sql(engine):
+    rng = np.random.default_rng(2)
+    pdf = pd.DataFrame(
+        dict(
+            name=np.array(["aa", "bb", "cc"])[rng.integers(0, 3, 2000)],
+            v=rng.random(2000),
+        )
+    )
+    res = fa.fugue_sql(
+        "SELECT name, SUM(v) AS s FROM t GROUP BY name",
+        t=pdf,
+        engine=engine,
+        as_fugue=True,
+    )
+    got = res.as_pandas().sort_values("name").reset_index(drop=True)
+    exp = pdf.groupby("name", as_index=False).agg(s=("v", "sum"))
+    np.testing.assert_allclose(got["s"], exp["s"], rtol=1e-9)
