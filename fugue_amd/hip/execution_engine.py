@@ -642,6 +642,8 @@ class HipExecutionEngine(ExecutionEngine):
         cols2 = [d2.col(k) for k in keys]
         if (
             len(keys) == 1
+            and not isinstance(cols1[0], StringDeviceColumn)
+            and not isinstance(cols2[0], StringDeviceColumn)
             and cols1[0].data.dtype == torch.int64
             and cols2[0].data.dtype == torch.int64
             and cols1[0].valid is None

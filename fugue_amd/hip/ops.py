@@ -209,6 +209,7 @@ def pack_keys(
     """
     if (
         len(key_cols) == 1
+        and not isinstance(key_cols[0], StringDeviceColumn)
         and key_cols[0].data.dtype == torch.int64
         and key_cols[0].valid is None
         and mins is None
