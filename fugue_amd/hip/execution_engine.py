@@ -890,12 +890,55 @@ class HipExecutionEngine(ExecutionEngine):
     def subtract(
         self, df1: DataFrame, df2: DataFrame, distinct: bool = True
     ) -> DataFrame:
+        res = self._device_setop(df1, df2, distinct, "anti")
+        if res is not None:
+            return res
         return self._setop_fallback(df1, df2, distinct, "except")
 
     def intersect(
         self, df1: DataFrame, df2: DataFrame, distinct: bool = True
     ) -> DataFrame:
+        res = self._device_setop(df1, df2, distinct, "semi")
+        if res is not None:
+            return res
         return self._setop_fallback(df1, df2, distinct, "intersect")
+
+    def _row_keys(
+        self, d: HipDataFrame
+    ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """128-bit row-content keys (nulls form their own value class)."""
+        cols = [d.col(n) for n in d.schema.names]
+        return (
+            dops.hash_rows(cols),
+            dops.hash_rows(cols, seed=dops._H2_SEED),
+        )
+
+    def _device_setop(
+        self, df1: DataFrame, df2: DataFrame, distinct: bool, how: str
+    ) -> Optional[DataFrame]:
+        d1 = self.to_df(df1)
+        d2 = self.to_df(df2)
+        if d1.schema != d2.schema:
+            raise ValueError(f"schema mismatch {d1.schema} vs {d2.schema}")
+        if not (
+            isinstance(d1, HipDataFrame) and isinstance(d2, HipDataFrame)
+        ):
+            return None
+        if not distinct and how == "semi":
+            return None  # INTERSECT ALL: rare, host path
+        try:
+            if self.is_distributed:
+                d1 = self._shuffle_by_columns(d1, d1.schema.names)
+                d2 = self._shuffle_by_columns(d2, d2.schema.names)
+            h11, h12 = self._row_keys(d1)
+            h21, h22 = self._row_keys(d2)
+            pi, _bi = dops.hash_join_indices(h11, h21, how, h12, h22)
+            res = d1.gather_rows(pi)
+            if distinct:
+                res = self.distinct(res)
+            return res
+        except (NotImplementedError, dops.HashCollisionError):
+            return None
 
     def _setop_fallback(
         self, df1: DataFrame, df2: DataFrame, distinct: bool, op: str
@@ -936,23 +979,15 @@ class HipExecutionEngine(ExecutionEngine):
 
     def distinct(self, df: DataFrame) -> DataFrame:
         d = self.to_df(df)
-        if isinstance(d, HipDataFrame) and self.is_distributed:
+        if isinstance(d, HipDataFrame):
             try:
-                d = self._shuffle_by_columns(d, d.schema.names)
-            except NotImplementedError:
-                local = self._gather_all(d)
-                from fugue_amd.utils.pandas_like import drop_duplicates
-
-                return self.to_df(
-                    PandasDataFrame(drop_duplicates(local.as_pandas()), d.schema),
-                    shard_replicated=True,
-                )
-            from fugue_amd.utils.pandas_like import drop_duplicates
-
-            return self.to_df(
-                PandasDataFrame(drop_duplicates(d.as_pandas()), d.schema),
-                shard_replicated=False,
-            )
+                if self.is_distributed:
+                    d = self._shuffle_by_columns(d, d.schema.names)
+                h1, h2 = self._row_keys(d)
+                reps = dops.distinct_reps(h1, h2)
+                return d.gather_rows(reps.to(h1.device))
+            except (NotImplementedError, dops.HashCollisionError):
+                pass
         from fugue_amd.utils.pandas_like import drop_duplicates
 
         return self.to_df(

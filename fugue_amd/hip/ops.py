@@ -567,6 +567,43 @@ def groupby_aggregate_hashed(
     return reps, out_aggs, counts
 
 
+def distinct_reps(h1: torch.Tensor, h2: torch.Tensor) -> torch.Tensor:
+    """One representative row index per distinct (h1, h2) row-content key
+    (raises HashCollisionError on an h1 collision — caller falls back)."""
+    n = int(h1.numel())
+    if _is_cpu(h1):
+        import pandas as pd
+
+        pdf = pd.DataFrame({"h1": h1.numpy(), "h2": h2.numpy()})
+        pdf["idx"] = np.arange(n, dtype=np.int64)
+        g = pdf.groupby("h1", sort=False)
+        if int((g["h2"].nunique() > 1).sum()) > 0:
+            raise HashCollisionError("h1 collision in distinct")
+        return torch.from_numpy(g["idx"].first().to_numpy())
+    ext = get_ext()
+    device = h1.device
+    if n > 65536:
+        sample = h1[:: max(1, n // 65536)]
+        _, cnts = torch.unique(sample, return_counts=True)
+        d = int(cnts.numel())
+        f1 = int((cnts == 1).sum().item())
+        f2 = int((cnts == 2).sum().item())
+        expected = max(d, min(n, d + (f1 * f1) // max(2 * f2, 1)))
+    else:
+        expected = max(1, n)
+    tsize = _next_pow2(max(16, int(expected * 2)))
+    vals = torch.zeros((1, max(n, 1)), dtype=torch.float64, device=device)
+    ops = torch.tensor([AGG_COUNT], dtype=torch.int32, device=device)
+    tkeys, _gaggs, _gcount = ext.gb_aggregate(
+        h1, vals, None, ops, tsize, expected <= 100_000
+    )
+    rep, _th2, conflict = ext.gb_mark_reps(h1, h2, tkeys, tsize)
+    if int(conflict.item()) > 0:
+        raise HashCollisionError("h1 collision in distinct")
+    occupied = (tkeys != GB_EMPTY).nonzero(as_tuple=True)[0]
+    return rep.index_select(0, occupied)
+
+
 def hash_join_indices(
     probe_keys: torch.Tensor,
     build_keys: torch.Tensor,
