@@ -1233,6 +1233,17 @@ class HipExecutionEngine(ExecutionEngine):
         # plan each output column
         plans: List[Tuple[str, str, Any]] = []  # (out_name, kind, info)
         partials: List[Tuple[str, int, str]] = []  # (src col, op, tmp name)
+        derived: Dict[str, DeviceColumn] = {}
+
+        def _derive(arg: ColumnExpr) -> str:
+            """Evaluate a compound aggregate argument to a derived device
+            column usable as an aggregation input."""
+            name = f"__expr{len(derived)}"
+            data, valid = eval_device_expr(arg, d)
+            derived[name] = DeviceColumn(
+                data.to(torch.float64), valid, pa.float64()
+            )
+            return name
 
         def _add_partial(src: str, op: int) -> str:
             tmp = f"__p{len(partials)}_{src}_{op}"
@@ -1268,11 +1279,12 @@ class HipExecutionEngine(ExecutionEngine):
                         # per-group row count — no agg column traffic needed
                         plans.append((name, "rowcount", None))
                     continue
-                if not isinstance(arg, _NamedColumnExpr):
-                    raise DeviceExprError("complex agg argument: fallback")
-                src = arg.name
-                if isinstance(d.col(src), StringDeviceColumn):
-                    raise DeviceExprError("string aggregation: fallback")
+                if isinstance(arg, _NamedColumnExpr):
+                    src = arg.name
+                    if isinstance(d.col(src), StringDeviceColumn):
+                        raise DeviceExprError("string aggregation: fallback")
+                else:
+                    src = _derive(arg)
                 if fname in ("SUM", "MIN", "MAX"):
                     tmp = _add_partial(src, _AGG_FUNC_TO_OP[fname])
                     plans.append((name, fname.lower(), (tmp, c)))
@@ -1286,6 +1298,15 @@ class HipExecutionEngine(ExecutionEngine):
             raise DeviceExprError("compound aggregate expression: fallback")
         if len(key_names) == 0:
             raise DeviceExprError("global aggregate: fallback")
+        if derived:
+            cols_map = dict(d.columns_map)
+            cols_map.update(derived)
+            fields_ext = list(d.schema.fields) + [
+                pa.field(n, pa.float64()) for n in derived
+            ]
+            d = HipDataFrame.from_columns(
+                cols_map, Schema(fields_ext), self._device
+            )
         key_cols = [d.col(k) for k in key_names]
         hashed_keys = any(
             isinstance(kc, StringDeviceColumn) for kc in key_cols

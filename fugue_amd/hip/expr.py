@@ -52,6 +52,10 @@ def eval_device_expr(
         if expr.as_type is not None:
             data = _cast(data, expr.as_type)
         return data, c.valid
+    if isinstance(expr, _BinaryOpExpr) and expr.op in ("==", "!="):
+        res = _try_string_equality(expr, df)
+        if res is not None:
+            return res
     if isinstance(expr, _NotOpExpr):
         d, v = eval_device_expr(expr.col, df)
         return ~d.to(torch.bool), v
@@ -134,6 +138,48 @@ def eval_device_expr(
             return out_d, out_v
         raise DeviceExprError(f"function {expr.func}")
     raise DeviceExprError(f"can't evaluate {expr} on device")
+
+
+def _try_string_equality(
+    expr: "_BinaryOpExpr", df: HipDataFrame
+) -> Optional[Tuple[torch.Tensor, Optional[torch.Tensor]]]:
+    """``col == 'literal'`` on a device string column: compared via the
+    128-bit row hash of the column vs the literal's hash (undetectable
+    mismatch probability ~2^-128)."""
+    sides = [expr.left, expr.right]
+    col_e = lit_e = None
+    for a, b in (sides, sides[::-1]):
+        if (
+            isinstance(a, _NamedColumnExpr)
+            and a.name in df.schema._index
+            and isinstance(df.col(a.name), StringDeviceColumn)
+            and isinstance(b, _LiteralColumnExpr)
+            and isinstance(b.value, str)
+        ):
+            col_e, lit_e = a, b
+            break
+    if col_e is None:
+        return None
+    from fugue_amd.hip import ops as dops
+    import pyarrow as pa
+
+    c = df.col(col_e.name)
+    h1 = dops.hash_rows([c])
+    h2 = dops.hash_rows([c], seed=dops._H2_SEED)
+    lit_col = StringDeviceColumn.from_arrow_strings(
+        pa.array([lit_e.value], type=pa.string()), df.device
+    )
+    l1 = dops.hash_rows([lit_col])[0]
+    l2 = dops.hash_rows([lit_col], seed=dops._H2_SEED)[0]
+    eq = (h1 == l1) & (h2 == l2)
+    if c.valid is not None:
+        eq = eq & c.valid
+    if expr.op == "!=":
+        res = ~eq
+        if c.valid is not None:
+            res = res & c.valid
+        return res, None
+    return eq, None
 
 
 def _merge_valid(

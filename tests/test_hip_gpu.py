@@ -298,3 +298,32 @@ def test_string_groupby_high_card_gpu(engine):
     )
     assert len(got) == len(exp)
     np.testing.assert_allclose(got["s"], exp["s"], rtol=1e-9)
+
+
+def test_q3_gpu(engine):
+    import os, sys
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+    from benchmarks.q3_bench import Q3, gen_tables
+
+    customer, orders, lineitem, _ = gen_tables(0.05, engine.device, 0)
+    res = fa.fugue_sql(
+        Q3, customer=customer, orders=orders, lineitem=lineitem,
+        engine=engine, as_fugue=True,
+    )
+    got = res.as_pandas()
+    c = customer.as_pandas()
+    o = orders.as_pandas()
+    l = lineitem.as_pandas()
+    m = (
+        c[c.mktsegment == "BUILDING"].merge(o, on="custkey").merge(l, on="orderkey")
+    )
+    m = m[(m.orderdate < 9204) & (m.shipdate > 9204)]
+    m["rev"] = m.extendedprice * (1 - m.discount)
+    exp = (
+        m.groupby(["orderkey", "orderdate", "shippriority"], as_index=False)
+        .agg(revenue=("rev", "sum"))
+        .nlargest(10, "revenue")
+    )
+    np.testing.assert_allclose(
+        got["revenue"].values, exp["revenue"].values, rtol=1e-9
+    )
