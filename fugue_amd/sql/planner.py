@@ -206,14 +206,99 @@ def _resolve_from(item: X.FromItem, tables: Dict[str, DataFrame], engine: Any) -
     return engine.to_df(tables[item.table])
 
 
+def _split_conjuncts(e: Optional[X.Expr]) -> List[X.Expr]:
+    if e is None:
+        return []
+    if isinstance(e, X.BinOp) and e.op == "AND":
+        return _split_conjuncts(e.left) + _split_conjuncts(e.right)
+    return [e]
+
+
+def _mentioned_cols(e: X.Expr) -> List[str]:
+    out: List[str] = []
+
+    def _walk(x: X.Expr) -> None:
+        if isinstance(x, X.ColRef):
+            out.append(x.name)
+        elif isinstance(x, X.BinOp):
+            _walk(x.left)
+            _walk(x.right)
+        elif isinstance(x, X.UnOp):
+            _walk(x.operand)
+        elif isinstance(x, X.Between):
+            _walk(x.expr)
+            _walk(x.low)
+            _walk(x.high)
+        elif isinstance(x, X.InList):
+            _walk(x.expr)
+            for v in x.values:
+                _walk(v)
+        elif isinstance(x, X.Cast):
+            _walk(x.expr)
+        elif isinstance(x, X.FuncCall):
+            for a in x.args:
+                _walk(a)
+
+    _walk(e)
+    return out
+
+
 def _execute_core(
     stmt: X.SelectStmt, tables: Dict[str, DataFrame], engine: Any
 ) -> DataFrame:
     if stmt.from_item is None:
         raise UnsupportedPlan("SELECT without FROM")
     res = _resolve_from(stmt.from_item, tables, engine)
-    for j in stmt.joins:
-        right = _resolve_from(j.item, tables, engine)
+    # predicate pushdown: when ALL joins are inner, single-table conjuncts
+    # of WHERE filter their source table before the joins
+    join_inputs: List[DataFrame] = []
+    residual_where: Optional[X.Expr] = stmt.where
+    pushed: Dict[int, List[X.Expr]] = {}
+    all_inner = all(j.how == "inner" for j in stmt.joins)
+    if stmt.joins and all_inner and stmt.where is not None:
+        frames: List[DataFrame] = [res] + [
+            _resolve_from(j.item, tables, engine) for j in stmt.joins
+        ]
+        remaining: List[X.Expr] = []
+        for conj in _split_conjuncts(stmt.where):
+            cols = set(_mentioned_cols(conj))
+            target = None
+            for idx, fr in enumerate(frames):
+                if cols <= set(fr.schema.names):
+                    # unique owner only (ambiguous cols stay post-join)
+                    owners = [
+                        i
+                        for i, f2 in enumerate(frames)
+                        if cols <= set(f2.schema.names)
+                    ]
+                    if len(owners) == 1:
+                        target = idx
+                    break
+            if target is None:
+                remaining.append(conj)
+            else:
+                pushed.setdefault(target, []).append(conj)
+        for idx, conjs in pushed.items():
+            fr = frames[idx]
+            for conj in conjs:
+                ce = _to_column_expr(conj, fr.schema, {})
+                fr = engine.filter(fr, ce)
+            frames[idx] = fr
+        res = frames[0]
+        join_inputs = frames[1:]
+        residual_where = None
+        for conj in remaining:
+            residual_where = (
+                conj
+                if residual_where is None
+                else X.BinOp("AND", residual_where, conj)
+            )
+    for ji, j in enumerate(stmt.joins):
+        right = (
+            join_inputs[ji]
+            if ji < len(join_inputs)
+            else _resolve_from(j.item, tables, engine)
+        )
         how = _JOIN_MAP.get(j.how)
         if how is None:
             raise UnsupportedPlan(f"join {j.how}")
@@ -251,8 +336,8 @@ def _execute_core(
     # select columns
     schema = res.schema
     where_expr: Optional[ColumnExpr] = None
-    if stmt.where is not None:
-        where_expr = _to_column_expr(stmt.where, schema, {})
+    if residual_where is not None:
+        where_expr = _to_column_expr(residual_where, schema, {})
     cols: List[ColumnExpr] = []
     from fugue_amd.column.expressions import all_cols
 
