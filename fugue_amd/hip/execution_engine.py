@@ -54,6 +54,7 @@ from fugue_amd.hip.frame import (
     supported_device_type,
 )
 from fugue_amd.parallel.comm import Communicator, get_communicator
+from fugue_amd.utils.stats import EngineStats
 from fugue_amd.utils.tracing import op_range
 from fugue_amd.schema import Schema
 from fugue_amd.utils.params import ParamDict
@@ -259,6 +260,7 @@ class HipExecutionEngine(ExecutionEngine):
         else:
             self._device = "cpu"
         self._comm = get_communicator(self._device)
+        self._stats = EngineStats()
 
     # ------------------------------------------------------------------ #
     @property
@@ -284,6 +286,12 @@ class HipExecutionEngine(ExecutionEngine):
     @property
     def device(self) -> str:
         return self._device
+
+    @property
+    def stats(self) -> EngineStats:
+        """Engine counters: shuffle_bytes/shuffle_rows/joins/aggregates/
+        broadcasts/spills (reset with ``engine.stats.reset()``)."""
+        return self._stats
 
     def create_default_map_engine(self) -> MapEngine:
         return HipMapEngine(self)
@@ -367,6 +375,9 @@ class HipExecutionEngine(ExecutionEngine):
         to rank i."""
         if not self.is_distributed:
             return df
+        self._stats.add("shuffles")
+        self._stats.add("shuffle_rows", float(df.count()))
+        self._stats.add("shuffle_bytes", float(df.num_bytes()))
         counts_matrix = self._comm.allgather_counts(bucket_counts.cpu())
         send_counts = bucket_counts.cpu().tolist()
         recv_counts = counts_matrix[:, self.rank].tolist()
@@ -538,6 +549,7 @@ class HipExecutionEngine(ExecutionEngine):
             if isinstance(hdf, DataFrame):
                 hdf.metadata["broadcasted"] = True
             return hdf
+        self._stats.add("broadcasts")
         res = self._gather_all(hdf)
         out = HipDataFrame(res.as_arrow(), hdf.schema, device=self._device)
         out.metadata["broadcasted"] = True
@@ -558,6 +570,8 @@ class HipExecutionEngine(ExecutionEngine):
             if storage == "host" or (
                 threshold > 0 and res.num_bytes() > threshold
             ):
+                self._stats.add("spills")
+                self._stats.add("spill_bytes", float(res.num_bytes()))
                 res = SpilledDataFrame(res)
         if df.has_metadata:
             res.reset_metadata(df.metadata)
@@ -573,6 +587,7 @@ class HipExecutionEngine(ExecutionEngine):
         how: str,
         on: Optional[List[str]] = None,
     ) -> DataFrame:
+        self._stats.add("joins")
         with op_range(f"fugue.join.{how}"):
             return self._join_impl(df1, df2, how, on)
 
@@ -1216,6 +1231,7 @@ class HipExecutionEngine(ExecutionEngine):
                 out_cols, Schema(fields), self._device
             )
         # aggregation path
+        self._stats.add("aggregates")
         with op_range("fugue.aggregate"):
             return self._device_aggregate(d, cols, having)
 
