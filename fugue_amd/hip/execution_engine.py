@@ -215,7 +215,16 @@ class HipMapEngine(MapEngine):
                 staged_pandas_batches,
             )
 
-            if can_fast_stage(sorted_df):
+            if map_func_format_hint == "pyarrow":
+                # arrow-native UDFs: zero pandas conversion
+                table = sorted_df.as_arrow()
+                for gi in range(len(bounds) - 1):
+                    start, end = bounds[gi], bounds[gi + 1]
+                    sub = table.slice(start, end - start)
+                    input_df = ArrowDataFrame(sub)
+                    cursor.set(lambda: input_df.peek_array(), gi, 0)
+                    results.append(map_func(cursor, input_df).as_pandas())
+            elif can_fast_stage(sorted_df):
                 # pinned, double-buffered D2H: batch k+1 copies on a side
                 # stream while the UDF processes batch k
                 for g0, g1, batch in staged_pandas_batches(sorted_df, bounds):
@@ -1331,9 +1340,34 @@ class HipExecutionEngine(ExecutionEngine):
             return self._device_aggregate_hashed(
                 d, key_names, partials, plans, having, cols
             )
+        pack_mins = pack_widths = None
+        if self.is_distributed:
+            # cross-rank partial merge requires a rank-consistent packing:
+            # allreduce per-key-column global min/max
+            needs_pack = not (
+                len(key_cols) == 1
+                and key_cols[0].data.dtype == torch.int64
+                and key_cols[0].valid is None
+            )
+            if needs_pack:
+                pack_mins, pack_widths = [], []
+                for kc in key_cols:
+                    lo = self._global_min(kc)
+                    hi = self._global_max(kc)
+                    pack_mins.append(lo)
+                    pack_widths.append(
+                        max(1, int(np.ceil(np.log2(max(2, hi - lo + 2)))))
+                    )
+                if sum(pack_widths) > 63:
+                    # not packable: co-shuffle rows, then exact local
+                    # hashed aggregation (globally exact)
+                    return self._device_aggregate_hashed(
+                        d, key_names, partials, plans, having, cols
+                    )
         # local partial aggregation
         out_keys, out_aggs, out_count, meta = dops.groupby_aggregate(
-            d, key_names, partials
+            d, key_names, partials, pack_mins=pack_mins,
+            pack_widths=pack_widths,
         )
         if self.is_distributed:
             # exchange partials by key hash, then re-aggregate

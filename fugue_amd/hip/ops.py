@@ -295,6 +295,8 @@ def groupby_aggregate(
     keys: List[str],
     aggs: List[Tuple[str, int, str]],  # (input col, op, output name)
     expected_groups: Optional[int] = None,
+    pack_mins: Optional[List[int]] = None,
+    pack_widths: Optional[List[int]] = None,
 ) -> Tuple[torch.Tensor, Dict[str, torch.Tensor], torch.Tensor, Optional[Dict[str, Any]]]:
     """Hash group-by: returns (group_keys_packed, {out_name: fp64 values},
     group_row_counts, pack_meta).
@@ -305,7 +307,7 @@ def groupby_aggregate(
     n = df.count()
     device = df.col(keys[0]).data.device if keys else torch.device(df.device)
     key_cols = [df.col(k) for k in keys]
-    packed, meta = pack_keys(key_cols)
+    packed, meta = pack_keys(key_cols, mins=pack_mins, widths=pack_widths)
     n_aggs = len(aggs)
     if (
         n_aggs == 1
