@@ -1522,11 +1522,42 @@ class HipExecutionEngine(ExecutionEngine):
     ) -> DataFrame:
         from fugue_amd.utils import io as _io
 
+        if (
+            isinstance(path, str)
+            and os.path.isdir(path)
+            and len(self._part_files(path)) > 0
+        ):
+            # partitioned parquet dataset: each rank reads its file subset
+            import pyarrow.parquet as pq
+
+            files = self._part_files(path)
+            mine = files[self.rank :: max(1, self.world_size)]
+            col_names = None
+            if isinstance(columns, list):
+                col_names = columns
+            elif columns is not None:
+                col_names = Schema(columns).names
+            if len(mine) > 0:
+                tables = [pq.read_table(f, columns=col_names) for f in mine]
+                table = (
+                    pa.concat_tables(tables) if len(tables) > 1 else tables[0]
+                )
+            else:
+                table = pq.read_table(files[0], columns=col_names).slice(0, 0)
+            return self.to_df(
+                ArrowDataFrame(table), shard_replicated=False
+            )
         pdf, schema = _io.load_df(
             path, format_hint=format_hint, columns=columns, **kwargs
         )
         src = PandasDataFrame(pdf, schema) if schema is not None else PandasDataFrame(pdf)
         return self.to_df(src, shard_replicated=True)
+
+    @staticmethod
+    def _part_files(path: str) -> List[str]:
+        import glob as _glob
+
+        return sorted(_glob.glob(os.path.join(path, "part-*.parquet")))
 
     def save_df(
         self,
@@ -1539,12 +1570,36 @@ class HipExecutionEngine(ExecutionEngine):
         **kwargs: Any,
     ) -> None:
         from fugue_amd.utils import io as _io
+        from fugue_amd.utils.io import infer_format
 
         d = self.to_df(df)
-        local = self._as_local(d)  # gathered on every rank
+        fmt = infer_format(path, format_hint if format_hint else None) if (
+            format_hint or "." in os.path.basename(path)
+        ) else "parquet"
+        if self.is_distributed and not force_single and fmt == "parquet":
+            # each rank writes its shard as a part file (no gather)
+            if self.rank == 0:
+                if os.path.exists(path) and mode == "overwrite":
+                    import shutil
+
+                    if os.path.isdir(path):
+                        shutil.rmtree(path)
+                    else:
+                        os.remove(path)
+                os.makedirs(path, exist_ok=True)
+            self._comm.barrier()
+            import pyarrow.parquet as pq
+
+            local = d.as_arrow() if hasattr(d, "as_arrow") else d.as_local().as_arrow()
+            pq.write_table(
+                local, os.path.join(path, f"part-{self.rank:05d}.parquet")
+            )
+            self._comm.barrier()
+            return
+        local_df = self._as_local(d)  # gathered on every rank
         if self.rank == 0:
             _io.save_df(
-                local.as_pandas(), d.schema, path,
+                local_df.as_pandas(), d.schema, path,
                 format_hint=format_hint, mode=mode, **kwargs
             )
         self._comm.barrier()

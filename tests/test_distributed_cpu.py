@@ -241,3 +241,33 @@ def test_distributed_q3():
         assert len(got) == len(exp)
         for g, x in zip(sorted(got), exp):
             assert g[0] == x[0] and abs(g[1] - x[1]) < 1e-6
+
+
+def _io_job(rank: int):
+    import tempfile
+
+    import fugue_amd.api as fa
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    pdf = pd.DataFrame(dict(a=np.arange(100), b=np.arange(100) * 1.5))
+    d = e.to_df(pdf)  # sharded
+    path = os.path.join(os.environ["FUGUE_TEST_TMP"], "out")
+    e.save_df(d, path)
+    back = e.load_df(path)
+    total = e.comm.allreduce_sum(back.count())
+    local = e._gather_all(back) if hasattr(back, "as_arrow") else back
+    s = sorted(r[0] for r in local.as_array())
+    return dict(total=total, keys=s)
+
+
+def test_distributed_parquet_parts(tmp_path):
+    os.environ["FUGUE_TEST_TMP"] = str(tmp_path)
+    try:
+        results = run_distributed(_io_job, 29521)
+        assert results[0]["total"] == 100
+        assert results[0]["keys"] == list(range(100))
+        files = os.listdir(os.path.join(str(tmp_path), "out"))
+        assert len([f for f in files if f.startswith("part-")]) == 2
+    finally:
+        os.environ.pop("FUGUE_TEST_TMP", None)
