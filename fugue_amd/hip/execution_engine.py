@@ -158,13 +158,42 @@ class HipMapEngine(MapEngine):
                     local, list(presort.keys()), list(presort.values())
                 )
                 local = local.gather_rows(perm)
-            pdf_local = local.as_pandas()
-            if pdf_local.shape[0] > 0 or not engine.is_distributed:
-                input_df = PandasDataFrame(
-                    pdf_local, Schema(df.schema), pandas_df_wrapper=True
+            n_local_parts = 1
+            if len(keys) == 0 and partition_spec.num_partitions != "0":
+                from fugue_amd.constants import (
+                    KEYWORD_CORECOUNT,
+                    KEYWORD_ROWCOUNT,
                 )
-                if not input_df.empty:
-                    cursor.set(lambda: input_df.peek_array(), engine.rank, 0)
+
+                total = partition_spec.get_num_partitions(
+                    **{
+                        KEYWORD_ROWCOUNT: lambda: engine.comm.allreduce_sum(
+                            local.count()
+                        ),
+                        KEYWORD_CORECOUNT: lambda: engine.world_size,
+                    }
+                )
+                n_local_parts = max(
+                    1, (total + engine.world_size - 1) // engine.world_size
+                )
+            pdf_local = local.as_pandas()
+            if len(pdf_local) > 0:
+                import numpy as _np
+
+                for p, subdf in enumerate(
+                    _np.array_split(pdf_local, n_local_parts)
+                ):
+                    if len(subdf) == 0:
+                        continue
+                    sub = subdf.reset_index(drop=True)
+                    input_df = PandasDataFrame(
+                        sub, Schema(df.schema), pandas_df_wrapper=True
+                    )
+                    cursor.set(
+                        lambda: input_df.peek_array(),
+                        engine.rank * n_local_parts + p,
+                        0,
+                    )
                     results.append(map_func(cursor, input_df).as_pandas())
         else:
             # 2. local segmented sort by keys (+presort), group boundaries

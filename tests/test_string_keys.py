@@ -109,3 +109,49 @@ def test_string_shuffle_sql(engine):
     got = res.as_pandas().sort_values("name").reset_index(drop=True)
     exp = pdf.groupby("name", as_index=False).agg(s=("v", "sum"))
     np.testing.assert_allclose(got["s"], exp["s"], rtol=1e-9)
+
+
+def test_datetime_keys_and_values(engine):
+    import datetime
+
+    pdf = pd.DataFrame(
+        dict(
+            ts=pd.to_datetime(
+                ["2024-01-01", "2024-01-02", "2024-01-01", "2024-01-03"]
+            ),
+            v=[1.0, 2.0, 3.0, 4.0],
+        )
+    )
+    res = fa.aggregate(
+        pdf, partition_by="ts", engine=engine, s=f.sum(col("v")), as_fugue=True
+    )
+    got = res.as_pandas().sort_values("ts").reset_index(drop=True)
+    exp = pdf.groupby("ts", as_index=False).agg(s=("v", "sum"))
+    assert got["ts"].tolist() == exp["ts"].tolist()
+    np.testing.assert_allclose(got["s"], exp["s"])
+    # datetime round-trip
+    d = engine.to_df(pdf)
+    back = d.as_pandas()
+    pd.testing.assert_frame_equal(back, pdf)
+    # join on datetime keys
+    dims = pd.DataFrame(
+        dict(ts=pd.to_datetime(["2024-01-01", "2024-01-03"]), w=[10.0, 20.0])
+    )
+    j = fa.join(pdf, dims, how="inner", engine=engine, as_fugue=True)
+    assert j.count() == 3
+
+
+def test_map_num_partitions(engine):
+    from typing import Any, List
+
+    pdf = pd.DataFrame(dict(x=np.arange(10)))
+
+    def count_rows(rows: List[List[Any]]) -> List[List[Any]]:
+        return [[len(rows)]]
+
+    res = fa.transform(
+        pdf, count_rows, schema="n:long", partition=dict(num=5), engine=engine
+    )
+    counts = sorted(r for r in pd.DataFrame(res)["n"].tolist())
+    assert sum(counts) == 10
+    assert len(counts) == 5
