@@ -12,7 +12,11 @@ from fugue_amd.sql.workflow import FugueSQLWorkflow
 
 
 def _capture_caller_vars(depth: int = 2) -> Dict[str, Any]:
-    frame = inspect.stack()[depth].frame
+    # sys._getframe, not inspect.stack(): the latter resolves source info
+    # for every frame (~20ms per call)
+    import sys
+
+    frame = sys._getframe(depth)
     res: Dict[str, Any] = {}
     res.update(frame.f_globals)
     res.update(frame.f_locals)
