@@ -1184,6 +1184,27 @@ class HipExecutionEngine(ExecutionEngine):
                 ).head(n)
             return pdf.reset_index(drop=True)
 
+        if (
+            isinstance(d, HipDataFrame)
+            and len(partition_spec.partition_by) == 0
+            and len(_presort) > 0
+            and na_position == "last"
+        ):
+            # device top-n: sort permutation on device, gather only n rows
+            try:
+                perm = dops.sort_indices(
+                    d, list(_presort.keys()), list(_presort.values())
+                )
+                local_top = d.gather_rows(perm[: min(n, d.count())])
+                if not self.is_distributed:
+                    return local_top
+                gathered = self._gather_all(local_top)
+                final = _take_pdf(gathered.as_pandas())
+                return self.to_df(
+                    PandasDataFrame(final, d.schema), shard_replicated=True
+                )
+            except NotImplementedError:
+                pass
         if self.is_distributed and isinstance(d, HipDataFrame):
             # prune to local candidates first (any global top-n row is in
             # some rank's local top-n), then gather only the candidates

@@ -699,14 +699,15 @@ def sort_indices(
             raise NotImplementedError("string sort keys not yet on device")
         vals = c.data.index_select(0, perm)
         if c.valid is not None:
-            # nulls last regardless of direction (pandas na_position default)
-            big = (
-                torch.iinfo(vals.dtype).max
-                if not vals.is_floating_point()
-                else float("inf")
-            )
+            # nulls last regardless of direction (pandas na_position
+            # default): sentinel is +extreme for asc, -extreme for desc
+            if vals.is_floating_point():
+                sentinel = float("inf") if asc else float("-inf")
+            else:
+                info = torch.iinfo(vals.dtype)
+                sentinel = info.max if asc else info.min
             v = c.valid.index_select(0, perm)
-            vals = torch.where(v, vals, torch.full_like(vals, big))
+            vals = torch.where(v, vals, torch.full_like(vals, sentinel))
         idx = torch.argsort(vals, stable=True, descending=not asc)
         perm = perm.index_select(0, idx)
     return perm
