@@ -56,6 +56,8 @@ void launch_hash_seed(uint64_t*, int64_t, uint64_t, hipStream_t);
 void launch_gb_mark_reps(const int64_t*, const int64_t*, const int64_t*,
                          int64_t, int64_t*, int64_t*, int64_t*, int64_t,
                          hipStream_t);
+void launch_compact8(const bool*, int64_t, int64_t*, const uint64_t**,
+                     uint64_t**, int, hipStream_t);
 }
 
 namespace {
@@ -342,6 +344,32 @@ std::vector<at::Tensor> gb_mark_reps(at::Tensor h1, at::Tensor h2,
   return {rep, th2, conflict};
 }
 
+// Compact up to 8 eight-byte columns by mask in one pass.
+// Returns the output tensors (sized by mask.sum(), computed on device
+// and read back once).
+std::vector<at::Tensor> compact_columns(at::Tensor mask,
+                                        std::vector<at::Tensor> cols,
+                                        int64_t out_n) {
+  check_gpu(mask, "mask");
+  TORCH_CHECK(cols.size() >= 1 && cols.size() <= 8, "1..8 columns");
+  int64_t n = mask.numel();
+  auto cursor = at::zeros({1}, mask.options().dtype(at::kLong));
+  const uint64_t* srcs[8];
+  uint64_t* dsts[8];
+  std::vector<at::Tensor> outs;
+  for (size_t c = 0; c < cols.size(); ++c) {
+    check_gpu(cols[c], "col");
+    TORCH_CHECK(cols[c].element_size() == 8, "8-byte columns only");
+    auto out = at::empty({out_n}, cols[c].options());
+    srcs[c] = reinterpret_cast<const uint64_t*>(cols[c].data_ptr());
+    dsts[c] = reinterpret_cast<uint64_t*>(out.data_ptr());
+    outs.push_back(out);
+  }
+  launch_compact8(mask.data_ptr<bool>(), n, cursor.data_ptr<int64_t>(),
+                  srcs, dsts, (int)cols.size(), current_stream());
+  return outs;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hash_column", &hash_column,
         "combine a column into the running row hash");
@@ -358,6 +386,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hash_seed", &hash_seed, "seed a row-hash buffer");
   m.def("gb_mark_reps", &gb_mark_reps,
         "representative rows + h2 verification for hashed group-by");
+  m.def("compact_columns", &compact_columns,
+        "fused masked compaction of 8-byte columns");
   m.def("join_count", &join_count, "count matches per probe row");
   m.def("join_emit", &join_emit, "emit join pairs");
   m.def("join_mark_build", &join_mark_build, "mark matched build rows");

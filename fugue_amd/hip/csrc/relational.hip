@@ -948,3 +948,78 @@ void launch_gb_mark_reps(const int64_t* h1, const int64_t* h2,
 }
 
 }  // extern "C"
+
+// ------------------------------------------------------------------ //
+// fused stream compaction: write the masked rows of up to 8 eight-byte
+// columns in one pass.  Each block reserves a contiguous output range
+// for its chunk (one global atomic), so writes coalesce; output order is
+// block-nondeterministic (frames are unordered collections).
+// ------------------------------------------------------------------ //
+#define COMPACT_CHUNK (BLOCK * 16)
+
+__global__ __launch_bounds__(BLOCK) void compact8_kernel(
+    const bool* __restrict__ mask, int64_t n,
+    int64_t* __restrict__ cursor,  // [1]
+    const uint64_t* __restrict__ s0, const uint64_t* __restrict__ s1,
+    const uint64_t* __restrict__ s2, const uint64_t* __restrict__ s3,
+    const uint64_t* __restrict__ s4, const uint64_t* __restrict__ s5,
+    const uint64_t* __restrict__ s6, const uint64_t* __restrict__ s7,
+    uint64_t* __restrict__ d0, uint64_t* __restrict__ d1,
+    uint64_t* __restrict__ d2, uint64_t* __restrict__ d3,
+    uint64_t* __restrict__ d4, uint64_t* __restrict__ d5,
+    uint64_t* __restrict__ d6, uint64_t* __restrict__ d7,
+    int ncols) {
+  const uint64_t* srcs[8] = {s0, s1, s2, s3, s4, s5, s6, s7};
+  uint64_t* dsts[8] = {d0, d1, d2, d3, d4, d5, d6, d7};
+  __shared__ int lcount;
+  __shared__ long long lbase;
+  for (int64_t start = (int64_t)blockIdx.x * COMPACT_CHUNK; start < n;
+       start += (int64_t)gridDim.x * COMPACT_CHUNK) {
+    int64_t end = start + COMPACT_CHUNK;
+    if (end > n) end = n;
+    if (threadIdx.x == 0) lcount = 0;
+    __syncthreads();
+    int local = 0;
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
+      if (mask[i]) ++local;
+    atomicAdd(&lcount, local);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      lbase = lcount > 0
+                  ? (long long)atomicAdd((unsigned long long*)cursor,
+                                         (unsigned long long)lcount)
+                  : 0;
+      lcount = 0;
+    }
+    __syncthreads();
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+      if (!mask[i]) continue;
+      int64_t pos = (int64_t)lbase + atomicAdd(&lcount, 1);
+      for (int c = 0; c < ncols; ++c) dsts[c][pos] = srcs[c][i];
+    }
+    __syncthreads();
+  }
+}
+
+extern "C" {
+
+void launch_compact8(const bool* mask, int64_t n, int64_t* cursor,
+                     const uint64_t** srcs, uint64_t** dsts, int ncols,
+                     hipStream_t stream) {
+  int64_t blocks = (n + COMPACT_CHUNK - 1) / COMPACT_CHUNK;
+  if (blocks > MAX_GRID) blocks = MAX_GRID;
+  if (blocks < 1) blocks = 1;
+  const uint64_t* z = nullptr;
+  hipLaunchKernelGGL(
+      compact8_kernel, dim3((int)blocks), dim3(BLOCK), 0, stream, mask, n,
+      cursor, ncols > 0 ? srcs[0] : z, ncols > 1 ? srcs[1] : z,
+      ncols > 2 ? srcs[2] : z, ncols > 3 ? srcs[3] : z,
+      ncols > 4 ? srcs[4] : z, ncols > 5 ? srcs[5] : z,
+      ncols > 6 ? srcs[6] : z, ncols > 7 ? srcs[7] : z,
+      ncols > 0 ? dsts[0] : nullptr, ncols > 1 ? dsts[1] : nullptr,
+      ncols > 2 ? dsts[2] : nullptr, ncols > 3 ? dsts[3] : nullptr,
+      ncols > 4 ? dsts[4] : nullptr, ncols > 5 ? dsts[5] : nullptr,
+      ncols > 6 ? dsts[6] : nullptr, ncols > 7 ? dsts[7] : nullptr, ncols);
+}
+
+}  // extern "C"

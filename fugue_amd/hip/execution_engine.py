@@ -1259,6 +1259,33 @@ class HipExecutionEngine(ExecutionEngine):
             out = PandasDataFrame(res)
         return self.to_df(out, shard_replicated=self.is_distributed)
 
+    def _filter_rows(self, d: HipDataFrame, mask: "torch.Tensor") -> HipDataFrame:
+        """Masked row filter; fused single-pass compaction kernel when all
+        columns are 8-byte, null-free and non-string."""
+        if (
+            mask.device.type == "cuda"
+            and all(
+                (not isinstance(c, StringDeviceColumn))
+                and c.valid is None
+                and c.data.element_size() == 8
+                for c in d.columns_map.values()
+            )
+            and 0 < len(d.columns_map) <= 8
+        ):
+            from fugue_amd.hip.ext import get_ext
+
+            out_n = int(mask.sum().item())
+            names = list(d.columns_map.keys())
+            outs = get_ext().compact_columns(
+                mask, [d.col(nm).data for nm in names], out_n
+            )
+            cols = {
+                nm: DeviceColumn(t, None, d.col(nm).pa_type)
+                for nm, t in zip(names, outs)
+            }
+            return HipDataFrame.from_columns(cols, d.schema, self._device)
+        return d.gather_rows(mask.nonzero(as_tuple=True)[0])
+
     def _device_select(
         self,
         d: HipDataFrame,
@@ -1269,7 +1296,7 @@ class HipExecutionEngine(ExecutionEngine):
         cols = columns.replace_wildcard(d.schema)
         if where is not None:
             mask = filter_mask(where, d)
-            d = d.gather_rows(mask.nonzero(as_tuple=True)[0])
+            d = self._filter_rows(d, mask)
         if not cols.has_agg:
             if cols.is_distinct:
                 raise DeviceExprError("distinct select: fallback")
