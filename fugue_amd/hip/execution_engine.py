@@ -1085,7 +1085,7 @@ class HipExecutionEngine(ExecutionEngine):
             keep = valid_count == len(names)
         else:
             keep = valid_count > 0
-        return d.gather_rows(keep.nonzero(as_tuple=True)[0])
+        return self._filter_rows(d, keep)
 
     def fillna(
         self, df: DataFrame, value: Any, subset: Optional[List[str]] = None

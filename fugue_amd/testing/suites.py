@@ -491,3 +491,115 @@ class DataFrameTestSuite:
 
         with pytest.raises(FugueDataFrameEmptyError):
             df.peek_array()
+
+
+class _SelectTopHelper:
+    @staticmethod
+    def select_top(cursor, data):
+        return ArrayDataFrame([cursor.row], cursor.row_schema)
+
+
+def _suite_select_top(cursor, data):
+    return ArrayDataFrame([cursor.row], cursor.row_schema)
+
+
+class ExecutionEngineEdgeCaseTests:
+    """Tricky semantics from the reference suite (null partition keys,
+    NaT datetimes, nested/binary columns through map)."""
+
+    @classmethod
+    def make_engine(cls) -> ExecutionEngine:  # pragma: no cover
+        raise NotImplementedError
+
+    @pytest.fixture(autouse=True)
+    def _engine(self):
+        self.engine = self.make_engine()
+        yield
+
+    def test_map_with_null_keys(self):
+        e = self.engine
+        o = e.to_df(
+            ArrayDataFrame(
+                [[1.0, None, 1], [1.0, None, 0], [None, None, 2]],
+                "a:double,b:double,c:int",
+            )
+        )
+        c = e.map_engine.map_dataframe(
+            o, _suite_select_top, o.schema, PartitionSpec(by=["a", "b"], presort="c")
+        )
+        assert _df_eq(
+            c,
+            [[1.0, None, 0], [None, None, 2]],
+            "a:double,b:double,c:int",
+            throw=True,
+        )
+
+    def test_map_with_nat(self):
+        import datetime as _dt
+
+        dt = _dt.datetime(2024, 5, 6, 7, 8, 9)
+        e = self.engine
+        o = e.to_df(
+            ArrayDataFrame(
+                [
+                    [dt, 2, 1.0],
+                    [None, 2, None],
+                    [None, 1, None],
+                    [dt, 5, 1.0],
+                    [None, 4, None],
+                ],
+                "a:datetime,b:int,c:double",
+            )
+        )
+        c = e.map_engine.map_dataframe(
+            o, _suite_select_top, o.schema, PartitionSpec(by=["a", "c"], presort="b DESC")
+        )
+        assert _df_eq(
+            c,
+            [[None, 4, None], [dt, 5, 1.0]],
+            "a:datetime,b:int,c:double",
+            throw=True,
+        )
+
+    def test_map_nested_and_binary(self):
+        import pickle
+
+        e = self.engine
+        # nested list column round-trips through map
+        o = e.to_df(ArrayDataFrame([[3, [1, 2]]], "a:long,b:[int]"))
+        c = e.map_engine.map_dataframe(
+            o, _suite_select_top, o.schema, PartitionSpec(by=["a"])
+        )
+        assert c.as_array() == [[3, [1, 2]]]
+        # binary column
+        def bmap(cursor, data):
+            rows = [[r[0] + b"x"] for r in data.as_array()]
+            return ArrayDataFrame(rows, "a:bytes")
+
+        o2 = e.to_df(ArrayDataFrame([[b"a"], [b"b"]], "a:bytes"))
+        c2 = e.map_engine.map_dataframe(o2, bmap, "a:bytes", PartitionSpec())
+        assert sorted(c2.as_array()) == [[b"ax"], [b"bx"]]
+
+    def test_take_na_first(self):
+        e = self.engine
+        o = e.to_df(
+            ArrayDataFrame([[1.0], [None], [3.0]], "a:double")
+        )
+        t = e.take(o, 1, presort="a", na_position="first")
+        assert t.as_array()[0][0] is None
+
+    def test_aggregate_with_nulls(self):
+        e = self.engine
+        o = e.to_df(
+            ArrayDataFrame(
+                [[1, 1.0], [1, None], [2, None]], "k:long,v:double"
+            )
+        )
+        r = e.aggregate(
+            o,
+            PartitionSpec(by=["k"]),
+            [f.sum(col("v")).alias("s"), f.count(col("v")).alias("n")],
+        )
+        rows = sorted(r.as_array())
+        assert rows[0][0] == 1 and rows[0][2] == 1
+        assert rows[1][0] == 2 and rows[1][2] == 0

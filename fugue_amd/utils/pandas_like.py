@@ -181,9 +181,12 @@ def _group_starts(df: pd.DataFrame, cols: List[str]):
             a, b = prev[:, j], cur[:, j]
             eq = np.array(
                 [
-                    (x is None and y is None)
-                    or (isinstance(x, float) and isinstance(y, float) and np.isnan(x) and np.isnan(y))
-                    or (x is not None and y is not None and not _is_nan(x) and not _is_nan(y) and x == y)
+                    (bool(pd.isna(x)) and bool(pd.isna(y)))
+                    or (
+                        not bool(pd.isna(x))
+                        and not bool(pd.isna(y))
+                        and x == y
+                    )
                     for x, y in zip(a, b)
                 ]
             )

@@ -71,3 +71,14 @@ def test_sql_planner_on_gpu():
     )
     got = res.as_pandas().sort_values("s", ascending=False).reset_index(drop=True)
     np.testing.assert_allclose(got["s"].values, exp["s"].values, rtol=1e-9)
+
+
+from fugue_amd.testing.suites import ExecutionEngineEdgeCaseTests
+
+
+class TestHipGpuEdgeCases(ExecutionEngineEdgeCaseTests):
+    @classmethod
+    def make_engine(cls):
+        from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+        return HipExecutionEngine()
