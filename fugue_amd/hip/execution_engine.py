@@ -783,7 +783,12 @@ class HipExecutionEngine(ExecutionEngine):
                 "semi": "semi",
                 "anti": "anti",
             }[how]
-            pi, bi = dops.hash_join_indices(k1, k2, mode, h21, h22)
+            if how == "inner" and k1.numel() < k2.numel():
+                # build the smaller side (left), probe with the right
+                bi2, pi2 = dops.hash_join_indices(k2, k1, "inner", h22, h21)
+                pi, bi = pi2, bi2
+            else:
+                pi, bi = dops.hash_join_indices(k1, k2, mode, h21, h22)
             res = self._emit_join_output(
                 d1v, d2v, pi, bi, keys, output_schema, probe_is_left=True
             )
