@@ -783,7 +783,7 @@ class HipExecutionEngine(ExecutionEngine):
                 "semi": "semi",
                 "anti": "anti",
             }[how]
-            if how == "inner" and k1.numel() < k2.numel():
+            if how == "inner" and k1.numel() * 4 < k2.numel():
                 # build the smaller side (left), probe with the right
                 bi2, pi2 = dops.hash_join_indices(k2, k1, "inner", h22, h21)
                 pi, bi = pi2, bi2
