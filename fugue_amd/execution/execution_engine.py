@@ -101,11 +101,16 @@ class SQLEngine(EngineFacet):
     """SQL facet: execute a raw SQL statement over named dataframes.
 
     Reference parity: ``fugue/execution/execution_engine.py:183``.
+    Tables are an in-memory registry of named engine frames (on the
+    MI355X engine these are HBM-resident — SURVEY.md §5 checkpoint note:
+    "table" storage = named device-resident tables).
     """
 
     def __init__(self, execution_engine: "ExecutionEngine"):
         super().__init__(execution_engine)
         self._uid = "_" + str(id(self))
+        self._tables: Dict[str, DataFrame] = {}
+        self._tables_lock = RLock()
 
     @property
     def dialect(self) -> Optional[str]:
@@ -126,7 +131,8 @@ class SQLEngine(EngineFacet):
         ...
 
     def table_exists(self, table: str) -> bool:
-        raise NotImplementedError(f"{self} doesn't support tables")
+        with self._tables_lock:
+            return table in self._tables
 
     def save_table(
         self,
@@ -136,10 +142,16 @@ class SQLEngine(EngineFacet):
         partition_spec: Optional[PartitionSpec] = None,
         **kwargs: Any,
     ) -> None:
-        raise NotImplementedError(f"{self} doesn't support tables")
+        with self._tables_lock:
+            if table in self._tables and mode == "error":
+                raise ValueError(f"table {table} exists")
+            self._tables[table] = self.execution_engine.persist(df)
 
     def load_table(self, table: str, **kwargs: Any) -> DataFrame:
-        raise NotImplementedError(f"{self} doesn't support tables")
+        with self._tables_lock:
+            if table not in self._tables:
+                raise KeyError(f"table {table} does not exist")
+            return self._tables[table]
 
 
 class MapEngine(EngineFacet):

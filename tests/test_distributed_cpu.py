@@ -271,3 +271,46 @@ def test_distributed_parquet_parts(tmp_path):
         assert len([f for f in files if f.startswith("part-")]) == 2
     finally:
         os.environ.pop("FUGUE_TEST_TMP", None)
+
+
+def _empty_shard_job(rank: int):
+    import fugue_amd.api as fa
+    from fugue_amd.column.expressions import col
+    from fugue_amd.column import functions as f
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    # rank 1's shard is emptied by the filter (values 50..99 live there)
+    pdf = pd.DataFrame(dict(k=np.arange(100) % 7, v=np.arange(100.0)))
+    d = e.to_df(pdf)
+    filtered = e.filter(d, col("v") < 50)
+    agg = fa.aggregate(
+        filtered, partition_by="k", engine=e, s=f.sum(col("v")), as_fugue=True
+    )
+    local = e._gather_all(agg)
+    j = fa.join(
+        agg,
+        pd.DataFrame(dict(k=np.arange(7), w=np.arange(7.0))),
+        how="inner",
+        engine=e,
+        as_fugue=True,
+    )
+    jn = e.comm.allreduce_sum(j.count())
+    return dict(rows=sorted(map(tuple, local.as_array())), join_count=jn)
+
+
+def test_distributed_empty_shard():
+    results = run_distributed(_empty_shard_job, 29523)
+    pdf = pd.DataFrame(dict(k=np.arange(100) % 7, v=np.arange(100.0)))
+    sub = pdf[pdf.v < 50]
+    expected = sorted(
+        map(
+            tuple,
+            sub.groupby("k", as_index=False).agg(s=("v", "sum")).values.tolist(),
+        )
+    )
+    for rank, got in results.items():
+        assert [tuple(map(float, r)) for r in got["rows"]] == [
+            tuple(map(float, r)) for r in expected
+        ]
+        assert got["join_count"] == 7

@@ -210,3 +210,24 @@ def test_runtime_exception_traceback():
 
     with pytest.raises(RuntimeError, match="inner failure"):
         transform(pd.DataFrame(dict(a=[1])), bad, schema="*")
+
+
+def test_yield_file_and_table_cross_workflow():
+    import tempfile
+
+    from fugue_amd.execution import NativeExecutionEngine
+
+    with tempfile.TemporaryDirectory() as tmp:
+        conf = {FUGUE_CONF_WORKFLOW_CHECKPOINT_PATH: tmp}
+        engine = NativeExecutionEngine(conf)
+        dag1 = FugueWorkflow()
+        dag1.df([[1], [2]], "x:long").yield_file_as("f1")
+        dag1.df([[3]], "x:long").yield_table_as("t1")
+        res1 = dag1.run(engine)
+        # second workflow consumes the yields
+        dag2 = FugueWorkflow()
+        a = dag2.df(res1["f1"])
+        b = dag2.df(res1["t1"])
+        a.union(b, distinct=False).yield_dataframe_as("out")
+        res2 = dag2.run(engine)
+        assert sorted(r[0] for r in res2["out"].result.as_array()) == [1, 2, 3]
