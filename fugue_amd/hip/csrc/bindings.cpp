@@ -36,11 +36,12 @@ void launch_gb_aggregate_part(const int64_t*, const double*, const int32_t*,
                               int, int64_t, const int64_t*, int64_t, int64_t*,
                               double*, int64_t*, int64_t, hipStream_t);
 void launch_gb_part_scatter_staged(const int64_t*, const double*, int64_t,
-                                   int, int64_t*, int64_t*, double*,
+                                   int, int64_t*, int64_t*, double*, int64_t,
                                    hipStream_t);
 void launch_gb_aggregate_part_big(const int64_t*, const double*,
                                   const int32_t*, int64_t, int64_t*,
-                                  double*, int64_t*, int64_t, hipStream_t);
+                                  double*, int64_t*, int64_t, int64_t,
+                                  hipStream_t);
 void launch_join_count(const int64_t*, int64_t, const int64_t*,
                        const int64_t*, const int64_t*, const int32_t*,
                        const int32_t*, int64_t, int32_t*, hipStream_t);
@@ -177,7 +178,7 @@ std::vector<at::Tensor> gb_aggregate(at::Tensor keys, at::Tensor vals,
 
 std::vector<at::Tensor> gb_aggregate_partitioned(
     at::Tensor keys, at::Tensor vals, at::Tensor ops, int64_t num_parts,
-    int64_t tsize) {
+    int64_t tsize, int64_t scatter_chunk, int64_t agg_chunk) {
   check_gpu(keys, "keys");
   check_gpu(vals, "vals");
   TORCH_CHECK((tsize & (tsize - 1)) == 0, "tsize must be a power of 2");
@@ -207,7 +208,7 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
     launch_gb_part_scatter_staged(
         keys.data_ptr<int64_t>(), vals.data_ptr<double>(), n, shift,
         cursor.data_ptr<int64_t>(), pkeys.data_ptr<int64_t>(),
-        pvals.data_ptr<double>(), stream);
+        pvals.data_ptr<double>(), scatter_chunk, stream);
   } else {
     launch_gb_part_scatter(keys.data_ptr<int64_t>(), vals.data_ptr<double>(),
                            n_aggs, n, shift, cursor.data_ptr<int64_t>(),
@@ -223,7 +224,8 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
     launch_gb_aggregate_part_big(
         pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),
         ops.data_ptr<int32_t>(), n, tkeys.data_ptr<int64_t>(),
-        gaggs.data_ptr<double>(), gcount.data_ptr<int64_t>(), tsize, stream);
+        gaggs.data_ptr<double>(), gcount.data_ptr<int64_t>(), tsize,
+        agg_chunk, stream);
   } else {
     launch_gb_aggregate_part(
         pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),

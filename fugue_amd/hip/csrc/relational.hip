@@ -490,15 +490,16 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
     int64_t n, int shift,
     int64_t* __restrict__ cursor,
     int64_t* __restrict__ out_keys,
-    double* __restrict__ out_vals) {
+    double* __restrict__ out_vals,
+    int64_t chunk) {
   __shared__ int lhist[STAGE_P];
   __shared__ int64_t lbase[STAGE_P];
   __shared__ int lcnt[STAGE_P];
   __shared__ int64_t skey[STAGE_P * STAGE_E];
   __shared__ double sval[STAGE_P * STAGE_E];
-  for (int64_t start = (int64_t)blockIdx.x * SCATTER_CHUNK; start < n;
-       start += (int64_t)gridDim.x * SCATTER_CHUNK) {
-    int64_t end = start + SCATTER_CHUNK;
+  for (int64_t start = (int64_t)blockIdx.x * chunk; start < n;
+       start += (int64_t)gridDim.x * chunk) {
+    int64_t end = start + chunk;
     if (end > n) end = n;
     for (int i = threadIdx.x; i < STAGE_P; i += blockDim.x) {
       lhist[i] = 0;
@@ -565,14 +566,14 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
     int64_t* __restrict__ tkeys,
     double* __restrict__ gaggs,
     int64_t* __restrict__ gcount,
-    int64_t tsize) {
+    int64_t tsize, int64_t chunk) {
   __shared__ int64_t lkeys[LDS_SLOTS_BIG];
   __shared__ double laggs[LDS_SLOTS_BIG];
   __shared__ int lcount[LDS_SLOTS_BIG];
   bool is_count = ops[0] == 3;
-  for (int64_t start = (int64_t)blockIdx.x * AGG_CHUNK; start < n;
-       start += (int64_t)gridDim.x * AGG_CHUNK) {
-    int64_t end = start + AGG_CHUNK;
+  for (int64_t start = (int64_t)blockIdx.x * chunk; start < n;
+       start += (int64_t)gridDim.x * chunk) {
+    int64_t end = start + chunk;
     if (end > n) end = n;
     for (int i = threadIdx.x; i < LDS_SLOTS_BIG; i += blockDim.x) {
       lkeys[i] = GB_EMPTY;
@@ -620,13 +621,14 @@ extern "C" {
 void launch_gb_part_scatter_staged(const int64_t* keys, const double* vals,
                                    int64_t n, int shift, int64_t* cursor,
                                    int64_t* out_keys, double* out_vals,
-                                   hipStream_t stream) {
-  int64_t blocks = (n + SCATTER_CHUNK - 1) / SCATTER_CHUNK;
+                                   int64_t chunk, hipStream_t stream) {
+  if (chunk <= 0) chunk = SCATTER_CHUNK;
+  int64_t blocks = (n + chunk - 1) / chunk;
   if (blocks > MAX_GRID) blocks = MAX_GRID;
   if (blocks < 1) blocks = 1;
   hipLaunchKernelGGL(gb_part_scatter_staged_kernel, dim3((int)blocks),
                      dim3(BLOCK), 0, stream, keys, vals, n, shift, cursor,
-                     out_keys, out_vals);
+                     out_keys, out_vals, chunk);
 }
 
 void launch_gb_aggregate_part_big(const int64_t* part_keys,
@@ -634,13 +636,14 @@ void launch_gb_aggregate_part_big(const int64_t* part_keys,
                                   const int32_t* ops, int64_t n,
                                   int64_t* tkeys, double* gaggs,
                                   int64_t* gcount, int64_t tsize,
-                                  hipStream_t stream) {
-  int64_t blocks = (n + AGG_CHUNK - 1) / AGG_CHUNK;
+                                  int64_t chunk, hipStream_t stream) {
+  if (chunk <= 0) chunk = AGG_CHUNK;
+  int64_t blocks = (n + chunk - 1) / chunk;
   if (blocks > MAX_GRID) blocks = MAX_GRID;
   if (blocks < 1) blocks = 1;
   hipLaunchKernelGGL(gb_aggregate_part_big_kernel, dim3((int)blocks),
                      dim3(BLOCK), 0, stream, part_keys, part_vals, ops, n,
-                     tkeys, gaggs, gcount, tsize);
+                     tkeys, gaggs, gcount, tsize, chunk);
 }
 
 }  // extern "C"

@@ -367,8 +367,12 @@ def groupby_aggregate(
             num_parts = 512
         else:
             num_parts = min(4096, _next_pow2(max(16, expected_groups // 512)))
+        import os as _os
+
+        sc_chunk = int(_os.environ.get("FUGUE_GB_SCATTER_CHUNK", "0"))
+        ag_chunk = int(_os.environ.get("FUGUE_GB_AGG_CHUNK", "0"))
         tkeys, gaggs, gcount = ext.gb_aggregate_partitioned(
-            packed, vals, ops, num_parts, tsize
+            packed, vals, ops, num_parts, tsize, sc_chunk, ag_chunk
         )
     else:
         use_lds = expected_groups <= 100_000 and sum_count_only
