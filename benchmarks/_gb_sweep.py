@@ -4,7 +4,11 @@ Times `ext.gb_aggregate_partitioned` directly on the bench's shape
 (125M rows, 1M groups, 1 fp64 SUM) across runtime scatter/agg chunk sizes.
 Run under gpurun; prints one line per config.
 """
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
@@ -19,7 +23,7 @@ def main() -> None:
     keys = torch.randint(0, 1_000_000, (n,), device=dev, generator=g)
     vals = torch.rand((1, n), device=dev, dtype=torch.float64, generator=g)
     ops = torch.tensor([0], dtype=torch.int32, device=dev)  # SUM
-    tsize = 4 * 1_000_000
+    tsize = 1 << 22  # 4.2M slots for ~1M groups
 
     def run(sc, ag, iters=6):
         torch.cuda.synchronize()
@@ -34,14 +38,10 @@ def main() -> None:
 
     base = run(0, 0)
     print(f"baseline (compiled defaults): {base:.3f} ms", flush=True)
-    for sc in (2048, 4096, 8192, 16384, 32768):
-        t = run(sc, 0)
-        print(f"scatter_chunk={sc:6d} agg=default: {t:.3f} ms", flush=True)
-    for ag in (32768, 65536, 131072, 262144):
+    for ag in (4096, 8192, 16384, 32768):
         t = run(0, ag)
         print(f"scatter=default agg_chunk={ag:6d}: {t:.3f} ms", flush=True)
-    # joint best candidates
-    for sc, ag in ((8192, 131072), (16384, 131072), (16384, 262144)):
+    for sc, ag in ((4096, 16384), (4096, 32768), (2048, 16384)):
         t = run(sc, ag)
         print(f"scatter_chunk={sc:6d} agg_chunk={ag:6d}: {t:.3f} ms", flush=True)
 

@@ -556,7 +556,7 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
 // atomic).  Chunk-striding restores full grid parallelism regardless of
 // the partition count.  4096 slots / 80KB LDS → 2 blocks/CU.
 #define LDS_SLOTS_BIG 4096
-#define AGG_CHUNK (BLOCK * 256)  // 65536 rows per chunk
+#define AGG_CHUNK (BLOCK * 128)  // 32768 rows per chunk (sweep-tuned)
 
 __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
     const int64_t* __restrict__ part_keys,
