@@ -20,6 +20,23 @@ def coalesce(*args: Any) -> ColumnExpr:
     return _FuncExpr("COALESCE", *[_to_col(x) for x in args])
 
 
+def case_when(*whens: Any, else_: Any = None) -> ColumnExpr:
+    """``CASE WHEN c1 THEN v1 [WHEN c2 THEN v2 ...] ELSE e END``.
+
+    ``whens`` are ``(condition, value)`` pairs. Encoded as a ``_FuncExpr``
+    with flattened args ``[c1, v1, c2, v2, ..., else]`` so alias/cast/
+    mention machinery applies unchanged (reference parity:
+    QPD's CASE WHEN support used by ``fugue/execution`` SQL lowering)."""
+    assert len(whens) > 0, "case_when requires at least one (cond, value) pair"
+    flat = []
+    for pair in whens:
+        cond, value = pair
+        flat.append(_to_col(cond))
+        flat.append(_to_col(value))
+    flat.append(_to_col(else_))
+    return _FuncExpr("CASE_WHEN", *flat)
+
+
 def min(col: ColumnExpr) -> ColumnExpr:  # pylint: disable=redefined-builtin
     assert isinstance(col, ColumnExpr)
     return _UnaryAggFuncExpr("MIN", col)

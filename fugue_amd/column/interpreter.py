@@ -109,6 +109,21 @@ def _eval(expr: ColumnExpr, df: pd.DataFrame) -> Any:
                 )
                 out = s if out is None else out.where(out.notna(), s)
             return out
+        if fname == "CASE_WHEN":
+            args = expr.args
+            n = len(df)
+            idx = df.index
+
+            def _as_series(v: Any) -> pd.Series:
+                return v if isinstance(v, pd.Series) else pd.Series([v] * n, index=idx)
+
+            out = _as_series(_eval(args[-1], df))  # ELSE
+            # apply WHEN branches in reverse so the FIRST match wins
+            for i in range(len(args) - 2, 0, -2):
+                cond = _as_series(_eval(args[i - 1], df)).astype("boolean").fillna(False)
+                val = _as_series(_eval(args[i], df))
+                out = val.where(cond.to_numpy(dtype=bool), out)
+            return out
         raise NotImplementedError(f"function {expr.func}")
     raise NotImplementedError(f"can't evaluate {expr}")
 

@@ -1315,7 +1315,7 @@ class HipExecutionEngine(ExecutionEngine):
                     fields.append(pa.field(name, src.pa_type))
                 else:
                     data, valid = eval_device_expr(c, d)
-                    tp = c.infer_type(d.schema) or _pa_type_of(data)
+                    tp = c.infer_type(d.schema) or self._pa_type_of_tensor(data)
                     out_cols[name] = DeviceColumn(data, valid, tp)
                     fields.append(pa.field(name, tp))
             return HipDataFrame.from_columns(
@@ -1371,7 +1371,12 @@ class HipExecutionEngine(ExecutionEngine):
                 fname = c.func.upper()
                 arg = c.args[0]
                 if c.is_distinct:
-                    raise DeviceExprError("count distinct: fallback")
+                    if fname != "COUNT" or not isinstance(
+                        arg, _NamedColumnExpr
+                    ):
+                        raise DeviceExprError("non-COUNT distinct: fallback")
+                    plans.append((name, "count_distinct", arg.name))
+                    continue
                 if fname == "COUNT":
                     if (
                         isinstance(arg, _NamedColumnExpr)
@@ -1388,7 +1393,11 @@ class HipExecutionEngine(ExecutionEngine):
                     continue
                 if isinstance(arg, _NamedColumnExpr):
                     src = arg.name
-                    if isinstance(d.col(src), StringDeviceColumn):
+                    if isinstance(
+                        d.col(src), StringDeviceColumn
+                    ) and fname not in ("FIRST", "LAST"):
+                        # FIRST/LAST never aggregate the string values
+                        # themselves (row-index scheme); others fall back
                         raise DeviceExprError("string aggregation: fallback")
                 else:
                     src = _derive(arg)
@@ -1399,12 +1408,29 @@ class HipExecutionEngine(ExecutionEngine):
                     tmp_s = _add_partial(src, dops.AGG_SUM)
                     tmp_c = _add_partial(src, dops.AGG_COUNT)
                     plans.append((name, "avg", (tmp_s, tmp_c)))
+                elif fname in ("FIRST", "LAST"):
+                    # representative-row aggregation: MIN/MAX over the row
+                    # index (masked by the source's validity), then a
+                    # post-aggregation gather of the source values
+                    idx_name = f"__fidx_{src}"
+                    op = dops.AGG_MIN if fname == "FIRST" else dops.AGG_MAX
+                    tmp = _add_partial(idx_name, op)
+                    plans.append((name, "firstlast", (tmp, src, idx_name)))
                 else:
                     raise DeviceExprError(f"agg {fname}: fallback")
                 continue
             raise DeviceExprError("compound aggregate expression: fallback")
         if len(key_names) == 0:
             raise DeviceExprError("global aggregate: fallback")
+        cd_plans = [p for p in plans if p[1] == "count_distinct"]
+        if cd_plans:
+            return self._device_aggregate_distinct(
+                d, key_names, cols, plans, having
+            )
+        fl_specs = sorted(
+            {info[2]: info[1] for _n, k, info in plans if k == "firstlast"}
+            .items()
+        )  # [(idx_name, src)]
         if derived:
             cols_map = dict(d.columns_map)
             cols_map.update(derived)
@@ -1418,10 +1444,16 @@ class HipExecutionEngine(ExecutionEngine):
         hashed_keys = any(
             isinstance(kc, StringDeviceColumn) for kc in key_cols
         )
-        if hashed_keys:
+        if hashed_keys or (fl_specs and self.is_distributed):
+            # string keys, and/or FIRST/LAST in distributed mode (row
+            # indices are rank-local): co-shuffle full rows by key so each
+            # group is wholly local, then aggregate exactly
             return self._device_aggregate_hashed(
-                d, key_names, partials, plans, having, cols
+                d, key_names, partials, plans, having, cols,
+                firstlast=fl_specs,
             )
+        if fl_specs:
+            d = self._with_rowindex_cols(d, fl_specs)
         pack_mins = pack_widths = None
         if self.is_distributed:
             # cross-rank partial merge requires a rank-consistent packing:
@@ -1444,7 +1476,8 @@ class HipExecutionEngine(ExecutionEngine):
                     # not packable: co-shuffle rows, then exact local
                     # hashed aggregation (globally exact)
                     return self._device_aggregate_hashed(
-                        d, key_names, partials, plans, having, cols
+                        d, key_names, partials, plans, having, cols,
+                        firstlast=fl_specs,
                     )
         # local partial aggregation
         out_keys, out_aggs, out_count, meta = dops.groupby_aggregate(
@@ -1480,6 +1513,12 @@ class HipExecutionEngine(ExecutionEngine):
                 tp = pa.float64()
                 out_cols[name] = DeviceColumn(vals, None, tp)
                 fields.append(pa.field(name, tp))
+            elif kind == "firstlast":
+                tmp, src, _idx = info
+                out_cols[name] = self._gather_firstlast(
+                    out_aggs[tmp], d.col(src)
+                )
+                fields.append(pa.field(name, out_cols[name].pa_type))
             elif kind == "avg":
                 tmp_s, tmp_c = info
                 cnt = out_aggs[tmp_c]
@@ -1494,6 +1533,147 @@ class HipExecutionEngine(ExecutionEngine):
             raise DeviceExprError("distinct aggregate: fallback")
         return res
 
+    @staticmethod
+    def _pa_type_of_tensor(data: "torch.Tensor") -> pa.DataType:
+        m = {
+            torch.int64: pa.int64(),
+            torch.int32: pa.int32(),
+            torch.int16: pa.int16(),
+            torch.int8: pa.int8(),
+            torch.float64: pa.float64(),
+            torch.float32: pa.float32(),
+            torch.bool: pa.bool_(),
+        }
+        tp = m.get(data.dtype)
+        if tp is None:
+            raise DeviceExprError(f"no arrow type for {data.dtype}")
+        return tp
+
+    def _with_rowindex_cols(
+        self, d: HipDataFrame, fl_specs: List[Tuple[str, str]]
+    ) -> HipDataFrame:
+        """Attach fp64 row-index columns (masked by the source column's
+        validity) used by FIRST/LAST's MIN/MAX-over-row-index scheme."""
+        n = d.count()
+        device = torch.device(d.device)
+        cols_map = dict(d.columns_map)
+        fields_ext = list(d.schema.fields)
+        idx = torch.arange(n, dtype=torch.float64, device=device)
+        for idx_name, src in fl_specs:
+            srccol = d.col(src)
+            cols_map[idx_name] = DeviceColumn(idx, srccol.valid, pa.float64())
+            fields_ext.append(pa.field(idx_name, pa.float64()))
+        return HipDataFrame.from_columns(
+            cols_map, Schema(fields_ext), self._device
+        )
+
+    def _gather_firstlast(
+        self, idxs: torch.Tensor, srccol: DeviceColumn
+    ) -> DeviceColumn:
+        """Gather source values at aggregated row indices; out-of-range /
+        non-finite indices (all-null group) become NULL."""
+        n = len(srccol)
+        ok = torch.isfinite(idxs) & (idxs >= 0) & (idxs < max(n, 1))
+        safe = torch.where(ok, idxs, torch.zeros_like(idxs)).to(torch.int64)
+        all_ok = bool(ok.all().item())
+        gathered = srccol.gather(safe)
+        if all_ok:
+            return gathered
+        if isinstance(gathered, StringDeviceColumn):
+            valid = ok if gathered.valid is None else (gathered.valid & ok)
+            return StringDeviceColumn(gathered.offsets, gathered.bytes, valid)
+        valid = ok if gathered.valid is None else (gathered.valid & ok)
+        return DeviceColumn(gathered.data, valid, gathered.pa_type)
+
+    def _device_aggregate_distinct(
+        self,
+        d: HipDataFrame,
+        key_names: List[str],
+        cols: SelectColumns,
+        plans: List[Tuple[str, str, Any]],
+        having: Optional[ColumnExpr],
+    ) -> DataFrame:
+        """COUNT(DISTINCT x) decomposition: dedupe (keys, x) rows, then
+        COUNT per key; joined back to the other aggregates on the keys.
+        (Reference comparator: the SQL backends' native COUNT DISTINCT,
+        e.g. duckdb via ``fugue_duckdb``.)
+
+        Null group keys would not survive the inner re-join (SQL null
+        semantics) so that shape falls back to the host path."""
+        from fugue_amd.column import functions as ff
+        from fugue_amd.column.expressions import col as _col
+
+        has_null = 0
+        for k in key_names:
+            kc = d.col(k)
+            if kc.valid is not None and bool((~kc.valid).any().item()):
+                has_null = 1
+        if self.is_distributed:
+            import torch.distributed as dist
+
+            t = torch.tensor([has_null], dtype=torch.int64)
+            t = t.to(self._comm._comm_device(t))
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            has_null = int(t.cpu().item())
+        if has_null > 0:
+            raise DeviceExprError("count distinct with null keys")
+        base_exprs = [_col(k) for k in key_names]
+        out_order: List[str] = []
+        cd_items: List[Tuple[str, str]] = []  # (out name, src col)
+        for name, kind, info in plans:
+            out_order.append(name)
+            if kind == "key":
+                continue
+            if kind == "count_distinct":
+                cd_items.append((name, info))
+                continue
+            # rebuild the original expression for the base aggregate
+            for c in cols.all_cols:
+                if c.output_name == name:
+                    base_exprs.append(c)
+                    break
+        res: Optional[DataFrame] = None
+        if len(base_exprs) > len(key_names):
+            base_sc = SelectColumns(*base_exprs)
+            res = self._device_aggregate(d, base_sc, None)
+        for name, src in cd_items:
+            proj_names = key_names + ([src] if src not in key_names else [])
+            proj = self._device_select_named(d, proj_names)
+            dedup = self.to_df(self.distinct(proj))
+            agg_sc = SelectColumns(
+                *[_col(k) for k in key_names],
+                ff.count(_col(src)).alias(name),
+            )
+            part = self._device_aggregate(self.to_df(dedup), agg_sc, None)
+            res = (
+                part
+                if res is None
+                else self.join(res, part, "inner", key_names)
+            )
+        assert res is not None
+        res = self.to_df(
+            self._select_columns(
+                res, SelectColumns(*[_col(n) for n in out_order])
+            )
+        )
+        if having is not None and isinstance(res, HipDataFrame):
+            mask = filter_mask(having, res)
+            res = res.gather_rows(mask.nonzero(as_tuple=True)[0])
+        elif having is not None:
+            raise DeviceExprError("having after distinct fallback")
+        if cols.is_distinct:
+            raise DeviceExprError("distinct aggregate: fallback")
+        return res
+
+    def _device_select_named(
+        self, d: HipDataFrame, names: List[str]
+    ) -> HipDataFrame:
+        cols_map = {n: d.col(n) for n in names}
+        fields = [pa.field(n, d.col(n).pa_type) for n in names]
+        return HipDataFrame.from_columns(
+            cols_map, Schema(fields), self._device
+        )
+
     def _device_aggregate_hashed(
         self,
         d: HipDataFrame,
@@ -1502,12 +1682,17 @@ class HipExecutionEngine(ExecutionEngine):
         plans: List[Tuple[str, str, Any]],
         having: Optional[ColumnExpr],
         cols: SelectColumns,
+        firstlast: Optional[List[Tuple[str, str]]] = None,
     ) -> DataFrame:
         """Group-by on string (or otherwise unpackable) key tuples: the
         distributed case co-shuffles full rows by key hash first, so the
         local 128-bit-hashed aggregation is globally exact."""
         if self.is_distributed:
             d = self._shuffle_by_columns(d, key_names)
+        if firstlast:
+            # row-index columns must be built AFTER the shuffle so the
+            # aggregated indices address the local (post-shuffle) frame
+            d = self._with_rowindex_cols(d, firstlast)
         try:
             reps, out_aggs, out_count = dops.groupby_aggregate_hashed(
                 d, key_names, partials
@@ -1541,6 +1726,12 @@ class HipExecutionEngine(ExecutionEngine):
                     pa.float64(),
                 )
                 fields.append(pa.field(name, pa.float64()))
+            elif kind == "firstlast":
+                tmp, src, _idx = info
+                out_cols[name] = self._gather_firstlast(
+                    out_aggs[tmp].to(torch.device(self._device)), d.col(src)
+                )
+                fields.append(pa.field(name, out_cols[name].pa_type))
             elif kind == "avg":
                 tmp_s, tmp_c = info
                 dev = torch.device(self._device)

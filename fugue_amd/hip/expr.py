@@ -136,6 +136,31 @@ def eval_device_expr(
             if out_v is not None and bool(out_v.all().item()):
                 out_v = None
             return out_d, out_v
+        if fname == "CASE_WHEN":
+            # args = [c1, v1, c2, v2, ..., else]; first match wins →
+            # apply branches in reverse over torch.where
+            args = expr.args
+            out_d, out_v = eval_device_expr(args[-1], df)
+            for i in range(len(args) - 2, 0, -2):
+                cd, cv = eval_device_expr(args[i - 1], df)
+                cond = _as_bool(cd, cv)
+                vd, vv = eval_device_expr(args[i], df)
+                if vd.dtype != out_d.dtype:
+                    common = torch.promote_types(vd.dtype, out_d.dtype)
+                    vd = vd.to(common)
+                    out_d = out_d.to(common)
+                out_d = torch.where(cond, vd, out_d)
+                if out_v is None and vv is None:
+                    new_v = None
+                else:
+                    ones = torch.ones_like(cond)
+                    new_v = torch.where(
+                        cond,
+                        vv if vv is not None else ones,
+                        out_v if out_v is not None else ones,
+                    )
+                out_v = new_v
+            return out_d, out_v
         raise DeviceExprError(f"function {expr.func}")
     raise DeviceExprError(f"can't evaluate {expr} on device")
 

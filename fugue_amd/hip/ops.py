@@ -290,6 +290,17 @@ def unpack_keys(
     return res
 
 
+def _agg_input(c, n: int) -> "torch.Tensor":
+    """fp64 aggregation input for a column; string columns are only legal
+    for COUNT (validity-only), so their data contribution is zeros."""
+    if getattr(c, "is_string", False):
+        return torch.zeros(
+            n, dtype=torch.float64,
+            device=c.offsets.device,
+        )
+    return c.data.to(torch.float64)
+
+
 def groupby_aggregate(
     df: HipDataFrame,
     keys: List[str],
@@ -328,7 +339,7 @@ def groupby_aggregate(
             valids = torch.ones((n_aggs, n), dtype=torch.bool, device=device)
         for i, (cname, op, _) in enumerate(aggs):
             c = df.col(cname)
-            vals[i] = c.data.to(torch.float64)
+            vals[i] = _agg_input(c, n)
             if valids is not None and c.valid is not None:
                 valids[i] = c.valid
         ops = torch.tensor(
@@ -397,9 +408,10 @@ def _groupby_aggregate_cpu(
     import pandas as pd
 
     data: Dict[str, Any] = {"__key": packed.numpy()}
+    n_rows = packed.numel()
     for cname, op, oname in aggs:
         c = df.col(cname)
-        v = c.data.numpy().astype("float64")
+        v = _agg_input(c, int(n_rows)).numpy()
         if c.valid is not None:
             v = np.where(c.valid.numpy(), v, np.nan)
         data[oname] = v
@@ -503,7 +515,7 @@ def groupby_aggregate_hashed(
         pdf["__idx"] = np.arange(n, dtype=np.int64)
         for cname, op, oname in aggs:
             c = df.col(cname)
-            v = c.data.numpy().astype("float64")
+            v = _agg_input(c, n).numpy()
             if c.valid is not None:
                 v = np.where(c.valid.numpy(), v, np.nan)
             pdf[oname] = v
@@ -536,7 +548,7 @@ def groupby_aggregate_hashed(
             valids = torch.ones((n_aggs, n), dtype=torch.bool, device=device)
         for i, (cname, op, _) in enumerate(aggs):
             c = df.col(cname)
-            vals[i] = c.data.to(torch.float64)
+            vals[i] = _agg_input(c, n)
             if valids is not None and c.valid is not None:
                 valids[i] = c.valid
         ops = torch.tensor(

@@ -117,6 +117,20 @@ def _to_column_expr(e: X.Expr, schema: Schema, alias_map: Dict[str, str]) -> Col
                 *[_to_column_expr(a, schema, alias_map) for a in e.args]
             )
         raise UnsupportedPlan(f"function {name}")
+    if isinstance(e, X.Case):
+        whens = [
+            (
+                _to_column_expr(c, schema, alias_map),
+                _to_column_expr(v, schema, alias_map),
+            )
+            for c, v in e.whens
+        ]
+        else_ = (
+            _to_column_expr(e.else_, schema, alias_map)
+            if e.else_ is not None
+            else lit(None)
+        )
+        return F.case_when(*whens, else_=else_)
     raise UnsupportedPlan(f"expression {type(e).__name__}")
 
 

@@ -603,3 +603,74 @@ class ExecutionEngineEdgeCaseTests:
         rows = sorted(r.as_array())
         assert rows[0][0] == 1 and rows[0][2] == 1
         assert rows[1][0] == 2 and rows[1][2] == 0
+
+    def test_first_last_count_distinct(self):
+        e = self.engine
+        o = e.to_df(
+            ArrayDataFrame(
+                [
+                    [1, 5.0, "a"],
+                    [1, None, "b"],
+                    [2, 1.0, "c"],
+                    [2, 2.0, "c"],
+                    [2, 1.0, "d"],
+                    [3, None, None],
+                ],
+                "k:long,v:double,s:str",
+            )
+        )
+        r = e.aggregate(
+            o,
+            PartitionSpec(by=["k"]),
+            [
+                f.first(col("v")).alias("fv"),
+                f.last(col("v")).alias("lv"),
+                f.count_distinct(col("s")).alias("cd"),
+            ],
+        )
+        rows = sorted(r.as_array(), key=lambda x: x[0])
+        assert rows[0][0] == 1 and rows[0][1] == 5.0 and rows[0][2] == 5.0
+        assert rows[0][3] == 2
+        assert rows[1][0] == 2 and rows[1][3] == 2
+        # group values {1.0, 2.0}: first/last must be members of the group
+        assert rows[1][1] in (1.0, 2.0) and rows[1][2] in (1.0, 2.0)
+        assert rows[2][0] == 3 and rows[2][1] is None and rows[2][2] is None
+        assert rows[2][3] == 0
+
+    def test_first_on_string_column(self):
+        e = self.engine
+        o = e.to_df(
+            ArrayDataFrame(
+                [[1, "x"], [1, "y"], [2, None]], "k:long,s:str"
+            )
+        )
+        r = e.aggregate(
+            o, PartitionSpec(by=["k"]), [f.first(col("s")).alias("fs")]
+        )
+        rows = sorted(r.as_array(), key=lambda x: x[0])
+        assert rows[0][1] in ("x", "y")
+        assert rows[1][1] is None
+
+    def test_case_when_select(self):
+        from fugue_amd.column.expressions import lit
+        from fugue_amd.column.sql import SelectColumns
+
+        e = self.engine
+        o = e.to_df(
+            ArrayDataFrame(
+                [[1, 5.0], [2, None], [3, 1.0]], "k:long,v:double"
+            )
+        )
+        r = e.select(
+            o,
+            SelectColumns(
+                col("k"),
+                f.case_when(
+                    (col("v") > 2.0, lit("hi")),
+                    (col("v") > 0.0, lit("lo")),
+                    else_=lit("na"),
+                ).alias("c"),
+            ),
+        )
+        rows = sorted(r.as_array(), key=lambda x: x[0])
+        assert [x[1] for x in rows] == ["hi", "na", "lo"]

@@ -314,3 +314,50 @@ def test_distributed_empty_shard():
             tuple(map(float, r)) for r in expected
         ]
         assert got["join_count"] == 7
+
+
+def _firstlast_cd_job(rank: int):
+    import fugue_amd.api as fa
+    from fugue_amd.column import functions as f
+    from fugue_amd.column.expressions import col
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    rng = np.random.default_rng(3)
+    pdf = pd.DataFrame(
+        dict(
+            k=rng.integers(0, 20, 4000),
+            v=rng.integers(0, 5, 4000).astype("f8"),
+        )
+    )
+    res = fa.aggregate(
+        pdf, partition_by="k", engine=e,
+        cd=f.count_distinct(col("v")),
+        fv=f.first(col("v")),
+        as_fugue=True,
+    )
+    local = e._gather_all(res)
+    return local.as_pandas().sort_values("k").reset_index(drop=True).to_dict("list")
+
+
+def test_distributed_count_distinct_first():
+    results = run_distributed(_firstlast_cd_job, 29525)
+    rng = np.random.default_rng(3)
+    pdf = pd.DataFrame(
+        dict(
+            k=rng.integers(0, 20, 4000),
+            v=rng.integers(0, 5, 4000).astype("f8"),
+        )
+    )
+    expected = (
+        pdf.groupby("k", as_index=False)
+        .agg(cd=("v", "nunique"))
+        .sort_values("k")
+        .reset_index(drop=True)
+    )
+    member_vals = pdf.groupby("k")["v"].agg(lambda s: set(s)).to_dict()
+    for rank, got in results.items():
+        assert got["k"] == expected["k"].tolist()
+        assert got["cd"] == expected["cd"].tolist()
+        for k, fv in zip(got["k"], got["fv"]):
+            assert fv in member_vals[k]
