@@ -23,6 +23,8 @@ from fugue_amd.dataframe import (
     as_fugue_df,
 )
 from fugue_amd.dataset import Dataset
+from fugue_amd.registry import load_entry_point_plugins
 from fugue_amd.registry import register_builtins as _register_builtins
 
 _register_builtins()
+load_entry_point_plugins()

@@ -77,5 +77,13 @@ def last(col: ColumnExpr) -> ColumnExpr:
     return _UnaryAggFuncExpr("LAST", col)
 
 
+def like(col: ColumnExpr, pattern: str) -> ColumnExpr:
+    """SQL ``LIKE`` predicate (``%`` any run, ``_`` any single char).
+    Negation via ``~like(...)``."""
+    assert isinstance(col, ColumnExpr)
+    assert isinstance(pattern, str)
+    return _FuncExpr("LIKE", col, _to_col(pattern))
+
+
 def is_agg(column: Any) -> bool:
     return _is_agg(column)

@@ -109,6 +109,21 @@ def _eval(expr: ColumnExpr, df: pd.DataFrame) -> Any:
                 )
                 out = s if out is None else out.where(out.notna(), s)
             return out
+        if fname == "LIKE":
+            import re as _re
+
+            v = _eval(expr.args[0], df)
+            pat = _eval(expr.args[1], df)
+            if isinstance(pat, pd.Series):
+                pat = pat.iloc[0]
+            regex = "^" + _re.escape(str(pat)).replace("%", ".*").replace(
+                "_", "."
+            ) + "$"
+            s = v if isinstance(v, pd.Series) else pd.Series([v] * len(df), index=df.index)
+            res = s.astype(str).str.match(regex).astype("boolean")
+            # SQL three-valued logic: NULL LIKE p → NULL (so NOT LIKE
+            # also excludes nulls)
+            return res.mask(s.isna())
         if fname == "CASE_WHEN":
             args = expr.args
             n = len(df)

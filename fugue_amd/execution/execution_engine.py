@@ -179,7 +179,31 @@ class MapEngine(EngineFacet):
         partition_spec: PartitionSpec,
         on_init: Optional[Callable] = None,
     ) -> Any:
-        raise NotImplementedError("map_bag is not supported by this engine")
+        """Apply ``map_func`` per bag partition (reference parity:
+        ``fugue/execution/execution_engine.py:318``).  Bags are unordered
+        schemaless item collections, so the partitioner only honors the
+        requested partition count (even item split)."""
+        from fugue_amd.bag.array_bag import ArrayBag
+        from fugue_amd.collections.partition import BagPartitionCursor
+
+        items = list(bag.as_array_iterable())
+        num = partition_spec.get_num_partitions(
+            ROWCOUNT=lambda: len(items),
+            CONCURRENCY=lambda: self.execution_engine.get_current_parallelism(),
+        )
+        num = max(1, min(num if num > 0 else 1, max(1, len(items))))
+        out: List[Any] = []
+        size = (len(items) + num - 1) // num if items else 0
+        for pno in range(num):
+            part = items[pno * size : (pno + 1) * size] if size else []
+            if len(part) == 0 and pno > 0:
+                continue
+            cursor = BagPartitionCursor(pno)
+            if on_init is not None:
+                on_init(pno, ArrayBag(part))
+            res = map_func(cursor, ArrayBag(part))
+            out.extend(res.as_array())
+        return ArrayBag(out)
 
 
 class ExecutionEngine(FugueEngineBase):
@@ -309,6 +333,19 @@ class ExecutionEngine(FugueEngineBase):
     @abstractmethod
     def get_current_parallelism(self) -> int:
         ...
+
+    def map_bag(
+        self,
+        bag: Any,
+        map_func: Callable,
+        partition_spec: PartitionSpec,
+        on_init: Optional[Callable] = None,
+    ) -> Any:
+        """Partition-wise map over a :class:`~fugue_amd.bag.bag.Bag`
+        (reference parity: ``fugue/execution/execution_engine.py:318``)."""
+        return self.map_engine.map_bag(
+            bag, map_func, partition_spec, on_init=on_init
+        )
 
     @abstractmethod
     def repartition(self, df: DataFrame, partition_spec: PartitionSpec) -> DataFrame:

@@ -136,6 +136,8 @@ def eval_device_expr(
             if out_v is not None and bool(out_v.all().item()):
                 out_v = None
             return out_d, out_v
+        if fname == "LIKE":
+            return _eval_like(expr, df)
         if fname == "CASE_WHEN":
             # args = [c1, v1, c2, v2, ..., else]; first match wins →
             # apply branches in reverse over torch.where
@@ -163,6 +165,89 @@ def eval_device_expr(
             return out_d, out_v
         raise DeviceExprError(f"function {expr.func}")
     raise DeviceExprError(f"can't evaluate {expr} on device")
+
+
+def _eval_like(
+    expr: "_FuncExpr", df: HipDataFrame
+) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+    """Device LIKE over the string column's flat byte buffer.
+
+    Supported shapes (pattern ``p`` with no ``_``, no interior ``%``):
+    exact, ``p%`` prefix, ``%p`` suffix, ``%p%`` contains.  Everything
+    else falls back (DeviceExprError → pandas interpreter)."""
+    col_e = expr.args[0]
+    pat_e = expr.args[1]
+    if not (
+        isinstance(col_e, _NamedColumnExpr)
+        and isinstance(pat_e, _LiteralColumnExpr)
+        and isinstance(pat_e.value, str)
+    ):
+        raise DeviceExprError("LIKE shape not supported on device")
+    c = df.col(col_e.name)
+    if not isinstance(c, StringDeviceColumn):
+        raise DeviceExprError("LIKE on non-string column")
+    pat = pat_e.value
+    if "_" in pat:
+        raise DeviceExprError("LIKE with _ wildcard")
+    prefix_any = pat.startswith("%")
+    suffix_any = pat.endswith("%")
+    core = pat.strip("%")
+    if "%" in core:
+        raise DeviceExprError("LIKE with interior %")
+    device = torch.device(df.device)
+    n = len(c)
+    lengths = c.offsets[1:] - c.offsets[:-1]
+    kb = core.encode("utf-8")
+    k = len(kb)
+    if k == 0:
+        # '%' / '%%' matches everything; '' matches only empty strings
+        res = (
+            torch.ones(n, dtype=torch.bool, device=device)
+            if prefix_any or suffix_any
+            else lengths == 0
+        )
+        return res, c.valid
+    patt = torch.tensor(list(kb), dtype=torch.uint8, device=device)
+    long_enough = lengths >= k
+    if not prefix_any:  # 'core%' or exact: compare k bytes at row start
+        starts = c.offsets[:-1]
+        res = long_enough.clone()
+        for j in range(k):
+            idx = torch.clamp(
+                starts + j, max=max(int(c.bytes.numel()) - 1, 0)
+            )
+            res = res & (c.bytes.index_select(0, idx) == patt[j])
+        if not suffix_any:
+            res = res & (lengths == k)
+    elif not suffix_any:  # '%core': compare k bytes at row end
+        ends = c.offsets[1:]
+        res = long_enough.clone()
+        for j in range(k):
+            idx = torch.clamp(ends - k + j, min=0)
+            res = res & (c.bytes.index_select(0, idx) == patt[j])
+    else:  # '%core%': substring search over the flat byte buffer
+        total = int(c.bytes.numel())
+        if total < k:
+            res = torch.zeros(n, dtype=torch.bool, device=device)
+        else:
+            w = total - k + 1
+            m = c.bytes[0:w] == patt[0]
+            for j in range(1, k):
+                m = m & (c.bytes[j : w + j] == patt[j])
+            pos = m.nonzero(as_tuple=True)[0]
+            # map byte position → row, require the window inside the row
+            row = torch.searchsorted(c.offsets[1:], pos, right=True)
+            ok = (pos + k) <= c.offsets.index_select(0, row + 1)
+            res = torch.zeros(n, dtype=torch.bool, device=device)
+            rows = row[ok]
+            if rows.numel() > 0:
+                res.index_put_(
+                    (rows,),
+                    torch.ones(
+                        rows.numel(), dtype=torch.bool, device=device
+                    ),
+                )
+    return res, c.valid
 
 
 def _try_string_equality(

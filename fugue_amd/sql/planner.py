@@ -117,6 +117,10 @@ def _to_column_expr(e: X.Expr, schema: Schema, alias_map: Dict[str, str]) -> Col
                 *[_to_column_expr(a, schema, alias_map) for a in e.args]
             )
         raise UnsupportedPlan(f"function {name}")
+    if isinstance(e, X.Like):
+        inner = _to_column_expr(e.expr, schema, alias_map)
+        res = F.like(inner, e.pattern)
+        return ~res if e.negate else res
     if isinstance(e, X.Case):
         whens = [
             (

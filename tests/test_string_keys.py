@@ -155,3 +155,55 @@ def test_map_num_partitions(engine):
     counts = sorted(r for r in pd.DataFrame(res)["n"].tolist())
     assert sum(counts) == 10
     assert len(counts) == 5
+
+
+def test_device_like_patterns():
+    import pandas as pd
+
+    from fugue_amd.column import functions as F
+    from fugue_amd.column.expressions import col
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    df = pd.DataFrame(
+        dict(
+            k=list(range(8)),
+            s=["apple", "banana", "grape", "applet", "", "nap", None, "app"],
+        )
+    )
+    d = e.to_df(df)
+    cases = [
+        ("app%", [0, 3, 7]),
+        ("%ap", [5]),
+        ("%ap%", [0, 2, 3, 5, 7]),
+        ("app", [7]),
+        ("%", [0, 1, 2, 3, 4, 5, 7]),
+        ("", [4]),
+        ("%zz%", []),
+    ]
+    for pat, exp in cases:
+        r = e.filter(d, F.like(col("s"), pat)).as_pandas()
+        assert sorted(r["k"].tolist()) == exp, pat
+    # NOT LIKE excludes nulls (three-valued logic)
+    r = e.filter(d, ~F.like(col("s"), "app%")).as_pandas()
+    assert sorted(r["k"].tolist()) == [1, 2, 4, 5]
+
+
+def test_sql_like_on_engine():
+    import pandas as pd
+
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+    from fugue_amd.sql.executor import parse_select
+    from fugue_amd.sql.planner import execute_plan
+
+    e = HipExecutionEngine()
+    df = pd.DataFrame(dict(k=[0, 1, 2], s=["foo", "bar", None]))
+    d = e.to_df(df)
+    r = execute_plan(
+        parse_select("SELECT k FROM a WHERE s LIKE 'f%'"), dict(a=d), e
+    ).as_pandas()
+    assert r["k"].tolist() == [0]
+    r2 = execute_plan(
+        parse_select("SELECT k FROM a WHERE s NOT LIKE 'f%'"), dict(a=d), e
+    ).as_pandas()
+    assert r2["k"].tolist() == [1]

@@ -45,3 +45,50 @@ def test_comment_schema():
     def no_comment(df):
         return df
     assert parse_output_schema_from_comment(no_comment) is None
+
+
+def test_entry_point_plugin_loading(monkeypatch):
+    import importlib.metadata as md
+
+    from fugue_amd import registry
+
+    calls = []
+
+    class _EP:
+        name = "demo"
+
+        def load(self):
+            def plug():
+                calls.append("loaded")
+
+            return plug
+
+    def fake_entry_points(*args, **kwargs):
+        assert kwargs.get("group") in registry.FUGUE_ENTRYPOINT_GROUPS
+        return [_EP()] if kwargs.get("group") == "fugue.plugins" else []
+
+    monkeypatch.setattr(md, "entry_points", fake_entry_points)
+    monkeypatch.setattr(registry, "_plugins_done", [False])
+    n = registry.load_entry_point_plugins()
+    assert n == 1 and calls == ["loaded"]
+    # second call is a no-op
+    assert registry.load_entry_point_plugins() == 0
+
+
+def test_map_bag():
+    from fugue_amd.bag.array_bag import ArrayBag
+    from fugue_amd.collections.partition import PartitionSpec
+    from fugue_amd.execution import NativeExecutionEngine
+
+    e = NativeExecutionEngine()
+    bag = ArrayBag([1, 2, 3, 4, 5])
+    inits = []
+
+    def m(cursor, b):
+        return ArrayBag([x * 10 for x in b.as_array()])
+
+    res = e.map_bag(
+        bag, m, PartitionSpec(num=2), on_init=lambda no, b: inits.append(no)
+    )
+    assert sorted(res.as_array()) == [10, 20, 30, 40, 50]
+    assert len(inits) == 2
