@@ -674,3 +674,26 @@ class ExecutionEngineEdgeCaseTests:
         )
         rows = sorted(r.as_array(), key=lambda x: x[0])
         assert [x[1] for x in rows] == ["hi", "na", "lo"]
+
+    def test_like_filter(self):
+        from fugue_amd.column import functions as ff
+
+        e = self.engine
+        o = e.to_df(
+            ArrayDataFrame(
+                [
+                    [0, "apple"],
+                    [1, "banana"],
+                    [2, "applet"],
+                    [3, None],
+                    [4, "nap"],
+                ],
+                "k:long,s:str",
+            )
+        )
+        r = e.filter(o, ff.like(col("s"), "app%"))
+        assert sorted(x[0] for x in r.as_array()) == [0, 2]
+        r2 = e.filter(o, ff.like(col("s"), "%ap%"))
+        assert sorted(x[0] for x in r2.as_array()) == [0, 2, 4]
+        r3 = e.filter(o, ~ff.like(col("s"), "app%"))
+        assert sorted(x[0] for x in r3.as_array()) == [1, 4]
