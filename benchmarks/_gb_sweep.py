@@ -25,25 +25,30 @@ def main() -> None:
     ops = torch.tensor([0], dtype=torch.int32, device=dev)  # SUM
     tsize = 1 << 22  # 4.2M slots for ~1M groups
 
-    def run(sc, ag, iters=6):
+    def run(sc, ag, nt, iters=8):
         torch.cuda.synchronize()
         # warmup
-        ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, sc, ag)
+        ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, sc, ag, nt)
         torch.cuda.synchronize()
         t0 = time.perf_counter()
         for _ in range(iters):
-            ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, sc, ag)
+            ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, sc, ag, nt)
         torch.cuda.synchronize()
         return (time.perf_counter() - t0) / iters * 1000
 
-    base = run(0, 0)
-    print(f"baseline (compiled defaults): {base:.3f} ms", flush=True)
-    for ag in (4096, 8192, 16384, 32768):
-        t = run(0, ag)
-        print(f"scatter=default agg_chunk={ag:6d}: {t:.3f} ms", flush=True)
-    for sc, ag in ((4096, 16384), (4096, 32768), (2048, 16384)):
-        t = run(sc, ag)
-        print(f"scatter_chunk={sc:6d} agg_chunk={ag:6d}: {t:.3f} ms", flush=True)
+    # correctness cross-check nt vs not
+    k0, a0, c0 = ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, 0, 0, 0)
+    k1, a1, c1 = ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, 0, 0, 1)
+    s0 = torch.argsort(k0); s1 = torch.argsort(k1)
+    assert torch.equal(k0[s0], k1[s1]), "keys mismatch"
+    assert torch.allclose(a0[0][s0], a1[0][s1]), "aggs mismatch"
+    print("nt correctness OK", flush=True)
+    for nt in (0, 1):
+        t = run(0, 0, nt)
+        print(f"defaults nt={nt}: {t:.3f} ms", flush=True)
+    for nt in (0, 1):
+        t = run(0, 16384, nt)
+        print(f"agg_chunk=16384 nt={nt}: {t:.3f} ms", flush=True)
 
 
 if __name__ == "__main__":

@@ -37,11 +37,11 @@ void launch_gb_aggregate_part(const int64_t*, const double*, const int32_t*,
                               double*, int64_t*, int64_t, hipStream_t);
 void launch_gb_part_scatter_staged(const int64_t*, const double*, int64_t,
                                    int, int64_t*, int64_t*, double*, int64_t,
-                                   hipStream_t);
+                                   int, hipStream_t);
 void launch_gb_aggregate_part_big(const int64_t*, const double*,
                                   const int32_t*, int64_t, int64_t*,
                                   double*, int64_t*, int64_t, int64_t,
-                                  hipStream_t);
+                                  int, hipStream_t);
 void launch_join_count(const int64_t*, int64_t, const int64_t*,
                        const int64_t*, const int64_t*, const int32_t*,
                        const int32_t*, int64_t, int32_t*, hipStream_t);
@@ -178,7 +178,7 @@ std::vector<at::Tensor> gb_aggregate(at::Tensor keys, at::Tensor vals,
 
 std::vector<at::Tensor> gb_aggregate_partitioned(
     at::Tensor keys, at::Tensor vals, at::Tensor ops, int64_t num_parts,
-    int64_t tsize, int64_t scatter_chunk, int64_t agg_chunk) {
+    int64_t tsize, int64_t scatter_chunk, int64_t agg_chunk, int64_t nt) {
   check_gpu(keys, "keys");
   check_gpu(vals, "vals");
   TORCH_CHECK((tsize & (tsize - 1)) == 0, "tsize must be a power of 2");
@@ -208,7 +208,7 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
     launch_gb_part_scatter_staged(
         keys.data_ptr<int64_t>(), vals.data_ptr<double>(), n, shift,
         cursor.data_ptr<int64_t>(), pkeys.data_ptr<int64_t>(),
-        pvals.data_ptr<double>(), scatter_chunk, stream);
+        pvals.data_ptr<double>(), scatter_chunk, (int)nt, stream);
   } else {
     launch_gb_part_scatter(keys.data_ptr<int64_t>(), vals.data_ptr<double>(),
                            n_aggs, n, shift, cursor.data_ptr<int64_t>(),
@@ -225,7 +225,7 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
         pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),
         ops.data_ptr<int32_t>(), n, tkeys.data_ptr<int64_t>(),
         gaggs.data_ptr<double>(), gcount.data_ptr<int64_t>(), tsize,
-        agg_chunk, stream);
+        agg_chunk, (int)nt, stream);
   } else {
     launch_gb_aggregate_part(
         pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),

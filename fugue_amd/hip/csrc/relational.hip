@@ -45,6 +45,21 @@ __device__ __forceinline__ uint64_t hash_combine(uint64_t h, uint64_t v) {
   return mix64(h ^ (v + 0x9e3779b97f4a7c15ULL + (h << 6) + (h >> 2)));
 }
 
+// streaming (nontemporal) access helpers: single-use data bypasses L2/
+// MALL so cached lines survive for data that IS reused (guide §6 G13)
+template <bool NT, typename T>
+__device__ __forceinline__ T stream_ld(const T* p) {
+  return NT ? __builtin_nontemporal_load(p) : *p;
+}
+template <bool NT, typename T>
+__device__ __forceinline__ void stream_st(T* p, T v) {
+  if (NT) {
+    __builtin_nontemporal_store(v, p);
+  } else {
+    *p = v;
+  }
+}
+
 // ------------------------------------------------------------------ //
 // hashing: combine one column into the running row-hash               //
 // ------------------------------------------------------------------ //
@@ -484,6 +499,7 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_kernel(
 #define STAGE_P 512
 #define STAGE_E 8
 
+template <bool NT>
 __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
     const int64_t* __restrict__ keys,
     const double* __restrict__ vals,   // [1, n]
@@ -524,13 +540,14 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
       int64_t key = keys[i];
       int p = (int)(mix64((uint64_t)key) >> shift);
       int pos = atomicAdd(&lcnt[p], 1);
+      double v = stream_ld<NT>(&vals[i]);
       if (pos < STAGE_E) {
         skey[p * STAGE_E + pos] = key;
-        sval[p * STAGE_E + pos] = vals[i];
+        sval[p * STAGE_E + pos] = v;
       } else {
         int64_t gpos = lbase[p] + pos;
-        out_keys[gpos] = key;
-        out_vals[gpos] = vals[i];
+        stream_st<NT>(&out_keys[gpos], key);
+        stream_st<NT>(&out_vals[gpos], v);
       }
     }
     __syncthreads();
@@ -541,8 +558,8 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
       int c = lcnt[p];
       if (e < c && e < STAGE_E) {
         int64_t gpos = lbase[p] + e;
-        out_keys[gpos] = skey[t];
-        out_vals[gpos] = sval[t];
+        stream_st<NT>(&out_keys[gpos], skey[t]);
+        stream_st<NT>(&out_vals[gpos], sval[t]);
       }
     }
     __syncthreads();
@@ -558,6 +575,7 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
 #define LDS_SLOTS_BIG 4096
 #define AGG_CHUNK (BLOCK * 128)  // 32768 rows per chunk (sweep-tuned)
 
+template <bool NT>
 __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
     const int64_t* __restrict__ part_keys,
     const double* __restrict__ part_vals,  // [1, n]
@@ -582,7 +600,7 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
     }
     __syncthreads();
     for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
-      int64_t key = part_keys[i];
+      int64_t key = stream_ld<NT>(&part_keys[i]);
       uint64_t h = mix64((uint64_t)key);
       int slot = (int)(h & (LDS_SLOTS_BIG - 1));
       bool in_lds = false;
@@ -593,7 +611,7 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
         if (prev == GB_EMPTY || prev == key) { in_lds = true; break; }
         slot = (slot + 1) & (LDS_SLOTS_BIG - 1);
       }
-      double v = is_count ? 1.0 : part_vals[i];
+      double v = is_count ? 1.0 : stream_ld<NT>(&part_vals[i]);
       if (in_lds) {
         atomicAdd(&lcount[slot], 1);
         atomicAdd(&laggs[slot], v);
@@ -621,14 +639,21 @@ extern "C" {
 void launch_gb_part_scatter_staged(const int64_t* keys, const double* vals,
                                    int64_t n, int shift, int64_t* cursor,
                                    int64_t* out_keys, double* out_vals,
-                                   int64_t chunk, hipStream_t stream) {
+                                   int64_t chunk, int nt,
+                                   hipStream_t stream) {
   if (chunk <= 0) chunk = SCATTER_CHUNK;
   int64_t blocks = (n + chunk - 1) / chunk;
   if (blocks > MAX_GRID) blocks = MAX_GRID;
   if (blocks < 1) blocks = 1;
-  hipLaunchKernelGGL(gb_part_scatter_staged_kernel, dim3((int)blocks),
-                     dim3(BLOCK), 0, stream, keys, vals, n, shift, cursor,
-                     out_keys, out_vals, chunk);
+  if (nt) {
+    hipLaunchKernelGGL(gb_part_scatter_staged_kernel<true>,
+                       dim3((int)blocks), dim3(BLOCK), 0, stream, keys, vals,
+                       n, shift, cursor, out_keys, out_vals, chunk);
+  } else {
+    hipLaunchKernelGGL(gb_part_scatter_staged_kernel<false>,
+                       dim3((int)blocks), dim3(BLOCK), 0, stream, keys, vals,
+                       n, shift, cursor, out_keys, out_vals, chunk);
+  }
 }
 
 void launch_gb_aggregate_part_big(const int64_t* part_keys,
@@ -636,14 +661,21 @@ void launch_gb_aggregate_part_big(const int64_t* part_keys,
                                   const int32_t* ops, int64_t n,
                                   int64_t* tkeys, double* gaggs,
                                   int64_t* gcount, int64_t tsize,
-                                  int64_t chunk, hipStream_t stream) {
+                                  int64_t chunk, int nt,
+                                  hipStream_t stream) {
   if (chunk <= 0) chunk = AGG_CHUNK;
   int64_t blocks = (n + chunk - 1) / chunk;
   if (blocks > MAX_GRID) blocks = MAX_GRID;
   if (blocks < 1) blocks = 1;
-  hipLaunchKernelGGL(gb_aggregate_part_big_kernel, dim3((int)blocks),
-                     dim3(BLOCK), 0, stream, part_keys, part_vals, ops, n,
-                     tkeys, gaggs, gcount, tsize, chunk);
+  if (nt) {
+    hipLaunchKernelGGL(gb_aggregate_part_big_kernel<true>, dim3((int)blocks),
+                       dim3(BLOCK), 0, stream, part_keys, part_vals, ops, n,
+                       tkeys, gaggs, gcount, tsize, chunk);
+  } else {
+    hipLaunchKernelGGL(gb_aggregate_part_big_kernel<false>, dim3((int)blocks),
+                       dim3(BLOCK), 0, stream, part_keys, part_vals, ops, n,
+                       tkeys, gaggs, gcount, tsize, chunk);
+  }
 }
 
 }  // extern "C"

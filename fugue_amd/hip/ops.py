@@ -382,8 +382,9 @@ def groupby_aggregate(
 
         sc_chunk = int(_os.environ.get("FUGUE_GB_SCATTER_CHUNK", "0"))
         ag_chunk = int(_os.environ.get("FUGUE_GB_AGG_CHUNK", "0"))
+        nt = int(_os.environ.get("FUGUE_GB_NT", "0"))
         tkeys, gaggs, gcount = ext.gb_aggregate_partitioned(
-            packed, vals, ops, num_parts, tsize, sc_chunk, ag_chunk
+            packed, vals, ops, num_parts, tsize, sc_chunk, ag_chunk, nt
         )
     else:
         use_lds = expected_groups <= 100_000 and sum_count_only
