@@ -25,30 +25,34 @@ def main() -> None:
     ops = torch.tensor([0], dtype=torch.int32, device=dev)  # SUM
     tsize = 1 << 22  # 4.2M slots for ~1M groups
 
-    def run(sc, ag, nt, iters=8):
+    def run(sc, ag, nt, narrow, iters=8):
         torch.cuda.synchronize()
         # warmup
-        ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, sc, ag, nt)
+        ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, sc, ag, nt, narrow)
         torch.cuda.synchronize()
         t0 = time.perf_counter()
         for _ in range(iters):
-            ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, sc, ag, nt)
+            ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, sc, ag, nt, narrow)
         torch.cuda.synchronize()
         return (time.perf_counter() - t0) / iters * 1000
 
-    # correctness cross-check nt vs not
-    k0, a0, c0 = ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, 0, 0, 0)
-    k1, a1, c1 = ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, 0, 0, 1)
-    s0 = torch.argsort(k0); s1 = torch.argsort(k1)
-    assert torch.equal(k0[s0], k1[s1]), "keys mismatch"
-    assert torch.allclose(a0[0][s0], a1[0][s1]), "aggs mismatch"
-    print("nt correctness OK", flush=True)
-    for nt in (0, 1):
-        t = run(0, 0, nt)
-        print(f"defaults nt={nt}: {t:.3f} ms", flush=True)
-    for nt in (0, 1):
-        t = run(0, 16384, nt)
-        print(f"agg_chunk=16384 nt={nt}: {t:.3f} ms", flush=True)
+    # correctness cross-check narrow vs not
+    k0, a0, c0 = ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, 0, 0, 0, 0)
+    k1, a1, c1 = ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, 0, 0, 0, 1)
+    m0 = k0 != -9223372036854775808
+    m1 = k1 != -9223372036854775808
+    kk0, kk1 = k0[m0], k1[m1]
+    aa0, aa1 = a0[0][m0], a1[0][m1]
+    s0 = torch.argsort(kk0); s1 = torch.argsort(kk1)
+    assert torch.equal(kk0[s0], kk1[s1]), "keys mismatch"
+    assert torch.allclose(aa0[s0], aa1[s1]), "aggs mismatch"
+    print("narrow correctness OK", flush=True)
+    for narrow in (0, 1):
+        t = run(0, 0, 0, narrow)
+        print(f"defaults narrow={narrow}: {t:.3f} ms", flush=True)
+    for narrow in (0, 1):
+        t = run(0, 16384, 0, narrow)
+        print(f"agg_chunk=16384 narrow={narrow}: {t:.3f} ms", flush=True)
 
 
 if __name__ == "__main__":

@@ -36,12 +36,12 @@ void launch_gb_aggregate_part(const int64_t*, const double*, const int32_t*,
                               int, int64_t, const int64_t*, int64_t, int64_t*,
                               double*, int64_t*, int64_t, hipStream_t);
 void launch_gb_part_scatter_staged(const int64_t*, const double*, int64_t,
-                                   int, int64_t*, int64_t*, double*, int64_t,
-                                   int, hipStream_t);
-void launch_gb_aggregate_part_big(const int64_t*, const double*,
+                                   int, int64_t*, void*, double*, int64_t,
+                                   int, int, hipStream_t);
+void launch_gb_aggregate_part_big(const void*, const double*,
                                   const int32_t*, int64_t, int64_t*,
                                   double*, int64_t*, int64_t, int64_t,
-                                  int, hipStream_t);
+                                  int, int, hipStream_t);
 void launch_join_count(const int64_t*, int64_t, const int64_t*,
                        const int64_t*, const int64_t*, const int32_t*,
                        const int32_t*, int64_t, int32_t*, hipStream_t);
@@ -178,7 +178,8 @@ std::vector<at::Tensor> gb_aggregate(at::Tensor keys, at::Tensor vals,
 
 std::vector<at::Tensor> gb_aggregate_partitioned(
     at::Tensor keys, at::Tensor vals, at::Tensor ops, int64_t num_parts,
-    int64_t tsize, int64_t scatter_chunk, int64_t agg_chunk, int64_t nt) {
+    int64_t tsize, int64_t scatter_chunk, int64_t agg_chunk, int64_t nt,
+    int64_t narrow) {
   check_gpu(keys, "keys");
   check_gpu(vals, "vals");
   TORCH_CHECK((tsize & (tsize - 1)) == 0, "tsize must be a power of 2");
@@ -201,14 +202,18 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
   offsets.narrow(0, 1, num_parts).copy_(at::cumsum(hist, 0));
   auto cursor = offsets.narrow(0, 0, num_parts).clone();
   // phase 2: scatter into partitioned order
-  auto pkeys = at::empty({n}, keys.options());
-  auto pvals = at::empty({n_aggs, n}, vals.options());
   bool staged = (n_aggs == 1 && num_parts == 512);
+  bool use_narrow = staged && narrow != 0;
+  auto pkeys = use_narrow
+                   ? at::empty({n}, keys.options().dtype(at::kInt))
+                   : at::empty({n}, keys.options());
+  auto pvals = at::empty({n_aggs, n}, vals.options());
   if (staged) {
     launch_gb_part_scatter_staged(
         keys.data_ptr<int64_t>(), vals.data_ptr<double>(), n, shift,
-        cursor.data_ptr<int64_t>(), pkeys.data_ptr<int64_t>(),
-        pvals.data_ptr<double>(), scatter_chunk, (int)nt, stream);
+        cursor.data_ptr<int64_t>(), pkeys.data_ptr(),
+        pvals.data_ptr<double>(), scatter_chunk, (int)nt,
+        use_narrow ? 1 : 0, stream);
   } else {
     launch_gb_part_scatter(keys.data_ptr<int64_t>(), vals.data_ptr<double>(),
                            n_aggs, n, shift, cursor.data_ptr<int64_t>(),
@@ -222,10 +227,10 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
   auto gcount = at::zeros({tsize}, keys.options());
   if (staged) {
     launch_gb_aggregate_part_big(
-        pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),
+        pkeys.data_ptr(), pvals.data_ptr<double>(),
         ops.data_ptr<int32_t>(), n, tkeys.data_ptr<int64_t>(),
         gaggs.data_ptr<double>(), gcount.data_ptr<int64_t>(), tsize,
-        agg_chunk, (int)nt, stream);
+        agg_chunk, (int)nt, use_narrow ? 1 : 0, stream);
   } else {
     launch_gb_aggregate_part(
         pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),

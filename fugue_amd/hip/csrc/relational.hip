@@ -499,19 +499,19 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_kernel(
 #define STAGE_P 512
 #define STAGE_E 8
 
-template <bool NT>
+template <bool NT, typename KT>
 __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
     const int64_t* __restrict__ keys,
     const double* __restrict__ vals,   // [1, n]
     int64_t n, int shift,
     int64_t* __restrict__ cursor,
-    int64_t* __restrict__ out_keys,
+    KT* __restrict__ out_keys,
     double* __restrict__ out_vals,
     int64_t chunk) {
   __shared__ int lhist[STAGE_P];
   __shared__ int64_t lbase[STAGE_P];
   __shared__ int lcnt[STAGE_P];
-  __shared__ int64_t skey[STAGE_P * STAGE_E];
+  __shared__ KT skey[STAGE_P * STAGE_E];
   __shared__ double sval[STAGE_P * STAGE_E];
   for (int64_t start = (int64_t)blockIdx.x * chunk; start < n;
        start += (int64_t)gridDim.x * chunk) {
@@ -542,11 +542,11 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
       int pos = atomicAdd(&lcnt[p], 1);
       double v = stream_ld<NT>(&vals[i]);
       if (pos < STAGE_E) {
-        skey[p * STAGE_E + pos] = key;
+        skey[p * STAGE_E + pos] = (KT)key;
         sval[p * STAGE_E + pos] = v;
       } else {
         int64_t gpos = lbase[p] + pos;
-        stream_st<NT>(&out_keys[gpos], key);
+        stream_st<NT>(&out_keys[gpos], (KT)key);
         stream_st<NT>(&out_vals[gpos], v);
       }
     }
@@ -575,9 +575,31 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
 #define LDS_SLOTS_BIG 4096
 #define AGG_CHUNK (BLOCK * 128)  // 32768 rows per chunk (sweep-tuned)
 
-template <bool NT>
+// empty-slot sentinel for the LDS table: int64 path uses GB_EMPTY;
+// int32 (narrow, packed width <=31 so keys are >=0) uses -1
+template <typename KT> __device__ __forceinline__ KT lds_empty();
+template <> __device__ __forceinline__ int64_t lds_empty<int64_t>() {
+  return GB_EMPTY;
+}
+template <> __device__ __forceinline__ int32_t lds_empty<int32_t>() {
+  return -1;
+}
+
+__device__ __forceinline__ long long lds_key_cas(int64_t* addr, int64_t cmp,
+                                                 int64_t val) {
+  return (long long)atomicCAS((unsigned long long*)addr,
+                              (unsigned long long)cmp,
+                              (unsigned long long)val);
+}
+__device__ __forceinline__ int lds_key_cas(int32_t* addr, int32_t cmp,
+                                           int32_t val) {
+  return (int)atomicCAS((unsigned int*)addr, (unsigned int)cmp,
+                        (unsigned int)val);
+}
+
+template <bool NT, typename KT>
 __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
-    const int64_t* __restrict__ part_keys,
+    const KT* __restrict__ part_keys,
     const double* __restrict__ part_vals,  // [1, n]
     const int32_t* __restrict__ ops,
     int64_t n,
@@ -585,30 +607,29 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
     double* __restrict__ gaggs,
     int64_t* __restrict__ gcount,
     int64_t tsize, int64_t chunk) {
-  __shared__ int64_t lkeys[LDS_SLOTS_BIG];
+  __shared__ KT lkeys[LDS_SLOTS_BIG];
   __shared__ double laggs[LDS_SLOTS_BIG];
   __shared__ int lcount[LDS_SLOTS_BIG];
+  const KT EMPTY = lds_empty<KT>();
   bool is_count = ops[0] == 3;
   for (int64_t start = (int64_t)blockIdx.x * chunk; start < n;
        start += (int64_t)gridDim.x * chunk) {
     int64_t end = start + chunk;
     if (end > n) end = n;
     for (int i = threadIdx.x; i < LDS_SLOTS_BIG; i += blockDim.x) {
-      lkeys[i] = GB_EMPTY;
+      lkeys[i] = EMPTY;
       lcount[i] = 0;
       laggs[i] = 0.0;
     }
     __syncthreads();
     for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
-      int64_t key = stream_ld<NT>(&part_keys[i]);
-      uint64_t h = mix64((uint64_t)key);
+      KT key = stream_ld<NT>(&part_keys[i]);
+      uint64_t h = mix64((uint64_t)(int64_t)key);
       int slot = (int)(h & (LDS_SLOTS_BIG - 1));
       bool in_lds = false;
       for (int probe = 0; probe < 32; ++probe) {
-        long long prev = (long long)atomicCAS(
-            (unsigned long long*)&lkeys[slot], (unsigned long long)GB_EMPTY,
-            (unsigned long long)key);
-        if (prev == GB_EMPTY || prev == key) { in_lds = true; break; }
+        KT prev = (KT)lds_key_cas(&lkeys[slot], EMPTY, key);
+        if (prev == EMPTY || prev == key) { in_lds = true; break; }
         slot = (slot + 1) & (LDS_SLOTS_BIG - 1);
       }
       double v = is_count ? 1.0 : stream_ld<NT>(&part_vals[i]);
@@ -616,16 +637,16 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
         atomicAdd(&lcount[slot], 1);
         atomicAdd(&laggs[slot], v);
       } else {
-        int64_t gslot = gb_probe_insert(key, tkeys, tsize);
+        int64_t gslot = gb_probe_insert((int64_t)key, tkeys, tsize);
         atomicAdd((unsigned long long*)&gcount[gslot], 1ULL);
         atomicAdd(&gaggs[gslot], v);
       }
     }
     __syncthreads();
     for (int i = threadIdx.x; i < LDS_SLOTS_BIG; i += blockDim.x) {
-      int64_t key = lkeys[i];
-      if (key == GB_EMPTY) continue;
-      int64_t gslot = gb_probe_insert(key, tkeys, tsize);
+      KT key = lkeys[i];
+      if (key == EMPTY) continue;
+      int64_t gslot = gb_probe_insert((int64_t)key, tkeys, tsize);
       atomicAdd((unsigned long long*)&gcount[gslot],
                 (unsigned long long)lcount[i]);
       atomicAdd(&gaggs[gslot], laggs[i]);
@@ -638,43 +659,69 @@ extern "C" {
 
 void launch_gb_part_scatter_staged(const int64_t* keys, const double* vals,
                                    int64_t n, int shift, int64_t* cursor,
-                                   int64_t* out_keys, double* out_vals,
-                                   int64_t chunk, int nt,
+                                   void* out_keys, double* out_vals,
+                                   int64_t chunk, int nt, int narrow,
                                    hipStream_t stream) {
   if (chunk <= 0) chunk = SCATTER_CHUNK;
   int64_t blocks = (n + chunk - 1) / chunk;
   if (blocks > MAX_GRID) blocks = MAX_GRID;
   if (blocks < 1) blocks = 1;
-  if (nt) {
-    hipLaunchKernelGGL(gb_part_scatter_staged_kernel<true>,
-                       dim3((int)blocks), dim3(BLOCK), 0, stream, keys, vals,
-                       n, shift, cursor, out_keys, out_vals, chunk);
+  dim3 g((int)blocks), b(BLOCK);
+  if (narrow) {
+    auto* ok = (int32_t*)out_keys;
+    if (nt)
+      hipLaunchKernelGGL((gb_part_scatter_staged_kernel<true, int32_t>), g, b,
+                         0, stream, keys, vals, n, shift, cursor, ok,
+                         out_vals, chunk);
+    else
+      hipLaunchKernelGGL((gb_part_scatter_staged_kernel<false, int32_t>), g,
+                         b, 0, stream, keys, vals, n, shift, cursor, ok,
+                         out_vals, chunk);
   } else {
-    hipLaunchKernelGGL(gb_part_scatter_staged_kernel<false>,
-                       dim3((int)blocks), dim3(BLOCK), 0, stream, keys, vals,
-                       n, shift, cursor, out_keys, out_vals, chunk);
+    auto* ok = (int64_t*)out_keys;
+    if (nt)
+      hipLaunchKernelGGL((gb_part_scatter_staged_kernel<true, int64_t>), g, b,
+                         0, stream, keys, vals, n, shift, cursor, ok,
+                         out_vals, chunk);
+    else
+      hipLaunchKernelGGL((gb_part_scatter_staged_kernel<false, int64_t>), g,
+                         b, 0, stream, keys, vals, n, shift, cursor, ok,
+                         out_vals, chunk);
   }
 }
 
-void launch_gb_aggregate_part_big(const int64_t* part_keys,
+void launch_gb_aggregate_part_big(const void* part_keys,
                                   const double* part_vals,
                                   const int32_t* ops, int64_t n,
                                   int64_t* tkeys, double* gaggs,
                                   int64_t* gcount, int64_t tsize,
-                                  int64_t chunk, int nt,
+                                  int64_t chunk, int nt, int narrow,
                                   hipStream_t stream) {
   if (chunk <= 0) chunk = AGG_CHUNK;
   int64_t blocks = (n + chunk - 1) / chunk;
   if (blocks > MAX_GRID) blocks = MAX_GRID;
   if (blocks < 1) blocks = 1;
-  if (nt) {
-    hipLaunchKernelGGL(gb_aggregate_part_big_kernel<true>, dim3((int)blocks),
-                       dim3(BLOCK), 0, stream, part_keys, part_vals, ops, n,
-                       tkeys, gaggs, gcount, tsize, chunk);
+  dim3 g((int)blocks), b(BLOCK);
+  if (narrow) {
+    auto* pk = (const int32_t*)part_keys;
+    if (nt)
+      hipLaunchKernelGGL((gb_aggregate_part_big_kernel<true, int32_t>), g, b,
+                         0, stream, pk, part_vals, ops, n, tkeys, gaggs,
+                         gcount, tsize, chunk);
+    else
+      hipLaunchKernelGGL((gb_aggregate_part_big_kernel<false, int32_t>), g, b,
+                         0, stream, pk, part_vals, ops, n, tkeys, gaggs,
+                         gcount, tsize, chunk);
   } else {
-    hipLaunchKernelGGL(gb_aggregate_part_big_kernel<false>, dim3((int)blocks),
-                       dim3(BLOCK), 0, stream, part_keys, part_vals, ops, n,
-                       tkeys, gaggs, gcount, tsize, chunk);
+    auto* pk = (const int64_t*)part_keys;
+    if (nt)
+      hipLaunchKernelGGL((gb_aggregate_part_big_kernel<true, int64_t>), g, b,
+                         0, stream, pk, part_vals, ops, n, tkeys, gaggs,
+                         gcount, tsize, chunk);
+    else
+      hipLaunchKernelGGL((gb_aggregate_part_big_kernel<false, int64_t>), g, b,
+                         0, stream, pk, part_vals, ops, n, tkeys, gaggs,
+                         gcount, tsize, chunk);
   }
 }
 

@@ -383,8 +383,20 @@ def groupby_aggregate(
         sc_chunk = int(_os.environ.get("FUGUE_GB_SCATTER_CHUNK", "0"))
         ag_chunk = int(_os.environ.get("FUGUE_GB_AGG_CHUNK", "0"))
         nt = int(_os.environ.get("FUGUE_GB_NT", "0"))
+        # int32 intermediate keys shrink the partitioned spill 16B->12B
+        # per row (and its MALL footprint) when the packed key range is
+        # known (pack meta) or measured to fit 31 bits
+        narrow = 0
+        if int(_os.environ.get("FUGUE_GB_NARROW", "1")):
+            if meta is not None and sum(meta["widths"]) <= 31:
+                narrow = 1
+            elif meta is None:
+                lo, hi = torch.aminmax(packed)
+                if int(lo.item()) >= 0 and int(hi.item()) < (1 << 31):
+                    narrow = 1
         tkeys, gaggs, gcount = ext.gb_aggregate_partitioned(
-            packed, vals, ops, num_parts, tsize, sc_chunk, ag_chunk, nt
+            packed, vals, ops, num_parts, tsize, sc_chunk, ag_chunk, nt,
+            narrow
         )
     else:
         use_lds = expected_groups <= 100_000 and sum_count_only
