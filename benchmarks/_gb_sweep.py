@@ -46,13 +46,16 @@ def main() -> None:
     s0 = torch.argsort(kk0); s1 = torch.argsort(kk1)
     assert torch.equal(kk0[s0], kk1[s1]), "keys mismatch"
     assert torch.allclose(aa0[s0], aa1[s1]), "aggs mismatch"
-    print("narrow correctness OK", flush=True)
-    for narrow in (0, 1):
+    k2, a2, c2 = ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, 0, 0, 0, -1)
+    m2 = k2 != -9223372036854775808
+    kk2, aa2 = k2[m2], a2[0][m2]
+    s2 = torch.argsort(kk2)
+    assert torch.equal(kk0[s0], kk2[s2]), "auto keys mismatch"
+    assert torch.allclose(aa0[s0], aa2[s2]), "auto aggs mismatch"
+    print("narrow correctness OK (explicit + auto)", flush=True)
+    for narrow in (0, 1, -1):
         t = run(0, 0, 0, narrow)
         print(f"defaults narrow={narrow}: {t:.3f} ms", flush=True)
-    for narrow in (0, 1):
-        t = run(0, 16384, 0, narrow)
-        print(f"agg_chunk=16384 narrow={narrow}: {t:.3f} ms", flush=True)
 
 
 if __name__ == "__main__":

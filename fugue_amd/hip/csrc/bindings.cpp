@@ -29,7 +29,7 @@ void launch_gb_aggregate(const int64_t*, const double*, const bool*,
 void launch_join_build(const int64_t*, int64_t, int32_t*, int32_t*, int64_t,
                        hipStream_t);
 void launch_gb_part_hist(const int64_t*, int64_t, int, int64_t*, int,
-                         hipStream_t);
+                         int64_t*, hipStream_t);
 void launch_gb_part_scatter(const int64_t*, const double*, int, int64_t, int,
                             int64_t*, int64_t*, double*, int, hipStream_t);
 void launch_gb_aggregate_part(const int64_t*, const double*, const int32_t*,
@@ -193,17 +193,33 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
     int64_t p = num_parts;
     while (p > 1) { --shift; p >>= 1; }
   }
-  // phase 1: histogram + scan
+  // phase 1: histogram + scan (also reduces key min/max when the
+  // caller asks for narrow auto-detection: narrow < 0)
   auto hist = at::zeros({num_parts}, keys.options());
   TORCH_CHECK(num_parts <= 4096, "num_parts must be <= 4096");
+  bool staged = (n_aggs == 1 && num_parts == 512);
+  bool auto_narrow = staged && narrow < 0;
+  at::Tensor minmax;
+  if (auto_narrow) {
+    minmax = at::empty({2}, keys.options());
+    minmax[0] = std::numeric_limits<int64_t>::max();
+    minmax[1] = std::numeric_limits<int64_t>::min();
+  }
   launch_gb_part_hist(keys.data_ptr<int64_t>(), n, shift,
-                      hist.data_ptr<int64_t>(), (int)num_parts, stream);
+                      hist.data_ptr<int64_t>(), (int)num_parts,
+                      auto_narrow ? minmax.data_ptr<int64_t>() : nullptr,
+                      stream);
+  if (auto_narrow) {
+    auto mm = minmax.cpu();
+    int64_t lo = mm[0].item<int64_t>();
+    int64_t hi = mm[1].item<int64_t>();
+    narrow = (n > 0 && lo >= 0 && hi < (int64_t(1) << 31)) ? 1 : 0;
+  }
   auto offsets = at::zeros({num_parts + 1}, keys.options());
   offsets.narrow(0, 1, num_parts).copy_(at::cumsum(hist, 0));
   auto cursor = offsets.narrow(0, 0, num_parts).clone();
   // phase 2: scatter into partitioned order
-  bool staged = (n_aggs == 1 && num_parts == 512);
-  bool use_narrow = staged && narrow != 0;
+  bool use_narrow = staged && narrow > 0;
   auto pkeys = use_narrow
                    ? at::empty({n}, keys.options().dtype(at::kInt))
                    : at::empty({n}, keys.options());

@@ -388,12 +388,10 @@ def groupby_aggregate(
         # known (pack meta) or measured to fit 31 bits
         narrow = 0
         if int(_os.environ.get("FUGUE_GB_NARROW", "1")):
-            if meta is not None and sum(meta["widths"]) <= 31:
-                narrow = 1
-            elif meta is None:
-                lo, hi = torch.aminmax(packed)
-                if int(lo.item()) >= 0 and int(hi.item()) < (1 << 31):
-                    narrow = 1
+            if meta is not None:
+                narrow = 1 if sum(meta["widths"]) <= 31 else 0
+            else:
+                narrow = -1  # auto: hist kernel reduces key min/max
         tkeys, gaggs, gcount = ext.gb_aggregate_partitioned(
             packed, vals, ops, num_parts, tsize, sc_chunk, ag_chunk, nt,
             narrow
