@@ -37,8 +37,8 @@ def main() -> None:
         return (time.perf_counter() - t0) / iters * 1000
 
     # correctness cross-check narrow vs not
-    k0, a0, c0 = ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, 0, 0, 0, 0)
-    k1, a1, c1 = ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, 0, 0, 0, 1)
+    k0, a0, c0, _ = ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, 0, 0, 0, 0)
+    k1, a1, c1, _ = ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, 0, 0, 0, 1)
     m0 = k0 != -9223372036854775808
     m1 = k1 != -9223372036854775808
     kk0, kk1 = k0[m0], k1[m1]
@@ -46,7 +46,7 @@ def main() -> None:
     s0 = torch.argsort(kk0); s1 = torch.argsort(kk1)
     assert torch.equal(kk0[s0], kk1[s1]), "keys mismatch"
     assert torch.allclose(aa0[s0], aa1[s1]), "aggs mismatch"
-    k2, a2, c2 = ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, 0, 0, 0, -1)
+    k2, a2, c2, _ = ext.gb_aggregate_partitioned(keys, vals, ops, 512, tsize, 0, 0, 0, -1)
     m2 = k2 != -9223372036854775808
     kk2, aa2 = k2[m2], a2[0][m2]
     s2 = torch.argsort(kk2)

@@ -5,6 +5,8 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
+#include <cstdio>
 #include <limits>
 #include <vector>
 
@@ -37,7 +39,7 @@ void launch_gb_aggregate_part(const int64_t*, const double*, const int32_t*,
                               double*, int64_t*, int64_t, hipStream_t);
 void launch_gb_part_scatter_staged(const int64_t*, const double*, int64_t,
                                    int, int64_t*, void*, double*, int64_t,
-                                   int, int, hipStream_t);
+                                   int, int, int*, hipStream_t);
 void launch_gb_aggregate_part_big(const void*, const double*,
                                   const int32_t*, int64_t, int64_t*,
                                   double*, int64_t*, int64_t, int64_t,
@@ -198,23 +200,14 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
   auto hist = at::zeros({num_parts}, keys.options());
   TORCH_CHECK(num_parts <= 4096, "num_parts must be <= 4096");
   bool staged = (n_aggs == 1 && num_parts == 512);
-  bool auto_narrow = staged && narrow < 0;
-  at::Tensor minmax;
-  if (auto_narrow) {
-    minmax = at::empty({2}, keys.options());
-    minmax[0] = std::numeric_limits<int64_t>::max();
-    minmax[1] = std::numeric_limits<int64_t>::min();
-  }
+  // narrow < 0 = speculative: run the int32 path with an in-kernel
+  // overflow flag; the CALLER checks the returned flag at its existing
+  // sync point and re-runs wide if set (no mid-pipeline sync here)
+  bool speculative = staged && narrow < 0;
+  if (speculative) narrow = 1;
   launch_gb_part_hist(keys.data_ptr<int64_t>(), n, shift,
-                      hist.data_ptr<int64_t>(), (int)num_parts,
-                      auto_narrow ? minmax.data_ptr<int64_t>() : nullptr,
+                      hist.data_ptr<int64_t>(), (int)num_parts, nullptr,
                       stream);
-  if (auto_narrow) {
-    auto mm = minmax.cpu();
-    int64_t lo = mm[0].item<int64_t>();
-    int64_t hi = mm[1].item<int64_t>();
-    narrow = (n > 0 && lo >= 0 && hi < (int64_t(1) << 31)) ? 1 : 0;
-  }
   auto offsets = at::zeros({num_parts + 1}, keys.options());
   offsets.narrow(0, 1, num_parts).copy_(at::cumsum(hist, 0));
   auto cursor = offsets.narrow(0, 0, num_parts).clone();
@@ -224,12 +217,15 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
                    ? at::empty({n}, keys.options().dtype(at::kInt))
                    : at::empty({n}, keys.options());
   auto pvals = at::empty({n_aggs, n}, vals.options());
+  auto ovf = at::zeros({1}, keys.options().dtype(at::kInt));
   if (staged) {
     launch_gb_part_scatter_staged(
         keys.data_ptr<int64_t>(), vals.data_ptr<double>(), n, shift,
         cursor.data_ptr<int64_t>(), pkeys.data_ptr(),
         pvals.data_ptr<double>(), scatter_chunk, (int)nt,
-        use_narrow ? 1 : 0, stream);
+        use_narrow ? 1 : 0,
+        (use_narrow && speculative) ? ovf.data_ptr<int32_t>() : nullptr,
+        stream);
   } else {
     launch_gb_part_scatter(keys.data_ptr<int64_t>(), vals.data_ptr<double>(),
                            n_aggs, n, shift, cursor.data_ptr<int64_t>(),
@@ -254,7 +250,7 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
         num_parts, tkeys.data_ptr<int64_t>(), gaggs.data_ptr<double>(),
         gcount.data_ptr<int64_t>(), tsize, stream);
   }
-  return {tkeys, gaggs, gcount};
+  return {tkeys, gaggs, gcount, ovf};
 }
 
 std::vector<at::Tensor> join_build(at::Tensor keys, int64_t tsize) {

@@ -354,23 +354,42 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_lds_kernel(
 
 #define PART_MAX 4096
 
-// LDS-privatized histogram (global atomics only on the per-block flush)
+// LDS-privatized histogram (global atomics only on the per-block flush);
+// optionally also reduces the key min/max (wave shuffle + one device
+// atomic per wave) for the narrow-intermediate auto-detection
 __global__ __launch_bounds__(BLOCK) void gb_part_hist_kernel(
     const int64_t* __restrict__ keys, int64_t n, int shift,
-    int64_t* __restrict__ hist, int num_parts) {
+    int64_t* __restrict__ hist, int num_parts,
+    int64_t* __restrict__ minmax /* [2]={min,max}, may be null */) {
   __shared__ int lhist[PART_MAX];
   for (int i = threadIdx.x; i < num_parts; i += blockDim.x) lhist[i] = 0;
   __syncthreads();
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t kmin = INT64_MAX, kmax = INT64_MIN;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
-    int p = (int)(mix64((uint64_t)keys[i]) >> shift);
+    int64_t k = keys[i];
+    int p = (int)(mix64((uint64_t)k) >> shift);
     atomicAdd(&lhist[p], 1);
+    if (k < kmin) kmin = k;
+    if (k > kmax) kmax = k;
   }
   __syncthreads();
   for (int i = threadIdx.x; i < num_parts; i += blockDim.x) {
     if (lhist[i] > 0)
       atomicAdd((unsigned long long*)&hist[i], (unsigned long long)lhist[i]);
+  }
+  if (minmax != nullptr && kmin <= kmax) {
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+      int64_t omin = __shfl_down(kmin, off, WAVE);
+      int64_t omax = __shfl_down(kmax, off, WAVE);
+      if (omin < kmin) kmin = omin;
+      if (omax > kmax) kmax = omax;
+    }
+    if ((threadIdx.x & (WAVE - 1)) == 0) {
+      atomicMin((long long*)&minmax[0], (long long)kmin);
+      atomicMax((long long*)&minmax[1], (long long)kmax);
+    }
   }
 }
 
@@ -507,7 +526,8 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
     int64_t* __restrict__ cursor,
     KT* __restrict__ out_keys,
     double* __restrict__ out_vals,
-    int64_t chunk) {
+    int64_t chunk,
+    int* __restrict__ ovf /* set when a key overflows KT; may be null */) {
   __shared__ int lhist[STAGE_P];
   __shared__ int64_t lbase[STAGE_P];
   __shared__ int lcnt[STAGE_P];
@@ -538,6 +558,12 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
     __syncthreads();
     for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
       int64_t key = keys[i];
+      if constexpr (sizeof(KT) == 4) {
+        // speculative narrow path: flag any key outside [0, 2^31)
+        if (ovf != nullptr && (uint64_t)key >= (1ULL << 31)) {
+          atomicOr(ovf, 1);
+        }
+      }
       int p = (int)(mix64((uint64_t)key) >> shift);
       int pos = atomicAdd(&lcnt[p], 1);
       double v = stream_ld<NT>(&vals[i]);
@@ -661,7 +687,7 @@ void launch_gb_part_scatter_staged(const int64_t* keys, const double* vals,
                                    int64_t n, int shift, int64_t* cursor,
                                    void* out_keys, double* out_vals,
                                    int64_t chunk, int nt, int narrow,
-                                   hipStream_t stream) {
+                                   int* ovf, hipStream_t stream) {
   if (chunk <= 0) chunk = SCATTER_CHUNK;
   int64_t blocks = (n + chunk - 1) / chunk;
   if (blocks > MAX_GRID) blocks = MAX_GRID;
@@ -672,21 +698,21 @@ void launch_gb_part_scatter_staged(const int64_t* keys, const double* vals,
     if (nt)
       hipLaunchKernelGGL((gb_part_scatter_staged_kernel<true, int32_t>), g, b,
                          0, stream, keys, vals, n, shift, cursor, ok,
-                         out_vals, chunk);
+                         out_vals, chunk, ovf);
     else
       hipLaunchKernelGGL((gb_part_scatter_staged_kernel<false, int32_t>), g,
                          b, 0, stream, keys, vals, n, shift, cursor, ok,
-                         out_vals, chunk);
+                         out_vals, chunk, ovf);
   } else {
     auto* ok = (int64_t*)out_keys;
     if (nt)
       hipLaunchKernelGGL((gb_part_scatter_staged_kernel<true, int64_t>), g, b,
                          0, stream, keys, vals, n, shift, cursor, ok,
-                         out_vals, chunk);
+                         out_vals, chunk, nullptr);
     else
       hipLaunchKernelGGL((gb_part_scatter_staged_kernel<false, int64_t>), g,
                          b, 0, stream, keys, vals, n, shift, cursor, ok,
-                         out_vals, chunk);
+                         out_vals, chunk, nullptr);
   }
 }
 
@@ -730,9 +756,10 @@ void launch_gb_aggregate_part_big(const void* part_keys,
 extern "C" {
 
 void launch_gb_part_hist(const int64_t* keys, int64_t n, int shift,
-                         int64_t* hist, int num_parts, hipStream_t stream) {
+                         int64_t* hist, int num_parts, int64_t* minmax,
+                         hipStream_t stream) {
   hipLaunchKernelGGL(gb_part_hist_kernel, dim3(grid_for(n, 4)), dim3(BLOCK),
-                     0, stream, keys, n, shift, hist, num_parts);
+                     0, stream, keys, n, shift, hist, num_parts, minmax);
 }
 
 void launch_gb_part_scatter(const int64_t* keys, const double* vals,

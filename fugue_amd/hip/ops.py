@@ -391,11 +391,17 @@ def groupby_aggregate(
             if meta is not None:
                 narrow = 1 if sum(meta["widths"]) <= 31 else 0
             else:
-                narrow = -1  # auto: hist kernel reduces key min/max
-        tkeys, gaggs, gcount = ext.gb_aggregate_partitioned(
+                narrow = -1  # speculative: overflow flag checked below
+        tkeys, gaggs, gcount, ovf = ext.gb_aggregate_partitioned(
             packed, vals, ops, num_parts, tsize, sc_chunk, ag_chunk, nt,
             narrow
         )
+        if narrow == -1 and int(ovf.item()) != 0:
+            # a key fell outside [0, 2^31): redo exactly with wide keys
+            tkeys, gaggs, gcount, ovf = ext.gb_aggregate_partitioned(
+                packed, vals, ops, num_parts, tsize, sc_chunk, ag_chunk,
+                nt, 0
+            )
     else:
         use_lds = expected_groups <= 100_000 and sum_count_only
         tkeys, gaggs, gcount = ext.gb_aggregate(
