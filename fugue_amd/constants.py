@@ -31,6 +31,8 @@ FUGUE_COMPILE_TIME_CONFIGS = {
     FUGUE_CONF_WORKFLOW_EXCEPTION_HIDE,
     FUGUE_CONF_WORKFLOW_EXCEPTION_INJECT,
     FUGUE_CONF_WORKFLOW_EXCEPTION_OPTIMIZE,
+    FUGUE_CONF_WORKFLOW_AUTO_PERSIST,
+    FUGUE_CONF_WORKFLOW_AUTO_PERSIST_VALUE,
     FUGUE_CONF_SQL_IGNORE_CASE,
     FUGUE_CONF_SQL_DIALECT,
 }

@@ -108,6 +108,14 @@ class PandasMapEngine(MapEngine):
                 )
             else:
                 input_df = df.as_local()
+                if isinstance(input_df, PandasDataFrame):
+                    # shallow copy: a UDF assigning columns in place must
+                    # not mutate the (possibly persisted) source frame
+                    input_df = PandasDataFrame(
+                        input_df.as_pandas().copy(deep=False),
+                        df.schema,
+                        pandas_df_wrapper=True,
+                    )
             if (
                 len(partition_spec.partition_by) == 0
                 and partition_spec.num_partitions != "0"

@@ -531,6 +531,7 @@ class FugueWorkflow:
         self._yields: Dict[str, Yielded] = {}
         self._computed = False
         self._last_df: Optional[WorkflowDataFrame] = None
+        self._consumers: Dict[int, int] = {}
 
     @property
     def conf(self) -> ParamDict:
@@ -563,6 +564,29 @@ class FugueWorkflow:
         else:
             self._tasks[uid] = task
             self._task_order.append(task)
+            # auto-persist: an upstream consumed by >1 downstream tasks is
+            # persisted so it isn't recomputed per consumer (reference
+            # ``workflow/workflow.py:2228-2241``)
+            from fugue_amd.constants import (
+                FUGUE_CONF_WORKFLOW_AUTO_PERSIST,
+                FUGUE_CONF_WORKFLOW_AUTO_PERSIST_VALUE,
+            )
+
+            for inp in task._inputs:
+                k = id(inp)
+                self._consumers[k] = self._consumers.get(k, 0) + 1
+                if (
+                    self._consumers[k] > 1
+                    and self._conf.get(FUGUE_CONF_WORKFLOW_AUTO_PERSIST, False)
+                    and inp._checkpoint.is_null
+                ):
+                    kwargs: Dict[str, Any] = {}
+                    level = self._conf.get_or_none(
+                        FUGUE_CONF_WORKFLOW_AUTO_PERSIST_VALUE, object
+                    )
+                    if level is not None:
+                        kwargs["level"] = level
+                    inp.set_checkpoint(WeakCheckpoint(lazy=False, **kwargs))
         res = WorkflowDataFrame(self, task)
         self._last_df = res
         return res

@@ -231,3 +231,26 @@ def test_yield_file_and_table_cross_workflow():
         a.union(b, distinct=False).yield_dataframe_as("out")
         res2 = dag2.run(engine)
         assert sorted(r[0] for r in res2["out"].result.as_array()) == [1, 2, 3]
+
+
+def test_auto_persist():
+    from fugue_amd.constants import FUGUE_CONF_WORKFLOW_AUTO_PERSIST
+
+    calls = []
+
+    def count_calls(df: pd.DataFrame) -> pd.DataFrame:
+        calls.append(1)
+        return df
+
+    dag = FugueWorkflow({FUGUE_CONF_WORKFLOW_AUTO_PERSIST: True})
+    a = dag.df([[1]], "x:long").transform(count_calls, schema="*")
+    a.transform(_double_x, schema="*").yield_dataframe_as("r1")
+    a.transform(lambda df: df, schema="*") if False else None
+    b = a.filter(col("x") >= 0)
+    b.yield_dataframe_as("r2")
+    res = dag.run()
+    assert res["r1"].result.as_array() == [[2]]
+    assert res["r2"].result.as_array() == [[1]]
+    # `a` is consumed twice; the auto-persist marks it with a weak
+    # checkpoint so its transform runs once
+    assert len(calls) == 1
