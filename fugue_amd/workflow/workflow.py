@@ -554,9 +554,18 @@ class FugueWorkflow:
 
     def add(self, task: FugueTask) -> WorkflowDataFrame:
         """Add a task, dedup by spec uuid (determinism)."""
+        from fugue_amd.constants import (
+            FUGUE_CONF_WORKFLOW_EXCEPTION_HIDE,
+            FUGUE_CONF_WORKFLOW_EXCEPTION_OPTIMIZE,
+        )
+
         inject = int(self._conf.get(FUGUE_CONF_WORKFLOW_EXCEPTION_INJECT, 3))
+        optimize = bool(self._conf.get(FUGUE_CONF_WORKFLOW_EXCEPTION_OPTIMIZE, True))
+        hide = str(self._conf.get(FUGUE_CONF_WORKFLOW_EXCEPTION_HIDE, "fugue_amd."))
+        prefixes = tuple(x for x in hide.split(",") if x != "")
         task.reset_traceback(
-            inject, make_prune_predicate(("fugue_amd",)) if inject > 0 else None
+            inject if optimize else 0,
+            make_prune_predicate(prefixes) if inject > 0 and optimize else None,
         )
         uid = task.__uuid__()
         if uid in self._tasks:
@@ -845,11 +854,34 @@ class FugueWorkflow:
 
     # --- run -------------------------------------------------------------- #
     def run(self, engine: Any = None, conf: Any = None, **kwargs: Any) -> FugueWorkflowResult:
+        import sys as _sys
+
+        from fugue_amd.constants import (
+            FUGUE_CONF_WORKFLOW_EXCEPTION_HIDE,
+            FUGUE_CONF_WORKFLOW_EXCEPTION_OPTIMIZE,
+        )
+        from fugue_amd.utils.exception import modify_traceback
+
         e = make_execution_engine(engine, conf, **kwargs)
         ctx = FugueWorkflowContext(
             e, yields_as_local=getattr(self, "_yields_as_local", False)
         )
-        ctx.run(self._task_order)
+        try:
+            ctx.run(self._task_order)
+        except Exception as ex:
+            if not bool(
+                self._conf.get(FUGUE_CONF_WORKFLOW_EXCEPTION_OPTIMIZE, True)
+            ):
+                raise
+            hide = str(self._conf.get(FUGUE_CONF_WORKFLOW_EXCEPTION_HIDE, ""))
+            prefixes = tuple(
+                x.rstrip(".") for x in hide.split(",") if x != ""
+            )
+            if len(prefixes) == 0:
+                raise
+            raise modify_traceback(
+                ex, None, make_prune_predicate(prefixes)
+            ) from None
         self._computed = True
         return FugueWorkflowResult(self._yields)
 
