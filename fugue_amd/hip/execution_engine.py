@@ -244,6 +244,29 @@ class HipMapEngine(MapEngine):
                 staged_pandas_batches,
             )
 
+            if map_func_format_hint == "device":
+                # device-resident UDFs (HipDataFrame-annotated): slice the
+                # sorted shard per logical partition, never leave HBM
+                dev_results: List[HipDataFrame] = []
+                for gi in range(len(bounds) - 1):
+                    start, end = bounds[gi], bounds[gi + 1]
+                    sub = sorted_df.slice_rows(start, end - start)
+                    cursor.set(lambda: sub.peek_array(), gi, 0)
+                    res = map_func(cursor, sub)
+                    d = engine.to_df(res)
+                    if not isinstance(d, HipDataFrame):
+                        d = HipDataFrame(
+                            res.as_arrow(), output_schema, engine._device
+                        )
+                    dev_results.append(d)
+                if len(dev_results) == 0:
+                    return engine.to_df(
+                        PandasDataFrame(
+                            output_schema.create_empty_pandas(), output_schema
+                        ),
+                        shard_replicated=False,
+                    )
+                return dev_results[0].concat_with(dev_results[1:])
             if map_func_format_hint == "pyarrow":
                 # arrow-native UDFs: zero pandas conversion
                 table = sorted_df.as_arrow()

@@ -207,3 +207,49 @@ def test_sql_like_on_engine():
         parse_select("SELECT k FROM a WHERE s NOT LIKE 'f%'"), dict(a=d), e
     ).as_pandas()
     assert r2["k"].tolist() == [1]
+
+
+def _device_group_sum(df):
+    import pyarrow as pa
+    import torch
+
+    from fugue_amd.hip.frame import DeviceColumn, HipDataFrame
+    from fugue_amd.schema import Schema
+
+    k = df.col("k").data[:1]
+    s = df.col("v").data.sum().reshape(1)
+    return HipDataFrame.from_columns(
+        {
+            "k": DeviceColumn(k, None, pa.int64()),
+            "s": DeviceColumn(s.to(torch.float64), None, pa.float64()),
+        },
+        Schema("k:long,s:double"),
+        df.device,
+    )
+
+
+def test_device_resident_udf():
+    """HipDataFrame-annotated transformers get the HBM shard directly
+    (fugue_polars pattern; reference fugue_polars/registry.py:24)."""
+    import pandas as pd
+
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+    from fugue_amd.hip.frame import HipDataFrame
+    from fugue_amd.workflow import transform
+
+    _device_group_sum.__annotations__ = {
+        "df": HipDataFrame,
+        "return": HipDataFrame,
+    }
+    e = HipExecutionEngine()
+    pdf = pd.DataFrame(dict(k=[1, 1, 2, 2, 2], v=[1.0, 2.0, 3.0, 4.0, 5.0]))
+    res = transform(
+        pdf,
+        _device_group_sum,
+        schema="k:long,s:double",
+        partition=dict(by=["k"]),
+        engine=e,
+    )
+    r = res if isinstance(res, pd.DataFrame) else res.as_pandas()
+    r = r.sort_values("k").reset_index(drop=True)
+    assert r["s"].tolist() == [3.0, 12.0]
