@@ -327,3 +327,30 @@ def test_q3_gpu(engine):
     np.testing.assert_allclose(
         got["revenue"].values, exp["revenue"].values, rtol=1e-9
     )
+
+
+def test_device_resident_udf_gpu():
+    import pandas as pd
+
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+    from fugue_amd.hip.frame import HipDataFrame
+    from fugue_amd.workflow import transform
+
+    def dev_top1(df: HipDataFrame) -> HipDataFrame:
+        import torch
+
+        v = df.col("v").data
+        i = torch.argmax(v).reshape(1)
+        return df.gather_rows(i)
+
+    e = HipExecutionEngine()
+    pdf = pd.DataFrame(
+        dict(k=[1, 1, 2, 2, 2], v=[1.0, 5.0, 3.0, 9.0, 2.0])
+    )
+    res = transform(
+        pdf, dev_top1, schema="k:long,v:double",
+        partition=dict(by=["k"]), engine=e,
+    )
+    r = res if isinstance(res, pd.DataFrame) else res.as_pandas()
+    r = r.sort_values("k").reset_index(drop=True)
+    assert r["v"].tolist() == [5.0, 9.0]
