@@ -256,9 +256,41 @@ def _mentioned_cols(e: X.Expr) -> List[str]:
         elif isinstance(x, X.FuncCall):
             for a in x.args:
                 _walk(a)
+        elif isinstance(x, X.Like):
+            _walk(x.expr)
+        elif isinstance(x, X.Case):
+            for cnd, val in x.whens:
+                _walk(cnd)
+                _walk(val)
+            if x.else_ is not None:
+                _walk(x.else_)
 
     _walk(e)
     return out
+
+
+def _needed_columns(stmt: X.SelectStmt) -> Optional[set]:
+    """All column names the statement can reference; None when a ``*``
+    makes every column live."""
+    needed: set = set()
+    for e, _alias in stmt.columns:
+        if isinstance(e, X.Star):
+            return None
+        needed.update(_mentioned_cols(e))
+    for e in (stmt.where, stmt.having):
+        if e is not None:
+            needed.update(_mentioned_cols(e))
+    for e in stmt.group_by:
+        if not isinstance(e, X.Star):
+            needed.update(_mentioned_cols(e))
+    for o in stmt.order_by:
+        needed.update(_mentioned_cols(o.expr))
+    for j in stmt.joins:
+        if j.on is not None:
+            needed.update(_mentioned_cols(j.on))
+        if j.using is not None:
+            needed.update(j.using)
+    return needed
 
 
 def _execute_core(
@@ -311,6 +343,29 @@ def _execute_core(
                 if residual_where is None
                 else X.BinOp("AND", residual_where, conj)
             )
+    # projection pushdown: drop base-table columns not referenced anywhere
+    # in the statement before joining (string columns in particular make
+    # join-output gathers expensive)
+    if stmt.joins:
+        frames2: List[DataFrame] = [res] + [
+            join_inputs[i]
+            if i < len(join_inputs)
+            else _resolve_from(stmt.joins[i].item, tables, engine)
+            for i in range(len(stmt.joins))
+        ]
+        needed = _needed_columns(stmt)
+        if needed is not None:
+            pruned: List[DataFrame] = []
+            for fr in frames2:
+                keep = [n for n in fr.schema.names if n in needed]
+                if 0 < len(keep) < len(fr.schema.names):
+                    fr = engine.select(
+                        fr, SelectColumns(*[col(n) for n in keep])
+                    )
+                pruned.append(fr)
+            frames2 = pruned
+        res = frames2[0]
+        join_inputs = frames2[1:]
     for ji, j in enumerate(stmt.joins):
         right = (
             join_inputs[ji]
