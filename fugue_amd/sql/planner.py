@@ -269,15 +269,18 @@ def _mentioned_cols(e: X.Expr) -> List[str]:
     return out
 
 
-def _needed_columns(stmt: X.SelectStmt) -> Optional[set]:
+def _needed_columns(
+    stmt: X.SelectStmt, where: Optional[X.Expr]
+) -> Optional[set]:
     """All column names the statement can reference; None when a ``*``
-    makes every column live."""
+    makes every column live.  ``where`` is the residual (post-pushdown)
+    predicate — columns only used by already-pushed filters are dead."""
     needed: set = set()
     for e, _alias in stmt.columns:
         if isinstance(e, X.Star):
             return None
         needed.update(_mentioned_cols(e))
-    for e in (stmt.where, stmt.having):
+    for e in (where, stmt.having):
         if e is not None:
             needed.update(_mentioned_cols(e))
     for e in stmt.group_by:
@@ -353,7 +356,7 @@ def _execute_core(
             else _resolve_from(stmt.joins[i].item, tables, engine)
             for i in range(len(stmt.joins))
         ]
-        needed = _needed_columns(stmt)
+        needed = _needed_columns(stmt, residual_where)
         if needed is not None:
             pruned: List[DataFrame] = []
             for fr in frames2:
