@@ -47,6 +47,13 @@ void launch_gb_aggregate_part_big(const void*, const double*,
 void launch_join_count(const int64_t*, int64_t, const int64_t*,
                        const int64_t*, const int64_t*, const int32_t*,
                        const int32_t*, int64_t, int32_t*, hipStream_t);
+void launch_join_total(const int64_t*, int64_t, const int64_t*,
+                       const int64_t*, const int64_t*, const int32_t*,
+                       const int32_t*, int64_t, int, int64_t*, hipStream_t);
+void launch_join_emit_chunked(const int64_t*, int64_t, const int64_t*,
+                              const int64_t*, const int64_t*, const int32_t*,
+                              const int32_t*, int64_t, int64_t*, int64_t*,
+                              int64_t*, int, hipStream_t);
 void launch_join_emit(const int64_t*, int64_t, const int64_t*, const int64_t*,
                       const int64_t*, const int32_t*, const int32_t*, int64_t,
                       const int64_t*, int64_t*, int64_t*, int, hipStream_t);
@@ -291,6 +298,39 @@ at::Tensor join_count(at::Tensor pkeys, at::Tensor bkeys,
   return counts;
 }
 
+std::vector<at::Tensor> join_pairs(at::Tensor pkeys, at::Tensor bkeys,
+                                   c10::optional<at::Tensor> ph2,
+                                   c10::optional<at::Tensor> bh2,
+                                   at::Tensor heads, at::Tensor next,
+                                   int64_t tsize, int64_t mode) {
+  // one-scalar total + chunked single-reservation emit (no per-row
+  // counts array / prefix sum)
+  check_gpu(pkeys, "pkeys");
+  int64_t np = pkeys.numel();
+  auto stream = current_stream();
+  auto total_t = at::zeros({1}, pkeys.options());
+  if (np > 0) {
+    launch_join_total(pkeys.data_ptr<int64_t>(), np,
+                      bkeys.data_ptr<int64_t>(), opt_i64_ptr(ph2),
+                      opt_i64_ptr(bh2), heads.data_ptr<int32_t>(),
+                      next.data_ptr<int32_t>(), tsize, (int)mode,
+                      total_t.data_ptr<int64_t>(), stream);
+  }
+  int64_t out_n = total_t.cpu().item<int64_t>();
+  auto out_p = at::empty({out_n}, pkeys.options());
+  auto out_b = at::empty({out_n}, pkeys.options());
+  if (np > 0 && out_n > 0) {
+    auto cursor = at::zeros({1}, pkeys.options());
+    launch_join_emit_chunked(
+        pkeys.data_ptr<int64_t>(), np, bkeys.data_ptr<int64_t>(),
+        opt_i64_ptr(ph2), opt_i64_ptr(bh2), heads.data_ptr<int32_t>(),
+        next.data_ptr<int32_t>(), tsize, cursor.data_ptr<int64_t>(),
+        out_p.data_ptr<int64_t>(), out_b.data_ptr<int64_t>(), (int)mode,
+        stream);
+  }
+  return {out_p, out_b};
+}
+
 std::vector<at::Tensor> join_emit(at::Tensor pkeys, at::Tensor bkeys,
                                   c10::optional<at::Tensor> ph2,
                                   c10::optional<at::Tensor> bh2,
@@ -409,5 +449,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused masked compaction of 8-byte columns");
   m.def("join_count", &join_count, "count matches per probe row");
   m.def("join_emit", &join_emit, "emit join pairs");
+  m.def("join_pairs", &join_pairs,
+        "total + chunked-reservation join pair emission");
   m.def("join_mark_build", &join_mark_build, "mark matched build rows");
 }

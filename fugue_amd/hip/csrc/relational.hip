@@ -893,6 +893,125 @@ __global__ __launch_bounds__(BLOCK) void join_emit_kernel(
   }
 }
 
+// single-scalar total of output rows for a join mode (no per-row counts
+// array, no prefix sum): per-thread count -> wave reduce -> one device
+// atomic per wave
+__global__ __launch_bounds__(BLOCK) void join_total_kernel(
+    const int64_t* __restrict__ pkeys, int64_t np,
+    const int64_t* __restrict__ bkeys,
+    const int64_t* __restrict__ ph2, const int64_t* __restrict__ bh2,
+    const int32_t* __restrict__ heads, const int32_t* __restrict__ next,
+    int64_t tsize, int mode, int64_t* __restrict__ total) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t mine = 0;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < np;
+       i += stride) {
+    int64_t key = pkeys[i];
+    uint64_t h = mix64((uint64_t)key);
+    int32_t cur = heads[(int64_t)(h & (uint64_t)(tsize - 1))];
+    int64_t c = 0;
+    while (cur >= 0) {
+      if (bkeys[cur] == key && (ph2 == nullptr || bh2[cur] == ph2[i])) {
+        ++c;
+        if (mode >= 2) break;  // semi/anti: presence is enough
+      }
+      cur = next[cur];
+    }
+    if (mode == 0) mine += c;
+    else if (mode == 1) mine += (c > 0 ? c : 1);
+    else if (mode == 2) mine += (c > 0 ? 1 : 0);
+    else mine += (c == 0 ? 1 : 0);
+  }
+  for (int off = WAVE / 2; off > 0; off >>= 1)
+    mine += __shfl_down(mine, off, WAVE);
+  if ((threadIdx.x & (WAVE - 1)) == 0 && mine > 0)
+    atomicAdd((unsigned long long*)total, (unsigned long long)mine);
+}
+
+// chunked single-reservation emit: each block counts its chunk's output
+// rows (chain walk; the chunk's keys and table lines stay hot in L1/L2
+// for the second walk), reserves one contiguous range with a single
+// global atomic, then emits.  Replaces the per-row counts array +
+// device-wide prefix sum + offset reads of the 2-pass scheme.
+#define JOIN_CHUNK (BLOCK * 16)
+// per-wave sub-chunks: each wave counts its rows (chain walk), does a
+// wave-local exclusive scan + ONE global atomic reservation, then emits
+// re-walking the same rows (hot in L1/L2).  No LDS, no cross-wave sync.
+#define JOIN_SUB (JOIN_CHUNK / (BLOCK / WAVE))
+__global__ __launch_bounds__(BLOCK) void join_emit_chunked_kernel(
+    const int64_t* __restrict__ pkeys, int64_t np,
+    const int64_t* __restrict__ bkeys,
+    const int64_t* __restrict__ ph2, const int64_t* __restrict__ bh2,
+    const int32_t* __restrict__ heads, const int32_t* __restrict__ next,
+    int64_t tsize, int64_t* __restrict__ cursor,
+    int64_t* __restrict__ out_p, int64_t* __restrict__ out_b, int mode) {
+  int wave = threadIdx.x / WAVE;
+  int lane = threadIdx.x & (WAVE - 1);
+  for (int64_t cstart = (int64_t)blockIdx.x * JOIN_CHUNK; cstart < np;
+       cstart += (int64_t)gridDim.x * JOIN_CHUNK) {
+    int64_t start = cstart + (int64_t)wave * JOIN_SUB;
+    int64_t end = start + JOIN_SUB;
+    if (end > np) end = np;
+    int64_t mine = 0;
+    for (int64_t i = start + lane; i < end; i += WAVE) {
+      int64_t key = pkeys[i];
+      uint64_t h = mix64((uint64_t)key);
+      int32_t cur = heads[(int64_t)(h & (uint64_t)(tsize - 1))];
+      int c = 0;
+      while (cur >= 0) {
+        if (bkeys[cur] == key && (ph2 == nullptr || bh2[cur] == ph2[i])) {
+          ++c;
+          if (mode >= 2) break;
+        }
+        cur = next[cur];
+      }
+      if (mode == 0) mine += c;
+      else if (mode == 1) mine += (c > 0 ? c : 1);
+      else if (mode == 2) mine += (c > 0 ? 1 : 0);
+      else mine += (c == 0 ? 1 : 0);
+    }
+    // wave-local inclusive scan -> exclusive offset + total
+    int64_t inc = mine;
+    for (int d = 1; d < WAVE; d <<= 1) {
+      int64_t nv = __shfl_up(inc, d, WAVE);
+      if (lane >= d) inc += nv;
+    }
+    int64_t tot = __shfl(inc, WAVE - 1, WAVE);
+    int64_t base = 0;
+    if (lane == 0 && tot > 0)
+      base = (int64_t)atomicAdd((unsigned long long*)cursor,
+                                (unsigned long long)tot);
+    base = __shfl(base, 0, WAVE);
+    if (tot == 0) continue;
+    int64_t pos = base + (inc - mine);
+    for (int64_t i = start + lane; i < end; i += WAVE) {
+      int64_t key = pkeys[i];
+      uint64_t h = mix64((uint64_t)key);
+      int32_t cur = heads[(int64_t)(h & (uint64_t)(tsize - 1))];
+      bool any = false;
+      while (cur >= 0) {
+        if (bkeys[cur] == key && (ph2 == nullptr || bh2[cur] == ph2[i])) {
+          any = true;
+          if (mode <= 2) {
+            out_p[pos] = i;
+            out_b[pos] = cur;
+            ++pos;
+            if (mode == 2) break;
+          } else {
+            break;
+          }
+        }
+        cur = next[cur];
+      }
+      if (!any && (mode == 1 || mode == 3)) {
+        out_p[pos] = i;
+        out_b[pos] = -1;
+        ++pos;
+      }
+    }
+  }
+}
+
 // match-flag on the build side (for right/full outer): mark matched rows
 __global__ __launch_bounds__(BLOCK) void join_mark_build_kernel(
     const int64_t* __restrict__ pkeys, int64_t np,
@@ -929,6 +1048,30 @@ void launch_join_count(const int64_t* pkeys, int64_t np, const int64_t* bkeys,
   hipLaunchKernelGGL(join_count_kernel, dim3(grid_for(np)), dim3(BLOCK), 0,
                      stream, pkeys, np, bkeys, ph2, bh2, heads, next, tsize,
                      counts);
+}
+
+void launch_join_total(const int64_t* pkeys, int64_t np,
+                       const int64_t* bkeys, const int64_t* ph2,
+                       const int64_t* bh2, const int32_t* heads,
+                       const int32_t* next, int64_t tsize, int mode,
+                       int64_t* total, hipStream_t stream) {
+  hipLaunchKernelGGL(join_total_kernel, dim3(grid_for(np, 4)), dim3(BLOCK), 0,
+                     stream, pkeys, np, bkeys, ph2, bh2, heads, next, tsize,
+                     mode, total);
+}
+
+void launch_join_emit_chunked(const int64_t* pkeys, int64_t np,
+                              const int64_t* bkeys, const int64_t* ph2,
+                              const int64_t* bh2, const int32_t* heads,
+                              const int32_t* next, int64_t tsize,
+                              int64_t* cursor, int64_t* out_p,
+                              int64_t* out_b, int mode, hipStream_t stream) {
+  int64_t blocks = (np + JOIN_CHUNK - 1) / JOIN_CHUNK;
+  if (blocks > MAX_GRID) blocks = MAX_GRID;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(join_emit_chunked_kernel, dim3((int)blocks), dim3(BLOCK),
+                     0, stream, pkeys, np, bkeys, ph2, bh2, heads, next,
+                     tsize, cursor, out_p, out_b, mode);
 }
 
 void launch_join_emit(const int64_t* pkeys, int64_t np, const int64_t* bkeys,

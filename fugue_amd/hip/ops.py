@@ -658,9 +658,14 @@ def hash_join_indices(
             probe_keys, build_keys, how, probe_h2, build_h2
         )
     ext = get_ext()
+    if how not in ("inner", "left", "semi", "anti"):
+        raise FugueBug(f"unsupported join mode {how}")
     nb = int(build_keys.numel())
     tsize = _next_pow2(max(16, nb * 2))
     heads, nxt = ext.join_build(build_keys, tsize)
+    # 2-pass count+prefix+emit; a 3-pass total+chunked-reservation variant
+    # (ext.join_pairs) measured SLOWER — the random chain walk dominates,
+    # not the streaming counts/cumsum (profiles/NOTES.md)
     counts = ext.join_count(
         probe_keys, build_keys, probe_h2, build_h2, heads, nxt, tsize
     )
@@ -671,10 +676,8 @@ def hash_join_indices(
         out_counts = torch.clamp(counts64, min=1)
     elif how == "semi":
         out_counts = (counts64 > 0).to(torch.int64)
-    elif how == "anti":
-        out_counts = (counts64 == 0).to(torch.int64)
     else:
-        raise FugueBug(f"unsupported join mode {how}")
+        out_counts = (counts64 == 0).to(torch.int64)
     offsets = torch.zeros_like(out_counts)
     if out_counts.numel() > 1:
         torch.cumsum(out_counts[:-1], 0, out=offsets[1:])
