@@ -361,3 +361,86 @@ def test_distributed_count_distinct_first():
         assert got["cd"] == expected["cd"].tolist()
         for k, fv in zip(got["k"], got["fv"]):
             assert fv in member_vals[k]
+
+
+def _rowops_job(rank: int):
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    rng = np.random.default_rng(5)
+    pdf = pd.DataFrame(
+        dict(
+            k=rng.integers(0, 30, 3000).astype("f8"),
+            v=rng.random(3000),
+        )
+    )
+    pdf.loc[pdf.index % 7 == 0, "k"] = np.nan
+    d = e.to_df(pdf)
+    out = {}
+    out["dropna"] = e._gather_all(e.dropna(d)).as_pandas().shape[0]
+    out["fillna"] = float(
+        e._gather_all(e.fillna(d, 0.0)).as_pandas()["k"].sum()
+    )
+    out["distinct"] = e._gather_all(
+        e.distinct(e._select_columns_by_names(d, ["k"]))
+        if hasattr(e, "_select_columns_by_names")
+        else e.distinct(d)
+    ).as_pandas().shape[0]
+    t = e._gather_all(e.take(d, 5, presort="v desc")).as_pandas()
+    out["take"] = t["v"].tolist()
+    s = e._gather_all(e.sample(d, frac=0.5, seed=3)).as_pandas().shape[0]
+    out["sample_n"] = s
+    return out
+
+
+def test_distributed_row_ops():
+    results = run_distributed(_rowops_job, 29527)
+    rng = np.random.default_rng(5)
+    pdf = pd.DataFrame(
+        dict(
+            k=rng.integers(0, 30, 3000).astype("f8"),
+            v=rng.random(3000),
+        )
+    )
+    pdf.loc[pdf.index % 7 == 0, "k"] = np.nan
+    exp_drop = pdf.dropna().shape[0]
+    exp_fill = float(pdf["k"].fillna(0.0).sum())
+    exp_take = pdf.nlargest(5, "v")["v"].tolist()
+    for rank, got in results.items():
+        assert got["dropna"] == exp_drop
+        assert abs(got["fillna"] - exp_fill) < 1e-6
+        assert got["take"] == exp_take
+        # bernoulli sample: expect roughly half (loose bounds)
+        assert 1000 < got["sample_n"] < 2000
+
+
+def _setops_job(rank: int):
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    rng = np.random.default_rng(6)
+    a = pd.DataFrame(dict(x=rng.integers(0, 200, 1000)))
+    b = pd.DataFrame(dict(x=rng.integers(100, 300, 1000)))
+    da, db = e.to_df(a), e.to_df(b)
+    out = {}
+    out["union"] = sorted(
+        e._gather_all(e.union(da, db, distinct=True)).as_pandas()["x"].tolist()
+    )
+    out["sub"] = sorted(
+        e._gather_all(e.subtract(da, db, distinct=True)).as_pandas()["x"].tolist()
+    )
+    out["inter"] = sorted(
+        e._gather_all(e.intersect(da, db, distinct=True)).as_pandas()["x"].tolist()
+    )
+    return out
+
+
+def test_distributed_set_ops():
+    results = run_distributed(_setops_job, 29529)
+    rng = np.random.default_rng(6)
+    a = set(rng.integers(0, 200, 1000).tolist())
+    b = set(rng.integers(100, 300, 1000).tolist())
+    for rank, got in results.items():
+        assert got["union"] == sorted(a | b)
+        assert got["sub"] == sorted(a - b)
+        assert got["inter"] == sorted(a & b)
