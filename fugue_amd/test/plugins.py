@@ -18,6 +18,30 @@ from fugue_amd.execution.factory import make_execution_engine
 
 _BACKENDS: Dict[str, Type["FugueTestBackend"]] = {}
 _LOCK = threading.RLock()
+_GLOBAL_TEST_CONF: Dict[str, Any] = {}
+
+
+def set_global_test_conf(conf: Dict[str, Any]) -> None:
+    """Install the parsed ``fugue_test_conf`` pytest-ini mapping (called
+    by ``fugue_amd.test.pytest_plugin.pytest_configure``)."""
+    with _LOCK:
+        _GLOBAL_TEST_CONF.clear()
+        _GLOBAL_TEST_CONF.update(conf)
+
+
+def _backend_conf(name: str, base: Dict[str, Any]) -> Dict[str, Any]:
+    """Merge backend class conf with ini entries: ``<name>.key`` entries
+    override (prefix stripped); unprefixed entries apply to all backends
+    (reference parity: ``fugue/test/plugins.py`` ``fugue_test_conf``)."""
+    out = dict(base)
+    with _LOCK:
+        for k, v in _GLOBAL_TEST_CONF.items():
+            if "." in k and k.split(".", 1)[0] in _BACKENDS:
+                if k.startswith(name + "."):
+                    out[k.split(".", 1)[1]] = v
+            else:
+                out[k] = v
+    return out
 
 
 @dataclass
@@ -41,7 +65,8 @@ class FugueTestBackend:
     def context(cls) -> Iterator[FugueTestContext]:
         with cls.session_context() as session:
             engine = make_execution_engine(
-                cls.name if session is None else session, cls.conf
+                cls.name if session is None else session,
+                _backend_conf(cls.name, cls.conf),
             )
             yield FugueTestContext(engine=engine, session=session, name=cls.name)
 

@@ -92,3 +92,28 @@ def test_map_bag():
     )
     assert sorted(res.as_array()) == [10, 20, 30, 40, 50]
     assert len(inits) == 2
+
+
+def test_fugue_test_conf_backend_merge():
+    from fugue_amd.test.plugins import (
+        _backend_conf,
+        set_global_test_conf,
+    )
+
+    set_global_test_conf(
+        {
+            "fugue.workflow.concurrency": 4,
+            "hip.fugue.hip.broadcast_threshold_bytes": 1024,
+            "native.some.key": "x",
+        }
+    )
+    try:
+        hip = _backend_conf("hip", {})
+        assert hip["fugue.workflow.concurrency"] == 4
+        assert hip["fugue.hip.broadcast_threshold_bytes"] == 1024
+        assert "some.key" not in hip
+        nat = _backend_conf("native", {})
+        assert nat["some.key"] == "x"
+        assert "fugue.hip.broadcast_threshold_bytes" not in nat
+    finally:
+        set_global_test_conf({})
