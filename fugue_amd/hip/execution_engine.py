@@ -1823,26 +1823,48 @@ class HipExecutionEngine(ExecutionEngine):
             and os.path.isdir(path)
             and len(self._part_files(path)) > 0
         ):
-            # partitioned parquet dataset: each rank reads its file subset
-            import pyarrow.parquet as pq
-
+            # partitioned dataset (parquet/csv/json part files): each
+            # rank reads its file subset
             files = self._part_files(path)
+            fmt = os.path.splitext(files[0])[1].lstrip(".")
             mine = files[self.rank :: max(1, self.world_size)]
             col_names = None
             if isinstance(columns, list):
                 col_names = columns
             elif columns is not None:
                 col_names = Schema(columns).names
-            if len(mine) > 0:
-                tables = [pq.read_table(f, columns=col_names) for f in mine]
-                table = (
-                    pa.concat_tables(tables) if len(tables) > 1 else tables[0]
-                )
+            if fmt == "parquet":
+                import pyarrow.parquet as pq
+
+                if len(mine) > 0:
+                    tables = [
+                        pq.read_table(f, columns=col_names) for f in mine
+                    ]
+                    table = (
+                        pa.concat_tables(tables)
+                        if len(tables) > 1
+                        else tables[0]
+                    )
+                else:
+                    table = pq.read_table(files[0], columns=col_names).slice(
+                        0, 0
+                    )
+                return self.to_df(ArrowDataFrame(table), shard_replicated=False)
+            read = mine if len(mine) > 0 else files[:1]
+            if fmt == "csv":
+                parts = [pd.read_csv(f, usecols=col_names) for f in read]
             else:
-                table = pq.read_table(files[0], columns=col_names).slice(0, 0)
-            return self.to_df(
-                ArrowDataFrame(table), shard_replicated=False
+                parts = [pd.read_json(f, orient="records", lines=True) for f in read]
+                if col_names is not None:
+                    parts = [p[col_names] for p in parts]
+            pdf = (
+                pd.concat(parts, ignore_index=True)
+                if len(parts) > 1
+                else parts[0]
             )
+            if len(mine) == 0:
+                pdf = pdf.head(0)
+            return self.to_df(PandasDataFrame(pdf), shard_replicated=False)
         pdf, schema = _io.load_df(
             path, format_hint=format_hint, columns=columns, **kwargs
         )
@@ -1853,7 +1875,10 @@ class HipExecutionEngine(ExecutionEngine):
     def _part_files(path: str) -> List[str]:
         import glob as _glob
 
-        return sorted(_glob.glob(os.path.join(path, "part-*.parquet")))
+        out: List[str] = []
+        for ext in ("parquet", "csv", "json"):
+            out.extend(_glob.glob(os.path.join(path, f"part-*.{ext}")))
+        return sorted(out)
 
     def save_df(
         self,
@@ -1872,7 +1897,11 @@ class HipExecutionEngine(ExecutionEngine):
         fmt = infer_format(path, format_hint if format_hint else None) if (
             format_hint or "." in os.path.basename(path)
         ) else "parquet"
-        if self.is_distributed and not force_single and fmt == "parquet":
+        if (
+            self.is_distributed
+            and not force_single
+            and fmt in ("parquet", "csv", "json")
+        ):
             # each rank writes its shard as a part file (no gather)
             if self.rank == 0:
                 if os.path.exists(path) and mode == "overwrite":
@@ -1884,12 +1913,22 @@ class HipExecutionEngine(ExecutionEngine):
                         os.remove(path)
                 os.makedirs(path, exist_ok=True)
             self._comm.barrier()
-            import pyarrow.parquet as pq
+            part = os.path.join(path, f"part-{self.rank:05d}.{fmt}")
+            if fmt == "parquet":
+                import pyarrow.parquet as pq
 
-            local = d.as_arrow() if hasattr(d, "as_arrow") else d.as_local().as_arrow()
-            pq.write_table(
-                local, os.path.join(path, f"part-{self.rank:05d}.parquet")
-            )
+                local = (
+                    d.as_arrow()
+                    if hasattr(d, "as_arrow")
+                    else d.as_local().as_arrow()
+                )
+                pq.write_table(local, part)
+            elif fmt == "csv":
+                d.as_pandas().to_csv(
+                    part, index=False, header=kwargs.get("header", True)
+                )
+            else:
+                d.as_pandas().to_json(part, orient="records", lines=True)
             self._comm.barrier()
             return
         local_df = self._as_local(d)  # gathered on every rank

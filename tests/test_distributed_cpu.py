@@ -258,7 +258,12 @@ def _io_job(rank: int):
     total = e.comm.allreduce_sum(back.count())
     local = e._gather_all(back) if hasattr(back, "as_arrow") else back
     s = sorted(r[0] for r in local.as_array())
-    return dict(total=total, keys=s)
+    # csv part-file round trip
+    cpath = os.path.join(os.environ["FUGUE_TEST_TMP"], "out.csv")
+    e.save_df(d, cpath, format_hint="csv")
+    cback = e.load_df(cpath)
+    ctotal = e.comm.allreduce_sum(cback.count())
+    return dict(total=total, keys=s, ctotal=ctotal)
 
 
 def test_distributed_parquet_parts(tmp_path):
@@ -267,6 +272,7 @@ def test_distributed_parquet_parts(tmp_path):
         results = run_distributed(_io_job, 29521)
         assert results[0]["total"] == 100
         assert results[0]["keys"] == list(range(100))
+        assert results[0]["ctotal"] == 100
         files = os.listdir(os.path.join(str(tmp_path), "out"))
         assert len([f for f in files if f.startswith("part-")]) == 2
     finally:
