@@ -231,10 +231,17 @@ __device__ __forceinline__ int64_t gb_probe_insert(
   uint64_t h = mix64((uint64_t)key);
   int64_t slot = (int64_t)(h & (uint64_t)(tsize - 1));
   while (true) {
-    long long prev = (long long)atomicCAS(
-        (unsigned long long*)&tkeys[slot], (unsigned long long)GB_EMPTY,
-        (unsigned long long)key);
-    if (prev == GB_EMPTY || prev == key) return slot;
+    // plain load first; the CAS (global RMW) only runs for empty slots
+    int64_t cur = tkeys[slot];
+    if (cur == key) return slot;
+    if (cur == GB_EMPTY) {
+      long long prev = (long long)atomicCAS(
+          (unsigned long long*)&tkeys[slot], (unsigned long long)GB_EMPTY,
+          (unsigned long long)key);
+      if (prev == GB_EMPTY || prev == key) return slot;
+      // lost the race to a different key: fall through and advance (a
+      // re-read could serve a stale line from the per-XCD cache)
+    }
     slot = (slot + 1) & (tsize - 1);
   }
 }
@@ -654,8 +661,14 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
       int slot = (int)(h & (LDS_SLOTS_BIG - 1));
       bool in_lds = false;
       for (int probe = 0; probe < 32; ++probe) {
-        KT prev = (KT)lds_key_cas(&lkeys[slot], EMPTY, key);
-        if (prev == EMPTY || prev == key) { in_lds = true; break; }
+        // plain ds_read first: with ~100x key repetition per chunk the
+        // slot almost always already owns the key — skip the LDS RMW
+        KT cur = lkeys[slot];
+        if (cur == key) { in_lds = true; break; }
+        if (cur == EMPTY) {
+          KT prev = (KT)lds_key_cas(&lkeys[slot], EMPTY, key);
+          if (prev == EMPTY || prev == key) { in_lds = true; break; }
+        }
         slot = (slot + 1) & (LDS_SLOTS_BIG - 1);
       }
       double v = is_count ? 1.0 : stream_ld<NT>(&part_vals[i]);
