@@ -697,3 +697,45 @@ class ExecutionEngineEdgeCaseTests:
         assert sorted(x[0] for x in r2.as_array()) == [0, 2, 4]
         r3 = e.filter(o, ~ff.like(col("s"), "app%"))
         assert sorted(x[0] for x in r3.as_array()) == [1, 4]
+
+
+    def test_any_column_name(self):
+        """Underscored/numbered column names flow through filter/select."""
+        e = self.engine
+        o = e.to_df(ArrayDataFrame([[1, "x"], [2, "y"]], "a_b:long,c1:str"))
+        r = e.filter(o, col("a_b") > 1)
+        assert r.as_array() == [[2, "y"]]
+
+
+class BagTestSuite:
+    """Bag conformance (reference parity: ``fugue_test/bag_suite.py``)."""
+
+    @classmethod
+    def make_bag(cls, data):  # pragma: no cover
+        from fugue_amd.bag.array_bag import ArrayBag
+
+        return ArrayBag(data)
+
+    def test_init_basic(self):
+        b = self.make_bag([2, 1, "a", None])
+        assert b.count() == 4
+        assert not b.empty
+        assert self.make_bag([]).empty
+
+    def test_peek(self):
+        b = self.make_bag([5])
+        assert b.peek() == 5
+
+    def test_as_array_special_values(self):
+        data = [1, "x", None, 2.5, b"bytes", [1, 2]]
+        b = self.make_bag(data)
+        assert list(b.as_array()) == data
+        assert list(b.as_array_iterable()) == data
+
+    def test_head(self):
+        b = self.make_bag(list(range(10)))
+        arr = b.as_array()
+        assert arr[:3] == [0, 1, 2]
+
+    def test_show(self):
+        self.make_bag([1, 2]).show()

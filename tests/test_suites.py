@@ -87,3 +87,10 @@ class TestHipCpuEdgeCases(ExecutionEngineEdgeCaseTests):
         from fugue_amd.hip.execution_engine import HipExecutionEngine
 
         return HipExecutionEngine()
+
+
+from fugue_amd.testing.suites import BagTestSuite
+
+
+class TestArrayBag(BagTestSuite):
+    pass

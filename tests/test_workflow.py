@@ -254,3 +254,60 @@ def test_auto_persist():
     # `a` is consumed twice; the auto-persist marks it with a weak
     # checkpoint so its transform runs once
     assert len(calls) == 1
+
+
+def test_module_composition():
+    from fugue_amd.workflow.module import module
+    from fugue_amd.workflow.workflow import WorkflowDataFrame
+
+    @module
+    def doubled(df: WorkflowDataFrame) -> WorkflowDataFrame:
+        return df.transform(_double_x, schema="*")
+
+    @module(as_method=True, name="tripled")
+    def _tripled(df: WorkflowDataFrame) -> WorkflowDataFrame:
+        # schema: *
+        def t3(pdf: pd.DataFrame) -> pd.DataFrame:
+            pdf["x"] = pdf["x"] * 3
+            return pdf
+
+        return df.transform(t3, schema="*")
+
+    dag = FugueWorkflow()
+    a = dag.df([[1]], "x:long")
+    doubled(a).yield_dataframe_as("d")
+    a.tripled().yield_dataframe_as("t")
+    res = dag.run()
+    assert res["d"].result.as_array() == [[2]]
+    assert res["t"].result.as_array() == [[3]]
+
+
+def test_save_and_use():
+    import tempfile
+
+    with tempfile.TemporaryDirectory() as tmp:
+        path = os.path.join(tmp, "out.parquet")
+        dag = FugueWorkflow()
+        a = dag.df([[1], [2]], "x:long")
+        b = a.save_and_use(path)
+        b.yield_dataframe_as("r")
+        dag.run()
+        assert os.path.exists(path)
+        # the saved file loads back with the same data
+        dag2 = FugueWorkflow()
+        dag2.load(path).yield_dataframe_as("r")
+        res2 = dag2.run()
+        assert sorted(r[0] for r in res2["r"].result.as_array()) == [1, 2]
+
+
+def test_transform_iterable_dfs_output():
+    from typing import Iterable
+
+    pdf = pd.DataFrame(dict(x=[1, 2, 3]))
+
+    def splitter(df: pd.DataFrame) -> Iterable[pd.DataFrame]:
+        for _, row in df.iterrows():
+            yield pd.DataFrame([row])
+
+    res = transform(pdf, splitter, schema="*")
+    assert sorted(res["x"].tolist()) == [1, 2, 3]
