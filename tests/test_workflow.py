@@ -311,3 +311,38 @@ def test_transform_iterable_dfs_output():
 
     res = transform(pdf, splitter, schema="*")
     assert sorted(res["x"].tolist()) == [1, 2, 3]
+
+
+def test_extension_registry_aliases():
+    from fugue_amd.extensions import (
+        register_creator,
+        register_outputter,
+        register_processor,
+        register_transformer,
+    )
+
+    # schema: x:long
+    def make_one(ctx=None) -> pd.DataFrame:
+        return pd.DataFrame(dict(x=[7]))
+
+    def double_proc(df: pd.DataFrame) -> pd.DataFrame:
+        df = df.copy()
+        df["x"] = df["x"] * 2
+        return df
+
+    seen = []
+
+    def sink(df: pd.DataFrame) -> None:
+        seen.append(df["x"].tolist())
+
+    register_creator("mk1", make_one)
+    register_transformer("dbl", double_proc)
+    register_outputter("snk", sink)
+
+    dag = FugueWorkflow()
+    a = dag.create("mk1")
+    b = a.transform("dbl", schema="*")
+    b.output("snk")
+    b.yield_dataframe_as("r")
+    dag.run()
+    assert seen == [[14]]
