@@ -492,6 +492,40 @@ class DataFrameTestSuite:
         with pytest.raises(FugueDataFrameEmptyError):
             df.peek_array()
 
+    def test_rename_invalid(self):
+        df = self.make_df([[1]], "x:long")
+        with pytest.raises(Exception):
+            df.rename({"nope": "y"})
+
+    def test_alter_columns_invalid(self):
+        df = self.make_df([[1]], "x:long")
+        with pytest.raises(Exception):
+            df.alter_columns("nope:str")
+
+    def test_show(self):
+        df = self.make_df([[1, "a"], [2, None]], "x:long,y:str")
+        df.show()
+        df.show(1, with_count=True, title="t")
+
+    supports_nested = True
+
+    def test_nested_types(self):
+        if not self.supports_nested:
+            pytest.skip("backend frame holds flat columns only")
+        df = self.make_df(
+            [[[1, 2], dict(a=1)], [[3], dict(a=2)]],
+            "x:[long],y:{a:long}",
+        )
+        arr = df.as_array()
+        assert list(arr[0][0]) == [1, 2]
+
+    def test_binary_type(self):
+        if not self.supports_nested:
+            pytest.skip("backend frame holds flat columns only")
+        df = self.make_df([[b"ab"], [b"c"]], "x:bytes")
+        got = df.as_array()
+        assert bytes(got[0][0]) == b"ab" and bytes(got[1][0]) == b"c"
+
 
 class _SelectTopHelper:
     @staticmethod

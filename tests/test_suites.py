@@ -63,6 +63,7 @@ class TestArrowDataFrame(DataFrameTestSuite):
 
 
 class TestHipDataFrameCpu(DataFrameTestSuite):
+    supports_nested = False  # device columns are flat (validity + data)
     @classmethod
     def make_df(cls, data: Any, schema: Any):
         from fugue_amd.hip.frame import HipDataFrame

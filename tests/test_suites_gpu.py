@@ -35,6 +35,8 @@ class TestHipGpuBuiltIn(BuiltInWorkflowTestSuite):
 
 
 class TestHipDataFrameGpu(DataFrameTestSuite):
+    supports_nested = False  # device columns are flat (validity + data)
+
     @classmethod
     def make_df(cls, data: Any, schema: Any):
         from fugue_amd.hip.frame import HipDataFrame
