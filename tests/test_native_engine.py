@@ -140,3 +140,42 @@ def test_fa_eager():
         pd.DataFrame(dict(k=[2, 3], y=[5, 6])),
     )
     assert len(j) == 1
+
+
+def test_engine_api_eager_ops(tmp_path):
+    """fa.* eager API parity sweep (reference fugue_test
+    execution_suite.test_engine_api)."""
+    import os
+
+    import fugue_amd.api as fa
+    from fugue_amd.column import functions as f
+    from fugue_amd.column.expressions import col, lit
+
+    a = pd.DataFrame(dict(k=[1, 1, 2], v=[1.0, 2.0, 3.0]))
+    b = pd.DataFrame(dict(k=[1, 3], w=[10.0, 30.0]))
+
+    assert fa.get_current_parallelism() >= 1
+    r = fa.select(a, col("k"), (col("v") * 2).alias("v2"), engine="native")
+    assert r["v2"].tolist() == [2.0, 4.0, 6.0]
+    r = fa.filter(a, col("v") > 1.5, engine="native")
+    assert r["v"].tolist() == [2.0, 3.0]
+    r = fa.assign(a, z=col("v") + 1, engine="native")
+    assert r["z"].tolist() == [2.0, 3.0, 4.0]
+    r = fa.aggregate(a, partition_by="k", s=f.sum(col("v")), engine="native")
+    assert sorted(r["s"].tolist()) == [3.0, 3.0]
+    r = fa.inner_join(a, b, engine="native")
+    assert r["w"].tolist() == [10.0, 10.0]
+    r = fa.union(a[["k"]], b[["k"]], distinct=True, engine="native")
+    assert sorted(r["k"].tolist()) == [1, 2, 3]
+    r = fa.subtract(a[["k"]], b[["k"]], engine="native")
+    assert r["k"].tolist() == [2]
+    r = fa.intersect(a[["k"]], b[["k"]], engine="native")
+    assert r["k"].tolist() == [1]
+    r = fa.distinct(a[["k"]], engine="native")
+    assert sorted(r["k"].tolist()) == [1, 2]
+    r = fa.take(a, 1, presort="v desc", engine="native")
+    assert r["v"].tolist() == [3.0]
+    p = os.path.join(str(tmp_path), "t.parquet")
+    fa.save(a, p, engine="native")
+    back = fa.load(p, engine="native")
+    assert len(back) == 3

@@ -346,3 +346,21 @@ def test_extension_registry_aliases():
     b.yield_dataframe_as("r")
     dag.run()
     assert seen == [[14]]
+
+
+def test_cotransformer_decorator():
+    from fugue_amd.extensions import cotransformer
+
+    @cotransformer("k:long,n1:long,n2:long")
+    def merge(df1: pd.DataFrame, df2: pd.DataFrame) -> pd.DataFrame:
+        return pd.DataFrame(
+            dict(k=[df1["k"].iloc[0]], n1=[len(df1)], n2=[len(df2)])
+        )
+
+    dag = FugueWorkflow()
+    a = dag.df([[1, "a"], [1, "b"], [2, "c"]], "k:long,x:str")
+    b = dag.df([[1, 1.0], [3, 2.0]], "k:long,y:double")
+    z = dag.zip(a, b)  # inner: only k=1 survives
+    z.transform(merge).yield_dataframe_as("r")
+    res = dag.run()
+    assert res["r"].result.as_array() == [[1, 2, 1]]
