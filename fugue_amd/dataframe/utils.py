@@ -142,20 +142,49 @@ def _rows_eq(r1: List[Any], r2: List[Any], digits: int) -> bool:
     return True
 
 
-def serialize_df(df: Optional[DataFrame]) -> Optional[bytes]:
-    """Serialize a local frame to Arrow IPC bytes."""
+_FILE_BLOB_MARKER = b"\x00FUGUE-BLOB-FILE\x00"
+
+
+def serialize_df(
+    df: Optional[DataFrame],
+    threshold: int = -1,
+    file_path_root: Optional[str] = None,
+) -> Optional[bytes]:
+    """Serialize a local frame to Arrow IPC bytes.  When ``threshold`` is
+    positive and the payload exceeds it, the bytes are written to a file
+    under ``file_path_root`` and a path marker is returned instead
+    (reference parity: ``to_file_threshold`` in
+    ``fugue/execution/execution_engine.py:968-979``)."""
     if df is None:
         return None
     table = df.as_arrow()
     sink = pa.BufferOutputStream()
     with pa.ipc.new_stream(sink, table.schema) as writer:
         writer.write_table(table)
-    return sink.getvalue().to_pybytes()
+    data = sink.getvalue().to_pybytes()
+    if threshold > 0 and len(data) > threshold:
+        import os
+        import tempfile
+        import uuid as _uuid
+
+        root = file_path_root or tempfile.gettempdir()
+        os.makedirs(root, exist_ok=True)
+        path = os.path.join(root, f"fugue-blob-{_uuid.uuid4().hex}.arrow")
+        with open(path, "wb") as fh:
+            fh.write(data)
+        return _FILE_BLOB_MARKER + path.encode("utf-8")
+    return data
 
 
 def deserialize_df(data: Optional[bytes]) -> Optional[LocalBoundedDataFrame]:
     if data is None:
         return None
+    if isinstance(data, (bytes, bytearray)) and bytes(data).startswith(
+        _FILE_BLOB_MARKER
+    ):
+        path = bytes(data)[len(_FILE_BLOB_MARKER):].decode("utf-8")
+        with open(path, "rb") as fh:
+            data = fh.read()
     with pa.ipc.open_stream(pa.BufferReader(data)) as reader:
         table = reader.read_all()
     return ArrowDataFrame(table)

@@ -595,6 +595,8 @@ class ExecutionEngine(FugueEngineBase):
                     partition_spec,
                     i,
                     pairs[i][0] if dfs.has_key else None,
+                    temp_path=temp_path,
+                    to_file_threshold=int(to_file_threshold),
                 )
             )
             schemas[pairs[i][0] if dfs.has_key else i] = pairs[i][1].schema
@@ -646,6 +648,8 @@ class ExecutionEngine(FugueEngineBase):
         partition_spec: PartitionSpec,
         df_no: int,
         df_name: Optional[str] = None,
+        temp_path: Optional[str] = None,
+        to_file_threshold: int = -1,
     ) -> DataFrame:
         on = [k for k in partition_spec.partition_by if k in df.schema]
         presort = [
@@ -659,7 +663,9 @@ class ExecutionEngine(FugueEngineBase):
             output_schema = (
                 partition_spec.get_key_schema(df.schema) + _FUGUE_SERIALIZED_BLOB_SCHEMA
             )
-        s = _PartitionSerializer(output_schema, df_no, df_name)
+        s = _PartitionSerializer(
+            output_schema, df_no, df_name, temp_path, to_file_threshold
+        )
         return self.map_engine.map_dataframe(df, s.run, output_schema, _spec)
 
     def __uuid__(self) -> str:
@@ -670,13 +676,24 @@ class ExecutionEngine(FugueEngineBase):
 
 
 class _PartitionSerializer:
-    def __init__(self, output_schema: Schema, no: int, name: Optional[str]):
+    def __init__(
+        self,
+        output_schema: Schema,
+        no: int,
+        name: Optional[str],
+        temp_path: Optional[str] = None,
+        to_file_threshold: int = -1,
+    ):
         self.output_schema = output_schema
         self.no = no
         self.name = name
+        self.temp_path = temp_path
+        self.to_file_threshold = to_file_threshold
 
     def run(self, cursor: PartitionCursor, df: LocalDataFrame) -> LocalDataFrame:
-        data = serialize_df(df)
+        data = serialize_df(
+            df, threshold=self.to_file_threshold, file_path_root=self.temp_path
+        )
         row = cursor.key_value_array + [data, self.no, self.name, 1]
         return ArrayDataFrame([row], self.output_schema)
 

@@ -364,3 +364,34 @@ def test_cotransformer_decorator():
     z.transform(merge).yield_dataframe_as("r")
     res = dag.run()
     assert res["r"].result.as_array() == [[1, 2, 1]]
+
+
+def test_zip_to_file_threshold(tmp_path):
+    from fugue_amd.execution import NativeExecutionEngine
+    from fugue_amd.dataframe.dataframes import DataFrames
+
+    e = NativeExecutionEngine()
+    a = e.to_df(pd.DataFrame(dict(k=[1] * 50, x=list(range(50)))))
+    b = e.to_df(pd.DataFrame(dict(k=[1] * 30, y=list(range(30)))))
+    z = e.zip(
+        DataFrames(a, b),
+        partition_spec=None,
+        temp_path=str(tmp_path),
+        to_file_threshold=64,  # tiny: force file-backed blobs
+    )
+    import os
+
+    blobs = [f for f in os.listdir(tmp_path) if f.startswith("fugue-blob-")]
+    assert len(blobs) > 0  # payloads spilled to files
+
+    def cm(cursor, dfs):
+        from fugue_amd.dataframe.array_dataframe import ArrayDataFrame
+
+        return ArrayDataFrame(
+            [[dfs[0].count(), dfs[1].count()]], "n1:long,n2:long"
+        )
+
+    from fugue_amd.collections.partition import PartitionSpec
+
+    res = e.comap(z, cm, "n1:long,n2:long", PartitionSpec())
+    assert res.as_array() == [[50, 30]]
