@@ -201,30 +201,35 @@ class FugueSQLParser:
             elif depth == 0 and t.kind == "PUNCT" and t.value == ";":
                 ts.next()
                 break
-            elif depth == 0 and t.kind == "NAME":
+            elif t.kind == "NAME":
                 up = t.upper
-                if up in _POST_CLAUSES:
-                    break
-                if up in (
-                    "TRANSFORM", "PROCESS", "OUTPUT", "PRINT", "SAVE",
-                    "TAKE", "ZIP", "OUTTRANSFORM", "LOAD", "CREATE",
-                ):
-                    break
-                nxt = ts.peek(1)
-                if (
-                    up not in _STMT_START
-                    and nxt is not None
-                    and nxt.kind == "OP"
-                    and nxt.value == "="
-                    and self._starts_line(t)
-                ):
-                    break  # next statement: NAME = ...
-                if up == "SELECT" and t.pos != start:
-                    prev = ts.tokens[ts.pos - 1] if ts.pos > 0 else None
-                    if prev is None or prev.upper not in (
-                        "UNION", "ALL", "INTERSECT", "EXCEPT", "(",
+                if depth == 0:
+                    if up in _POST_CLAUSES:
+                        break
+                    if up in (
+                        "TRANSFORM", "PROCESS", "OUTPUT", "PRINT", "SAVE",
+                        "TAKE", "ZIP", "OUTTRANSFORM", "LOAD", "CREATE",
                     ):
-                        break  # a new top-level SELECT statement
+                        break
+                    nxt = ts.peek(1)
+                    if (
+                        up not in _STMT_START
+                        and nxt is not None
+                        and nxt.kind == "OP"
+                        and nxt.value == "="
+                        and self._starts_line(t)
+                    ):
+                        break  # next statement: NAME = ...
+                    if up == "SELECT" and t.pos != start:
+                        prev = ts.tokens[ts.pos - 1] if ts.pos > 0 else None
+                        if prev is None or prev.upper not in (
+                            "UNION", "ALL", "INTERSECT", "EXCEPT", "(",
+                        ):
+                            break  # a new top-level SELECT statement
+                        from_context = False
+                # table-ref replacement tracks FROM/JOIN context at ANY
+                # depth so subqueries resolve captured frames too
+                if up == "SELECT":
                     from_context = False
                 elif up in ("FROM", "JOIN"):
                     from_context = True

@@ -190,3 +190,32 @@ def test_sql_caller_var_capture():
 def test_sql_errors():
     with pytest.raises((FugueSQLSyntaxError, FugueSQLError)):
         fa.fugue_sql("NONSENSE STATEMENT HERE")
+
+
+def test_subquery_in_from():
+    import fugue_amd.api as fa
+
+    df = pd.DataFrame(dict(k=[1, 1, 2, 2, 3], v=[1.0, 2.0, 3.0, 4.0, 5.0]))
+    res = fa.fugue_sql(
+        """
+        SELECT k, s FROM (
+          SELECT k, SUM(v) AS s FROM df GROUP BY k HAVING SUM(v) > 3
+        ) WHERE s < 8
+        UNION ALL
+        SELECT k, v AS s FROM df WHERE k = 3
+        """,
+        df=df,
+    )
+    got = sorted((int(a), float(b)) for a, b in res.values.tolist())
+    assert got == [(2, 7.0), (3, 5.0), (3, 5.0)]
+
+
+def test_case_when_in_fsql():
+    import fugue_amd.api as fa
+
+    df = pd.DataFrame(dict(k=[1, 2], v=[1.0, 3.0]))
+    res = fa.fugue_sql(
+        "SELECT k, CASE WHEN v > 2.5 THEN 'big' ELSE 'small' END AS c FROM df",
+        df=df,
+    )
+    assert res["c"].tolist() == ["small", "big"]
