@@ -196,6 +196,156 @@ class FuncCall(Expr):
         raise NotImplementedError(name)
 
 
+class WindowFunc(Expr):
+    """``func(args) OVER (PARTITION BY ... ORDER BY ...)``.
+
+    Implemented in the executor itself (the reference delegates window
+    functions to its SQL backends, e.g. duckdb/spark).  With ORDER BY,
+    aggregate windows are cumulative (ROWS UNBOUNDED PRECEDING..CURRENT
+    ROW); without, they span the whole partition."""
+
+    def __init__(
+        self,
+        func: "FuncCall",
+        partition_by: List[Expr],
+        order_by: List["OrderItem"],
+    ):
+        self.func = func
+        self.partition_by = partition_by
+        self.order_by = order_by
+
+    @property
+    def is_agg(self) -> bool:
+        return False
+
+    def default_name(self) -> str:
+        return self.func.name.lower()
+
+    def eval(self, ctx: "Scope") -> Any:
+        import numpy as np
+
+        frame = ctx.frame
+        n = len(frame)
+        if n == 0:
+            return pd.Series([], dtype="float64")
+        work = pd.DataFrame(index=pd.RangeIndex(n))
+        pcols: List[str] = []
+        for i, e in enumerate(self.partition_by):
+            v = e.eval(ctx)
+            work[f"__p{i}"] = (
+                v.reset_index(drop=True)
+                if isinstance(v, pd.Series)
+                else pd.Series([v] * n)
+            )
+            pcols.append(f"__p{i}")
+        if not pcols:
+            work["__p0"] = 0
+            pcols = ["__p0"]
+        ocols: List[str] = []
+        oasc: List[bool] = []
+        for i, o in enumerate(self.order_by):
+            v = o.expr.eval(ctx)
+            work[f"__o{i}"] = (
+                v.reset_index(drop=True)
+                if isinstance(v, pd.Series)
+                else pd.Series([v] * n)
+            )
+            ocols.append(f"__o{i}")
+            oasc.append(o.asc)
+        fname = self.func.name
+        args = self.func.args
+        val: Optional[pd.Series] = None
+        if len(args) > 0 and not isinstance(args[0], Star):
+            v = args[0].eval(ctx)
+            val = (
+                v.reset_index(drop=True)
+                if isinstance(v, pd.Series)
+                else pd.Series([v] * n)
+            )
+            work["__v"] = val
+        sorted_w = work.sort_values(
+            pcols + ocols,
+            ascending=[True] * len(pcols) + oasc,
+            kind="stable",
+        )
+        g = sorted_w.groupby(pcols, sort=False, dropna=False)
+        rn = g.cumcount() + 1
+        if fname == "ROW_NUMBER":
+            res = rn
+        elif fname in ("RANK", "DENSE_RANK"):
+            if not ocols:
+                raise SyntaxError(f"{fname} requires ORDER BY")
+            prev = sorted_w[pcols + ocols].shift(1)
+            same_part = (
+                (sorted_w[pcols] == prev[pcols]) | (sorted_w[pcols].isna() & prev[pcols].isna())
+            ).all(axis=1)
+            same_ord = (
+                (sorted_w[ocols].values == prev[ocols].values)
+                | (sorted_w[ocols].isna().values & prev[ocols].isna().values)
+            ).all(axis=1)
+            new_peer = ~(same_part.to_numpy() & same_ord)
+            if fname == "RANK":
+                res = rn.where(pd.Series(new_peer, index=rn.index))
+                res = res.groupby(
+                    [sorted_w[c] for c in pcols], sort=False, dropna=False
+                ).ffill()
+            else:
+                res = (
+                    pd.Series(new_peer.astype("int64"), index=sorted_w.index)
+                    .groupby(
+                        [sorted_w[c] for c in pcols], sort=False, dropna=False
+                    )
+                    .cumsum()
+                )
+        elif fname in ("LAG", "LEAD"):
+            if val is None:
+                raise SyntaxError(f"{fname} requires an argument")
+            k = 1
+            default = None
+            if len(args) > 1 and isinstance(args[1], Lit):
+                k = int(args[1].value)
+            if len(args) > 2 and isinstance(args[2], Lit):
+                default = args[2].value
+            shift = k if fname == "LAG" else -k
+            res = g["__v"].shift(shift)
+            if default is not None:
+                res = res.fillna(default)
+        elif fname in ("FIRST_VALUE", "LAST_VALUE", "FIRST", "LAST"):
+            if val is None:
+                raise SyntaxError(f"{fname} requires an argument")
+            which = "first" if fname in ("FIRST_VALUE", "FIRST") else "last"
+            res = g["__v"].transform(which)
+        elif fname in ("SUM", "MIN", "MAX", "AVG", "MEAN", "COUNT"):
+            if fname == "COUNT" and val is None:
+                res = g.cumcount() + 1 if ocols else g["__p0" if "__p0" in sorted_w else pcols[0]].transform("size")
+            else:
+                if val is None:
+                    raise SyntaxError(f"{fname} requires an argument")
+                if ocols:
+                    gv = g["__v"]
+                    if fname == "SUM":
+                        res = gv.cumsum()
+                    elif fname == "MIN":
+                        res = gv.cummin()
+                    elif fname == "MAX":
+                        res = gv.cummax()
+                    elif fname in ("AVG", "MEAN"):
+                        res = gv.expanding().mean().reset_index(
+                            level=list(range(len(pcols))), drop=True
+                        )
+                    else:  # COUNT(expr)
+                        res = gv.expanding().count().reset_index(
+                            level=list(range(len(pcols))), drop=True
+                        )
+                else:
+                    m = {"SUM": "sum", "MIN": "min", "MAX": "max",
+                         "AVG": "mean", "MEAN": "mean", "COUNT": "count"}
+                    res = g["__v"].transform(m[fname])
+        else:
+            raise NotImplementedError(f"window function {fname}")
+        return pd.Series(res, index=sorted_w.index).sort_index()
+
+
 class Case(Expr):
     def __init__(self, whens: List[Tuple[Expr, Expr]], else_: Optional[Expr]):
         self.whens = whens
@@ -780,7 +930,39 @@ def _parse_primary(ts: TokenStream) -> Expr:
                     if not ts.take_punct(","):
                         break
             ts.expect_punct(")")
-            return FuncCall(name, args, distinct=distinct)
+            fc = FuncCall(name, args, distinct=distinct)
+            if ts.take_kw("OVER"):
+                ts.expect_punct("(")
+                partition_by: List[Expr] = []
+                order_by: List[OrderItem] = []
+                if ts.take_kw("PARTITION"):
+                    ts.expect_kw("BY")
+                    while True:
+                        partition_by.append(_parse_expr(ts))
+                        if not ts.take_punct(","):
+                            break
+                if ts.take_kw("ORDER"):
+                    ts.expect_kw("BY")
+                    while True:
+                        e = _parse_expr(ts)
+                        asc = True
+                        if ts.take_kw("DESC"):
+                            asc = False
+                        elif ts.take_kw("ASC"):
+                            pass
+                        order_by.append(OrderItem(e, asc))
+                        if not ts.take_punct(","):
+                            break
+                nt = ts.peek()
+                if nt is not None and nt.kind == "NAME" and nt.upper in (
+                    "ROWS", "RANGE", "GROUPS",
+                ):
+                    raise NotImplementedError(
+                        "explicit window frame specs are not supported"
+                    )
+                ts.expect_punct(")")
+                return WindowFunc(fc, partition_by, order_by)
+            return fc
         # qualified / unqualified column, or alias.*
         name = ts.next().value
         if ts.match_punct(".") and ts.peek(1) is not None:

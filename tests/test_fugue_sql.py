@@ -219,3 +219,52 @@ def test_case_when_in_fsql():
         df=df,
     )
     assert res["c"].tolist() == ["small", "big"]
+
+
+def test_window_functions():
+    from fugue_amd.sql.executor import run_sql_on_pandas
+
+    df = pd.DataFrame(dict(k=["a", "a", "a", "b", "b"], v=[3.0, 1.0, 2.0, 5.0, 4.0]))
+    r = run_sql_on_pandas(
+        "SELECT k, v, ROW_NUMBER() OVER (PARTITION BY k ORDER BY v) AS rn, "
+        "SUM(v) OVER (PARTITION BY k ORDER BY v) AS cs, "
+        "SUM(v) OVER (PARTITION BY k) AS ts, "
+        "LAG(v) OVER (PARTITION BY k ORDER BY v) AS pv, "
+        "RANK() OVER (ORDER BY v DESC) AS r FROM a",
+        dict(a=df),
+    )[0]
+    assert r["rn"].tolist() == [3, 1, 2, 2, 1]
+    assert r["cs"].tolist() == [6.0, 1.0, 3.0, 9.0, 4.0]
+    assert r["ts"].tolist() == [6.0, 6.0, 6.0, 9.0, 9.0]
+    assert r["r"].tolist() == [3, 5, 4, 1, 2]
+    assert pd.isna(r["pv"].iloc[1]) and r["pv"].iloc[2] == 1.0
+
+
+def test_window_dedup_pattern():
+    import fugue_amd.api as fa
+
+    df = pd.DataFrame(dict(k=["a", "a", "b"], v=[1.0, 3.0, 5.0]))
+    r = fa.fugue_sql(
+        "SELECT k, v FROM ("
+        "  SELECT k, v, ROW_NUMBER() OVER (PARTITION BY k ORDER BY v DESC) AS rn"
+        "  FROM df) WHERE rn = 1",
+        df=df,
+    )
+    assert sorted(map(tuple, r.values.tolist())) == [("a", 3.0), ("b", 5.0)]
+
+
+def test_window_rank_dense_and_values():
+    from fugue_amd.sql.executor import run_sql_on_pandas
+
+    df = pd.DataFrame(dict(k=["a", "a", "a", "a"], v=[1.0, 2.0, 2.0, 3.0]))
+    r = run_sql_on_pandas(
+        "SELECT v, RANK() OVER (ORDER BY v) AS r, "
+        "DENSE_RANK() OVER (ORDER BY v) AS d, "
+        "FIRST_VALUE(v) OVER (PARTITION BY k ORDER BY v) AS fv, "
+        "LAST_VALUE(v) OVER (PARTITION BY k) AS lv FROM a",
+        dict(a=df),
+    )[0]
+    assert r["r"].tolist() == [1, 2, 2, 4]
+    assert r["d"].tolist() == [1, 2, 2, 3]
+    assert r["fv"].tolist() == [1.0] * 4
+    assert r["lv"].tolist() == [3.0] * 4
