@@ -196,6 +196,51 @@ class FuncCall(Expr):
         raise NotImplementedError(name)
 
 
+class SubqueryExpr(Expr):
+    """Uncorrelated scalar subquery in expression position."""
+
+    def __init__(self, stmt: "SelectStmt"):
+        self.stmt = stmt
+
+    @property
+    def is_agg(self) -> bool:
+        return False
+
+    def eval(self, ctx: "Scope") -> Any:
+        res = _execute_select(self.stmt, ctx.tables)
+        if res.shape[1] != 1:
+            raise SyntaxError("scalar subquery must return one column")
+        if len(res) == 0:
+            return None
+        if len(res) > 1:
+            raise ValueError("scalar subquery returned more than one row")
+        return res.iloc[0, 0]
+
+
+class InSubquery(Expr):
+    """``expr [NOT] IN (SELECT ...)`` (uncorrelated)."""
+
+    def __init__(self, expr: Expr, stmt: "SelectStmt", negate: bool):
+        self.expr = expr
+        self.stmt = stmt
+        self.negate = negate
+
+    @property
+    def is_agg(self) -> bool:
+        return self.expr.is_agg
+
+    def eval(self, ctx: "Scope") -> Any:
+        res = _execute_select(self.stmt, ctx.tables)
+        if res.shape[1] != 1:
+            raise SyntaxError("IN subquery must return one column")
+        values = set(res.iloc[:, 0].dropna().tolist())
+        v = self.expr.eval(ctx)
+        s = v if isinstance(v, pd.Series) else pd.Series([v] * len(ctx.frame))
+        m = s.isin(values)
+        m = m.mask(s.isna())  # NULL IN (...) -> NULL
+        return (~m.astype("boolean")) if self.negate else m.astype("boolean")
+
+
 class WindowFunc(Expr):
     """``func(args) OVER (PARTITION BY ... ORDER BY ...)``.
 
@@ -444,12 +489,14 @@ class Cast(Expr):
 # --------------------------------------------------------------------- #
 class Scope:
     """Column resolution over an internal frame whose columns are
-    ``alias␟name`` pairs."""
+    ``alias␟name`` pairs.  ``tables`` carries the statement's source
+    tables so uncorrelated subquery expressions can execute."""
 
     SEP = "␟"
 
-    def __init__(self, frame: pd.DataFrame):
+    def __init__(self, frame: pd.DataFrame, tables: Optional[Dict[str, pd.DataFrame]] = None):
         self.frame = frame
+        self.tables = tables or {}
 
     def resolve(self, name: str, qualifier: Optional[str]) -> pd.Series:
         if qualifier is not None:
@@ -801,6 +848,10 @@ def _parse_comparison(ts: TokenStream) -> Expr:
         negate = True
     if ts.take_kw("IN"):
         ts.expect_punct("(")
+        if ts.match_kw("SELECT"):
+            sub = _parse_select_stmt(ts)
+            ts.expect_punct(")")
+            return InSubquery(left, sub, negate)
         vals: List[Expr] = []
         while True:
             vals.append(_parse_expr(ts))
@@ -867,6 +918,10 @@ def _parse_primary(ts: TokenStream) -> Expr:
         ts.next()
         return Lit(t.value[1:-1].replace("''", "'"))
     if ts.take_punct("("):
+        if ts.match_kw("SELECT"):
+            sub = _parse_select_stmt(ts)
+            ts.expect_punct(")")
+            return SubqueryExpr(sub)
         e = _parse_expr(ts)
         ts.expect_punct(")")
         return e
@@ -1144,12 +1199,12 @@ def _execute_core(stmt: SelectStmt, tables: Dict[str, pd.DataFrame]) -> pd.DataF
         frame = _resolve_from(stmt.from_item, tables)
         for j in stmt.joins:
             frame = _apply_join(frame, j, tables)
-    scope = Scope(frame)
+    scope = Scope(frame, tables)
     # WHERE
     if stmt.where is not None:
         mask = _as_bool(_as_series(stmt.where.eval(scope), scope))
         frame = frame[mask.to_numpy(dtype=bool)].reset_index(drop=True)
-        scope = Scope(frame)
+        scope = Scope(frame, tables)
     has_agg = any(c[0].is_agg for c in stmt.columns) or len(stmt.group_by) > 0
     if not has_agg:
         out: Dict[str, Any] = {}

@@ -268,3 +268,22 @@ def test_window_rank_dense_and_values():
     assert r["d"].tolist() == [1, 2, 2, 3]
     assert r["fv"].tolist() == [1.0] * 4
     assert r["lv"].tolist() == [3.0] * 4
+
+
+def test_in_and_scalar_subqueries():
+    from fugue_amd.sql.executor import run_sql_on_pandas
+
+    df = pd.DataFrame(dict(k=[1, 2, 3], v=[1.0, 2.0, 3.0]))
+    d2 = pd.DataFrame(dict(k=[2, 3]))
+    r = run_sql_on_pandas(
+        "SELECT k FROM a WHERE k IN (SELECT k FROM b)", dict(a=df, b=d2)
+    )[0]
+    assert r["k"].tolist() == [2, 3]
+    r = run_sql_on_pandas(
+        "SELECT k FROM a WHERE k NOT IN (SELECT k FROM b)", dict(a=df, b=d2)
+    )[0]
+    assert r["k"].tolist() == [1]
+    r = run_sql_on_pandas(
+        "SELECT k FROM a WHERE v > (SELECT AVG(v) FROM a)", dict(a=df)
+    )[0]
+    assert r["k"].tolist() == [3]
