@@ -241,6 +241,21 @@ class InSubquery(Expr):
         return (~m.astype("boolean")) if self.negate else m.astype("boolean")
 
 
+class ExistsSubquery(Expr):
+    """``EXISTS (SELECT ...)`` (uncorrelated)."""
+
+    def __init__(self, stmt: "SelectStmt"):
+        self.stmt = stmt
+
+    @property
+    def is_agg(self) -> bool:
+        return False
+
+    def eval(self, ctx: "Scope") -> Any:
+        res = _execute_select(self.stmt, ctx.tables)
+        return len(res) > 0
+
+
 class WindowFunc(Expr):
     """``func(args) OVER (PARTITION BY ... ORDER BY ...)``.
 
@@ -964,6 +979,12 @@ def _parse_primary(ts: TokenStream) -> Expr:
             ts.next()
             s = ts.next().value[1:-1]
             return Lit(pd.Timestamp(s))
+        if up == "EXISTS" and ts.peek(1) is not None and ts.peek(1).value == "(":
+            ts.next()
+            ts.expect_punct("(")
+            sub = _parse_select_stmt(ts)
+            ts.expect_punct(")")
+            return ExistsSubquery(sub)
         if up == "NULL":
             ts.next()
             return Lit(None)
