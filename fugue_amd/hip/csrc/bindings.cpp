@@ -32,6 +32,8 @@ void launch_join_build(const int64_t*, int64_t, int32_t*, int32_t*, int64_t,
                        hipStream_t);
 void launch_gb_part_hist(const int64_t*, int64_t, int, int64_t*, int,
                          int64_t*, hipStream_t);
+void launch_reduce_cols(const double*, const bool*, const int32_t*, int,
+                        int64_t, double*, int64_t*, hipStream_t);
 void launch_gb_part_scatter(const int64_t*, const double*, int, int64_t, int,
                             int64_t*, int64_t*, double*, int, hipStream_t);
 void launch_gb_aggregate_part(const int64_t*, const double*, const int32_t*,
@@ -429,6 +431,33 @@ std::vector<at::Tensor> compact_columns(at::Tensor mask,
   return outs;
 }
 
+std::vector<at::Tensor> reduce_columns(at::Tensor vals,
+                                       c10::optional<at::Tensor> valids,
+                                       at::Tensor ops) {
+  check_gpu(vals, "vals");
+  int n_aggs = (int)vals.size(0);
+  int64_t n = vals.size(1);
+  auto out = at::zeros({n_aggs}, vals.options());
+  auto ops_cpu = ops.cpu();
+  for (int a = 0; a < n_aggs; ++a) {
+    int op = (int)ops_cpu[a].item<int32_t>();
+    if (op == 1) out[a] = std::numeric_limits<double>::infinity();
+    if (op == 2) out[a] = -std::numeric_limits<double>::infinity();
+  }
+  auto cnt = at::zeros({n_aggs}, vals.options().dtype(at::kLong));
+  const bool* vptr = nullptr;
+  if (valids.has_value()) {
+    vptr = valids.value().data_ptr<bool>();
+  }
+  if (n > 0) {
+    launch_reduce_cols(vals.data_ptr<double>(), vptr,
+                       ops.data_ptr<int32_t>(), n_aggs, n,
+                       out.data_ptr<double>(), cnt.data_ptr<int64_t>(),
+                       current_stream());
+  }
+  return {out, cnt};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hash_column", &hash_column,
         "combine a column into the running row hash");
@@ -449,6 +478,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused masked compaction of 8-byte columns");
   m.def("join_count", &join_count, "count matches per probe row");
   m.def("join_emit", &join_emit, "emit join pairs");
+  m.def("reduce_columns", &reduce_columns,
+        "global column reductions (MFMA-reduced sums)");
   m.def("join_pairs", &join_pairs,
         "total + chunked-reservation join pair emission");
   m.def("join_mark_build", &join_mark_build, "mark matched build rows");

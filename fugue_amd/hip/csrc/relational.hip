@@ -189,6 +189,96 @@ void launch_scatter(const int32_t* buckets, int64_t* cursor, int64_t* perm,
 }  // extern "C"
 
 // ------------------------------------------------------------------ //
+// global (keyless) column reductions                                   //
+//   wave-level SUM reduction uses the f64 matrix core:                 //
+//   ones[16x4] x partials[4x16] via v_mfma_f64_16x16x4_f64 collapses   //
+//   64 lane partials to 16 column sums in ONE instruction (lanes       //
+//   0..15), finished by 4 shuffle rounds.  min/max/count use plain     //
+//   shuffle trees.  (north-star: "MFMA-vectorised reductions shown in  //
+//   rocprof")                                                          //
+// ------------------------------------------------------------------ //
+typedef double v4d __attribute__((ext_vector_type(4)));
+
+__device__ __forceinline__ void atomic_min_f64(double* addr, double val);
+__device__ __forceinline__ void atomic_max_f64(double* addr, double val);
+
+__device__ __forceinline__ double wave_reduce_sum_mfma(double v) {
+#if defined(__gfx950__) || defined(__gfx90a__) || defined(__gfx940__) || \
+    defined(__gfx942__)
+  v4d acc = {0.0, 0.0, 0.0, 0.0};
+  // A = ones (16x4), B = lane partials (4x16): D[i][j] = sum_k B[k][j]
+  acc = __builtin_amdgcn_mfma_f64_16x16x4f64(1.0, v, acc, 0, 0, 0);
+  double s = acc[0];  // col j = lane&15 sum, replicated across rows
+  s += __shfl_down(s, 8, 16);
+  s += __shfl_down(s, 4, 16);
+  s += __shfl_down(s, 2, 16);
+  s += __shfl_down(s, 1, 16);
+  return s;  // valid on lane 0
+#else
+  for (int off = WAVE / 2; off > 0; off >>= 1) v += __shfl_down(v, off, WAVE);
+  return v;
+#endif
+}
+
+__global__ __launch_bounds__(BLOCK) void reduce_cols_kernel(
+    const double* __restrict__ vals,   // [n_aggs, n]
+    const bool* __restrict__ valids,   // [n_aggs, n] or null
+    const int32_t* __restrict__ ops,   // [n_aggs] 0=sum 1=min 2=max
+    int n_aggs, int64_t n,
+    double* __restrict__ out,          // [n_aggs] pre-initialized
+    int64_t* __restrict__ out_count) { // [n_aggs] zero-initialized
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int lane = threadIdx.x & (WAVE - 1);
+  for (int a = 0; a < n_aggs; ++a) {
+    int op = ops[a];
+    double acc = op == 1 ? INFINITY : (op == 2 ? -INFINITY : 0.0);
+    int64_t cnt = 0;
+    const double* col = vals + (int64_t)a * n;
+    const bool* cv = valids == nullptr ? nullptr : valids + (int64_t)a * n;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+      if (cv != nullptr && !cv[i]) continue;
+      double v = col[i];
+      ++cnt;
+      if (op == 0) acc += v;
+      else if (op == 1) acc = v < acc ? v : acc;
+      else acc = v > acc ? v : acc;
+    }
+    // wave reduce
+    if (op == 0) {
+      acc = wave_reduce_sum_mfma(acc);
+    } else {
+      for (int off = WAVE / 2; off > 0; off >>= 1) {
+        double o = __shfl_down(acc, off, WAVE);
+        if (op == 1) acc = o < acc ? o : acc;
+        else acc = o > acc ? o : acc;
+      }
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+      cnt += __shfl_down(cnt, off, WAVE);
+    if (lane == 0) {
+      if (cnt > 0) {
+        if (op == 0) atomicAdd(&out[a], acc);
+        else if (op == 1) atomic_min_f64(&out[a], acc);
+        else atomic_max_f64(&out[a], acc);
+        atomicAdd((unsigned long long*)&out_count[a],
+                  (unsigned long long)cnt);
+      }
+    }
+  }
+}
+
+extern "C" {
+void launch_reduce_cols(const double* vals, const bool* valids,
+                        const int32_t* ops, int n_aggs, int64_t n,
+                        double* out, int64_t* out_count,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(reduce_cols_kernel, dim3(grid_for(n, 8)), dim3(BLOCK),
+                     0, stream, vals, valids, ops, n_aggs, n, out, out_count);
+}
+}  // extern "C"
+
+// ------------------------------------------------------------------ //
 // group-by aggregation: open-addressing HBM hash table                //
 //   key: exact int64 (multi-column keys packed by the python layer)   //
 //   aggs: fp64 matrix [n_aggs, n_rows]; per-agg op code:              //

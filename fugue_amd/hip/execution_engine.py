@@ -1444,7 +1444,9 @@ class HipExecutionEngine(ExecutionEngine):
                 continue
             raise DeviceExprError("compound aggregate expression: fallback")
         if len(key_names) == 0:
-            raise DeviceExprError("global aggregate: fallback")
+            return self._device_global_aggregate(
+                d, plans, partials, derived, having, cols
+            )
         cd_plans = [p for p in plans if p[1] == "count_distinct"]
         if cd_plans:
             return self._device_aggregate_distinct(
@@ -1571,6 +1573,79 @@ class HipExecutionEngine(ExecutionEngine):
         if tp is None:
             raise DeviceExprError(f"no arrow type for {data.dtype}")
         return tp
+
+    def _device_global_aggregate(
+        self,
+        d: HipDataFrame,
+        plans: List[Tuple[str, str, Any]],
+        partials: List[Tuple[str, int, str]],
+        derived: Dict[str, DeviceColumn],
+        having: Optional[ColumnExpr],
+        cols: SelectColumns,
+    ) -> DataFrame:
+        """Keyless aggregation (``SELECT SUM(v) FROM t``): single-pass
+        column reductions (MFMA-reduced f64 sums, ``reduce_cols_kernel``),
+        allreduce-merged across ranks.  FIRST/LAST/DISTINCT shapes fall
+        back to the host path."""
+        for _n, kind, _i in plans:
+            if kind in ("firstlast", "count_distinct"):
+                raise DeviceExprError("global first/last/distinct: fallback")
+        if cols.is_distinct:
+            raise DeviceExprError("distinct aggregate: fallback")
+        if derived:
+            cols_map = dict(d.columns_map)
+            cols_map.update(derived)
+            fields_ext = list(d.schema.fields) + [
+                pa.field(n, pa.float64()) for n in derived
+            ]
+            d = HipDataFrame.from_columns(
+                cols_map, Schema(fields_ext), self._device
+            )
+        vals, counts = dops.global_aggregate(d, partials)
+        n_rows = d.count()
+        if self.is_distributed:
+            import torch.distributed as dist
+
+            n_rows = self._comm.allreduce_sum(n_rows)
+            for nm, op, tmp in partials:
+                red = (
+                    dist.ReduceOp.MIN
+                    if op == dops.AGG_MIN
+                    else dist.ReduceOp.MAX
+                    if op == dops.AGG_MAX
+                    else dist.ReduceOp.SUM
+                )
+                t = torch.tensor([vals[tmp]], dtype=torch.float64)
+                t = t.to(self._comm._comm_device(t))
+                dist.all_reduce(t, op=red)
+                vals[tmp] = float(t.cpu().item())
+                ct = torch.tensor([counts[tmp]], dtype=torch.int64)
+                ct = ct.to(self._comm._comm_device(ct))
+                dist.all_reduce(ct, op=dist.ReduceOp.SUM)
+                counts[tmp] = int(ct.cpu().item())
+        row: Dict[str, Any] = {}
+        for name, kind, info in plans:
+            if kind == "rowcount":
+                row[name] = n_rows
+            elif kind == "count":
+                row[name] = counts[info]
+            elif kind in ("sum", "min", "max"):
+                tmp, _c = info
+                row[name] = vals[tmp] if counts[tmp] > 0 else None
+            elif kind == "avg":
+                tmp_s, tmp_c = info
+                cnt = counts[tmp_s]
+                row[name] = (vals[tmp_s] / cnt) if cnt > 0 else None
+            else:
+                raise DeviceExprError(f"global plan {kind}: fallback")
+        res = pd.DataFrame([row])
+        if having is not None:
+            from fugue_amd.column.interpreter import eval_filter
+
+            res = eval_filter(res, having)
+        if self.is_distributed and self.rank != 0:
+            res = res.head(0)
+        return self.to_df(PandasDataFrame(res), shard_replicated=False)
 
     def _with_rowindex_cols(
         self, d: HipDataFrame, fl_specs: List[Tuple[str, str]]

@@ -507,6 +507,71 @@ class HashCollisionError(RuntimeError):
 _H2_SEED = 0x5851F42D4C957F2D
 
 
+def global_aggregate(
+    df: HipDataFrame,
+    aggs: List[Tuple[str, int, str]],  # (input col, op, output name)
+) -> Tuple[Dict[str, float], Dict[str, int]]:
+    """Keyless whole-column reductions: returns ({name: value},
+    {name: non-null count}).  SUM wave-reduction runs on the f64 matrix
+    core (``v_mfma_f64_16x16x4_f64`` — see ``reduce_cols_kernel``)."""
+    n = df.count()
+    out_vals: Dict[str, float] = {}
+    out_counts: Dict[str, int] = {}
+    if len(aggs) == 0:
+        return out_vals, out_counts
+    c0 = df.col(aggs[0][0])
+    if _is_cpu(c0.data if not getattr(c0, "is_string", False) else c0.offsets):
+        import pandas as pd
+
+        for cname, op, oname in aggs:
+            c = df.col(cname)
+            v = _agg_input(c, n).numpy().astype("float64")
+            if c.valid is not None:
+                v = np.where(c.valid.numpy(), v, np.nan)
+            sr = pd.Series(v)
+            cnt = int(sr.notna().sum())
+            if op == AGG_SUM:
+                val = float(sr.sum()) if cnt > 0 else 0.0
+            elif op == AGG_MIN:
+                val = float(sr.min()) if cnt > 0 else float("inf")
+            elif op == AGG_MAX:
+                val = float(sr.max()) if cnt > 0 else float("-inf")
+            else:  # AGG_COUNT
+                val = float(cnt)
+            out_vals[oname] = val
+            out_counts[oname] = cnt
+        return out_vals, out_counts
+    ext = get_ext()
+    device = torch.device(df.device)
+    n_aggs = len(aggs)
+    vals = torch.empty((n_aggs, max(n, 1)), dtype=torch.float64, device=device)
+    valids: Optional[torch.Tensor] = None
+    if any(df.col(c).valid is not None for c, _, _ in aggs):
+        valids = torch.ones((n_aggs, max(n, 1)), dtype=torch.bool, device=device)
+    kernel_ops = []
+    for i, (cname, op, _) in enumerate(aggs):
+        c = df.col(cname)
+        vals[i, :n] = _agg_input(c, n)
+        if valids is not None:
+            if c.valid is not None:
+                valids[i, :n] = c.valid
+            if n < vals.shape[1]:
+                valids[i, n:] = False
+        kernel_ops.append(AGG_SUM if op == AGG_COUNT else op)
+    if n == 0 and valids is None:
+        valids = torch.zeros((n_aggs, 1), dtype=torch.bool, device=device)
+    ops_t = torch.tensor(kernel_ops, dtype=torch.int32, device=device)
+    out, cnt = ext.reduce_columns(vals[:, :max(n, 1)], valids, ops_t)
+    out_h = out.cpu().tolist()
+    cnt_h = cnt.cpu().tolist()
+    for i, (_, op, oname) in enumerate(aggs):
+        out_counts[oname] = int(cnt_h[i])
+        out_vals[oname] = (
+            float(cnt_h[i]) if op == AGG_COUNT else float(out_h[i])
+        )
+    return out_vals, out_counts
+
+
 def groupby_aggregate_hashed(
     df: HipDataFrame,
     keys: List[str],

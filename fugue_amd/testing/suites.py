@@ -741,6 +741,29 @@ class ExecutionEngineEdgeCaseTests:
         assert r.as_array() == [[2, "y"]]
 
 
+    def test_global_aggregate(self):
+        e = self.engine
+        o = e.to_df(
+            ArrayDataFrame(
+                [[1.0, 1], [2.0, 2], [None, 3], [4.0, 4]], "v:double,w:long"
+            )
+        )
+        r = e.aggregate(
+            o,
+            None,
+            [
+                f.sum(col("v")).alias("s"),
+                f.min(col("v")).alias("mn"),
+                f.max(col("w")).alias("mx"),
+                f.avg(col("v")).alias("av"),
+                f.count(col("v")).alias("c"),
+            ],
+        )
+        row = r.as_array()[0]
+        assert row[0] == 7.0 and row[1] == 1.0 and row[2] == 4.0
+        assert abs(row[3] - 7.0 / 3) < 1e-9 and row[4] == 3
+
+
 class BagTestSuite:
     """Bag conformance (reference parity: ``fugue_test/bag_suite.py``)."""
 
@@ -773,3 +796,4 @@ class BagTestSuite:
 
     def test_show(self):
         self.make_bag([1, 2]).show()
+

@@ -450,3 +450,29 @@ def test_distributed_set_ops():
         assert got["union"] == sorted(a | b)
         assert got["sub"] == sorted(a - b)
         assert got["inter"] == sorted(a & b)
+
+
+def _global_agg_job(rank: int):
+    from fugue_amd.column import functions as f
+    from fugue_amd.column.expressions import col
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    pdf = pd.DataFrame(dict(v=np.arange(100, dtype="f8")))
+    d = e.to_df(pdf)  # sharded
+    r = e.aggregate(
+        d, None,
+        [f.sum(col("v")).alias("s"), f.count(col("v")).alias("c"),
+         f.min(col("v")).alias("mn"), f.avg(col("v")).alias("av")],
+    )
+    local = e._gather_all(r)
+    return local.as_array()
+
+
+def test_distributed_global_aggregate():
+    results = run_distributed(_global_agg_job, 29535)
+    for rank, rows in results.items():
+        assert len(rows) == 1
+        s, c, mn, av = rows[0]
+        assert s == sum(range(100)) and c == 100 and mn == 0.0
+        assert abs(av - 49.5) < 1e-9
