@@ -203,7 +203,7 @@ def main() -> None:
                     ms_per_step=elapsed / args.steps * 1000.0,
                     higher_is_better=True,
                     scaling="weak",
-                    vs_baseline=None,
+                    vs_baseline=round(rows_per_sec / 1.63e6, 1),
                     dtype="int64+fp64+str",
                     data="synthetic",
                     config=dict(
