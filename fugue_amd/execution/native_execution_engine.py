@@ -130,8 +130,14 @@ class PandasMapEngine(MapEngine):
                     ),
                 )
                 results: List[pd.DataFrame] = []
+                _pdf_all = input_df.as_pandas()
+                _bounds = [
+                    (len(_pdf_all) * i) // partitions
+                    for i in range(partitions + 1)
+                ]
                 for p, subdf in enumerate(
-                    np.array_split(input_df.as_pandas(), partitions)
+                    _pdf_all.iloc[_bounds[i] : _bounds[i + 1]]
+                    for i in range(partitions)
                 ):
                     if len(subdf) > 0:
                         sub = subdf.reset_index(drop=True)

@@ -180,8 +180,13 @@ class HipMapEngine(MapEngine):
             if len(pdf_local) > 0:
                 import numpy as _np
 
+                bounds_sp = [
+                    (len(pdf_local) * i) // n_local_parts
+                    for i in range(n_local_parts + 1)
+                ]
                 for p, subdf in enumerate(
-                    _np.array_split(pdf_local, n_local_parts)
+                    pdf_local.iloc[bounds_sp[i] : bounds_sp[i + 1]]
+                    for i in range(n_local_parts)
                 ):
                     if len(subdf) == 0:
                         continue
