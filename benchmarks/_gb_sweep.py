@@ -53,9 +53,16 @@ def main() -> None:
     assert torch.equal(kk0[s0], kk2[s2]), "auto keys mismatch"
     assert torch.allclose(aa0[s0], aa2[s2]), "auto aggs mismatch"
     print("narrow correctness OK (explicit + auto)", flush=True)
-    for narrow in (0, 1, -1):
+    import os as _os
+
+    for ilp, narrow in ((2, 1), (1, 1), (2, 1), (1, 1), (2, 0), (1, 0)):
+        if ilp == 1:
+            _os.environ["FUGUE_GB_ILP"] = "1"
+        else:
+            _os.environ.pop("FUGUE_GB_ILP", None)
         t = run(0, 0, 0, narrow)
-        print(f"defaults narrow={narrow}: {t:.3f} ms", flush=True)
+        print(f"ilp={ilp} narrow={narrow}: {t:.3f} ms", flush=True)
+    _os.environ.pop("FUGUE_GB_ILP", None)
 
 
 if __name__ == "__main__":

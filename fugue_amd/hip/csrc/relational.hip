@@ -18,6 +18,7 @@
 
 #include <hip/hip_runtime.h>
 #include <algorithm>
+#include <cstdlib>
 #include <cstdint>
 
 #define WAVE 64
@@ -720,6 +721,35 @@ __device__ __forceinline__ int lds_key_cas(int32_t* addr, int32_t cmp,
                         (unsigned int)val);
 }
 
+// slow path of the per-row LDS upsert: first-slot read missed (empty or
+// different key) — probe/claim, spilling to the global table after 32
+// displacements
+template <typename KT>
+__device__ __forceinline__ void gb_lds_upsert(
+    KT key, double v, int slot, KT EMPTY, KT* __restrict__ lkeys,
+    double* __restrict__ laggs, int* __restrict__ lcount,
+    int64_t* __restrict__ tkeys, double* __restrict__ gaggs,
+    int64_t* __restrict__ gcount, int64_t tsize) {
+  bool in_lds = false;
+  for (int probe = 0; probe < 32; ++probe) {
+    KT cur = lkeys[slot];
+    if (cur == key) { in_lds = true; break; }
+    if (cur == EMPTY) {
+      KT prev = (KT)lds_key_cas(&lkeys[slot], EMPTY, key);
+      if (prev == EMPTY || prev == key) { in_lds = true; break; }
+    }
+    slot = (slot + 1) & (LDS_SLOTS_BIG - 1);
+  }
+  if (in_lds) {
+    atomicAdd(&lcount[slot], 1);
+    atomicAdd(&laggs[slot], v);
+  } else {
+    int64_t gslot = gb_probe_insert((int64_t)key, tkeys, tsize);
+    atomicAdd((unsigned long long*)&gcount[gslot], 1ULL);
+    atomicAdd(&gaggs[gslot], v);
+  }
+}
+
 template <bool NT, typename KT>
 __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
     const KT* __restrict__ part_keys,
@@ -745,30 +775,99 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
       laggs[i] = 0.0;
     }
     __syncthreads();
-    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
-      KT key = stream_ld<NT>(&part_keys[i]);
-      uint64_t h = mix64((uint64_t)(int64_t)key);
-      int slot = (int)(h & (LDS_SLOTS_BIG - 1));
-      bool in_lds = false;
-      for (int probe = 0; probe < 32; ++probe) {
-        // plain ds_read first: with ~100x key repetition per chunk the
-        // slot almost always already owns the key — skip the LDS RMW
-        KT cur = lkeys[slot];
-        if (cur == key) { in_lds = true; break; }
-        if (cur == EMPTY) {
-          KT prev = (KT)lds_key_cas(&lkeys[slot], EMPTY, key);
-          if (prev == EMPTY || prev == key) { in_lds = true; break; }
-        }
-        slot = (slot + 1) & (LDS_SLOTS_BIG - 1);
+    // paired (ILP-2) row processing: two independent key loads, hash
+    // computations and first-slot LDS reads in flight per iteration —
+    // hides LDS latency at the low occupancy (2 blocks/CU) this
+    // kernel's 80KB LDS footprint allows
+    int64_t i = start + threadIdx.x;
+    for (; i + blockDim.x < end; i += 2 * (int64_t)blockDim.x) {
+      int64_t i2 = i + blockDim.x;
+      KT k1 = stream_ld<NT>(&part_keys[i]);
+      KT k2 = stream_ld<NT>(&part_keys[i2]);
+      int s1 = (int)(mix64((uint64_t)(int64_t)k1) & (LDS_SLOTS_BIG - 1));
+      int s2 = (int)(mix64((uint64_t)(int64_t)k2) & (LDS_SLOTS_BIG - 1));
+      KT c1 = lkeys[s1];
+      KT c2 = lkeys[s2];
+      double v1 = is_count ? 1.0 : stream_ld<NT>(&part_vals[i]);
+      double v2 = is_count ? 1.0 : stream_ld<NT>(&part_vals[i2]);
+      if (c1 == k1) {
+        atomicAdd(&lcount[s1], 1);
+        atomicAdd(&laggs[s1], v1);
+      } else {
+        gb_lds_upsert<KT>(k1, v1, s1, EMPTY, lkeys, laggs, lcount, tkeys,
+                          gaggs, gcount, tsize);
       }
+      if (c2 == k2) {
+        atomicAdd(&lcount[s2], 1);
+        atomicAdd(&laggs[s2], v2);
+      } else {
+        gb_lds_upsert<KT>(k2, v2, s2, EMPTY, lkeys, laggs, lcount, tkeys,
+                          gaggs, gcount, tsize);
+      }
+    }
+    for (; i < end; i += blockDim.x) {
+      KT key = stream_ld<NT>(&part_keys[i]);
+      int slot = (int)(mix64((uint64_t)(int64_t)key) & (LDS_SLOTS_BIG - 1));
       double v = is_count ? 1.0 : stream_ld<NT>(&part_vals[i]);
-      if (in_lds) {
+      KT cur = lkeys[slot];
+      if (cur == key) {
         atomicAdd(&lcount[slot], 1);
         atomicAdd(&laggs[slot], v);
       } else {
-        int64_t gslot = gb_probe_insert((int64_t)key, tkeys, tsize);
-        atomicAdd((unsigned long long*)&gcount[gslot], 1ULL);
-        atomicAdd(&gaggs[gslot], v);
+        gb_lds_upsert<KT>(key, v, slot, EMPTY, lkeys, laggs, lcount, tkeys,
+                          gaggs, gcount, tsize);
+      }
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < LDS_SLOTS_BIG; i += blockDim.x) {
+      KT key = lkeys[i];
+      if (key == EMPTY) continue;
+      int64_t gslot = gb_probe_insert((int64_t)key, tkeys, tsize);
+      atomicAdd((unsigned long long*)&gcount[gslot],
+                (unsigned long long)lcount[i]);
+      atomicAdd(&gaggs[gslot], laggs[i]);
+    }
+    __syncthreads();
+  }
+}
+
+// legacy (ILP-1) row loop kept for same-box A/B (FUGUE_GB_ILP=1)
+template <bool NT, typename KT>
+__global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel_v1(
+    const KT* __restrict__ part_keys,
+    const double* __restrict__ part_vals,
+    const int32_t* __restrict__ ops,
+    int64_t n,
+    int64_t* __restrict__ tkeys,
+    double* __restrict__ gaggs,
+    int64_t* __restrict__ gcount,
+    int64_t tsize, int64_t chunk) {
+  __shared__ KT lkeys[LDS_SLOTS_BIG];
+  __shared__ double laggs[LDS_SLOTS_BIG];
+  __shared__ int lcount[LDS_SLOTS_BIG];
+  const KT EMPTY = lds_empty<KT>();
+  bool is_count = ops[0] == 3;
+  for (int64_t start = (int64_t)blockIdx.x * chunk; start < n;
+       start += (int64_t)gridDim.x * chunk) {
+    int64_t end = start + chunk;
+    if (end > n) end = n;
+    for (int i = threadIdx.x; i < LDS_SLOTS_BIG; i += blockDim.x) {
+      lkeys[i] = EMPTY;
+      lcount[i] = 0;
+      laggs[i] = 0.0;
+    }
+    __syncthreads();
+    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+      KT key = stream_ld<NT>(&part_keys[i]);
+      int slot = (int)(mix64((uint64_t)(int64_t)key) & (LDS_SLOTS_BIG - 1));
+      double v = is_count ? 1.0 : stream_ld<NT>(&part_vals[i]);
+      KT cur = lkeys[slot];
+      if (cur == key) {
+        atomicAdd(&lcount[slot], 1);
+        atomicAdd(&laggs[slot], v);
+      } else {
+        gb_lds_upsert<KT>(key, v, slot, EMPTY, lkeys, laggs, lcount, tkeys,
+                          gaggs, gcount, tsize);
       }
     }
     __syncthreads();
@@ -831,9 +930,15 @@ void launch_gb_aggregate_part_big(const void* part_keys,
   if (blocks > MAX_GRID) blocks = MAX_GRID;
   if (blocks < 1) blocks = 1;
   dim3 g((int)blocks), b(BLOCK);
+  const char* ilp = std::getenv("FUGUE_GB_ILP");
+  bool legacy = ilp != nullptr && ilp[0] == '1';
   if (narrow) {
     auto* pk = (const int32_t*)part_keys;
-    if (nt)
+    if (legacy)
+      hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v1<false, int32_t>), g,
+                         b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
+                         gcount, tsize, chunk);
+    else if (nt)
       hipLaunchKernelGGL((gb_aggregate_part_big_kernel<true, int32_t>), g, b,
                          0, stream, pk, part_vals, ops, n, tkeys, gaggs,
                          gcount, tsize, chunk);
@@ -843,7 +948,11 @@ void launch_gb_aggregate_part_big(const void* part_keys,
                          gcount, tsize, chunk);
   } else {
     auto* pk = (const int64_t*)part_keys;
-    if (nt)
+    if (legacy)
+      hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v1<false, int64_t>), g,
+                         b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
+                         gcount, tsize, chunk);
+    else if (nt)
       hipLaunchKernelGGL((gb_aggregate_part_big_kernel<true, int64_t>), g, b,
                          0, stream, pk, part_vals, ops, n, tkeys, gaggs,
                          gcount, tsize, chunk);
