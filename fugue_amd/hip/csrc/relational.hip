@@ -640,9 +640,20 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
       lcnt[i] = 0;
     }
     __syncthreads();
-    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
-      int p = (int)(mix64((uint64_t)keys[i]) >> shift);
-      atomicAdd(&lhist[p], 1);
+    {
+      int64_t i = start + threadIdx.x;
+      for (; i + blockDim.x < end; i += 2 * (int64_t)blockDim.x) {
+        int64_t k1 = keys[i];
+        int64_t k2 = keys[i + blockDim.x];
+        int p1 = (int)(mix64((uint64_t)k1) >> shift);
+        int p2 = (int)(mix64((uint64_t)k2) >> shift);
+        atomicAdd(&lhist[p1], 1);
+        atomicAdd(&lhist[p2], 1);
+      }
+      for (; i < end; i += blockDim.x) {
+        int p = (int)(mix64((uint64_t)keys[i]) >> shift);
+        atomicAdd(&lhist[p], 1);
+      }
     }
     __syncthreads();
     for (int i = threadIdx.x; i < STAGE_P; i += blockDim.x) {
@@ -654,24 +665,59 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
               : 0;
     }
     __syncthreads();
-    for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
-      int64_t key = keys[i];
-      if constexpr (sizeof(KT) == 4) {
-        // speculative narrow path: flag any key outside [0, 2^31)
-        if (ovf != nullptr && (uint64_t)key >= (1ULL << 31)) {
-          atomicOr(ovf, 1);
+    {
+      int64_t i = start + threadIdx.x;
+      for (; i + blockDim.x < end; i += 2 * (int64_t)blockDim.x) {
+        int64_t i2 = i + blockDim.x;
+        int64_t k1 = keys[i];
+        int64_t k2 = keys[i2];
+        if constexpr (sizeof(KT) == 4) {
+          if (ovf != nullptr && ((uint64_t)k1 >= (1ULL << 31) ||
+                                 (uint64_t)k2 >= (1ULL << 31))) {
+            atomicOr(ovf, 1);
+          }
+        }
+        int p1 = (int)(mix64((uint64_t)k1) >> shift);
+        int p2 = (int)(mix64((uint64_t)k2) >> shift);
+        double v1 = stream_ld<NT>(&vals[i]);
+        double v2 = stream_ld<NT>(&vals[i2]);
+        int pos1 = atomicAdd(&lcnt[p1], 1);
+        int pos2 = atomicAdd(&lcnt[p2], 1);
+        if (pos1 < STAGE_E) {
+          skey[p1 * STAGE_E + pos1] = (KT)k1;
+          sval[p1 * STAGE_E + pos1] = v1;
+        } else {
+          int64_t gpos = lbase[p1] + pos1;
+          stream_st<NT>(&out_keys[gpos], (KT)k1);
+          stream_st<NT>(&out_vals[gpos], v1);
+        }
+        if (pos2 < STAGE_E) {
+          skey[p2 * STAGE_E + pos2] = (KT)k2;
+          sval[p2 * STAGE_E + pos2] = v2;
+        } else {
+          int64_t gpos = lbase[p2] + pos2;
+          stream_st<NT>(&out_keys[gpos], (KT)k2);
+          stream_st<NT>(&out_vals[gpos], v2);
         }
       }
-      int p = (int)(mix64((uint64_t)key) >> shift);
-      int pos = atomicAdd(&lcnt[p], 1);
-      double v = stream_ld<NT>(&vals[i]);
-      if (pos < STAGE_E) {
-        skey[p * STAGE_E + pos] = (KT)key;
-        sval[p * STAGE_E + pos] = v;
-      } else {
-        int64_t gpos = lbase[p] + pos;
-        stream_st<NT>(&out_keys[gpos], (KT)key);
-        stream_st<NT>(&out_vals[gpos], v);
+      for (; i < end; i += blockDim.x) {
+        int64_t key = keys[i];
+        if constexpr (sizeof(KT) == 4) {
+          if (ovf != nullptr && (uint64_t)key >= (1ULL << 31)) {
+            atomicOr(ovf, 1);
+          }
+        }
+        int p = (int)(mix64((uint64_t)key) >> shift);
+        int pos = atomicAdd(&lcnt[p], 1);
+        double v = stream_ld<NT>(&vals[i]);
+        if (pos < STAGE_E) {
+          skey[p * STAGE_E + pos] = (KT)key;
+          sval[p * STAGE_E + pos] = v;
+        } else {
+          int64_t gpos = lbase[p] + pos;
+          stream_st<NT>(&out_keys[gpos], (KT)key);
+          stream_st<NT>(&out_vals[gpos], v);
+        }
       }
     }
     __syncthreads();
