@@ -464,8 +464,20 @@ __global__ __launch_bounds__(BLOCK) void gb_part_hist_kernel(
   __syncthreads();
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   int64_t kmin = INT64_MAX, kmax = INT64_MIN;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i + stride < n; i += 2 * stride) {
+    int64_t k1 = keys[i];
+    int64_t k2 = keys[i + stride];
+    int p1 = (int)(mix64((uint64_t)k1) >> shift);
+    int p2 = (int)(mix64((uint64_t)k2) >> shift);
+    atomicAdd(&lhist[p1], 1);
+    atomicAdd(&lhist[p2], 1);
+    if (k1 < kmin) kmin = k1;
+    if (k1 > kmax) kmax = k1;
+    if (k2 < kmin) kmin = k2;
+    if (k2 > kmax) kmax = k2;
+  }
+  for (; i < n; i += stride) {
     int64_t k = keys[i];
     int p = (int)(mix64((uint64_t)k) >> shift);
     atomicAdd(&lhist[p], 1);
