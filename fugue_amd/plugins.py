@@ -1,0 +1,32 @@
+"""Public plugin points (reference parity: ``fugue/plugins.py``).
+
+Third-party packages extend fugue_amd through these:
+
+- ``register_execution_engine`` / ``register_sql_engine`` /
+  ``register_default_execution_engine`` — engine aliases and types
+- ``register_annotated_param`` — new UDF parameter frame types
+  (the MI355X engine registers :class:`~fugue_amd.hip.udf.HipDataFrameParam`
+  through this)
+- ``register_creator`` / ``register_processor`` / ``register_outputter``
+  / ``register_transformer`` — named extensions
+- ``register_global_conf`` — configuration defaults
+- setuptools entry points in the ``fugue.plugins`` / ``fugue_amd.plugins``
+  groups are loaded at import (``fugue_amd.registry.load_entry_point_plugins``)
+"""
+# flake8: noqa
+from fugue_amd.constants import register_global_conf
+from fugue_amd.dataframe.function_wrapper import register_annotated_param
+from fugue_amd.execution.factory import (
+    make_execution_engine,
+    make_sql_engine,
+    register_default_execution_engine,
+    register_execution_engine,
+    register_sql_engine,
+)
+from fugue_amd.extensions import (
+    register_creator,
+    register_outputter,
+    register_processor,
+    register_transformer,
+)
+from fugue_amd.registry import load_entry_point_plugins
