@@ -56,6 +56,7 @@ _STMT_START = {
     "FILL",
     "CONNECT",
     "WITH",
+    "SUB",
 }
 
 _POST_CLAUSES = {
@@ -143,6 +144,8 @@ class FugueSQLParser:
             df = self._parse_sample()
         elif kw == "FILL":
             df = self._parse_fill()
+        elif kw == "SUB":
+            df = self._parse_sub()
         else:
             raise FugueSQLSyntaxError(
                 f"unexpected token {t.value!r} at {t.pos} in FugueSQL"
@@ -240,8 +243,16 @@ class FugueSQLParser:
                     from_context = False
                 elif from_context and self.hooks.has_var(t.value):
                     parts.append((False, self.code[seg_start : t.pos]))
-                    parts.append((True, t.value))
                     ts.next()
+                    ref_end = t.pos + len(t.value)
+                    if ts.match_punct("["):
+                        # dfs[key] / dfs[0]: resolve to the object here
+                        obj = self._maybe_index(self.hooks.get_var(t.value))
+                        last = ts.tokens[ts.pos - 1]  # the closing ']'
+                        ref_end = last.pos + len(last.value)
+                        parts.append((True, obj))
+                    else:
+                        parts.append((True, t.value))
                     # keep the original name visible as a table alias unless
                     # the query supplies its own alias
                     nxt2 = ts.peek()
@@ -262,14 +273,14 @@ class FugueSQLParser:
                     )
                     if not has_alias:
                         parts.append((False, f" AS {t.value} "))
-                    seg_start = t.pos + len(t.value)
+                    seg_start = ref_end
                     end = seg_start
                     continue
             ts.next()
             end = t.pos + len(t.value)
         parts.append((False, self.code[seg_start:end]))
         return self.hooks.select_statement(
-            [(r, v) for r, v in parts if v != ""]
+            [(r, v) for r, v in parts if r or v != ""]
         )
 
     def _token_end(self, idx: int) -> int:
@@ -417,12 +428,58 @@ class FugueSQLParser:
                 not in ("USING", "PREPARTITION", "ROWS", "ROWCOUNT", "TITLE", "BY", "PRESORT", "FROM", "SCHEMA", "CALLBACK", "OVERWRITE", "APPEND", "TO", "ROW", "AND")
                 and self.hooks.has_var(t.value)
             ):
-                dfs.append(self.hooks.get_var(ts.next().value))
+                dfs.append(self._maybe_index(self.hooks.get_var(ts.next().value)))
                 if not ts.take_punct(","):
                     break
             else:
                 break
         return dfs
+
+    def _maybe_index(self, obj: Any) -> Any:
+        """Apply ``[key]`` / ``[0]`` indexing after a dataframes variable
+        reference (reference test ``print dfs[a1]``)."""
+        ts = self.ts
+        while ts.take_punct("["):
+            t = ts.next()
+            if t.kind == "NUMBER":
+                key: Any = int(t.value)
+            elif t.kind == "STRING":
+                key = t.value[1:-1]
+            else:
+                key = t.value
+            ts.expect_punct("]")
+            obj = obj[key]
+        return obj
+
+    def _parse_sub(self) -> Any:
+        """``SUB [df | key:df, ...] USING module [(params)]`` — invoke a
+        :func:`fugue_amd.workflow.module.module` function (reference
+        ``_visitors.py:697`` ``visitFugueModuleTask``)."""
+        ts = self.ts
+        ts.expect_kw("SUB")
+        ordered: List[Any] = []
+        named: Dict[str, Any] = {}
+        while True:
+            t = ts.peek()
+            if t is None or t.kind != "NAME" or t.upper == "USING":
+                break
+            nxt = ts.peek(1)
+            if nxt is not None and nxt.kind == "PUNCT" and nxt.value == ":":
+                key = ts.next().value
+                ts.next()  # ':'
+                named[key] = self._maybe_index(
+                    self.hooks.get_var(ts.next().value)
+                )
+            elif self.hooks.has_var(t.value):
+                ordered.append(self._maybe_index(self.hooks.get_var(ts.next().value)))
+            else:
+                break
+            if not ts.take_punct(","):
+                break
+        ts.expect_kw("USING")
+        ext = self._parse_extension_name()
+        params = self._parse_params()
+        return self.hooks.sql_module(ordered, named, ext, params)
 
     def _parse_prepartition(self) -> Optional[PartitionSpec]:
         ts = self.ts

@@ -104,7 +104,7 @@ class FugueSQLWorkflow(FugueWorkflow):
         statements: List[Any] = []
         for is_ref, v in parts:
             if is_ref:
-                statements.append(self.get_var(v))
+                statements.append(self.get_var(v) if isinstance(v, str) else v)
             else:
                 statements.append(v)
         return self.select(*statements)
@@ -270,6 +270,46 @@ class FugueSQLWorkflow(FugueWorkflow):
 
     def sql_fillna(self, dfs: List[Any], value: Any) -> WorkflowDataFrame:
         return self._dfs_or_last(dfs)[0].fillna(value)
+
+    def sql_module(
+        self,
+        ordered: List[Any],
+        named: Dict[str, Any],
+        ext: Any,
+        params: ParamDict,
+    ) -> Any:
+        """``SUB`` statement: invoke a workflow module (reference
+        ``_visitors.py:697``).  Input dispatch: ``WorkflowDataFrames``
+        annotation gets the whole collection; otherwise positional
+        frames (or the previous statement's frame when none given)."""
+        from fugue_amd.workflow.module import to_module
+        from fugue_amd.workflow.workflow import WorkflowDataFrames
+
+        sub = to_module(self._resolve_ext(ext), self._captured)
+        if sub.has_input:
+            if len(ordered) == 0 and len(named) == 0:
+                dfs = WorkflowDataFrames(self._dfs_or_last([])[0])
+            elif len(named) > 0:
+                dfs = WorkflowDataFrames(**named)
+            else:
+                dfs = WorkflowDataFrames(*ordered)
+        else:
+            dfs = WorkflowDataFrames()
+        p = dict(params)
+        if sub.has_dfs_input:
+            result = sub(dfs, **p)
+        elif len(dfs) == 0:
+            result = sub(self, **p)
+        elif len(dfs) == 1 or not dfs.has_key:
+            result = sub(*list(dfs.values()), **p)
+        else:
+            result = sub(**dfs, **p)
+        if sub.has_single_output:
+            self._last_df = result
+            return result
+        if sub.has_multiple_output:
+            return result
+        return None
 
     # --- post clauses ---------------------------------------------------- #
     def sql_persist(self, df: Any) -> Any:

@@ -287,3 +287,69 @@ def test_in_and_scalar_subqueries():
         "SELECT k FROM a WHERE v > (SELECT AVG(v) FROM a)", dict(a=df)
     )[0]
     assert r["k"].tolist() == [3]
+
+
+def test_sql_module_sub():
+    """SUB statement invoking @module functions (reference
+    tests/fugue/sql/test_workflow_parse.py:711 test_module scenario)."""
+    from fugue_amd.workflow.module import module
+    from fugue_amd.workflow.workflow import (
+        FugueWorkflow,
+        WorkflowDataFrame,
+        WorkflowDataFrames,
+    )
+
+    def create(wf: FugueWorkflow, n: int = 1) -> WorkflowDataFrame:
+        return wf.df([[n]], "a:int")
+
+    def merge(
+        df1: WorkflowDataFrame, df2: WorkflowDataFrame, k: str = "aa"
+    ) -> WorkflowDataFrames:
+        return WorkflowDataFrames({k: df1, "bb": df2})
+
+    def merge2(
+        wf: FugueWorkflow, dfs: WorkflowDataFrames, k: int = 0
+    ) -> WorkflowDataFrame:
+        return dfs[k]
+
+    def merge3(df1: WorkflowDataFrame, df2: WorkflowDataFrame) -> WorkflowDataFrames:
+        return WorkflowDataFrames(df1, df2)
+
+    collected = []
+
+    @module()
+    def out1(wf: FugueWorkflow, df: WorkflowDataFrame) -> None:
+        def grab(pdf: pd.DataFrame) -> None:
+            collected.append(pdf["a"].tolist())
+
+        df.output(grab)
+
+    dag = FugueSQLWorkflow()
+    dag(
+        """
+        a=sub using create
+        b=sub using create(n=2)
+        dfs=sub a,b using merge(k="a1")
+        r1=select * from dfs[a1]
+        r1 yield dataframe as r1
+        r2=select * from dfs[bb]
+        r2 yield dataframe as r2
+        m2=sub a,b using merge2(k=1)
+        sub using out1
+        dfs2=sub df2:a,df1:b using merge3
+        r3=select * from dfs2[0]
+        r3 yield dataframe as r3
+        """,
+        create=create,
+        merge=merge,
+        merge2=merge2,
+        merge3=merge3,
+        out1=out1,
+    )
+    res = dag.run()
+    assert res["r1"].result.as_array() == [[1]]
+    assert res["r2"].result.as_array() == [[2]]
+    # merge2 selects dfs[1] == b == [[2]]; out1 prints the last frame (m2)
+    assert collected == [[2]]
+    # merge3 named refs: df1=b, df2=a → positional order df1,df2 = b,a
+    assert res["r3"].result.as_array() == [[2]]
