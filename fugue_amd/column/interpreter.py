@@ -172,9 +172,12 @@ def _eval_agg_series(expr: _UnaryAggFuncExpr, df: pd.DataFrame) -> Any:
         raise NotImplementedError(f"{fname}(*)")
     s = eval_scalar_expr(argexpr, df)
     if expr.is_distinct:
-        if fname != "COUNT":
-            raise NotImplementedError("DISTINCT only supported for COUNT")
-        return s.dropna().nunique()
+        if fname == "COUNT":
+            return s.dropna().nunique()
+        if fname in ("SUM", "AVG", "MIN", "MAX"):
+            s = s.drop_duplicates()
+        else:
+            raise NotImplementedError(f"DISTINCT not supported for {fname}")
     if fname == "COUNT":
         return int(s.notna().sum())
     if fname == "MIN":

@@ -1398,12 +1398,20 @@ class HipExecutionEngine(ExecutionEngine):
             if isinstance(c, _UnaryAggFuncExpr):
                 fname = c.func.upper()
                 arg = c.args[0]
-                if c.is_distinct:
-                    if fname != "COUNT" or not isinstance(
+                if c.is_distinct and fname in ("MIN", "MAX"):
+                    pass  # DISTINCT is a no-op under MIN/MAX: plain path
+                elif c.is_distinct:
+                    if fname not in ("COUNT", "SUM", "AVG") or not isinstance(
                         arg, _NamedColumnExpr
                     ):
-                        raise DeviceExprError("non-COUNT distinct: fallback")
-                    plans.append((name, "count_distinct", arg.name))
+                        raise DeviceExprError(
+                            f"{fname} distinct on expression: fallback"
+                        )
+                    # dedupe (keys, x) then aggregate — handled by
+                    # _device_aggregate_distinct
+                    plans.append(
+                        (name, f"{fname.lower()}_distinct", arg.name)
+                    )
                     continue
                 if fname == "COUNT":
                     if (
@@ -1452,7 +1460,11 @@ class HipExecutionEngine(ExecutionEngine):
             return self._device_global_aggregate(
                 d, plans, partials, derived, having, cols
             )
-        cd_plans = [p for p in plans if p[1] == "count_distinct"]
+        cd_plans = [
+            p
+            for p in plans
+            if p[1] in ("count_distinct", "sum_distinct", "avg_distinct")
+        ]
         if cd_plans:
             return self._device_aggregate_distinct(
                 d, key_names, cols, plans, having
@@ -1593,7 +1605,7 @@ class HipExecutionEngine(ExecutionEngine):
         allreduce-merged across ranks.  FIRST/LAST/DISTINCT shapes fall
         back to the host path."""
         for _n, kind, _i in plans:
-            if kind in ("firstlast", "count_distinct"):
+            if kind == "firstlast" or kind.endswith("_distinct"):
                 raise DeviceExprError("global first/last/distinct: fallback")
         if cols.is_distinct:
             raise DeviceExprError("distinct aggregate: fallback")
@@ -1696,10 +1708,11 @@ class HipExecutionEngine(ExecutionEngine):
         plans: List[Tuple[str, str, Any]],
         having: Optional[ColumnExpr],
     ) -> DataFrame:
-        """COUNT(DISTINCT x) decomposition: dedupe (keys, x) rows, then
-        COUNT per key; joined back to the other aggregates on the keys.
-        (Reference comparator: the SQL backends' native COUNT DISTINCT,
-        e.g. duckdb via ``fugue_duckdb``.)
+        """COUNT/SUM/AVG(DISTINCT x) decomposition: dedupe (keys, x)
+        rows, then aggregate per key; joined back to the other aggregates
+        on the keys.  (MIN/MAX DISTINCT are planned as plain MIN/MAX.)
+        (Reference comparator: the SQL backends' native DISTINCT
+        aggregates, e.g. duckdb via ``fugue_duckdb``.)
 
         Null group keys would not survive the inner re-join (SQL null
         semantics) so that shape falls back to the host path."""
@@ -1722,13 +1735,13 @@ class HipExecutionEngine(ExecutionEngine):
             raise DeviceExprError("count distinct with null keys")
         base_exprs = [_col(k) for k in key_names]
         out_order: List[str] = []
-        cd_items: List[Tuple[str, str]] = []  # (out name, src col)
+        cd_items: List[Tuple[str, str, str]] = []  # (out name, src col, func)
         for name, kind, info in plans:
             out_order.append(name)
             if kind == "key":
                 continue
-            if kind == "count_distinct":
-                cd_items.append((name, info))
+            if kind.endswith("_distinct"):
+                cd_items.append((name, info, kind[: -len("_distinct")]))
                 continue
             # rebuild the original expression for the base aggregate
             for c in cols.all_cols:
@@ -1739,13 +1752,14 @@ class HipExecutionEngine(ExecutionEngine):
         if len(base_exprs) > len(key_names):
             base_sc = SelectColumns(*base_exprs)
             res = self._device_aggregate(d, base_sc, None)
-        for name, src in cd_items:
+        _agg_fn = {"count": ff.count, "sum": ff.sum, "avg": ff.avg}
+        for name, src, func in cd_items:
             proj_names = key_names + ([src] if src not in key_names else [])
             proj = self._device_select_named(d, proj_names)
             dedup = self.to_df(self.distinct(proj))
             agg_sc = SelectColumns(
                 *[_col(k) for k in key_names],
-                ff.count(_col(src)).alias(name),
+                _agg_fn[func](_col(src)).alias(name),
             )
             part = self._device_aggregate(self.to_df(dedup), agg_sc, None)
             res = (

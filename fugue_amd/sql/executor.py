@@ -179,6 +179,8 @@ class FuncCall(Expr):
             return int(s.notna().sum())
         s = _as_series(self.args[0].eval(ctx), ctx)
         nn = s.dropna()
+        if self.distinct:
+            nn = nn.drop_duplicates()
         if len(nn) == 0:
             return None
         if name == "SUM":

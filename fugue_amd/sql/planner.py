@@ -111,7 +111,12 @@ def _to_column_expr(e: X.Expr, schema: Schema, alias_map: Dict[str, str]) -> Col
                 "FIRST": F.first,
                 "LAST": F.last,
             }[name]
-            return fn(inner)
+            res = fn(inner)
+            if e.distinct and name in ("SUM", "AVG", "MIN", "MAX"):
+                from fugue_amd.column.expressions import _UnaryAggFuncExpr
+
+                res = _UnaryAggFuncExpr(name, inner, arg_distinct=True)
+            return res
         if name == "COALESCE":
             return F.coalesce(
                 *[_to_column_expr(a, schema, alias_map) for a in e.args]

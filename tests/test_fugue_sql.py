@@ -353,3 +353,19 @@ def test_sql_module_sub():
     assert collected == [[2]]
     # merge3 named refs: df1=b, df2=a → positional order df1,df2 = b,a
     assert res["r3"].result.as_array() == [[2]]
+
+
+def test_distinct_aggregates_sql_executor():
+    from fugue_amd.sql.executor import run_sql_on_pandas
+
+    df = pd.DataFrame(dict(k=[1, 1, 1, 2, 2], v=[1.0, 1.0, 3.0, 2.0, 2.0]))
+    r = run_sql_on_pandas(
+        "SELECT k, SUM(DISTINCT v) AS s, AVG(DISTINCT v) AS a, "
+        "MIN(DISTINCT v) AS mn, COUNT(DISTINCT v) AS c "
+        "FROM t GROUP BY k ORDER BY k",
+        dict(t=df),
+    )[0]
+    assert r["s"].tolist() == [4.0, 2.0]
+    assert r["a"].tolist() == [2.0, 2.0]
+    assert r["mn"].tolist() == [1.0, 2.0]
+    assert r["c"].tolist() == [2, 1]

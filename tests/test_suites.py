@@ -95,3 +95,47 @@ from fugue_amd.testing.suites import BagTestSuite
 
 class TestArrayBag(BagTestSuite):
     pass
+
+
+def test_hip_cpu_distinct_aggregates():
+    """SUM/AVG/MIN DISTINCT on the device engine (CPU tensors), checked
+    against the pandas groupby comparator."""
+    import numpy as np
+    import pandas as pd
+
+    import fugue_amd.api as fa
+    from fugue_amd.column import functions as f
+    from fugue_amd.column.expressions import _UnaryAggFuncExpr, col
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    rng = np.random.default_rng(11)
+    pdf = pd.DataFrame(
+        dict(
+            k=rng.integers(0, 15, 2000),
+            v=rng.integers(0, 6, 2000).astype("f8"),
+        )
+    )
+    res = fa.aggregate(
+        pdf,
+        partition_by="k",
+        engine=e,
+        sd=_UnaryAggFuncExpr("SUM", col("v"), arg_distinct=True),
+        ad=_UnaryAggFuncExpr("AVG", col("v"), arg_distinct=True),
+        md=_UnaryAggFuncExpr("MIN", col("v"), arg_distinct=True),
+        tot=f.sum(col("v")),
+    )
+    got = fa.as_pandas(res).sort_values("k").reset_index(drop=True)
+    exp = (
+        pdf.groupby("k", as_index=False)
+        .agg(
+            sd=("v", lambda s: s.drop_duplicates().sum()),
+            ad=("v", lambda s: s.drop_duplicates().mean()),
+            md=("v", "min"),
+            tot=("v", "sum"),
+        )
+        .sort_values("k")
+        .reset_index(drop=True)
+    )
+    for c in ("sd", "ad", "md", "tot"):
+        assert np.allclose(got[c].to_numpy(float), exp[c].to_numpy(float)), c
