@@ -2,7 +2,7 @@
 each lowers to a HIP kernel; fusion of the scalar pipeline comes from
 torch's elementwise fuser, while the relational ops use the hand-written
 kernels in ``csrc/relational.hip``)."""
-from typing import Any, Optional, Tuple
+from typing import Any, List, Optional, Tuple
 
 import pyarrow as pa
 import torch
@@ -187,13 +187,12 @@ def _eval_like(
     if not isinstance(c, StringDeviceColumn):
         raise DeviceExprError("LIKE on non-string column")
     pat = pat_e.value
-    if "_" in pat:
-        raise DeviceExprError("LIKE with _ wildcard")
     prefix_any = pat.startswith("%")
     suffix_any = pat.endswith("%")
     core = pat.strip("%")
-    if "%" in core:
-        raise DeviceExprError("LIKE with interior %")
+    if "_" in pat or "%" in core:
+        # general pattern: segment matcher (greedy earliest-occurrence)
+        return _eval_like_general(c, pat, torch.device(df.device))
     device = torch.device(df.device)
     n = len(c)
     lengths = c.offsets[1:] - c.offsets[:-1]
@@ -247,6 +246,106 @@ def _eval_like(
                         rows.numel(), dtype=torch.bool, device=device
                     ),
                 )
+    return res, c.valid
+
+
+def _eval_like_general(
+    c: "StringDeviceColumn", pat: str, device: torch.device
+) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
+    """Full LIKE semantics on device tensors: the pattern is split on
+    ``%`` into segments; the first/last segments are anchored at row
+    start/end, middle segments are located greedily (earliest match ≥
+    the running cursor, which is optimal for glob matching).  ``_``
+    matches one byte, so patterns containing ``_`` fall back when the
+    column holds non-ASCII data (multi-byte characters)."""
+    n = len(c)
+    total = int(c.bytes.numel())
+    if "_" in pat and total > 0 and int(c.bytes.max().item()) >= 0x80:
+        raise DeviceExprError("LIKE _ on non-ASCII data")
+    starts = c.offsets[:-1]
+    ends = c.offsets[1:]
+    lengths = ends - starts
+
+    def seg_bytes(seg: str) -> List[Optional[int]]:
+        return [None if ch == "_" else b for ch, b in
+                zip(seg, seg.encode("utf-8"))] if seg.isascii() else _multi(seg)
+
+    def _multi(seg: str) -> List[Optional[int]]:
+        out: List[Optional[int]] = []
+        for ch in seg:
+            if ch == "_":
+                out.append(None)
+            else:
+                out.extend(ch.encode("utf-8"))
+        return out
+
+    def window_mask(sb: List[Optional[int]]) -> torch.Tensor:
+        """bool over byte positions [0, total-k+1): segment matches."""
+        k = len(sb)
+        if total < k:
+            return torch.zeros(0, dtype=torch.bool, device=device)
+        w = total - k + 1
+        m = torch.ones(w, dtype=torch.bool, device=device)
+        for j, b in enumerate(sb):
+            if b is None:
+                continue
+            m = m & (c.bytes[j : w + j] == b)
+        return m
+
+    def match_at(sb: List[Optional[int]], pos: torch.Tensor) -> torch.Tensor:
+        """bool per row: segment matches at byte position pos (caller
+        guarantees pos+k is inside the row when the row is eligible)."""
+        k = len(sb)
+        res = torch.ones(n, dtype=torch.bool, device=device)
+        cap = max(total - 1, 0)
+        for j, b in enumerate(sb):
+            if b is None:
+                continue
+            idx = torch.clamp(pos + j, min=0, max=cap)
+            res = res & (c.bytes.index_select(0, idx) == b)
+        return res
+
+    segs = pat.split("%")
+    first, last, middles = segs[0], segs[-1] if len(segs) > 1 else "", [
+        s for s in segs[1:-1] if s != ""
+    ]
+    if len(segs) == 1:
+        # no %: fixed-length pattern with _ wildcards
+        sb = seg_bytes(first)
+        res = (lengths == len(sb)) & match_at(sb, starts)
+        return res, c.valid
+    res = torch.ones(n, dtype=torch.bool, device=device)
+    cur = starts.clone()
+    if first != "":
+        sb = seg_bytes(first)
+        res = res & (lengths >= len(sb)) & match_at(sb, starts)
+        cur = starts + len(sb)
+    for seg in middles:
+        sb = seg_bytes(seg)
+        k = len(sb)
+        m = window_mask(sb)
+        pos = m.nonzero(as_tuple=True)[0]
+        # keep only matches fully inside a row
+        row = torch.searchsorted(ends, pos, right=True)
+        ok = (pos + k) <= ends.index_select(0, torch.clamp(row, max=n - 1))
+        pos, row = pos[ok], row[ok]
+        if pos.numel() == 0:
+            res = torch.zeros(n, dtype=torch.bool, device=device)
+            break
+        # earliest match position >= cur, required to be in the same row
+        idx = torch.searchsorted(pos, cur)
+        found = idx < pos.numel()
+        idx_c = torch.clamp(idx, max=pos.numel() - 1)
+        p = pos.index_select(0, idx_c)
+        r = row.index_select(0, idx_c)
+        ok_row = found & (r == torch.arange(n, device=device))
+        res = res & ok_row
+        cur = torch.where(ok_row, p + k, cur)
+    if last != "":
+        sb = seg_bytes(last)
+        k = len(sb)
+        at = ends - k
+        res = res & (at >= cur) & match_at(sb, at)
     return res, c.valid
 
 

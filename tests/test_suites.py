@@ -139,3 +139,53 @@ def test_hip_cpu_distinct_aggregates():
     )
     for c in ("sd", "ad", "md", "tot"):
         assert np.allclose(got[c].to_numpy(float), exp[c].to_numpy(float)), c
+
+
+def test_hip_cpu_like_general_patterns():
+    """Full LIKE wildcard support on the device engine (CPU tensors):
+    `_`, interior `%`, and mixes, vs a regex comparator."""
+    import re as _re
+
+    import numpy as np
+    import pandas as pd
+
+    import fugue_amd.api as fa
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    rng = np.random.default_rng(7)
+    alphabet = list("abcdez")
+    vals = [
+        "".join(rng.choice(alphabet, rng.integers(0, 8)).tolist())
+        for _ in range(500)
+    ] + ["", "abc", "abcde", "azc", "aXc"]
+    pdf = pd.DataFrame(dict(s=vals, i=range(len(vals))))
+
+    def like_to_re(p):
+        out = "^"
+        for ch in p:
+            if ch == "%":
+                out += ".*"
+            elif ch == "_":
+                out += "."
+            else:
+                out += _re.escape(ch)
+        return out + "$"
+
+    patterns = [
+        "a_c", "%b_d%", "a%c%e", "_bc", "a__%", "%_z", "abc",
+        "a%c", "%de", "%%", "a_%_e", "__", "%a%b%",
+    ]
+    from fugue_amd.hip import expr as hexpr
+
+    for p in patterns:
+        got = fa.as_pandas(
+            fa.fugue_sql(
+                f"SELECT i FROM pdf WHERE s LIKE '{p}'", engine=e
+            )
+        )["i"].sort_values().tolist()
+        rx = _re.compile(like_to_re(p))
+        exp = pdf[pdf["s"].map(lambda s: rx.match(s) is not None)][
+            "i"
+        ].sort_values().tolist()
+        assert got == exp, (p, got[:10], exp[:10])
