@@ -354,3 +354,67 @@ def test_device_resident_udf_gpu():
     r = res if isinstance(res, pd.DataFrame) else res.as_pandas()
     r = r.sort_values("k").reset_index(drop=True)
     assert r["v"].tolist() == [5.0, 9.0]
+
+
+def test_distinct_aggregates_gpu(engine):
+    """SUM/AVG DISTINCT decomposition on device vs pandas comparator."""
+    from fugue_amd.column.expressions import _UnaryAggFuncExpr
+
+    rng = np.random.default_rng(11)
+    pdf = pd.DataFrame(
+        dict(
+            k=rng.integers(0, 50, 100_000),
+            v=rng.integers(0, 9, 100_000).astype("f8"),
+        )
+    )
+    res = fa.as_pandas(
+        fa.aggregate(
+            pdf,
+            partition_by="k",
+            engine=engine,
+            sd=_UnaryAggFuncExpr("SUM", col("v"), arg_distinct=True),
+            ad=_UnaryAggFuncExpr("AVG", col("v"), arg_distinct=True),
+            tot=f.sum(col("v")),
+        )
+    ).sort_values("k").reset_index(drop=True)
+    exp = (
+        pdf.groupby("k", as_index=False)
+        .agg(
+            sd=("v", lambda s: s.drop_duplicates().sum()),
+            ad=("v", lambda s: s.drop_duplicates().mean()),
+            tot=("v", "sum"),
+        )
+        .sort_values("k")
+        .reset_index(drop=True)
+    )
+    for c in ("sd", "ad", "tot"):
+        assert np.allclose(res[c].to_numpy(float), exp[c].to_numpy(float))
+
+
+def test_like_general_gpu(engine):
+    """General LIKE (_ and interior %) on device vs regex comparator."""
+    import re as _re
+
+    rng = np.random.default_rng(7)
+    alphabet = list("abcdez")
+    vals = [
+        "".join(rng.choice(alphabet, rng.integers(0, 8)).tolist())
+        for _ in range(20_000)
+    ] + ["", "abc", "abcde", "azc"]
+    pdf = pd.DataFrame(dict(s=vals, i=range(len(vals))))
+
+    def like_to_re(p):
+        out = "^"
+        for ch in p:
+            out += ".*" if ch == "%" else "." if ch == "_" else _re.escape(ch)
+        return out + "$"
+
+    for p in ("a_c", "%b_d%", "a%c%e", "_bc", "a__%", "%_z", "a_%_e"):
+        got = fa.as_pandas(
+            fa.fugue_sql(f"SELECT i FROM pdf WHERE s LIKE '{p}'", engine=engine)
+        )["i"].sort_values().tolist()
+        rx = _re.compile(like_to_re(p))
+        exp = pdf[pdf["s"].map(lambda s: rx.match(s) is not None)][
+            "i"
+        ].sort_values().tolist()
+        assert got == exp, p
