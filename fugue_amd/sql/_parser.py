@@ -112,6 +112,18 @@ class FugueSQLParser:
         kw = t.upper
         if kw == "SELECT" or kw == "WITH":
             df = self._parse_select_statement()
+        elif kw == "CONNECT":
+            # CONNECT <engine>[(params)] SELECT ... : per-query SQL engine
+            # (reference ``_visitors.py:728`` visitFugueSqlEngine)
+            ts.next()
+            eng = self._parse_extension_name()
+            eng_params = self._parse_params()
+            nt = ts.peek()
+            if nt is None or nt.upper not in ("SELECT", "WITH"):
+                raise FugueSQLSyntaxError("CONNECT must be followed by SELECT")
+            df = self._parse_select_statement(
+                sql_engine=eng, sql_engine_params=eng_params
+            )
         elif kw == "CREATE":
             df = self._parse_create()
         elif kw == "LOAD":
@@ -185,7 +197,9 @@ class FugueSQLParser:
                 return df
 
     # ------------------------------------------------------------------ #
-    def _parse_select_statement(self) -> Any:
+    def _parse_select_statement(
+        self, sql_engine: Any = None, sql_engine_params: Any = None
+    ) -> Any:
         """Collect the SELECT body tokens until a statement/post-clause
         boundary; table references resolved through the hook."""
         ts = self.ts
@@ -280,7 +294,9 @@ class FugueSQLParser:
             end = t.pos + len(t.value)
         parts.append((False, self.code[seg_start:end]))
         return self.hooks.select_statement(
-            [(r, v) for r, v in parts if r or v != ""]
+            [(r, v) for r, v in parts if r or v != ""],
+            sql_engine=sql_engine,
+            sql_engine_params=sql_engine_params,
         )
 
     def _token_end(self, idx: int) -> int:

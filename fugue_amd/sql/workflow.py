@@ -78,7 +78,12 @@ class FugueSQLWorkflow(FugueWorkflow):
         return [self.last_df]
 
     # --- statement hooks ------------------------------------------------ #
-    def select_statement(self, parts: List[Tuple[bool, str]]) -> WorkflowDataFrame:
+    def select_statement(
+        self,
+        parts: List[Tuple[bool, str]],
+        sql_engine: Any = None,
+        sql_engine_params: Any = None,
+    ) -> WorkflowDataFrame:
         has_ref = any(r for r, _ in parts)
         text = "".join(v for r, v in parts if not r)
         norm = " " + text.upper().replace("\n", " ") + " "
@@ -107,7 +112,13 @@ class FugueSQLWorkflow(FugueWorkflow):
                 statements.append(self.get_var(v) if isinstance(v, str) else v)
             else:
                 statements.append(v)
-        return self.select(*statements)
+        return self.select(
+            *statements,
+            sql_engine=self._resolve_ext(sql_engine)
+            if sql_engine is not None
+            else None,
+            sql_engine_params=sql_engine_params,
+        )
 
     def sql_create(self, ext: str, schema: Any, params: ParamDict) -> WorkflowDataFrame:
         return self.create(self._resolve_ext(ext), schema=schema, params=params)

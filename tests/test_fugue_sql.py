@@ -369,3 +369,39 @@ def test_distinct_aggregates_sql_executor():
     assert r["a"].tolist() == [2.0, 2.0]
     assert r["mn"].tolist() == [1.0, 2.0]
     assert r["c"].tolist() == [2, 1]
+
+
+def test_connect_sql_engine():
+    """CONNECT <engine> SELECT ...: per-query SQL engine (reference
+    test_workflow_parse.py test_select_plus_engine)."""
+    from fugue_amd.execution.execution_engine import SQLEngine
+    from fugue_amd.execution.factory import register_sql_engine
+
+    seen = []
+
+    class MockEngine(SQLEngine):
+        def __init__(self, execution_engine, p: int = 0):
+            super().__init__(execution_engine)
+            self.p = p
+
+        @property
+        def is_distributed(self):
+            return False
+
+        def select(self, dfs, statement):
+            seen.append(self.p)
+            from fugue_amd.sql.executor import run_sql_on_pandas
+
+            sql = statement.construct(dialect=None)
+            pdfs = {k: v.as_pandas() for k, v in dfs.items()}
+            from fugue_amd.dataframe.pandas_dataframe import PandasDataFrame
+
+            return PandasDataFrame(run_sql_on_pandas(sql, pdfs)[0])
+
+    register_sql_engine("_mock_sql", lambda e, **kw: MockEngine(e, **kw))
+    df = pd.DataFrame(dict(a=[1, 2, 3]))
+    r = fugue_sql("connect _mock_sql(p=2) select a from df where a>1")
+    assert sorted(r["a"].tolist()) == [2, 3]
+    r2 = fugue_sql("connect MockEngine select a from df where a>2")
+    assert r2["a"].tolist() == [3]
+    assert seen == [2, 0]
