@@ -295,6 +295,29 @@ class ExecutionEngineTestSuite:
                     throw=True,
                 ), fmt
 
+    def test_load_glob_and_single_save(self):
+        """Glob loads + force_single saves incl. headerless CSV
+        (reference execution_suite test_load_*_folder,
+        test_save_single_and_load_csv_no_header)."""
+        with tempfile.TemporaryDirectory() as tmp:
+            a = self.df([[1, "a"], [2, "b"]], "x:long,y:str")
+            b = self.df([[3, "c"]], "x:long,y:str")
+            p1 = os.path.join(tmp, "part1.parquet")
+            p2 = os.path.join(tmp, "part2.parquet")
+            self.engine.save_df(a, p1, force_single=True)
+            self.engine.save_df(b, p2, force_single=True)
+            back = self.engine.to_df(
+                self.engine.load_df(os.path.join(tmp, "*.parquet"))
+            )
+            assert sorted(back.as_array()) == [[1, "a"], [2, "b"], [3, "c"]]
+            # headerless single csv round-trip (schema supplied on load)
+            pc = os.path.join(tmp, "single.csv")
+            self.engine.save_df(a, pc, force_single=True, header=False)
+            back2 = self.engine.to_df(
+                self.engine.load_df(pc, columns="x:long,y:str", header=False)
+            )
+            assert sorted(back2.as_array()) == [[1, "a"], [2, "b"]]
+
     # --- sql facet --------------------------------------------------------------- #
     def test_sql_select(self):
         from fugue_amd.collections.sql import StructuredRawSQL
