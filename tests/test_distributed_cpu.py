@@ -476,3 +476,55 @@ def test_distributed_global_aggregate():
         s, c, mn, av = rows[0]
         assert s == sum(range(100)) and c == 100 and mn == 0.0
         assert abs(av - 49.5) < 1e-9
+
+
+def _sum_distinct_job(rank: int):
+    import fugue_amd.api as fa
+    from fugue_amd.column import functions as f
+    from fugue_amd.column.expressions import _UnaryAggFuncExpr, col
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    rng = np.random.default_rng(13)
+    pdf = pd.DataFrame(
+        dict(
+            k=rng.integers(0, 12, 3000),
+            v=rng.integers(0, 7, 3000).astype("f8"),
+        )
+    )
+    res = fa.aggregate(
+        pdf, partition_by="k", engine=e,
+        sd=_UnaryAggFuncExpr("SUM", col("v"), arg_distinct=True),
+        tot=f.sum(col("v")),
+        as_fugue=True,
+    )
+    local = e._gather_all(res)
+    return (
+        local.as_pandas().sort_values("k").reset_index(drop=True).to_dict("list")
+    )
+
+
+def test_distributed_sum_distinct():
+    """SUM DISTINCT through the dedupe decomposition across 2 ranks
+    (each rank holds a copy of the frame → sharded by the engine)."""
+    results = run_distributed(_sum_distinct_job, 29531)
+    rng = np.random.default_rng(13)
+    pdf = pd.DataFrame(
+        dict(
+            k=rng.integers(0, 12, 3000),
+            v=rng.integers(0, 7, 3000).astype("f8"),
+        )
+    )
+    expected = (
+        pdf.groupby("k", as_index=False)
+        .agg(
+            sd=("v", lambda s: s.drop_duplicates().sum()),
+            tot=("v", "sum"),
+        )
+        .sort_values("k")
+        .reset_index(drop=True)
+    )
+    for rank, got in results.items():
+        assert got["k"] == expected["k"].tolist()
+        assert got["sd"] == expected["sd"].tolist()
+        assert got["tot"] == pytest.approx(expected["tot"].tolist())
