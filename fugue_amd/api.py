@@ -8,7 +8,10 @@ Usage::
 # flake8: noqa
 from fugue_amd.dataset.api import (
     as_fugue_dataset,
+    as_local,
+    as_local_bounded,
     count,
+    get_num_partitions,
     is_bounded,
     is_empty,
     is_local,
@@ -38,6 +41,7 @@ from fugue_amd.dataframe.dataframe import as_fugue_df
 from fugue_amd.execution.api import (
     aggregate,
     anti_join,
+    as_fugue_engine_df,
     assign,
     broadcast,
     clear_global_engine,

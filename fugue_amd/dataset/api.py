@@ -34,3 +34,30 @@ def is_bounded(data: Any) -> bool:
 
 def is_empty(data: Any) -> bool:
     return as_fugue_dataset(data).empty
+
+
+def as_local(data: Any) -> Any:
+    """Convert the dataset to a local dataset (reference
+    ``fugue/dataset/api.py:39``); plugin-overridable via
+    ``as_local_bounded``."""
+    return as_local_bounded(data)
+
+
+def as_local_bounded(data: Any) -> Any:
+    """Convert the dataset to a local bounded dataset (reference
+    ``fugue/dataset/api.py:48``)."""
+    ok, res = try_run_plugin("as_local_bounded", data)
+    if ok:
+        return res
+    ds = as_fugue_dataset(data)
+    if hasattr(ds, "as_local_bounded"):
+        return ds.as_local_bounded()
+    raise NotImplementedError(
+        f"no registered function to convert {type(data)} to a local bounded dataset"
+    )
+
+
+def get_num_partitions(data: Any) -> int:
+    """Number of partitions of the dataset (reference
+    ``fugue/dataset/api.py:95``)."""
+    return as_fugue_dataset(data).num_partitions
