@@ -70,3 +70,61 @@ from fugue_amd.workflow._workflow_context import (  # noqa: E402
 )
 from fugue_amd.sql.api import fugue_sql, fugue_sql_flow  # noqa: E402
 from fugue_amd.sql.workflow import FugueSQLWorkflow  # noqa: E402
+
+# --- full reference top-level surface (reference fugue/__init__.py) --- #
+from fugue_amd.bag.bag import BagDisplay  # noqa: E402
+from fugue_amd.dataframe.dataframe import (  # noqa: E402
+    AnyDataFrame,
+    DataFrameDisplay,
+)
+from fugue_amd.dataset.api import as_fugue_dataset  # noqa: E402
+from fugue_amd.dataset.dataset import (  # noqa: E402
+    AnyDataset,
+    DatasetDisplay,
+    get_dataset_display,
+)
+from fugue_amd.execution.execution_engine import (  # noqa: E402
+    AnyExecutionEngine,
+    EngineFacet,
+)
+from fugue_amd.execution.native_execution_engine import (  # noqa: E402
+    PandasMapEngine,
+    PandasSQLEngine,
+)
+
+# the reference exports its qpd-backed pandas SQL engine under this name
+# (fugue/execution/native_execution_engine.py:42); here the same role is
+# played by the built-in pandas SQL executor
+QPDPandasEngine = PandasSQLEngine
+from fugue_amd.execution.factory import (  # noqa: E402
+    is_pandas_or,
+    make_sql_engine,
+    register_default_execution_engine,
+    register_default_sql_engine,
+    register_sql_engine,
+)
+from fugue_amd.extensions import (  # noqa: E402
+    cotransformer,
+    creator,
+    output_cotransformer,
+    output_transformer,
+    outputter,
+    processor,
+    register_creator,
+    register_output_transformer,
+    register_outputter,
+    register_processor,
+    register_transformer,
+    transformer,
+)
+from fugue_amd.rpc import (  # noqa: E402
+    EmptyRPCHandler,
+    RPCClient,
+    RPCFunc,
+    RPCHandler,
+    RPCServer,
+    make_rpc_server,
+    to_rpc_handler,
+)
+from fugue_amd.workflow.workflow import WorkflowDataFrames  # noqa: E402
+from fugue_amd.sql.api import fugue_sql_flow as fsql  # noqa: E402

@@ -10,6 +10,9 @@ from fugue_amd.utils.params import ParamDict
 from fugue_amd.utils.registry import run_plugin, try_run_plugin
 
 
+AnyDataset = Any  # anything convertible to a Dataset (reference fugue/dataset/dataset.py)
+
+
 class Dataset(ABC):
     def __init__(self):
         self._metadata: Optional[ParamDict] = None

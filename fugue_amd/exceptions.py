@@ -53,6 +53,10 @@ class FugueWorkflowRuntimeError(FugueWorkflowError):
     """Workflow runtime error"""
 
 
+class FugueSQLRuntimeError(FugueWorkflowRuntimeError):
+    """FugueSQL runtime error (reference ``fugue/exceptions.py:65``)"""
+
+
 class FugueWorkflowRuntimeValidationError(FugueWorkflowRuntimeError):
     """Validation error at runtime"""
 
