@@ -1156,8 +1156,10 @@ void launch_gb_part_scatter_staged(const int64_t* keys, const double* vals,
   if (blocks > MAX_GRID) blocks = MAX_GRID;
   if (blocks < 1) blocks = 1;
   dim3 g((int)blocks), b(BLOCK);
+  // ILP depth of the scatter row loop: 4 (default, measured 5.06 vs
+  // 5.43 ms/step on the 125M-row bench), 2 = paired
   const char* silp = std::getenv("FUGUE_SC_ILP");
-  bool squad = silp != nullptr && silp[0] == '4';
+  bool squad = silp == nullptr || silp[0] == '4';
   if (narrow) {
     auto* ok = (int32_t*)out_keys;
     if (squad)
