@@ -8,3 +8,4 @@ from fugue_amd.column.expressions import (
 )
 from fugue_amd.column import functions
 from fugue_amd.column.sql import SelectColumns, SQLExpressionGenerator
+from fugue_amd.column.functions import is_agg  # noqa: E402

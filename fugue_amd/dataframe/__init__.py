@@ -33,3 +33,14 @@ from fugue_amd.dataframe.utils import (
     parse_join_type,
     serialize_df,
 )
+from fugue_amd.dataframe.api import (  # noqa: E402
+    get_column_names,
+    rename,
+)
+from fugue_amd.dataframe.function_wrapper import (  # noqa: E402
+    fugue_annotated_param,
+    register_annotated_param,
+)
+from fugue_amd.dataframe.utils import (  # noqa: E402
+    normalize_dataframe_column_names,
+)

@@ -484,7 +484,9 @@ _PARAM_CLASSES: List[type] = []
 
 def register_annotated_param(cls: type, prepend: bool = False) -> type:
     """Register a custom AnnotatedParam subclass (plugin point; the HIP
-    engine registers device-frame param types through this)."""
+    engine registers device-frame param types through this).  Also
+    exposed as ``fugue_annotated_param`` for reference-name parity
+    (``fugue/dataframe/function_wrapper.py:151``)."""
     if prepend:
         _PARAM_CLASSES.insert(0, cls)
     else:
@@ -631,3 +633,6 @@ class DataFrameFunctionWrapper:
         if self._rt.code in "dlspq":
             return self._rt.to_output_df(rt, output_schema, ctx=ctx)
         return rt
+
+
+fugue_annotated_param = register_annotated_param

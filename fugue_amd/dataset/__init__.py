@@ -7,3 +7,4 @@ from fugue_amd.dataset.api import (
     is_local,
     show,
 )
+from fugue_amd.dataset.dataset import AnyDataset  # noqa: E402

@@ -22,3 +22,15 @@ from fugue_amd.execution.factory import (
     register_sql_engine,
     try_get_context_execution_engine,
 )
+from fugue_amd.execution.native_execution_engine import (  # noqa: E402
+    PandasMapEngine,
+    PandasSQLEngine,
+)
+
+# reference name for the pandas SQL facet (see fugue_amd/__init__.py)
+QPDPandasEngine = PandasSQLEngine
+from fugue_amd.execution.factory import (  # noqa: E402
+    infer_execution_engine,
+    parse_execution_engine,
+    parse_sql_engine,
+)

@@ -15,6 +15,13 @@ from fugue_amd.execution.execution_engine import (
     _GLOBAL_ENGINE,
 )
 from fugue_amd.utils.params import ParamDict
+from fugue_amd.utils.registry import ConditionalDispatcher
+
+# plugin points (reference ``parse_execution_engine`` factory.py:342 /
+# ``parse_sql_engine`` :508): candidates convert non-standard engine
+# descriptors (e.g. namespace tuples) before the built-in resolution
+parse_execution_engine = ConditionalDispatcher("parse_execution_engine")
+parse_sql_engine = ConditionalDispatcher("parse_sql_engine")
 
 _LOCK = threading.RLock()
 _ENGINE_REGISTRY: Dict[str, Callable[[Any], ExecutionEngine]] = {}
@@ -133,6 +140,9 @@ def make_execution_engine(
     for tp, func in items:
         if isinstance(engine, tp):
             return func(engine, merged)
+    ok, parsed = parse_execution_engine.run(engine, merged)
+    if ok:
+        return parsed
     raise ValueError(f"can't make execution engine from {engine!r}")
 
 
@@ -174,6 +184,9 @@ def make_sql_engine(
         return func(execution_engine, **kwargs)
     if isinstance(engine, type) and issubclass(engine, SQLEngine):
         return engine(execution_engine)
+    ok, parsed = parse_sql_engine.run(engine, execution_engine, **kwargs)
+    if ok:
+        return parsed
     raise ValueError(f"can't make sql engine from {engine!r}")
 
 

@@ -33,3 +33,14 @@ from fugue_amd.extensions.transformer.convert import (
     _to_output_transformer,
     _to_transformer,
 )
+from fugue_amd.extensions._utils import (  # noqa: E402
+    is_namespace_extension,
+    namespace_candidate,
+)
+from fugue_amd.extensions.creator.convert import parse_creator  # noqa: E402
+from fugue_amd.extensions.processor.convert import parse_processor  # noqa: E402
+from fugue_amd.extensions.outputter.convert import parse_outputter  # noqa: E402
+from fugue_amd.extensions.transformer.convert import (  # noqa: E402
+    parse_output_transformer,
+    parse_transformer,
+)

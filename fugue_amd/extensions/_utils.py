@@ -13,6 +13,35 @@ from fugue_amd.exceptions import (
 from fugue_amd.schema import Schema
 from fugue_amd.utils.interfaceless import parse_comment_annotation
 
+def is_namespace_extension(obj: Any) -> bool:
+    """Whether ``obj`` is a namespace extension reference: a 2-tuple of
+    (namespace str, payload) (reference ``fugue/extensions/_utils.py:14``)."""
+    return (
+        isinstance(obj, tuple)
+        and len(obj) == 2
+        and isinstance(obj[0], str)
+        and obj[0] != ""
+    )
+
+
+def namespace_candidate(
+    namespace: str, matcher: Callable[..., bool]
+) -> Callable[..., bool]:
+    """Build a plugin matcher that fires only for ``(namespace, payload)``
+    tuples whose payload passes ``matcher`` (reference
+    ``fugue/extensions/_utils.py:25``); used with the ``parse_*``
+    dispatchers to register e.g. ``("sparksql", "...")``-style creators."""
+
+    def _matcher(obj: Any, *args: Any, **kwargs: Any) -> bool:
+        return (
+            is_namespace_extension(obj)
+            and obj[0] == namespace
+            and matcher(obj[1], *args, **kwargs)
+        )
+
+    return _matcher
+
+
 _VALIDATION_KEYS = [
     "partitionby_has",
     "partitionby_is",

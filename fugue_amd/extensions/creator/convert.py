@@ -10,6 +10,7 @@ from fugue_amd.extensions.creator.creator import Creator
 from fugue_amd.schema import Schema
 from fugue_amd.utils.convert import to_function, to_instance
 from fugue_amd.utils.hash import to_uuid
+from fugue_amd.utils.registry import ConditionalDispatcher
 from fugue_amd.utils.interfaceless import parse_output_schema_from_comment
 
 _CREATOR_REGISTRY = ExtensionRegistry()
@@ -26,12 +27,19 @@ def creator(schema: Any = None) -> Callable[[Callable], "_FuncAsCreator"]:
     return deco
 
 
+# plugin point (reference ``parse_creator`` conditional dispatcher)
+parse_creator = ConditionalDispatcher("parse_creator")
+
+
 def _to_creator(
     obj: Any,
     schema: Any = None,
     global_vars: Optional[Dict[str, Any]] = None,
     local_vars: Optional[Dict[str, Any]] = None,
 ) -> Creator:
+    ok, parsed = parse_creator.run(obj)
+    if ok:
+        obj = parsed
     if isinstance(obj, str):
         reg = _CREATOR_REGISTRY.get(obj)
         if reg is not None:

@@ -13,6 +13,7 @@ from fugue_amd.extensions._utils import (
 from fugue_amd.extensions.outputter.outputter import Outputter
 from fugue_amd.utils.convert import to_function, to_instance
 from fugue_amd.utils.hash import to_uuid
+from fugue_amd.utils.registry import ConditionalDispatcher
 
 _OUTPUTTER_REGISTRY = ExtensionRegistry()
 
@@ -30,12 +31,19 @@ def outputter(**validation_rules: Any) -> Callable[[Callable], "_FuncAsOutputter
     return deco
 
 
+# plugin point (reference ``parse_outputter`` conditional dispatcher)
+parse_outputter = ConditionalDispatcher("parse_outputter")
+
+
 def _to_outputter(
     obj: Any,
     global_vars: Optional[Dict[str, Any]] = None,
     local_vars: Optional[Dict[str, Any]] = None,
     validation_rules: Optional[Dict[str, Any]] = None,
 ) -> Outputter:
+    ok, parsed = parse_outputter.run(obj)
+    if ok:
+        obj = parsed
     if isinstance(obj, str):
         reg = _OUTPUTTER_REGISTRY.get(obj)
         if reg is not None:

@@ -15,6 +15,7 @@ from fugue_amd.extensions.processor.processor import Processor
 from fugue_amd.schema import Schema
 from fugue_amd.utils.convert import to_function, to_instance
 from fugue_amd.utils.hash import to_uuid
+from fugue_amd.utils.registry import ConditionalDispatcher
 from fugue_amd.utils.interfaceless import parse_output_schema_from_comment
 
 _PROCESSOR_REGISTRY = ExtensionRegistry()
@@ -33,6 +34,10 @@ def processor(schema: Any = None, **validation_rules: Any) -> Callable[[Callable
     return deco
 
 
+# plugin point (reference ``parse_processor`` conditional dispatcher)
+parse_processor = ConditionalDispatcher("parse_processor")
+
+
 def _to_processor(
     obj: Any,
     schema: Any = None,
@@ -40,6 +45,9 @@ def _to_processor(
     local_vars: Optional[Dict[str, Any]] = None,
     validation_rules: Optional[Dict[str, Any]] = None,
 ) -> Processor:
+    ok, parsed = parse_processor.run(obj)
+    if ok:
+        obj = parsed
     if isinstance(obj, str):
         reg = _PROCESSOR_REGISTRY.get(obj)
         if reg is not None:

@@ -28,6 +28,7 @@ from fugue_amd.extensions.transformer.transformer import (
 )
 from fugue_amd.schema import Schema
 from fugue_amd.utils.convert import get_full_type_path, to_function, to_instance
+from fugue_amd.utils.registry import ConditionalDispatcher
 from fugue_amd.utils.hash import to_uuid
 from fugue_amd.utils.interfaceless import parse_output_schema_from_comment
 from fugue_amd.utils.params import ParamDict
@@ -122,6 +123,14 @@ def _to_general_transformer(
     )
 
 
+# plugin points (reference ``parse_transformer``/``parse_output_transformer``
+# conditional dispatchers, ``fugue/extensions/transformer/convert.py:27``):
+# backends add candidates (often with ``namespace_candidate``) that turn
+# non-standard objects into transformers before the default chain runs
+parse_transformer = ConditionalDispatcher("parse_transformer")
+parse_output_transformer = ConditionalDispatcher("parse_output_transformer")
+
+
 def _to_transformer(
     obj: Any,
     schema: Any = None,
@@ -129,6 +138,9 @@ def _to_transformer(
     local_vars: Optional[Dict[str, Any]] = None,
     validation_rules: Optional[Dict[str, Any]] = None,
 ) -> Union[Transformer, CoTransformer]:
+    ok, parsed = parse_transformer.run(obj)
+    if ok:
+        obj = parsed
     return _to_general_transformer(
         obj,
         schema,
@@ -152,6 +164,9 @@ def _to_output_transformer(
     local_vars: Optional[Dict[str, Any]] = None,
     validation_rules: Optional[Dict[str, Any]] = None,
 ) -> Union[Transformer, CoTransformer]:
+    ok, parsed = parse_output_transformer.run(obj)
+    if ok:
+        obj = parsed
     return _to_general_transformer(
         obj,
         None,

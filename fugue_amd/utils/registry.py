@@ -52,6 +52,31 @@ def try_run_plugin(name: str, *args: Any, **kwargs: Any) -> Tuple[bool, Any]:
         return False, None
 
 
+class ConditionalDispatcher:
+    """A named plugin point with the reference's dispatcher ergonomics
+    (``@parse_transformer.candidate(matcher)`` — triad
+    ``conditional_dispatcher`` in ``fugue/_utils/registry.py:9``),
+    backed by this module's registry."""
+
+    def __init__(self, name: str):
+        self.name = name
+
+    def candidate(
+        self, matcher: Callable[..., bool], priority: float = 1.0
+    ) -> Callable[[Callable], Callable]:
+        def deco(func: Callable) -> Callable:
+            register_plugin(self.name, matcher, func, priority)
+            return func
+
+        return deco
+
+    def run(self, *args: Any, **kwargs: Any) -> Tuple[bool, Any]:
+        return try_run_plugin(self.name, *args, **kwargs)
+
+    def __call__(self, *args: Any, **kwargs: Any) -> Any:
+        return run_plugin(self.name, *args, **kwargs)
+
+
 def fugue_plugin(name: str, priority: float = 1.0):
     """Decorator form: ``@fugue_plugin("parse_execution_engine")`` with a
     ``matcher`` attribute on the function, or pass matcher via this factory."""

@@ -395,3 +395,28 @@ def test_zip_to_file_threshold(tmp_path):
 
     res = e.comap(z, cm, "n1:long,n2:long", PartitionSpec())
     assert res.as_array() == [[50, 30]]
+
+
+def test_namespace_extension_plugins():
+    """parse_creator + namespace_candidate plugin point (reference
+    fugue/plugins.py dispatchers): a ("myns", payload) tuple resolves
+    to a creator through a registered candidate."""
+    from fugue_amd.extensions import namespace_candidate, parse_creator
+    from fugue_amd.workflow.workflow import FugueWorkflow
+
+    @parse_creator.candidate(
+        namespace_candidate("myns", lambda x: isinstance(x, str))
+    )
+    def _parse(obj):
+        n = int(obj[1])
+
+        # schema: a:int
+        def create() -> List[List[Any]]:
+            return [[n]]
+
+        return create
+
+    dag = FugueWorkflow()
+    dag.create(("myns", "7")).yield_dataframe_as("r")
+    res = dag.run()
+    assert res["r"].result.as_array() == [[7]]
