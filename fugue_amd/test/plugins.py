@@ -106,6 +106,38 @@ def with_backend(*names: str) -> Callable:
     return deco
 
 
+class FugueTestSuite:
+    """Base class for user test suites bound to a backend with
+    :func:`fugue_test_suite` (reference ``fugue/test/plugins.py:139``):
+    exposes ``engine`` (lazily built from the bound backend) and
+    ``df_eq``."""
+
+    backend: str = ""
+    __test__ = False
+    _engine: Any = None
+
+    @property
+    def engine(self) -> Any:
+        if type(self)._engine is None:
+            type(self)._engine = type(self).make_engine()  # type: ignore
+        return type(self)._engine
+
+    def df_eq(self, *args: Any, **kwargs: Any) -> bool:
+        return _df_eq(*args, **kwargs)
+
+
+def extract_conf(
+    conf: Dict[str, Any], prefix: str, remove_prefix: bool
+) -> Dict[str, Any]:
+    """Extract config entries under ``prefix`` (reference
+    ``fugue/test/plugins.py:315``)."""
+    res: Dict[str, Any] = {}
+    for k, v in conf.items():
+        if k.startswith(prefix):
+            res[k[len(prefix):] if remove_prefix else k] = v
+    return res
+
+
 def fugue_test_suite(backend: str, mark_test: Optional[Any] = None) -> Callable:
     """Class decorator binding a conformance suite to a backend."""
 
@@ -120,7 +152,11 @@ def fugue_test_suite(backend: str, mark_test: Optional[Any] = None) -> Callable:
                 )
 
         cls.make_engine = make_engine  # type: ignore
-        cls.df_eq = staticmethod(_df_eq)  # type: ignore
+        if not hasattr(cls, "df_eq"):
+            cls.df_eq = staticmethod(_df_eq)  # type: ignore
+        cls.backend = backend  # type: ignore
+        if isinstance(cls, type) and issubclass(cls, FugueTestSuite):
+            cls.__test__ = True  # re-enable collection for bound suites
         if mark_test is not None:
             cls = mark_test(cls)
         return cls

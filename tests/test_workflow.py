@@ -420,3 +420,23 @@ def test_namespace_extension_plugins():
     dag.create(("myns", "7")).yield_dataframe_as("r")
     res = dag.run()
     assert res["r"].result.as_array() == [[7]]
+
+
+def test_fugue_test_suite_base():
+    """ft.FugueTestSuite + fugue_test_suite binding (reference
+    fugue/test/plugins.py:139)."""
+    import fugue_amd.test as ft
+
+    @ft.fugue_test_suite("pandas")
+    class MySuite(ft.FugueTestSuite):
+        pass
+
+    s = MySuite()
+    assert s.backend == "pandas"
+    assert s.engine is s.engine  # lazily built, cached
+    from fugue_amd import PandasDataFrame
+
+    assert s.df_eq(
+        PandasDataFrame(pd.DataFrame(dict(a=[1]))), [[1]], "a:long"
+    )
+    assert ft.extract_conf({"x.a": 1, "y.b": 2}, "x.", True) == {"a": 1}
