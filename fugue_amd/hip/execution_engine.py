@@ -1457,6 +1457,10 @@ class HipExecutionEngine(ExecutionEngine):
                 continue
             raise DeviceExprError("compound aggregate expression: fallback")
         if len(key_names) == 0:
+            if any(k.endswith("_distinct") for _n, k, _i in plans):
+                return self._device_global_aggregate_distinct(
+                    d, plans, partials, derived, having, cols
+                )
             return self._device_global_aggregate(
                 d, plans, partials, derived, having, cols
             )
@@ -1590,6 +1594,77 @@ class HipExecutionEngine(ExecutionEngine):
         if tp is None:
             raise DeviceExprError(f"no arrow type for {data.dtype}")
         return tp
+
+    def _device_global_aggregate_distinct(
+        self,
+        d: HipDataFrame,
+        plans: List[Tuple[str, str, Any]],
+        partials: List[Tuple[str, int, str]],
+        derived: Dict[str, DeviceColumn],
+        having: Optional[ColumnExpr],
+        cols: SelectColumns,
+    ) -> DataFrame:
+        """Keyless COUNT/SUM/AVG(DISTINCT x): global dedupe of each source
+        column, then a direct reduction (allreduce-merged), merged with
+        the non-distinct aggregates' 1-row result.  (MIN/MAX DISTINCT are
+        planned as plain MIN/MAX upstream.)"""
+        if cols.is_distinct:
+            raise DeviceExprError("distinct aggregate: fallback")
+        cd_vals: Dict[str, Any] = {}
+        for name, kind, src in plans:
+            if not kind.endswith("_distinct"):
+                continue
+            func = kind[: -len("_distinct")]
+            dedup = self.to_df(
+                self.distinct(self._device_select_named(d, [src]))
+            )
+            vcol = dedup.col(src)
+            if isinstance(vcol, StringDeviceColumn):
+                raise DeviceExprError("string distinct aggregate: fallback")
+            data = vcol.data.to(torch.float64)
+            if vcol.valid is not None:
+                data = data[vcol.valid]
+            s = float(data.sum().item()) if data.numel() > 0 else 0.0
+            n = int(data.numel())
+            if self.is_distributed:
+                import torch.distributed as dist
+
+                t = torch.tensor([s, float(n)], dtype=torch.float64)
+                t = t.to(self._comm._comm_device(t))
+                dist.all_reduce(t, op=dist.ReduceOp.SUM)
+                s, n = float(t.cpu()[0].item()), int(t.cpu()[1].item())
+            if func == "count":
+                cd_vals[name] = n
+            elif func == "sum":
+                cd_vals[name] = s if n > 0 else None
+            else:  # avg
+                cd_vals[name] = (s / n) if n > 0 else None
+        base_plans = [p for p in plans if not p[1].endswith("_distinct")]
+        base_row: Dict[str, Any] = {}
+        if base_plans:
+            base = self._device_global_aggregate(
+                d, base_plans, partials, derived, None, cols
+            )
+            bp = base.as_pandas()
+            if len(bp) > 0:
+                base_row = bp.iloc[0].to_dict()
+        row = {}
+        for name, kind, _i in plans:
+            if name in cd_vals:
+                row[name] = cd_vals[name]
+            else:
+                # typed placeholder on non-root ranks (frame is emptied
+                # below) keeps the inferred schema rank-consistent
+                dflt = 0 if kind in ("count", "rowcount") else 0.0
+                row[name] = base_row.get(name, dflt)
+        res = pd.DataFrame([row])
+        if self.is_distributed and self.rank != 0:
+            res = res.head(0)
+        if having is not None:
+            from fugue_amd.column.interpreter import eval_filter
+
+            res = eval_filter(res, having)
+        return self.to_df(PandasDataFrame(res), shard_replicated=False)
 
     def _device_global_aggregate(
         self,

@@ -189,3 +189,30 @@ def test_hip_cpu_like_general_patterns():
             "i"
         ].sort_values().tolist()
         assert got == exp, (p, got[:10], exp[:10])
+
+
+def test_hip_cpu_global_distinct_aggregates():
+    """Keyless SUM/AVG/COUNT DISTINCT on the device engine (CPU
+    tensors) vs pandas comparator."""
+    import numpy as np
+    import pandas as pd
+
+    import fugue_amd.api as fa
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    rng = np.random.default_rng(21)
+    pdf = pd.DataFrame(dict(v=rng.integers(0, 9, 5000).astype("f8"),
+                            w=rng.random(5000)))
+    r = fa.as_pandas(
+        fa.fugue_sql(
+            "SELECT SUM(DISTINCT v) AS s, AVG(DISTINCT v) AS a, "
+            "COUNT(DISTINCT v) AS c, SUM(w) AS tw FROM pdf",
+            engine=e,
+        )
+    )
+    dd = pdf["v"].drop_duplicates()
+    assert float(r["s"][0]) == dd.sum()
+    assert abs(float(r["a"][0]) - dd.mean()) < 1e-9
+    assert int(r["c"][0]) == dd.nunique()
+    assert abs(float(r["tw"][0]) - pdf["w"].sum()) < 1e-6
