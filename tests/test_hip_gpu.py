@@ -418,3 +418,20 @@ def test_like_general_gpu(engine):
             "i"
         ].sort_values().tolist()
         assert got == exp, p
+
+
+def test_global_distinct_aggregates_gpu(engine):
+    """Keyless SUM/AVG/COUNT DISTINCT on device vs pandas."""
+    rng = np.random.default_rng(21)
+    pdf = pd.DataFrame(dict(v=rng.integers(0, 9, 50_000).astype("f8")))
+    r = fa.as_pandas(
+        fa.fugue_sql(
+            "SELECT SUM(DISTINCT v) AS s, AVG(DISTINCT v) AS a, "
+            "COUNT(DISTINCT v) AS c FROM pdf",
+            engine=engine,
+        )
+    )
+    dd = pdf["v"].drop_duplicates()
+    assert float(r["s"][0]) == dd.sum()
+    assert abs(float(r["a"][0]) - dd.mean()) < 1e-9
+    assert int(r["c"][0]) == dd.nunique()
