@@ -18,8 +18,10 @@
 
 #include <hip/hip_runtime.h>
 #include <algorithm>
+#include <cstdio>
 #include <cstdlib>
 #include <cstdint>
+#include <cstring>
 
 #define WAVE 64
 #define BLOCK 256
@@ -1146,6 +1148,25 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel_v1(
 
 extern "C" {
 
+// Strict ILP env parsing: exact "1"/"2"/"4" only; anything else (typos,
+// "40", empty) warns once and selects the default depth.
+static int parse_ilp_env(const char* name, int dflt) {
+  const char* v = std::getenv(name);
+  if (v == nullptr) return dflt;
+  if (std::strcmp(v, "1") == 0) return 1;
+  if (std::strcmp(v, "2") == 0) return 2;
+  if (std::strcmp(v, "4") == 0) return 4;
+  static bool warned = false;
+  if (!warned) {
+    std::fprintf(stderr,
+                 "fugue_amd: unrecognized %s='%s' (expected 1/2/4); using "
+                 "default %d\n",
+                 name, v, dflt);
+    warned = true;
+  }
+  return dflt;
+}
+
 void launch_gb_part_scatter_staged(const int64_t* keys, const double* vals,
                                    int64_t n, int shift, int64_t* cursor,
                                    void* out_keys, double* out_vals,
@@ -1158,11 +1179,14 @@ void launch_gb_part_scatter_staged(const int64_t* keys, const double* vals,
   dim3 g((int)blocks), b(BLOCK);
   // ILP depth of the scatter row loop: 4 (default, measured 5.06 vs
   // 5.43 ms/step on the 125M-row bench), 2 = paired
-  const char* silp = std::getenv("FUGUE_SC_ILP");
-  bool squad = silp == nullptr || silp[0] == '4';
+  bool squad = parse_ilp_env("FUGUE_SC_ILP", 4) == 4;
   if (narrow) {
     auto* ok = (int32_t*)out_keys;
-    if (squad)
+    if (squad && nt)
+      hipLaunchKernelGGL((gb_part_scatter_staged_kernel_v4<true, int32_t>),
+                         g, b, 0, stream, keys, vals, n, shift, cursor, ok,
+                         out_vals, chunk, ovf);
+    else if (squad)
       hipLaunchKernelGGL((gb_part_scatter_staged_kernel_v4<false, int32_t>),
                          g, b, 0, stream, keys, vals, n, shift, cursor, ok,
                          out_vals, chunk, ovf);
@@ -1176,7 +1200,11 @@ void launch_gb_part_scatter_staged(const int64_t* keys, const double* vals,
                          out_vals, chunk, ovf);
   } else {
     auto* ok = (int64_t*)out_keys;
-    if (squad)
+    if (squad && nt)
+      hipLaunchKernelGGL((gb_part_scatter_staged_kernel_v4<true, int64_t>),
+                         g, b, 0, stream, keys, vals, n, shift, cursor, ok,
+                         out_vals, chunk, nullptr);
+    else if (squad)
       hipLaunchKernelGGL((gb_part_scatter_staged_kernel_v4<false, int64_t>),
                          g, b, 0, stream, keys, vals, n, shift, cursor, ok,
                          out_vals, chunk, nullptr);
@@ -1205,12 +1233,16 @@ void launch_gb_aggregate_part_big(const void* part_keys,
   dim3 g((int)blocks), b(BLOCK);
   // ILP depth of the row loop: 4 (default, measured 5.55 vs 5.63
   // ms/step on the 125M-row bench), 2 = paired, 1 = legacy scalar
-  const char* ilp = std::getenv("FUGUE_GB_ILP");
-  bool legacy = ilp != nullptr && ilp[0] == '1';
-  bool quad = ilp == nullptr || ilp[0] == '4';
+  int ilp_d = parse_ilp_env("FUGUE_GB_ILP", 4);
+  bool legacy = ilp_d == 1;
+  bool quad = ilp_d == 4;
   if (narrow) {
     auto* pk = (const int32_t*)part_keys;
-    if (quad)
+    if (quad && nt)
+      hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v4<true, int32_t>), g,
+                         b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
+                         gcount, tsize, chunk);
+    else if (quad)
       hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v4<false, int32_t>), g,
                          b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
                          gcount, tsize, chunk);
@@ -1228,7 +1260,11 @@ void launch_gb_aggregate_part_big(const void* part_keys,
                          gcount, tsize, chunk);
   } else {
     auto* pk = (const int64_t*)part_keys;
-    if (quad)
+    if (quad && nt)
+      hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v4<true, int64_t>), g,
+                         b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
+                         gcount, tsize, chunk);
+    else if (quad)
       hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v4<false, int64_t>), g,
                          b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
                          gcount, tsize, chunk);

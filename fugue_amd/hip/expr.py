@@ -206,6 +206,9 @@ def _eval_like(
             else lengths == 0
         )
         return res, c.valid
+    if int(c.bytes.numel()) == 0:
+        # every row is the empty string; a non-empty literal never matches
+        return torch.zeros(n, dtype=torch.bool, device=device), c.valid
     patt = torch.tensor(list(kb), dtype=torch.uint8, device=device)
     long_enough = lengths >= k
     if not prefix_any:  # 'core%' or exact: compare k bytes at row start
@@ -260,7 +263,13 @@ def _eval_like_general(
     column holds non-ASCII data (multi-byte characters)."""
     n = len(c)
     total = int(c.bytes.numel())
-    if "_" in pat and total > 0 and int(c.bytes.max().item()) >= 0x80:
+    if total == 0:
+        # every row is the empty string: the pattern matches iff it has
+        # no literal bytes and no ``_`` (i.e. it is all ``%`` / empty)
+        empty_ok = pat.replace("%", "") == ""
+        res = torch.full((n,), empty_ok, dtype=torch.bool, device=device)
+        return res, c.valid
+    if "_" in pat and int(c.bytes.max().item()) >= 0x80:
         raise DeviceExprError("LIKE _ on non-ASCII data")
     starts = c.offsets[:-1]
     ends = c.offsets[1:]

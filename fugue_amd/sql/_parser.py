@@ -483,9 +483,17 @@ class FugueSQLParser:
             if nxt is not None and nxt.kind == "PUNCT" and nxt.value == ":":
                 key = ts.next().value
                 ts.next()  # ':'
-                named[key] = self._maybe_index(
-                    self.hooks.get_var(ts.next().value)
-                )
+                vt = ts.next()
+                if vt is None or vt.kind != "NAME" or not self.hooks.has_var(
+                    vt.value
+                ):
+                    where = f" at {vt.pos}" if vt is not None else ""
+                    raise FugueSQLSyntaxError(
+                        "unknown dataframe variable "
+                        f"{vt.value if vt is not None else '<eof>'!r} "
+                        f"for SUB input {key!r}{where}"
+                    )
+                named[key] = self._maybe_index(self.hooks.get_var(vt.value))
             elif self.hooks.has_var(t.value):
                 ordered.append(self._maybe_index(self.hooks.get_var(ts.next().value)))
             else:

@@ -111,6 +111,11 @@ def _to_column_expr(e: X.Expr, schema: Schema, alias_map: Dict[str, str]) -> Col
                 "FIRST": F.first,
                 "LAST": F.last,
             }[name]
+            if e.distinct and name in ("FIRST", "LAST"):
+                # FIRST/LAST over a deduplicated argument has no device
+                # decomposition and silently dropping the qualifier can
+                # change the answer — route to the host executor instead
+                raise UnsupportedPlan(f"{name}(DISTINCT ...)")
             res = fn(inner)
             if e.distinct and name in ("SUM", "AVG", "MIN", "MAX"):
                 from fugue_amd.column.expressions import _UnaryAggFuncExpr

@@ -216,3 +216,64 @@ def test_hip_cpu_global_distinct_aggregates():
     assert abs(float(r["a"][0]) - dd.mean()) < 1e-9
     assert int(r["c"][0]) == dd.nunique()
     assert abs(float(r["tw"][0]) - pdf["w"].sum()) < 1e-6
+
+
+def test_hip_cpu_like_all_empty_strings():
+    """LIKE over a column where every row is the empty string: the byte
+    buffer is empty and the simple/general matchers must not index into
+    it (r01 advisor finding — IndexError on index_select)."""
+    import pandas as pd
+
+    import fugue_amd.api as fa
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    pdf = pd.DataFrame(dict(s=["", "", ""], i=[0, 1, 2]))
+    for p, expect in [
+        ("a", []), ("%a%", []), ("a%", []), ("%a", []),
+        ("%", [0, 1, 2]), ("%%", [0, 1, 2]), ("", [0, 1, 2]),
+        ("_", []), ("a_c", []), ("%_%", []), ("a%c%e", []),
+    ]:
+        got = fa.as_pandas(
+            fa.fugue_sql(
+                f"SELECT i FROM pdf WHERE s LIKE '{p}'", engine=e
+            )
+        )["i"].sort_values().tolist()
+        assert got == expect, (p, got)
+
+
+def test_sql_first_last_distinct_not_dropped():
+    """FIRST(DISTINCT x)/LAST(DISTINCT x) must not silently plan as
+    plain FIRST/LAST on the device path (r01 advisor finding): the
+    planner refuses the distinct qualifier and the host executor
+    computes it over the deduplicated sequence (LAST over [1,2,1]
+    distinct-ordered [1,2] is 2, not 1)."""
+    import pandas as pd
+
+    import fugue_amd.api as fa
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    pdf = pd.DataFrame(dict(k=[1, 1, 1], v=[1, 2, 1]))
+    r = fa.as_pandas(
+        fa.fugue_sql(
+            "SELECT k, LAST(DISTINCT v) AS lv, FIRST(DISTINCT v) AS fv "
+            "FROM pdf GROUP BY k",
+            engine=e,
+        )
+    )
+    assert int(r["lv"][0]) == 2
+    assert int(r["fv"][0]) == 1
+
+
+def test_sql_sub_unknown_named_input_errors():
+    """SUB with an unknown named input variable raises a syntax error
+    naming the variable, not a bare KeyError (r01 advisor finding)."""
+    import pytest as _pytest
+
+    import fugue_amd.api as fa
+    from fugue_amd.exceptions import FugueSQLSyntaxError
+
+    with _pytest.raises(FugueSQLSyntaxError) as ei:
+        fa.fugue_sql_flow("SUB a:unknown_df USING mymod")
+    assert "unknown_df" in str(ei.value)
