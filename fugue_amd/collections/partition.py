@@ -13,6 +13,53 @@ from fugue_amd.utils.params import ParamDict
 
 _VALID_ALGOS = ("", "default", "hash", "rand", "even", "coarse")
 
+# canonical field names for the spec-dict aliases users may pass
+_FIELD_ALIASES = {"by": "partition_by", "num": "num_partitions"}
+
+
+def _spec_shorthand(a: Any) -> Optional[Dict[str, Any]]:
+    """Interpret the one-positional-argument shorthands
+    (``PartitionSpec(4)``, ``PartitionSpec("a")``, ``PartitionSpec(["a"])``,
+    ``PartitionSpec("per_row")``); returns None when ``a`` is a full spec
+    (dict / JSON string / PartitionSpec) that the merge path handles."""
+    if isinstance(a, bool):
+        return None
+    if isinstance(a, int):
+        return {"num_partitions": str(a)}
+    if isinstance(a, (list, tuple)):
+        return {"partition_by": list(a)}
+    if isinstance(a, str) and not a.startswith("{"):
+        if a.lower() == "per_row":
+            return {"algo": "even", "num_partitions": "ROWCOUNT"}
+        return {"partition_by": [a]} if a != "" else {}
+    return None
+
+
+def _merge_spec_sources(
+    args: Tuple[Any, ...], kwargs: Dict[str, Any]
+) -> Dict[str, Any]:
+    """Fold positional spec sources (PartitionSpec / dict / JSON string)
+    left-to-right, then keyword overrides, normalizing field aliases."""
+    merged: Dict[str, Any] = {}
+
+    def absorb(u: Dict[str, Any]) -> None:
+        for k, v in u.items():
+            merged[_FIELD_ALIASES.get(k, k)] = v
+
+    for a in args:
+        if a is None:
+            continue
+        if isinstance(a, PartitionSpec):
+            absorb(a.jsondict)
+        elif isinstance(a, dict):
+            absorb(a)
+        elif isinstance(a, str):
+            absorb(json.loads(a))
+        else:
+            raise TypeError(f"{a} is not supported by PartitionSpec")
+    absorb(kwargs)
+    return merged
+
 
 def parse_presort_exp(presort: Any) -> Dict[str, bool]:
     """Parse ``"b desc, c asc"`` (or pair lists) into an ordered
@@ -63,57 +110,31 @@ class PartitionSpec:
     presort.  See reference docs for the user-facing semantics."""
 
     def __init__(self, *args: Any, **kwargs: Any):
-        p: Dict[str, Any] = {}
-        if len(args) == 1 and len(kwargs) == 0:
-            a = args[0]
-            if isinstance(a, str):
-                if a.lower() == "per_row":
-                    p["algo"] = "even"
-                    p["num_partitions"] = "ROWCOUNT"
-                elif not a.startswith("{") and a != "":
-                    p["partition_by"] = [a]
-            elif isinstance(a, int):
-                p["num_partitions"] = str(a)
-            elif isinstance(a, (list, tuple)):
-                p["partition_by"] = list(a)
-        if len(p) == 0:
-            for a in args:
-                if a is None:
-                    continue
-                if isinstance(a, PartitionSpec):
-                    self._update_dict(p, a.jsondict)
-                elif isinstance(a, dict):
-                    self._update_dict(p, a)
-                elif isinstance(a, str):
-                    self._update_dict(p, json.loads(a))
-                else:
-                    raise TypeError(f"{a} is not supported by PartitionSpec")
-            self._update_dict(p, kwargs)
+        p: Optional[Dict[str, Any]] = (
+            _spec_shorthand(args[0])
+            if len(args) == 1 and len(kwargs) == 0
+            else None
+        )
+        if not p:  # no shorthand matched (or it was empty): merge sources
+            p = _merge_spec_sources(args, kwargs)
+        self._init_fields(p)
+
+    def _init_fields(self, p: Dict[str, Any]) -> None:
         self._num_partitions = str(p.get("num_partitions", "0"))
         self._algo = str(p.get("algo", "")).lower()
         if self._algo not in _VALID_ALGOS:
             raise SyntaxError(f"invalid partition algo {self._algo!r}")
         by = p.get("partition_by", [])
-        if isinstance(by, str):
-            by = [by]
-        self._partition_by: List[str] = list(by)
+        self._partition_by: List[str] = [by] if isinstance(by, str) else list(by)
         if len(self._partition_by) != len(set(self._partition_by)):
             raise SyntaxError(f"{self._partition_by} has duplicated keys")
         self._presort = parse_presort_exp(p.get("presort", None))
-        if any(x in self._presort for x in self._partition_by):
+        overlap = set(self._presort) & set(self._partition_by)
+        if overlap:
             raise SyntaxError(
                 f"partition by overlaps with presort: "
                 f"{self._partition_by}, {list(self._presort)}"
             )
-
-    @staticmethod
-    def _update_dict(d: Dict[str, Any], u: Dict[str, Any]) -> None:
-        for k, v in u.items():
-            if k == "by":
-                k = "partition_by"
-            if k == "num":
-                k = "num_partitions"
-            d[k] = v
 
     def __repr__(self) -> str:
         return (

@@ -1,17 +1,10 @@
 """Built-in outputters (reference: ``fugue/extensions/_builtins/outputters.py``)."""
-from typing import Any, List, Type
-
-from fugue_amd.collections.partition import PartitionCursor
-from fugue_amd.dataframe.dataframe import DataFrame, LocalDataFrame
 from fugue_amd.dataframe.dataframes import DataFrames
 from fugue_amd.dataframe.utils import _df_eq
 from fugue_amd.exceptions import FugueWorkflowError
 from fugue_amd.extensions.outputter.outputter import Outputter
 from fugue_amd.extensions.transformer.convert import _to_output_transformer
-from fugue_amd.extensions.transformer.transformer import CoTransformer, Transformer
-from fugue_amd.rpc import EmptyRPCHandler, to_rpc_handler
-from fugue_amd.utils.convert import to_type
-from fugue_amd.utils.params import ParamDict
+from fugue_amd.extensions.transformer.transformer import Transformer
 
 
 class Show(Outputter):
@@ -64,72 +57,54 @@ class Save(Outputter):
 
 
 class RunOutputTransformer(Outputter):
+    """Runs an output (co)transformer for side effects: same execution
+    shape as ``RunTransformer`` but the mapped result is forced to
+    materialize via an eager persist and then discarded."""
+
     def process(self, dfs: DataFrames) -> None:
-        df = dfs[0]
-        tf = _to_output_transformer(
-            self.params.get_or_none("transformer", object),
+        from fugue_amd.extensions._builtins.processors import (
+            _comap_empty_inputs,
+            _CoTransformerRunner,
+            _prepare_transformer,
+            _TransformerRunner,
         )
-        tf._workflow_conf = self.execution_engine.conf
-        tf._params = ParamDict(self.params.get("params", ParamDict()))
-        tf._partition_spec = self.partition_spec
-        rpc_handler = to_rpc_handler(self.params.get_or_throw("rpc_handler", object))
-        if not isinstance(rpc_handler, EmptyRPCHandler):
-            tf._rpc_client = self.rpc_server.make_client(rpc_handler)
-            tf._has_rpc_client = True
-        ie = self.params.get("ignore_errors", [])
-        self._ignore_errors = [to_type(x, Exception) for x in ie]
-        tf.validate_on_compile()
-        tf.validate_on_runtime(df)
-        if isinstance(tf, Transformer):
-            self.transform(df, tf)
-        else:
-            self.cotransform(df, tf)
-
-    def transform(self, df: DataFrame, tf: Transformer) -> None:
-        from fugue_amd.extensions._builtins.processors import _TransformerRunner
-        from fugue_amd.schema import Schema
-
-        tf._key_schema = self.partition_spec.get_key_schema(df.schema)
-        tf._output_schema = Schema(tf.get_output_schema(df))
-        tr = _TransformerRunner(df, tf, self._ignore_errors)
-        self.execution_engine.persist(
-            self.execution_engine.map_engine.map_dataframe(
-                df=df,
-                map_func=tr.run,
-                output_schema=tf.output_schema,
-                partition_spec=tf.partition_spec,
-                on_init=tr.on_init,
-            ),
-            lazy=False,
-        )
-
-    def cotransform(self, df: DataFrame, tf: CoTransformer) -> None:
-        from fugue_amd.dataframe.array_dataframe import ArrayDataFrame
         from fugue_amd.execution.execution_engine import (
             _FUGUE_SERIALIZED_BLOB_SCHEMA,
         )
-        from fugue_amd.extensions._builtins.processors import _CoTransformerRunner
         from fugue_amd.schema import Schema
 
-        if not df.metadata.get("serialized", False):
-            raise FugueWorkflowError("must use serialized (zipped) dataframe")
-        tf._key_schema = df.schema - _FUGUE_SERIALIZED_BLOB_SCHEMA
-        schemas = df.metadata["schemas"]
-        named = df.metadata.get("serialized_has_name", False)
-        empty_dfs = (
-            DataFrames({k: ArrayDataFrame([], v) for k, v in schemas.items()})
-            if named
-            else DataFrames([ArrayDataFrame([], v) for v in schemas.values()])
+        df = dfs[0]
+        tf, ignorable = _prepare_transformer(
+            self, convert=_to_output_transformer, with_schema=False
         )
-        tf._output_schema = Schema(tf.get_output_schema(empty_dfs))
-        tr = _CoTransformerRunner(df, tf, self._ignore_errors)
-        self.execution_engine.persist(
-            self.execution_engine.comap(
+        tf.validate_on_runtime(df)
+        eng = self.execution_engine
+        if isinstance(tf, Transformer):
+            tf._key_schema = self.partition_spec.get_key_schema(df.schema)
+            tf._output_schema = Schema(tf.get_output_schema(df))
+            runner = _TransformerRunner(df, tf, list(ignorable))
+            mapped = eng.map_engine.map_dataframe(
                 df=df,
-                map_func=tr.run,
+                map_func=runner.run,
                 output_schema=tf.output_schema,
                 partition_spec=tf.partition_spec,
-                on_init=tr.on_init,
-            ),
-            lazy=False,
-        )
+                on_init=runner.on_init,
+            )
+        else:
+            if not df.metadata.get("serialized", False):
+                raise FugueWorkflowError(
+                    "must use serialized (zipped) dataframe"
+                )
+            tf._key_schema = df.schema - _FUGUE_SERIALIZED_BLOB_SCHEMA
+            tf._output_schema = Schema(
+                tf.get_output_schema(_comap_empty_inputs(df))
+            )
+            co_runner = _CoTransformerRunner(df, tf, list(ignorable))
+            mapped = eng.comap(
+                df=df,
+                map_func=co_runner.run,
+                output_schema=tf.output_schema,
+                partition_spec=tf.partition_spec,
+                on_init=co_runner.on_init,
+            )
+        eng.persist(mapped, lazy=False)
