@@ -176,6 +176,40 @@ class HipMapEngine(MapEngine):
                 n_local_parts = max(
                     1, (total + engine.world_size - 1) // engine.world_size
                 )
+            if map_func_format_hint == "device":
+                # device-resident UDF on physical partitions: slice the
+                # shard (usually 1 slice per rank) without leaving HBM
+                n_rows = local.count()
+                dev_results: List[HipDataFrame] = []
+                bounds_d = [
+                    (n_rows * i) // n_local_parts
+                    for i in range(n_local_parts + 1)
+                ]
+                for p in range(n_local_parts):
+                    start, end = bounds_d[p], bounds_d[p + 1]
+                    if end <= start:
+                        continue
+                    sub = local.slice_rows(start, end - start)
+                    cursor.set(
+                        lambda: sub.peek_array(),
+                        engine.rank * n_local_parts + p,
+                        0,
+                    )
+                    res = map_func(cursor, sub)
+                    d = engine.to_df(res)
+                    if not isinstance(d, HipDataFrame):
+                        d = HipDataFrame(
+                            res.as_arrow(), output_schema, engine._device
+                        )
+                    dev_results.append(d)
+                if len(dev_results) == 0:
+                    return engine.to_df(
+                        PandasDataFrame(
+                            output_schema.create_empty_pandas(), output_schema
+                        ),
+                        shard_replicated=False,
+                    )
+                return dev_results[0].concat_with(dev_results[1:])
             pdf_local = local.as_pandas()
             if len(pdf_local) > 0:
                 import numpy as _np

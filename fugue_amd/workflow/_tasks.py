@@ -46,6 +46,7 @@ class FugueTask:
         self._executed = False
         self._traceback = None
         self.name = ""
+        self._uuid: Optional[str] = None
 
     def reset_traceback(self, limit: int, should_prune: Optional[Callable] = None) -> None:
         """Capture the user's compile-time call site for error splicing
@@ -69,16 +70,20 @@ class FugueTask:
         return self._inputs
 
     def __uuid__(self) -> str:
-        return to_uuid(
-            str(type(self).__name__),
-            self._extension.__uuid__()
-            if hasattr(self._extension, "__uuid__")
-            else str(self._extension),
-            dict(self._params),
-            self._partition_spec,
-            [t.__uuid__() for t in self._inputs],
-            self._input_names,
-        )
+        # identity fields are fixed at construction, so the spec UUID is
+        # computed once (the DAG recomputes uuids heavily for dedup)
+        if self._uuid is None:
+            self._uuid = to_uuid(
+                str(type(self).__name__),
+                self._extension.__uuid__()
+                if hasattr(self._extension, "__uuid__")
+                else str(self._extension),
+                dict(self._params),
+                self._partition_spec,
+                [t.__uuid__() for t in self._inputs],
+                self._input_names,
+            )
+        return self._uuid
 
     def set_checkpoint(self, checkpoint: Checkpoint) -> "FugueTask":
         self._checkpoint = checkpoint

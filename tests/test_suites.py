@@ -277,3 +277,63 @@ def test_sql_sub_unknown_named_input_errors():
     with _pytest.raises(FugueSQLSyntaxError) as ei:
         fa.fugue_sql_flow("SUB a:unknown_df USING mymod")
     assert "unknown_df" in str(ei.value)
+
+
+def test_keyless_device_transform():
+    """A HipDataFrame-annotated transformer with no partition keys must
+    stay on the device path (no pandas round trip) — the bench's
+    TRANSFORM stage shape.  CPU tensors here; same code path on GPU."""
+    import pandas as pd
+    import pyarrow as pa
+    import torch
+
+    import fugue_amd.api as fa
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+    from fugue_amd.hip.frame import DeviceColumn, HipDataFrame
+    from fugue_amd.schema import Schema
+
+    e = HipExecutionEngine()
+    n = 1000
+    fact = HipDataFrame.from_columns(
+        {
+            "k": DeviceColumn(
+                torch.arange(n, dtype=torch.int64), None, pa.int64()
+            ),
+            "v": DeviceColumn(
+                torch.ones(n, dtype=torch.float64), None, pa.float64()
+            ),
+        },
+        Schema("k:long,v:double"),
+        e.device,
+    )
+
+    calls = []
+
+    def scale(df: HipDataFrame) -> HipDataFrame:
+        calls.append(type(df).__name__)
+        v = df.col("v")
+        return HipDataFrame.from_columns(
+            {"k": df.col("k"),
+             "v": DeviceColumn(v.data * 2.0, v.valid, pa.float64())},
+            Schema("k:long,v:double"),
+            df.device,
+        )
+
+    out = fa.transform(fact, scale, schema="k:long,v:double", engine=e,
+                       as_fugue=True)
+    assert calls == ["HipDataFrame"]
+    assert isinstance(out, HipDataFrame) or isinstance(
+        e.to_df(out), HipDataFrame
+    )
+    pdf = out.as_pandas()
+    assert len(pdf) == n and (pdf["v"] == 2.0).all()
+
+    # same through FugueSQL TRANSFORM + SELECT (the bench pipeline)
+    res = fa.as_pandas(
+        fa.fugue_sql(
+            "t = TRANSFORM fact USING scale SCHEMA k:long,v:double\n"
+            "SELECT SUM(v) AS s FROM t",
+            fact=fact, scale=scale, engine=e,
+        )
+    )
+    assert float(res["s"][0]) == 2.0 * n
