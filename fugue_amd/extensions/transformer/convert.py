@@ -5,7 +5,8 @@ instance / decorated function / plain function with ``# schema:`` comment /
 registered alias all become Transformer/CoTransformer objects.
 """
 import copy
-from typing import Any, Callable, Dict, List, Optional, Union
+import weakref
+from typing import Any, Callable, Dict, List, Optional, Tuple, Union
 
 from fugue_amd.dataframe.array_dataframe import ArrayDataFrame
 from fugue_amd.dataframe.dataframe import DataFrame, LocalDataFrame
@@ -83,6 +84,11 @@ def output_cotransformer(**validation_rules: Any) -> Callable[[Callable], "_Func
     return deco
 
 
+# conversion prototypes per callable (weak keys: dropping the function
+# drops its cache entry); values: {(schema, rules, is_output): prototype}
+_CONVERT_CACHE: "weakref.WeakKeyDictionary" = weakref.WeakKeyDictionary()
+
+
 def _to_general_transformer(
     obj: Any,
     schema: Any,
@@ -106,6 +112,19 @@ def _to_general_transformer(
             return to_instance(obj)
     except Exception as e:
         exp = e
+    # plain-callable conversions are memoized: signature inspection and
+    # `# schema:` comment parsing are pure in (func, schema, rules), and
+    # the call sites receive a shallow copy exactly like the
+    # Transformer-instance path above
+    cache_key: Optional[Tuple[Any, ...]] = None
+    if callable(obj) and not isinstance(obj, type):
+        try:
+            cache_key = (repr(schema), repr(validation_rules), is_output)
+            proto_map = _CONVERT_CACHE.get(obj)
+            if proto_map is not None and cache_key in proto_map:
+                return copy.copy(proto_map[cache_key])
+        except TypeError:  # not weak-referenceable
+            cache_key = None
     try:
         f = to_function(obj, global_vars={**(global_vars or {}), **(local_vars or {})})
         # single vs co: co if first param is DataFrames or multiple df params
@@ -114,8 +133,15 @@ def _to_general_transformer(
         n_df = len([c for c in code if c in "dlspq"])
         use_dfs = "c" in code
         if use_dfs or n_df > 1:
-            return func_co(f, schema, validation_rules)
-        return func_single(f, schema, validation_rules)
+            res = func_co(f, schema, validation_rules)
+        else:
+            res = func_single(f, schema, validation_rules)
+        if cache_key is not None:
+            try:
+                _CONVERT_CACHE.setdefault(obj, {})[cache_key] = res
+            except TypeError:
+                pass
+        return copy.copy(res) if cache_key is not None else res
     except Exception as e:
         exp = e
     raise FugueInterfacelessError(

@@ -30,6 +30,23 @@ _TOKEN_RE = re.compile(
 
 
 def tokenize(sql: str) -> List[Token]:
+    """Tokenize with a small memo: identical statement texts (the common
+    case for repeated queries) skip the regex scan.  Callers treat the
+    token list as read-only (TokenStream only advances an index)."""
+    hit = _TOKEN_CACHE.get(sql)
+    if hit is not None:
+        return hit
+    tokens = _tokenize_uncached(sql)
+    if len(_TOKEN_CACHE) > 256:
+        _TOKEN_CACHE.clear()
+    _TOKEN_CACHE[sql] = tokens
+    return tokens
+
+
+_TOKEN_CACHE: dict = {}
+
+
+def _tokenize_uncached(sql: str) -> List[Token]:
     tokens: List[Token] = []
     pos = 0
     while pos < len(sql):

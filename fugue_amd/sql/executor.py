@@ -682,6 +682,23 @@ class SelectStmt:
 
 
 def parse_select(sql: str) -> SelectStmt:
+    """Parse a SELECT into its AST, memoized by statement text: the AST
+    is read-only downstream (planner/executor never mutate it), so
+    repeated identical queries skip the parse entirely."""
+    hit = _SELECT_CACHE.get(sql)
+    if hit is not None:
+        return hit
+    stmt = _parse_select_uncached(sql)
+    if len(_SELECT_CACHE) > 256:
+        _SELECT_CACHE.clear()
+    _SELECT_CACHE[sql] = stmt
+    return stmt
+
+
+_SELECT_CACHE: Dict[str, SelectStmt] = {}
+
+
+def _parse_select_uncached(sql: str) -> SelectStmt:
     ts = TokenStream(tokenize(sql))
     stmt = _parse_select_stmt(ts)
     if not ts.eof and not ts.match_punct(";"):
