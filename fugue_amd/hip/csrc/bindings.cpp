@@ -41,11 +41,11 @@ void launch_gb_aggregate_part(const int64_t*, const double*, const int32_t*,
                               double*, int64_t*, int64_t, hipStream_t);
 void launch_gb_part_scatter_staged(const int64_t*, const double*, int64_t,
                                    int, int64_t*, void*, double*, int64_t,
-                                   int, int, int*, hipStream_t);
+                                   int, int, int*, int, hipStream_t);
 void launch_gb_aggregate_part_big(const void*, const double*,
                                   const int32_t*, int64_t, int64_t*,
                                   double*, int64_t*, int64_t, int64_t,
-                                  int, int, hipStream_t);
+                                  int, int, int, hipStream_t);
 void launch_join_count(const int64_t*, int64_t, const int64_t*,
                        const int64_t*, const int64_t*, const int32_t*,
                        const int32_t*, int64_t, int32_t*, hipStream_t);
@@ -208,7 +208,12 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
   // caller asks for narrow auto-detection: narrow < 0)
   auto hist = at::zeros({num_parts}, keys.options());
   TORCH_CHECK(num_parts <= 4096, "num_parts must be <= 4096");
-  bool staged = (n_aggs == 1 && num_parts == 512);
+  bool staged =
+      (n_aggs == 1 &&
+       (num_parts == 512 || num_parts == 1024 || num_parts == 2048));
+  // phase-3 LDS table sized for ~0.5 load at the partition granularity:
+  // 512 parts -> 4096 slots, 1024 -> 2048, 2048 -> 1024
+  int slots = num_parts == 1024 ? 2048 : (num_parts == 2048 ? 1024 : 4096);
   // narrow < 0 = speculative: run the int32 path with an in-kernel
   // overflow flag; the CALLER checks the returned flag at its existing
   // sync point and re-runs wide if set (no mid-pipeline sync here)
@@ -234,7 +239,7 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
         pvals.data_ptr<double>(), scatter_chunk, (int)nt,
         use_narrow ? 1 : 0,
         (use_narrow && speculative) ? ovf.data_ptr<int32_t>() : nullptr,
-        stream);
+        (int)num_parts, stream);
   } else {
     launch_gb_part_scatter(keys.data_ptr<int64_t>(), vals.data_ptr<double>(),
                            n_aggs, n, shift, cursor.data_ptr<int64_t>(),
@@ -251,7 +256,7 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
         pkeys.data_ptr(), pvals.data_ptr<double>(),
         ops.data_ptr<int32_t>(), n, tkeys.data_ptr<int64_t>(),
         gaggs.data_ptr<double>(), gcount.data_ptr<int64_t>(), tsize,
-        agg_chunk, (int)nt, use_narrow ? 1 : 0, stream);
+        agg_chunk, (int)nt, use_narrow ? 1 : 0, slots, stream);
   } else {
     launch_gb_aggregate_part(
         pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),

@@ -752,8 +752,10 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel(
 
 // ILP-4 variant of the staged scatter row loop (FUGUE_SC_ILP=4): four
 // key loads / hashes / value loads in flight before the LDS cursor
-// atomics, same staging scheme
-template <bool NT, typename KT>
+// atomics, same staging scheme.  Templated on the partition count P and
+// staging depth E so occupancy/granularity variants (512/8, 1024/4,
+// 2048/2) can be A/B-tested at runtime (FUGUE_GB_PARTS).
+template <bool NT, typename KT, int P, int E>
 __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel_v4(
     const int64_t* __restrict__ keys,
     const double* __restrict__ vals,
@@ -763,16 +765,16 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel_v4(
     double* __restrict__ out_vals,
     int64_t chunk,
     int* __restrict__ ovf) {
-  __shared__ int lhist[STAGE_P];
-  __shared__ int64_t lbase[STAGE_P];
-  __shared__ int lcnt[STAGE_P];
-  __shared__ KT skey[STAGE_P * STAGE_E];
-  __shared__ double sval[STAGE_P * STAGE_E];
+  __shared__ int lhist[P];
+  __shared__ int64_t lbase[P];
+  __shared__ int lcnt[P];
+  __shared__ KT skey[P * E];
+  __shared__ double sval[P * E];
   for (int64_t start = (int64_t)blockIdx.x * chunk; start < n;
        start += (int64_t)gridDim.x * chunk) {
     int64_t end = start + chunk;
     if (end > n) end = n;
-    for (int i = threadIdx.x; i < STAGE_P; i += blockDim.x) {
+    for (int i = threadIdx.x; i < P; i += blockDim.x) {
       lhist[i] = 0;
       lcnt[i] = 0;
     }
@@ -794,7 +796,7 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel_v4(
       }
     }
     __syncthreads();
-    for (int i = threadIdx.x; i < STAGE_P; i += blockDim.x) {
+    for (int i = threadIdx.x; i < P; i += blockDim.x) {
       int c = lhist[i];
       lbase[i] =
           c > 0
@@ -826,9 +828,9 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel_v4(
 #pragma unroll
         for (int u = 0; u < 4; ++u) {
           int pos = atomicAdd(&lcnt[p[u]], 1);
-          if (pos < STAGE_E) {
-            skey[p[u] * STAGE_E + pos] = (KT)k[u];
-            sval[p[u] * STAGE_E + pos] = v[u];
+          if (pos < E) {
+            skey[p[u] * E + pos] = (KT)k[u];
+            sval[p[u] * E + pos] = v[u];
           } else {
             int64_t gpos = lbase[p[u]] + pos;
             stream_st<NT>(&out_keys[gpos], (KT)k[u]);
@@ -846,9 +848,9 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel_v4(
         int p = (int)(mix64((uint64_t)key) >> shift);
         int pos = atomicAdd(&lcnt[p], 1);
         double v = stream_ld<NT>(&vals[i]);
-        if (pos < STAGE_E) {
-          skey[p * STAGE_E + pos] = (KT)key;
-          sval[p * STAGE_E + pos] = v;
+        if (pos < E) {
+          skey[p * E + pos] = (KT)key;
+          sval[p * E + pos] = v;
         } else {
           int64_t gpos = lbase[p] + pos;
           stream_st<NT>(&out_keys[gpos], (KT)key);
@@ -857,11 +859,11 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_staged_kernel_v4(
       }
     }
     __syncthreads();
-    for (int t = threadIdx.x; t < STAGE_P * STAGE_E; t += blockDim.x) {
-      int p = t / STAGE_E;
-      int e = t % STAGE_E;
+    for (int t = threadIdx.x; t < P * E; t += blockDim.x) {
+      int p = t / E;
+      int e = t % E;
       int c = lcnt[p];
-      if (e < c && e < STAGE_E) {
+      if (e < c && e < E) {
         int64_t gpos = lbase[p] + e;
         stream_st<NT>(&out_keys[gpos], skey[t]);
         stream_st<NT>(&out_vals[gpos], sval[t]);
@@ -905,7 +907,7 @@ __device__ __forceinline__ int lds_key_cas(int32_t* addr, int32_t cmp,
 // slow path of the per-row LDS upsert: first-slot read missed (empty or
 // different key) — probe/claim, spilling to the global table after 32
 // displacements
-template <typename KT>
+template <int SLOTS, typename KT>
 __device__ __forceinline__ void gb_lds_upsert(
     KT key, double v, int slot, KT EMPTY, KT* __restrict__ lkeys,
     double* __restrict__ laggs, int* __restrict__ lcount,
@@ -919,7 +921,7 @@ __device__ __forceinline__ void gb_lds_upsert(
       KT prev = (KT)lds_key_cas(&lkeys[slot], EMPTY, key);
       if (prev == EMPTY || prev == key) { in_lds = true; break; }
     }
-    slot = (slot + 1) & (LDS_SLOTS_BIG - 1);
+    slot = (slot + 1) & (SLOTS - 1);
   }
   if (in_lds) {
     atomicAdd(&lcount[slot], 1);
@@ -975,14 +977,14 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
         atomicAdd(&lcount[s1], 1);
         atomicAdd(&laggs[s1], v1);
       } else {
-        gb_lds_upsert<KT>(k1, v1, s1, EMPTY, lkeys, laggs, lcount, tkeys,
+        gb_lds_upsert<LDS_SLOTS_BIG, KT>(k1, v1, s1, EMPTY, lkeys, laggs, lcount, tkeys,
                           gaggs, gcount, tsize);
       }
       if (c2 == k2) {
         atomicAdd(&lcount[s2], 1);
         atomicAdd(&laggs[s2], v2);
       } else {
-        gb_lds_upsert<KT>(k2, v2, s2, EMPTY, lkeys, laggs, lcount, tkeys,
+        gb_lds_upsert<LDS_SLOTS_BIG, KT>(k2, v2, s2, EMPTY, lkeys, laggs, lcount, tkeys,
                           gaggs, gcount, tsize);
       }
     }
@@ -995,7 +997,7 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
         atomicAdd(&lcount[slot], 1);
         atomicAdd(&laggs[slot], v);
       } else {
-        gb_lds_upsert<KT>(key, v, slot, EMPTY, lkeys, laggs, lcount, tkeys,
+        gb_lds_upsert<LDS_SLOTS_BIG, KT>(key, v, slot, EMPTY, lkeys, laggs, lcount, tkeys,
                           gaggs, gcount, tsize);
       }
     }
@@ -1015,7 +1017,7 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel(
 // ILP-4 row loop variant (FUGUE_GB_ILP=4): four independent key loads /
 // hashes / first-slot LDS reads in flight per iteration — deeper
 // latency hiding at the 2-blocks/CU occupancy this kernel runs at
-template <bool NT, typename KT>
+template <bool NT, typename KT, int SLOTS>
 __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel_v4(
     const KT* __restrict__ part_keys,
     const double* __restrict__ part_vals,
@@ -1025,16 +1027,16 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel_v4(
     double* __restrict__ gaggs,
     int64_t* __restrict__ gcount,
     int64_t tsize, int64_t chunk) {
-  __shared__ KT lkeys[LDS_SLOTS_BIG];
-  __shared__ double laggs[LDS_SLOTS_BIG];
-  __shared__ int lcount[LDS_SLOTS_BIG];
+  __shared__ KT lkeys[SLOTS];
+  __shared__ double laggs[SLOTS];
+  __shared__ int lcount[SLOTS];
   const KT EMPTY = lds_empty<KT>();
   bool is_count = ops[0] == 3;
   for (int64_t start = (int64_t)blockIdx.x * chunk; start < n;
        start += (int64_t)gridDim.x * chunk) {
     int64_t end = start + chunk;
     if (end > n) end = n;
-    for (int i = threadIdx.x; i < LDS_SLOTS_BIG; i += blockDim.x) {
+    for (int i = threadIdx.x; i < SLOTS; i += blockDim.x) {
       lkeys[i] = EMPTY;
       lcount[i] = 0;
       laggs[i] = 0.0;
@@ -1051,7 +1053,7 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel_v4(
       for (int u = 0; u < 4; ++u) k[u] = stream_ld<NT>(&part_keys[i + u * stride]);
 #pragma unroll
       for (int u = 0; u < 4; ++u)
-        s[u] = (int)(mix64((uint64_t)(int64_t)k[u]) & (LDS_SLOTS_BIG - 1));
+        s[u] = (int)(mix64((uint64_t)(int64_t)k[u]) & (SLOTS - 1));
 #pragma unroll
       for (int u = 0; u < 4; ++u) c[u] = lkeys[s[u]];
 #pragma unroll
@@ -1063,26 +1065,26 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel_v4(
           atomicAdd(&lcount[s[u]], 1);
           atomicAdd(&laggs[s[u]], v[u]);
         } else {
-          gb_lds_upsert<KT>(k[u], v[u], s[u], EMPTY, lkeys, laggs, lcount,
+          gb_lds_upsert<SLOTS, KT>(k[u], v[u], s[u], EMPTY, lkeys, laggs, lcount,
                             tkeys, gaggs, gcount, tsize);
         }
       }
     }
     for (; i < end; i += stride) {
       KT key = stream_ld<NT>(&part_keys[i]);
-      int slot = (int)(mix64((uint64_t)(int64_t)key) & (LDS_SLOTS_BIG - 1));
+      int slot = (int)(mix64((uint64_t)(int64_t)key) & (SLOTS - 1));
       double v = is_count ? 1.0 : stream_ld<NT>(&part_vals[i]);
       KT cur = lkeys[slot];
       if (cur == key) {
         atomicAdd(&lcount[slot], 1);
         atomicAdd(&laggs[slot], v);
       } else {
-        gb_lds_upsert<KT>(key, v, slot, EMPTY, lkeys, laggs, lcount, tkeys,
+        gb_lds_upsert<SLOTS, KT>(key, v, slot, EMPTY, lkeys, laggs, lcount, tkeys,
                           gaggs, gcount, tsize);
       }
     }
     __syncthreads();
-    for (int i = threadIdx.x; i < LDS_SLOTS_BIG; i += blockDim.x) {
+    for (int i = threadIdx.x; i < SLOTS; i += blockDim.x) {
       KT key = lkeys[i];
       if (key == EMPTY) continue;
       int64_t gslot = gb_probe_insert((int64_t)key, tkeys, tsize);
@@ -1129,7 +1131,7 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel_v1(
         atomicAdd(&lcount[slot], 1);
         atomicAdd(&laggs[slot], v);
       } else {
-        gb_lds_upsert<KT>(key, v, slot, EMPTY, lkeys, laggs, lcount, tkeys,
+        gb_lds_upsert<LDS_SLOTS_BIG, KT>(key, v, slot, EMPTY, lkeys, laggs, lcount, tkeys,
                           gaggs, gcount, tsize);
       }
     }
@@ -1145,6 +1147,65 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel_v1(
     __syncthreads();
   }
 }
+
+// P/E-variant dispatch for the v4 staged scatter: 512/8 (baseline),
+// 1024/4 and 2048/2 trade staging depth for finer partitions (smaller
+// phase-3 LDS tables -> higher phase-3 occupancy)
+template <bool NT, typename KT>
+static void launch_scatter_v4_dispatch(int num_parts, dim3 g, dim3 b,
+                                       hipStream_t stream,
+                                       const int64_t* keys,
+                                       const double* vals, int64_t n,
+                                       int shift, int64_t* cursor, KT* ok,
+                                       double* out_vals, int64_t chunk,
+                                       int* ovf) {
+  switch (num_parts) {
+    case 1024:
+      hipLaunchKernelGGL((gb_part_scatter_staged_kernel_v4<NT, KT, 1024, 4>),
+                         g, b, 0, stream, keys, vals, n, shift, cursor, ok,
+                         out_vals, chunk, ovf);
+      break;
+    case 2048:
+      hipLaunchKernelGGL((gb_part_scatter_staged_kernel_v4<NT, KT, 2048, 2>),
+                         g, b, 0, stream, keys, vals, n, shift, cursor, ok,
+                         out_vals, chunk, ovf);
+      break;
+    default:
+      hipLaunchKernelGGL((gb_part_scatter_staged_kernel_v4<NT, KT, 512, 8>),
+                         g, b, 0, stream, keys, vals, n, shift, cursor, ok,
+                         out_vals, chunk, ovf);
+  }
+}
+
+
+// SLOTS-variant dispatch for the v4 aggregate: 4096 (64KB LDS, 2
+// blocks/CU), 2048 (32KB, 5 blocks/CU), 1024 (16KB, 8 blocks/CU)
+template <bool NT, typename KT>
+static void launch_agg_v4_dispatch(int slots, dim3 g, dim3 b,
+                                   hipStream_t stream, const KT* pk,
+                                   const double* part_vals,
+                                   const int32_t* ops, int64_t n,
+                                   int64_t* tkeys, double* gaggs,
+                                   int64_t* gcount, int64_t tsize,
+                                   int64_t chunk) {
+  switch (slots) {
+    case 1024:
+      hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v4<NT, KT, 1024>), g,
+                         b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
+                         gcount, tsize, chunk);
+      break;
+    case 2048:
+      hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v4<NT, KT, 2048>), g,
+                         b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
+                         gcount, tsize, chunk);
+      break;
+    default:
+      hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v4<NT, KT, 4096>), g,
+                         b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
+                         gcount, tsize, chunk);
+  }
+}
+
 
 extern "C" {
 
@@ -1171,25 +1232,27 @@ void launch_gb_part_scatter_staged(const int64_t* keys, const double* vals,
                                    int64_t n, int shift, int64_t* cursor,
                                    void* out_keys, double* out_vals,
                                    int64_t chunk, int nt, int narrow,
-                                   int* ovf, hipStream_t stream) {
+                                   int* ovf, int num_parts,
+                                   hipStream_t stream) {
   if (chunk <= 0) chunk = SCATTER_CHUNK;
   int64_t blocks = (n + chunk - 1) / chunk;
   if (blocks > MAX_GRID) blocks = MAX_GRID;
   if (blocks < 1) blocks = 1;
   dim3 g((int)blocks), b(BLOCK);
   // ILP depth of the scatter row loop: 4 (default, measured 5.06 vs
-  // 5.43 ms/step on the 125M-row bench), 2 = paired
-  bool squad = parse_ilp_env("FUGUE_SC_ILP", 4) == 4;
+  // 5.43 ms/step on the 125M-row bench), 2 = paired (512 parts only)
+  bool squad = parse_ilp_env("FUGUE_SC_ILP", 4) == 4 || num_parts != 512;
   if (narrow) {
     auto* ok = (int32_t*)out_keys;
     if (squad && nt)
-      hipLaunchKernelGGL((gb_part_scatter_staged_kernel_v4<true, int32_t>),
-                         g, b, 0, stream, keys, vals, n, shift, cursor, ok,
-                         out_vals, chunk, ovf);
+      launch_scatter_v4_dispatch<true, int32_t>(num_parts, g, b, stream,
+                                                keys, vals, n, shift, cursor,
+                                                ok, out_vals, chunk, ovf);
     else if (squad)
-      hipLaunchKernelGGL((gb_part_scatter_staged_kernel_v4<false, int32_t>),
-                         g, b, 0, stream, keys, vals, n, shift, cursor, ok,
-                         out_vals, chunk, ovf);
+      launch_scatter_v4_dispatch<false, int32_t>(num_parts, g, b, stream,
+                                                 keys, vals, n, shift,
+                                                 cursor, ok, out_vals, chunk,
+                                                 ovf);
     else if (nt)
       hipLaunchKernelGGL((gb_part_scatter_staged_kernel<true, int32_t>), g, b,
                          0, stream, keys, vals, n, shift, cursor, ok,
@@ -1201,13 +1264,14 @@ void launch_gb_part_scatter_staged(const int64_t* keys, const double* vals,
   } else {
     auto* ok = (int64_t*)out_keys;
     if (squad && nt)
-      hipLaunchKernelGGL((gb_part_scatter_staged_kernel_v4<true, int64_t>),
-                         g, b, 0, stream, keys, vals, n, shift, cursor, ok,
-                         out_vals, chunk, nullptr);
+      launch_scatter_v4_dispatch<true, int64_t>(num_parts, g, b, stream,
+                                                keys, vals, n, shift, cursor,
+                                                ok, out_vals, chunk, nullptr);
     else if (squad)
-      hipLaunchKernelGGL((gb_part_scatter_staged_kernel_v4<false, int64_t>),
-                         g, b, 0, stream, keys, vals, n, shift, cursor, ok,
-                         out_vals, chunk, nullptr);
+      launch_scatter_v4_dispatch<false, int64_t>(num_parts, g, b, stream,
+                                                 keys, vals, n, shift,
+                                                 cursor, ok, out_vals, chunk,
+                                                 nullptr);
     else if (nt)
       hipLaunchKernelGGL((gb_part_scatter_staged_kernel<true, int64_t>), g, b,
                          0, stream, keys, vals, n, shift, cursor, ok,
@@ -1225,8 +1289,11 @@ void launch_gb_aggregate_part_big(const void* part_keys,
                                   int64_t* tkeys, double* gaggs,
                                   int64_t* gcount, int64_t tsize,
                                   int64_t chunk, int nt, int narrow,
-                                  hipStream_t stream) {
-  if (chunk <= 0) chunk = AGG_CHUNK;
+                                  int slots, hipStream_t stream) {
+  // default chunk sweep-tuned per table size: 32K rows at 4096 slots
+  // (2 blocks/CU), 16K at 2048 slots (5 blocks/CU; measured 4.03 vs
+  // 4.42 ms on the 125M-row bench shape)
+  if (chunk <= 0) chunk = (slots == 2048) ? (BLOCK * 64) : AGG_CHUNK;
   int64_t blocks = (n + chunk - 1) / chunk;
   if (blocks > MAX_GRID) blocks = MAX_GRID;
   if (blocks < 1) blocks = 1;
@@ -1236,16 +1303,17 @@ void launch_gb_aggregate_part_big(const void* part_keys,
   int ilp_d = parse_ilp_env("FUGUE_GB_ILP", 4);
   bool legacy = ilp_d == 1;
   bool quad = ilp_d == 4;
+  if (slots != 4096) legacy = false, quad = true;  // variants are v4-only
   if (narrow) {
     auto* pk = (const int32_t*)part_keys;
     if (quad && nt)
-      hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v4<true, int32_t>), g,
-                         b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
-                         gcount, tsize, chunk);
+      launch_agg_v4_dispatch<true, int32_t>(slots, g, b, stream, pk,
+                                            part_vals, ops, n, tkeys, gaggs,
+                                            gcount, tsize, chunk);
     else if (quad)
-      hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v4<false, int32_t>), g,
-                         b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
-                         gcount, tsize, chunk);
+      launch_agg_v4_dispatch<false, int32_t>(slots, g, b, stream, pk,
+                                             part_vals, ops, n, tkeys, gaggs,
+                                             gcount, tsize, chunk);
     else if (legacy)
       hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v1<false, int32_t>), g,
                          b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
@@ -1261,13 +1329,13 @@ void launch_gb_aggregate_part_big(const void* part_keys,
   } else {
     auto* pk = (const int64_t*)part_keys;
     if (quad && nt)
-      hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v4<true, int64_t>), g,
-                         b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
-                         gcount, tsize, chunk);
+      launch_agg_v4_dispatch<true, int64_t>(slots, g, b, stream, pk,
+                                            part_vals, ops, n, tkeys, gaggs,
+                                            gcount, tsize, chunk);
     else if (quad)
-      hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v4<false, int64_t>), g,
-                         b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,
-                         gcount, tsize, chunk);
+      launch_agg_v4_dispatch<false, int64_t>(slots, g, b, stream, pk,
+                                             part_vals, ops, n, tkeys, gaggs,
+                                             gcount, tsize, chunk);
     else if (legacy)
       hipLaunchKernelGGL((gb_aggregate_part_big_kernel_v1<false, int64_t>), g,
                          b, 0, stream, pk, part_vals, ops, n, tkeys, gaggs,

@@ -375,7 +375,15 @@ def groupby_aggregate(
         # single-agg path with moderate cardinality uses the LDS
         # write-staged scatter (512 parts, 4096-slot phase-3 table)
         if len(aggs) <= 1 and expected_groups <= 1_500_000:
-            num_parts = 512
+            import os as _os0
+
+            # staged-variant partition count: 512 (8-deep staging,
+            # 4096-slot phase-3), 1024 (4-deep, 2048-slot) or 2048
+            # (2-deep, 1024-slot) — smaller tables run phase 3 at
+            # higher occupancy; A/B via FUGUE_GB_PARTS
+            num_parts = int(_os0.environ.get("FUGUE_GB_PARTS", "1024"))
+            if num_parts not in (512, 1024, 2048):
+                num_parts = 512
         else:
             num_parts = min(4096, _next_pow2(max(16, expected_groups // 512)))
         import os as _os
