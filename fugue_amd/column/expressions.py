@@ -503,6 +503,19 @@ class _UnaryAggFuncExpr(_FuncExpr):
             self.func, self.args[0], arg_distinct=self.is_distinct
         )
 
+    def infer_type(self, schema: Schema) -> Optional[pa.DataType]:
+        if self.as_type is not None:
+            return self.as_type
+        f = self.func.upper()
+        if f in ("COUNT", "COUNT_DISTINCT"):
+            return pa.int64()
+        if f == "AVG":
+            return pa.float64()
+        if f in ("MIN", "MAX", "FIRST", "LAST", "SUM"):
+            arg = self.args[0]
+            return arg.infer_type(schema) if isinstance(arg, ColumnExpr) else None
+        return None
+
 
 def _is_agg(column: Any) -> bool:
     if isinstance(column, _UnaryAggFuncExpr):

@@ -245,11 +245,24 @@ def eval_select(
                     if name in row:
                         continue
                     row[name] = _eval_one_agg_cell(c, sub)
+                if having is not None and _is_agg(having):
+                    # HAVING with aggregate terms (possibly absent from
+                    # the select list): evaluate per group
+                    row["__fugue_having"] = bool(
+                        _eval_one_agg_cell(having, sub)
+                    )
                 rows.append(row)
             names = [c.output_name for c in cols.all_cols]
+            if having is not None and _is_agg(having):
+                names = names + ["__fugue_having"]
             res = pd.DataFrame(rows, columns=names) if rows else pd.DataFrame(
                 columns=names
             )
+            if having is not None and _is_agg(having):
+                res = res[res["__fugue_having"]].drop(
+                    columns=["__fugue_having"]
+                )
+                having = None
         if having is not None:
             res = eval_filter(res, having)
     if cols.is_distinct:
@@ -278,6 +291,22 @@ def _eval_one_agg_cell(c: ColumnExpr, df: pd.DataFrame) -> Any:
                 v = left * right
             elif op == "/":
                 v = left / right
+            elif op == "==":
+                v = left == right
+            elif op == "!=":
+                v = left != right
+            elif op == "<":
+                v = left < right
+            elif op == "<=":
+                v = left <= right
+            elif op == ">":
+                v = left > right
+            elif op == ">=":
+                v = left >= right
+            elif op == "&":
+                v = bool(left) and bool(right)
+            elif op == "|":
+                v = bool(left) or bool(right)
             else:
                 raise NotImplementedError(f"agg op {op}")
         except TypeError:

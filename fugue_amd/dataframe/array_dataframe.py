@@ -42,8 +42,10 @@ class ArrayDataFrame(LocalBoundedDataFrame):
     def native(self) -> List[List[Any]]:
         return self._native
 
-    def native_as_df(self) -> List[List[Any]]:
-        return self._native
+    def native_as_df(self) -> Any:
+        # the raw array carries no schema; the dataframe form of this
+        # frame is pandas (reference ``dataframe.py:316``)
+        return self.as_pandas()
 
     @property
     def empty(self) -> bool:
@@ -68,7 +70,15 @@ class ArrayDataFrame(LocalBoundedDataFrame):
         if columns is None and not type_safe:
             return self._native
         pos = self._pos(columns)
-        return [[row[i] for i in pos] for row in self._native]
+        rows = [[row[i] for i in pos] for row in self._native]
+        if type_safe:
+            from fugue_amd.dataframe.coerce import coerce_rows
+
+            sub = (
+                self.schema if columns is None else self.schema.extract(columns)
+            )
+            rows = coerce_rows(rows, sub)
+        return rows
 
     def as_array_iterable(
         self, columns: Optional[List[str]] = None, type_safe: bool = False

@@ -76,7 +76,9 @@ class DataFrame(Dataset):
         return self.as_arrow().to_pandas()
 
     def as_arrow(self, type_safe: bool = False) -> pa.Table:
-        rows = self.as_array()
+        from fugue_amd.dataframe.coerce import coerce_rows
+
+        rows = coerce_rows(self.as_array(), self.schema)
         cols = self.columns
         pylist = [{c: row[i] for i, c in enumerate(cols)} for row in rows]
         return pa.Table.from_pylist(pylist, schema=self.schema.pa_schema)

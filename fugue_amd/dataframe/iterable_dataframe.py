@@ -78,7 +78,8 @@ class IterableDataFrame(LocalUnboundedDataFrame):
         return self._native
 
     def native_as_df(self) -> Any:
-        return self._native
+        # the raw iterable carries no schema; present as pandas
+        return self.as_pandas()
 
     @property
     def empty(self) -> bool:

@@ -476,6 +476,13 @@ def aggregate(
     as_fugue: bool = False,
     **agg_kwcols: ColumnExpr,
 ) -> AnyDataFrame:
+    from fugue_amd.column.expressions import ColumnExpr as _CE
+
+    bad = [k for k, v in agg_kwcols.items() if not isinstance(v, _CE)]
+    if bad:
+        raise ValueError(
+            f"aggregate args must be column expressions: {bad}"
+        )
     cols = [v.alias(k) for k, v in agg_kwcols.items()]
     spec = (
         PartitionSpec(by=partition_by)

@@ -227,6 +227,14 @@ class ExecutionEngine(FugueEngineBase):
         self._stop_engine_called = False
         self._lock = RLock()
 
+    def __copy__(self) -> "ExecutionEngine":
+        # engines hold live device/process state: copies are the engine
+        # itself (reference contract, ``execution_engine.py:1176``)
+        return self
+
+    def __deepcopy__(self, memo: Any) -> "ExecutionEngine":
+        return self
+
     def __enter__(self) -> "ExecutionEngine":
         raise FugueInvalidOperation(
             "use engine.as_context() instead of `with engine:`"
@@ -471,10 +479,29 @@ class ExecutionEngine(FugueEngineBase):
 
         pdf = df.as_pandas()
         res = eval_select(pdf, df.schema, columns, where=where, having=having)
-        inferred = columns.replace_wildcard(df.schema).infer_schema(df.schema)
+        target = columns.replace_wildcard(df.schema)
+        inferred = target.infer_schema(df.schema)
         if inferred is not None:
             return self.to_df(PandasDataFrame(res, inferred))
-        return self.to_df(PandasDataFrame(res))
+        # partial correction: columns whose expression type IS known must
+        # come out with that type even when the full schema can't be
+        # inferred (reference ``SQLExpressionGenerator.correct_select_schema``)
+        out = PandasDataFrame(res)
+        import pyarrow as _pa
+
+        fields = []
+        for c in target.all_cols:
+            tp = c.infer_type(df.schema)
+            name = c.output_name or c.name
+            if (
+                tp is not None
+                and name in out.schema
+                and out.schema[name].type != tp
+            ):
+                fields.append(_pa.field(name, tp))
+        if fields:
+            out = out.alter_columns(Schema(fields))
+        return self.to_df(out)
 
     def select(
         self,

@@ -925,13 +925,25 @@ class HipExecutionEngine(ExecutionEngine):
         build_invalid = (bi < 0) if has_null_build else None
         cols: Dict[str, DeviceColumn] = {}
         probe_gather = probe.gather_rows(pi)
-        build_gather = build.gather_rows(bi_safe)
+        need_build = any(
+            f.name not in probe.schema._index for f in output_schema.fields
+        )
+        # empty build side: every bi is -1, nothing to gather (semi/anti
+        # emit probe columns only; outer joins pad with nulls)
+        build_gather = (
+            build.gather_rows(bi_safe)
+            if need_build and build.count() > 0
+            else None
+        )
         for f in output_schema.fields:
             name = f.name
             if name in probe.schema._index:
                 src = probe_gather.col(name)
                 # keys come from the left frame per fugue semantics
                 cols[name] = src
+            elif build_gather is None:
+                cols[name] = self._null_column(f.type, int(pi.numel()))
+                continue
             else:
                 src = build_gather.col(name)
                 if build_invalid is not None:

@@ -286,7 +286,7 @@ class ExecutionEngineTestSuite:
                 back = self.engine.load_df(
                     path,
                     columns="x:long,y:str" if fmt != "parquet" else None,
-                    **(dict(infer_schema=True, header=True) if fmt == "csv" else {}),
+                    **(dict(header=True) if fmt == "csv" else {}),
                 )
                 assert _df_eq(
                     self.engine.to_df(back),

@@ -38,6 +38,17 @@ def _expand(path: Union[str, List[str]]) -> List[str]:
             if len(matches) == 0:
                 raise FileNotFoundError(f"no files match {p}")
             res.extend(matches)
+        elif os.path.isdir(p):
+            # a folder of part files; marker/hidden files (_SUCCESS,
+            # .crc) are not data
+            inner = sorted(
+                os.path.join(p, f)
+                for f in os.listdir(p)
+                if not f.startswith((".", "_"))
+            )
+            if len(inner) == 0:
+                raise FileNotFoundError(f"no data files in folder {p}")
+            res.extend(inner)
         else:
             res.append(p)
     return res
@@ -70,6 +81,10 @@ def load_df(
     elif fmt == "csv":
         header = kwargs.pop("header", True)
         infer_schema = kwargs.pop("infer_schema", False)
+        if infer_schema and schema is not None:
+            raise ValueError(
+                "can't set schema in columns when infer_schema is true"
+            )
         dfs = []
         for f in files:
             if header:
