@@ -84,3 +84,18 @@ class TestHipGpuEdgeCases(ExecutionEngineEdgeCaseTests):
         from fugue_amd.hip.execution_engine import HipExecutionEngine
 
         return HipExecutionEngine()
+
+
+from fugue_amd.testing.execution_conformance import ExecutionEngineConformance
+
+
+class TestHipGpuExecutionConformance(ExecutionEngineConformance):
+    """All 42 reference ExecutionEngineTests cases on device tensors."""
+
+    native_is_fugue = True  # HipDataFrame is both native and fugue
+
+    @classmethod
+    def make_engine(cls):
+        from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+        return HipExecutionEngine()
