@@ -46,6 +46,11 @@ class ColumnExpr:
     def infer_type(self, schema: Schema) -> Optional[pa.DataType]:
         return self.as_type
 
+    def infer_alias(self) -> "ColumnExpr":
+        """Infer the output name from underlying columns when unnamed
+        (reference ``expressions.py:111``)."""
+        return self
+
     def __str__(self) -> str:
         res = self.body_str
         if self.as_type is not None:
@@ -333,6 +338,12 @@ class _UnaryOpExpr(ColumnExpr):
             return f"{self.col.body_str} IS NOT NULL"
         return f"{self.op}({self.col.body_str})"
 
+    def infer_alias(self) -> ColumnExpr:
+        if self.output_name != "":
+            return self
+        inner = self.col.infer_alias().output_name
+        return self.alias(inner) if inner != "" else self
+
     def alias(self, as_name: str) -> ColumnExpr:
         other = type(self)(self.op, self.col)
         other._as_name = as_name
@@ -502,6 +513,17 @@ class _UnaryAggFuncExpr(_FuncExpr):
         return _UnaryAggFuncExpr(
             self.func, self.args[0], arg_distinct=self.is_distinct
         )
+
+    def infer_alias(self) -> ColumnExpr:
+        if self.output_name != "":
+            return self
+        inner = self.args[0]
+        name = (
+            inner.infer_alias().output_name
+            if isinstance(inner, ColumnExpr)
+            else ""
+        )
+        return self.alias(name) if name != "" else self
 
     def infer_type(self, schema: Schema) -> Optional[pa.DataType]:
         if self.as_type is not None:

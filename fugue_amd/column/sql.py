@@ -26,7 +26,17 @@ from fugue_amd.utils.hash import to_uuid
 class SelectColumns:
     def __init__(self, *cols: ColumnExpr, arg_distinct: bool = False):
         self._distinct = arg_distinct
-        self._cols = [c.infer_alias() for c in cols]
+        from fugue_amd.column.expressions import all_cols as _all_cols
+
+        # normalize col("*") into the canonical wildcard expression
+        self._cols = [
+            _all_cols()
+            if isinstance(c, _NamedColumnExpr)
+            and c.name == "*"
+            and c.as_name == ""
+            else c.infer_alias()
+            for c in cols
+        ]
         if len(self._cols) == 0:
             raise ValueError("select columns can't be empty")
         self._literals = [
@@ -60,6 +70,17 @@ class SelectColumns:
         names = [c.output_name for c in self._cols if c.output_name != ""]
         if len(names) != len(set(names)):
             raise ValueError(f"duplicated output names in {names}")
+        if len(self._wildcards) > 0:
+            # plain unaliased columns next to * duplicate its output
+            for c in self._cols:
+                if (
+                    isinstance(c, _NamedColumnExpr)
+                    and c.name != "*"
+                    and c.as_name == ""
+                ):
+                    raise ValueError(
+                        f"with *, column {c} must have an alias"
+                    )
 
     def __str__(self):
         return ", ".join(str(c) for c in self._cols)

@@ -6,7 +6,7 @@ immutable frame views with lazy schema discovery, conversion
 The implementation is new, written against pandas/pyarrow directly.
 """
 from abc import abstractmethod
-from typing import Any, Callable, Dict, Iterable, List, Optional, Tuple, Union
+from typing import TypeVar, Any, Callable, Dict, Iterable, List, Optional, Tuple, Union
 
 import pandas as pd
 import pyarrow as pa
@@ -17,7 +17,9 @@ from fugue_amd.schema import Schema
 from fugue_amd.utils.display import PrettyTable
 from fugue_amd.utils.registry import register_plugin, try_run_plugin
 
-AnyDataFrame = Any
+# a TypeVar (not typing.Any) so annotation dispatch can target it
+# (reference ``dataframe.py:26``)
+AnyDataFrame = TypeVar("AnyDataFrame", "DataFrame", object)
 
 
 class DataFrame(Dataset):

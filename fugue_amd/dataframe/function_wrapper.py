@@ -312,6 +312,57 @@ class _EmptyAwareIterableDictParam(_IterableDictParam):
         return make_empty_aware(df.as_dict_iterable())
 
 
+class _AnyDataFrameParam(AnnotatedParam):
+    """``AnyDataFrame`` annotation: any supported frame object
+    (reference ``_AnyDataFrameParam``, code "d")."""
+
+    code = "d"
+    need_schema_ = None
+
+    @staticmethod
+    def matches(anno: Any) -> bool:
+        from fugue_amd.dataframe.dataframe import AnyDataFrame
+
+        return anno is AnyDataFrame
+
+    def to_input_data(self, df: DataFrame, ctx: Any) -> Any:
+        return df
+
+    def to_output_df(self, output: Any, schema: Any, ctx: Any) -> DataFrame:
+        from fugue_amd.dataframe.dataframe import as_fugue_df
+
+        res = as_fugue_df(output) if schema is None else as_fugue_df(
+            output, schema=schema
+        )
+        return res
+
+    def count(self, obj: Any) -> int:
+        from fugue_amd.dataframe.dataframe import as_fugue_df
+
+        return as_fugue_df(obj).count()
+
+
+class _DictRowParam(AnnotatedParam):
+    """``Dict[str, Any]`` output: one row (reference ``DictParam``,
+    code "r")."""
+
+    code = "r"
+    need_schema_ = True
+
+    @staticmethod
+    def matches(anno: Any) -> bool:
+        import typing
+
+        return anno == Dict[str, Any] or anno == typing.Dict[str, Any]
+
+    def to_output_df(self, output: Any, schema: Any, ctx: Any) -> DataFrame:
+        sc = schema if isinstance(schema, Schema) else Schema(schema)
+        return ArrayDataFrame([[output.get(n, None) for n in sc.names]], sc)
+
+    def count(self, obj: Any) -> int:
+        return 1
+
+
 class _PandasParam(AnnotatedParam):
     code = "p"
     need_schema_ = False
@@ -503,6 +554,8 @@ for _c in [
     _ListDictParam,
     _IterableDictParam,
     _EmptyAwareIterableDictParam,
+    _DictRowParam,
+    _AnyDataFrameParam,
     _PandasParam,
     _IterablePandasParam,
     _PyArrowTableParam,
@@ -630,7 +683,7 @@ class DataFrameFunctionWrapper:
                 for _ in rt:
                     pass
             return None
-        if self._rt.code in "dlspq":
+        if self._rt.code in "dlspqr":
             return self._rt.to_output_df(rt, output_schema, ctx=ctx)
         return rt
 

@@ -183,7 +183,7 @@ def make_sql_engine(
             raise ValueError(f"sql engine {engine!r} is not registered")
         return func(execution_engine, **kwargs)
     if isinstance(engine, type) and issubclass(engine, SQLEngine):
-        return engine(execution_engine)
+        return engine(execution_engine, **kwargs)
     ok, parsed = parse_sql_engine.run(engine, execution_engine, **kwargs)
     if ok:
         return parsed

@@ -377,6 +377,13 @@ class NativeExecutionEngine(ExecutionEngine):
         force_single: bool = False,
         **kwargs: Any,
     ) -> None:
+        keys = list(partition_spec.partition_by) if partition_spec else []
+        if len(keys) > 0 and not force_single:
+            _io.save_df_partitioned(
+                df.as_pandas(), df.schema, path, keys,
+                format_hint=format_hint, mode=mode, **kwargs
+            )
+            return
         _io.save_df(
             df.as_pandas(), df.schema, path, format_hint=format_hint, mode=mode, **kwargs
         )

@@ -223,7 +223,10 @@ class Fillna(_FrameOp):
 
     def run_op(self, df: DataFrame, value: Any,
                subset: Optional[List[str]]) -> DataFrame:
-        if value is None:
+        if value is None or (
+            isinstance(value, dict)
+            and (len(value) == 0 or any(v is None for v in value.values()))
+        ):
             raise FugueWorkflowError("fillna value cannot be None")
         return self.execution_engine.fillna(df, value=value, subset=subset)
 

@@ -276,7 +276,7 @@ class _FuncAsTransformer(Transformer):
                 f"schema is required for transformer {func}"
             )
         tr = _FuncAsTransformer()
-        tr._wrapper = DataFrameFunctionWrapper(func, "^[dlspq][fF]?x*z?$", "^[dlspq]$")
+        tr._wrapper = DataFrameFunctionWrapper(func, "^[dlspq][fF]?x*z?$", "^[dlspqr]$")
         tr._output_schema_arg = schema
         tr._validation_rules = validation_rules
         tr._uses_callback = "f" in tr._wrapper.input_code.lower()
@@ -305,7 +305,7 @@ class _FuncAsOutputTransformer(_FuncAsTransformer):
         validation_rules.update(parse_validation_rules_from_comment(func))
         tr = _FuncAsOutputTransformer()
         tr._wrapper = DataFrameFunctionWrapper(
-            func, "^[dlspq][fF]?x*z?$", "^[dlspqn]$"
+            func, "^[dlspq][fF]?x*z?$", "^[dlspqrn]$"
         )
         tr._output_schema_arg = None
         tr._validation_rules = validation_rules
@@ -389,7 +389,7 @@ class _FuncAsCoTransformer(CoTransformer):
             )
         tr = _FuncAsCoTransformer()
         tr._wrapper = DataFrameFunctionWrapper(
-            func, "^(c|[dlspq]+)[fF]?x*z?$", "^[dlspq]$"
+            func, "^(c|[dlspq]+)[fF]?x*z?$", "^[dlspqr]$"
         )
         tr._dfs_input = tr._wrapper.input_code.startswith("c")
         tr._output_schema_arg = schema
@@ -434,7 +434,7 @@ class _FuncAsOutputCoTransformer(_FuncAsCoTransformer):
         validation_rules.update(parse_validation_rules_from_comment(func))
         tr = _FuncAsOutputCoTransformer()
         tr._wrapper = DataFrameFunctionWrapper(
-            func, "^(c|[dlspq]+)[fF]?x*z?$", "^[dlspqn]$"
+            func, "^(c|[dlspq]+)[fF]?x*z?$", "^[dlspqrn]$"
         )
         tr._dfs_input = tr._wrapper.input_code.startswith("c")
         tr._output_schema_arg = None

@@ -1623,7 +1623,47 @@ class HipExecutionEngine(ExecutionEngine):
             res = res.gather_rows(mask.nonzero(as_tuple=True)[0])
         if cols.is_distinct:
             raise DeviceExprError("distinct aggregate: fallback")
-        return res
+        return self._correct_output_types(res, cols, d.schema)
+
+    def _correct_output_types(
+        self, res: HipDataFrame, cols: SelectColumns, in_schema: Schema
+    ) -> HipDataFrame:
+        """Cast output columns whose expression type is known (casts /
+        int aggregates) — the kernels aggregate in fp64 (reference
+        ``correct_select_schema`` semantics)."""
+        m = {
+            pa.int64(): torch.int64,
+            pa.int32(): torch.int32,
+            pa.int16(): torch.int16,
+            pa.float64(): torch.float64,
+            pa.float32(): torch.float32,
+            pa.bool_(): torch.bool,
+        }
+        out: Dict[str, DeviceColumn] = {}
+        fields = []
+        changed = False
+        by_name = {c.output_name or c.name: c for c in cols.all_cols}
+        for f in res.schema.fields:
+            src = res.col(f.name)
+            expr = by_name.get(f.name)
+            tp = expr.infer_type(in_schema) if expr is not None else None
+            if (
+                tp is not None
+                and tp != f.type
+                and tp in m
+                and not isinstance(src, StringDeviceColumn)
+            ):
+                out[f.name] = DeviceColumn(
+                    src.data.to(m[tp]), src.valid, tp
+                )
+                fields.append(pa.field(f.name, tp))
+                changed = True
+            else:
+                out[f.name] = src
+                fields.append(f)
+        if not changed:
+            return res
+        return HipDataFrame.from_columns(out, Schema(fields), self._device)
 
     @staticmethod
     def _pa_type_of_tensor(data: "torch.Tensor") -> pa.DataType:
@@ -1983,7 +2023,7 @@ class HipExecutionEngine(ExecutionEngine):
             res = res.gather_rows(mask.nonzero(as_tuple=True)[0])
         if cols.is_distinct:
             raise DeviceExprError("distinct aggregate: fallback")
-        return res
+        return self._correct_output_types(res, cols, d.schema)
 
     def _merge_partials(
         self,
@@ -2112,9 +2152,11 @@ class HipExecutionEngine(ExecutionEngine):
         fmt = infer_format(path, format_hint if format_hint else None) if (
             format_hint or "." in os.path.basename(path)
         ) else "parquet"
+        part_keys = list(partition_spec.partition_by) if partition_spec else []
         if (
             self.is_distributed
             and not force_single
+            and len(part_keys) == 0
             and fmt in ("parquet", "csv", "json")
         ):
             # each rank writes its shard as a part file (no gather)
@@ -2147,9 +2189,16 @@ class HipExecutionEngine(ExecutionEngine):
             self._comm.barrier()
             return
         local_df = self._as_local(d)  # gathered on every rank
+        keys = list(partition_spec.partition_by) if partition_spec else []
         if self.rank == 0:
-            _io.save_df(
-                local_df.as_pandas(), d.schema, path,
-                format_hint=format_hint, mode=mode, **kwargs
-            )
+            if len(keys) > 0 and not force_single:
+                _io.save_df_partitioned(
+                    local_df.as_pandas(), d.schema, path, keys,
+                    format_hint=format_hint, mode=mode, **kwargs
+                )
+            else:
+                _io.save_df(
+                    local_df.as_pandas(), d.schema, path,
+                    format_hint=format_hint, mode=mode, **kwargs
+                )
         self._comm.barrier()

@@ -17,12 +17,23 @@ _plugins_done = [False]
 FUGUE_ENTRYPOINT_GROUPS = ("fugue.plugins", "fugue_amd.plugins")
 
 
+def _make_pandas_sql(engine: Any, conf: Any = None, **kwargs: Any):
+    from fugue_amd.execution.native_execution_engine import PandasSQLEngine
+
+    return PandasSQLEngine(engine, **kwargs)
+
+
 def register_builtins() -> None:
     if _done[0]:
         return
     _done[0] = True
+    from fugue_amd.execution.factory import register_sql_engine
+
     for alias in ("native", "pandas"):
         register_execution_engine(alias, _make_native, on_dup="ignore")
+    # the local pandas SQL facet (reference alias "qpdpandas")
+    for alias in ("native", "pandas", "qpdpandas"):
+        register_sql_engine(alias, _make_pandas_sql, on_dup="ignore")
 
 
 def load_entry_point_plugins() -> int:

@@ -88,6 +88,13 @@ _TYPE_TO_EXPR: Dict[pa.DataType, str] = {
 }
 
 
+def _quote_name(name: str) -> str:
+    """Backtick-quote a column name when it isn't a plain identifier."""
+    if name != "" and all(c.isalnum() or c == "_" for c in name):
+        return name
+    return "`" + name + "`"
+
+
 def _type_to_expression(tp: pa.DataType) -> str:
     if tp in _TYPE_TO_EXPR:
         return _TYPE_TO_EXPR[tp]
@@ -108,7 +115,9 @@ def _type_to_expression(tp: pa.DataType) -> str:
             + ">"
         )
     if pa.types.is_struct(tp):
-        inner = ",".join(f.name + ":" + _type_to_expression(f.type) for f in tp)
+        inner = ",".join(
+            _quote_name(f.name) + ":" + _type_to_expression(f.type) for f in tp
+        )
         return "{" + inner + "}"
     raise ValueError(f"can't convert {tp} to schema expression")
 
@@ -132,10 +141,23 @@ class _Tokenizer:
 
     def read_name(self) -> str:
         self.skip_ws()
+        if self.peek() == "`":
+            # backtick-quoted name: any characters until the closing tick
+            self.next()
+            start = self.pos
+            while self.peek() != "" and self.peek() != "`":
+                self.pos += 1
+            if self.peek() != "`":
+                raise SyntaxError(f"unclosed ` at {start} in {self.expr!r}")
+            name = self.expr[start : self.pos]
+            self.next()
+            if name == "":
+                raise SyntaxError(f"empty quoted name at {start}")
+            return name
         start = self.pos
-        while self.peek() and (self.peek().isalnum() or self.peek() in "_`"):
+        while self.peek() and (self.peek().isalnum() or self.peek() == "_"):
             self.pos += 1
-        name = self.expr[start : self.pos].strip("`")
+        name = self.expr[start : self.pos]
         if name == "":
             raise SyntaxError(f"expected name at {start} in {self.expr!r}")
         return name
@@ -253,7 +275,9 @@ def expression_to_schema(expr: str) -> pa.Schema:
 
 
 def schema_to_expression(schema: pa.Schema) -> str:
-    return ",".join(f.name + ":" + _type_to_expression(f.type) for f in schema)
+    return ",".join(
+        _quote_name(f.name) + ":" + _type_to_expression(f.type) for f in schema
+    )
 
 
 def to_pa_datatype(obj: Any) -> pa.DataType:
@@ -314,7 +338,9 @@ class Schema:
         if len(set(names)) != len(names):
             raise SchemaError(f"duplicate columns in {names}")
         for n in names:
-            if n == "" or n.strip() == "":
+            # whitespace-only names are legal when quoted (`` ` ``);
+            # only truly empty names are invalid
+            if n == "":
                 raise SchemaError("empty column name")
         self._schema = pa.schema(fields)
         self._index = {f.name: i for i, f in enumerate(fields)}

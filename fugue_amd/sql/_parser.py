@@ -258,12 +258,12 @@ class FugueSQLParser:
                 elif from_context and self.hooks.has_var(t.value):
                     parts.append((False, self.code[seg_start : t.pos]))
                     ts.next()
-                    ref_end = t.pos + len(t.value)
+                    ref_end = t.src_end
                     if ts.match_punct("["):
                         # dfs[key] / dfs[0]: resolve to the object here
                         obj = self._maybe_index(self.hooks.get_var(t.value))
                         last = ts.tokens[ts.pos - 1]  # the closing ']'
-                        ref_end = last.pos + len(last.value)
+                        ref_end = last.src_end
                         parts.append((True, obj))
                     else:
                         parts.append((True, t.value))
@@ -291,7 +291,7 @@ class FugueSQLParser:
                     end = seg_start
                     continue
             ts.next()
-            end = t.pos + len(t.value)
+            end = t.src_end
         parts.append((False, self.code[seg_start:end]))
         return self.hooks.select_statement(
             [(r, v) for r, v in parts if r or v != ""],
@@ -301,7 +301,7 @@ class FugueSQLParser:
 
     def _token_end(self, idx: int) -> int:
         t = self.ts.tokens[idx]
-        return t.pos + len(t.value)
+        return t.src_end
 
     def _starts_line(self, t: Token) -> bool:
         """Whether this token is the first non-whitespace on its line."""
@@ -584,7 +584,10 @@ class FugueSQLParser:
         if ts.take_kw("COLUMNS"):
             columns = self._parse_schema()
             if ":" not in columns:
-                columns = [c.strip() for c in columns.split(",")]
+                # a bare column list; names may be backtick-quoted
+                columns = [
+                    c.strip().strip("`") for c in columns.split(",")
+                ]
         return self.hooks.sql_load(path, fmt, columns, params)
 
     def _parse_transform(self, output: bool) -> Any:

@@ -9,10 +9,16 @@ class Token(NamedTuple):
     kind: str  # KW, NAME, NUMBER, STRING, OP, PUNCT
     value: str
     pos: int
+    end: int = -1  # source end offset (pos + raw length; value may be
+    #              shorter than the raw text for `quoted` names)
 
     @property
     def upper(self) -> str:
         return self.value.upper()
+
+    @property
+    def src_end(self) -> int:
+        return self.end if self.end >= 0 else self.pos + len(self.value)
 
 
 _TOKEN_RE = re.compile(
@@ -58,7 +64,7 @@ def _tokenize_uncached(sql: str) -> List[Token]:
         if kind not in ("WS", "COMMENT"):
             if kind == "NAME" and value.startswith("`"):
                 value = value[1:-1]
-            tokens.append(Token(kind, value, pos))
+            tokens.append(Token(kind, value, pos, m.end()))
         pos = m.end()
     return tokens
 

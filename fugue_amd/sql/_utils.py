@@ -12,7 +12,7 @@ def fill_sql_template(sql: str, params: Dict[str, Any]) -> str:
     try:
         from jinja2 import Template
 
-        return Template(sql).render(**params)
+        return Template(sql).render(params)
     except ImportError:
         def _sub(m: "re.Match") -> str:
             name = m.group(1)

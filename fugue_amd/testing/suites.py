@@ -354,7 +354,7 @@ class BuiltInWorkflowTestSuite:
         a = dag.df([[1, "a"]], "x:long,y:str")
         a.yield_dataframe_as("r")
         res = self.run_dag(dag)
-        assert res["r"].result.as_array() == [[1, "a"]]
+        assert res["r"].as_array() == [[1, "a"]]
 
     def test_transformer_styles(self):
         # pandas in/out
@@ -378,9 +378,9 @@ class BuiltInWorkflowTestSuite:
         a.transform(t2, schema="*,z:long").yield_dataframe_as("r2")
         a.transform(t3, schema="x:long,z:long").yield_dataframe_as("r3")
         res = self.run_dag(dag)
-        assert sorted(res["r1"].result.as_array()) == [[1, 2], [2, 4]]
-        assert sorted(res["r2"].result.as_array()) == [[1, 2], [2, 3]]
-        assert sorted(res["r3"].result.as_array()) == [[1, 10], [2, 20]]
+        assert sorted(res["r1"].as_array()) == [[1, 2], [2, 4]]
+        assert sorted(res["r2"].as_array()) == [[1, 2], [2, 3]]
+        assert sorted(res["r3"].as_array()) == [[1, 10], [2, 20]]
 
     def test_transform_api(self):
         pdf = pd.DataFrame(dict(g=["a", "a", "b"], v=[1, 2, 3]))
@@ -404,7 +404,7 @@ class BuiltInWorkflowTestSuite:
         r = dag.select("SELECT k, SUM(v) AS s FROM", a, "GROUP BY k")
         r.yield_dataframe_as("r")
         res = self.run_dag(dag)
-        assert sorted(res["r"].result.as_array()) == [[1, 6.0], [2, 6.0]]
+        assert sorted(res["r"].as_array()) == [[1, 6.0], [2, 6.0]]
 
     def test_checkpoint_and_persist(self):
         with tempfile.TemporaryDirectory() as tmp:
@@ -415,7 +415,7 @@ class BuiltInWorkflowTestSuite:
                 self.engine,
                 {"fugue.workflow.checkpoint.path": tmp},
             )
-            assert res["r"].result.as_array() == [[1]]
+            assert res["r"].as_array() == [[1]]
 
     def test_callbacks(self):
         collected = []
@@ -445,7 +445,7 @@ class BuiltInWorkflowTestSuite:
             bad, schema="*", ignore_errors=[ValueError]
         ).yield_dataframe_as("r")
         res = self.run_dag(dag)
-        assert res["r"].result.as_array() == [[1, 2]]
+        assert res["r"].as_array() == [[1, 2]]
 
     def test_out_transform(self):
         side: List[int] = []
@@ -466,8 +466,8 @@ class BuiltInWorkflowTestSuite:
         a.inner_join(b).yield_dataframe_as("j")
         a.union(a, distinct=False).yield_dataframe_as("u")
         res = self.run_dag(dag)
-        assert res["j"].result.as_array() == [[1, "x", 5.0]]
-        assert res["u"].result.count() == 2
+        assert res["j"].as_array() == [[1, "x", 5.0]]
+        assert res["u"].count() == 2
 
 
 class DataFrameTestSuite:

@@ -62,9 +62,15 @@ _SMALL_FRAME_ROWS = 10_000
 
 def _is_raw_frame(obj: Any) -> bool:
     cls = type(obj).__module__ + "." + type(obj).__name__
-    return cls in (
-        "pandas.core.frame.DataFrame",
-        "pyarrow.lib.Table",
+    if cls in ("pandas.core.frame.DataFrame", "pyarrow.lib.Table"):
+        return True
+    # fugue local bounded frames: identity must include the data, not
+    # just the repr/schema (two distinct literals must hash apart)
+    return (
+        hasattr(obj, "as_array")
+        and hasattr(obj, "schema")
+        and getattr(obj, "is_local", False)
+        and getattr(obj, "is_bounded", False)
     )
 
 
@@ -75,8 +81,9 @@ def _frame_token(obj: Any) -> str:
     such tasks are simply not deterministic across runs)."""
     try:
         n = len(obj)
-    except TypeError:  # pyarrow Table has num_rows
-        n = obj.num_rows
+    except TypeError:
+        # pyarrow Table has num_rows; fugue frames have count()
+        n = obj.num_rows if hasattr(obj, "num_rows") else obj.count()
     if n <= _SMALL_FRAME_ROWS:
         try:
             import pandas as pd
@@ -85,7 +92,10 @@ def _frame_token(obj: Any) -> str:
                 return (
                     "c" + str(list(obj.columns)) + repr(obj.values.tolist())
                 )
-            return "c" + str(obj.schema) + repr(obj.to_pylist())
+            if hasattr(obj, "to_pylist"):  # pyarrow Table
+                return "c" + str(obj.schema) + repr(obj.to_pylist())
+            # fugue local bounded frame
+            return "c" + str(obj.schema) + repr(obj.as_array())
         except Exception:
             pass
     return "i" + str(id(obj))

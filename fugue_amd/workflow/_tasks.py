@@ -188,6 +188,7 @@ class Process(FugueTask):
         self._bind(ctx)
         self._extension.validate_on_compile()
         dfs = self._collect_inputs(ctx)
+        self._extension.validate_on_runtime(dfs)
         df = self._extension.process(dfs)
         self.set_result(ctx, ctx.execution_engine.to_df(df))
 
@@ -217,6 +218,7 @@ class Output(FugueTask):
         self._bind(ctx)
         self._extension.validate_on_compile()
         dfs = DataFrames([t.result for t in self._inputs])
+        self._extension.validate_on_runtime(dfs)
         self._extension.process(dfs)
         self._executed = True
         # outputs pass through their first input so downstream `show` chains work

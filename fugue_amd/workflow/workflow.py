@@ -76,6 +76,21 @@ from fugue_amd.workflow._workflow_context import FugueWorkflowContext
 _DEFAULT_IGNORE_ERRORS: List[Any] = []
 
 
+def _compile_validate(ext: Any, spec: PartitionSpec, callback: Any) -> None:
+    """DAG-build-time validation: partition-spec rules
+    (``partitionby_has``/``partitionby_is``) and required-callback checks
+    fail at compile, before any engine runs (reference
+    ``builtin_suite`` validation tests)."""
+    from fugue_amd.exceptions import FugueInterfacelessError
+
+    ext._partition_spec = spec
+    ext.validate_on_compile()
+    if getattr(ext, "_requires_callback", False) and callback is None:
+        raise FugueInterfacelessError(
+            f"{ext} requires a callback but none was provided"
+        )
+
+
 class WorkflowDataFrame:
     """A lazy node handle in the workflow DAG."""
 
@@ -103,6 +118,11 @@ class WorkflowDataFrame:
 
     def __uuid__(self) -> str:
         return to_uuid(self._task.__uuid__(), self._metadata.get("pre_partition", ""))
+
+    def spec_uuid(self) -> str:
+        """Deterministic spec id of this node (reference
+        ``workflow.py`` WorkflowDataFrame.spec_uuid)."""
+        return self._task.__uuid__()
 
     @property
     def result(self) -> DataFrame:
@@ -216,7 +236,13 @@ class WorkflowDataFrame:
     def semi_join(self, *dfs: Any, on: Optional[List[str]] = None) -> "WorkflowDataFrame":
         return self.join(*dfs, how="semi", on=on)
 
+    def left_semi_join(self, *dfs: Any, on: Optional[List[str]] = None) -> "WorkflowDataFrame":
+        return self.join(*dfs, how="semi", on=on)
+
     def anti_join(self, *dfs: Any, on: Optional[List[str]] = None) -> "WorkflowDataFrame":
+        return self.join(*dfs, how="anti", on=on)
+
+    def left_anti_join(self, *dfs: Any, on: Optional[List[str]] = None) -> "WorkflowDataFrame":
         return self.join(*dfs, how="anti", on=on)
 
     def left_outer_join(self, *dfs: Any, on: Optional[List[str]] = None) -> "WorkflowDataFrame":
@@ -269,6 +295,8 @@ class WorkflowDataFrame:
         replace: bool = False,
         seed: Optional[int] = None,
     ) -> "WorkflowDataFrame":
+        if (n is None) == (frac is None):
+            raise ValueError("one and only one of n and frac must be set")
         params: Dict[str, Any] = dict(replace=replace)
         if n is not None:
             params["n"] = n
@@ -279,6 +307,10 @@ class WorkflowDataFrame:
         return self._op(Sample, params)
 
     def take(self, n: int, presort: str = "", na_position: str = "last") -> "WorkflowDataFrame":
+        if not isinstance(n, int):
+            raise ValueError("n must be an integer")
+        if na_position not in ("first", "last"):
+            raise ValueError("na_position must be 'first' or 'last'")
         task = Process(
             Take(),
             [self.task],
@@ -409,12 +441,43 @@ class WorkflowDataFrame:
         self._task.broadcast()
         return self
 
+    def _assert_can_yield_physical(self) -> None:
+        # physical yields are allowed on unpersisted nodes (the yield
+        # itself becomes a strong checkpoint) or on deterministic
+        # checkpoints; any other checkpoint state is a compile error
+        # (persist: ValueError — reference workflow.py:987 note)
+        ck = self._task._checkpoint
+        if ck.is_null or ck.deterministic:
+            return
+        if not ck.to_file:
+            raise ValueError(
+                "can't yield file/table after persist; use "
+                "deterministic_checkpoint"
+            )
+        raise FugueWorkflowCompileError(
+            "can't yield file/table after a non-deterministic checkpoint"
+        )
+
+    def _physical_yield_key(self) -> str:
+        # a direct yield is not reusable across executions, so its
+        # identity is random per DAG build; yielding a deterministic
+        # checkpoint is stable (consumers of dag.yields[...] then hash
+        # identically across builds — reference test_yield_file)
+        ck = self._task._checkpoint
+        if not ck.is_null and ck.deterministic:
+            return self.__uuid__()
+        import uuid as _uuid_mod
+
+        return str(_uuid_mod.uuid4())
+
     def yield_file_as(self, name: str) -> None:
-        y = PhysicalYielded(self.__uuid__(), "file")
+        self._assert_can_yield_physical()
+        y = PhysicalYielded(self._physical_yield_key(), "file")
         self.workflow._register_yield(name, y, self._task)
 
     def yield_table_as(self, name: str) -> None:
-        y = PhysicalYielded(self.__uuid__(), "table")
+        self._assert_can_yield_physical()
+        y = PhysicalYielded(self._physical_yield_key(), "table")
         self.workflow._register_yield(name, y, self._task)
 
     def yield_dataframe_as(self, name: str, as_local: bool = False) -> None:
@@ -515,7 +578,10 @@ class FugueWorkflowResult:
         return self._yields
 
     def __getitem__(self, name: str) -> Any:
-        return self._yields[name]
+        y = self._yields[name]
+        # dataframe yields resolve to their result (reference
+        # FugueWorkflowResult is a DataFrames of results)
+        return y.result if isinstance(y, YieldedDataFrame) else y
 
 
 class FugueWorkflow:
@@ -541,6 +607,11 @@ class FugueWorkflow:
     def yields(self) -> Dict[str, Yielded]:
         return self._yields
 
+    def spec_uuid(self) -> str:
+        """Deterministic id of the whole DAG spec: stable across runs
+        for identical workflows (reference FugueWorkflow.spec_uuid)."""
+        return to_uuid([t.__uuid__() for t in self._tasks.values()])
+
     @property
     def last_df(self) -> Optional[WorkflowDataFrame]:
         return self._last_df
@@ -549,8 +620,9 @@ class FugueWorkflow:
         return self
 
     def __exit__(self, exc_type: Any, exc_val: Any, exc_tb: Any) -> None:
-        if exc_type is None:
-            self.run()
+        # the `with` form only scopes the DAG build; execution is the
+        # explicit dag.run(...) (reference ``workflow.py:1619``)
+        return
 
     def add(self, task: FugueTask) -> WorkflowDataFrame:
         """Add a task, dedup by spec uuid (determinism)."""
@@ -642,6 +714,13 @@ class FugueWorkflow:
     def create(
         self, using: Any, schema: Any = None, params: Any = None
     ) -> WorkflowDataFrame:
+        # a dataframe object used as a "creator" is just data: emit the
+        # same CreateData task as dag.df so the spec uuids coincide
+        import pandas as _pd
+        import pyarrow as _pa
+
+        if isinstance(using, (DataFrame, Yielded, _pd.DataFrame, _pa.Table)):
+            return self.create_data(using, schema)
         creator = _to_creator(using, schema)
         task = Create(creator, params=dict(params=ParamDict(params)))
         return self.add(task)
@@ -670,6 +749,7 @@ class FugueWorkflow:
     ) -> WorkflowDataFrame:
         _dfs = self._build_dataframes(dfs)
         proc = _to_processor(using, schema)
+        _compile_validate(proc, PartitionSpec(pre_partition), None)
         task = Process(
             proc,
             [d.task for d in _dfs.values()],
@@ -684,6 +764,7 @@ class FugueWorkflow:
     ) -> None:
         _dfs = self._build_dataframes(dfs)
         out = _to_outputter(using)
+        _compile_validate(out, PartitionSpec(pre_partition), None)
         task = Output(
             out,
             [d.task for d in _dfs.values()],
@@ -705,6 +786,8 @@ class FugueWorkflow:
         if len(dfs) != 1:
             raise NotImplementedError("transform can only take one input dataframe")
         tf = _to_transformer(using, schema)
+        spec = PartitionSpec(pre_partition)
+        _compile_validate(tf, spec, callback)
         handler = to_rpc_handler(callback)
         _dfs = self._build_dataframes(dfs)
         wdf = list(_dfs.values())[0]
@@ -738,6 +821,7 @@ class FugueWorkflow:
         if len(dfs) != 1:
             raise NotImplementedError("transform can only take one input dataframe")
         tf = _to_output_transformer(using)
+        _compile_validate(tf, PartitionSpec(pre_partition), callback)
         handler = to_rpc_handler(callback)
         _dfs = self._build_dataframes(dfs)
         wdf = list(_dfs.values())[0]
@@ -817,6 +901,16 @@ class FugueWorkflow:
                     names[tid] = f"_fugue_tmp_{len(names)}"
                     deps.append(wdf.task)
                 parts.append((True, names[tid]))
+        # "SELECT" may be omitted (reference: dag.select("* FROM", df))
+        first_str = next((p1 for has, p1 in parts if not has), "")
+        first_tok = first_str.strip().split(" ")[0].upper() if first_str.strip() else ""
+        lead_is_df = len(parts) > 0 and parts[0][0]
+        if first_tok != "SELECT" or (lead_is_df and parts[0][0]):
+            if not any(
+                not has and p1.strip().upper().startswith("SELECT")
+                for has, p1 in parts[:1]
+            ):
+                parts = [(False, "SELECT ")] + parts
         statement = StructuredRawSQL(parts, dialect=dialect)
         task = Process(
             RunSQLSelect(),
@@ -886,6 +980,21 @@ class FugueWorkflow:
         return FugueWorkflowResult(self._yields)
 
     def _build_dataframes(self, dfs: Tuple[Any, ...]) -> WorkflowDataFrames:
+        if (
+            len(dfs) == 1
+            and isinstance(dfs[0], (list, tuple))
+            and not (
+                isinstance(dfs[0], tuple)
+                and len(dfs[0]) == 2
+                and isinstance(dfs[0][0], str)
+            )
+            and all(
+                isinstance(x, (WorkflowDataFrame, DataFrame, Yielded))
+                for x in dfs[0]
+            )
+        ):
+            # a single list argument is the unpacked input set
+            dfs = tuple(dfs[0])
         if len(dfs) == 1 and isinstance(dfs[0], dict):
             res = WorkflowDataFrames()
             for k, v in dfs[0].items():

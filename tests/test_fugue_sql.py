@@ -100,7 +100,7 @@ def test_yield_and_checkpoint():
         """
     )
     res = dag.run()
-    assert sorted(r[0] for r in res["out"].result.as_array()) == [2, 3]
+    assert sorted(r[0] for r in res["out"].as_array()) == [2, 3]
 
 
 def test_print_and_outtransform(capsys):
@@ -347,12 +347,12 @@ def test_sql_module_sub():
         out1=out1,
     )
     res = dag.run()
-    assert res["r1"].result.as_array() == [[1]]
-    assert res["r2"].result.as_array() == [[2]]
+    assert res["r1"].as_array() == [[1]]
+    assert res["r2"].as_array() == [[2]]
     # merge2 selects dfs[1] == b == [[2]]; out1 prints the last frame (m2)
     assert collected == [[2]]
     # merge3 named refs: df1=b, df2=a → positional order df1,df2 = b,a
-    assert res["r3"].result.as_array() == [[2]]
+    assert res["r3"].as_array() == [[2]]
 
 
 def test_distinct_aggregates_sql_executor():

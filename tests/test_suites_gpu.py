@@ -99,3 +99,16 @@ class TestHipGpuExecutionConformance(ExecutionEngineConformance):
         from fugue_amd.hip.execution_engine import HipExecutionEngine
 
         return HipExecutionEngine()
+
+
+from fugue_amd.testing.builtin_conformance import BuiltInConformance
+
+
+class TestHipGpuBuiltInConformance(BuiltInConformance):
+    """All 45 reference BuiltInTests cases on device tensors."""
+
+    @classmethod
+    def make_engine(cls):
+        from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+        return HipExecutionEngine()

@@ -23,7 +23,7 @@ def test_basic_dag_run():
     b = a.filter(col("x") > 1)
     b.yield_dataframe_as("r")
     res = dag.run()
-    assert res["r"].result.as_array() == [[2], [3]]
+    assert res["r"].as_array() == [[2], [3]]
 
 
 def test_dag_determinism():
@@ -130,8 +130,8 @@ def test_workflow_join_setops():
     u = a.union(a, distinct=False)
     u.yield_dataframe_as("u")
     res = dag.run()
-    assert res["j"].result.as_array() == [[2, "y", 5.0]]
-    assert res["u"].result.count() == 4
+    assert res["j"].as_array() == [[2, "y", 5.0]]
+    assert res["u"].count() == 4
 
 
 def test_checkpoints_and_persist():
@@ -141,7 +141,7 @@ def test_checkpoints_and_persist():
         b = a.strong_checkpoint()
         b.yield_dataframe_as("r")
         res = dag.run(None, {FUGUE_CONF_WORKFLOW_CHECKPOINT_PATH: tmp})
-        assert res["r"].result.as_array() == [[1]]
+        assert res["r"].as_array() == [[1]]
 
 
 def test_deterministic_checkpoint_reuse():
@@ -153,7 +153,7 @@ def test_deterministic_checkpoint_reuse():
             a = dag.df([[1]], "x:long")
             b = a.transform(_double_x, schema="*").deterministic_checkpoint()
             b.yield_dataframe_as("r")
-            return dag.run(None, conf)["r"].result.as_array()
+            return dag.run(None, conf)["r"].as_array()
 
         assert run_once() == [[2]]
         files = os.listdir(tmp)
@@ -191,7 +191,7 @@ def test_zip_comap_workflow():
     r = z.transform(merge_counts, schema="n1:long,n2:long")
     r.yield_dataframe_as("r")
     res = dag.run()
-    assert res["r"].result.as_array() == [[1, 2]]
+    assert res["r"].as_array() == [[1, 2]]
 
 
 def test_workflow_parallelism():
@@ -201,7 +201,7 @@ def test_workflow_parallelism():
         a.transform(_double_x, schema="*").yield_dataframe_as(f"r{i}")
     res = dag.run(None, {"fugue.workflow.concurrency": 4})
     for i in range(5):
-        assert res[f"r{i}"].result.as_array() == [[2]]
+        assert res[f"r{i}"].as_array() == [[2]]
 
 
 def test_runtime_exception_traceback():
@@ -230,7 +230,7 @@ def test_yield_file_and_table_cross_workflow():
         b = dag2.df(res1["t1"])
         a.union(b, distinct=False).yield_dataframe_as("out")
         res2 = dag2.run(engine)
-        assert sorted(r[0] for r in res2["out"].result.as_array()) == [1, 2, 3]
+        assert sorted(r[0] for r in res2["out"].as_array()) == [1, 2, 3]
 
 
 def test_auto_persist():
@@ -249,8 +249,8 @@ def test_auto_persist():
     b = a.filter(col("x") >= 0)
     b.yield_dataframe_as("r2")
     res = dag.run()
-    assert res["r1"].result.as_array() == [[2]]
-    assert res["r2"].result.as_array() == [[1]]
+    assert res["r1"].as_array() == [[2]]
+    assert res["r2"].as_array() == [[1]]
     # `a` is consumed twice; the auto-persist marks it with a weak
     # checkpoint so its transform runs once
     assert len(calls) == 1
@@ -278,8 +278,8 @@ def test_module_composition():
     doubled(a).yield_dataframe_as("d")
     a.tripled().yield_dataframe_as("t")
     res = dag.run()
-    assert res["d"].result.as_array() == [[2]]
-    assert res["t"].result.as_array() == [[3]]
+    assert res["d"].as_array() == [[2]]
+    assert res["t"].as_array() == [[3]]
 
 
 def test_save_and_use():
@@ -297,7 +297,7 @@ def test_save_and_use():
         dag2 = FugueWorkflow()
         dag2.load(path).yield_dataframe_as("r")
         res2 = dag2.run()
-        assert sorted(r[0] for r in res2["r"].result.as_array()) == [1, 2]
+        assert sorted(r[0] for r in res2["r"].as_array()) == [1, 2]
 
 
 def test_transform_iterable_dfs_output():
@@ -363,7 +363,7 @@ def test_cotransformer_decorator():
     z = dag.zip(a, b)  # inner: only k=1 survives
     z.transform(merge).yield_dataframe_as("r")
     res = dag.run()
-    assert res["r"].result.as_array() == [[1, 2, 1]]
+    assert res["r"].as_array() == [[1, 2, 1]]
 
 
 def test_zip_to_file_threshold(tmp_path):
@@ -419,7 +419,7 @@ def test_namespace_extension_plugins():
     dag = FugueWorkflow()
     dag.create(("myns", "7")).yield_dataframe_as("r")
     res = dag.run()
-    assert res["r"].result.as_array() == [[7]]
+    assert res["r"].as_array() == [[7]]
 
 
 def test_fugue_test_suite_base():
