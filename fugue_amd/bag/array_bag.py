@@ -27,8 +27,13 @@ class ArrayBag(LocalBag):
 
     def peek(self) -> Any:
         if self.empty:
-            raise IndexError("bag is empty")
+            from fugue_amd.exceptions import FugueDatasetEmptyError
+
+            raise FugueDatasetEmptyError("bag is empty")
         return self._native[0]
+
+    def head(self, n: int) -> "ArrayBag":
+        return ArrayBag(self._native[:n])
 
     def as_array(self) -> List[Any]:
         return list(self._native)

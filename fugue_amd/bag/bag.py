@@ -7,6 +7,15 @@ from fugue_amd.utils.registry import register_plugin
 
 
 class Bag(Dataset):
+    def __copy__(self) -> "Bag":
+        return self
+
+    def __deepcopy__(self, memo: Any) -> "Bag":
+        return self
+
+    def head(self, n: int) -> "LocalBag":
+        raise NotImplementedError  # pragma: no cover
+
     @abstractmethod
     def as_local(self) -> "LocalBag":
         ...

@@ -69,22 +69,25 @@ class ArrayDataFrame(LocalBoundedDataFrame):
     ) -> List[Any]:
         if columns is None and not type_safe:
             return self._native
-        pos = self._pos(columns)
-        rows = [[row[i] for i in pos] for row in self._native]
         if type_safe:
-            from fugue_amd.dataframe.coerce import coerce_rows
+            # round-trip through arrow: enforces the schema exactly
+            # (NaT/NaN -> None, string datetimes parsed, struct fields
+            # normalized) — reference triad semantics
+            from fugue_amd.dataframe.arrow_dataframe import ArrowDataFrame
 
-            sub = (
-                self.schema if columns is None else self.schema.extract(columns)
+            return ArrowDataFrame(self.as_arrow()).as_array(
+                columns, type_safe=True
             )
-            rows = coerce_rows(rows, sub)
-        return rows
+        pos = self._pos(columns)
+        return [[row[i] for i in pos] for row in self._native]
 
     def as_array_iterable(
         self, columns: Optional[List[str]] = None, type_safe: bool = False
     ) -> Iterable[Any]:
         if columns is None and not type_safe:
             yield from self._native
+        elif type_safe:
+            yield from self.as_array(columns, type_safe=True)
         else:
             pos = self._pos(columns)
             for row in self._native:

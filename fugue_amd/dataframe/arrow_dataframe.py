@@ -123,7 +123,21 @@ class ArrowDataFrame(LocalBoundedDataFrame):
         schema = self._get_altered_schema(columns)
         if schema == self.schema:
             return self
-        return ArrowDataFrame(self._native.cast(schema.pa_schema), schema)
+        # per-column cast via the pandas casting rules so temporal->str
+        # uses python formatting, str->bool is case-insensitive, etc.
+        from fugue_amd.utils.pandas_like import _cast_series
+
+        arrays = []
+        for f in schema.fields:
+            col = self._native.column(f.name).combine_chunks()
+            if col.type == f.type:
+                arrays.append(col)
+            else:
+                casted = _cast_series(col.to_pandas(), f.type)
+                arrays.append(pa.Array.from_pandas(casted, type=f.type))
+        return ArrowDataFrame(
+            pa.Table.from_arrays(arrays, schema=schema.pa_schema), schema
+        )
 
     def head(
         self, n: int, columns: Optional[List[str]] = None

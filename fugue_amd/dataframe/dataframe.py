@@ -115,7 +115,10 @@ class DataFrame(Dataset):
         ...
 
     def drop(self, columns: List[str]) -> "DataFrame":
-        schema = self.schema.exclude(columns)
+        try:
+            schema = self.schema.exclude(columns)
+        except Exception as e:
+            raise FugueDataFrameOperationError(str(e)) from e
         if len(schema) == 0:
             raise FugueDataFrameOperationError("can't drop all columns")
         if len(schema) + len(columns) != len(self.schema):
@@ -127,7 +130,14 @@ class DataFrame(Dataset):
     def __getitem__(self, columns: List[Any]) -> "DataFrame":
         if not isinstance(columns, list):
             columns = [columns]
-        schema = self.schema.extract(columns)
+        try:
+            schema = self.schema.extract(columns)
+            if len(schema) == 0:
+                raise FugueDataFrameOperationError("can't select no columns")
+        except FugueDataFrameOperationError:
+            raise
+        except Exception as e:
+            raise FugueDataFrameOperationError(str(e)) from e
         return self._select_cols(schema.names)
 
     @abstractmethod
@@ -143,7 +153,7 @@ class DataFrame(Dataset):
         self, columns: Optional[List[str]] = None
     ) -> Iterable[Dict[str, Any]]:
         cols = columns if columns is not None else self.columns
-        for row in self.as_array_iterable(columns):
+        for row in self.as_array_iterable(columns, type_safe=True):
             yield {n: row[i] for i, n in enumerate(cols)}
 
     def get_info_str(self) -> str:

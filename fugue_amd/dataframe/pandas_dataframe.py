@@ -52,7 +52,13 @@ class PandasDataFrame(LocalBoundedDataFrame):
                 pdf = cast_pandas(pdf, schema)
             elif isinstance(df, (list, tuple)) or isinstance(df, Iterable):
                 schema = Schema(schema).assert_not_empty()
-                rows = [{c: row[i] for i, c in enumerate(schema.names)} for row in df]
+                from fugue_amd.dataframe.coerce import coerce_rows
+
+                crows = coerce_rows([list(r) for r in df], schema)
+                rows = [
+                    {c: row[i] for i, c in enumerate(schema.names)}
+                    for row in crows
+                ]
                 pdf = pa.Table.from_pylist(rows, schema=schema.pa_schema).to_pandas()
             else:
                 raise ValueError(f"{type(df)} is incompatible with PandasDataFrame")

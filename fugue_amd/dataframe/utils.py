@@ -91,7 +91,15 @@ def _df_eq(
         else:
             df2 = ArrayDataFrame(data, schema if schema is not None else df.schema)
         if check_schema and df1.schema != df2.schema:
-            raise AssertionError(f"schema mismatch {df1.schema} vs {df2.schema}")
+            # column order is not identity (reference Schema.is_like):
+            # same name->type mapping in any order is a match
+            f1 = {f.name: f.type for f in df1.schema.fields}
+            f2 = {f.name: f.type for f in df2.schema.fields}
+            if f1 != f2:
+                raise AssertionError(
+                    f"schema mismatch {df1.schema} vs {df2.schema}"
+                )
+            df1 = df1[df2.schema.names].as_local_bounded()
         if not check_content:
             return True
         a1 = df1.as_array(type_safe=True)

@@ -49,8 +49,11 @@ def as_local_bounded(data: Any) -> Any:
     ok, res = try_run_plugin("as_local_bounded", data)
     if ok:
         return res
-    ds = as_fugue_dataset(data)
-    if hasattr(ds, "as_local_bounded"):
+    try:
+        ds = as_fugue_dataset(data)
+    except Exception:
+        ds = None
+    if ds is not None and hasattr(ds, "as_local_bounded"):
         return ds.as_local_bounded()
     raise NotImplementedError(
         f"no registered function to convert {type(data)} to a local bounded dataset"

@@ -25,12 +25,13 @@ class FugueDataFrameInitError(FugueDataFrameError):
     """DataFrame construction error"""
 
 
-class FugueDataFrameEmptyError(FugueDataFrameError):
-    """Peeking an empty DataFrame"""
-
-
-class FugueDatasetEmptyError(FugueDataFrameEmptyError):
+class FugueDatasetEmptyError(FugueDataFrameError):
     """Empty dataset error"""
+
+
+class FugueDataFrameEmptyError(FugueDatasetEmptyError):
+    """Peeking an empty DataFrame (subclass of the dataset-level empty
+    error so either type can be caught, as in the reference)"""
 
 
 class FugueWorkflowError(FugueError):

@@ -49,6 +49,8 @@ def supported_device_type(tp: pa.DataType) -> bool:
         or pa.types.is_date(tp)
         or pa.types.is_string(tp)
         or pa.types.is_large_string(tp)
+        or pa.types.is_binary(tp)
+        or pa.types.is_large_binary(tp)
     )
 
 
@@ -114,6 +116,10 @@ class DeviceColumn:
             arr = arr.combine_chunks()
         if pa.types.is_string(tp) or pa.types.is_large_string(tp):
             return StringDeviceColumn.from_arrow_strings(arr, device)
+        if pa.types.is_binary(tp) or pa.types.is_large_binary(tp):
+            return StringDeviceColumn.from_arrow_strings(
+                arr, device, pa_type=pa.binary()
+            )
         np_arr = arr.to_numpy(zero_copy_only=False)
         valid_t: Optional[torch.Tensor] = None
         if arr.null_count > 0:
@@ -165,9 +171,12 @@ class StringDeviceColumn(DeviceColumn):
         offsets: torch.Tensor,
         bytes_: torch.Tensor,
         valid: Optional[torch.Tensor],
+        pa_type: Optional[pa.DataType] = None,
     ):
-        # ``data`` property is the offsets tensor (n+1 elements)
-        super().__init__(offsets, valid, pa.string())
+        # ``data`` property is the offsets tensor (n+1 elements);
+        # pa_type selects string (default) vs binary payloads — the
+        # HBM layout (offsets + byte buffer) is identical
+        super().__init__(offsets, valid, pa_type or pa.string())
         self.offsets = offsets
         self.bytes = bytes_
 
@@ -201,6 +210,7 @@ class StringDeviceColumn(DeviceColumn):
             new_offsets,
             new_bytes,
             None if self.valid is None else self.valid.index_select(0, idx),
+            pa_type=self.pa_type,
         )
 
     def slice(self, start: int, length: int) -> "StringDeviceColumn":
@@ -211,6 +221,7 @@ class StringDeviceColumn(DeviceColumn):
             offs - b0,
             self.bytes[b0:b1],
             None if self.valid is None else self.valid[start : start + length],
+            pa_type=self.pa_type,
         )
 
     def concat_with(self, others: List["DeviceColumn"]) -> "StringDeviceColumn":
@@ -236,8 +247,11 @@ class StringDeviceColumn(DeviceColumn):
         return StringDeviceColumn(offsets, bytes_, valid)
 
     @staticmethod
-    def from_arrow_strings(arr: pa.Array, device: str) -> "StringDeviceColumn":
-        arr = arr.cast(pa.large_string())
+    def from_arrow_strings(
+        arr: pa.Array, device: str, pa_type: Optional[pa.DataType] = None
+    ) -> "StringDeviceColumn":
+        is_bin = pa_type is not None and pa.types.is_binary(pa_type)
+        arr = arr.cast(pa.large_binary() if is_bin else pa.large_string())
         buffers = arr.buffers()
         # buffers: [validity, offsets(int64), data]
         offsets_np = np.frombuffer(
@@ -259,13 +273,19 @@ class StringDeviceColumn(DeviceColumn):
             torch.from_numpy(offsets_np).to(device),
             torch.from_numpy(data_np).to(device),
             valid_t,
+            pa_type=pa_type,
         )
 
     def to_arrow(self) -> pa.Array:
+        is_bin = pa.types.is_binary(self.pa_type) or pa.types.is_large_binary(
+            self.pa_type
+        )
+        big = pa.large_binary() if is_bin else pa.large_string()
+        final = pa.binary() if is_bin else pa.string()
         offsets_np = self.offsets.cpu().numpy()
         bytes_np = self.bytes.cpu().numpy()
         arr = pa.Array.from_buffers(
-            pa.large_string(),
+            big,
             len(self),
             [
                 None,
@@ -273,13 +293,13 @@ class StringDeviceColumn(DeviceColumn):
                 pa.py_buffer(bytes_np.tobytes()),
             ],
         )
-        arr = arr.cast(pa.string())
+        arr = arr.cast(final)
         if self.valid is not None:
             mask = self.valid.cpu().numpy()
             # rebuild with nulls
             py = arr.to_pylist()
             py = [v if m else None for v, m in zip(py, mask)]
-            arr = pa.array(py, type=pa.string())
+            arr = pa.array(py, type=final)
         return arr
 
 
@@ -306,7 +326,8 @@ class SpilledDataFrame(LocalBoundedDataFrame):
             valid = None if c.valid is None else _to_host(c.valid)
             if isinstance(c, StringDeviceColumn):
                 self._host_cols[name] = StringDeviceColumn(
-                    _to_host(c.offsets), _to_host(c.bytes), valid
+                    _to_host(c.offsets), _to_host(c.bytes), valid,
+                    pa_type=c.pa_type,
                 )
             else:
                 self._host_cols[name] = DeviceColumn(
@@ -325,7 +346,7 @@ class SpilledDataFrame(LocalBoundedDataFrame):
             valid = None if c.valid is None else _up(c.valid)
             if isinstance(c, StringDeviceColumn):
                 cols[name] = StringDeviceColumn(
-                    _up(c.offsets), _up(c.bytes), valid
+                    _up(c.offsets), _up(c.bytes), valid, pa_type=c.pa_type
                 )
             else:
                 cols[name] = DeviceColumn(_up(c.data), valid, c.pa_type)
@@ -365,7 +386,10 @@ class SpilledDataFrame(LocalBoundedDataFrame):
     def as_local_bounded(self) -> LocalBoundedDataFrame:
         from fugue_amd.dataframe.arrow_dataframe import ArrowDataFrame
 
-        return ArrowDataFrame(self.as_arrow())
+        res = ArrowDataFrame(self.as_arrow())
+        if self.has_metadata:
+            res.reset_metadata(self.metadata)
+        return res
 
     def as_array(self, columns=None, type_safe: bool = False) -> List[Any]:
         return self._as_host_frame().as_array(columns, type_safe=type_safe)
@@ -537,7 +561,10 @@ class HipDataFrame(LocalBoundedDataFrame):
     def as_local_bounded(self) -> LocalBoundedDataFrame:
         from fugue_amd.dataframe.arrow_dataframe import ArrowDataFrame
 
-        return ArrowDataFrame(self.as_arrow())
+        res = ArrowDataFrame(self.as_arrow())
+        if self.has_metadata:
+            res.reset_metadata(self.metadata)
+        return res
 
     def as_array(
         self, columns: Optional[List[str]] = None, type_safe: bool = False
@@ -579,9 +606,14 @@ class HipDataFrame(LocalBoundedDataFrame):
         schema = self._get_altered_schema(columns)
         if schema == self.schema:
             return self
-        # cast through arrow (rare path; device-native casts later)
+        # cast through the host frame (pandas casting rules: str(datetime)
+        # formatting, case-insensitive str->bool — reference triad
+        # semantics); device-native casts later
+        from fugue_amd.dataframe.arrow_dataframe import ArrowDataFrame
+
+        altered = ArrowDataFrame(self.as_arrow()).alter_columns(columns)
         return HipDataFrame(
-            self.as_arrow().cast(schema.pa_schema), schema, device=self._device
+            altered.as_arrow(), altered.schema, device=self._device
         )
 
     def head(

@@ -35,6 +35,16 @@ def pandas_to_schema(df: pd.DataFrame) -> pa.Schema:
 def _cast_series(s: pd.Series, tp: pa.DataType) -> pd.Series:
     arr = pa.Array.from_pandas(s)
     if arr.type != tp:
+        if pa.types.is_string(tp) and (
+            pa.types.is_timestamp(arr.type) or pa.types.is_date(arr.type)
+        ):
+            # arrow's timestamp->string keeps trailing fractional zeros
+            # ("03:04:05.000000"); the contract is python str(datetime)
+            vals = [
+                None if v is None else str(v)
+                for v in arr.to_pylist()
+            ]
+            return pd.Series(vals, dtype=object)
         arr = arr.cast(tp, safe=False)
     return arr.to_pandas()
 

@@ -112,3 +112,21 @@ class TestHipGpuBuiltInConformance(BuiltInConformance):
         from fugue_amd.hip.execution_engine import HipExecutionEngine
 
         return HipExecutionEngine()
+
+
+from fugue_amd.testing.dataframe_conformance import DataFrameConformance
+
+
+class TestHipGpuDataFrameConformance(DataFrameConformance):
+    """All 24 reference DataFrameTests cases on device tensors."""
+
+    supports_nested = False  # flat device columns (documented deviation)
+    supports_map = False
+    native_is_fugue = True
+
+    def df(self, data: Any = None, schema: Any = None):
+        from fugue_amd.hip.frame import HipDataFrame
+
+        return HipDataFrame(
+            ArrayDataFrame(data, schema).as_arrow(), schema, device="cuda:0"
+        )
