@@ -679,6 +679,8 @@ def distinct_reps(h1: torch.Tensor, h2: torch.Tensor) -> torch.Tensor:
     """One representative row index per distinct (h1, h2) row-content key
     (raises HashCollisionError on an h1 collision — caller falls back)."""
     n = int(h1.numel())
+    if n == 0:
+        return torch.empty(0, dtype=torch.int64, device=h1.device)
     if _is_cpu(h1):
         import pandas as pd
 
