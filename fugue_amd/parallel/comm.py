@@ -64,9 +64,15 @@ class Communicator:
         dist.all_gather(gathered, local)
         return torch.stack([t.cpu() for t in gathered], dim=0)
 
-    def _comm_device(self, t: torch.Tensor) -> torch.device:
+    def _comm_device(self, t: Optional[torch.Tensor] = None) -> torch.device:
         if self.backend == "nccl":
-            return torch.device(self.device or "cuda")
+            if self.device is not None:
+                return torch.device(self.device)
+            # one process per GPU: LOCAL_RANK names this rank's device
+            local = int(os.environ.get("LOCAL_RANK", "0"))
+            return torch.device(
+                f"cuda:{local % max(1, torch.cuda.device_count())}"
+            )
         return torch.device("cpu")
 
     def all_to_all_v(
@@ -132,11 +138,24 @@ class Communicator:
         if self.rank == src:
             assert t is not None
             meta = [(list(t.shape), str(t.dtype).replace("torch.", ""))]
-        dist.broadcast_object_list(meta, src=src)
+        # NCCL transports object payloads through a device tensor: pin
+        # the staging device explicitly so ranks with a non-default
+        # current device can't deadlock on mismatched buffers
+        if self.backend == "nccl":
+            dist.broadcast_object_list(
+                meta, src=src, device=self._comm_device(None)
+            )
+        else:
+            dist.broadcast_object_list(meta, src=src)
         shape, dtype_name = meta[0]
+        default_dev = (
+            self._comm_device(None) if self.backend == "nccl" else "cpu"
+        )
         if self.rank != src:
             t = torch.empty(
-                shape, dtype=getattr(torch, dtype_name), device=device or "cpu"
+                shape,
+                dtype=getattr(torch, dtype_name),
+                device=device or default_dev,
             )
         dist.broadcast(t, src=src)
         return t
