@@ -238,16 +238,30 @@ class HipMapEngine(MapEngine):
             # 2. local segmented sort by keys (+presort), group boundaries
             key_cols = [hdf.col(k) for k in keys]
             try:
-                packed, _ = dops.pack_keys(key_cols)
-                sort_by = keys + [
-                    k for k in presort.keys() if k not in keys
-                ]
-                asc = [True] * len(keys) + [
-                    presort[k] for k in presort.keys() if k not in keys
-                ]
-                perm = dops.sort_indices(hdf, sort_by, asc)
+                ps_names = [k for k in presort.keys() if k not in keys]
+                ps_asc = [presort[k] for k in ps_names]
+                try:
+                    packed, _ = dops.pack_keys(key_cols)
+                    h2v: Optional[torch.Tensor] = None
+                    perm = dops.sort_indices(
+                        hdf, keys + ps_names, [True] * len(keys) + ps_asc
+                    )
+                except NotImplementedError:
+                    # string keys: group identity by 64-bit row hash
+                    # (verified with an independent second hash — an h1
+                    # collision falls back to the exact host path)
+                    packed = dops.hash_rows(key_cols)
+                    h2v = dops.hash_rows(key_cols, seed=dops._H2_SEED)
+                    perm = dops.sort_perm_keys_first(
+                        hdf, packed, ps_names, ps_asc
+                    )
                 sorted_df = hdf.gather_rows(perm)
                 sorted_keys = packed.index_select(0, perm)
+                if h2v is not None and sorted_keys.numel() > 1:
+                    s2 = h2v.index_select(0, perm)
+                    same1 = sorted_keys[1:] == sorted_keys[:-1]
+                    if bool((same1 & (s2[1:] != s2[:-1])).any().item()):
+                        raise NotImplementedError("string key hash collision")
                 bounds = dops.group_boundaries(sorted_keys).cpu().tolist()
                 n = sorted_df.count()
                 bounds.append(n)
@@ -316,18 +330,64 @@ class HipMapEngine(MapEngine):
                     cursor.set(lambda: input_df.peek_array(), gi, 0)
                     results.append(map_func(cursor, input_df).as_pandas())
             elif can_fast_stage(sorted_df):
-                # pinned, double-buffered D2H: batch k+1 copies on a side
-                # stream while the UDF processes batch k
-                for g0, g1, batch in staged_pandas_batches(sorted_df, bounds):
-                    base = bounds[g0]
-                    for gi in range(g0, g1):
-                        start, end = bounds[gi] - base, bounds[gi + 1] - base
-                        sub = batch.iloc[start:end].reset_index(drop=True)
-                        input_df = PandasDataFrame(
-                            sub, Schema(df.schema), pandas_df_wrapper=True
-                        )
-                        cursor.set(lambda: input_df.peek_array(), gi, 0)
-                        results.append(map_func(cursor, input_df).as_pandas())
+                # 3-stage pipeline: D2H of batch k+1 on a side stream,
+                # UDF on batch k on this thread, and H2D of batch k-1's
+                # results on an upload worker — all concurrent
+                from concurrent.futures import ThreadPoolExecutor
+
+                from fugue_amd.hip.frame import supported_device_type
+
+                out_on_device = all(
+                    supported_device_type(f.type)
+                    for f in output_schema.fields
+                )
+
+                def _upload(batch_results: List[pd.DataFrame]) -> DataFrame:
+                    out = pd.concat(batch_results, ignore_index=True)
+                    host = PandasDataFrame(out, output_schema)
+                    if not out_on_device:
+                        return host  # nested/decimal outputs stay on host
+                    return HipDataFrame(
+                        host.as_arrow(), output_schema, device=engine._device
+                    )
+
+                futures = []
+                with ThreadPoolExecutor(max_workers=1) as pool:
+                    for g0, g1, batch in staged_pandas_batches(
+                        sorted_df, bounds
+                    ):
+                        base = bounds[g0]
+                        batch_res: List[pd.DataFrame] = []
+                        for gi in range(g0, g1):
+                            start = bounds[gi] - base
+                            end = bounds[gi + 1] - base
+                            sub = batch.iloc[start:end].reset_index(drop=True)
+                            input_df = PandasDataFrame(
+                                sub, Schema(df.schema), pandas_df_wrapper=True
+                            )
+                            cursor.set(lambda: input_df.peek_array(), gi, 0)
+                            batch_res.append(
+                                map_func(cursor, input_df).as_pandas()
+                            )
+                        if batch_res:
+                            futures.append(pool.submit(_upload, batch_res))
+                parts = [f.result() for f in futures]
+                if len(parts) == 0:
+                    return engine.to_df(
+                        PandasDataFrame(
+                            output_schema.create_empty_pandas(), output_schema
+                        ),
+                        shard_replicated=False,
+                    )
+                if out_on_device:
+                    return parts[0].concat_with(parts[1:])
+                merged = pd.concat(
+                    [p.as_pandas() for p in parts], ignore_index=True
+                )
+                return engine.to_df(
+                    PandasDataFrame(merged, output_schema),
+                    shard_replicated=False,
+                )
             else:
                 pdf_all = sorted_df.as_pandas()
                 for gi in range(len(bounds) - 1):

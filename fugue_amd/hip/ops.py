@@ -824,6 +824,39 @@ def sort_indices(
     return perm
 
 
+def sort_perm_keys_first(
+    df: HipDataFrame,
+    key_tensor: torch.Tensor,
+    presort: List[str],
+    ascending: List[bool],
+) -> torch.Tensor:
+    """Stable sort permutation by (key_tensor, presort columns): the
+    int64 key tensor is the most significant key.  Used for string-keyed
+    grouping where the key identity is a 64-bit row hash (strings have
+    no device ordering; group boundaries only need identity)."""
+    n = df.count()
+    device = key_tensor.device
+    perm = torch.arange(n, dtype=torch.int64, device=device)
+    for name, asc in reversed(list(zip(presort, ascending))):
+        c = df.col(name)
+        if isinstance(c, StringDeviceColumn):
+            raise NotImplementedError("string presort keys not on device")
+        vals = c.data.index_select(0, perm)
+        if c.valid is not None:
+            if vals.is_floating_point():
+                sentinel = float("inf") if asc else float("-inf")
+            else:
+                info = torch.iinfo(vals.dtype)
+                sentinel = info.max if asc else info.min
+            v = c.valid.index_select(0, perm)
+            vals = torch.where(v, vals, torch.full_like(vals, sentinel))
+        idx = torch.argsort(vals, stable=True, descending=not asc)
+        perm = perm.index_select(0, idx)
+    kv = key_tensor.index_select(0, perm)
+    idx = torch.argsort(kv, stable=True)
+    return perm.index_select(0, idx)
+
+
 def group_boundaries(sorted_keys: torch.Tensor) -> torch.Tensor:
     """Start offsets of each group in a key-sorted int64 tensor."""
     n = sorted_keys.numel()

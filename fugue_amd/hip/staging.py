@@ -5,25 +5,94 @@ pandas/pyarrow UDFs via pinned ``hipMemcpyAsync`` device↔host staging on a
 side stream, overlapped with the next batch's transfer — a 2-deep
 double-buffered pipeline (copy batch k+1 while the UDF chews batch k).
 
-The fast path applies to numeric/bool/datetime columns (zero-copy
-numpy→pandas on the host side); frames containing strings or validity
-masks take the arrow conversion path (correct, not overlapped).
+The fast path covers numeric/bool/datetime columns, validity-masked
+numeric columns (the mask is staged alongside the data) and string
+columns (offset + byte buffers staged; host-side reassembly is a
+zero-copy arrow ``from_buffers``).  Nested/decimal columns take the
+arrow conversion path (correct, not overlapped).
 """
-from typing import Dict, Iterator, List, Optional, Tuple
+from typing import Any, Dict, Iterator, List, Optional, Tuple
 
 import numpy as np
 import pandas as pd
+import pyarrow as pa
 import torch
 
-from fugue_amd.hip.frame import HipDataFrame, StringDeviceColumn
+from fugue_amd.hip.frame import DeviceColumn, HipDataFrame, StringDeviceColumn
 from fugue_amd.schema import Schema
 
 
 def can_fast_stage(df: HipDataFrame) -> bool:
-    for c in df.columns_map.values():
-        if isinstance(c, StringDeviceColumn) or c.valid is not None:
-            return False
-    return True
+    # flat data (+ optional mask) and offsets+bytes (string/binary)
+    # layouts are stageable; nested/decimal columns take the arrow path
+    from fugue_amd.hip.frame import supported_device_type
+
+    return all(supported_device_type(f.type) for f in df.schema.fields)
+
+
+class _ColStager:
+    """Per-column staging: issues async pinned D2H copies for a row
+    range and reassembles the host pandas column."""
+
+    def __init__(self, c: DeviceColumn, pa_type: pa.DataType):
+        self.c = c
+        self.pa_type = pa_type
+        self.is_str = isinstance(c, StringDeviceColumn)
+        if self.is_str:
+            # offsets fetched once up front: byte ranges per batch are
+            # then known host-side without extra syncs
+            self.offsets_np = c.offsets.cpu().numpy()
+
+    def _pinned_like(self, src: torch.Tensor) -> torch.Tensor:
+        return torch.empty(src.shape, dtype=src.dtype, pin_memory=True)
+
+    def start(self, lo: int, hi: int, use_cuda: bool) -> Dict[str, Any]:
+        out: Dict[str, Any] = {}
+
+        def cp(src: torch.Tensor) -> torch.Tensor:
+            if not use_cuda:
+                return src
+            dst = self._pinned_like(src)
+            dst.copy_(src, non_blocking=True)
+            return dst
+
+        if self.is_str:
+            b0 = int(self.offsets_np[lo])
+            b1 = int(self.offsets_np[hi])
+            out["bytes"] = cp(self.c.bytes[b0:b1])
+            out["off"] = self.offsets_np[lo : hi + 1] - b0
+        else:
+            out["data"] = cp(self.c.data[lo:hi])
+        if self.c.valid is not None:
+            out["valid"] = cp(self.c.valid[lo:hi])
+        return out
+
+    def finish(self, staged: Dict[str, Any]) -> pd.Series:
+        valid_np = (
+            staged["valid"].numpy() if "valid" in staged else None
+        )
+        if self.is_str:
+            n = len(staged["off"]) - 1
+            is_bin = pa.types.is_binary(self.pa_type) or pa.types.is_large_binary(
+                self.pa_type
+            )
+            big = pa.large_binary() if is_bin else pa.large_string()
+            bufs = [
+                None if valid_np is None else pa.array(valid_np).buffers()[1],
+                pa.py_buffer(np.ascontiguousarray(staged["off"]).tobytes()),
+                pa.py_buffer(staged["bytes"].numpy().tobytes()),
+            ]
+            arr = pa.Array.from_buffers(big, n, bufs)
+            return arr.to_pandas()
+        data_np = staged["data"].numpy()
+        if pa.types.is_timestamp(self.pa_type):
+            data_np = data_np.astype("datetime64[us]")
+        elif pa.types.is_date(self.pa_type):
+            data_np = data_np.astype("datetime64[D]")
+        if valid_np is None:
+            return pd.Series(data_np)
+        arr = pa.array(data_np, mask=~valid_np)
+        return arr.to_pandas()
 
 
 def staged_pandas_batches(
@@ -51,13 +120,23 @@ def staged_pandas_batches(
             h += 1
         batches.append((g, h))
         g = h
-    names = list(df.columns_map.keys())
-    cols = [df.col(n) for n in names]
+    names = [f.name for f in df.schema.fields]
+    stagers = [
+        _ColStager(df.col(f.name), f.type) for f in df.schema.fields
+    ]
+
+    def assemble(staged_cols: List[Dict[str, Any]]) -> pd.DataFrame:
+        out = {
+            n: st.finish(sc)
+            for n, st, sc in zip(names, stagers, staged_cols)
+        }
+        return pd.DataFrame(out, columns=names)
+
     if not use_cuda:
         for g0, g1 in batches:
             lo, hi = bounds[g0], bounds[g1]
-            data = {n: c.data[lo:hi].numpy() for n, c in zip(names, cols)}
-            yield g0, g1, _to_pandas(data, df.schema)
+            staged = [st.start(lo, hi, False) for st in stagers]
+            yield g0, g1, assemble(staged)
         return
 
     side = torch.cuda.Stream()
@@ -65,43 +144,19 @@ def staged_pandas_batches(
     def start_copy(b: int):
         g0, g1 = batches[b]
         lo, hi = bounds[g0], bounds[g1]
-        host: Dict[str, torch.Tensor] = {}
         with torch.cuda.stream(side):
-            for n, c in zip(names, cols):
-                pinned = torch.empty(
-                    hi - lo, dtype=c.data.dtype, pin_memory=True
-                )
-                pinned.copy_(c.data[lo:hi], non_blocking=True)
-                host[n] = pinned
+            staged = [st.start(lo, hi, True) for st in stagers]
             ev = torch.cuda.Event()
             ev.record(side)
-        return host, ev
+        return staged, ev
 
     pending = start_copy(0)
     for b in range(len(batches)):
-        host, ev = pending
-        if b + 1 < len(batches):
-            nxt = start_copy(b + 1)
-        else:
-            nxt = None
+        staged, ev = pending
+        nxt = start_copy(b + 1) if b + 1 < len(batches) else None
         ev.synchronize()
-        data = {n: t.numpy() for n, t in host.items()}
         g0, g1 = batches[b]
-        yield g0, g1, _to_pandas(data, df.schema)
+        yield g0, g1, assemble(staged)
         if nxt is None:
             return
         pending = nxt
-
-
-def _to_pandas(data: Dict[str, np.ndarray], schema: Schema) -> pd.DataFrame:
-    import pyarrow as pa
-
-    out = {}
-    for f in schema.fields:
-        arr = data[f.name]
-        if pa.types.is_timestamp(f.type):
-            arr = arr.astype("datetime64[us]")
-        elif pa.types.is_date(f.type):
-            arr = arr.astype("datetime64[D]")
-        out[f.name] = arr
-    return pd.DataFrame(out, columns=schema.names)
