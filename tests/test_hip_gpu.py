@@ -435,3 +435,71 @@ def test_global_distinct_aggregates_gpu(engine):
     assert float(r["s"][0]) == dd.sum()
     assert abs(float(r["a"][0]) - dd.mean()) < 1e-9
     assert int(r["c"][0]) == dd.nunique()
+
+
+def test_string_keyed_map_10m_gpu(engine):
+    """10M-row string-keyed transform stays on the device grouping path
+    (hash-identity sort + boundaries; no host safe_groupby_apply —
+    VERDICT r01 item 5 'Done' criterion)."""
+    import pyarrow as pa
+
+    import fugue_amd.api as fa
+    from fugue_amd.hip import ops as dops
+    from fugue_amd.hip.frame import HipDataFrame, StringDeviceColumn, DeviceColumn
+    from fugue_amd.schema import Schema
+    from fugue_amd.utils import pandas_like
+
+    n = 10_000_000
+    n_groups = 1000
+    dev = torch.device(engine.device)
+    gen = torch.Generator(device=dev)
+    gen.manual_seed(11)
+    codes = torch.randint(0, n_groups, (n,), device=dev, generator=gen)
+    # group names 'g0000'..'g0999' (5 bytes each): build bytes directly
+    import numpy as np
+
+    names = np.array([f"g{i:04d}" for i in range(n_groups)])
+    name_bytes = torch.from_numpy(
+        np.frombuffer("".join(names).encode(), dtype=np.uint8).copy()
+    ).to(dev)
+    offsets = torch.zeros(n + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(torch.full((n,), 5, dtype=torch.int64, device=dev), 0,
+                 out=offsets[1:])
+    byte_idx = (codes * 5).unsqueeze(1) + torch.arange(5, device=dev)
+    flat = name_bytes.index_select(0, byte_idx.reshape(-1))
+    col = StringDeviceColumn(offsets, flat, None)
+    vals = torch.rand(n, dtype=torch.float64, device=dev, generator=gen)
+    df = HipDataFrame.from_columns(
+        {"k": col, "v": DeviceColumn(vals, None, pa.float64())},
+        Schema("k:str,v:double"),
+        engine.device,
+    )
+
+    calls = {"host_grouping": 0}
+    orig = pandas_like.safe_groupby_apply
+
+    def spy(*a, **k):
+        calls["host_grouping"] += 1
+        return orig(*a, **k)
+
+    pandas_like.safe_groupby_apply = spy
+    try:
+        # schema: k:str,s:double,n:long
+        def agg(pdf):
+            import pandas as pd
+
+            return pd.DataFrame(
+                dict(k=[pdf["k"].iloc[0]], s=[pdf["v"].sum()],
+                     n=[len(pdf)])
+            )
+
+        res = fa.transform(
+            df, agg, partition=dict(by=["k"]), engine=engine, as_fugue=True
+        )
+        out = res.as_pandas().sort_values("k").reset_index(drop=True)
+    finally:
+        pandas_like.safe_groupby_apply = orig
+    assert calls["host_grouping"] == 0, "fell back to host grouping"
+    assert len(out) == n_groups
+    assert abs(out["s"].sum() - float(vals.sum().item())) < 1e-3
+    assert out["n"].sum() == n
