@@ -24,6 +24,8 @@ def main() -> None:
     ap.add_argument("--rows", type=int, default=100_000_000)
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--parts", type=int, default=32,
+                    help="logical partitions per rank (pipeline depth)")
     ap.add_argument("--naive", action="store_true")
     args = ap.parse_args()
 
@@ -66,7 +68,8 @@ def main() -> None:
 
     def step():
         return fa.transform(
-            df, identity, schema="*", engine=engine, as_fugue=True
+            df, identity, schema="*", engine=engine, as_fugue=True,
+            partition=dict(num=args.parts),
         )
 
     for _ in range(args.warmup):
@@ -96,6 +99,7 @@ def main() -> None:
                 data="synthetic",
                 config=dict(
                     model="pandas-identity-transform",
+                    parts=args.parts,
                     rows=n,
                     out_rows=out.count() if out is not None else 0,
                 ),

@@ -210,18 +210,25 @@ class HipMapEngine(MapEngine):
                         shard_replicated=False,
                     )
                 return dev_results[0].concat_with(dev_results[1:])
+            from fugue_amd.hip.staging import can_fast_stage
+
+            n_rows = local.count()
+            bounds_sp = [
+                (n_rows * i) // n_local_parts
+                for i in range(n_local_parts + 1)
+            ]
+            if n_rows > 0 and can_fast_stage(local):
+                # same 3-stage pipeline as the keyed path: D2H of the
+                # next slice overlaps the UDF, result H2D overlaps both
+                return _run_staged_udf(
+                    engine, local, bounds_sp, Schema(df.schema),
+                    output_schema, map_func, cursor,
+                    lambda gi: engine.rank * n_local_parts + gi,
+                )
             pdf_local = local.as_pandas()
             if len(pdf_local) > 0:
-                import numpy as _np
-
-                bounds_sp = [
-                    (len(pdf_local) * i) // n_local_parts
-                    for i in range(n_local_parts + 1)
-                ]
-                for p, subdf in enumerate(
-                    pdf_local.iloc[bounds_sp[i] : bounds_sp[i + 1]]
-                    for i in range(n_local_parts)
-                ):
+                for p in range(n_local_parts):
+                    subdf = pdf_local.iloc[bounds_sp[p] : bounds_sp[p + 1]]
                     if len(subdf) == 0:
                         continue
                     sub = subdf.reset_index(drop=True)
@@ -330,63 +337,9 @@ class HipMapEngine(MapEngine):
                     cursor.set(lambda: input_df.peek_array(), gi, 0)
                     results.append(map_func(cursor, input_df).as_pandas())
             elif can_fast_stage(sorted_df):
-                # 3-stage pipeline: D2H of batch k+1 on a side stream,
-                # UDF on batch k on this thread, and H2D of batch k-1's
-                # results on an upload worker — all concurrent
-                from concurrent.futures import ThreadPoolExecutor
-
-                from fugue_amd.hip.frame import supported_device_type
-
-                out_on_device = all(
-                    supported_device_type(f.type)
-                    for f in output_schema.fields
-                )
-
-                def _upload(batch_results: List[pd.DataFrame]) -> DataFrame:
-                    out = pd.concat(batch_results, ignore_index=True)
-                    host = PandasDataFrame(out, output_schema)
-                    if not out_on_device:
-                        return host  # nested/decimal outputs stay on host
-                    return HipDataFrame(
-                        host.as_arrow(), output_schema, device=engine._device
-                    )
-
-                futures = []
-                with ThreadPoolExecutor(max_workers=1) as pool:
-                    for g0, g1, batch in staged_pandas_batches(
-                        sorted_df, bounds
-                    ):
-                        base = bounds[g0]
-                        batch_res: List[pd.DataFrame] = []
-                        for gi in range(g0, g1):
-                            start = bounds[gi] - base
-                            end = bounds[gi + 1] - base
-                            sub = batch.iloc[start:end].reset_index(drop=True)
-                            input_df = PandasDataFrame(
-                                sub, Schema(df.schema), pandas_df_wrapper=True
-                            )
-                            cursor.set(lambda: input_df.peek_array(), gi, 0)
-                            batch_res.append(
-                                map_func(cursor, input_df).as_pandas()
-                            )
-                        if batch_res:
-                            futures.append(pool.submit(_upload, batch_res))
-                parts = [f.result() for f in futures]
-                if len(parts) == 0:
-                    return engine.to_df(
-                        PandasDataFrame(
-                            output_schema.create_empty_pandas(), output_schema
-                        ),
-                        shard_replicated=False,
-                    )
-                if out_on_device:
-                    return parts[0].concat_with(parts[1:])
-                merged = pd.concat(
-                    [p.as_pandas() for p in parts], ignore_index=True
-                )
-                return engine.to_df(
-                    PandasDataFrame(merged, output_schema),
-                    shard_replicated=False,
+                return _run_staged_udf(
+                    engine, sorted_df, bounds, Schema(df.schema),
+                    output_schema, map_func, cursor, lambda gi: gi,
                 )
             else:
                 pdf_all = sorted_df.as_pandas()
@@ -405,6 +358,71 @@ class HipMapEngine(MapEngine):
         return engine.to_df(
             PandasDataFrame(out_pdf, output_schema), shard_replicated=False
         )
+
+
+def _run_staged_udf(
+    engine: "HipExecutionEngine",
+    frame: HipDataFrame,
+    bounds: List[int],
+    in_schema: Schema,
+    output_schema: Schema,
+    map_func: Any,
+    cursor: Any,
+    part_no_of: Any,
+) -> DataFrame:
+    """3-stage UDF pipeline shared by the keyed and keyless map paths:
+    D2H of slice k+1 on a side stream, the user UDF on slice k on this
+    thread, and H2D of slice k-1's results on an upload worker."""
+    from concurrent.futures import ThreadPoolExecutor
+
+    from fugue_amd.hip.frame import supported_device_type
+    from fugue_amd.hip.staging import staged_pandas_batches
+
+    out_on_device = all(
+        supported_device_type(f.type) for f in output_schema.fields
+    )
+
+    def _upload(batch_results: List[pd.DataFrame]) -> DataFrame:
+        out = pd.concat(batch_results, ignore_index=True)
+        host = PandasDataFrame(out, output_schema)
+        if not out_on_device:
+            return host  # nested/decimal outputs stay on host
+        return HipDataFrame(
+            host.as_arrow(), output_schema, device=engine._device
+        )
+
+    futures = []
+    with ThreadPoolExecutor(max_workers=1) as pool:
+        for g0, g1, batch in staged_pandas_batches(frame, bounds):
+            base = bounds[g0]
+            batch_res: List[pd.DataFrame] = []
+            for gi in range(g0, g1):
+                start = bounds[gi] - base
+                end = bounds[gi + 1] - base
+                if end <= start:
+                    continue
+                sub = batch.iloc[start:end].reset_index(drop=True)
+                input_df = PandasDataFrame(
+                    sub, in_schema, pandas_df_wrapper=True
+                )
+                cursor.set(lambda: input_df.peek_array(), part_no_of(gi), 0)
+                batch_res.append(map_func(cursor, input_df).as_pandas())
+            if batch_res:
+                futures.append(pool.submit(_upload, batch_res))
+    parts = [f.result() for f in futures]
+    if len(parts) == 0:
+        return engine.to_df(
+            PandasDataFrame(
+                output_schema.create_empty_pandas(), output_schema
+            ),
+            shard_replicated=False,
+        )
+    if out_on_device:
+        return parts[0].concat_with(parts[1:])
+    merged = pd.concat([p.as_pandas() for p in parts], ignore_index=True)
+    return engine.to_df(
+        PandasDataFrame(merged, output_schema), shard_replicated=False
+    )
 
 
 class HipExecutionEngine(ExecutionEngine):

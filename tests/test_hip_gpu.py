@@ -485,9 +485,7 @@ def test_string_keyed_map_10m_gpu(engine):
     pandas_like.safe_groupby_apply = spy
     try:
         # schema: k:str,s:double,n:long
-        def agg(pdf):
-            import pandas as pd
-
+        def agg(pdf: pd.DataFrame) -> pd.DataFrame:
             return pd.DataFrame(
                 dict(k=[pdf["k"].iloc[0]], s=[pdf["v"].sum()],
                      n=[len(pdf)])
