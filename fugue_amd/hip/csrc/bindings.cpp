@@ -46,6 +46,8 @@ void launch_gb_aggregate_part_big(const void*, const double*,
                                   const int32_t*, int64_t, int64_t*,
                                   double*, int64_t*, int64_t, int64_t,
                                   int, int, int, hipStream_t);
+void launch_gather_cols(const uint64_t*, const uint64_t*, int, int,
+                        const int64_t*, int64_t, hipStream_t);
 void launch_join_count(const int64_t*, int64_t, const int64_t*,
                        const int64_t*, const int64_t*, const int32_t*,
                        const int32_t*, int64_t, int32_t*, hipStream_t);
@@ -436,6 +438,37 @@ std::vector<at::Tensor> compact_columns(at::Tensor mask,
   return outs;
 }
 
+std::vector<at::Tensor> gather_columns(at::Tensor idx,
+                                       std::vector<at::Tensor> cols) {
+  check_gpu(idx, "idx");
+  TORCH_CHECK(cols.size() >= 1 && cols.size() <= 16, "1..16 columns");
+  int64_t n = idx.numel();
+  // group by element width: one fused launch per width class
+  std::vector<at::Tensor> outs(cols.size());
+  for (int width : {8, 4, 2, 1}) {
+    uint64_t srcs[16];
+    uint64_t dsts[16];
+    int k = 0;
+    for (size_t c = 0; c < cols.size(); ++c) {
+      if (cols[c].element_size() != width) continue;
+      check_gpu(cols[c], "col");
+      TORCH_CHECK(cols[c].is_contiguous(), "columns must be contiguous");
+      auto out = at::empty({n}, cols[c].options());
+      srcs[k] = reinterpret_cast<uint64_t>(cols[c].data_ptr());
+      dsts[k] = reinterpret_cast<uint64_t>(out.data_ptr());
+      outs[c] = out;
+      ++k;
+    }
+    if (k > 0 && n > 0) {
+      launch_gather_cols(srcs, dsts, k, width, idx.data_ptr<int64_t>(), n,
+                         current_stream());
+    } else if (k > 0) {
+      // n == 0: outputs already sized 0
+    }
+  }
+  return outs;
+}
+
 std::vector<at::Tensor> reduce_columns(at::Tensor vals,
                                        c10::optional<at::Tensor> valids,
                                        at::Tensor ops) {
@@ -479,6 +512,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hash_seed", &hash_seed, "seed a row-hash buffer");
   m.def("gb_mark_reps", &gb_mark_reps,
         "representative rows + h2 verification for hashed group-by");
+  m.def("gather_columns", &gather_columns,
+        "fused multi-column row gather");
   m.def("compact_columns", &compact_columns,
         "fused masked compaction of 8-byte columns");
   m.def("join_count", &join_count, "count matches per probe row");

@@ -1148,6 +1148,59 @@ __global__ __launch_bounds__(BLOCK) void gb_aggregate_part_big_kernel_v1(
   }
 }
 
+// ---------------------------------------------------------------------
+// fused multi-column row gather: one pass reads idx once per row and
+// materializes up to GATHER_MAX_COLS columns of one element width
+// (replaces per-column at::index_select launches on the join/sort
+// output-materialization path)
+#define GATHER_MAX_COLS 16
+
+struct GatherPtrs {
+  uint64_t src[GATHER_MAX_COLS];
+  uint64_t dst[GATHER_MAX_COLS];
+};
+
+template <typename T>
+__global__ __launch_bounds__(BLOCK) void gather_cols_kernel(
+    GatherPtrs ptrs, int ncols, const int64_t* __restrict__ idx,
+    int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    int64_t j = idx[i];
+#pragma unroll 4
+    for (int c = 0; c < ncols; ++c) {
+      reinterpret_cast<T*>(ptrs.dst[c])[i] =
+          reinterpret_cast<const T*>(ptrs.src[c])[j];
+    }
+  }
+}
+
+extern "C" {
+void launch_gather_cols(const uint64_t* src_ptrs, const uint64_t* dst_ptrs,
+                        int ncols, int elem_size, const int64_t* idx,
+                        int64_t n, hipStream_t stream) {
+  GatherPtrs ptrs;
+  for (int c = 0; c < ncols; ++c) {
+    ptrs.src[c] = src_ptrs[c];
+    ptrs.dst[c] = dst_ptrs[c];
+  }
+  dim3 g(grid_for(n, 2)), b(BLOCK);
+  if (elem_size == 8)
+    hipLaunchKernelGGL(gather_cols_kernel<uint64_t>, g, b, 0, stream, ptrs,
+                       ncols, idx, n);
+  else if (elem_size == 4)
+    hipLaunchKernelGGL(gather_cols_kernel<uint32_t>, g, b, 0, stream, ptrs,
+                       ncols, idx, n);
+  else if (elem_size == 2)
+    hipLaunchKernelGGL(gather_cols_kernel<uint16_t>, g, b, 0, stream, ptrs,
+                       ncols, idx, n);
+  else
+    hipLaunchKernelGGL(gather_cols_kernel<uint8_t>, g, b, 0, stream, ptrs,
+                       ncols, idx, n);
+}
+}  // extern "C"
+
 // P/E-variant dispatch for the v4 staged scatter: 512/8 (baseline),
 // 1024/4 and 2048/2 trade staging depth for finer partitions (smaller
 // phase-3 LDS tables -> higher phase-3 occupancy)

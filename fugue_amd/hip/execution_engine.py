@@ -1347,12 +1347,37 @@ class HipExecutionEngine(ExecutionEngine):
             and len(_presort) > 0
             and na_position == "last"
         ):
-            # device top-n: sort permutation on device, gather only n rows
+            # device top-n: radix-select (torch.topk → rocPRIM select)
+            # for one sort key — O(n) instead of a full sort; multi-key
+            # presorts take the argsort path
             try:
-                perm = dops.sort_indices(
-                    d, list(_presort.keys()), list(_presort.values())
-                )
-                local_top = d.gather_rows(perm[: min(n, d.count())])
+                k = min(n, d.count())
+                keys_l = list(_presort.keys())
+                if len(keys_l) == 1 and k > 0:
+                    c = d.col(keys_l[0])
+                    if isinstance(c, StringDeviceColumn):
+                        raise NotImplementedError("string presort")
+                    asc = _presort[keys_l[0]]
+                    vals = c.data
+                    if c.valid is not None:
+                        if vals.is_floating_point():
+                            sentinel: Any = (
+                                float("inf") if asc else float("-inf")
+                            )
+                        else:
+                            info = torch.iinfo(vals.dtype)
+                            sentinel = info.max if asc else info.min
+                        vals = torch.where(
+                            c.valid, vals, torch.full_like(vals, sentinel)
+                        )
+                    _, perm = torch.topk(
+                        vals, k, largest=not asc, sorted=True
+                    )
+                else:
+                    perm = dops.sort_indices(
+                        d, keys_l, list(_presort.values())
+                    )
+                local_top = d.gather_rows(perm[:k])
                 if not self.is_distributed:
                     return local_top
                 gathered = self._gather_all(local_top)

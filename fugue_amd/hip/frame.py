@@ -531,8 +531,48 @@ class HipDataFrame(LocalBoundedDataFrame):
 
     # --- device ops ------------------------------------------------------ #
     def gather_rows(self, idx: torch.Tensor) -> "HipDataFrame":
-        cols = {n: c.gather(idx) for n, c in self._cols.items()}
+        cols = self._gather_cols_fused(idx)
         return HipDataFrame.from_columns(cols, self.schema, self._device)
+
+    def _gather_cols_fused(self, idx: torch.Tensor) -> Dict[str, "DeviceColumn"]:
+        """Row gather across all columns: flat columns (data + validity
+        tensors) go through one fused HIP kernel per element width
+        (``gather_cols_kernel``); strings keep their offsets-aware
+        path."""
+        flat_items = [
+            (n, c)
+            for n, c in self._cols.items()
+            if not isinstance(c, StringDeviceColumn)
+        ]
+        if idx.device.type != "cuda" or len(flat_items) <= 1:
+            return {n: c.gather(idx) for n, c in self._cols.items()}
+        from fugue_amd.hip.ext import get_ext
+
+        ext = get_ext()
+        tensors: List[torch.Tensor] = []
+        slots: List[Tuple[str, str]] = []  # (col name, "data"|"valid")
+        for n, c in flat_items:
+            tensors.append(c.data.contiguous())
+            slots.append((n, "data"))
+            if c.valid is not None:
+                tensors.append(c.valid.contiguous())
+                slots.append((n, "valid"))
+        gathered: Dict[Tuple[str, str], torch.Tensor] = {}
+        for i in range(0, len(tensors), 16):
+            outs = ext.gather_columns(idx, tensors[i : i + 16])
+            for slot, out in zip(slots[i : i + 16], outs):
+                gathered[slot] = out
+        cols: Dict[str, DeviceColumn] = {}
+        for n, c in self._cols.items():
+            if isinstance(c, StringDeviceColumn):
+                cols[n] = c.gather(idx)
+            else:
+                cols[n] = DeviceColumn(
+                    gathered[(n, "data")],
+                    gathered.get((n, "valid")),
+                    c.pa_type,
+                )
+        return cols
 
     def slice_rows(self, start: int, length: int) -> "HipDataFrame":
         cols = {n: c.slice(start, length) for n, c in self._cols.items()}
