@@ -7,6 +7,7 @@
 
 #include <cstdlib>
 #include <cstdio>
+#include <cstring>
 #include <limits>
 #include <vector>
 
@@ -48,6 +49,7 @@ void launch_gb_aggregate_part_big(const void*, const double*,
                                   int, int, int, hipStream_t);
 void launch_gather_cols(const uint64_t*, const uint64_t*, int, int,
                         const int64_t*, int64_t, hipStream_t);
+void launch_expr_filter(const void*, int64_t, bool*, hipStream_t);
 void launch_join_count(const int64_t*, int64_t, const int64_t*,
                        const int64_t*, const int64_t*, const int32_t*,
                        const int32_t*, int64_t, int32_t*, hipStream_t);
@@ -438,6 +440,63 @@ std::vector<at::Tensor> compact_columns(at::Tensor mask,
   return outs;
 }
 
+// host mirror of ExprProg in relational.hip (layout must match)
+struct ExprProgHost {
+  int n_ops;
+  unsigned char op[48];
+  signed char aux[48];
+  unsigned long long imm[12];
+  unsigned long long col_data[12];
+  unsigned long long col_valid[12];
+  unsigned char col_dt[12];
+};
+
+at::Tensor expr_filter(at::Tensor ops, at::Tensor aux, at::Tensor imm,
+                       std::vector<at::Tensor> col_data,
+                       std::vector<at::Tensor> col_valid,
+                       at::Tensor col_dt, int64_t n) {
+  TORCH_CHECK(ops.numel() <= 48, "program too long");
+  TORCH_CHECK(col_data.size() <= 12, "too many columns");
+  TORCH_CHECK(imm.numel() <= 12, "too many immediates");
+  ExprProgHost prog;
+  std::memset(&prog, 0, sizeof(prog));
+  prog.n_ops = (int)ops.numel();
+  auto ops_c = ops.to(at::kByte).cpu();
+  auto aux_c = aux.to(at::kChar).cpu();
+  auto imm_c = imm.cpu();  // int64 bit patterns
+  auto dt_c = col_dt.to(at::kByte).cpu();
+  for (int i = 0; i < prog.n_ops; ++i) {
+    prog.op[i] = ops_c[i].item<uint8_t>();
+    prog.aux[i] = aux_c[i].item<int8_t>();
+  }
+  for (int i = 0; i < (int)imm_c.numel(); ++i) {
+    prog.imm[i] = (unsigned long long)imm_c[i].item<int64_t>();
+  }
+  at::Tensor out;
+  for (size_t c = 0; c < col_data.size(); ++c) {
+    check_gpu(col_data[c], "col");
+    TORCH_CHECK(col_data[c].is_contiguous(), "columns must be contiguous");
+    prog.col_data[c] = reinterpret_cast<unsigned long long>(
+        col_data[c].data_ptr());
+    prog.col_dt[c] = dt_c[c].item<uint8_t>();
+    if (col_valid[c].defined() && col_valid[c].numel() > 0) {
+      check_gpu(col_valid[c], "valid");
+      prog.col_valid[c] = reinterpret_cast<unsigned long long>(
+          col_valid[c].data_ptr());
+    }
+    if (!out.defined()) {
+      out = at::empty({n}, col_data[c].options().dtype(at::kBool));
+    }
+  }
+  if (!out.defined()) {
+    TORCH_CHECK(false, "expr_filter needs at least one column");
+  }
+  if (n > 0) {
+    launch_expr_filter(&prog, n, out.data_ptr<bool>(), current_stream());
+  }
+  return out;
+}
+
 std::vector<at::Tensor> gather_columns(at::Tensor idx,
                                        std::vector<at::Tensor> cols) {
   check_gpu(idx, "idx");
@@ -512,6 +571,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hash_seed", &hash_seed, "seed a row-hash buffer");
   m.def("gb_mark_reps", &gb_mark_reps,
         "representative rows + h2 verification for hashed group-by");
+  m.def("expr_filter", &expr_filter,
+        "fused filter-predicate interpreter");
   m.def("gather_columns", &gather_columns,
         "fused multi-column row gather");
   m.def("compact_columns", &compact_columns,

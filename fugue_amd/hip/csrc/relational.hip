@@ -1201,6 +1201,207 @@ void launch_gather_cols(const uint64_t* src_ptrs, const uint64_t* dst_ptrs,
 }
 }  // extern "C"
 
+// ---------------------------------------------------------------------
+// fused filter-predicate interpreter: evaluates a postfix expression
+// program over typed columns in ONE pass, replacing the chain of
+// elementwise compare/and/arith kernel launches the torch evaluator
+// emits.  Values ride a tagged 8-byte stack (exact int64 compares, no
+// double round-trip); null semantics match the python evaluator
+// (arith propagates invalid, comparisons on null -> false, AND/OR
+// treat null as false, IS NULL inspects the validity bit).
+#define EXPR_MAX_OPS 48
+#define EXPR_MAX_COLS 12
+#define EXPR_STACK 12
+
+// opcodes
+enum {
+  XOP_COL = 0,   // aux = column index
+  XOP_LIT_D = 1, // aux = imm index (double)
+  XOP_LIT_I = 2, // aux = imm index (int64 bits)
+  XOP_ADD = 3, XOP_SUB = 4, XOP_MUL = 5, XOP_DIV = 6,
+  XOP_LT = 7, XOP_LE = 8, XOP_GT = 9, XOP_GE = 10, XOP_EQ = 11,
+  XOP_NE = 12,
+  XOP_AND = 13, XOP_OR = 14, XOP_NOT = 15,
+  XOP_ISNULL = 16, XOP_NOTNULL = 17,
+  XOP_NEG = 18,
+};
+
+// column dtype tags
+enum { XDT_F64 = 0, XDT_F32 = 1, XDT_I64 = 2, XDT_I32 = 3, XDT_I16 = 4,
+       XDT_I8 = 5, XDT_BOOL = 6 };
+
+struct ExprProg {
+  int n_ops;
+  unsigned char op[EXPR_MAX_OPS];
+  signed char aux[EXPR_MAX_OPS];
+  unsigned long long imm[EXPR_MAX_COLS];
+  unsigned long long col_data[EXPR_MAX_COLS];
+  unsigned long long col_valid[EXPR_MAX_COLS];  // 0 = none
+  unsigned char col_dt[EXPR_MAX_COLS];
+};
+
+struct XVal {
+  long long bits;   // double bits, int64, or bool(0/1)
+  unsigned char tag;  // 0=double 1=int64 2=bool
+  bool valid;
+};
+
+__device__ __forceinline__ double xv_as_f64(const XVal& v) {
+  if (v.tag == 0) return __longlong_as_double(v.bits);
+  return (double)v.bits;
+}
+
+__global__ __launch_bounds__(BLOCK) void expr_filter_kernel(
+    ExprProg prog, int64_t n, bool* __restrict__ out) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    XVal st[EXPR_STACK];
+    int sp = 0;
+    for (int p = 0; p < prog.n_ops; ++p) {
+      int op = prog.op[p];
+      int aux = prog.aux[p];
+      if (op == XOP_COL) {
+        const void* dp = reinterpret_cast<const void*>(prog.col_data[aux]);
+        XVal v;
+        v.valid = true;
+        switch (prog.col_dt[aux]) {
+          case XDT_F64:
+            v.bits = reinterpret_cast<const long long*>(dp)[i];
+            v.tag = 0;
+            break;
+          case XDT_F32: {
+            double d = (double)reinterpret_cast<const float*>(dp)[i];
+            v.bits = __double_as_longlong(d);
+            v.tag = 0;
+            break;
+          }
+          case XDT_I64:
+            v.bits = reinterpret_cast<const long long*>(dp)[i];
+            v.tag = 1;
+            break;
+          case XDT_I32:
+            v.bits = (long long)reinterpret_cast<const int*>(dp)[i];
+            v.tag = 1;
+            break;
+          case XDT_I16:
+            v.bits = (long long)reinterpret_cast<const short*>(dp)[i];
+            v.tag = 1;
+            break;
+          case XDT_I8:
+            v.bits = (long long)reinterpret_cast<const signed char*>(dp)[i];
+            v.tag = 1;
+            break;
+          default:  // bool
+            v.bits = reinterpret_cast<const bool*>(dp)[i] ? 1 : 0;
+            v.tag = 2;
+        }
+        if (prog.col_valid[aux] != 0ULL) {
+          v.valid = reinterpret_cast<const bool*>(prog.col_valid[aux])[i];
+        }
+        st[sp++] = v;
+      } else if (op == XOP_LIT_D) {
+        XVal v;
+        v.bits = (long long)prog.imm[aux];
+        v.tag = 0;
+        v.valid = true;
+        st[sp++] = v;
+      } else if (op == XOP_LIT_I) {
+        XVal v;
+        v.bits = (long long)prog.imm[aux];
+        v.tag = 1;
+        v.valid = true;
+        st[sp++] = v;
+      } else if (op == XOP_NOT) {
+        XVal& a = st[sp - 1];
+        bool b = a.valid && (a.tag == 0 ? xv_as_f64(a) != 0.0 : a.bits != 0);
+        a.bits = b ? 0 : 1;
+        a.tag = 2;
+        a.valid = true;
+      } else if (op == XOP_ISNULL || op == XOP_NOTNULL) {
+        XVal& a = st[sp - 1];
+        bool b = (op == XOP_ISNULL) ? !a.valid : a.valid;
+        a.bits = b ? 1 : 0;
+        a.tag = 2;
+        a.valid = true;
+      } else if (op == XOP_NEG) {
+        XVal& a = st[sp - 1];
+        if (a.tag == 0)
+          a.bits = __double_as_longlong(-xv_as_f64(a));
+        else
+          a.bits = -a.bits;
+      } else if (op == XOP_AND || op == XOP_OR) {
+        XVal b = st[--sp];
+        XVal& a = st[sp - 1];
+        bool ab = a.valid && (a.tag == 0 ? xv_as_f64(a) != 0.0 : a.bits != 0);
+        bool bb = b.valid && (b.tag == 0 ? xv_as_f64(b) != 0.0 : b.bits != 0);
+        bool r = (op == XOP_AND) ? (ab && bb) : (ab || bb);
+        a.bits = r ? 1 : 0;
+        a.tag = 2;
+        a.valid = true;
+      } else {
+        XVal b = st[--sp];
+        XVal& a = st[sp - 1];
+        bool both_int = (a.tag != 0 && b.tag != 0);
+        bool valid = a.valid && b.valid;
+        if (op >= XOP_ADD && op <= XOP_DIV) {
+          if (op == XOP_DIV || !both_int) {
+            double x = xv_as_f64(a), y = xv_as_f64(b);
+            double r = op == XOP_ADD   ? x + y
+                       : op == XOP_SUB ? x - y
+                       : op == XOP_MUL ? x * y
+                                       : x / y;
+            a.bits = __double_as_longlong(r);
+            a.tag = 0;
+          } else {
+            long long x = a.bits, y = b.bits;
+            long long r = op == XOP_ADD   ? x + y
+                          : op == XOP_SUB ? x - y
+                                          : x * y;
+            a.bits = r;
+            a.tag = 1;
+          }
+          a.valid = valid;
+        } else {  // comparisons: null -> false, result always valid
+          bool r;
+          if (both_int) {
+            long long x = a.bits, y = b.bits;
+            r = op == XOP_LT   ? x < y
+                : op == XOP_LE ? x <= y
+                : op == XOP_GT ? x > y
+                : op == XOP_GE ? x >= y
+                : op == XOP_EQ ? x == y
+                               : x != y;
+          } else {
+            double x = xv_as_f64(a), y = xv_as_f64(b);
+            r = op == XOP_LT   ? x < y
+                : op == XOP_LE ? x <= y
+                : op == XOP_GT ? x > y
+                : op == XOP_GE ? x >= y
+                : op == XOP_EQ ? x == y
+                               : x != y;
+          }
+          a.bits = (r && valid) ? 1 : 0;
+          a.tag = 2;
+          a.valid = true;
+        }
+      }
+    }
+    const XVal& top = st[0];
+    out[i] = top.valid && (top.tag == 0 ? xv_as_f64(top) != 0.0
+                                        : top.bits != 0);
+  }
+}
+
+extern "C" {
+void launch_expr_filter(const void* prog, int64_t n, bool* out,
+                        hipStream_t stream) {
+  const ExprProg* pg = reinterpret_cast<const ExprProg*>(prog);
+  hipLaunchKernelGGL(expr_filter_kernel, dim3(grid_for(n, 2)), dim3(BLOCK),
+                     0, stream, *pg, n, out);
+}
+}  // extern "C"
+
 // P/E-variant dispatch for the v4 staged scatter: 512/8 (baseline),
 // 1024/4 and 2048/2 trade staging depth for finer partitions (smaller
 // phase-3 LDS tables -> higher phase-3 occupancy)

@@ -2,7 +2,7 @@
 each lowers to a HIP kernel; fusion of the scalar pipeline comes from
 torch's elementwise fuser, while the relational ops use the hand-written
 kernels in ``csrc/relational.hip``)."""
-from typing import Any, List, Optional, Tuple
+from typing import Any, Dict, List, Optional, Tuple
 
 import pyarrow as pa
 import torch
@@ -435,6 +435,141 @@ def _cast(data: torch.Tensor, tp: pa.DataType) -> torch.Tensor:
     raise DeviceExprError(f"cast to {tp} not supported on device")
 
 
+# ------------------------------------------------------------------ #
+# fused filter programs: compile a predicate tree to a postfix program
+# executed by the one-pass device interpreter (``expr_filter_kernel``)
+# instead of a chain of elementwise torch launches
+# ------------------------------------------------------------------ #
+_XOP = dict(
+    COL=0, LIT_D=1, LIT_I=2, ADD=3, SUB=4, MUL=5, DIV=6,
+    LT=7, LE=8, GT=9, GE=10, EQ=11, NE=12, AND=13, OR=14, NOT=15,
+    ISNULL=16, NOTNULL=17, NEG=18,
+)
+_XDT = {
+    torch.float64: 0, torch.float32: 1, torch.int64: 2, torch.int32: 3,
+    torch.int16: 4, torch.int8: 5, torch.bool: 6,
+}
+_X_BINOPS = {
+    "+": "ADD", "-": "SUB", "*": "MUL", "/": "DIV",
+    "<": "LT", "<=": "LE", ">": "GT", ">=": "GE", "==": "EQ", "!=": "NE",
+    "&": "AND", "|": "OR",
+}
+
+
+class _ProgBuilder:
+    def __init__(self, df: HipDataFrame):
+        self.df = df
+        self.ops: List[int] = []
+        self.aux: List[int] = []
+        self.imm: List[int] = []
+        self.cols: List[Any] = []
+        self.col_idx: Dict[str, int] = {}
+        self.depth = 0
+        self.max_depth = 0
+
+    def _push(self, op: str, aux: int = 0, dstack: int = 1) -> None:
+        self.ops.append(_XOP[op])
+        self.aux.append(aux)
+        self.depth += dstack
+        self.max_depth = max(self.max_depth, self.depth)
+
+    def emit(self, e: ColumnExpr) -> None:
+        if isinstance(e, _LiteralColumnExpr):
+            v = e.value
+            if v is None or isinstance(v, str):
+                raise DeviceExprError("literal not fusable")
+            if isinstance(v, bool) or isinstance(v, int):
+                bits = int(v)
+                self.imm.append(bits)
+                self._push("LIT_I", len(self.imm) - 1)
+            elif isinstance(v, float):
+                import struct as _struct
+
+                bits = _struct.unpack("<q", _struct.pack("<d", v))[0]
+                self.imm.append(bits)
+                self._push("LIT_D", len(self.imm) - 1)
+            else:
+                raise DeviceExprError("literal not fusable")
+            return
+        if isinstance(e, _NamedColumnExpr):
+            if e.as_type is not None:
+                raise DeviceExprError("cast not fusable")
+            c = self.df.col(e.name)
+            if isinstance(c, StringDeviceColumn) or c.data.dtype not in _XDT:
+                raise DeviceExprError("column not fusable")
+            if e.name not in self.col_idx:
+                self.col_idx[e.name] = len(self.cols)
+                self.cols.append(c)
+            self._push("COL", self.col_idx[e.name])
+            return
+        if isinstance(e, _NotOpExpr):
+            self.emit(e.col)
+            self._push("NOT", dstack=0)
+            return
+        if isinstance(e, _UnaryOpExpr):
+            self.emit(e.col)
+            if e.op == "IS_NULL":
+                self._push("ISNULL", dstack=0)
+            elif e.op == "NOT_NULL":
+                self._push("NOTNULL", dstack=0)
+            elif e.op == "-":
+                self._push("NEG", dstack=0)
+            else:
+                raise DeviceExprError(f"unary {e.op} not fusable")
+            return
+        if isinstance(e, _BinaryOpExpr):
+            name = _X_BINOPS.get(e.op)
+            if name is None or e.as_type is not None:
+                raise DeviceExprError(f"binary {e.op} not fusable")
+            self.emit(e.left)
+            self.emit(e.right)
+            self._push(name, dstack=-1)
+            return
+        raise DeviceExprError(f"{type(e).__name__} not fusable")
+
+
+def try_fused_filter(
+    expr: ColumnExpr, df: HipDataFrame
+) -> Optional[torch.Tensor]:
+    """One-pass device evaluation of a filter predicate; None when the
+    expression (or frame location) isn't fusable."""
+    if not df.device.startswith("cuda"):
+        return None
+    b = _ProgBuilder(df)
+    try:
+        b.emit(expr)
+    except DeviceExprError:
+        return None
+    if (
+        len(b.ops) > 48
+        or len(b.cols) == 0
+        or len(b.cols) > 12
+        or len(b.imm) > 12
+        or b.max_depth > 12
+    ):
+        return None
+    from fugue_amd.hip.ext import get_ext
+
+    ext = get_ext()
+    device = torch.device(df.device)
+    empty = torch.empty(0)
+    return ext.expr_filter(
+        torch.tensor(b.ops, dtype=torch.uint8),
+        torch.tensor(b.aux, dtype=torch.int8),
+        torch.tensor(b.imm, dtype=torch.int64),
+        [c.data.contiguous() for c in b.cols],
+        [
+            c.valid.contiguous() if c.valid is not None else empty
+            for c in b.cols
+        ],
+        torch.tensor([_XDT[c.data.dtype] for c in b.cols], dtype=torch.uint8),
+        df.count(),
+    )
+
+
 def filter_mask(expr: ColumnExpr, df: HipDataFrame) -> torch.Tensor:
+    fused = try_fused_filter(expr, df)
+    if fused is not None:
+        return fused
     d, v = eval_device_expr(expr, df)
     return _as_bool(d, v)

@@ -501,3 +501,51 @@ def test_string_keyed_map_10m_gpu(engine):
     assert len(out) == n_groups
     assert abs(out["s"].sum() - float(vals.sum().item())) < 1e-3
     assert out["n"].sum() == n
+
+
+def test_fused_filter_program_gpu(engine):
+    """The one-pass filter interpreter must agree with the torch
+    elementwise evaluator on arithmetic/comparison/logic predicates over
+    int and float columns with nulls."""
+    import pyarrow as pa
+
+    from fugue_amd.column.expressions import col, lit
+    from fugue_amd.hip import expr as hexpr
+    from fugue_amd.hip.frame import DeviceColumn, HipDataFrame
+    from fugue_amd.schema import Schema
+
+    dev = torch.device(engine.device)
+    gen = torch.Generator(device=dev)
+    gen.manual_seed(3)
+    n = 1_000_000
+    a = torch.randint(-1000, 1000, (n,), dtype=torch.int64, device=dev,
+                      generator=gen)
+    b = torch.rand(n, dtype=torch.float64, device=dev, generator=gen)
+    v = torch.rand(n, device=dev, generator=gen) > 0.1  # 10% nulls on b
+    big = torch.randint(
+        (1 << 55), (1 << 56), (n,), dtype=torch.int64, device=dev,
+        generator=gen,
+    )
+    df = HipDataFrame.from_columns(
+        {
+            "a": DeviceColumn(a, None, pa.int64()),
+            "b": DeviceColumn(b, v, pa.float64()),
+            "big": DeviceColumn(big, None, pa.int64()),
+        },
+        Schema("a:long,b:double,big:long"),
+        engine.device,
+    )
+    exprs = [
+        (col("a") > lit(0)) & (col("b") < lit(0.5)),
+        (col("a") + lit(3)) * lit(2) >= col("a") - lit(1),
+        col("b").is_null() | (col("a") == lit(7)),
+        ~(col("b").not_null()) | (col("b") / lit(2.0) > lit(0.2)),
+        (col("big") == col("big")) & (col("big") > lit((1 << 55) + 12345)),
+        -col("a") > lit(500),
+    ]
+    for e in exprs:
+        fused = hexpr.try_fused_filter(e, df)
+        assert fused is not None, f"not fused: {e}"
+        d, vv = hexpr.eval_device_expr(e, df)
+        ref = hexpr._as_bool(d, vv)
+        assert torch.equal(fused, ref), f"mismatch for {e}"
