@@ -50,6 +50,8 @@ void launch_gb_aggregate_part_big(const void*, const double*,
 void launch_gather_cols(const uint64_t*, const uint64_t*, int, int,
                         const int64_t*, int64_t, hipStream_t);
 void launch_expr_filter(const void*, int64_t, bool*, hipStream_t);
+void launch_expr_value(const void*, int64_t, int, void*, bool*,
+                       hipStream_t);
 void launch_join_count(const int64_t*, int64_t, const int64_t*,
                        const int64_t*, const int64_t*, const int32_t*,
                        const int32_t*, int64_t, int32_t*, hipStream_t);
@@ -451,10 +453,11 @@ struct ExprProgHost {
   unsigned char col_dt[12];
 };
 
-at::Tensor expr_filter(at::Tensor ops, at::Tensor aux, at::Tensor imm,
-                       std::vector<at::Tensor> col_data,
-                       std::vector<at::Tensor> col_valid,
-                       at::Tensor col_dt, int64_t n) {
+static ExprProgHost build_expr_prog(at::Tensor& ops, at::Tensor& aux,
+                                    at::Tensor& imm,
+                                    std::vector<at::Tensor>& col_data,
+                                    std::vector<at::Tensor>& col_valid,
+                                    at::Tensor& col_dt) {
   TORCH_CHECK(ops.numel() <= 48, "program too long");
   TORCH_CHECK(col_data.size() <= 12, "too many columns");
   TORCH_CHECK(imm.numel() <= 12, "too many immediates");
@@ -463,7 +466,7 @@ at::Tensor expr_filter(at::Tensor ops, at::Tensor aux, at::Tensor imm,
   prog.n_ops = (int)ops.numel();
   auto ops_c = ops.to(at::kByte).cpu();
   auto aux_c = aux.to(at::kChar).cpu();
-  auto imm_c = imm.cpu();  // int64 bit patterns
+  auto imm_c = imm.cpu();
   auto dt_c = col_dt.to(at::kByte).cpu();
   for (int i = 0; i < prog.n_ops; ++i) {
     prog.op[i] = ops_c[i].item<uint8_t>();
@@ -472,7 +475,6 @@ at::Tensor expr_filter(at::Tensor ops, at::Tensor aux, at::Tensor imm,
   for (int i = 0; i < (int)imm_c.numel(); ++i) {
     prog.imm[i] = (unsigned long long)imm_c[i].item<int64_t>();
   }
-  at::Tensor out;
   for (size_t c = 0; c < col_data.size(); ++c) {
     check_gpu(col_data[c], "col");
     TORCH_CHECK(col_data[c].is_contiguous(), "columns must be contiguous");
@@ -484,13 +486,35 @@ at::Tensor expr_filter(at::Tensor ops, at::Tensor aux, at::Tensor imm,
       prog.col_valid[c] = reinterpret_cast<unsigned long long>(
           col_valid[c].data_ptr());
     }
-    if (!out.defined()) {
-      out = at::empty({n}, col_data[c].options().dtype(at::kBool));
-    }
   }
-  if (!out.defined()) {
-    TORCH_CHECK(false, "expr_filter needs at least one column");
+  return prog;
+}
+
+std::vector<at::Tensor> expr_value(at::Tensor ops, at::Tensor aux,
+                                   at::Tensor imm,
+                                   std::vector<at::Tensor> col_data,
+                                   std::vector<at::Tensor> col_valid,
+                                   at::Tensor col_dt, int64_t n,
+                                   int64_t out_int) {
+  TORCH_CHECK(col_data.size() >= 1, "needs at least one column");
+  auto prog = build_expr_prog(ops, aux, imm, col_data, col_valid, col_dt);
+  auto opts = col_data[0].options();
+  auto out = at::empty({n}, opts.dtype(out_int ? at::kLong : at::kDouble));
+  auto valid = at::empty({n}, opts.dtype(at::kBool));
+  if (n > 0) {
+    launch_expr_value(&prog, n, (int)out_int, out.data_ptr(),
+                      valid.data_ptr<bool>(), current_stream());
   }
+  return {out, valid};
+}
+
+at::Tensor expr_filter(at::Tensor ops, at::Tensor aux, at::Tensor imm,
+                       std::vector<at::Tensor> col_data,
+                       std::vector<at::Tensor> col_valid,
+                       at::Tensor col_dt, int64_t n) {
+  TORCH_CHECK(col_data.size() >= 1, "needs at least one column");
+  auto prog = build_expr_prog(ops, aux, imm, col_data, col_valid, col_dt);
+  auto out = at::empty({n}, col_data[0].options().dtype(at::kBool));
   if (n > 0) {
     launch_expr_filter(&prog, n, out.data_ptr<bool>(), current_stream());
   }
@@ -571,6 +595,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hash_seed", &hash_seed, "seed a row-hash buffer");
   m.def("gb_mark_reps", &gb_mark_reps,
         "representative rows + h2 verification for hashed group-by");
+  m.def("expr_value", &expr_value,
+        "fused value-expression interpreter");
   m.def("expr_filter", &expr_filter,
         "fused filter-predicate interpreter");
   m.def("gather_columns", &gather_columns,

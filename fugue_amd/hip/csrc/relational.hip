@@ -1251,6 +1251,136 @@ __device__ __forceinline__ double xv_as_f64(const XVal& v) {
   return (double)v.bits;
 }
 
+__device__ __forceinline__ void expr_step(
+    const ExprProg& prog, int op, int aux, XVal* st, int& sp, int64_t i) {
+
+      if (op == XOP_COL) {
+    const void* dp = reinterpret_cast<const void*>(prog.col_data[aux]);
+    XVal v;
+    v.valid = true;
+    switch (prog.col_dt[aux]) {
+      case XDT_F64:
+        v.bits = reinterpret_cast<const long long*>(dp)[i];
+        v.tag = 0;
+        break;
+      case XDT_F32: {
+        double d = (double)reinterpret_cast<const float*>(dp)[i];
+        v.bits = __double_as_longlong(d);
+        v.tag = 0;
+        break;
+      }
+      case XDT_I64:
+        v.bits = reinterpret_cast<const long long*>(dp)[i];
+        v.tag = 1;
+        break;
+      case XDT_I32:
+        v.bits = (long long)reinterpret_cast<const int*>(dp)[i];
+        v.tag = 1;
+        break;
+      case XDT_I16:
+        v.bits = (long long)reinterpret_cast<const short*>(dp)[i];
+        v.tag = 1;
+        break;
+      case XDT_I8:
+        v.bits = (long long)reinterpret_cast<const signed char*>(dp)[i];
+        v.tag = 1;
+        break;
+      default:  // bool
+        v.bits = reinterpret_cast<const bool*>(dp)[i] ? 1 : 0;
+        v.tag = 2;
+    }
+    if (prog.col_valid[aux] != 0ULL) {
+      v.valid = reinterpret_cast<const bool*>(prog.col_valid[aux])[i];
+    }
+    st[sp++] = v;
+  } else if (op == XOP_LIT_D) {
+    XVal v;
+    v.bits = (long long)prog.imm[aux];
+    v.tag = 0;
+    v.valid = true;
+    st[sp++] = v;
+  } else if (op == XOP_LIT_I) {
+    XVal v;
+    v.bits = (long long)prog.imm[aux];
+    v.tag = 1;
+    v.valid = true;
+    st[sp++] = v;
+  } else if (op == XOP_NOT) {
+    XVal& a = st[sp - 1];
+    bool b = a.valid && (a.tag == 0 ? xv_as_f64(a) != 0.0 : a.bits != 0);
+    a.bits = b ? 0 : 1;
+    a.tag = 2;
+    a.valid = true;
+  } else if (op == XOP_ISNULL || op == XOP_NOTNULL) {
+    XVal& a = st[sp - 1];
+    bool b = (op == XOP_ISNULL) ? !a.valid : a.valid;
+    a.bits = b ? 1 : 0;
+    a.tag = 2;
+    a.valid = true;
+  } else if (op == XOP_NEG) {
+    XVal& a = st[sp - 1];
+    if (a.tag == 0)
+      a.bits = __double_as_longlong(-xv_as_f64(a));
+    else
+      a.bits = -a.bits;
+  } else if (op == XOP_AND || op == XOP_OR) {
+    XVal b = st[--sp];
+    XVal& a = st[sp - 1];
+    bool ab = a.valid && (a.tag == 0 ? xv_as_f64(a) != 0.0 : a.bits != 0);
+    bool bb = b.valid && (b.tag == 0 ? xv_as_f64(b) != 0.0 : b.bits != 0);
+    bool r = (op == XOP_AND) ? (ab && bb) : (ab || bb);
+    a.bits = r ? 1 : 0;
+    a.tag = 2;
+    a.valid = true;
+  } else {
+    XVal b = st[--sp];
+    XVal& a = st[sp - 1];
+    bool both_int = (a.tag != 0 && b.tag != 0);
+    bool valid = a.valid && b.valid;
+    if (op >= XOP_ADD && op <= XOP_DIV) {
+      if (op == XOP_DIV || !both_int) {
+        double x = xv_as_f64(a), y = xv_as_f64(b);
+        double r = op == XOP_ADD   ? x + y
+                   : op == XOP_SUB ? x - y
+                   : op == XOP_MUL ? x * y
+                                   : x / y;
+        a.bits = __double_as_longlong(r);
+        a.tag = 0;
+      } else {
+        long long x = a.bits, y = b.bits;
+        long long r = op == XOP_ADD   ? x + y
+                      : op == XOP_SUB ? x - y
+                                      : x * y;
+        a.bits = r;
+        a.tag = 1;
+      }
+      a.valid = valid;
+    } else {  // comparisons: null -> false, result always valid
+      bool r;
+      if (both_int) {
+        long long x = a.bits, y = b.bits;
+        r = op == XOP_LT   ? x < y
+            : op == XOP_LE ? x <= y
+            : op == XOP_GT ? x > y
+            : op == XOP_GE ? x >= y
+            : op == XOP_EQ ? x == y
+                           : x != y;
+      } else {
+        double x = xv_as_f64(a), y = xv_as_f64(b);
+        r = op == XOP_LT   ? x < y
+            : op == XOP_LE ? x <= y
+            : op == XOP_GT ? x > y
+            : op == XOP_GE ? x >= y
+            : op == XOP_EQ ? x == y
+                           : x != y;
+      }
+      a.bits = (r && valid) ? 1 : 0;
+      a.tag = 2;
+      a.valid = true;
+    }
+  }
+}
+
 __global__ __launch_bounds__(BLOCK) void expr_filter_kernel(
     ExprProg prog, int64_t n, bool* __restrict__ out) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -1259,137 +1389,39 @@ __global__ __launch_bounds__(BLOCK) void expr_filter_kernel(
     XVal st[EXPR_STACK];
     int sp = 0;
     for (int p = 0; p < prog.n_ops; ++p) {
-      int op = prog.op[p];
-      int aux = prog.aux[p];
-      if (op == XOP_COL) {
-        const void* dp = reinterpret_cast<const void*>(prog.col_data[aux]);
-        XVal v;
-        v.valid = true;
-        switch (prog.col_dt[aux]) {
-          case XDT_F64:
-            v.bits = reinterpret_cast<const long long*>(dp)[i];
-            v.tag = 0;
-            break;
-          case XDT_F32: {
-            double d = (double)reinterpret_cast<const float*>(dp)[i];
-            v.bits = __double_as_longlong(d);
-            v.tag = 0;
-            break;
-          }
-          case XDT_I64:
-            v.bits = reinterpret_cast<const long long*>(dp)[i];
-            v.tag = 1;
-            break;
-          case XDT_I32:
-            v.bits = (long long)reinterpret_cast<const int*>(dp)[i];
-            v.tag = 1;
-            break;
-          case XDT_I16:
-            v.bits = (long long)reinterpret_cast<const short*>(dp)[i];
-            v.tag = 1;
-            break;
-          case XDT_I8:
-            v.bits = (long long)reinterpret_cast<const signed char*>(dp)[i];
-            v.tag = 1;
-            break;
-          default:  // bool
-            v.bits = reinterpret_cast<const bool*>(dp)[i] ? 1 : 0;
-            v.tag = 2;
-        }
-        if (prog.col_valid[aux] != 0ULL) {
-          v.valid = reinterpret_cast<const bool*>(prog.col_valid[aux])[i];
-        }
-        st[sp++] = v;
-      } else if (op == XOP_LIT_D) {
-        XVal v;
-        v.bits = (long long)prog.imm[aux];
-        v.tag = 0;
-        v.valid = true;
-        st[sp++] = v;
-      } else if (op == XOP_LIT_I) {
-        XVal v;
-        v.bits = (long long)prog.imm[aux];
-        v.tag = 1;
-        v.valid = true;
-        st[sp++] = v;
-      } else if (op == XOP_NOT) {
-        XVal& a = st[sp - 1];
-        bool b = a.valid && (a.tag == 0 ? xv_as_f64(a) != 0.0 : a.bits != 0);
-        a.bits = b ? 0 : 1;
-        a.tag = 2;
-        a.valid = true;
-      } else if (op == XOP_ISNULL || op == XOP_NOTNULL) {
-        XVal& a = st[sp - 1];
-        bool b = (op == XOP_ISNULL) ? !a.valid : a.valid;
-        a.bits = b ? 1 : 0;
-        a.tag = 2;
-        a.valid = true;
-      } else if (op == XOP_NEG) {
-        XVal& a = st[sp - 1];
-        if (a.tag == 0)
-          a.bits = __double_as_longlong(-xv_as_f64(a));
-        else
-          a.bits = -a.bits;
-      } else if (op == XOP_AND || op == XOP_OR) {
-        XVal b = st[--sp];
-        XVal& a = st[sp - 1];
-        bool ab = a.valid && (a.tag == 0 ? xv_as_f64(a) != 0.0 : a.bits != 0);
-        bool bb = b.valid && (b.tag == 0 ? xv_as_f64(b) != 0.0 : b.bits != 0);
-        bool r = (op == XOP_AND) ? (ab && bb) : (ab || bb);
-        a.bits = r ? 1 : 0;
-        a.tag = 2;
-        a.valid = true;
-      } else {
-        XVal b = st[--sp];
-        XVal& a = st[sp - 1];
-        bool both_int = (a.tag != 0 && b.tag != 0);
-        bool valid = a.valid && b.valid;
-        if (op >= XOP_ADD && op <= XOP_DIV) {
-          if (op == XOP_DIV || !both_int) {
-            double x = xv_as_f64(a), y = xv_as_f64(b);
-            double r = op == XOP_ADD   ? x + y
-                       : op == XOP_SUB ? x - y
-                       : op == XOP_MUL ? x * y
-                                       : x / y;
-            a.bits = __double_as_longlong(r);
-            a.tag = 0;
-          } else {
-            long long x = a.bits, y = b.bits;
-            long long r = op == XOP_ADD   ? x + y
-                          : op == XOP_SUB ? x - y
-                                          : x * y;
-            a.bits = r;
-            a.tag = 1;
-          }
-          a.valid = valid;
-        } else {  // comparisons: null -> false, result always valid
-          bool r;
-          if (both_int) {
-            long long x = a.bits, y = b.bits;
-            r = op == XOP_LT   ? x < y
-                : op == XOP_LE ? x <= y
-                : op == XOP_GT ? x > y
-                : op == XOP_GE ? x >= y
-                : op == XOP_EQ ? x == y
-                               : x != y;
-          } else {
-            double x = xv_as_f64(a), y = xv_as_f64(b);
-            r = op == XOP_LT   ? x < y
-                : op == XOP_LE ? x <= y
-                : op == XOP_GT ? x > y
-                : op == XOP_GE ? x >= y
-                : op == XOP_EQ ? x == y
-                               : x != y;
-          }
-          a.bits = (r && valid) ? 1 : 0;
-          a.tag = 2;
-          a.valid = true;
-        }
-      }
+      expr_step(prog, prog.op[p], prog.aux[p], st, sp, i);
     }
     const XVal& top = st[0];
     out[i] = top.valid && (top.tag == 0 ? xv_as_f64(top) != 0.0
                                         : top.bits != 0);
+  }
+}
+
+// value-producing variant: writes the top-of-stack value (as f64 or
+// exact i64 per OUT_INT) plus a validity mask — fuses compound
+// arithmetic in SELECT/assign/aggregate-input expressions
+template <bool OUT_INT>
+__global__ __launch_bounds__(BLOCK) void expr_value_kernel(
+    ExprProg prog, int64_t n, void* __restrict__ out_v,
+    bool* __restrict__ out_valid) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    XVal st[EXPR_STACK];
+    int sp = 0;
+    for (int p = 0; p < prog.n_ops; ++p) {
+      int op = prog.op[p];
+      int aux = prog.aux[p];
+      expr_step(prog, op, aux, st, sp, i);
+    }
+    const XVal& top = st[0];
+    if (OUT_INT) {
+      long long r = top.tag == 0 ? (long long)xv_as_f64(top) : top.bits;
+      reinterpret_cast<long long*>(out_v)[i] = r;
+    } else {
+      reinterpret_cast<double*>(out_v)[i] = xv_as_f64(top);
+    }
+    out_valid[i] = top.valid;
   }
 }
 
@@ -1399,6 +1431,17 @@ void launch_expr_filter(const void* prog, int64_t n, bool* out,
   const ExprProg* pg = reinterpret_cast<const ExprProg*>(prog);
   hipLaunchKernelGGL(expr_filter_kernel, dim3(grid_for(n, 2)), dim3(BLOCK),
                      0, stream, *pg, n, out);
+}
+
+void launch_expr_value(const void* prog, int64_t n, int out_int, void* out_v,
+                       bool* out_valid, hipStream_t stream) {
+  const ExprProg* pg = reinterpret_cast<const ExprProg*>(prog);
+  if (out_int)
+    hipLaunchKernelGGL(expr_value_kernel<true>, dim3(grid_for(n, 2)),
+                       dim3(BLOCK), 0, stream, *pg, n, out_v, out_valid);
+  else
+    hipLaunchKernelGGL(expr_value_kernel<false>, dim3(grid_for(n, 2)),
+                       dim3(BLOCK), 0, stream, *pg, n, out_v, out_valid);
 }
 }  // extern "C"
 

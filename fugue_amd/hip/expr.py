@@ -28,6 +28,10 @@ def eval_device_expr(
     expr: ColumnExpr, df: HipDataFrame
 ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
     """Evaluate a non-aggregate expression; returns (data, valid_or_None)."""
+    if isinstance(expr, _BinaryOpExpr):
+        fused = try_fused_value(expr, df)
+        if fused is not None:
+            return fused
     if isinstance(expr, _LiteralColumnExpr):
         n = df.count()
         v = expr.value
@@ -466,12 +470,46 @@ class _ProgBuilder:
         self.col_idx: Dict[str, int] = {}
         self.depth = 0
         self.max_depth = 0
+        self.tags: List[int] = []
+        self.int64_seen = False
 
     def _push(self, op: str, aux: int = 0, dstack: int = 1) -> None:
         self.ops.append(_XOP[op])
         self.aux.append(aux)
         self.depth += dstack
         self.max_depth = max(self.max_depth, self.depth)
+        # static result-tag simulation (mirrors the kernel's promotion)
+        t = self.tags
+        if op == "COL":
+            dt = self.cols[aux].data.dtype
+            t.append(
+                0 if dt in (torch.float64, torch.float32)
+                else (2 if dt == torch.bool else 1)
+            )
+            if dt == torch.int64:
+                self.int64_seen = True
+        elif op == "LIT_D":
+            t.append(0)
+        elif op == "LIT_I":
+            t.append(1)
+            self.int64_seen = True
+        elif op in ("NOT", "ISNULL", "NOTNULL"):
+            t[-1] = 2
+        elif op == "NEG":
+            pass
+        elif op in ("AND", "OR"):
+            t.pop()
+            t[-1] = 2
+        elif op in ("ADD", "SUB", "MUL"):
+            b = t.pop()
+            a = t[-1]
+            t[-1] = 1 if (a != 0 and b != 0) else 0
+        elif op == "DIV":
+            t.pop()
+            t[-1] = 0
+        else:  # comparisons
+            t.pop()
+            t[-1] = 2
 
     def emit(self, e: ColumnExpr) -> None:
         if isinstance(e, _LiteralColumnExpr):
@@ -528,6 +566,69 @@ class _ProgBuilder:
         raise DeviceExprError(f"{type(e).__name__} not fusable")
 
 
+def _run_program(
+    b: "_ProgBuilder", df: HipDataFrame, mode: str
+) -> Any:
+    from fugue_amd.hip.ext import get_ext
+
+    ext = get_ext()
+    empty = torch.empty(0)
+    args = (
+        torch.tensor(b.ops, dtype=torch.uint8),
+        torch.tensor(b.aux, dtype=torch.int8),
+        torch.tensor(b.imm, dtype=torch.int64),
+        [c.data.contiguous() for c in b.cols],
+        [
+            c.valid.contiguous() if c.valid is not None else empty
+            for c in b.cols
+        ],
+        torch.tensor([_XDT[c.data.dtype] for c in b.cols], dtype=torch.uint8),
+        df.count(),
+    )
+    if mode == "filter":
+        return ext.expr_filter(*args)
+    return ext.expr_value(*args, 1 if mode == "int" else 0)
+
+
+def _fusable(b: "_ProgBuilder") -> bool:
+    return (
+        len(b.ops) <= 48
+        and 0 < len(b.cols) <= 12
+        and len(b.imm) <= 12
+        and b.max_depth <= 12
+    )
+
+
+def try_fused_value(
+    expr: ColumnExpr, df: HipDataFrame
+) -> Optional[Tuple[torch.Tensor, Optional[torch.Tensor]]]:
+    """One-pass evaluation of a compound arithmetic expression; returns
+    (data, valid) like eval_device_expr, or None when not fusable /
+    not worth fusing (fewer than 3 fused launches saved)."""
+    if not df.device.startswith("cuda"):
+        return None
+    if not isinstance(expr, _BinaryOpExpr) or expr.op not in "+-*/":
+        return None
+    if expr.as_type is not None:
+        return None
+    b = _ProgBuilder(df)
+    try:
+        b.emit(expr)
+    except DeviceExprError:
+        return None
+    n_arith = sum(1 for o in b.ops if _XOP["ADD"] <= o <= _XOP["NE"])
+    if n_arith < 2 or not _fusable(b) or b.tags[-1] == 2:
+        return None
+    out_int = b.tags[-1] == 1
+    if out_int and not b.int64_seen:
+        # torch promotion would keep a narrower int dtype; skip fusion
+        return None
+    vals, valid = _run_program(b, df, "int" if out_int else "value")
+    if all(c.valid is None for c in b.cols):
+        valid = None
+    return vals, valid
+
+
 def try_fused_filter(
     expr: ColumnExpr, df: HipDataFrame
 ) -> Optional[torch.Tensor]:
@@ -540,31 +641,9 @@ def try_fused_filter(
         b.emit(expr)
     except DeviceExprError:
         return None
-    if (
-        len(b.ops) > 48
-        or len(b.cols) == 0
-        or len(b.cols) > 12
-        or len(b.imm) > 12
-        or b.max_depth > 12
-    ):
+    if not _fusable(b):
         return None
-    from fugue_amd.hip.ext import get_ext
-
-    ext = get_ext()
-    device = torch.device(df.device)
-    empty = torch.empty(0)
-    return ext.expr_filter(
-        torch.tensor(b.ops, dtype=torch.uint8),
-        torch.tensor(b.aux, dtype=torch.int8),
-        torch.tensor(b.imm, dtype=torch.int64),
-        [c.data.contiguous() for c in b.cols],
-        [
-            c.valid.contiguous() if c.valid is not None else empty
-            for c in b.cols
-        ],
-        torch.tensor([_XDT[c.data.dtype] for c in b.cols], dtype=torch.uint8),
-        df.count(),
-    )
+    return _run_program(b, df, "filter")
 
 
 def filter_mask(expr: ColumnExpr, df: HipDataFrame) -> torch.Tensor:

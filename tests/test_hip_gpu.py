@@ -549,3 +549,60 @@ def test_fused_filter_program_gpu(engine):
         d, vv = hexpr.eval_device_expr(e, df)
         ref = hexpr._as_bool(d, vv)
         assert torch.equal(fused, ref), f"mismatch for {e}"
+
+
+def test_fused_value_program_gpu(engine):
+    """The value-producing interpreter must agree with the torch
+    evaluator on compound arithmetic (float result, exact int64 result,
+    null propagation)."""
+    import pyarrow as pa
+
+    from fugue_amd.column.expressions import col, lit
+    from fugue_amd.hip import expr as hexpr
+    from fugue_amd.hip.frame import DeviceColumn, HipDataFrame
+    from fugue_amd.schema import Schema
+
+    dev = torch.device(engine.device)
+    gen = torch.Generator(device=dev)
+    gen.manual_seed(5)
+    n = 1_000_000
+    p = torch.rand(n, dtype=torch.float64, device=dev, generator=gen) * 1e5
+    dsc = torch.rand(n, dtype=torch.float64, device=dev, generator=gen) * 0.1
+    q = torch.randint(1, 50, (n,), dtype=torch.int64, device=dev,
+                      generator=gen)
+    vmask = torch.rand(n, device=dev, generator=gen) > 0.05
+    df = HipDataFrame.from_columns(
+        {
+            "p": DeviceColumn(p, vmask, pa.float64()),
+            "d": DeviceColumn(dsc, None, pa.float64()),
+            "q": DeviceColumn(q, None, pa.int64()),
+        },
+        Schema("p:double,d:double,q:long"),
+        engine.device,
+    )
+    exprs = [
+        col("p") * (lit(1) - col("d")),
+        (col("p") + col("d")) / (col("q") + lit(1)),
+        (col("q") * lit(3) + lit(7)) - col("q"),
+    ]
+    for e in exprs:
+        fused = hexpr.try_fused_value(e, df)
+        assert fused is not None, f"not fused: {e}"
+        fd, fv = fused
+        # reference: the torch evaluator with fusion disabled
+        orig = hexpr.try_fused_value
+        hexpr.try_fused_value = lambda *_: None
+        try:
+            rd, rv = hexpr.eval_device_expr(e, df)
+        finally:
+            hexpr.try_fused_value = orig
+        assert fd.dtype == rd.dtype, (e, fd.dtype, rd.dtype)
+        if fd.is_floating_point():
+            mask = rv if rv is not None else torch.ones_like(vmask)
+            assert torch.allclose(fd[mask], rd[mask], rtol=1e-12)
+        else:
+            assert torch.equal(fd, rd)
+        if rv is None:
+            assert fv is None or bool(fv.all().item())
+        else:
+            assert fv is not None and torch.equal(fv, rv)
