@@ -29,7 +29,12 @@ void launch_scatter(const int32_t*, int64_t*, int64_t*, int64_t, hipStream_t);
 void launch_gb_aggregate(const int64_t*, const double*, const bool*,
                          const int32_t*, int, int64_t, int64_t*, double*,
                          int64_t*, int64_t, int, hipStream_t);
+void launch_join_emit_unique(const int64_t*, int64_t, const int64_t*,
+                             const int64_t*, const int64_t*,
+                             const int32_t*, const int32_t*, int64_t, int,
+                             int64_t*, int64_t*, int64_t*, hipStream_t);
 void launch_join_build(const int64_t*, int64_t, int32_t*, int32_t*, int64_t,
+                       int32_t*,
                        hipStream_t);
 void launch_gb_part_hist(const int64_t*, int64_t, int, int64_t*, int,
                          int64_t*, hipStream_t);
@@ -95,6 +100,27 @@ const bool* opt_valid_ptr(const c10::optional<at::Tensor>& v) {
 }
 
 }  // namespace
+
+std::vector<at::Tensor> join_emit_unique(
+    at::Tensor pkeys, at::Tensor bkeys, c10::optional<at::Tensor> ph2,
+    c10::optional<at::Tensor> bh2, at::Tensor heads, at::Tensor next,
+    int64_t mode) {
+  check_gpu(pkeys, "pkeys");
+  int64_t np = pkeys.numel();
+  int64_t tsize = heads.numel();
+  auto out_pi = at::empty({np}, pkeys.options());
+  auto out_bi = at::empty({np}, pkeys.options());
+  auto cursor = at::zeros({1}, pkeys.options());
+  if (np > 0) {
+    launch_join_emit_unique(
+        pkeys.data_ptr<int64_t>(), np, bkeys.data_ptr<int64_t>(),
+        opt_i64_ptr(ph2), opt_i64_ptr(bh2), heads.data_ptr<int32_t>(),
+        next.data_ptr<int32_t>(), tsize, (int)mode,
+        out_pi.data_ptr<int64_t>(), out_bi.data_ptr<int64_t>(),
+        cursor.data_ptr<int64_t>(), current_stream());
+  }
+  return {out_pi, out_bi, cursor};
+}
 
 // Combine one column into the running row hash (out int64 viewed as u64).
 void hash_column(at::Tensor data, c10::optional<at::Tensor> valid,
@@ -280,12 +306,15 @@ std::vector<at::Tensor> join_build(at::Tensor keys, int64_t tsize) {
   auto heads = at::full({tsize}, -1, keys.options().dtype(at::kInt));
   auto next = at::empty({std::max<int64_t>(n, 1)},
                         keys.options().dtype(at::kInt));
+  auto dup = at::zeros({1}, keys.options().dtype(at::kInt));
   if (n > 0) {
     launch_join_build(keys.data_ptr<int64_t>(), n, heads.data_ptr<int32_t>(),
-                      next.data_ptr<int32_t>(), tsize, current_stream());
+                      next.data_ptr<int32_t>(), tsize,
+                      dup.data_ptr<int32_t>(), current_stream());
   }
-  return {heads, next};
+  return {heads, next, dup};
 }
+
 
 namespace {
 const int64_t* opt_i64_ptr(const c10::optional<at::Tensor>& v) {
@@ -590,6 +619,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gb_aggregate_partitioned", &gb_aggregate_partitioned,
         "partitioned (2-phase) hash group-by aggregation");
   m.def("join_build", &join_build, "build chained hash table");
+  m.def("join_emit_unique", &join_emit_unique,
+        "single-pass join emit for unique build keys");
   m.def("hash_string_column", &hash_string_column,
         "combine a string column into the running row hash");
   m.def("hash_seed", &hash_seed, "seed a row-hash buffer");

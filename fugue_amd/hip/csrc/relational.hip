@@ -1713,14 +1713,91 @@ __global__ __launch_bounds__(BLOCK) void join_build_kernel(
     const int64_t* __restrict__ keys, int64_t n,
     int32_t* __restrict__ heads,   // [tsize] init -1
     int32_t* __restrict__ next,    // [n]
-    int64_t tsize) {
+    int64_t tsize,
+    int32_t* __restrict__ dup /* set when a key occurs twice; may be
+                                 null */) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
-    uint64_t h = mix64((uint64_t)keys[i]);
+    int64_t key = keys[i];
+    uint64_t h = mix64((uint64_t)key);
     int64_t slot = (int64_t)(h & (uint64_t)(tsize - 1));
     int32_t old = atomicExch(&heads[slot], (int32_t)i);
     next[i] = old;
+    if (dup != nullptr && *dup == 0) {
+      // the chain behind us is fully linked; if our key is in it, the
+      // build side has duplicates (disables the single-pass emit)
+      int32_t cur = old;
+      while (cur >= 0) {
+        if (keys[cur] == key) {
+          atomicOr(dup, 1);
+          break;
+        }
+        cur = next[cur];
+      }
+    }
+  }
+}
+
+// single-pass emit for UNIQUE build keys (≤1 match per probe): walks
+// the chain once and appends via a global cursor — replaces the
+// count+scan+emit 3-kernel pipeline.  mode: 0=inner 2=semi 3=anti
+// (left is handled positionally by join_left_unique_kernel).
+__global__ __launch_bounds__(BLOCK) void join_emit_unique_kernel(
+    const int64_t* __restrict__ pkeys, int64_t np,
+    const int64_t* __restrict__ bkeys,
+    const int64_t* __restrict__ ph2, const int64_t* __restrict__ bh2,
+    const int32_t* __restrict__ heads, const int32_t* __restrict__ next,
+    int64_t tsize, int mode,
+    int64_t* __restrict__ out_pi, int64_t* __restrict__ out_bi,
+    int64_t* __restrict__ cursor) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < np;
+       i += stride) {
+    int64_t key = pkeys[i];
+    uint64_t h = mix64((uint64_t)key);
+    int32_t cur = heads[(int64_t)(h & (uint64_t)(tsize - 1))];
+    int32_t match = -1;
+    while (cur >= 0) {
+      if (bkeys[cur] == key && (ph2 == nullptr || bh2[cur] == ph2[i])) {
+        match = cur;
+        break;
+      }
+      cur = next[cur];
+    }
+    bool emit = mode == 3 ? (match < 0) : (match >= 0);
+    if (emit) {
+      int64_t pos = atomicAdd((unsigned long long*)cursor, 1ULL);
+      out_pi[pos] = i;
+      if (out_bi != nullptr) out_bi[pos] = mode == 3 ? -1 : match;
+    }
+  }
+}
+
+// positional left join for unique build keys: out slot i = probe i
+__global__ __launch_bounds__(BLOCK) void join_left_unique_kernel(
+    const int64_t* __restrict__ pkeys, int64_t np,
+    const int64_t* __restrict__ bkeys,
+    const int64_t* __restrict__ ph2, const int64_t* __restrict__ bh2,
+    const int32_t* __restrict__ heads, const int32_t* __restrict__ next,
+    int64_t tsize,
+    int64_t* __restrict__ out_pi, int64_t* __restrict__ out_bi) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < np;
+       i += stride) {
+    int64_t key = pkeys[i];
+    uint64_t h = mix64((uint64_t)key);
+    int32_t cur = heads[(int64_t)(h & (uint64_t)(tsize - 1))];
+    int32_t match = -1;
+    while (cur >= 0) {
+      if (bkeys[cur] == key && (ph2 == nullptr || bh2[cur] == ph2[i])) {
+        match = cur;
+        break;
+      }
+      cur = next[cur];
+    }
+    out_pi[i] = i;
+    out_bi[i] = match;
   }
 }
 
@@ -1933,9 +2010,27 @@ __global__ __launch_bounds__(BLOCK) void join_mark_build_kernel(
 extern "C" {
 
 void launch_join_build(const int64_t* keys, int64_t n, int32_t* heads,
-                       int32_t* next, int64_t tsize, hipStream_t stream) {
+                       int32_t* next, int64_t tsize, int32_t* dup,
+                       hipStream_t stream) {
   hipLaunchKernelGGL(join_build_kernel, dim3(grid_for(n)), dim3(BLOCK), 0,
-                     stream, keys, n, heads, next, tsize);
+                     stream, keys, n, heads, next, tsize, dup);
+}
+
+void launch_join_emit_unique(const int64_t* pkeys, int64_t np,
+                             const int64_t* bkeys, const int64_t* ph2,
+                             const int64_t* bh2, const int32_t* heads,
+                             const int32_t* next, int64_t tsize, int mode,
+                             int64_t* out_pi, int64_t* out_bi,
+                             int64_t* cursor, hipStream_t stream) {
+  if (mode == 1) {
+    hipLaunchKernelGGL(join_left_unique_kernel, dim3(grid_for(np)),
+                       dim3(BLOCK), 0, stream, pkeys, np, bkeys, ph2, bh2,
+                       heads, next, tsize, out_pi, out_bi);
+  } else {
+    hipLaunchKernelGGL(join_emit_unique_kernel, dim3(grid_for(np)),
+                       dim3(BLOCK), 0, stream, pkeys, np, bkeys, ph2, bh2,
+                       heads, next, tsize, mode, out_pi, out_bi, cursor);
+  }
 }
 
 void launch_join_count(const int64_t* pkeys, int64_t np, const int64_t* bkeys,

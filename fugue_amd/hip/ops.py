@@ -737,10 +737,22 @@ def hash_join_indices(
         raise FugueBug(f"unsupported join mode {how}")
     nb = int(build_keys.numel())
     tsize = _next_pow2(max(16, nb * 2))
-    heads, nxt = ext.join_build(build_keys, tsize)
-    # 2-pass count+prefix+emit; a 3-pass total+chunked-reservation variant
-    # (ext.join_pairs) measured SLOWER — the random chain walk dominates,
-    # not the streaming counts/cumsum (profiles/NOTES.md)
+    heads, nxt, dup = ext.join_build(build_keys, tsize)
+    mode = {"inner": 0, "left": 1, "semi": 2, "anti": 3}[how]
+    if int(dup.item()) == 0:
+        # unique build keys (≤1 match per probe): single-pass emit —
+        # no count pass, no prefix scan; output ≤ n_probe
+        out_p, out_b, cursor = ext.join_emit_unique(
+            probe_keys, build_keys, probe_h2, build_h2, heads, nxt, mode
+        )
+        if how == "left":
+            return out_p, out_b
+        total = int(cursor.item())
+        return out_p[:total], out_b[:total]
+    # duplicate build keys: 2-pass count+prefix+emit; a 3-pass
+    # total+chunked-reservation variant (ext.join_pairs) measured
+    # SLOWER — the random chain walk dominates, not the streaming
+    # counts/cumsum (profiles/NOTES.md)
     counts = ext.join_count(
         probe_keys, build_keys, probe_h2, build_h2, heads, nxt, tsize
     )
@@ -757,7 +769,6 @@ def hash_join_indices(
     if out_counts.numel() > 1:
         torch.cumsum(out_counts[:-1], 0, out=offsets[1:])
     total = int(out_counts.sum().item())
-    mode = {"inner": 0, "left": 1, "semi": 2, "anti": 3}[how]
     out_p, out_b = ext.join_emit(
         probe_keys, build_keys, probe_h2, build_h2, heads, nxt, tsize,
         offsets, total, mode
@@ -790,7 +801,7 @@ def mark_matched_build_rows(
     ext = get_ext()
     nb = int(build_keys.numel())
     tsize = _next_pow2(max(16, nb * 2))
-    heads, nxt = ext.join_build(build_keys, tsize)
+    heads, nxt, _dup = ext.join_build(build_keys, tsize)
     return ext.join_mark_build(
         probe_keys, build_keys, probe_h2, build_h2, heads, nxt, tsize, nb
     )
