@@ -101,26 +101,6 @@ const bool* opt_valid_ptr(const c10::optional<at::Tensor>& v) {
 
 }  // namespace
 
-std::vector<at::Tensor> join_emit_unique(
-    at::Tensor pkeys, at::Tensor bkeys, c10::optional<at::Tensor> ph2,
-    c10::optional<at::Tensor> bh2, at::Tensor heads, at::Tensor next,
-    int64_t mode) {
-  check_gpu(pkeys, "pkeys");
-  int64_t np = pkeys.numel();
-  int64_t tsize = heads.numel();
-  auto out_pi = at::empty({np}, pkeys.options());
-  auto out_bi = at::empty({np}, pkeys.options());
-  auto cursor = at::zeros({1}, pkeys.options());
-  if (np > 0) {
-    launch_join_emit_unique(
-        pkeys.data_ptr<int64_t>(), np, bkeys.data_ptr<int64_t>(),
-        opt_i64_ptr(ph2), opt_i64_ptr(bh2), heads.data_ptr<int32_t>(),
-        next.data_ptr<int32_t>(), tsize, (int)mode,
-        out_pi.data_ptr<int64_t>(), out_bi.data_ptr<int64_t>(),
-        cursor.data_ptr<int64_t>(), current_stream());
-  }
-  return {out_pi, out_bi, cursor};
-}
 
 // Combine one column into the running row hash (out int64 viewed as u64).
 void hash_column(at::Tensor data, c10::optional<at::Tensor> valid,
@@ -322,6 +302,27 @@ const int64_t* opt_i64_ptr(const c10::optional<at::Tensor>& v) {
   return v->data_ptr<int64_t>();
 }
 }  // namespace
+
+std::vector<at::Tensor> join_emit_unique(
+    at::Tensor pkeys, at::Tensor bkeys, c10::optional<at::Tensor> ph2,
+    c10::optional<at::Tensor> bh2, at::Tensor heads, at::Tensor next,
+    int64_t mode) {
+  check_gpu(pkeys, "pkeys");
+  int64_t np = pkeys.numel();
+  int64_t tsize = heads.numel();
+  auto out_pi = at::empty({np}, pkeys.options());
+  auto out_bi = at::empty({np}, pkeys.options());
+  auto cursor = at::zeros({1}, pkeys.options());
+  if (np > 0) {
+    launch_join_emit_unique(
+        pkeys.data_ptr<int64_t>(), np, bkeys.data_ptr<int64_t>(),
+        opt_i64_ptr(ph2), opt_i64_ptr(bh2), heads.data_ptr<int32_t>(),
+        next.data_ptr<int32_t>(), tsize, (int)mode,
+        out_pi.data_ptr<int64_t>(), out_bi.data_ptr<int64_t>(),
+        cursor.data_ptr<int64_t>(), current_stream());
+  }
+  return {out_pi, out_bi, cursor};
+}
 
 at::Tensor join_count(at::Tensor pkeys, at::Tensor bkeys,
                       c10::optional<at::Tensor> ph2,
