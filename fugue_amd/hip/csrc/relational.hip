@@ -1713,28 +1713,38 @@ __global__ __launch_bounds__(BLOCK) void join_build_kernel(
     const int64_t* __restrict__ keys, int64_t n,
     int32_t* __restrict__ heads,   // [tsize] init -1
     int32_t* __restrict__ next,    // [n]
-    int64_t tsize,
-    int32_t* __restrict__ dup /* set when a key occurs twice; may be
-                                 null */) {
+    int64_t tsize) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
-    int64_t key = keys[i];
-    uint64_t h = mix64((uint64_t)key);
+    uint64_t h = mix64((uint64_t)keys[i]);
     int64_t slot = (int64_t)(h & (uint64_t)(tsize - 1));
     int32_t old = atomicExch(&heads[slot], (int32_t)i);
     next[i] = old;
-    if (dup != nullptr && *dup == 0) {
-      // the chain behind us is fully linked; if our key is in it, the
-      // build side has duplicates (disables the single-pass emit)
-      int32_t cur = old;
-      while (cur >= 0) {
-        if (keys[cur] == key) {
-          atomicOr(dup, 1);
-          break;
-        }
-        cur = next[cur];
+  }
+}
+
+// post-build pass: set dup when any key occurs twice (chains are fully
+// linked here — doing this inside the build kernel would race with
+// other threads' unwritten next[] entries)
+__global__ __launch_bounds__(BLOCK) void join_dup_check_kernel(
+    const int64_t* __restrict__ keys, int64_t n,
+    const int32_t* __restrict__ heads, const int32_t* __restrict__ next,
+    int64_t tsize, int32_t* __restrict__ dup) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    if (*dup != 0) return;
+    int64_t key = keys[i];
+    uint64_t h = mix64((uint64_t)key);
+    int32_t cur = heads[(int64_t)(h & (uint64_t)(tsize - 1))];
+    // a duplicate exists iff a DIFFERENT row with my key precedes me
+    while (cur >= 0 && cur != (int32_t)i) {
+      if (keys[cur] == key) {
+        atomicOr(dup, 1);
+        return;
       }
+      cur = next[cur];
     }
   }
 }
@@ -2013,7 +2023,11 @@ void launch_join_build(const int64_t* keys, int64_t n, int32_t* heads,
                        int32_t* next, int64_t tsize, int32_t* dup,
                        hipStream_t stream) {
   hipLaunchKernelGGL(join_build_kernel, dim3(grid_for(n)), dim3(BLOCK), 0,
-                     stream, keys, n, heads, next, tsize, dup);
+                     stream, keys, n, heads, next, tsize);
+  if (dup != nullptr) {
+    hipLaunchKernelGGL(join_dup_check_kernel, dim3(grid_for(n)), dim3(BLOCK),
+                       0, stream, keys, n, heads, next, tsize, dup);
+  }
 }
 
 void launch_join_emit_unique(const int64_t* pkeys, int64_t np,
