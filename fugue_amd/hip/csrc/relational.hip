@@ -1776,10 +1776,23 @@ __global__ __launch_bounds__(BLOCK) void join_emit_unique_kernel(
       cur = next[cur];
     }
     bool emit = mode == 3 ? (match < 0) : (match >= 0);
-    if (emit) {
-      int64_t pos = atomicAdd((unsigned long long*)cursor, 1ULL);
-      out_pi[pos] = i;
-      if (out_bi != nullptr) out_bi[pos] = mode == 3 ? -1 : match;
+    // wave-aggregated cursor reservation: one atomic per 64 lanes
+    unsigned long long ballot = __ballot(emit);
+    if (ballot != 0ULL) {
+      int lane = __lane_id();
+      int leader = __ffsll((long long)ballot) - 1;
+      long long base = 0;
+      if (lane == leader) {
+        base = (long long)atomicAdd((unsigned long long*)cursor,
+                                    (unsigned long long)__popcll(ballot));
+      }
+      base = __shfl(base, leader);
+      if (emit) {
+        int64_t pos =
+            base + __popcll(ballot & ((1ULL << lane) - 1ULL));
+        out_pi[pos] = i;
+        if (out_bi != nullptr) out_bi[pos] = mode == 3 ? -1 : match;
+      }
     }
   }
 }
