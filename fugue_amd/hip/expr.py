@@ -605,6 +605,10 @@ def try_fused_value(
     """One-pass evaluation of a compound arithmetic expression; returns
     (data, valid) like eval_device_expr, or None when not fusable /
     not worth fusing (fewer than 3 fused launches saved)."""
+    import os as _os
+
+    if _os.environ.get("FUGUE_EXPR_FUSE", "1") == "0":
+        return None
     if not df.device.startswith("cuda"):
         return None
     if not isinstance(expr, _BinaryOpExpr) or expr.op not in "+-*/":

@@ -739,7 +739,12 @@ def hash_join_indices(
     tsize = _next_pow2(max(16, nb * 2))
     heads, nxt, dup = ext.join_build(build_keys, tsize)
     mode = {"inner": 0, "left": 1, "semi": 2, "anti": 3}[how]
-    if int(dup.item()) == 0:
+    import os as _os_j
+
+    if (
+        _os_j.environ.get("FUGUE_JOIN_UNIQUE", "1") != "0"
+        and int(dup.item()) == 0
+    ):
         # unique build keys (≤1 match per probe): single-pass emit —
         # no count pass, no prefix scan; output ≤ n_probe
         out_p, out_b, cursor = ext.join_emit_unique(
