@@ -745,15 +745,28 @@ def hash_join_indices(
         _os_j.environ.get("FUGUE_JOIN_UNIQUE", "1") != "0"
         and int(dup.item()) == 0
     ):
-        # unique build keys (≤1 match per probe): single-pass emit —
-        # no count pass, no prefix scan; output ≤ n_probe
-        out_p, out_b, cursor = ext.join_emit_unique(
-            probe_keys, build_keys, probe_h2, build_h2, heads, nxt, mode
+        # unique build keys (≤1 match per probe): ONE chain walk writes
+        # the match index positionally; inner/semi/anti then compact
+        # with the block-scan compaction kernel (no global-cursor
+        # contention, no second walk)
+        np_ = int(probe_keys.numel())
+        out_p, out_b, _cur = ext.join_emit_unique(
+            probe_keys, build_keys, probe_h2, build_h2, heads, nxt, 1
         )
         if how == "left":
             return out_p, out_b
-        total = int(cursor.item())
-        return out_p[:total], out_b[:total]
+        matched = out_b >= 0
+        if how == "anti":
+            matched = ~matched
+        total = int(matched.sum().item())
+        arange = torch.arange(
+            np_, dtype=torch.int64, device=probe_keys.device
+        )
+        if how == "semi" or how == "anti":
+            (pi,) = ext.compact_columns(matched, [arange], total)
+            return pi, torch.full_like(pi, -1) if how == "anti" else                 ext.compact_columns(matched, [out_b], total)[0]
+        pi, bi = ext.compact_columns(matched, [arange, out_b], total)
+        return pi, bi
     # duplicate build keys: 2-pass count+prefix+emit; a 3-pass
     # total+chunked-reservation variant (ext.join_pairs) measured
     # SLOWER — the random chain walk dominates, not the streaming
