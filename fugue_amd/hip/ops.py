@@ -762,9 +762,9 @@ def hash_join_indices(
         arange = torch.arange(
             np_, dtype=torch.int64, device=probe_keys.device
         )
-        if how == "semi" or how == "anti":
+        if how == "anti":
             (pi,) = ext.compact_columns(matched, [arange], total)
-            return pi, torch.full_like(pi, -1) if how == "anti" else                 ext.compact_columns(matched, [out_b], total)[0]
+            return pi, torch.full_like(pi, -1)
         pi, bi = ext.compact_columns(matched, [arange, out_b], total)
         return pi, bi
     # duplicate build keys: 2-pass count+prefix+emit; a 3-pass
