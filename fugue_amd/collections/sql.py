@@ -30,9 +30,57 @@ class TempTableName:
 _TMP_RE = re.compile(r"<tmpdf:(?P<key>[^>]+)>")
 
 
+# dialect families that share the grammar this engine's SELECT parser
+# accepts (spark-flavored: backtick identifiers, standard functions)
+_SPARKLIKE = {"spark", "hive", "databricks", "trino", "presto"}
+
+# function spellings normalized to the internal dialect
+_FN_RENAMES = {
+    "ifnull": "COALESCE",
+    "nvl": "COALESCE",
+}
+
+
+def _builtin_transpile(raw: str, from_dialect: str, to_dialect: str) -> str:
+    """Minimal cross-dialect normalization for the supported grammar:
+    double-quoted identifiers (duckdb/postgres style) become backticked,
+    and a small set of function aliases are rewritten.  Constructs
+    outside the engine's grammar surface still fail in the parser with
+    a position-annotated error."""
+    out = []
+    i, n = 0, len(raw)
+    while i < n:
+        ch = raw[i]
+        if ch == "'":  # string literal: copy verbatim ('' escapes)
+            j = i + 1
+            while j < n:
+                if raw[j] == "'" and (j + 1 >= n or raw[j + 1] != "'"):
+                    break
+                j += 2 if raw[j] == "'" else 1
+            out.append(raw[i : j + 1])
+            i = j + 1
+            continue
+        if ch == '"' and from_dialect not in _SPARKLIKE:
+            j = raw.find('"', i + 1)
+            if j < 0:
+                out.append(raw[i:])
+                break
+            out.append("`" + raw[i + 1 : j] + "`")
+            i = j + 1
+            continue
+        out.append(ch)
+        i += 1
+    res = "".join(out)
+    for name, repl in _FN_RENAMES.items():
+        res = re.sub(rf"\b{name}\s*\(", repl + "(", res, flags=re.I)
+    return res
+
+
 def transpile_sql(raw: str, from_dialect: Optional[str], to_dialect: Optional[str]) -> str:
-    """Dialect transpile hook; identity unless a transpiler plugin is
-    registered (the reference used sqlglot here)."""
+    """Dialect transpile: plugin-overridable; the built-in fallback
+    normalizes quote styles and function aliases between the supported
+    dialect families (reference used sqlglot, ``fugue/collections/
+    sql.py:24``)."""
     from fugue_amd.utils.registry import try_run_plugin
 
     if (
@@ -44,7 +92,9 @@ def transpile_sql(raw: str, from_dialect: Optional[str], to_dialect: Optional[st
     ok, res = try_run_plugin("transpile_sql", raw, from_dialect, to_dialect)
     if ok:
         return res
-    return raw
+    if from_dialect in _SPARKLIKE and to_dialect in _SPARKLIKE:
+        return raw
+    return _builtin_transpile(raw, from_dialect, to_dialect)
 
 
 class StructuredRawSQL:

@@ -53,6 +53,16 @@ def _feed(obj: Any, parts: list) -> None:
         if name is None:
             name = type(obj).__qualname__
         parts.append("\0f" + mod + "." + name)
+        # distinct closures over the same code differ by their captured
+        # values — include them so such functions don't hash-collide
+        # (still process-stable: values, never object ids)
+        clo = getattr(obj, "__closure__", None)
+        if clo:
+            for cell in clo:
+                try:
+                    _feed(cell.cell_contents, parts)
+                except Exception:  # pragma: no cover - self cycles etc.
+                    parts.append("\0?")
     else:
         parts.append("\0o" + repr(obj))
 

@@ -35,15 +35,22 @@ def _torchrun(nproc: int, script: str, *args: str, timeout: int = 420) -> str:
         "--master-port", str(_free_port()),
         os.path.join(REPO, script), *args,
     ]
-    res = subprocess.run(
-        cmd, cwd=REPO, env=env, capture_output=True, text=True,
-        timeout=timeout,
+    last = None
+    for attempt in range(2):  # gloo rendezvous can flake under load
+        res = subprocess.run(
+            cmd,
+            cwd=REPO, env=env, capture_output=True, text=True,
+            timeout=timeout,
+        )
+        if res.returncode == 0:
+            return res.stdout
+        last = res
+        cmd[cmd.index("--master-port") + 1] = str(_free_port())
+    assert last is not None and last.returncode == 0, (
+        f"torchrun failed\nSTDOUT:\n{last.stdout[-4000:]}\n"
+        f"STDERR:\n{last.stderr[-4000:]}"
     )
-    assert res.returncode == 0, (
-        f"torchrun failed\nSTDOUT:\n{res.stdout[-4000:]}\n"
-        f"STDERR:\n{res.stderr[-4000:]}"
-    )
-    return res.stdout
+    return last.stdout
 
 
 def _last_json(out: str) -> dict:

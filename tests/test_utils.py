@@ -117,3 +117,27 @@ def test_fugue_test_conf_backend_merge():
         assert "fugue.hip.broadcast_threshold_bytes" not in nat
     finally:
         set_global_test_conf({})
+
+
+def test_transpile_sql_builtin():
+    """Cross-dialect normalization: double-quoted
+    identifiers and function aliases from duckdb-style SQL are rewritten
+    for the internal spark-flavored grammar; spark-family pairs pass
+    through; plugins can override."""
+    from fugue_amd.collections.sql import StructuredRawSQL, transpile_sql
+
+    assert (
+        transpile_sql('SELECT "a b", ifnull(x, 1) FROM t', "duckdb", "spark")
+        == "SELECT `a b`, COALESCE(x, 1) FROM t"
+    )
+    # string literals untouched
+    assert (
+        transpile_sql("SELECT 'has \"quotes\"' FROM t", "duckdb", "spark")
+        == "SELECT 'has \"quotes\"' FROM t"
+    )
+    # same family: identity
+    assert transpile_sql("SELECT `a` FROM t", "hive", "spark") == "SELECT `a` FROM t"
+    # end to end through StructuredRawSQL.construct
+    s = StructuredRawSQL([(False, 'SELECT "x y" FROM '), (True, "t")],
+                         dialect="duckdb")
+    assert "`x y`" in s.construct(dialect="spark")
