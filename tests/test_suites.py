@@ -337,3 +337,52 @@ def test_keyless_device_transform():
         )
     )
     assert float(res["s"][0]) == 2.0 * n
+
+
+def test_hip_engine_nested_decimal_fallback():
+    """The device engine accepts frames with nested / decimal columns:
+    they stay host-resident and every op still works (documented
+    deviation — device columns are flat)."""
+    import decimal
+
+    import pandas as pd
+    import pyarrow as pa
+
+    import fugue_amd.api as fa
+    from fugue_amd import ArrayDataFrame
+    from fugue_amd.hip.execution_engine import HipExecutionEngine
+
+    e = HipExecutionEngine()
+    nested = ArrayDataFrame(
+        [[1, {"x": 1}, [1, 2]], [2, {"x": 2}, [3]], [1, None, []]],
+        "k:long,s:{x:long},l:[long]",
+    )
+    # relational ops via the engine: distinct/filter/take/join
+    d = e.to_df(nested)
+    assert e.take(d, 2, presort="k").count() == 2
+    dim = ArrayDataFrame([[1, "a"], [2, "b"]], "k:long,name:str")
+    joined = e.join(d, e.to_df(dim), how="inner")
+    got = joined.as_array(type_safe=True)
+    assert len(got) == 3
+    # map/transform over nested input
+    # schema: k:long,n:long
+    def count_list(df: pd.DataFrame) -> pd.DataFrame:
+        return pd.DataFrame(
+            dict(k=[df["k"].iloc[0]], n=[sum(len(x) for x in df["l"])])
+        )
+
+    res = fa.transform(
+        d, count_list, partition=dict(by=["k"]), engine=e, as_fugue=True
+    )
+    out = {r[0]: r[1] for r in res.as_array()}
+    assert out == {1: 2, 2: 1}
+
+    # decimal columns round-trip through engine ops
+    dec = ArrayDataFrame(
+        [[1, decimal.Decimal("1.50")], [2, decimal.Decimal("2.25")]],
+        "k:long,v:decimal(10,2)",
+    )
+    dd = e.to_df(dec)
+    assert e.distinct(dd).count() == 2
+    arr = e.take(dd, 1, presort="k desc").as_array(type_safe=True)
+    assert arr[0][0] == 2
