@@ -81,6 +81,16 @@ void launch_gb_mark_reps(const int64_t*, const int64_t*, const int64_t*,
                          hipStream_t);
 void launch_compact8(const bool*, int64_t, int64_t*, const uint64_t**,
                      uint64_t**, int, hipStream_t);
+void launch_gb_key_stats(const void**, const bool**, const int*, int,
+                         int64_t, int64_t, int64_t*, int32_t*, int, int,
+                         uint64_t*, hipStream_t);
+void launch_pack_cols(const void**, const bool**, const int*, const int64_t*,
+                      const int*, int, int64_t, int64_t*, hipStream_t);
+void launch_unpack_col(const int64_t*, int64_t, int, int64_t, int64_t, int,
+                       int, void*, bool*, hipStream_t);
+void launch_gb_compact(const int64_t*, const int64_t*, const double**,
+                       double**, int, int64_t, int64_t*, int64_t*, int64_t*,
+                       int64_t*, const int64_t*, int64_t*, hipStream_t);
 }
 
 namespace {
@@ -472,6 +482,32 @@ std::vector<at::Tensor> compact_columns(at::Tensor mask,
   return outs;
 }
 
+// Capacity-mode compaction: outputs are allocated at mask length and the
+// caller narrows using the returned cursor (total) — skips the separate
+// mask.sum() reduction + its host sync (profiles/NOTES.md r02).
+std::vector<at::Tensor> compact_columns_cap(at::Tensor mask,
+                                            std::vector<at::Tensor> cols) {
+  check_gpu(mask, "mask");
+  TORCH_CHECK(cols.size() >= 1 && cols.size() <= 8, "1..8 columns");
+  int64_t n = mask.numel();
+  auto cursor = at::zeros({1}, mask.options().dtype(at::kLong));
+  const uint64_t* srcs[8];
+  uint64_t* dsts[8];
+  std::vector<at::Tensor> outs;
+  for (size_t c = 0; c < cols.size(); ++c) {
+    check_gpu(cols[c], "col");
+    TORCH_CHECK(cols[c].element_size() == 8, "8-byte columns only");
+    auto out = at::empty({n}, cols[c].options());
+    srcs[c] = reinterpret_cast<const uint64_t*>(cols[c].data_ptr());
+    dsts[c] = reinterpret_cast<uint64_t*>(out.data_ptr());
+    outs.push_back(out);
+  }
+  launch_compact8(mask.data_ptr<bool>(), n, cursor.data_ptr<int64_t>(),
+                  srcs, dsts, (int)cols.size(), current_stream());
+  outs.push_back(cursor);
+  return outs;
+}
+
 // host mirror of ExprProg in relational.hip (layout must match)
 struct ExprProgHost {
   int n_ops;
@@ -609,6 +645,139 @@ std::vector<at::Tensor> reduce_columns(at::Tensor vals,
   return {out, cnt};
 }
 
+
+// ---- fused group-by key preparation --------------------------------- //
+
+// One launch set + ONE host readback for everything the group-by sizing
+// needs: per-column min/max (2*ncols leading slots, signed order
+// preserved via the s2u bias) and the sampled distinct estimate
+// (d, f1, f2 in the last 3 slots).  Replaces per-column
+// min().item()/max().item() and the torch.unique Chao83 sampling
+// (profiles/NOTES.md r02 host-sync findings).
+at::Tensor gb_key_stats(std::vector<at::Tensor> cols,
+                        std::vector<c10::optional<at::Tensor>> valids,
+                        int64_t nsamples, int64_t tsize, bool do_minmax) {
+  int k = (int)cols.size();
+  TORCH_CHECK(k >= 1 && k <= 8, "1..8 key columns");
+  TORCH_CHECK((tsize & (tsize - 1)) == 0, "tsize must be a power of 2");
+  int64_t n = cols[0].numel();
+  const void* data[8];
+  const bool* valid[8];
+  int dwidth[8];
+  for (int c = 0; c < k; ++c) {
+    check_gpu(cols[c], "key col");
+    int es = (int)cols[c].element_size();
+    TORCH_CHECK(es == 8 || es == 4 || es == 2, "2/4/8-byte key columns");
+    data[c] = cols[c].data_ptr();
+    valid[c] = opt_valid_ptr(valids[c]);
+    dwidth[c] = es;
+  }
+  auto out = at::empty({2 * k + 3}, cols[0].options().dtype(at::kLong));
+  auto slots = at::empty({tsize}, cols[0].options().dtype(at::kLong));
+  auto counts = at::empty({tsize}, cols[0].options().dtype(at::kInt));
+  launch_gb_key_stats(data, valid, dwidth, k, n, nsamples,
+                      slots.data_ptr<int64_t>(), counts.data_ptr<int32_t>(),
+                      (int)tsize, do_minmax ? 1 : 0,
+                      reinterpret_cast<uint64_t*>(out.data_ptr<int64_t>()),
+                      current_stream());
+  return out;
+}
+
+// Pack up to 8 integer key columns into one int64 group key in a single
+// pass (code 0 = NULL; valid values offset by 1 from the column min).
+at::Tensor pack_columns(std::vector<at::Tensor> cols,
+                        std::vector<c10::optional<at::Tensor>> valids,
+                        std::vector<int64_t> mins,
+                        std::vector<int64_t> shifts) {
+  int k = (int)cols.size();
+  TORCH_CHECK(k >= 1 && k <= 8, "1..8 key columns");
+  int64_t n = cols[0].numel();
+  const void* data[8];
+  const bool* valid[8];
+  int dwidth[8];
+  int shift_i[8];
+  for (int c = 0; c < k; ++c) {
+    check_gpu(cols[c], "key col");
+    int es = (int)cols[c].element_size();
+    TORCH_CHECK(es == 8 || es == 4 || es == 2, "2/4/8-byte key columns");
+    data[c] = cols[c].data_ptr();
+    valid[c] = opt_valid_ptr(valids[c]);
+    dwidth[c] = es;
+    shift_i[c] = (int)shifts[c];
+  }
+  auto out = at::empty({n}, cols[0].options().dtype(at::kLong));
+  launch_pack_cols(data, valid, dwidth, mins.data(), shift_i, k, n,
+                   out.data_ptr<int64_t>(), current_stream());
+  return out;
+}
+
+// Unpack one column from packed group keys; returns (data, valid) where
+// valid is an empty tensor when the source column had no nulls.
+std::vector<at::Tensor> unpack_column(at::Tensor packed, int64_t shift,
+                                      int64_t width, int64_t lo,
+                                      int64_t dwidth, bool has_nulls) {
+  check_gpu(packed, "packed");
+  int64_t n = packed.numel();
+  auto dt = dwidth == 8 ? at::kLong : (dwidth == 4 ? at::kInt : at::kShort);
+  auto out = at::empty({n}, packed.options().dtype(dt));
+  at::Tensor valid;
+  if (has_nulls) {
+    valid = at::empty({n}, packed.options().dtype(at::kBool));
+  } else {
+    valid = at::empty({0}, packed.options().dtype(at::kBool));
+  }
+  int64_t mask = width >= 64 ? -1 : ((int64_t(1) << width) - 1);
+  launch_unpack_col(packed.data_ptr<int64_t>(), n, (int)shift, mask, lo,
+                    (int)dwidth, has_nulls ? 1 : 0, out.data_ptr(),
+                    has_nulls ? valid.data_ptr<bool>() : nullptr,
+                    current_stream());
+  return {out, valid};
+}
+
+// Deterministic compaction of the group hash table: drops EMPTY slots
+// from tkeys/gcount/gaggs in slot order.  Returns capacity-sized outputs
+// plus the block-base scan whose last element is the total (single host
+// read; caller narrows).
+std::vector<at::Tensor> gb_compact(at::Tensor tkeys, at::Tensor gcount,
+                                   at::Tensor gaggs,
+                                   c10::optional<at::Tensor> extra) {
+  check_gpu(tkeys, "tkeys");
+  check_gpu(gcount, "gcount");
+  int64_t tsize = tkeys.numel();
+  int n_aggs = gaggs.numel() > 0 ? (int)gaggs.size(0) : 0;
+  TORCH_CHECK(n_aggs <= 6, "at most 6 aggregate columns");
+  const int64_t chunk = 256 * 32;
+  int64_t nblocks = (tsize + chunk - 1) / chunk;
+  auto bcounts = at::empty({nblocks}, tkeys.options());
+  auto bases = at::empty({nblocks + 1}, tkeys.options());
+  auto out_keys = at::empty({tsize}, tkeys.options());
+  auto out_count = at::empty({tsize}, gcount.options());
+  at::Tensor out_aggs = at::empty({n_aggs, tsize}, gaggs.options());
+  const double* asrc[6];
+  double* adst[6];
+  for (int a = 0; a < n_aggs; ++a) {
+    asrc[a] = gaggs[a].data_ptr<double>();
+    adst[a] = out_aggs[a].data_ptr<double>();
+  }
+  at::Tensor out_extra;
+  const int64_t* esrc = nullptr;
+  int64_t* edst = nullptr;
+  if (extra.has_value()) {
+    check_gpu(*extra, "extra");
+    out_extra = at::empty({tsize}, extra->options());
+    esrc = extra->data_ptr<int64_t>();
+    edst = out_extra.data_ptr<int64_t>();
+  } else {
+    out_extra = at::empty({0}, tkeys.options());
+  }
+  launch_gb_compact(tkeys.data_ptr<int64_t>(), gcount.data_ptr<int64_t>(),
+                    asrc, adst, n_aggs, tsize, bcounts.data_ptr<int64_t>(),
+                    bases.data_ptr<int64_t>(), out_keys.data_ptr<int64_t>(),
+                    out_count.data_ptr<int64_t>(), esrc, edst,
+                    current_stream());
+  return {out_keys, out_count, out_aggs, out_extra, bases};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hash_column", &hash_column,
         "combine a column into the running row hash");
@@ -633,6 +802,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused filter-predicate interpreter");
   m.def("gather_columns", &gather_columns,
         "fused multi-column row gather");
+  m.def("gb_key_stats", &gb_key_stats,
+        "fused key min/max + sampled distinct estimate");
+  m.def("pack_columns", &pack_columns, "pack key columns into int64");
+  m.def("unpack_column", &unpack_column, "unpack one key column");
+  m.def("gb_compact", &gb_compact,
+        "deterministic group-table compaction");
+  m.def("compact_columns_cap", &compact_columns_cap,
+        "mask compaction, capacity outputs + cursor");
   m.def("compact_columns", &compact_columns,
         "fused masked compaction of 8-byte columns");
   m.def("join_count", &join_count, "count matches per probe row");

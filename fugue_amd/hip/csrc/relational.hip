@@ -2294,3 +2294,443 @@ void launch_compact8(const bool* mask, int64_t n, int64_t* cursor,
 }
 
 }  // extern "C"
+
+// ------------------------------------------------------------------ //
+// fused key-column min/max + pack (group-by key preparation)          //
+//                                                                     //
+// Replaces the torch residue measured in profiles/NOTES.md r02: per   //
+// key column min().item() + max().item() (2 reduce kernels + 2 host   //
+// syncs EACH) and the 6+ elementwise shift/or/where packing ops.      //
+// One pass computes all columns' min/max (one host read), a second    //
+// packs every column into the int64 group key.                        //
+// ------------------------------------------------------------------ //
+
+struct PackCols {
+  const void* data[8];   // int64/int32/int16 per dwidth
+  const bool* valid[8];  // may be null
+  int dwidth[8];         // element bytes: 8, 4 or 2
+  int64_t mins[8];       // pack phase only
+  int shifts[8];         // pack phase only
+};
+
+__device__ __forceinline__ int64_t pc_load(const PackCols& pc, int c,
+                                           int64_t i) {
+  switch (pc.dwidth[c]) {
+    case 8:
+      return ((const int64_t*)pc.data[c])[i];
+    case 4:
+      return (int64_t)((const int32_t*)pc.data[c])[i];
+    default:
+      return (int64_t)((const int16_t*)pc.data[c])[i];
+  }
+}
+
+// signed→unsigned order-preserving map so atomicMin/Max work on u64
+__device__ __forceinline__ uint64_t s2u(int64_t v) {
+  return (uint64_t)v ^ 0x8000000000000000ULL;
+}
+
+__global__ __launch_bounds__(BLOCK) void minmax_init_kernel(
+    uint64_t* __restrict__ out, int ncols) {
+  int i = threadIdx.x;
+  if (i < ncols) {
+    out[2 * i] = ~0ULL;      // min slot: u64 max
+    out[2 * i + 1] = 0ULL;   // max slot: u64 min
+  }
+}
+
+template <int NC>
+__global__ __launch_bounds__(BLOCK) void minmax_cols_kernel(
+    PackCols pc, int64_t n, uint64_t* __restrict__ out) {
+  uint64_t lmin[NC], lmax[NC];
+#pragma unroll
+  for (int c = 0; c < NC; ++c) {
+    lmin[c] = ~0ULL;
+    lmax[c] = 0ULL;
+  }
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      uint64_t u = s2u(pc_load(pc, c, i));
+      lmin[c] = u < lmin[c] ? u : lmin[c];
+      lmax[c] = u > lmax[c] ? u : lmax[c];
+    }
+  }
+  // wave reduce then one atomic per wave per col
+#pragma unroll
+  for (int c = 0; c < NC; ++c) {
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+      uint64_t omin = __shfl_down(lmin[c], off, WAVE);
+      uint64_t omax = __shfl_down(lmax[c], off, WAVE);
+      lmin[c] = omin < lmin[c] ? omin : lmin[c];
+      lmax[c] = omax > lmax[c] ? omax : lmax[c];
+    }
+    if ((threadIdx.x & (WAVE - 1)) == 0) {
+      atomicMin((unsigned long long*)&out[2 * c], lmin[c]);
+      atomicMax((unsigned long long*)&out[2 * c + 1], lmax[c]);
+    }
+  }
+}
+
+template <int NC>
+__global__ __launch_bounds__(BLOCK) void pack_cols_kernel(
+    PackCols pc, int64_t n, int64_t* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t packed = 0;
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      int64_t code = pc_load(pc, c, i) - pc.mins[c] + 1;  // 0 = NULL
+      if (pc.valid[c] != nullptr && !pc.valid[c][i]) code = 0;
+      packed |= code << pc.shifts[c];
+    }
+    out[i] = packed;
+  }
+}
+
+// unpack one column out of the packed group key (code 0 = NULL)
+__global__ __launch_bounds__(BLOCK) void unpack_col_kernel(
+    const int64_t* __restrict__ packed, int64_t n, int shift, int64_t mask,
+    int64_t lo, int dwidth, int has_nulls, void* __restrict__ out,
+    bool* __restrict__ valid) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t code = (packed[i] >> shift) & mask;
+    int64_t v = code - 1 + lo;
+    if (has_nulls) valid[i] = code != 0;
+    switch (dwidth) {
+      case 8:
+        ((int64_t*)out)[i] = v;
+        break;
+      case 4:
+        ((int32_t*)out)[i] = (int32_t)v;
+        break;
+      default:
+        ((int16_t*)out)[i] = (int16_t)v;
+        break;
+    }
+  }
+}
+
+
+static void launch_minmax_dispatch(int ncols, dim3 g, dim3 b,
+                                   hipStream_t stream, PackCols pc,
+                                   int64_t n, uint64_t* out) {
+  switch (ncols) {
+    case 1: hipLaunchKernelGGL(minmax_cols_kernel<1>, g, b, 0, stream, pc, n, out); break;
+    case 2: hipLaunchKernelGGL(minmax_cols_kernel<2>, g, b, 0, stream, pc, n, out); break;
+    case 3: hipLaunchKernelGGL(minmax_cols_kernel<3>, g, b, 0, stream, pc, n, out); break;
+    case 4: hipLaunchKernelGGL(minmax_cols_kernel<4>, g, b, 0, stream, pc, n, out); break;
+    case 5: hipLaunchKernelGGL(minmax_cols_kernel<5>, g, b, 0, stream, pc, n, out); break;
+    case 6: hipLaunchKernelGGL(minmax_cols_kernel<6>, g, b, 0, stream, pc, n, out); break;
+    case 7: hipLaunchKernelGGL(minmax_cols_kernel<7>, g, b, 0, stream, pc, n, out); break;
+    default: hipLaunchKernelGGL(minmax_cols_kernel<8>, g, b, 0, stream, pc, n, out); break;
+  }
+}
+
+static void launch_pack_dispatch(int ncols, dim3 g, dim3 b,
+                                 hipStream_t stream, PackCols pc, int64_t n,
+                                 int64_t* out) {
+  switch (ncols) {
+    case 1: hipLaunchKernelGGL(pack_cols_kernel<1>, g, b, 0, stream, pc, n, out); break;
+    case 2: hipLaunchKernelGGL(pack_cols_kernel<2>, g, b, 0, stream, pc, n, out); break;
+    case 3: hipLaunchKernelGGL(pack_cols_kernel<3>, g, b, 0, stream, pc, n, out); break;
+    case 4: hipLaunchKernelGGL(pack_cols_kernel<4>, g, b, 0, stream, pc, n, out); break;
+    case 5: hipLaunchKernelGGL(pack_cols_kernel<5>, g, b, 0, stream, pc, n, out); break;
+    case 6: hipLaunchKernelGGL(pack_cols_kernel<6>, g, b, 0, stream, pc, n, out); break;
+    case 7: hipLaunchKernelGGL(pack_cols_kernel<7>, g, b, 0, stream, pc, n, out); break;
+    default: hipLaunchKernelGGL(pack_cols_kernel<8>, g, b, 0, stream, pc, n, out); break;
+  }
+}
+
+extern "C" {
+
+void launch_minmax_cols(const void** data, const bool** valid,
+                        const int* dwidth, int ncols, int64_t n,
+                        uint64_t* out, hipStream_t stream) {
+  PackCols pc;
+  memset(&pc, 0, sizeof(pc));
+  for (int c = 0; c < ncols; ++c) {
+    pc.data[c] = data[c];
+    pc.valid[c] = valid[c];
+    pc.dwidth[c] = dwidth[c];
+  }
+  hipLaunchKernelGGL(minmax_init_kernel, dim3(1), dim3(BLOCK), 0, stream,
+                     out, ncols);
+  launch_minmax_dispatch(ncols, dim3(grid_for(n, 4)), dim3(BLOCK), stream,
+                         pc, n, out);
+}
+
+void launch_pack_cols(const void** data, const bool** valid,
+                      const int* dwidth, const int64_t* mins,
+                      const int* shifts, int ncols, int64_t n, int64_t* out,
+                      hipStream_t stream) {
+  PackCols pc;
+  memset(&pc, 0, sizeof(pc));
+  for (int c = 0; c < ncols; ++c) {
+    pc.data[c] = data[c];
+    pc.valid[c] = valid[c];
+    pc.dwidth[c] = dwidth[c];
+    pc.mins[c] = mins[c];
+    pc.shifts[c] = shifts[c];
+  }
+  launch_pack_dispatch(ncols, dim3(grid_for(n, 2)), dim3(BLOCK), stream, pc,
+                       n, out);
+}
+
+void launch_unpack_col(const int64_t* packed, int64_t n, int shift,
+                       int64_t mask, int64_t lo, int dwidth, int has_nulls,
+                       void* out, bool* valid, hipStream_t stream) {
+  hipLaunchKernelGGL(unpack_col_kernel, dim3(grid_for(n, 2)), dim3(BLOCK), 0,
+                     stream, packed, n, shift, mask, lo, dwidth, has_nulls,
+                     out, valid);
+}
+
+}  // extern "C"
+
+// ------------------------------------------------------------------ //
+// sampled distinct-count estimator (Chao83 inputs d/f1/f2)            //
+//                                                                     //
+// Replaces the torch.unique(sample) path (rocPRIM merge-sort + multi  //
+// sync) with one hash-insert pass + one table scan; the host reads a  //
+// single 3-element tensor.                                            //
+// ------------------------------------------------------------------ //
+
+#define DS_EMPTY 0x8000000000000001LL  // improbable sentinel key
+
+__device__ __forceinline__ uint64_t pc_row_hash(const PackCols& pc,
+                                                int ncols, int64_t i) {
+  uint64_t h = 0x9e3779b97f4a7c15ULL;
+  for (int c = 0; c < ncols; ++c) {
+    uint64_t v = (uint64_t)pc_load(pc, c, i);
+    if (pc.valid[c] != nullptr && !pc.valid[c][i]) v = 0xdeadULL;
+    h = hash_combine(h, v);
+  }
+  return h;
+}
+
+__global__ __launch_bounds__(BLOCK) void distinct_init_kernel(
+    int64_t* __restrict__ slots, int32_t* __restrict__ counts, int tsize,
+    int64_t* __restrict__ out3) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < 3) out3[i] = 0;
+  for (; i < tsize; i += gridDim.x * blockDim.x) {
+    slots[i] = DS_EMPTY;
+    counts[i] = 0;
+  }
+}
+
+__global__ __launch_bounds__(BLOCK) void distinct_insert_kernel(
+    PackCols pc, int ncols, int64_t nsamples, int64_t stride,
+    int64_t* __restrict__ slots, int32_t* __restrict__ counts, int tmask) {
+  for (int64_t s = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       s < nsamples; s += (int64_t)gridDim.x * blockDim.x) {
+    int64_t k = (int64_t)pc_row_hash(pc, ncols, s * stride);
+    if (k == DS_EMPTY) k ^= 1;  // estimator: sentinel remap is harmless
+    uint32_t h = (uint32_t)mix64((uint64_t)k) & tmask;
+    for (;;) {
+      int64_t prev = (int64_t)atomicCAS((unsigned long long*)&slots[h],
+                                        (unsigned long long)DS_EMPTY,
+                                        (unsigned long long)k);
+      if (prev == DS_EMPTY || prev == k) {
+        atomicAdd(&counts[h], 1);
+        break;
+      }
+      h = (h + 1) & tmask;
+    }
+  }
+}
+
+__global__ __launch_bounds__(BLOCK) void distinct_stats_kernel(
+    const int32_t* __restrict__ counts, int tsize,
+    int64_t* __restrict__ out3) {
+  int d = 0, f1 = 0, f2 = 0;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < tsize;
+       i += gridDim.x * blockDim.x) {
+    int c = counts[i];
+    if (c > 0) ++d;
+    if (c == 1) ++f1;
+    if (c == 2) ++f2;
+  }
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    d += __shfl_down(d, off, WAVE);
+    f1 += __shfl_down(f1, off, WAVE);
+    f2 += __shfl_down(f2, off, WAVE);
+  }
+  if ((threadIdx.x & (WAVE - 1)) == 0) {
+    atomicAdd((unsigned long long*)&out3[0], (unsigned long long)d);
+    atomicAdd((unsigned long long*)&out3[1], (unsigned long long)f1);
+    atomicAdd((unsigned long long*)&out3[2], (unsigned long long)f2);
+  }
+}
+
+extern "C" {
+
+// One fused launch set for group-by key preparation: per-column min/max
+// (first 2*ncols slots of `out`) and a sampled distinct-row estimate
+// (d, f1, f2 in the last 3 slots).  The host reads `out` ONCE — this
+// replaces 2 reduce kernels + 2 syncs per key column plus a
+// torch.unique (rocPRIM sort) sampling pass (profiles/NOTES.md r02).
+void launch_gb_key_stats(const void** data, const bool** valid,
+                         const int* dwidth, int ncols, int64_t n,
+                         int64_t nsamples, int64_t* slots, int32_t* counts,
+                         int tsize, int do_minmax, uint64_t* out,
+                         hipStream_t stream) {
+  PackCols pc;
+  memset(&pc, 0, sizeof(pc));
+  for (int c = 0; c < ncols; ++c) {
+    pc.data[c] = data[c];
+    pc.valid[c] = valid[c];
+    pc.dwidth[c] = dwidth[c];
+  }
+  int64_t* out3 = (int64_t*)out + 2 * ncols;
+  if (do_minmax) {
+    hipLaunchKernelGGL(minmax_init_kernel, dim3(1), dim3(BLOCK), 0, stream,
+                       out, ncols);
+    launch_minmax_dispatch(ncols, dim3(grid_for(n, 4)), dim3(BLOCK), stream,
+                           pc, n, out);
+  }
+  int64_t stride = n / nsamples;
+  if (stride < 1) stride = 1;
+  int64_t eff = n / stride;
+  if (eff > nsamples) eff = nsamples;
+  hipLaunchKernelGGL(distinct_init_kernel, dim3(grid_for(tsize)), dim3(BLOCK),
+                     0, stream, slots, counts, tsize, out3);
+  hipLaunchKernelGGL(distinct_insert_kernel, dim3(grid_for(eff)), dim3(BLOCK),
+                     0, stream, pc, ncols, eff, stride, slots, counts,
+                     tsize - 1);
+  hipLaunchKernelGGL(distinct_stats_kernel, dim3(grid_for(tsize)),
+                     dim3(BLOCK), 0, stream, counts, tsize, out3);
+}
+
+}  // extern "C"
+
+// ------------------------------------------------------------------ //
+// deterministic group-table compaction                                //
+//                                                                     //
+// Replaces occupied = (tkeys != EMPTY).nonzero() + per-column          //
+// index_select with a 3-kernel own pipeline (count / scan / emit) that //
+// preserves slot order (deterministic output) and moves each byte      //
+// once.  total lands in bases[nblocks] for a single host read.         //
+// ------------------------------------------------------------------ //
+
+#define GBC_CHUNK (BLOCK * 32)
+
+__global__ __launch_bounds__(BLOCK) void gbc_count_kernel(
+    const int64_t* __restrict__ tkeys, int64_t tsize,
+    int64_t* __restrict__ bcounts) {
+  __shared__ int lcount;
+  int64_t start = (int64_t)blockIdx.x * GBC_CHUNK;
+  if (threadIdx.x == 0) lcount = 0;
+  __syncthreads();
+  int64_t end = start + GBC_CHUNK < tsize ? start + GBC_CHUNK : tsize;
+  int local = 0;
+  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
+    if (tkeys[i] != GB_EMPTY) ++local;
+  atomicAdd(&lcount, local);
+  __syncthreads();
+  if (threadIdx.x == 0) bcounts[blockIdx.x] = lcount;
+}
+
+// single-block exclusive scan of bcounts (nblocks <= 65536); writes the
+// total into bases[nblocks]
+__global__ __launch_bounds__(BLOCK) void gbc_scan_kernel(
+    const int64_t* __restrict__ bcounts, int nblocks,
+    int64_t* __restrict__ bases) {
+  __shared__ int64_t carry;
+  __shared__ int64_t tile[BLOCK];
+  if (threadIdx.x == 0) carry = 0;
+  __syncthreads();
+  for (int t0 = 0; t0 < nblocks; t0 += BLOCK) {
+    int i = t0 + threadIdx.x;
+    int64_t v = i < nblocks ? bcounts[i] : 0;
+    tile[threadIdx.x] = v;
+    __syncthreads();
+    // inclusive scan in LDS
+    for (int off = 1; off < BLOCK; off <<= 1) {
+      int64_t add = threadIdx.x >= off ? tile[threadIdx.x - off] : 0;
+      __syncthreads();
+      tile[threadIdx.x] += add;
+      __syncthreads();
+    }
+    if (i < nblocks) bases[i] = carry + tile[threadIdx.x] - v;  // exclusive
+    __syncthreads();
+    if (threadIdx.x == 0) carry += tile[BLOCK - 1];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) bases[nblocks] = carry;  // total
+}
+
+struct GbcAggs {
+  const double* src[6];
+  double* dst[6];
+};
+
+__global__ __launch_bounds__(BLOCK) void gbc_emit_kernel(
+    const int64_t* __restrict__ tkeys, const int64_t* __restrict__ gcount,
+    GbcAggs ag, int n_aggs, int64_t tsize,
+    const int64_t* __restrict__ bases, int64_t* __restrict__ out_keys,
+    int64_t* __restrict__ out_count, const int64_t* __restrict__ extra_src,
+    int64_t* __restrict__ extra_dst) {
+  __shared__ int64_t wbase[BLOCK / WAVE];
+  __shared__ int64_t sbase;
+  int64_t start = (int64_t)blockIdx.x * GBC_CHUNK;
+  int64_t end = start + GBC_CHUNK < tsize ? start + GBC_CHUNK : tsize;
+  if (threadIdx.x == 0) sbase = bases[blockIdx.x];
+  int wave = threadIdx.x / WAVE;
+  int lane = threadIdx.x % WAVE;
+  __syncthreads();
+  for (int64_t i0 = start; i0 < end; i0 += blockDim.x) {
+    int64_t i = i0 + threadIdx.x;
+    bool occ = i < end && tkeys[i] != GB_EMPTY;
+    uint64_t mask = __ballot(occ);
+    int rank = __popcll(mask & ((1ULL << lane) - 1ULL));
+    if (lane == 0) wbase[wave] = __popcll(mask);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      int64_t acc = sbase;
+      for (int w = 0; w < BLOCK / WAVE; ++w) {
+        int64_t c = wbase[w];
+        wbase[w] = acc;
+        acc += c;
+      }
+      sbase = acc;
+    }
+    __syncthreads();
+    if (occ) {
+      int64_t pos = wbase[wave] + rank;
+      out_keys[pos] = tkeys[i];
+      out_count[pos] = gcount[i];
+      if (extra_src != nullptr) extra_dst[pos] = extra_src[i];
+      for (int a = 0; a < n_aggs; ++a) ag.dst[a][pos] = ag.src[a][i];
+    }
+    __syncthreads();
+  }
+}
+
+extern "C" {
+
+void launch_gb_compact(const int64_t* tkeys, const int64_t* gcount,
+                       const double** agg_src, double** agg_dst, int n_aggs,
+                       int64_t tsize, int64_t* bcounts, int64_t* bases,
+                       int64_t* out_keys, int64_t* out_count,
+                       const int64_t* extra_src, int64_t* extra_dst,
+                       hipStream_t stream) {
+  int nblocks = (int)((tsize + GBC_CHUNK - 1) / GBC_CHUNK);
+  hipLaunchKernelGGL(gbc_count_kernel, dim3(nblocks), dim3(BLOCK), 0, stream,
+                     tkeys, tsize, bcounts);
+  hipLaunchKernelGGL(gbc_scan_kernel, dim3(1), dim3(BLOCK), 0, stream,
+                     bcounts, nblocks, bases);
+  GbcAggs ag;
+  memset(&ag, 0, sizeof(ag));
+  for (int a = 0; a < n_aggs; ++a) {
+    ag.src[a] = agg_src[a];
+    ag.dst[a] = agg_dst[a];
+  }
+  hipLaunchKernelGGL(gbc_emit_kernel, dim3(nblocks), dim3(BLOCK), 0, stream,
+                     tkeys, gcount, ag, n_aggs, tsize, bases, out_keys,
+                     out_count, extra_src, extra_dst);
+}
+
+}  // extern "C"

@@ -930,7 +930,8 @@ class HipExecutionEngine(ExecutionEngine):
             else:
                 pi, bi = dops.hash_join_indices(k1, k2, mode, h21, h22)
             res = self._emit_join_output(
-                d1v, d2v, pi, bi, keys, output_schema, probe_is_left=True
+                d1v, d2v, pi, bi, keys, output_schema, probe_is_left=True,
+                may_have_unmatched=(mode == "left"),
             )
             if how in ("left_outer", "anti") and d1n.count() > 0:
                 nulls = self._pad_right_nulls(d1n, output_schema)
@@ -997,10 +998,18 @@ class HipExecutionEngine(ExecutionEngine):
         keys: List[str],
         output_schema: Schema,
         probe_is_left: bool,
+        may_have_unmatched: bool = True,
     ) -> HipDataFrame:
-        has_null_build = bool((bi < 0).any().item()) if bi.numel() > 0 else False
-        bi_safe = torch.clamp(bi, min=0)
-        build_invalid = (bi < 0) if has_null_build else None
+        # inner/semi/anti probes never carry -1 build indices; skipping
+        # the (bi < 0).any().item() probe avoids a host sync per join.
+        # For outer modes the mask is built unconditionally (an all-true
+        # validity mask is semantically equivalent, no sync needed).
+        if may_have_unmatched and bi.numel() > 0:
+            build_invalid = bi < 0
+            bi_safe = torch.clamp(bi, min=0)
+        else:
+            build_invalid = None
+            bi_safe = bi
         cols: Dict[str, DeviceColumn] = {}
         probe_gather = probe.gather_rows(pi)
         need_build = any(
@@ -1456,14 +1465,14 @@ class HipExecutionEngine(ExecutionEngine):
         ):
             from fugue_amd.hip.ext import get_ext
 
-            out_n = int(mask.sum().item())
             names = list(d.columns_map.keys())
-            outs = get_ext().compact_columns(
-                mask, [d.col(nm).data for nm in names], out_n
+            outs = get_ext().compact_columns_cap(
+                mask, [d.col(nm).data for nm in names]
             )
+            out_n = int(outs[-1].item())
             cols = {
-                nm: DeviceColumn(t, None, d.col(nm).pa_type)
-                for nm, t in zip(names, outs)
+                nm: DeviceColumn(t.narrow(0, 0, out_n), None, d.col(nm).pa_type)
+                for nm, t in zip(names, outs[:-1])
             }
             return HipDataFrame.from_columns(cols, d.schema, self._device)
         return d.gather_rows(mask.nonzero(as_tuple=True)[0])

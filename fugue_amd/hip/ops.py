@@ -8,6 +8,8 @@ SURVEY.md §2.3.
 """
 from typing import Any, Dict, List, Optional, Sequence, Tuple
 
+import weakref
+
 import numpy as np
 import pyarrow as pa
 import torch
@@ -196,6 +198,69 @@ def rand_buckets(n: int, num_buckets: int, seed: Optional[int], device) -> torch
     )
 
 
+
+
+# ---------------------------------------------------------------- #
+# fused group-key statistics (own kernels, single host readback)    #
+# ---------------------------------------------------------------- #
+
+_KEY_STATS_MEMO: "Dict[Tuple[int, ...], Tuple[Any, Any]]" = {}
+_KEY_STATS_MEMO_CAP = 128
+
+
+def _key_stats_device(
+    datas: "List[torch.Tensor]",
+    valids: "List[Optional[torch.Tensor]]",
+    with_minmax: bool,
+) -> Tuple[Optional[List[int]], Optional[List[int]], int]:
+    """(mins, maxs, estimated_distinct) for device key columns via ONE
+    launch set + ONE host readback (``gb_key_stats``), memoized by the
+    column tensors' identities (frames are immutable, so an identical
+    tensor tuple ⇒ identical stats; repeated plans skip the kernels AND
+    the device sync entirely).  Replaces per-column min/max reductions
+    + torch.unique Chao sampling (profiles/NOTES.md r02)."""
+    n = int(datas[0].numel())
+    key = tuple(id(t) for t in datas) + (n, with_minmax)
+    hit = _KEY_STATS_MEMO.get(key)
+    if hit is not None:
+        refs, value = hit
+        if all(r() is t for r, t in zip(refs, datas)):
+            return value
+        del _KEY_STATS_MEMO[key]
+    ext = get_ext()
+    nsamples = min(65536, n)
+    st = ext.gb_key_stats(datas, valids, nsamples, 131072, with_minmax)
+    st = st.cpu().tolist()  # the single host readback
+    k = len(datas)
+    BIAS = 1 << 63
+    mins = maxs = None
+    if with_minmax:
+        u = [x & ((1 << 64) - 1) for x in st[: 2 * k]]
+        mins = [(v ^ BIAS) - BIAS for v in u[0::2]]
+        maxs = [(v ^ BIAS) - BIAS for v in u[1::2]]
+    d, f1, f2 = st[2 * k], st[2 * k + 1], st[2 * k + 2]
+    if n > nsamples:
+        est = d + (f1 * f1) // max(2 * f2, 1)
+        est = max(d, min(n, est))
+    else:
+        est = max(1, d)
+    value = (mins, maxs, est)
+    if len(_KEY_STATS_MEMO) >= _KEY_STATS_MEMO_CAP:
+        _KEY_STATS_MEMO.pop(next(iter(_KEY_STATS_MEMO)))
+    _KEY_STATS_MEMO[key] = (tuple(weakref.ref(t) for t in datas), value)
+    return value
+
+
+def _device_packable(key_cols: "Sequence[DeviceColumn]") -> bool:
+    return all(
+        (not isinstance(c, StringDeviceColumn))
+        and c.data.is_cuda
+        and c.data.element_size() in (2, 4, 8)
+        and c.data.is_contiguous()
+        for c in key_cols
+    )
+
+
 def pack_keys(
     key_cols: Sequence[DeviceColumn],
     mins: Optional[List[int]] = None,
@@ -215,12 +280,47 @@ def pack_keys(
         and mins is None
     ):
         return key_cols[0].data, None
-    datas = []
-    comp_mins: List[int] = []
-    comp_widths: List[int] = []
-    for i, c in enumerate(key_cols):
+    n = len(key_cols[0])
+    for c in key_cols:
         if isinstance(c, StringDeviceColumn):
             raise NotImplementedError("string group keys not yet on device")
+    # device fused path: one min/max pass + one pack pass (own kernels,
+    # single host readback; memoized per tensor identity)
+    if n > 0 and _device_packable(key_cols):
+        kd = [c.data for c in key_cols]
+        kv = [c.valid for c in key_cols]
+        est = None
+        if mins is None:
+            comp_mins, comp_maxs, est = _key_stats_device(kd, kv, True)
+            comp_widths = [
+                max(1, int(np.ceil(np.log2(max(2, hi - lo + 2)))))
+                for lo, hi in zip(comp_mins, comp_maxs)
+            ]
+        else:
+            comp_mins = list(mins)
+            comp_widths = list(widths)
+        total_bits = sum(comp_widths)
+        if total_bits <= 63:
+            shifts: List[int] = []
+            shift = 0
+            for w in reversed(comp_widths):
+                shifts.append(shift)
+                shift += w
+            shifts.reverse()
+            packed = get_ext().pack_columns(kd, kv, comp_mins, shifts)
+            return packed, dict(
+                mode="pack",
+                mins=comp_mins,
+                widths=comp_widths,
+                nulls=[c.valid is not None for c in key_cols],
+                device_pack=True,
+                est=est,
+            )
+        # >63 bits: fall through to the torch dense re-encode below
+    datas = []
+    comp_mins = []
+    comp_widths = []
+    for i, c in enumerate(key_cols):
         d = c.data.to(torch.int64)
         if mins is None:
             lo = int(d.min().item()) if d.numel() > 0 else 0
@@ -244,7 +344,12 @@ def pack_keys(
         for d, w in zip(reversed(datas), reversed(comp_widths)):
             packed = packed | (d << shift)
             shift += w
-        return packed, dict(mode="pack", mins=comp_mins, widths=comp_widths)
+        return packed, dict(
+            mode="pack",
+            mins=comp_mins,
+            widths=comp_widths,
+            nulls=[c.valid is not None for c in key_cols],
+        )
     # fallback: dense re-encode via torch.unique on stacked keys
     stacked = torch.stack(datas, dim=1)
     uniq, inverse = torch.unique(stacked, dim=0, return_inverse=True)
@@ -263,14 +368,36 @@ def unpack_keys(
         return [DeviceColumn(packed, None, key_cols[0].pa_type)]
     res: List[DeviceColumn] = []
     if meta["mode"] == "pack":
+        nulls = meta.get("nulls")
         shift = sum(meta["widths"])
+        if packed.is_cuda and all(
+            c.data.element_size() in (2, 4, 8) for c in key_cols
+        ):
+            ext = get_ext()
+            for i, (c, lo, w) in enumerate(
+                zip(key_cols, meta["mins"], meta["widths"])
+            ):
+                shift -= w
+                has_nulls = nulls[i] if nulls is not None else True
+                data, valid = ext.unpack_column(
+                    packed, shift, w, lo, c.data.element_size(), has_nulls
+                )
+                res.append(
+                    DeviceColumn(
+                        data, valid if has_nulls else None, c.pa_type
+                    )
+                )
+            return res
         for i, (c, lo, w) in enumerate(
             zip(key_cols, meta["mins"], meta["widths"])
         ):
             shift -= w
             code = (packed >> shift) & ((1 << w) - 1)
-            valid = code != 0
             data = (code - 1 + lo).to(c.data.dtype)
+            if nulls is not None and not nulls[i]:
+                res.append(DeviceColumn(data, None, c.pa_type))
+                continue
+            valid = code != 0
             res.append(
                 DeviceColumn(
                     data, None if bool(valid.all().item()) else valid, c.pa_type
@@ -353,20 +480,27 @@ def groupby_aggregate(
     if _is_cpu(packed):
         return _groupby_aggregate_cpu(packed, aggs, df, meta)
     ext = get_ext()
+    key_lo = key_hi = None
     if expected_groups is None:
         # sample-based distinct-count estimate (Chao83: D ≈ d + f1²/(2·f2),
         # robust when the sample is mostly singletons — a linear scale-up
-        # would estimate ~n for any high-ish cardinality)
-        if n > 65536:
-            sample = packed[:: max(1, n // 65536)]
-            _, counts = torch.unique(sample, return_counts=True)
-            d = int(counts.numel())
-            f1 = int((counts == 1).sum().item())
-            f2 = int((counts == 2).sum().item())
-            est = d + (f1 * f1) // max(2 * f2, 1)
-            expected_groups = max(d, min(n, est))
+        # would estimate ~n for any high-ish cardinality).  Own kernel +
+        # single host readback; for meta-None keys the same readback
+        # carries min/max so the int32-narrowing decision needs no
+        # speculative overflow check.
+        if meta is None and n > 0:
+            key_lo_l, key_hi_l, expected_groups = _key_stats_device(
+                [packed], [None], True
+            )
+            key_lo, key_hi = key_lo_l[0], key_hi_l[0]
+        elif n > 0:
+            expected_groups = (meta or {}).get("est")
+            if expected_groups is None:
+                _, _, expected_groups = _key_stats_device(
+                    [packed], [None], False
+                )
         else:
-            expected_groups = max(1, n)
+            expected_groups = 1
     tsize = _next_pow2(max(16, int(expected_groups * 2)))
     sum_count_only = all(op in (AGG_SUM, AGG_COUNT) for _, op, _ in aggs)
     if expected_groups > 100_000 and valids is None and sum_count_only:
@@ -398,6 +532,8 @@ def groupby_aggregate(
         if int(_os.environ.get("FUGUE_GB_NARROW", "1")):
             if meta is not None:
                 narrow = 1 if sum(meta["widths"]) <= 31 else 0
+            elif key_lo is not None:
+                narrow = 1 if key_lo >= 0 and key_hi < (1 << 31) else 0
             else:
                 narrow = -1  # speculative: overflow flag checked below
         tkeys, gaggs, gcount, ovf = ext.gb_aggregate_partitioned(
@@ -415,12 +551,20 @@ def groupby_aggregate(
         tkeys, gaggs, gcount = ext.gb_aggregate(
             packed, vals, valids, ops, tsize, use_lds
         )
-    occupied = (tkeys != GB_EMPTY).nonzero(as_tuple=True)[0]
-    out_keys = tkeys.index_select(0, occupied)
-    out_count = gcount.index_select(0, occupied)
+    # deterministic own-kernel compaction of the group table (replaces
+    # nonzero + per-column index_select); ONE host read for the total
+    ck, cc, ca, _ce, bases = ext.gb_compact(
+        tkeys, gcount,
+        gaggs if len(aggs) > 0
+        else torch.empty((0, 0), dtype=torch.float64, device=device),
+        None,
+    )
+    total = int(bases[-1].item())
+    out_keys = ck.narrow(0, 0, total)
+    out_count = cc.narrow(0, 0, total)
     out_aggs: Dict[str, torch.Tensor] = {}
     for i, (_, op, oname) in enumerate(aggs):
-        out_aggs[oname] = gaggs[i].index_select(0, occupied)
+        out_aggs[oname] = ca[i].narrow(0, 0, total)
     return out_keys, out_aggs, out_count, meta
 
 
@@ -648,17 +792,8 @@ def groupby_aggregate_hashed(
         vals = torch.zeros((1, n), dtype=torch.float64, device=device)
         valids = None
         ops = torch.tensor([AGG_COUNT], dtype=torch.int32, device=device)
-    # distinct estimate over h1
-    if n > 65536:
-        sample = h1[:: max(1, n // 65536)]
-        _, cnts = torch.unique(sample, return_counts=True)
-        d = int(cnts.numel())
-        f1 = int((cnts == 1).sum().item())
-        f2 = int((cnts == 2).sum().item())
-        est = d + (f1 * f1) // max(2 * f2, 1)
-        expected = max(d, min(n, est))
-    else:
-        expected = max(1, n)
+    # distinct estimate over h1 (own kernel, one readback, memoized)
+    _, _, expected = _key_stats_device([h1], [None], False)
     tsize = _next_pow2(max(16, int(expected * 2)))
     tkeys, gaggs, gcount = ext.gb_aggregate(
         h1, vals, valids, ops, tsize, expected <= 100_000
@@ -666,12 +801,13 @@ def groupby_aggregate_hashed(
     rep, th2, conflict = ext.gb_mark_reps(h1, h2, tkeys, tsize)
     if int(conflict.item()) > 0:
         raise HashCollisionError("h1 collision on string keys")
-    occupied = (tkeys != GB_EMPTY).nonzero(as_tuple=True)[0]
-    reps = rep.index_select(0, occupied)
-    counts = gcount.index_select(0, occupied)
+    ck, cc, ca, ce, bases = ext.gb_compact(tkeys, gcount, gaggs, rep)
+    total = int(bases[-1].item())
+    reps = ce.narrow(0, 0, total)
+    counts = cc.narrow(0, 0, total)
     out_aggs: Dict[str, torch.Tensor] = {}
     for i, (_, op, oname) in enumerate(aggs):
-        out_aggs[oname] = gaggs[i].index_select(0, occupied)
+        out_aggs[oname] = ca[i].narrow(0, 0, total)
     return reps, out_aggs, counts
 
 
@@ -692,15 +828,7 @@ def distinct_reps(h1: torch.Tensor, h2: torch.Tensor) -> torch.Tensor:
         return torch.from_numpy(g["idx"].first().to_numpy())
     ext = get_ext()
     device = h1.device
-    if n > 65536:
-        sample = h1[:: max(1, n // 65536)]
-        _, cnts = torch.unique(sample, return_counts=True)
-        d = int(cnts.numel())
-        f1 = int((cnts == 1).sum().item())
-        f2 = int((cnts == 2).sum().item())
-        expected = max(d, min(n, d + (f1 * f1) // max(2 * f2, 1)))
-    else:
-        expected = max(1, n)
+    _, _, expected = _key_stats_device([h1], [None], False)
     tsize = _next_pow2(max(16, int(expected * 2)))
     vals = torch.zeros((1, max(n, 1)), dtype=torch.float64, device=device)
     ops = torch.tensor([AGG_COUNT], dtype=torch.int32, device=device)
@@ -710,8 +838,11 @@ def distinct_reps(h1: torch.Tensor, h2: torch.Tensor) -> torch.Tensor:
     rep, _th2, conflict = ext.gb_mark_reps(h1, h2, tkeys, tsize)
     if int(conflict.item()) > 0:
         raise HashCollisionError("h1 collision in distinct")
-    occupied = (tkeys != GB_EMPTY).nonzero(as_tuple=True)[0]
-    return rep.index_select(0, occupied)
+    ck, cc, _ca, _ce, bases = ext.gb_compact(
+        tkeys, rep,
+        torch.empty((0, 0), dtype=torch.float64, device=device), None,
+    )
+    return cc.narrow(0, 0, int(bases[-1].item()))
 
 
 def hash_join_indices(
@@ -741,32 +872,41 @@ def hash_join_indices(
     mode = {"inner": 0, "left": 1, "semi": 2, "anti": 3}[how]
     import os as _os_j
 
-    if (
-        _os_j.environ.get("FUGUE_JOIN_UNIQUE", "1") != "0"
-        and int(dup.item()) == 0
-    ):
+    if _os_j.environ.get("FUGUE_JOIN_UNIQUE", "1") != "0":
         # unique build keys (≤1 match per probe): ONE chain walk writes
         # the match index positionally; inner/semi/anti then compact
         # with the block-scan compaction kernel (no global-cursor
-        # contention, no second walk)
+        # contention, no second walk).  The walk is SPECULATIVE: it runs
+        # before the duplicate flag is read so the dup check, the
+        # compaction total and the emit all resolve in ONE host sync
+        # (dup build keys are rare — dim joins and groupby outputs are
+        # unique — and merely discard this walk).
         np_ = int(probe_keys.numel())
         out_p, out_b, _cur = ext.join_emit_unique(
             probe_keys, build_keys, probe_h2, build_h2, heads, nxt, 1
         )
         if how == "left":
-            return out_p, out_b
-        matched = out_b >= 0
-        if how == "anti":
-            matched = ~matched
-        total = int(matched.sum().item())
-        arange = torch.arange(
-            np_, dtype=torch.int64, device=probe_keys.device
-        )
-        if how == "anti":
-            (pi,) = ext.compact_columns(matched, [arange], total)
-            return pi, torch.full_like(pi, -1)
-        pi, bi = ext.compact_columns(matched, [arange, out_b], total)
-        return pi, bi
+            if int(dup.item()) == 0:
+                return out_p, out_b
+        else:
+            matched = (out_b >= 0) if how != "anti" else (out_b < 0)
+            arange = torch.arange(
+                np_, dtype=torch.int64, device=probe_keys.device
+            )
+            cols = [arange] if how == "anti" else [arange, out_b]
+            outs = ext.compact_columns_cap(matched, cols)
+            flags = torch.cat(
+                [dup.reshape(1), outs[-1].reshape(1)]
+            ).cpu()
+            if int(flags[0].item()) == 0:
+                total = int(flags[1].item())
+                if how == "anti":
+                    pi = outs[0].narrow(0, 0, total)
+                    return pi, torch.full_like(pi, -1)
+                return (
+                    outs[0].narrow(0, 0, total),
+                    outs[1].narrow(0, 0, total),
+                )
     # duplicate build keys: 2-pass count+prefix+emit; a 3-pass
     # total+chunked-reservation variant (ext.join_pairs) measured
     # SLOWER — the random chain walk dominates, not the streaming
