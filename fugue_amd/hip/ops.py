@@ -233,11 +233,16 @@ def _key_stats_device(
     st = st.cpu().tolist()  # the single host readback
     k = len(datas)
     BIAS = 1 << 63
+    U64 = (1 << 64) - 1
+
+    def _dec(x: int) -> int:
+        raw = (x & U64) ^ BIAS  # undo the order-preserving bias
+        return raw - (1 << 64) if raw >= BIAS else raw
+
     mins = maxs = None
     if with_minmax:
-        u = [x & ((1 << 64) - 1) for x in st[: 2 * k]]
-        mins = [(v ^ BIAS) - BIAS for v in u[0::2]]
-        maxs = [(v ^ BIAS) - BIAS for v in u[1::2]]
+        mins = [_dec(v) for v in st[: 2 * k : 2]]
+        maxs = [_dec(v) for v in st[1 : 2 * k : 2]]
     d, f1, f2 = st[2 * k], st[2 * k + 1], st[2 * k + 2]
     if n > nsamples:
         est = d + (f1 * f1) // max(2 * f2, 1)

@@ -663,10 +663,15 @@ def test_gb_key_stats_gpu(engine):
     kd = k.cuda()
     st = get_ext().gb_key_stats([kd], [None], 65536, 131072, True).cpu()
     BIAS = 1 << 63
-    u = [int(x) & ((1 << 64) - 1) for x in st.tolist()]
-    lo, hi = (u[0] ^ BIAS) - BIAS, (u[1] ^ BIAS) - BIAS
+    U64 = (1 << 64) - 1
+
+    def dec(x):
+        raw = (int(x) & U64) ^ BIAS
+        return raw - (1 << 64) if raw >= BIAS else raw
+
+    lo, hi = dec(st[0]), dec(st[1])
     assert lo == int(k.min()) and hi == int(k.max())
-    d, f1, f2 = u[2], u[3], u[4]
+    d, f1, f2 = int(st[2]), int(st[3]), int(st[4])
     est = d + (f1 * f1) // max(2 * f2, 1)
     assert true_d / 2 <= est <= true_d * 2
 
