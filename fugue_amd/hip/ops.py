@@ -558,6 +558,14 @@ def groupby_aggregate(
         )
     # deterministic own-kernel compaction of the group table (replaces
     # nonzero + per-column index_select); ONE host read for the total
+    if len(aggs) > 6:  # beyond the kernel's by-value pointer pack
+        occupied = (tkeys != GB_EMPTY).nonzero(as_tuple=True)[0]
+        out_keys = tkeys.index_select(0, occupied)
+        out_count = gcount.index_select(0, occupied)
+        out_aggs: Dict[str, torch.Tensor] = {}
+        for i, (_, op, oname) in enumerate(aggs):
+            out_aggs[oname] = gaggs[i].index_select(0, occupied)
+        return out_keys, out_aggs, out_count, meta
     ck, cc, ca, _ce, bases = ext.gb_compact(
         tkeys, gcount,
         gaggs if len(aggs) > 0
@@ -567,7 +575,7 @@ def groupby_aggregate(
     total = int(bases[-1].item())
     out_keys = ck.narrow(0, 0, total)
     out_count = cc.narrow(0, 0, total)
-    out_aggs: Dict[str, torch.Tensor] = {}
+    out_aggs = {}
     for i, (_, op, oname) in enumerate(aggs):
         out_aggs[oname] = ca[i].narrow(0, 0, total)
     return out_keys, out_aggs, out_count, meta
@@ -806,11 +814,19 @@ def groupby_aggregate_hashed(
     rep, th2, conflict = ext.gb_mark_reps(h1, h2, tkeys, tsize)
     if int(conflict.item()) > 0:
         raise HashCollisionError("h1 collision on string keys")
+    if len(aggs) > 6:  # beyond the kernel's by-value pointer pack
+        occupied = (tkeys != GB_EMPTY).nonzero(as_tuple=True)[0]
+        reps = rep.index_select(0, occupied)
+        counts = gcount.index_select(0, occupied)
+        out_aggs: Dict[str, torch.Tensor] = {}
+        for i, (_, op, oname) in enumerate(aggs):
+            out_aggs[oname] = gaggs[i].index_select(0, occupied)
+        return reps, out_aggs, counts
     ck, cc, ca, ce, bases = ext.gb_compact(tkeys, gcount, gaggs, rep)
     total = int(bases[-1].item())
     reps = ce.narrow(0, 0, total)
     counts = cc.narrow(0, 0, total)
-    out_aggs: Dict[str, torch.Tensor] = {}
+    out_aggs = {}
     for i, (_, op, oname) in enumerate(aggs):
         out_aggs[oname] = ca[i].narrow(0, 0, total)
     return reps, out_aggs, counts

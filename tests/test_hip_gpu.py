@@ -705,3 +705,27 @@ def test_gb_compact_gpu(engine):
         assert torch.equal(
             ca[a].narrow(0, 0, total), gaggs[a].index_select(0, occ)
         )
+
+
+def test_many_aggregates_gpu(engine):
+    """>6 aggregate columns exercises the index_select fallback behind
+    the gb_compact kernel's 6-slot pointer pack."""
+    pdf = pd.DataFrame(
+        dict(
+            k=np.arange(5000) % 97,
+            **{f"v{i}": np.random.default_rng(i).random(5000) for i in range(4)},
+        )
+    )
+    aggs = {}
+    for i in range(4):
+        aggs[f"s{i}"] = f.sum(col(f"v{i}"))
+        aggs[f"m{i}"] = f.max(col(f"v{i}"))
+    res = fa.aggregate(pdf, partition_by="k", engine=engine, as_fugue=True, **aggs)
+    got = res.as_pandas().sort_values("k").reset_index(drop=True)
+    exp = pdf.groupby("k").agg(
+        **{f"s{i}": (f"v{i}", "sum") for i in range(4)},
+        **{f"m{i}": (f"v{i}", "max") for i in range(4)},
+    ).reset_index().sort_values("k").reset_index(drop=True)
+    for i in range(4):
+        np.testing.assert_allclose(got[f"s{i}"], exp[f"s{i}"], rtol=1e-9)
+        np.testing.assert_allclose(got[f"m{i}"], exp[f"m{i}"], rtol=1e-12)
