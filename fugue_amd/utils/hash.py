@@ -6,6 +6,7 @@ is stable across processes and runs: it never uses object ids.  Objects may
 customize by defining ``__uuid__``.
 """
 import uuid
+import weakref
 from typing import Any, Iterable
 
 _NAMESPACE = uuid.UUID("6b1f84e2-13a1-4a73-9a33-0d67b6f0a1c5")
@@ -84,11 +85,35 @@ def _is_raw_frame(obj: Any) -> bool:
     )
 
 
+_FRAME_TOKEN_MEMO: dict = {}
+_FRAME_TOKEN_MEMO_CAP = 256
+
+
 def _frame_token(obj: Any) -> str:
-    """Identity token for a raw pandas/arrow frame: content-based for
-    small frames (so identical literals dedup deterministically), id-based
-    for large ones (content hashing a 1e8-row frame is not acceptable —
-    such tasks are simply not deterministic across runs)."""
+    """Identity token for a raw pandas/arrow frame, memoized per object
+    identity (frames are treated as immutable, so the same object always
+    tokenizes the same — repeated workflow builds over one frame skip
+    re-serialization)."""
+    key = id(obj)
+    hit = _FRAME_TOKEN_MEMO.get(key)
+    if hit is not None and hit[0]() is obj:
+        return hit[1]
+    tok = _frame_token_impl(obj)
+    try:
+        ref = weakref.ref(obj)
+    except TypeError:
+        return tok
+    if len(_FRAME_TOKEN_MEMO) >= _FRAME_TOKEN_MEMO_CAP:
+        _FRAME_TOKEN_MEMO.pop(next(iter(_FRAME_TOKEN_MEMO)))
+    _FRAME_TOKEN_MEMO[key] = (ref, tok)
+    return tok
+
+
+def _frame_token_impl(obj: Any) -> str:
+    """Content-based for small frames (so identical literals dedup
+    deterministically), id-based for large ones (content hashing a
+    1e8-row frame is not acceptable — such tasks are simply not
+    deterministic across runs)."""
     try:
         n = len(obj)
     except TypeError:
