@@ -88,6 +88,10 @@ void launch_pack_cols(const void**, const bool**, const int*, const int64_t*,
                       const int*, int, int64_t, int64_t*, hipStream_t);
 void launch_unpack_col(const int64_t*, int64_t, int, int64_t, int64_t, int,
                        int, void*, bool*, hipStream_t);
+void launch_cmp_imm(const void*, const bool*, int, int64_t, double, int,
+                    int, int64_t, bool*, hipStream_t);
+void launch_cmp_col(const void*, const bool*, const void*, const bool*, int,
+                    int, int64_t, bool*, hipStream_t);
 void launch_gb_compact(const int64_t*, const int64_t*, const double**,
                        double**, int, int64_t, int64_t*, int64_t*, int64_t*,
                        int64_t*, const int64_t*, int64_t*, hipStream_t);
@@ -508,6 +512,46 @@ std::vector<at::Tensor> compact_columns_cap(at::Tensor mask,
   return outs;
 }
 
+
+namespace {
+int cmp_dtype_code(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kLong: return 0;
+    case at::kInt: return 1;
+    case at::kShort: return 2;
+    case at::kDouble: return 3;
+    case at::kFloat: return 4;
+    default: TORCH_CHECK(false, "unsupported compare dtype"); return -1;
+  }
+}
+}  // namespace
+
+// Fast path for single-comparison WHERE filters (see relational.hip).
+at::Tensor cmp_imm(at::Tensor a, c10::optional<at::Tensor> valid,
+                   int64_t imm_i, double imm_d, bool use_int, int64_t op) {
+  check_gpu(a, "a");
+  int dt = cmp_dtype_code(a);
+  bool ui = use_int && dt <= 2;
+  auto out = at::empty({a.numel()}, a.options().dtype(at::kBool));
+  launch_cmp_imm(a.data_ptr(), opt_valid_ptr(valid), dt, imm_i, imm_d,
+                 ui ? 1 : 0, (int)op, a.numel(), out.data_ptr<bool>(),
+                 current_stream());
+  return out;
+}
+
+at::Tensor cmp_col(at::Tensor a, c10::optional<at::Tensor> va, at::Tensor b,
+                   c10::optional<at::Tensor> vb, int64_t op) {
+  check_gpu(a, "a");
+  check_gpu(b, "b");
+  TORCH_CHECK(a.scalar_type() == b.scalar_type(), "same dtype required");
+  int dt = cmp_dtype_code(a);
+  auto out = at::empty({a.numel()}, a.options().dtype(at::kBool));
+  launch_cmp_col(a.data_ptr(), opt_valid_ptr(va), b.data_ptr(),
+                 opt_valid_ptr(vb), dt, (int)op, a.numel(),
+                 out.data_ptr<bool>(), current_stream());
+  return out;
+}
+
 // host mirror of ExprProg in relational.hip (layout must match)
 struct ExprProgHost {
   int n_ops;
@@ -810,6 +854,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "deterministic group-table compaction");
   m.def("compact_columns_cap", &compact_columns_cap,
         "mask compaction, capacity outputs + cursor");
+  m.def("cmp_imm", &cmp_imm, "single col-vs-literal comparison mask");
+  m.def("cmp_col", &cmp_col, "single col-vs-col comparison mask");
   m.def("compact_columns", &compact_columns,
         "fused masked compaction of 8-byte columns");
   m.def("join_count", &join_count, "count matches per probe row");

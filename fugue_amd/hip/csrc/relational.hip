@@ -2357,7 +2357,10 @@ __global__ __launch_bounds__(BLOCK) void minmax_cols_kernel(
       lmax[c] = u > lmax[c] ? u : lmax[c];
     }
   }
-  // wave reduce then one atomic per wave per col
+  // wave reduce -> LDS -> one atomic per block per col (same-address
+  // atomics serialize in L2 even wave-aggregated — profiles/NOTES.md)
+  __shared__ uint64_t smin[NC][BLOCK / WAVE];
+  __shared__ uint64_t smax[NC][BLOCK / WAVE];
 #pragma unroll
   for (int c = 0; c < NC; ++c) {
     for (int off = WAVE / 2; off > 0; off >>= 1) {
@@ -2367,8 +2370,21 @@ __global__ __launch_bounds__(BLOCK) void minmax_cols_kernel(
       lmax[c] = omax > lmax[c] ? omax : lmax[c];
     }
     if ((threadIdx.x & (WAVE - 1)) == 0) {
-      atomicMin((unsigned long long*)&out[2 * c], lmin[c]);
-      atomicMax((unsigned long long*)&out[2 * c + 1], lmax[c]);
+      smin[c][threadIdx.x / WAVE] = lmin[c];
+      smax[c][threadIdx.x / WAVE] = lmax[c];
+    }
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      uint64_t bmin = smin[c][0], bmax = smax[c][0];
+      for (int w = 1; w < BLOCK / WAVE; ++w) {
+        bmin = smin[c][w] < bmin ? smin[c][w] : bmin;
+        bmax = smax[c][w] > bmax ? smax[c][w] : bmax;
+      }
+      atomicMin((unsigned long long*)&out[2 * c], bmin);
+      atomicMax((unsigned long long*)&out[2 * c + 1], bmax);
     }
   }
 }
@@ -2458,8 +2474,9 @@ void launch_minmax_cols(const void** data, const bool** valid,
   }
   hipLaunchKernelGGL(minmax_init_kernel, dim3(1), dim3(BLOCK), 0, stream,
                      out, ncols);
-  launch_minmax_dispatch(ncols, dim3(grid_for(n, 4)), dim3(BLOCK), stream,
-                         pc, n, out);
+  int mg = grid_for(n, 8);
+  if (mg > 4096) mg = 4096;
+  launch_minmax_dispatch(ncols, dim3(mg), dim3(BLOCK), stream, pc, n, out);
 }
 
 void launch_pack_cols(const void** data, const bool** valid,
@@ -2553,12 +2570,24 @@ __global__ __launch_bounds__(BLOCK) void distinct_stats_kernel(
     if (c == 1) ++f1;
     if (c == 2) ++f2;
   }
+  __shared__ int sd[BLOCK / WAVE], sf1[BLOCK / WAVE], sf2[BLOCK / WAVE];
   for (int off = WAVE / 2; off > 0; off >>= 1) {
     d += __shfl_down(d, off, WAVE);
     f1 += __shfl_down(f1, off, WAVE);
     f2 += __shfl_down(f2, off, WAVE);
   }
   if ((threadIdx.x & (WAVE - 1)) == 0) {
+    sd[threadIdx.x / WAVE] = d;
+    sf1[threadIdx.x / WAVE] = f1;
+    sf2[threadIdx.x / WAVE] = f2;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < BLOCK / WAVE; ++w) {
+      d += sd[w];
+      f1 += sf1[w];
+      f2 += sf2[w];
+    }
     atomicAdd((unsigned long long*)&out3[0], (unsigned long long)d);
     atomicAdd((unsigned long long*)&out3[1], (unsigned long long)f1);
     atomicAdd((unsigned long long*)&out3[2], (unsigned long long)f2);
@@ -2588,8 +2617,9 @@ void launch_gb_key_stats(const void** data, const bool** valid,
   if (do_minmax) {
     hipLaunchKernelGGL(minmax_init_kernel, dim3(1), dim3(BLOCK), 0, stream,
                        out, ncols);
-    launch_minmax_dispatch(ncols, dim3(grid_for(n, 4)), dim3(BLOCK), stream,
-                           pc, n, out);
+    int mg = grid_for(n, 8);
+    if (mg > 4096) mg = 4096;
+    launch_minmax_dispatch(ncols, dim3(mg), dim3(BLOCK), stream, pc, n, out);
   }
   int64_t stride = n / nsamples;
   if (stride < 1) stride = 1;
@@ -2731,6 +2761,113 @@ void launch_gb_compact(const int64_t* tkeys, const int64_t* gcount,
   hipLaunchKernelGGL(gbc_emit_kernel, dim3(nblocks), dim3(BLOCK), 0, stream,
                      tkeys, gcount, ag, n_aggs, tsize, bases, out_keys,
                      out_count, extra_src, extra_dst);
+}
+
+}  // extern "C"
+
+// ------------------------------------------------------------------ //
+// fast single-comparison filters                                      //
+//                                                                     //
+// The general expr_filter interpreter keeps its value stack in        //
+// scratch memory (dynamically indexed), which costs ~10x SOL on a     //
+// simple `col < imm` predicate (profiles r02c).  Single comparisons   //
+// — the dominant WHERE shape — get dedicated one-pass kernels.        //
+// ------------------------------------------------------------------ //
+
+template <typename CT>
+__device__ __forceinline__ bool cmp_apply(CT a, CT b, int op) {
+  switch (op) {
+    case 0: return a == b;
+    case 1: return a != b;
+    case 2: return a < b;
+    case 3: return a <= b;
+    case 4: return a > b;
+    default: return a >= b;
+  }
+}
+
+// col vs immediate; CT is the comparison domain (int64_t or double)
+template <typename T, typename CT>
+__global__ __launch_bounds__(BLOCK) void cmp_imm_kernel(
+    const T* __restrict__ a, const bool* __restrict__ valid, CT imm, int op,
+    int64_t n, bool* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bool r = cmp_apply<CT>((CT)a[i], imm, op);
+    out[i] = (valid == nullptr || valid[i]) && r;
+  }
+}
+
+// col vs col (same storage type); comparison in CT
+template <typename T, typename CT>
+__global__ __launch_bounds__(BLOCK) void cmp_col_kernel(
+    const T* __restrict__ a, const bool* __restrict__ va,
+    const T* __restrict__ b, const bool* __restrict__ vb, int op, int64_t n,
+    bool* __restrict__ out) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bool r = cmp_apply<CT>((CT)a[i], (CT)b[i], op);
+    out[i] = (va == nullptr || va[i]) && (vb == nullptr || vb[i]) && r;
+  }
+}
+
+extern "C" {
+
+// dtype codes: 0 i64, 1 i32, 2 i16, 3 f64, 4 f32 (int compare domain for
+// 0-2 with an integral immediate, double otherwise)
+void launch_cmp_imm(const void* a, const bool* valid, int dtype,
+                    int64_t imm_i, double imm_d, int use_int, int op,
+                    int64_t n, bool* out, hipStream_t stream) {
+  dim3 g(grid_for(n, 4)), b(BLOCK);
+#define CI(T)                                                              \
+  if (use_int) {                                                           \
+    hipLaunchKernelGGL((cmp_imm_kernel<T, int64_t>), g, b, 0, stream,      \
+                       (const T*)a, valid, (int64_t)imm_i, op, n, out);    \
+  } else {                                                                 \
+    hipLaunchKernelGGL((cmp_imm_kernel<T, double>), g, b, 0, stream,       \
+                       (const T*)a, valid, imm_d, op, n, out);             \
+  }
+  switch (dtype) {
+    case 0: CI(int64_t) break;
+    case 1: CI(int32_t) break;
+    case 2: CI(int16_t) break;
+    case 3: CI(double) break;
+    default: CI(float) break;
+  }
+#undef CI
+}
+
+void launch_cmp_col(const void* a, const bool* va, const void* b,
+                    const bool* vb, int dtype, int op, int64_t n, bool* out,
+                    hipStream_t stream) {
+  dim3 g(grid_for(n, 4)), blk(BLOCK);
+  switch (dtype) {
+    case 0:
+      hipLaunchKernelGGL((cmp_col_kernel<int64_t, int64_t>), g, blk, 0,
+                         stream, (const int64_t*)a, va, (const int64_t*)b,
+                         vb, op, n, out);
+      break;
+    case 1:
+      hipLaunchKernelGGL((cmp_col_kernel<int32_t, int64_t>), g, blk, 0,
+                         stream, (const int32_t*)a, va, (const int32_t*)b,
+                         vb, op, n, out);
+      break;
+    case 2:
+      hipLaunchKernelGGL((cmp_col_kernel<int16_t, int64_t>), g, blk, 0,
+                         stream, (const int16_t*)a, va, (const int16_t*)b,
+                         vb, op, n, out);
+      break;
+    case 3:
+      hipLaunchKernelGGL((cmp_col_kernel<double, double>), g, blk, 0,
+                         stream, (const double*)a, va, (const double*)b, vb,
+                         op, n, out);
+      break;
+    default:
+      hipLaunchKernelGGL((cmp_col_kernel<float, double>), g, blk, 0, stream,
+                         (const float*)a, va, (const float*)b, vb, op, n,
+                         out);
+      break;
+  }
 }
 
 }  // extern "C"

@@ -633,6 +633,63 @@ def try_fused_value(
     return vals, valid
 
 
+_CMP_OP_CODE = {"==": 0, "!=": 1, "<": 2, "<=": 3, ">": 4, ">=": 5}
+_CMP_DTYPES = (
+    torch.int64, torch.int32, torch.int16, torch.float64, torch.float32
+)
+
+
+def _try_simple_cmp(
+    expr: ColumnExpr, df: HipDataFrame
+) -> Optional[torch.Tensor]:
+    """Dedicated one-pass kernels for the dominant WHERE shapes
+    (``col <op> literal`` / ``col <op> col``): the general interpreter
+    keeps its value stack in scratch memory and runs ~10x off SOL on
+    these (profiles/NOTES.md r02c)."""
+    if not isinstance(expr, _BinaryOpExpr) or expr.as_type is not None:
+        return None
+    code = _CMP_OP_CODE.get(expr.op)
+    if code is None:
+        return None
+    left, right = expr.left, expr.right
+    swap = False
+    if isinstance(left, _LiteralColumnExpr) and isinstance(
+        right, _NamedColumnExpr
+    ):
+        left, right = right, left
+        swap = True
+        code = {0: 0, 1: 1, 2: 4, 3: 5, 4: 2, 5: 3}[code]  # mirror op
+    if not isinstance(left, _NamedColumnExpr) or left.as_type is not None:
+        return None
+    if left.name not in df.schema._index:
+        return None
+    a = df.col(left.name)
+    if isinstance(a, StringDeviceColumn) or a.data.dtype not in _CMP_DTYPES:
+        return None
+    from fugue_amd.hip.ext import get_ext
+
+    if isinstance(right, _LiteralColumnExpr):
+        v = right.value
+        if isinstance(v, bool) or v is None:
+            return None
+        if isinstance(v, int):
+            return get_ext().cmp_imm(a.data, a.valid, v, float(v), True, code)
+        if isinstance(v, float):
+            return get_ext().cmp_imm(a.data, a.valid, 0, v, False, code)
+        return None
+    if isinstance(right, _NamedColumnExpr) and right.as_type is None:
+        if right.name not in df.schema._index:
+            return None
+        b = df.col(right.name)
+        if (
+            isinstance(b, StringDeviceColumn)
+            or b.data.dtype != a.data.dtype
+        ):
+            return None
+        return get_ext().cmp_col(a.data, a.valid, b.data, b.valid, code)
+    return None
+
+
 def try_fused_filter(
     expr: ColumnExpr, df: HipDataFrame
 ) -> Optional[torch.Tensor]:
@@ -640,6 +697,9 @@ def try_fused_filter(
     expression (or frame location) isn't fusable."""
     if not df.device.startswith("cuda"):
         return None
+    simple = _try_simple_cmp(expr, df)
+    if simple is not None:
+        return simple
     b = _ProgBuilder(df)
     try:
         b.emit(expr)

@@ -729,3 +729,38 @@ def test_many_aggregates_gpu(engine):
     for i in range(4):
         np.testing.assert_allclose(got[f"s{i}"], exp[f"s{i}"], rtol=1e-9)
         np.testing.assert_allclose(got[f"m{i}"], exp[f"m{i}"], rtol=1e-12)
+
+
+def test_simple_cmp_filter_gpu(engine):
+    """Dedicated compare kernels agree with pandas across dtypes,
+    literal sides and null handling."""
+    rng = np.random.default_rng(9)
+    n = 100_000
+    pdf = pd.DataFrame(
+        dict(
+            a=rng.integers(-1000, 1000, n),
+            b=rng.integers(-1000, 1000, n).astype(np.int32),
+            x=rng.random(n),
+            y=rng.random(n),
+        )
+    )
+    pdf.loc[rng.random(n) < 0.05, "x"] = None
+    for cond, mask in [
+        (col("a") > lit(250), pdf.a > 250),
+        (lit(250) > col("a"), 250 > pdf.a),
+        (col("b") <= lit(-10), pdf.b <= -10),
+        (col("x") < lit(0.5), pdf.x < 0.5),
+        (col("a") != lit(0), pdf.a != 0),
+        (col("x") >= col("y"), pdf.x >= pdf.y),
+        (col("a") == col("a"), pdf.a == pdf.a),
+    ]:
+        got = (
+            fa.filter(pdf, cond, engine=engine, as_fugue=True)
+            .as_pandas()
+            .sort_values(["a", "b"])
+            .reset_index(drop=True)
+        )
+        exp = pdf[mask.fillna(False)].sort_values(["a", "b"]).reset_index(
+            drop=True
+        )
+        pd.testing.assert_frame_equal(got, exp, check_dtype=False)
