@@ -88,6 +88,11 @@ void launch_pack_cols(const void**, const bool**, const int*, const int64_t*,
                       const int*, int, int64_t, int64_t*, hipStream_t);
 void launch_unpack_col(const int64_t*, int64_t, int, int64_t, int64_t, int,
                        int, void*, bool*, hipStream_t);
+void launch_topk(const void*, int, int, int64_t, int64_t, void*, int64_t*,
+                 void*, int64_t*, hipStream_t);
+void launch_eq2_mask(const int64_t*, const int64_t*, const int64_t*,
+                     const int64_t*, const bool*, int, int64_t, bool*,
+                     hipStream_t);
 void launch_cmp_imm(const void*, const bool*, int, int64_t, double, int,
                     int, int64_t, bool*, hipStream_t);
 void launch_cmp_col(const void*, const bool*, const void*, const bool*, int,
@@ -489,8 +494,8 @@ std::vector<at::Tensor> compact_columns(at::Tensor mask,
 // Capacity-mode compaction: outputs are allocated at mask length and the
 // caller narrows using the returned cursor (total) — skips the separate
 // mask.sum() reduction + its host sync (profiles/NOTES.md r02).
-std::vector<at::Tensor> compact_columns_cap(at::Tensor mask,
-                                            std::vector<at::Tensor> cols) {
+std::vector<at::Tensor> compact_columns_cap(
+    at::Tensor mask, std::vector<c10::optional<at::Tensor>> cols) {
   check_gpu(mask, "mask");
   TORCH_CHECK(cols.size() >= 1 && cols.size() <= 8, "1..8 columns");
   int64_t n = mask.numel();
@@ -499,10 +504,16 @@ std::vector<at::Tensor> compact_columns_cap(at::Tensor mask,
   uint64_t* dsts[8];
   std::vector<at::Tensor> outs;
   for (size_t c = 0; c < cols.size(); ++c) {
-    check_gpu(cols[c], "col");
-    TORCH_CHECK(cols[c].element_size() == 8, "8-byte columns only");
-    auto out = at::empty({n}, cols[c].options());
-    srcs[c] = reinterpret_cast<const uint64_t*>(cols[c].data_ptr());
+    at::Tensor out;
+    if (cols[c].has_value()) {  // nullopt -> emit the row index
+      check_gpu(*cols[c], "col");
+      TORCH_CHECK(cols[c]->element_size() == 8, "8-byte columns only");
+      srcs[c] = reinterpret_cast<const uint64_t*>(cols[c]->data_ptr());
+      out = at::empty({n}, cols[c]->options());
+    } else {
+      srcs[c] = nullptr;
+      out = at::empty({n}, mask.options().dtype(at::kLong));
+    }
     dsts[c] = reinterpret_cast<uint64_t*>(out.data_ptr());
     outs.push_back(out);
   }
@@ -525,6 +536,48 @@ int cmp_dtype_code(const at::Tensor& t) {
   }
 }
 }  // namespace
+
+
+
+// Own top-k select for k <= 16 (see relational.hip); returns
+// (values[k], indices[k]) sorted, index-tiebroken (deterministic).
+std::vector<at::Tensor> topk_select(at::Tensor vals, int64_t k,
+                                    bool largest) {
+  check_gpu(vals, "vals");
+  TORCH_CHECK(k >= 1 && k <= 16, "k must be in [1, 16]");
+  int dt;
+  if (vals.scalar_type() == at::kLong) {
+    dt = 0;
+  } else if (vals.scalar_type() == at::kDouble) {
+    dt = 1;
+  } else {
+    TORCH_CHECK(false, "topk_select supports int64/float64");
+  }
+  int64_t n = vals.numel();
+  TORCH_CHECK(n >= 1, "empty input");
+  auto cand_v = at::empty({512 * k}, vals.options());
+  auto cand_i = at::empty({512 * k}, vals.options().dtype(at::kLong));
+  auto out_v = at::empty({k}, vals.options());
+  auto out_i = at::empty({k}, vals.options().dtype(at::kLong));
+  launch_topk(vals.data_ptr(), dt, largest ? 1 : 0, n, k,
+              cand_v.data_ptr(), cand_i.data_ptr<int64_t>(),
+              out_v.data_ptr(), out_i.data_ptr<int64_t>(),
+              current_stream());
+  return {out_v, out_i};
+}
+
+at::Tensor eq2_mask(at::Tensor h1, at::Tensor h2, at::Tensor l1,
+                    at::Tensor l2, c10::optional<at::Tensor> valid,
+                    bool neg) {
+  check_gpu(h1, "h1");
+  check_gpu(h2, "h2");
+  auto out = at::empty({h1.numel()}, h1.options().dtype(at::kBool));
+  launch_eq2_mask(h1.data_ptr<int64_t>(), h2.data_ptr<int64_t>(),
+                  l1.data_ptr<int64_t>(), l2.data_ptr<int64_t>(),
+                  opt_valid_ptr(valid), neg ? 1 : 0, h1.numel(),
+                  out.data_ptr<bool>(), current_stream());
+  return out;
+}
 
 // Fast path for single-comparison WHERE filters (see relational.hip).
 at::Tensor cmp_imm(at::Tensor a, c10::optional<at::Tensor> valid,
@@ -854,6 +907,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "deterministic group-table compaction");
   m.def("compact_columns_cap", &compact_columns_cap,
         "mask compaction, capacity outputs + cursor");
+  m.def("topk_select", &topk_select, "own top-k (k<=16) select");
+  m.def("eq2_mask", &eq2_mask, "128-bit string equality mask");
   m.def("cmp_imm", &cmp_imm, "single col-vs-literal comparison mask");
   m.def("cmp_col", &cmp_col, "single col-vs-col comparison mask");
   m.def("compact_columns", &compact_columns,

@@ -2266,7 +2266,8 @@ __global__ __launch_bounds__(BLOCK) void compact8_kernel(
     for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
       if (!mask[i]) continue;
       int64_t pos = (int64_t)lbase + atomicAdd(&lcount, 1);
-      for (int c = 0; c < ncols; ++c) dsts[c][pos] = srcs[c][i];
+      for (int c = 0; c < ncols; ++c)
+        dsts[c][pos] = srcs[c] != nullptr ? srcs[c][i] : (uint64_t)i;
     }
     __syncthreads();
   }
@@ -2870,4 +2871,194 @@ void launch_cmp_col(const void* a, const bool* va, const void* b,
   }
 }
 
+}  // extern "C"
+
+// 128-bit string-equality mask: (h1==l1 && h2==l2 [&& valid]) ^ neg,
+// literal hashes read from device memory (no host sync, no torch
+// compare_scalar residue).
+__global__ __launch_bounds__(BLOCK) void eq2_mask_kernel(
+    const int64_t* __restrict__ h1, const int64_t* __restrict__ h2,
+    const int64_t* __restrict__ l1, const int64_t* __restrict__ l2,
+    const bool* __restrict__ valid, int neg, int64_t n,
+    bool* __restrict__ out) {
+  int64_t a = l1[0], b = l2[0];
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    bool eq = h1[i] == a && h2[i] == b;
+    if (neg) eq = !eq;
+    out[i] = eq && (valid == nullptr || valid[i]);
+  }
+}
+
+extern "C" {
+void launch_eq2_mask(const int64_t* h1, const int64_t* h2, const int64_t* l1,
+                     const int64_t* l2, const bool* valid, int neg,
+                     int64_t n, bool* out, hipStream_t stream) {
+  hipLaunchKernelGGL(eq2_mask_kernel, dim3(grid_for(n, 4)), dim3(BLOCK), 0,
+                     stream, h1, h2, l1, l2, valid, neg, n, out);
+}
+}  // extern "C"
+
+// ------------------------------------------------------------------ //
+// own top-k select (k <= 16): torch.topk's rocPRIM path runs a        //
+// merge-sort cascade (~0.3 ms/step on q3's LIMIT 10); two own passes  //
+// — per-thread register top-k + LDS log-merge per block, then one     //
+// block over the per-block candidates — read the data once.          //
+// ------------------------------------------------------------------ //
+
+#define TOPK_MAX 16
+#define TOPK_BLOCKS 512
+
+template <typename T, bool LARGEST>
+__device__ __forceinline__ bool tk_before(T a, int64_t ia, T b, int64_t ib) {
+  // strict ordering with index tiebreak (deterministic)
+  if (a != b) return LARGEST ? (a > b) : (a < b);
+  return ia < ib;
+}
+
+template <typename T, bool LARGEST>
+__device__ void tk_insert(T v, int64_t idx, T* tv, int64_t* ti, int k) {
+  if (ti[k - 1] >= 0 && !tk_before<T, LARGEST>(v, idx, tv[k - 1], ti[k - 1]))
+    return;
+  int p = k - 1;
+  while (p > 0 &&
+         (ti[p - 1] < 0 || tk_before<T, LARGEST>(v, idx, tv[p - 1], ti[p - 1]))) {
+    tv[p] = tv[p - 1];
+    ti[p] = ti[p - 1];
+    --p;
+  }
+  tv[p] = v;
+  ti[p] = idx;
+}
+
+// merge two sorted k-lists (a <- top k of a ∪ b); idx<0 marks empty
+template <typename T, bool LARGEST>
+__device__ void tk_merge(T* av, int64_t* ai, const T* bv, const int64_t* bi,
+                         int k) {
+  T mv[TOPK_MAX];
+  int64_t mi[TOPK_MAX];
+  int pa = 0, pb = 0;
+  for (int o = 0; o < k; ++o) {
+    bool take_a;
+    if (pa < k && ai[pa] >= 0) {
+      take_a = !(pb < k && bi[pb] >= 0) ||
+               tk_before<T, LARGEST>(av[pa], ai[pa], bv[pb], bi[pb]);
+    } else {
+      take_a = false;
+    }
+    if (take_a) {
+      mv[o] = av[pa];
+      mi[o] = ai[pa];
+      ++pa;
+    } else if (pb < k && bi[pb] >= 0) {
+      mv[o] = bv[pb];
+      mi[o] = bi[pb];
+      ++pb;
+    } else {
+      mi[o] = -1;
+    }
+  }
+  for (int o = 0; o < k; ++o) {
+    av[o] = mv[o];
+    ai[o] = mi[o];
+  }
+}
+
+template <typename T, bool LARGEST>
+__global__ __launch_bounds__(BLOCK) void topk_stage1_kernel(
+    const T* __restrict__ vals, int64_t n, int k, T* __restrict__ cand_v,
+    int64_t* __restrict__ cand_i) {
+  __shared__ T sv[BLOCK * TOPK_MAX];
+  __shared__ int64_t si[BLOCK * TOPK_MAX];
+  T tv[TOPK_MAX];
+  int64_t ti[TOPK_MAX];
+  for (int j = 0; j < k; ++j) ti[j] = -1;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    tk_insert<T, LARGEST>(vals[i], i, tv, ti, k);
+  for (int j = 0; j < k; ++j) {
+    sv[threadIdx.x * k + j] = tv[j];
+    si[threadIdx.x * k + j] = ti[j];
+  }
+  __syncthreads();
+  for (int stride = 1; stride < BLOCK; stride <<= 1) {
+    if ((threadIdx.x & (2 * stride - 1)) == 0) {
+      tk_merge<T, LARGEST>(&sv[threadIdx.x * k], &si[threadIdx.x * k],
+                           &sv[(threadIdx.x + stride) * k],
+                           &si[(threadIdx.x + stride) * k], k);
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    for (int j = 0; j < k; ++j) {
+      cand_v[blockIdx.x * k + j] = sv[j];
+      cand_i[blockIdx.x * k + j] = si[j];
+    }
+  }
+}
+
+template <typename T, bool LARGEST>
+__global__ __launch_bounds__(BLOCK) void topk_stage2_kernel(
+    const T* __restrict__ cand_v, const int64_t* __restrict__ cand_i,
+    int nblocks, int k, T* __restrict__ out_v, int64_t* __restrict__ out_i) {
+  __shared__ T sv[BLOCK * TOPK_MAX];
+  __shared__ int64_t si[BLOCK * TOPK_MAX];
+  T tv[TOPK_MAX];
+  int64_t ti[TOPK_MAX];
+  for (int j = 0; j < k; ++j) ti[j] = -1;
+  int64_t total = (int64_t)nblocks * k;
+  for (int64_t i = threadIdx.x; i < total; i += blockDim.x)
+    if (cand_i[i] >= 0)
+      tk_insert<T, LARGEST>(cand_v[i], cand_i[i], tv, ti, k);
+  for (int j = 0; j < k; ++j) {
+    sv[threadIdx.x * k + j] = tv[j];
+    si[threadIdx.x * k + j] = ti[j];
+  }
+  __syncthreads();
+  for (int stride = 1; stride < BLOCK; stride <<= 1) {
+    if ((threadIdx.x & (2 * stride - 1)) == 0) {
+      tk_merge<T, LARGEST>(&sv[threadIdx.x * k], &si[threadIdx.x * k],
+                           &sv[(threadIdx.x + stride) * k],
+                           &si[(threadIdx.x + stride) * k], k);
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    for (int j = 0; j < k; ++j) {
+      out_v[j] = sv[j];
+      out_i[j] = si[j];
+    }
+  }
+}
+
+static void launch_topk_dispatch(int dtype, int largest, dim3 g1, dim3 b,
+                                 hipStream_t stream, const void* vals,
+                                 int64_t n, int k, void* cand_v,
+                                 int64_t* cand_i, void* out_v,
+                                 int64_t* out_i) {
+#define TK(T, L)                                                            \
+  do {                                                                      \
+    hipLaunchKernelGGL((topk_stage1_kernel<T, L>), g1, b, 0, stream,        \
+                       (const T*)vals, n, k, (T*)cand_v, cand_i);           \
+    hipLaunchKernelGGL((topk_stage2_kernel<T, L>), dim3(1), b, 0, stream,   \
+                       (const T*)cand_v, cand_i, (int)g1.x, k, (T*)out_v,   \
+                       out_i);                                              \
+  } while (0)
+  if (dtype == 0) {
+    if (largest) TK(int64_t, true); else TK(int64_t, false);
+  } else {
+    if (largest) TK(double, true); else TK(double, false);
+  }
+#undef TK
+}
+
+extern "C" {
+void launch_topk(const void* vals, int dtype, int largest, int64_t n, int k,
+                 void* cand_v, int64_t* cand_i, void* out_v, int64_t* out_i,
+                 hipStream_t stream) {
+  int blocks = grid_for(n, 8);
+  if (blocks > TOPK_BLOCKS) blocks = TOPK_BLOCKS;
+  launch_topk_dispatch(dtype, largest, dim3(blocks), dim3(BLOCK), stream,
+                       vals, n, k, cand_v, cand_i, out_v, out_i);
+}
 }  // extern "C"

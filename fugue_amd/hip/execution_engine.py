@@ -1379,9 +1379,20 @@ class HipExecutionEngine(ExecutionEngine):
                         vals = torch.where(
                             c.valid, vals, torch.full_like(vals, sentinel)
                         )
-                    _, perm = torch.topk(
-                        vals, k, largest=not asc, sorted=True
-                    )
+                    if k <= 16 and vals.dtype in (
+                        torch.int64, torch.float64
+                    ) and vals.is_cuda:
+                        # own 2-pass register/LDS top-k (k<=16): one data
+                        # read instead of rocPRIM's merge-sort cascade
+                        from fugue_amd.hip.ext import get_ext
+
+                        _, perm = get_ext().topk_select(
+                            vals.contiguous(), k, not asc
+                        )
+                    else:
+                        _, perm = torch.topk(
+                            vals, k, largest=not asc, sorted=True
+                        )
                 else:
                     perm = dops.sort_indices(
                         d, keys_l, list(_presort.values())

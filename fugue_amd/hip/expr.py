@@ -391,9 +391,18 @@ def _try_string_equality(
     lit_col = StringDeviceColumn.from_arrow_strings(
         pa.array([lit_e.value], type=pa.string()), df.device
     )
-    l1 = dops.hash_rows([lit_col])[0]
-    l2 = dops.hash_rows([lit_col], seed=dops._H2_SEED)[0]
-    eq = (h1 == l1) & (h2 == l2)
+    l1 = dops.hash_rows([lit_col])
+    l2 = dops.hash_rows([lit_col], seed=dops._H2_SEED)
+    if h1.is_cuda:
+        from fugue_amd.hip.ext import get_ext
+
+        return (
+            get_ext().eq2_mask(
+                h1, h2, l1, l2, c.valid, expr.op == "!="
+            ),
+            None,
+        )
+    eq = (h1 == l1[0]) & (h2 == l2[0])
     if c.valid is not None:
         eq = eq & c.valid
     if expr.op == "!=":

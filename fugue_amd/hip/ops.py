@@ -911,10 +911,8 @@ def hash_join_indices(
                 return out_p, out_b
         else:
             matched = (out_b >= 0) if how != "anti" else (out_b < 0)
-            arange = torch.arange(
-                np_, dtype=torch.int64, device=probe_keys.device
-            )
-            cols = [arange] if how == "anti" else [arange, out_b]
+            # None column = "emit the row index" (no arange materialized)
+            cols = [None] if how == "anti" else [None, out_b]
             outs = ext.compact_columns_cap(matched, cols)
             flags = torch.cat(
                 [dup.reshape(1), outs[-1].reshape(1)]

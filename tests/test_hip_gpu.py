@@ -764,3 +764,40 @@ def test_simple_cmp_filter_gpu(engine):
             drop=True
         )
         pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_topk_select_gpu(engine):
+    """Own top-k kernel vs torch.topk / numpy across dtypes, directions
+    and sizes (incl. n < 512*k and duplicate values)."""
+    import torch
+
+    from fugue_amd.hip.ext import get_ext
+
+    rng = np.random.default_rng(17)
+    for n in (37, 5000, 2_000_000):
+        for dt in (np.float64, np.int64):
+            vals = rng.integers(-(10**6), 10**6, n).astype(dt)
+            vt = torch.tensor(vals).cuda()
+            for largest in (True, False):
+                for k in (1, 10, 16):
+                    kk = min(k, n)
+                    ov, oi = get_ext().topk_select(vt, kk, largest)
+                    exp = np.sort(vals)[::-1][:kk] if largest else np.sort(vals)[:kk]
+                    got = ov.cpu().numpy()
+                    np.testing.assert_array_equal(got, exp.astype(dt))
+                    # indices must address the right values
+                    np.testing.assert_array_equal(
+                        vals[oi.cpu().numpy()], exp.astype(dt)
+                    )
+
+
+def test_take_presort_matches_pandas_gpu(engine):
+    """take() with a single presort key goes through the own top-k
+    kernel and matches pandas nsmallest/nlargest."""
+    rng = np.random.default_rng(23)
+    pdf = pd.DataFrame(dict(a=rng.integers(0, 10**9, 300_000), b=rng.random(300_000)))
+    res = fa.take(pdf, 10, presort="b desc", engine=engine, as_fugue=True)
+    exp = pdf.nlargest(10, "b").reset_index(drop=True)
+    pd.testing.assert_frame_equal(
+        res.as_pandas().reset_index(drop=True), exp, check_dtype=False
+    )
