@@ -325,11 +325,14 @@ const int64_t* opt_i64_ptr(const c10::optional<at::Tensor>& v) {
 std::vector<at::Tensor> join_emit_unique(
     at::Tensor pkeys, at::Tensor bkeys, c10::optional<at::Tensor> ph2,
     c10::optional<at::Tensor> bh2, at::Tensor heads, at::Tensor next,
-    int64_t mode) {
+    int64_t mode, bool want_pi) {
   check_gpu(pkeys, "pkeys");
   int64_t np = pkeys.numel();
   int64_t tsize = heads.numel();
-  auto out_pi = at::empty({np}, pkeys.options());
+  // positional mode (1) with want_pi=false skips the pi write — the
+  // compaction kernel emits row indices itself
+  auto out_pi = at::empty({(mode == 1 && !want_pi) ? 0 : np},
+                          pkeys.options());
   auto out_bi = at::empty({np}, pkeys.options());
   auto cursor = at::zeros({1}, pkeys.options());
   if (np > 0) {
@@ -337,8 +340,9 @@ std::vector<at::Tensor> join_emit_unique(
         pkeys.data_ptr<int64_t>(), np, bkeys.data_ptr<int64_t>(),
         opt_i64_ptr(ph2), opt_i64_ptr(bh2), heads.data_ptr<int32_t>(),
         next.data_ptr<int32_t>(), tsize, (int)mode,
-        out_pi.data_ptr<int64_t>(), out_bi.data_ptr<int64_t>(),
-        cursor.data_ptr<int64_t>(), current_stream());
+        out_pi.numel() > 0 ? out_pi.data_ptr<int64_t>() : nullptr,
+        out_bi.data_ptr<int64_t>(), cursor.data_ptr<int64_t>(),
+        current_stream());
   }
   return {out_pi, out_bi, cursor};
 }

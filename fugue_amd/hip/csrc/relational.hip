@@ -1798,6 +1798,9 @@ __global__ __launch_bounds__(BLOCK) void join_emit_unique_kernel(
 }
 
 // positional left join for unique build keys: out slot i = probe i
+// ILP-4: four interleaved chain walks per thread hide the dependent
+// random-load latency (heads -> bkeys/next) that bounds this kernel;
+// out_pi is optional — the compaction path emits indices itself.
 __global__ __launch_bounds__(BLOCK) void join_left_unique_kernel(
     const int64_t* __restrict__ pkeys, int64_t np,
     const int64_t* __restrict__ bkeys,
@@ -1806,21 +1809,47 @@ __global__ __launch_bounds__(BLOCK) void join_left_unique_kernel(
     int64_t tsize,
     int64_t* __restrict__ out_pi, int64_t* __restrict__ out_bi) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < np;
-       i += stride) {
-    int64_t key = pkeys[i];
-    uint64_t h = mix64((uint64_t)key);
-    int32_t cur = heads[(int64_t)(h & (uint64_t)(tsize - 1))];
-    int32_t match = -1;
-    while (cur >= 0) {
-      if (bkeys[cur] == key && (ph2 == nullptr || bh2[cur] == ph2[i])) {
-        match = cur;
-        break;
+  uint64_t tmask = (uint64_t)(tsize - 1);
+  for (int64_t base = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       base < np; base += stride * 4) {
+    int64_t idx[4], key[4], h2v[4];
+    int32_t cur[4], match[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      idx[j] = base + (int64_t)j * stride;
+      match[j] = -1;
+      if (idx[j] < np) {
+        key[j] = pkeys[idx[j]];
+        h2v[j] = ph2 != nullptr ? ph2[idx[j]] : 0;
+        cur[j] = heads[(int64_t)(mix64((uint64_t)key[j]) & tmask)];
+      } else {
+        cur[j] = -1;
       }
-      cur = next[cur];
     }
-    out_pi[i] = i;
-    out_bi[i] = match;
+    bool any = true;
+    while (any) {
+      any = false;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        if (cur[j] >= 0) {
+          int32_t c = cur[j];
+          if (bkeys[c] == key[j] && (ph2 == nullptr || bh2[c] == h2v[j])) {
+            match[j] = c;
+            cur[j] = -1;
+          } else {
+            cur[j] = next[c];
+            if (cur[j] >= 0) any = true;
+          }
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      if (idx[j] < np) {
+        if (out_pi != nullptr) out_pi[idx[j]] = idx[j];
+        out_bi[idx[j]] = match[j];
+      }
+    }
   }
 }
 
@@ -2050,7 +2079,7 @@ void launch_join_emit_unique(const int64_t* pkeys, int64_t np,
                              int64_t* out_pi, int64_t* out_bi,
                              int64_t* cursor, hipStream_t stream) {
   if (mode == 1) {
-    hipLaunchKernelGGL(join_left_unique_kernel, dim3(grid_for(np)),
+    hipLaunchKernelGGL(join_left_unique_kernel, dim3(grid_for(np, 4)),
                        dim3(BLOCK), 0, stream, pkeys, np, bkeys, ph2, bh2,
                        heads, next, tsize, out_pi, out_bi);
   } else {

@@ -904,7 +904,8 @@ def hash_join_indices(
         # unique — and merely discard this walk).
         np_ = int(probe_keys.numel())
         out_p, out_b, _cur = ext.join_emit_unique(
-            probe_keys, build_keys, probe_h2, build_h2, heads, nxt, 1
+            probe_keys, build_keys, probe_h2, build_h2, heads, nxt, 1,
+            how == "left",
         )
         if how == "left":
             if int(dup.item()) == 0:
