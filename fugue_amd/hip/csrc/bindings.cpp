@@ -32,7 +32,8 @@ void launch_gb_aggregate(const int64_t*, const double*, const bool*,
 void launch_join_emit_unique(const int64_t*, int64_t, const int64_t*,
                              const int64_t*, const int64_t*,
                              const int32_t*, const int32_t*, int64_t, int,
-                             int64_t*, int64_t*, int64_t*, hipStream_t);
+                             int64_t*, int64_t*, int64_t*, bool*, int,
+                             hipStream_t);
 void launch_join_build(const int64_t*, int64_t, int32_t*, int32_t*, int64_t,
                        int32_t*,
                        hipStream_t);
@@ -325,16 +326,19 @@ const int64_t* opt_i64_ptr(const c10::optional<at::Tensor>& v) {
 std::vector<at::Tensor> join_emit_unique(
     at::Tensor pkeys, at::Tensor bkeys, c10::optional<at::Tensor> ph2,
     c10::optional<at::Tensor> bh2, at::Tensor heads, at::Tensor next,
-    int64_t mode, bool want_pi) {
+    int64_t mode, bool want_pi, bool want_mask, bool mask_neg) {
   check_gpu(pkeys, "pkeys");
   int64_t np = pkeys.numel();
   int64_t tsize = heads.numel();
   // positional mode (1) with want_pi=false skips the pi write — the
-  // compaction kernel emits row indices itself
+  // compaction kernel emits row indices itself; want_mask emits the
+  // matched mask in the same pass (no separate compare kernel)
   auto out_pi = at::empty({(mode == 1 && !want_pi) ? 0 : np},
                           pkeys.options());
   auto out_bi = at::empty({np}, pkeys.options());
   auto cursor = at::zeros({1}, pkeys.options());
+  auto out_mask = at::empty({want_mask && mode == 1 ? np : 0},
+                            pkeys.options().dtype(at::kBool));
   if (np > 0) {
     launch_join_emit_unique(
         pkeys.data_ptr<int64_t>(), np, bkeys.data_ptr<int64_t>(),
@@ -342,9 +346,10 @@ std::vector<at::Tensor> join_emit_unique(
         next.data_ptr<int32_t>(), tsize, (int)mode,
         out_pi.numel() > 0 ? out_pi.data_ptr<int64_t>() : nullptr,
         out_bi.data_ptr<int64_t>(), cursor.data_ptr<int64_t>(),
-        current_stream());
+        out_mask.numel() > 0 ? out_mask.data_ptr<bool>() : nullptr,
+        mask_neg ? 1 : 0, current_stream());
   }
-  return {out_pi, out_bi, cursor};
+  return {out_pi, out_bi, cursor, out_mask};
 }
 
 at::Tensor join_count(at::Tensor pkeys, at::Tensor bkeys,

@@ -1807,7 +1807,8 @@ __global__ __launch_bounds__(BLOCK) void join_left_unique_kernel(
     const int64_t* __restrict__ ph2, const int64_t* __restrict__ bh2,
     const int32_t* __restrict__ heads, const int32_t* __restrict__ next,
     int64_t tsize,
-    int64_t* __restrict__ out_pi, int64_t* __restrict__ out_bi) {
+    int64_t* __restrict__ out_pi, int64_t* __restrict__ out_bi,
+    bool* __restrict__ out_mask, int mask_neg) {
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   uint64_t tmask = (uint64_t)(tsize - 1);
   for (int64_t base = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1848,6 +1849,8 @@ __global__ __launch_bounds__(BLOCK) void join_left_unique_kernel(
       if (idx[j] < np) {
         if (out_pi != nullptr) out_pi[idx[j]] = idx[j];
         out_bi[idx[j]] = match[j];
+        if (out_mask != nullptr)
+          out_mask[idx[j]] = mask_neg ? match[j] < 0 : match[j] >= 0;
       }
     }
   }
@@ -2077,11 +2080,13 @@ void launch_join_emit_unique(const int64_t* pkeys, int64_t np,
                              const int64_t* bh2, const int32_t* heads,
                              const int32_t* next, int64_t tsize, int mode,
                              int64_t* out_pi, int64_t* out_bi,
-                             int64_t* cursor, hipStream_t stream) {
+                             int64_t* cursor, bool* out_mask, int mask_neg,
+                             hipStream_t stream) {
   if (mode == 1) {
     hipLaunchKernelGGL(join_left_unique_kernel, dim3(grid_for(np, 4)),
                        dim3(BLOCK), 0, stream, pkeys, np, bkeys, ph2, bh2,
-                       heads, next, tsize, out_pi, out_bi);
+                       heads, next, tsize, out_pi, out_bi, out_mask,
+                       mask_neg);
   } else {
     hipLaunchKernelGGL(join_emit_unique_kernel, dim3(grid_for(np)),
                        dim3(BLOCK), 0, stream, pkeys, np, bkeys, ph2, bh2,

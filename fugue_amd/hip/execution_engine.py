@@ -27,8 +27,12 @@ from fugue_amd.collections.partition import (
 from fugue_amd.collections.sql import StructuredRawSQL
 from fugue_amd.column.expressions import (
     ColumnExpr,
+    _BinaryOpExpr,
+    _FuncExpr,
+    _LiteralColumnExpr,
     _NamedColumnExpr,
     _UnaryAggFuncExpr,
+    _UnaryOpExpr,
     col as col_expr,
 )
 from fugue_amd.column.sql import SelectColumns
@@ -1498,6 +1502,21 @@ class HipExecutionEngine(ExecutionEngine):
         cols = columns.replace_wildcard(d.schema)
         if where is not None:
             mask = filter_mask(where, d)
+            # compact only the columns the projection reads: predicate-
+            # only columns (strings especially) never pay the gather
+            if not cols.has_agg:
+                refs = _referenced_cols(cols)
+                if refs is not None and refs < set(d.schema.names):
+                    sub = Schema(
+                        [
+                            (f.name, f.type)
+                            for f in d.schema.fields
+                            if f.name in refs
+                        ]
+                    )
+                    d = HipDataFrame.from_columns(
+                        {n: d.col(n) for n in sub.names}, sub, self._device
+                    )
             d = self._filter_rows(d, mask)
         if not cols.has_agg:
             if cols.is_distinct:
@@ -2325,3 +2344,30 @@ class HipExecutionEngine(ExecutionEngine):
                     format_hint=format_hint, mode=mode, **kwargs
                 )
         self._comm.barrier()
+
+
+def _referenced_cols(cols: SelectColumns) -> Optional[set]:
+    """Column names a projection reads, or None when not statically
+    determinable (wildcards are already expanded by the caller)."""
+    out: set = set()
+
+    def walk(e: Any) -> bool:
+        if isinstance(e, _NamedColumnExpr):
+            if e.wildcard:
+                return False
+            out.add(e.name)
+            return True
+        if isinstance(e, _BinaryOpExpr):
+            return walk(e.left) and walk(e.right)
+        if isinstance(e, _UnaryOpExpr):
+            return walk(e.col)
+        if isinstance(e, _LiteralColumnExpr):
+            return True
+        if isinstance(e, _FuncExpr):
+            return all(walk(a) for a in e.args)
+        return False
+
+    for c in cols.all_cols:
+        if not walk(c):
+            return None
+    return out

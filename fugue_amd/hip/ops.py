@@ -903,15 +903,15 @@ def hash_join_indices(
         # (dup build keys are rare — dim joins and groupby outputs are
         # unique — and merely discard this walk).
         np_ = int(probe_keys.numel())
-        out_p, out_b, _cur = ext.join_emit_unique(
+        out_p, out_b, _cur, matched = ext.join_emit_unique(
             probe_keys, build_keys, probe_h2, build_h2, heads, nxt, 1,
-            how == "left",
+            how == "left", how != "left", how == "anti",
         )
         if how == "left":
             if int(dup.item()) == 0:
                 return out_p, out_b
         else:
-            matched = (out_b >= 0) if how != "anti" else (out_b < 0)
+            # matched mask came out of the walk kernel itself;
             # None column = "emit the row index" (no arange materialized)
             cols = [None] if how == "anti" else [None, out_b]
             outs = ext.compact_columns_cap(matched, cols)

@@ -341,12 +341,6 @@ def _execute_core(
                 remaining.append(conj)
             else:
                 pushed.setdefault(target, []).append(conj)
-        for idx, conjs in pushed.items():
-            fr = frames[idx]
-            for conj in conjs:
-                ce = _to_column_expr(conj, fr.schema, {})
-                fr = engine.filter(fr, ce)
-            frames[idx] = fr
         res = frames[0]
         join_inputs = frames[1:]
         residual_where = None
@@ -356,9 +350,10 @@ def _execute_core(
                 if residual_where is None
                 else X.BinOp("AND", residual_where, conj)
             )
-    # projection pushdown: drop base-table columns not referenced anywhere
-    # in the statement before joining (string columns in particular make
-    # join-output gathers expensive)
+    # projection pushdown fused with the pushed filters: each base table
+    # keeps only the columns referenced downstream, and a pushed filter
+    # runs as select(keep, where=...) so the compaction never gathers
+    # columns (strings especially) that the filter output drops
     if stmt.joins:
         frames2: List[DataFrame] = [res] + [
             join_inputs[i]
@@ -367,16 +362,31 @@ def _execute_core(
             for i in range(len(stmt.joins))
         ]
         needed = _needed_columns(stmt, residual_where)
-        if needed is not None:
-            pruned: List[DataFrame] = []
-            for fr in frames2:
+        pruned: List[DataFrame] = []
+        for idx, fr in enumerate(frames2):
+            conjs = pushed.get(idx, [])
+            where_ce = None
+            for conj in conjs:
+                ce = _to_column_expr(conj, fr.schema, {})
+                where_ce = ce if where_ce is None else (where_ce & ce)
+            if needed is not None:
                 keep = [n for n in fr.schema.names if n in needed]
-                if 0 < len(keep) < len(fr.schema.names):
-                    fr = engine.select(
-                        fr, SelectColumns(*[col(n) for n in keep])
-                    )
-                pruned.append(fr)
-            frames2 = pruned
+            else:
+                keep = list(fr.schema.names)
+            if len(keep) == 0:
+                keep = [fr.schema.names[0]]
+            if where_ce is not None:
+                fr = engine.select(
+                    fr,
+                    SelectColumns(*[col(n) for n in keep]),
+                    where=where_ce,
+                )
+            elif len(keep) < len(fr.schema.names):
+                fr = engine.select(
+                    fr, SelectColumns(*[col(n) for n in keep])
+                )
+            pruned.append(fr)
+        frames2 = pruned
         res = frames2[0]
         join_inputs = frames2[1:]
     for ji, j in enumerate(stmt.joins):
