@@ -257,10 +257,12 @@ def _key_stats_device(
 
 
 def _device_packable(key_cols: "Sequence[DeviceColumn]") -> bool:
+    # integer dtypes only: the kernels load raw values by width, while
+    # float keys go through the torch path's value-cast to int64
     return all(
         (not isinstance(c, StringDeviceColumn))
         and c.data.is_cuda
-        and c.data.element_size() in (2, 4, 8)
+        and c.data.dtype in (torch.int64, torch.int32, torch.int16)
         and c.data.is_contiguous()
         for c in key_cols
     )

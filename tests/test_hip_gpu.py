@@ -801,3 +801,15 @@ def test_take_presort_matches_pandas_gpu(engine):
     pd.testing.assert_frame_equal(
         res.as_pandas().reset_index(drop=True), exp, check_dtype=False
     )
+
+
+def test_float_group_keys_gpu(engine):
+    """Float group keys take the value-cast path, not the raw-bits pack
+    (regression: r02f test_select conformance failure)."""
+    pdf = pd.DataFrame(dict(a=[1.0, 1.0, 3.0, None, None], b=[1, 1, 4, 3, 4]))
+    res = fa.select(
+        pdf, col("a"), f.sum(col("b")).cast(float).alias("b"),
+        engine=engine, as_fugue=True,
+    )
+    got = sorted(res.as_array(), key=lambda r: (r[0] is None, r[0]))
+    assert got == [[1.0, 2.0], [3.0, 4.0], [None, 7.0]]
