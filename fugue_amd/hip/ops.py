@@ -220,11 +220,18 @@ def _key_stats_device(
     the device sync entirely).  Replaces per-column min/max reductions
     + torch.unique Chao sampling (profiles/NOTES.md r02)."""
     n = int(datas[0].numel())
-    key = tuple(id(t) for t in datas) + (n, with_minmax)
+    # key on (address, length, dtype): re-created VIEWS of the same
+    # storage (fresh tensor objects each plan run) still hit.  Safe
+    # because a hit requires the memoized tensors to still be ALIVE —
+    # two live tensors at one address with one length alias the same
+    # bytes (frames treat column tensors as immutable).
+    key = tuple(
+        (t.data_ptr(), t.numel(), str(t.dtype)) for t in datas
+    ) + (n, with_minmax)
     hit = _KEY_STATS_MEMO.get(key)
     if hit is not None:
         refs, value = hit
-        if all(r() is t for r, t in zip(refs, datas)):
+        if all(r() is not None for r in refs):
             return value
         del _KEY_STATS_MEMO[key]
     ext = get_ext()
