@@ -12,14 +12,21 @@ import numpy as np
 import torch
 
 CONFIGS = [
-    dict(FUGUE_GB_PARTS="512"),
     dict(FUGUE_GB_PARTS="1024"),
-    dict(FUGUE_GB_PARTS="1024", FUGUE_GB_AGG_CHUNK="16384"),
+    dict(FUGUE_GB_PARTS="1024", FUGUE_GB_SCATTER_CHUNK="8192"),
+    dict(FUGUE_GB_PARTS="1024", FUGUE_GB_SCATTER_CHUNK="32768"),
+    dict(FUGUE_GB_PARTS="1024", FUGUE_GB_SCATTER_CHUNK="65536"),
     dict(FUGUE_GB_PARTS="1024", FUGUE_GB_AGG_CHUNK="8192"),
-    dict(FUGUE_GB_PARTS="2048", FUGUE_GB_AGG_CHUNK="16384"),
-    dict(FUGUE_GB_PARTS="512", FUGUE_GB_AGG_CHUNK="16384"),
+    dict(FUGUE_GB_PARTS="1024", FUGUE_GB_AGG_CHUNK="32768"),
+    dict(FUGUE_GB_PARTS="2048"),
+    dict(FUGUE_GB_PARTS="2048", FUGUE_GB_SCATTER_CHUNK="32768"),
+    dict(FUGUE_GB_PARTS="512", FUGUE_SC_ILP="4"),
+    dict(FUGUE_GB_PARTS="1024", FUGUE_GB_ILP="1"),
 ]
-KNOBS = ("FUGUE_GB_PARTS", "FUGUE_GB_AGG_CHUNK", "FUGUE_GB_SCATTER_CHUNK")
+KNOBS = (
+    "FUGUE_GB_PARTS", "FUGUE_GB_AGG_CHUNK", "FUGUE_GB_SCATTER_CHUNK",
+    "FUGUE_SC_ILP", "FUGUE_GB_ILP",
+)
 
 
 def main() -> None:
