@@ -23,6 +23,10 @@ class FugueSQLWorkflow(FugueWorkflow):
         super().__init__(compile_conf)
         self._sql_vars: Dict[str, WorkflowDataFrame] = {}
         self._captured: Dict[str, Any] = {}
+        # (name -> guard) for every caller variable the script build
+        # consulted; the fugue_sql plan cache replays a built DAG only
+        # while every guard still holds (same object / value / absence)
+        self._var_guards: Dict[str, Any] = {}
 
     def __call__(self, code: str, *args: Any, **kwargs: Any) -> None:
         self._sql(code, *args, **kwargs)
@@ -38,7 +42,12 @@ class FugueSQLWorkflow(FugueWorkflow):
             for k, v in variables.items()
             if not self._is_dfable(v)
         }
-        code = fill_sql_template(code, template_vars)
+        rendered = fill_sql_template(code, template_vars)
+        if rendered != code:
+            # jinja-templated scripts depend on arbitrary rendered text;
+            # the plan cache must not replay them
+            self._var_guards["__template__"] = ("nocache",)
+        code = rendered
         self._captured.update(variables)
         parser = FugueSQLParser(code, self)
         parser.parse()
@@ -54,15 +63,34 @@ class FugueSQLWorkflow(FugueWorkflow):
     def set_var(self, name: str, df: WorkflowDataFrame) -> None:
         self._sql_vars[name] = df
 
+    def _touch(self, name: str) -> None:
+        if name in self._sql_vars or name in self._var_guards:
+            return
+        if name in self._captured:
+            v = self._captured[name]
+            if isinstance(v, (str, int, float, bool, bytes)):
+                self._var_guards[name] = ("val", type(v), v)
+            else:
+                import weakref as _wr
+
+                try:
+                    self._var_guards[name] = ("ref", _wr.ref(v))
+                except TypeError:
+                    self._var_guards[name] = ("nocache",)
+        else:
+            self._var_guards[name] = ("absent",)
+
     def has_var(self, name: str) -> bool:
         if name in self._sql_vars:
             return True
+        self._touch(name)
         v = self._captured.get(name)
         return v is not None and self._is_dfable(v)
 
     def get_var(self, name: str) -> WorkflowDataFrame:
         if name in self._sql_vars:
             return self._sql_vars[name]
+        self._touch(name)
         v = self._captured.get(name)
         if v is not None and self._is_dfable(v):
             wdf = self.df(v)
@@ -151,6 +179,8 @@ class FugueSQLWorkflow(FugueWorkflow):
         )
 
     def _resolve_ext(self, ext: Any) -> Any:
+        if isinstance(ext, str):
+            self._touch(ext)
         if isinstance(ext, str) and ext in self._captured:
             return self._captured[ext]
         return ext
@@ -158,6 +188,8 @@ class FugueSQLWorkflow(FugueWorkflow):
     def _resolve_callback(self, callback: Any) -> Any:
         if callback is None:
             return None
+        if isinstance(callback, str):
+            self._touch(callback)
         if isinstance(callback, str) and callback in self._captured:
             return self._captured[callback]
         return callback

@@ -612,6 +612,13 @@ class FugueWorkflow:
         for identical workflows (reference FugueWorkflow.spec_uuid)."""
         return to_uuid([t.__uuid__() for t in self._tasks.values()])
 
+    def reset_execution(self) -> None:
+        """Clear task results so a built DAG can run again (used by the
+        fugue_sql plan cache: construction is reused, execution is not)."""
+        for t in self._tasks.values():
+            t._executed = False
+            t._result = None
+
     @property
     def last_df(self) -> Optional[WorkflowDataFrame]:
         return self._last_df

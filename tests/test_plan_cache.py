@@ -1,0 +1,94 @@
+"""fugue_sql plan cache: construction is reused across identical calls,
+execution is not; any consulted-variable change invalidates."""
+import pandas as pd
+import pytest
+
+import fugue_amd.api as fa
+from fugue_amd.sql import api as sql_api
+
+
+@pytest.fixture(autouse=True)
+def _fresh_cache():
+    sql_api.clear_plan_cache()
+    yield
+    sql_api.clear_plan_cache()
+
+
+Q = """
+t = TRANSFORM df USING tr SCHEMA a:long,b:long
+SELECT a, SUM(b) AS s FROM t GROUP BY a
+"""
+
+
+# schema: a:long,b:long
+def _tr(pdf: pd.DataFrame) -> pd.DataFrame:
+    return pdf.assign(b=pdf.b + 1)
+
+
+def test_repeated_call_reexecutes():
+    df = pd.DataFrame(dict(a=[0, 0, 1], b=[1, 2, 3]))
+    calls = []
+
+    def tr(pdf: pd.DataFrame) -> pd.DataFrame:
+        calls.append(1)
+        return pdf.assign(b=pdf.b + 1)
+
+    tr.__annotations__ = _tr.__annotations__
+    tr._schema_hint = None
+    r1 = fa.fugue_sql(Q.replace("SCHEMA a:long,b:long", "SCHEMA *"),
+                      df=df, tr=tr)
+    n1 = len(calls)
+    r2 = fa.fugue_sql(Q.replace("SCHEMA a:long,b:long", "SCHEMA *"),
+                      df=df, tr=tr)
+    assert len(calls) > n1, "cached plan must still execute the UDF"
+    pd.testing.assert_frame_equal(
+        r1.sort_values("a").reset_index(drop=True),
+        r2.sort_values("a").reset_index(drop=True),
+    )
+
+
+def test_cache_hit_skips_rebuild_and_matches():
+    df = pd.DataFrame(dict(a=[0, 1, 1], b=[5, 6, 7]))
+    r1 = fa.fugue_sql("SELECT a, SUM(b) AS s FROM df GROUP BY a", df=df)
+    assert len(sql_api._PLAN_CACHE) == 1
+    r2 = fa.fugue_sql("SELECT a, SUM(b) AS s FROM df GROUP BY a", df=df)
+    entries = next(iter(sql_api._PLAN_CACHE.values()))
+    assert len(entries) == 1  # replayed, not re-stored
+    pd.testing.assert_frame_equal(r1, r2)
+
+
+def test_new_frame_object_invalidates():
+    df1 = pd.DataFrame(dict(a=[0], b=[1]))
+    df2 = pd.DataFrame(dict(a=[0], b=[100]))
+    r1 = fa.fugue_sql("SELECT a, SUM(b) AS s FROM df GROUP BY a", df=df1)
+    r2 = fa.fugue_sql("SELECT a, SUM(b) AS s FROM df GROUP BY a", df=df2)
+    assert r1.s.tolist() == [1]
+    assert r2.s.tolist() == [100]
+
+
+def test_templated_scripts_not_cached():
+    df = pd.DataFrame(dict(a=[0, 1], b=[1, 2]))
+    r1 = fa.fugue_sql(
+        "SELECT a, SUM(b) AS s FROM df GROUP BY a HAVING SUM(b) > {{lim}}",
+        df=df, lim=0,
+    )
+    r2 = fa.fugue_sql(
+        "SELECT a, SUM(b) AS s FROM df GROUP BY a HAVING SUM(b) > {{lim}}",
+        df=df, lim=1,
+    )
+    assert len(r1) == 2 and len(r2) == 1
+    assert len(sql_api._PLAN_CACHE) == 0
+
+
+def test_plan_cache_disabled_flag():
+    df = pd.DataFrame(dict(a=[0], b=[1]))
+    fa.fugue_sql("SELECT a, SUM(b) AS s FROM df GROUP BY a", df=df,
+                 plan_cache=False)
+    assert len(sql_api._PLAN_CACHE) == 0
+
+
+def test_immutable_binding_change_invalidates():
+    df = pd.DataFrame(dict(a=["x", "y"], b=[1, 2]))
+    r1 = fa.fugue_sql("SELECT * FROM df WHERE a = 'x'", df=df)
+    r2 = fa.fugue_sql("SELECT * FROM df WHERE a = 'y'", df=df)
+    assert r1.a.tolist() == ["x"] and r2.a.tolist() == ["y"]
