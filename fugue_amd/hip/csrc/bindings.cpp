@@ -89,6 +89,10 @@ void launch_pack_cols(const void**, const bool**, const int*, const int64_t*,
                       const int*, int, int64_t, int64_t*, hipStream_t);
 void launch_unpack_col(const int64_t*, int64_t, int, int64_t, int64_t, int,
                        int, void*, bool*, hipStream_t);
+void launch_joinoa_build(const int64_t*, int64_t, int64_t*, int64_t,
+                         int64_t*, hipStream_t);
+void launch_joinoa_probe(const int64_t*, int64_t, const int64_t*, int64_t,
+                         int64_t*, int64_t*, bool*, int, hipStream_t);
 void launch_topk(const void*, int, int, int64_t, int64_t, void*, int64_t*,
                  void*, int64_t*, hipStream_t);
 void launch_eq2_mask(const int64_t*, const int64_t*, const int64_t*,
@@ -548,6 +552,40 @@ int cmp_dtype_code(const at::Tensor& t) {
 
 
 
+
+// Open-addressed unique join for int64 keys (no h2): one 16B (key,idx)
+// entry per slot — ~1 random cache line per probe vs ~3 for the
+// chained layout.  flags = [dup_build_key, sentinel_key_seen]; either
+// nonzero means the caller must fall back to the chained join.
+std::vector<at::Tensor> join_open_unique(at::Tensor pkeys, at::Tensor bkeys,
+                                         bool want_pi, bool want_mask,
+                                         bool mask_neg) {
+  check_gpu(pkeys, "pkeys");
+  check_gpu(bkeys, "bkeys");
+  int64_t np = pkeys.numel();
+  int64_t nb = bkeys.numel();
+  int64_t tsize = 16;
+  while (tsize < nb * 2) tsize <<= 1;
+  auto table = at::empty({2 * tsize}, bkeys.options());
+  auto flags = at::zeros({2}, bkeys.options());
+  launch_joinoa_build(nb > 0 ? bkeys.data_ptr<int64_t>() : nullptr, nb,
+                      table.data_ptr<int64_t>(), tsize,
+                      flags.data_ptr<int64_t>(), current_stream());
+  auto out_pi = at::empty({want_pi ? np : 0}, pkeys.options());
+  auto out_bi = at::empty({np}, pkeys.options());
+  auto out_mask =
+      at::empty({want_mask ? np : 0}, pkeys.options().dtype(at::kBool));
+  if (np > 0) {
+    launch_joinoa_probe(
+        pkeys.data_ptr<int64_t>(), np, table.data_ptr<int64_t>(), tsize,
+        want_pi ? out_pi.data_ptr<int64_t>() : nullptr,
+        out_bi.data_ptr<int64_t>(),
+        want_mask ? out_mask.data_ptr<bool>() : nullptr, mask_neg ? 1 : 0,
+        current_stream());
+  }
+  return {out_pi, out_bi, out_mask, flags};
+}
+
 // Own top-k select for k <= 16 (see relational.hip); returns
 // (values[k], indices[k]) sorted, index-tiebroken (deterministic).
 std::vector<at::Tensor> topk_select(at::Tensor vals, int64_t k,
@@ -916,6 +954,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "deterministic group-table compaction");
   m.def("compact_columns_cap", &compact_columns_cap,
         "mask compaction, capacity outputs + cursor");
+  m.def("join_open_unique", &join_open_unique,
+        "open-addressed unique int join (build+probe)");
   m.def("topk_select", &topk_select, "own top-k (k<=16) select");
   m.def("eq2_mask", &eq2_mask, "128-bit string equality mask");
   m.def("cmp_imm", &cmp_imm, "single col-vs-literal comparison mask");

@@ -3096,3 +3096,130 @@ void launch_topk(const void* vals, int dtype, int largest, int64_t n, int k,
                        vals, n, k, cand_v, cand_i, out_v, out_i);
 }
 }  // extern "C"
+
+// ------------------------------------------------------------------ //
+// open-addressed unique join (int64 keys, no h2)                      //
+//                                                                     //
+// The chained layout costs ~3 random cache lines per probe (heads ->  //
+// bkeys -> next); a 16-byte (key, idx) entry costs ~1 line at 50%     //
+// load factor.  Reserved key sentinel: a build key equal to it sets   //
+// a flag and the caller falls back to the chained join (read in the   //
+// same single sync as the dup flag).                                  //
+// ------------------------------------------------------------------ //
+
+#define JOA_EMPTY 0x8000000000000000LL
+
+__global__ __launch_bounds__(BLOCK) void joinoa_init_kernel(
+    int64_t* __restrict__ table, int64_t tsize) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < tsize;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    table[2 * i] = JOA_EMPTY;
+  }
+}
+
+__global__ __launch_bounds__(BLOCK) void joinoa_build_kernel(
+    const int64_t* __restrict__ bkeys, int64_t nb,
+    int64_t* __restrict__ table, int64_t tsize,
+    int64_t* __restrict__ flags) {  // flags[0]=dup, flags[1]=sentinel
+  uint64_t tmask = (uint64_t)(tsize - 1);
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nb;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t k = bkeys[i];
+    if (k == JOA_EMPTY) {
+      flags[1] = 1;  // racy write of constant is fine
+      continue;
+    }
+    uint64_t h = mix64((uint64_t)k) & tmask;
+    for (;;) {
+      int64_t prev = (int64_t)atomicCAS((unsigned long long*)&table[2 * h],
+                                        (unsigned long long)JOA_EMPTY,
+                                        (unsigned long long)k);
+      if (prev == JOA_EMPTY) {
+        table[2 * h + 1] = i;  // idx read only after this kernel completes
+        break;
+      }
+      if (prev == k) {
+        flags[0] = 1;  // duplicate build key
+        break;
+      }
+      h = (h + 1) & tmask;
+    }
+  }
+}
+
+// ILP-4 positional probe: out_bi[i] = build idx or -1; optional pi/mask
+__global__ __launch_bounds__(BLOCK) void joinoa_probe_kernel(
+    const int64_t* __restrict__ pkeys, int64_t np,
+    const int64_t* __restrict__ table, int64_t tsize,
+    int64_t* __restrict__ out_pi, int64_t* __restrict__ out_bi,
+    bool* __restrict__ out_mask, int mask_neg) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  uint64_t tmask = (uint64_t)(tsize - 1);
+  for (int64_t base = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       base < np; base += stride * 4) {
+    int64_t idx[4], key[4], match[4];
+    uint64_t h[4];
+    bool act[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      idx[j] = base + (int64_t)j * stride;
+      match[j] = -1;
+      act[j] = idx[j] < np;
+      if (act[j]) {
+        key[j] = pkeys[idx[j]];
+        h[j] = mix64((uint64_t)key[j]) & tmask;
+      }
+    }
+    bool any = true;
+    while (any) {
+      any = false;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        if (act[j]) {
+          int64_t k = table[2 * h[j]];
+          if (k == key[j]) {
+            match[j] = table[2 * h[j] + 1];
+            act[j] = false;
+          } else if (k == JOA_EMPTY) {
+            act[j] = false;
+          } else {
+            h[j] = (h[j] + 1) & tmask;
+            any = true;
+          }
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      if (idx[j] < np) {
+        if (out_pi != nullptr) out_pi[idx[j]] = idx[j];
+        out_bi[idx[j]] = match[j];
+        if (out_mask != nullptr)
+          out_mask[idx[j]] = mask_neg ? match[j] < 0 : match[j] >= 0;
+      }
+    }
+  }
+}
+
+extern "C" {
+
+void launch_joinoa_build(const int64_t* bkeys, int64_t nb, int64_t* table,
+                         int64_t tsize, int64_t* flags, hipStream_t stream) {
+  hipLaunchKernelGGL(joinoa_init_kernel, dim3(grid_for(tsize, 2)),
+                     dim3(BLOCK), 0, stream, table, tsize);
+  if (nb > 0) {
+    hipLaunchKernelGGL(joinoa_build_kernel, dim3(grid_for(nb)), dim3(BLOCK),
+                       0, stream, bkeys, nb, table, tsize, flags);
+  }
+}
+
+void launch_joinoa_probe(const int64_t* pkeys, int64_t np,
+                         const int64_t* table, int64_t tsize,
+                         int64_t* out_pi, int64_t* out_bi, bool* out_mask,
+                         int mask_neg, hipStream_t stream) {
+  hipLaunchKernelGGL(joinoa_probe_kernel, dim3(grid_for(np, 4)), dim3(BLOCK),
+                     0, stream, pkeys, np, table, tsize, out_pi, out_bi,
+                     out_mask, mask_neg);
+}
+
+}  // extern "C"

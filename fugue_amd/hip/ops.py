@@ -897,12 +897,49 @@ def hash_join_indices(
     if how not in ("inner", "left", "semi", "anti"):
         raise FugueBug(f"unsupported join mode {how}")
     nb = int(build_keys.numel())
+    import os as _os_j
+
+    skip_unique_walk = False
+    if probe_h2 is None and _os_j.environ.get("FUGUE_JOIN_OA", "1") != "0":
+        # open-addressed unique join: ~1 random cache line per probe
+        # (16B key+idx entries) vs ~3 for the chained layout.  Build
+        # duplicates or a sentinel-valued key (flags) fall back to the
+        # chained join below; the check shares the single host sync
+        # with the compaction total.
+        np_ = int(probe_keys.numel())
+        out_p, out_b, matched, flags = ext.join_open_unique(
+            probe_keys, build_keys, how == "left", how != "left",
+            how == "anti",
+        )
+        if how == "left":
+            fl = flags.cpu()
+            if int(fl[0]) == 0 and int(fl[1]) == 0:
+                return out_p, out_b
+            skip_unique_walk = int(fl[0]) != 0 and int(fl[1]) == 0
+        else:
+            cols = [None] if how == "anti" else [None, out_b]
+            outs = ext.compact_columns_cap(matched, cols)
+            vals = torch.cat([flags, outs[-1].reshape(1)]).cpu()
+            if int(vals[0]) == 0 and int(vals[1]) == 0:
+                total = int(vals[2])
+                if how == "anti":
+                    pi = outs[0].narrow(0, 0, total)
+                    return pi, torch.full_like(pi, -1)
+                return (
+                    outs[0].narrow(0, 0, total),
+                    outs[1].narrow(0, 0, total),
+                )
+            # dup with no sentinel: the chained unique walk would fail
+            # the same way — go straight to the duplicate-emit path
+            skip_unique_walk = int(vals[0]) != 0 and int(vals[1]) == 0
+
     tsize = _next_pow2(max(16, nb * 2))
     heads, nxt, dup = ext.join_build(build_keys, tsize)
     mode = {"inner": 0, "left": 1, "semi": 2, "anti": 3}[how]
-    import os as _os_j
 
-    if _os_j.environ.get("FUGUE_JOIN_UNIQUE", "1") != "0":
+    if not skip_unique_walk and _os_j.environ.get(
+        "FUGUE_JOIN_UNIQUE", "1"
+    ) != "0":
         # unique build keys (≤1 match per probe): ONE chain walk writes
         # the match index positionally; inner/semi/anti then compact
         # with the block-scan compaction kernel (no global-cursor

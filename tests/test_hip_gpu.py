@@ -813,3 +813,59 @@ def test_float_group_keys_gpu(engine):
     )
     got = sorted(res.as_array(), key=lambda r: (r[0] is None, r[0]))
     assert got == [[1.0, 2.0], [3.0, 4.0], [None, 7.0]]
+
+
+def test_open_addressed_join_gpu(engine):
+    """OA unique join agrees with pandas for all modes, incl. the
+    sentinel-key (INT64_MIN) and duplicate-build fallbacks."""
+    import torch
+
+    rng = np.random.default_rng(31)
+    n = 200_000
+    left = pd.DataFrame(dict(k=rng.integers(0, 50_000, n), v=rng.random(n)))
+    right = pd.DataFrame(
+        dict(k=np.arange(0, 60_000, 2), w=rng.random(30_000))
+    )
+    for how in ("inner", "left_outer", "semi", "anti"):
+        got = (
+            fa.join(left, right, how=how, engine=engine, as_fugue=True)
+            .as_pandas()
+            .sort_values(["k", "v"])
+            .reset_index(drop=True)
+        )
+        pandas_how = {
+            "inner": "inner", "left_outer": "left",
+        }.get(how)
+        if pandas_how:
+            exp = left.merge(right, on="k", how=pandas_how)
+        elif how == "semi":
+            exp = left[left.k.isin(right.k)]
+        else:
+            exp = left[~left.k.isin(right.k)]
+        exp = exp.sort_values(["k", "v"]).reset_index(drop=True)
+        pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+    # sentinel key in the build side forces the chained fallback
+    right2 = pd.DataFrame(
+        dict(k=np.array([-(2**63), 0, 2], dtype=np.int64), w=[1.0, 2.0, 3.0])
+    )
+    left2 = pd.DataFrame(
+        dict(k=np.array([-(2**63), 1, 2], dtype=np.int64), v=[9.0, 8.0, 7.0])
+    )
+    got = (
+        fa.join(left2, right2, how="inner", engine=engine, as_fugue=True)
+        .as_pandas().sort_values("k").reset_index(drop=True)
+    )
+    exp = left2.merge(right2, on="k").sort_values("k").reset_index(drop=True)
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+    # duplicate build keys fall back to the chained duplicate emit
+    right3 = pd.DataFrame(dict(k=[1, 1, 2], w=[1.0, 2.0, 3.0]))
+    got = (
+        fa.join(left2, right3, how="inner", engine=engine, as_fugue=True)
+        .as_pandas().sort_values(["k", "w"]).reset_index(drop=True)
+    )
+    exp = left2.merge(right3, on="k").sort_values(["k", "w"]).reset_index(
+        drop=True
+    )
+    pd.testing.assert_frame_equal(got, exp, check_dtype=False)
