@@ -900,12 +900,13 @@ def hash_join_indices(
     import os as _os_j
 
     skip_unique_walk = False
-    if probe_h2 is None and _os_j.environ.get("FUGUE_JOIN_OA", "1") != "0":
-        # open-addressed unique join: ~1 random cache line per probe
-        # (16B key+idx entries) vs ~3 for the chained layout.  Build
-        # duplicates or a sentinel-valued key (flags) fall back to the
-        # chained join below; the check shares the single host sync
-        # with the compaction total.
+    if probe_h2 is None and _os_j.environ.get("FUGUE_JOIN_OA", "0") == "1":
+        # open-addressed unique join (16B key+idx entries, ~1 random
+        # cache line per probe).  MEASURED SLOWER than chains on q3
+        # (4.36 vs 4.22 ms/step): the doubled table footprint falls out
+        # of the 256MB Infinity Cache while the chained layout fits —
+        # kept env-gated for small-build workloads.  Build duplicates /
+        # sentinel keys fall back to the chained join below.
         np_ = int(probe_keys.numel())
         out_p, out_b, matched, flags = ext.join_open_unique(
             probe_keys, build_keys, how == "left", how != "left",
