@@ -815,10 +815,13 @@ def test_float_group_keys_gpu(engine):
     assert got == [[1.0, 2.0], [3.0, 4.0], [None, 7.0]]
 
 
-def test_open_addressed_join_gpu(engine):
-    """OA unique join agrees with pandas for all modes, incl. the
-    sentinel-key (INT64_MIN) and duplicate-build fallbacks."""
+def test_open_addressed_join_gpu(engine, monkeypatch):
+    """OA unique join (env-gated variant) agrees with pandas for all
+    modes, incl. the sentinel-key (INT64_MIN) and duplicate-build
+    fallbacks."""
     import torch
+
+    monkeypatch.setenv("FUGUE_JOIN_OA", "1")
 
     rng = np.random.default_rng(31)
     n = 200_000
