@@ -18,6 +18,7 @@ import numpy as np
 import pandas as pd
 import pyarrow as pa
 import torch
+import weakref
 
 from fugue_amd.collections.partition import (
     PartitionCursor,
@@ -531,6 +532,14 @@ class HipExecutionEngine(ExecutionEngine):
             res.reset_metadata(src.metadata)
         return res
 
+    @property
+    def _global_bytes_memo(self) -> Dict[int, Any]:
+        m = getattr(self, "_gb_memo", None)
+        if m is None:
+            m = {}
+            self._gb_memo = m
+        return m
+
     def _gather_all(self, df: HipDataFrame) -> LocalDataFrame:
         """Replicate the full (all-rank) contents locally as an arrow-backed
         frame (fallback path / SQL facet)."""
@@ -990,7 +999,24 @@ class HipExecutionEngine(ExecutionEngine):
         raise NotImplementedError(how)
 
     def _global_bytes(self, df: HipDataFrame) -> int:
-        return self._comm.allreduce_sum(df.num_bytes())
+        # one allreduce per distinct frame: repeated plans (plan cache,
+        # iterating drivers) re-ask for the same persistent frame every
+        # step — memoized by object identity (frames are immutable).
+        # SPMD-symmetric by construction: every rank runs the same
+        # program, so hits/misses (and FIFO evictions) line up and the
+        # collective is entered by all ranks or none
+        key = id(df)
+        hit = self._global_bytes_memo.get(key)
+        if hit is not None and hit[0]() is df:
+            return hit[1]
+        val = self._comm.allreduce_sum(df.num_bytes())
+        if len(self._global_bytes_memo) >= 64:
+            self._global_bytes_memo.pop(next(iter(self._global_bytes_memo)))
+        try:
+            self._global_bytes_memo[key] = (weakref.ref(df), val)
+        except TypeError:
+            pass
+        return val
 
     def _replicate(self, df: HipDataFrame) -> HipDataFrame:
         local = self._gather_all(df)
