@@ -1019,8 +1019,51 @@ class HipExecutionEngine(ExecutionEngine):
         return val
 
     def _replicate(self, df: HipDataFrame) -> HipDataFrame:
-        local = self._gather_all(df)
-        return HipDataFrame(local.as_arrow(), df.schema, device=self._device)
+        """Replicate a (small) frame to every rank.  Device-resident:
+        each rank tiles its shard world_size times and the standard
+        all-to-all exchange delivers every shard to every rank in rank
+        order — no host pickle round-trip.  Memoized per frame identity
+        (SPMD-symmetric, same argument as _global_bytes): a broadcast
+        join against a persistent dimension table pays the collective
+        once, not per step."""
+        key = id(df)
+        hit = self._replicate_memo.get(key)
+        if hit is not None and hit[0]() is df:
+            return hit[1]
+        if not self.is_distributed:
+            res = df
+        else:
+            n = df.count()
+            world = self.world_size
+            dev = torch.device(self._device)
+            if df.count() == 0:
+                tiled = df
+            else:
+                idx = torch.arange(
+                    n, dtype=torch.int64, device=dev
+                ).repeat(world)
+                tiled = df.gather_rows(idx)
+            res = self._exchange(
+                tiled,
+                torch.full((world,), n, dtype=torch.int64, device=dev),
+            )
+            res.reset_metadata(dict(df.metadata) if df.has_metadata else {})
+            res.metadata["broadcasted"] = True
+        if len(self._replicate_memo) >= 8:
+            self._replicate_memo.pop(next(iter(self._replicate_memo)))
+        try:
+            self._replicate_memo[key] = (weakref.ref(df), res)
+        except TypeError:
+            pass
+        return res
+
+    @property
+    def _replicate_memo(self) -> Dict[int, Any]:
+        m = getattr(self, "_rep_memo", None)
+        if m is None:
+            m = {}
+            self._rep_memo = m
+        return m
 
     def _split_null_keys(
         self, df: HipDataFrame, keys: List[str]
