@@ -194,7 +194,13 @@ class HipMapEngine(MapEngine):
                     start, end = bounds_d[p], bounds_d[p + 1]
                     if end <= start:
                         continue
-                    sub = local.slice_rows(start, end - start)
+                    # full-range "slice" must preserve column tensor
+                    # identity: downstream sizing memos key on it
+                    sub = (
+                        local
+                        if start == 0 and end == n_rows
+                        else local.slice_rows(start, end - start)
+                    )
                     cursor.set(
                         lambda: sub.peek_array(),
                         engine.rank * n_local_parts + p,

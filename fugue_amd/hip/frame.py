@@ -575,6 +575,8 @@ class HipDataFrame(LocalBoundedDataFrame):
         return cols
 
     def slice_rows(self, start: int, length: int) -> "HipDataFrame":
+        if start == 0 and length >= self.count():
+            return self  # preserve column tensor identity (sizing memos)
         cols = {n: c.slice(start, length) for n, c in self._cols.items()}
         return HipDataFrame.from_columns(cols, self.schema, self._device)
 
