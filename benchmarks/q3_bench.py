@@ -218,6 +218,14 @@ def main() -> None:
             flush=True,
         )
 
+    # orderly distributed teardown: a rank exiting while peers still
+    # hold gloo/NCCL state can SIGABRT in the transport destructor
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.barrier()
+        dist.destroy_process_group()
+
 
 if __name__ == "__main__":
     main()
