@@ -42,7 +42,10 @@ void launch_gb_part_hist(const int64_t*, int64_t, int, int64_t*, int,
 void launch_reduce_cols(const double*, const bool*, const int32_t*, int,
                         int64_t, double*, int64_t*, hipStream_t);
 void launch_gb_part_scatter(const int64_t*, const double*, int, int64_t, int,
-                            int64_t*, int64_t*, double*, int, hipStream_t);
+                            int64_t*, int64_t*, double*, int, int32_t*,
+                            hipStream_t);
+void launch_scatter_by_pos(const double*, const int32_t*, int64_t, double*,
+                           hipStream_t);
 void launch_gb_aggregate_part(const int64_t*, const double*, const int32_t*,
                               int, int64_t, const int64_t*, int64_t, int64_t*,
                               double*, int64_t*, int64_t, hipStream_t);
@@ -226,7 +229,7 @@ std::vector<at::Tensor> gb_aggregate(at::Tensor keys, at::Tensor vals,
 std::vector<at::Tensor> gb_aggregate_partitioned(
     at::Tensor keys, at::Tensor vals, at::Tensor ops, int64_t num_parts,
     int64_t tsize, int64_t scatter_chunk, int64_t agg_chunk, int64_t nt,
-    int64_t narrow) {
+    int64_t narrow, bool record_layout) {
   check_gpu(keys, "keys");
   check_gpu(vals, "vals");
   TORCH_CHECK((tsize & (tsize - 1)) == 0, "tsize must be a power of 2");
@@ -247,6 +250,11 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
   bool staged =
       (n_aggs == 1 &&
        (num_parts == 512 || num_parts == 1024 || num_parts == 2048));
+  // layout recording goes through the simple scatter (it has each
+  // row's spill position in hand); replay restores the staged-path
+  // spill format (narrow keys done by the caller)
+  record_layout = record_layout && n_aggs == 1 && n < (int64_t(1) << 31);
+  if (record_layout) staged = false;
   // phase-3 LDS table sized for ~0.5 load at the partition granularity:
   // 512 parts -> 4096 slots, 1024 -> 2048, 2048 -> 1024
   int slots = num_parts == 1024 ? 2048 : (num_parts == 2048 ? 1024 : 4096);
@@ -268,6 +276,8 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
                    : at::empty({n}, keys.options());
   auto pvals = at::empty({n_aggs, n}, vals.options());
   auto ovf = at::zeros({1}, keys.options().dtype(at::kInt));
+  auto pos = at::empty({record_layout ? n : 0},
+                       keys.options().dtype(at::kInt));
   if (staged) {
     launch_gb_part_scatter_staged(
         keys.data_ptr<int64_t>(), vals.data_ptr<double>(), n, shift,
@@ -280,7 +290,9 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
     launch_gb_part_scatter(keys.data_ptr<int64_t>(), vals.data_ptr<double>(),
                            n_aggs, n, shift, cursor.data_ptr<int64_t>(),
                            pkeys.data_ptr<int64_t>(), pvals.data_ptr<double>(),
-                           (int)num_parts, stream);
+                           (int)num_parts,
+                           record_layout ? pos.data_ptr<int32_t>() : nullptr,
+                           stream);
   }
   // phase 3: per-partition LDS aggregation into the global table
   auto tkeys = at::full({tsize}, (int64_t)0x8000000000000000LL,
@@ -300,7 +312,37 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
         num_parts, tkeys.data_ptr<int64_t>(), gaggs.data_ptr<double>(),
         gcount.data_ptr<int64_t>(), tsize, stream);
   }
-  return {tkeys, gaggs, gcount, ovf};
+  return {tkeys, gaggs, gcount, ovf, pkeys, pos};
+}
+
+// Phase 2+3 only, with a recorded layout: place the fresh values at
+// the recorded spill positions and aggregate against the cached
+// partitioned keys (phase 1 + key scatter skipped entirely).
+std::vector<at::Tensor> gb_aggregate_replay(at::Tensor pkeys_cached,
+                                            at::Tensor pos, at::Tensor vals,
+                                            at::Tensor ops, int64_t tsize,
+                                            int64_t agg_chunk, int64_t nt,
+                                            int64_t slots) {
+  check_gpu(pkeys_cached, "pkeys");
+  check_gpu(pos, "pos");
+  check_gpu(vals, "vals");
+  TORCH_CHECK(vals.size(0) == 1, "replay supports single-agg spills");
+  int64_t n = pos.numel();
+  auto stream = current_stream();
+  auto pvals = at::empty({1, n}, vals.options());
+  launch_scatter_by_pos(vals.data_ptr<double>(), pos.data_ptr<int32_t>(), n,
+                        pvals.data_ptr<double>(), stream);
+  auto tkeys = at::full({tsize}, (int64_t)0x8000000000000000LL,
+                        pos.options().dtype(at::kLong));
+  auto gaggs = at::zeros({1, tsize}, vals.options());
+  auto gcount = at::zeros({tsize}, pos.options().dtype(at::kLong));
+  launch_gb_aggregate_part_big(
+      pkeys_cached.data_ptr(), pvals.data_ptr<double>(),
+      ops.data_ptr<int32_t>(), n, tkeys.data_ptr<int64_t>(),
+      gaggs.data_ptr<double>(), gcount.data_ptr<int64_t>(), tsize,
+      agg_chunk, (int)nt, pkeys_cached.element_size() == 4 ? 1 : 0,
+      (int)slots, stream);
+  return {tkeys, gaggs, gcount};
 }
 
 std::vector<at::Tensor> join_build(at::Tensor keys, int64_t tsize) {
@@ -930,6 +972,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bucket_scatter", &bucket_scatter,
         "scatter row indices into bucket-contiguous order");
   m.def("gb_aggregate", &gb_aggregate, "hash group-by aggregation");
+  m.def("gb_aggregate_replay", &gb_aggregate_replay,
+        "re-aggregate with a recorded shuffle layout");
   m.def("gb_aggregate_partitioned", &gb_aggregate_partitioned,
         "partitioned (2-phase) hash group-by aggregation");
   m.def("join_build", &join_build, "build chained hash table");

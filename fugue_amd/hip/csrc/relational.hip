@@ -519,7 +519,8 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_kernel(
     int64_t* __restrict__ cursor,      // [P] exclusive offsets (mutated)
     int64_t* __restrict__ out_keys,
     double* __restrict__ out_vals,
-    int num_parts) {
+    int num_parts,
+    int32_t* __restrict__ out_pos) {  // optional layout record
   __shared__ int lhist[PART_MAX];
   __shared__ int64_t lbase[PART_MAX];
   for (int64_t start = (int64_t)blockIdx.x * SCATTER_CHUNK; start < n;
@@ -548,6 +549,7 @@ __global__ __launch_bounds__(BLOCK) void gb_part_scatter_kernel(
       int p = (int)(mix64((uint64_t)key) >> shift);
       int64_t pos = lbase[p] + atomicAdd(&lhist[p], 1);
       out_keys[pos] = key;
+      if (out_pos != nullptr) out_pos[i] = (int32_t)pos;
       for (int a = 0; a < n_aggs; ++a)
         out_vals[(int64_t)a * n + pos] = vals[(int64_t)a * n + i];
     }
@@ -1662,13 +1664,14 @@ void launch_gb_part_hist(const int64_t* keys, int64_t n, int shift,
 void launch_gb_part_scatter(const int64_t* keys, const double* vals,
                             int n_aggs, int64_t n, int shift, int64_t* cursor,
                             int64_t* out_keys, double* out_vals,
-                            int num_parts, hipStream_t stream) {
+                            int num_parts, int32_t* out_pos,
+                            hipStream_t stream) {
   int64_t blocks = (n + SCATTER_CHUNK - 1) / SCATTER_CHUNK;
   if (blocks > MAX_GRID) blocks = MAX_GRID;
   if (blocks < 1) blocks = 1;
   hipLaunchKernelGGL(gb_part_scatter_kernel, dim3((int)blocks), dim3(BLOCK),
                      0, stream, keys, vals, n_aggs, n, shift, cursor,
-                     out_keys, out_vals, num_parts);
+                     out_keys, out_vals, num_parts, out_pos);
 }
 
 void launch_gb_aggregate_part(const int64_t* part_keys,
@@ -3222,4 +3225,32 @@ void launch_joinoa_probe(const int64_t* pkeys, int64_t np,
                      out_mask, mask_neg);
 }
 
+}  // extern "C"
+
+
+// ------------------------------------------------------------------ //
+// shuffle-layout reuse (Spark shuffle-reuse analog)                   //
+//                                                                     //
+// Re-aggregating the SAME key column with fresh values (iterative     //
+// pipelines) skips hist + key scatter entirely: the recorded per-row  //
+// spill position places the new values, and the cached partitioned    //
+// keys feed phase 3 unchanged.                                        //
+// ------------------------------------------------------------------ //
+
+__global__ __launch_bounds__(BLOCK) void scatter_by_pos_kernel(
+    const double* __restrict__ vals, const int32_t* __restrict__ pos,
+    int64_t n, double* __restrict__ out_vals) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    out_vals[pos[i]] = vals[i];
+  }
+}
+
+extern "C" {
+void launch_scatter_by_pos(const double* vals, const int32_t* pos, int64_t n,
+                           double* out_vals, hipStream_t stream) {
+  hipLaunchKernelGGL(scatter_by_pos_kernel, dim3(grid_for(n, 2)), dim3(BLOCK),
+                     0, stream, vals, pos, n, out_vals);
+}
 }  // extern "C"

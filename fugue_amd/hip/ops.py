@@ -207,6 +207,13 @@ def rand_buckets(n: int, num_buckets: int, seed: Optional[int], device) -> torch
 _KEY_STATS_MEMO: "Dict[Tuple[int, ...], Tuple[Any, Any]]" = {}
 _KEY_STATS_MEMO_CAP = 128
 
+# shuffle-layout reuse (Spark shuffle-reuse analog): for a repeated key
+# tensor, the partition layout (spill keys + per-row positions) is
+# recorded once and fresh values replay through it.  Two entries ~1GB
+# each at 125M rows — bound the cache tightly.
+_LAYOUT_MEMO: "Dict[Tuple[int, ...], Dict[str, Any]]" = {}
+_LAYOUT_MEMO_CAP = 2
+
 
 def _key_stats_device(
     datas: "List[torch.Tensor]",
@@ -550,16 +557,51 @@ def groupby_aggregate(
                 narrow = 1 if key_lo >= 0 and key_hi < (1 << 31) else 0
             else:
                 narrow = -1  # speculative: overflow flag checked below
-        tkeys, gaggs, gcount, ovf = ext.gb_aggregate_partitioned(
-            packed, vals, ops, num_parts, tsize, sc_chunk, ag_chunk, nt,
-            narrow
+        slots = {512: 4096, 1024: 2048, 2048: 1024}.get(num_parts, 4096)
+        reuse = (
+            _os.environ.get("FUGUE_GB_LAYOUT_REUSE", "1") != "0"
+            and len(aggs) == 1
+            and n < (1 << 31)
+            and num_parts in (512, 1024, 2048)
         )
-        if narrow == -1 and int(ovf.item()) != 0:
-            # a key fell outside [0, 2^31): redo exactly with wide keys
-            tkeys, gaggs, gcount, ovf = ext.gb_aggregate_partitioned(
-                packed, vals, ops, num_parts, tsize, sc_chunk, ag_chunk,
-                nt, 0
+        lkey = (packed.data_ptr(), n, num_parts, tsize, narrow)
+        ent = _LAYOUT_MEMO.get(lkey) if reuse else None
+        if ent is not None and ent["ref"]() is not None:
+            # replay: fresh values through the recorded layout — phase 1
+            # (hist) and the key scatter are skipped entirely
+            tkeys, gaggs, gcount = ext.gb_aggregate_replay(
+                ent["pkeys"], ent["pos"], vals, ops, tsize, ag_chunk, nt,
+                slots,
             )
+        elif reuse:
+            tkeys, gaggs, gcount, ovf, pkeys_l, pos = (
+                ext.gb_aggregate_partitioned(
+                    packed, vals, ops, num_parts, tsize, sc_chunk,
+                    ag_chunk, nt, 0, True
+                )
+            )
+            if narrow == 1:
+                pkeys_l = pkeys_l.to(torch.int32)
+            if len(_LAYOUT_MEMO) >= _LAYOUT_MEMO_CAP:
+                _LAYOUT_MEMO.pop(next(iter(_LAYOUT_MEMO)))
+            _LAYOUT_MEMO[lkey] = dict(
+                ref=weakref.ref(packed), pkeys=pkeys_l, pos=pos
+            )
+        else:
+            tkeys, gaggs, gcount, ovf, _pk, _pos = (
+                ext.gb_aggregate_partitioned(
+                    packed, vals, ops, num_parts, tsize, sc_chunk,
+                    ag_chunk, nt, narrow, False
+                )
+            )
+            if narrow == -1 and int(ovf.item()) != 0:
+                # a key fell outside [0, 2^31): redo exactly, wide keys
+                tkeys, gaggs, gcount, ovf, _pk, _pos = (
+                    ext.gb_aggregate_partitioned(
+                        packed, vals, ops, num_parts, tsize, sc_chunk,
+                        ag_chunk, nt, 0, False
+                    )
+                )
     else:
         use_lds = expected_groups <= 100_000 and sum_count_only
         tkeys, gaggs, gcount = ext.gb_aggregate(

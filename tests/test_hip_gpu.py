@@ -872,3 +872,67 @@ def test_open_addressed_join_gpu(engine, monkeypatch):
         drop=True
     )
     pd.testing.assert_frame_equal(got, exp, check_dtype=False)
+
+
+def test_groupby_layout_reuse_gpu(engine):
+    """Shuffle-layout reuse: repeated aggregation over the SAME key
+    tensor with fresh values replays the recorded layout and stays
+    exact; a new key tensor re-records."""
+    import pyarrow as pa
+    import torch
+
+    from fugue_amd.hip.frame import DeviceColumn, HipDataFrame
+    from fugue_amd.schema import Schema
+    from fugue_amd.hip import ops as dops
+
+    n = 2_000_000
+    g = torch.Generator(device="cuda").manual_seed(5)
+    keys = torch.randint(0, 300_000, (n,), dtype=torch.int64, device="cuda",
+                         generator=g)
+    kcol = DeviceColumn(keys, None, pa.int64())
+    exp_counts = None
+    for rep in range(3):
+        vals = torch.rand(n, dtype=torch.float64, device="cuda",
+                          generator=g)
+        df = HipDataFrame.from_columns(
+            {"k": kcol, "v": DeviceColumn(vals, None, pa.float64())},
+            Schema("k:long,v:double"), "cuda",
+        )
+        out_keys, out_aggs, out_count, meta = dops.groupby_aggregate(
+            df, ["k"], [("v", dops.AGG_SUM, "s")]
+        )
+        # exact check against torch scatter-add reference
+        ref = torch.zeros(300_000, dtype=torch.float64, device="cuda")
+        ref.scatter_add_(0, keys, vals)
+        perm = torch.argsort(out_keys)
+        got_k = out_keys.index_select(0, perm)
+        got_s = out_aggs["s"].index_select(0, perm)
+        nz = (ref != 0).nonzero(as_tuple=True)[0]
+        assert torch.equal(got_k, nz)
+        assert torch.allclose(got_s, ref.index_select(0, nz), rtol=1e-12)
+        if exp_counts is None:
+            exp_counts = out_count.index_select(0, perm)
+        else:
+            assert torch.equal(out_count.index_select(0, perm), exp_counts)
+    assert len(dops._LAYOUT_MEMO) >= 1
+    # a different key tensor must not hit the stale layout
+    keys2 = torch.randint(0, 300_000, (n,), dtype=torch.int64,
+                          device="cuda", generator=g)
+    vals2 = torch.rand(n, dtype=torch.float64, device="cuda", generator=g)
+    df2 = HipDataFrame.from_columns(
+        {"k": DeviceColumn(keys2, None, pa.int64()),
+         "v": DeviceColumn(vals2, None, pa.float64())},
+        Schema("k:long,v:double"), "cuda",
+    )
+    ok2, oa2, oc2, _ = dops.groupby_aggregate(
+        df2, ["k"], [("v", dops.AGG_SUM, "s")]
+    )
+    ref2 = torch.zeros(300_000, dtype=torch.float64, device="cuda")
+    ref2.scatter_add_(0, keys2, vals2)
+    perm2 = torch.argsort(ok2)
+    nz2 = (ref2 != 0).nonzero(as_tuple=True)[0]
+    assert torch.equal(ok2.index_select(0, perm2), nz2)
+    assert torch.allclose(
+        oa2["s"].index_select(0, perm2), ref2.index_select(0, nz2),
+        rtol=1e-12,
+    )
