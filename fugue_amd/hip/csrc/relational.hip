@@ -2500,23 +2500,6 @@ static void launch_pack_dispatch(int ncols, dim3 g, dim3 b,
 
 extern "C" {
 
-void launch_minmax_cols(const void** data, const bool** valid,
-                        const int* dwidth, int ncols, int64_t n,
-                        uint64_t* out, hipStream_t stream) {
-  PackCols pc;
-  memset(&pc, 0, sizeof(pc));
-  for (int c = 0; c < ncols; ++c) {
-    pc.data[c] = data[c];
-    pc.valid[c] = valid[c];
-    pc.dwidth[c] = dwidth[c];
-  }
-  hipLaunchKernelGGL(minmax_init_kernel, dim3(1), dim3(BLOCK), 0, stream,
-                     out, ncols);
-  int mg = grid_for(n, 8);
-  if (mg > 4096) mg = 4096;
-  launch_minmax_dispatch(ncols, dim3(mg), dim3(BLOCK), stream, pc, n, out);
-}
-
 void launch_pack_cols(const void** data, const bool** valid,
                       const int* dwidth, const int64_t* mins,
                       const int* shifts, int ncols, int64_t n, int64_t* out,
