@@ -8,6 +8,7 @@ SURVEY.md §2.3.
 """
 from typing import Any, Dict, List, Optional, Sequence, Tuple
 
+import os
 import weakref
 
 import numpy as np
@@ -213,6 +214,11 @@ _KEY_STATS_MEMO_CAP = 128
 # each at 125M rows — bound the cache tightly.
 _LAYOUT_MEMO: "Dict[Tuple[int, ...], Dict[str, Any]]" = {}
 _LAYOUT_MEMO_CAP = 2
+# key tensors seen once (candidates): recording uses the slower simple
+# scatter, so it only happens on the SECOND sight of the same keys —
+# one-shot group-bys never pay the record cost
+_LAYOUT_SEEN: "Dict[Tuple[int, ...], Any]" = {}
+_LAYOUT_SEEN_CAP = 16
 
 
 def _key_stats_device(
@@ -268,6 +274,19 @@ def _key_stats_device(
         _KEY_STATS_MEMO.pop(next(iter(_KEY_STATS_MEMO)))
     _KEY_STATS_MEMO[key] = (tuple(weakref.ref(t) for t in datas), value)
     return value
+
+
+def _layout_second_sight(lkey: tuple, packed: "torch.Tensor") -> bool:
+    """True when this key tensor was already aggregated once (same
+    address, still alive) — the signal that recording its layout will
+    pay off."""
+    hit = _LAYOUT_SEEN.get(lkey)
+    if hit is not None and hit() is not None:
+        return True
+    if len(_LAYOUT_SEEN) >= _LAYOUT_SEEN_CAP:
+        _LAYOUT_SEEN.pop(next(iter(_LAYOUT_SEEN)))
+    _LAYOUT_SEEN[lkey] = weakref.ref(packed)
+    return False
 
 
 def _device_packable(key_cols: "Sequence[DeviceColumn]") -> bool:
@@ -529,7 +548,10 @@ def groupby_aggregate(
         # rows so each partition's groups fit the per-workgroup LDS table)
         # single-agg path with moderate cardinality uses the LDS
         # write-staged scatter (512 parts, 4096-slot phase-3 table)
-        if len(aggs) <= 1 and expected_groups <= 1_500_000:
+        staged_max = int(
+            os.environ.get("FUGUE_GB_STAGED_MAX", "1200000")
+        )
+        if len(aggs) <= 1 and expected_groups <= staged_max:
             import os as _os0
 
             # staged-variant partition count: 512 (8-deep staging,
@@ -573,6 +595,22 @@ def groupby_aggregate(
                 ent["pkeys"], ent["pos"], vals, ops, tsize, ag_chunk, nt,
                 slots,
             )
+        elif reuse and not _layout_second_sight(lkey, packed):
+            # first sight of these keys: normal path (recording uses the
+            # slower simple scatter; only repeat keys justify it)
+            tkeys, gaggs, gcount, ovf, _pk, _pos = (
+                ext.gb_aggregate_partitioned(
+                    packed, vals, ops, num_parts, tsize, sc_chunk,
+                    ag_chunk, nt, narrow, False
+                )
+            )
+            if narrow == -1 and int(ovf.item()) != 0:
+                tkeys, gaggs, gcount, ovf, _pk, _pos = (
+                    ext.gb_aggregate_partitioned(
+                        packed, vals, ops, num_parts, tsize, sc_chunk,
+                        ag_chunk, nt, 0, False
+                    )
+                )
         elif reuse:
             tkeys, gaggs, gcount, ovf, pkeys_l, pos = (
                 ext.gb_aggregate_partitioned(
