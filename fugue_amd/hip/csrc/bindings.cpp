@@ -229,7 +229,7 @@ std::vector<at::Tensor> gb_aggregate(at::Tensor keys, at::Tensor vals,
 std::vector<at::Tensor> gb_aggregate_partitioned(
     at::Tensor keys, at::Tensor vals, at::Tensor ops, int64_t num_parts,
     int64_t tsize, int64_t scatter_chunk, int64_t agg_chunk, int64_t nt,
-    int64_t narrow, bool record_layout) {
+    int64_t narrow, bool record_layout, bool force_simple) {
   check_gpu(keys, "keys");
   check_gpu(vals, "vals");
   TORCH_CHECK((tsize & (tsize - 1)) == 0, "tsize must be a power of 2");
@@ -254,7 +254,7 @@ std::vector<at::Tensor> gb_aggregate_partitioned(
   // row's spill position in hand); replay restores the staged-path
   // spill format (narrow keys done by the caller)
   record_layout = record_layout && n_aggs == 1 && n < (int64_t(1) << 31);
-  if (record_layout) staged = false;
+  if (record_layout || force_simple) staged = false;
   // phase-3 LDS table sized for ~0.5 load at the partition granularity:
   // 512 parts -> 4096 slots, 1024 -> 2048, 2048 -> 1024
   int slots = num_parts == 1024 ? 2048 : (num_parts == 2048 ? 1024 : 4096);

@@ -580,6 +580,12 @@ def groupby_aggregate(
             else:
                 narrow = -1  # speculative: overflow flag checked below
         slots = {512: 4096, 1024: 2048, 2048: 1024}.get(num_parts, 4096)
+        # the LDS write-staged scatter pays off only on big spills; at
+        # q3-like sizes the simple per-chunk-reservation variant is
+        # faster (same-box A/B, profiles/NOTES.md r02)
+        force_simple = n < int(
+            _os.environ.get("FUGUE_GB_STAGED_MIN_ROWS", "48000000")
+        )
         reuse = (
             _os.environ.get("FUGUE_GB_LAYOUT_REUSE", "1") != "0"
             and len(aggs) == 1
@@ -601,21 +607,21 @@ def groupby_aggregate(
             tkeys, gaggs, gcount, ovf, _pk, _pos = (
                 ext.gb_aggregate_partitioned(
                     packed, vals, ops, num_parts, tsize, sc_chunk,
-                    ag_chunk, nt, narrow, False
+                    ag_chunk, nt, narrow, False, force_simple
                 )
             )
             if narrow == -1 and int(ovf.item()) != 0:
                 tkeys, gaggs, gcount, ovf, _pk, _pos = (
                     ext.gb_aggregate_partitioned(
                         packed, vals, ops, num_parts, tsize, sc_chunk,
-                        ag_chunk, nt, 0, False
+                        ag_chunk, nt, 0, False, force_simple
                     )
                 )
         elif reuse:
             tkeys, gaggs, gcount, ovf, pkeys_l, pos = (
                 ext.gb_aggregate_partitioned(
                     packed, vals, ops, num_parts, tsize, sc_chunk,
-                    ag_chunk, nt, 0, True
+                    ag_chunk, nt, 0, True, False
                 )
             )
             if narrow == 1:
@@ -629,7 +635,7 @@ def groupby_aggregate(
             tkeys, gaggs, gcount, ovf, _pk, _pos = (
                 ext.gb_aggregate_partitioned(
                     packed, vals, ops, num_parts, tsize, sc_chunk,
-                    ag_chunk, nt, narrow, False
+                    ag_chunk, nt, narrow, False, force_simple
                 )
             )
             if narrow == -1 and int(ovf.item()) != 0:
@@ -637,7 +643,7 @@ def groupby_aggregate(
                 tkeys, gaggs, gcount, ovf, _pk, _pos = (
                     ext.gb_aggregate_partitioned(
                         packed, vals, ops, num_parts, tsize, sc_chunk,
-                        ag_chunk, nt, 0, False
+                        ag_chunk, nt, 0, False, force_simple
                     )
                 )
     else:
