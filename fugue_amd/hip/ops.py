@@ -548,31 +548,25 @@ def groupby_aggregate(
         # rows so each partition's groups fit the per-workgroup LDS table)
         # single-agg path with moderate cardinality uses the LDS
         # write-staged scatter (512 parts, 4096-slot phase-3 table)
-        staged_max = int(
-            os.environ.get("FUGUE_GB_STAGED_MAX", "1200000")
-        )
+        staged_max = int(os.environ.get("FUGUE_GB_STAGED_MAX", "1200000"))
         if len(aggs) <= 1 and expected_groups <= staged_max:
-            import os as _os0
-
             # staged-variant partition count: 512 (8-deep staging,
             # 4096-slot phase-3), 1024 (4-deep, 2048-slot) or 2048
             # (2-deep, 1024-slot) — smaller tables run phase 3 at
             # higher occupancy; A/B via FUGUE_GB_PARTS
-            num_parts = int(_os0.environ.get("FUGUE_GB_PARTS", "1024"))
+            num_parts = int(os.environ.get("FUGUE_GB_PARTS", "1024"))
             if num_parts not in (512, 1024, 2048):
                 num_parts = 512
         else:
             num_parts = min(4096, _next_pow2(max(16, expected_groups // 512)))
-        import os as _os
-
-        sc_chunk = int(_os.environ.get("FUGUE_GB_SCATTER_CHUNK", "0"))
-        ag_chunk = int(_os.environ.get("FUGUE_GB_AGG_CHUNK", "0"))
-        nt = int(_os.environ.get("FUGUE_GB_NT", "0"))
+        sc_chunk = int(os.environ.get("FUGUE_GB_SCATTER_CHUNK", "0"))
+        ag_chunk = int(os.environ.get("FUGUE_GB_AGG_CHUNK", "0"))
+        nt = int(os.environ.get("FUGUE_GB_NT", "0"))
         # int32 intermediate keys shrink the partitioned spill 16B->12B
         # per row (and its MALL footprint) when the packed key range is
         # known (pack meta) or measured to fit 31 bits
         narrow = 0
-        if int(_os.environ.get("FUGUE_GB_NARROW", "1")):
+        if int(os.environ.get("FUGUE_GB_NARROW", "1")):
             if meta is not None:
                 narrow = 1 if sum(meta["widths"]) <= 31 else 0
             elif key_lo is not None:
@@ -584,10 +578,10 @@ def groupby_aggregate(
         # q3-like sizes the simple per-chunk-reservation variant is
         # faster (same-box A/B, profiles/NOTES.md r02)
         force_simple = n < int(
-            _os.environ.get("FUGUE_GB_STAGED_MIN_ROWS", "48000000")
+            os.environ.get("FUGUE_GB_STAGED_MIN_ROWS", "48000000")
         )
         reuse = (
-            _os.environ.get("FUGUE_GB_LAYOUT_REUSE", "1") != "0"
+            os.environ.get("FUGUE_GB_LAYOUT_REUSE", "1") != "0"
             and len(aggs) == 1
             and n < (1 << 31)
             and num_parts in (512, 1024, 2048)
