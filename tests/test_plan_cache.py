@@ -92,3 +92,32 @@ def test_immutable_binding_change_invalidates():
     r1 = fa.fugue_sql("SELECT * FROM df WHERE a = 'x'", df=df)
     r2 = fa.fugue_sql("SELECT * FROM df WHERE a = 'y'", df=df)
     assert r1.a.tolist() == ["x"] and r2.a.tolist() == ["y"]
+
+
+def test_plan_cache_concurrent_calls():
+    """Two threads issuing the same query concurrently: one replays or
+    both build (lock contention falls back to a fresh build) — results
+    stay correct either way."""
+    import threading
+
+    df = pd.DataFrame(dict(a=[0, 1, 0, 1], b=[1, 2, 3, 4]))
+    out = [None, None]
+    errs = []
+
+    def run(i):
+        try:
+            for _ in range(10):
+                out[i] = fa.fugue_sql(
+                    "SELECT a, SUM(b) AS s FROM df GROUP BY a", df=df
+                )
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    ts = [threading.Thread(target=run, args=(i,)) for i in range(2)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert not errs
+    for r in out:
+        assert sorted(r.s.tolist()) == [4, 6]
