@@ -33,18 +33,18 @@ def main():
         torch.cuda.synchronize()
         return (time.perf_counter() - t0) / k * 1000
 
-    CFG = ("1", "48000000", "400000000")  # always-staged / mid / always-simple
+    CFG = ("512", "1024", "2048")
     for v in CFG:
-        os.environ["FUGUE_GB_STAGED_MIN_ROWS"] = v
+        os.environ["FUGUE_GB_PARTS"] = v
         timed(3)
     res = {v: [] for v in CFG}
     for _ in range(5):
         for v in CFG:
-            os.environ["FUGUE_GB_STAGED_MIN_ROWS"] = v
+            os.environ["FUGUE_GB_PARTS"] = v
             res[v].append(timed(8))
     for v in CFG:
         med = sorted(res[v])[len(res[v]) // 2]
-        print(f"q3 STAGED_MIN_ROWS={v}: median {med:.3f} ms "
+        print(f"q3 simple-path PARTS={v}: median {med:.3f} ms "
               f"all={[round(x,2) for x in res[v]]}", flush=True)
 
 
